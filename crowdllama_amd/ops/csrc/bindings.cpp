@@ -1,0 +1,103 @@
+// pybind11 module `crowdllama_amd.ops._core` — Python face of the
+// MI355X-native engine.
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "engine.h"
+
+namespace py = pybind11;
+using namespace cla;
+
+namespace cla {
+void launch_gemv_test(const void*, const void*, const float*, const float*,
+                      float*, int, int, int, int, int, size_t, size_t);
+}
+
+PYBIND11_MODULE(_core, m) {
+    m.doc() = "crowdllama-amd MI355X (gfx950) HIP inference engine";
+
+    m.def("device_count", [] {
+        int n = 0;
+        if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+        return n;
+    });
+    m.def("device_props", [](int dev) {
+        hipDeviceProp_t p;
+        HIP_CHECK(hipGetDeviceProperties(&p, dev));
+        py::dict d;
+        d["name"] = std::string(p.gcnArchName);
+        d["total_mem_gb"] = (double)p.totalGlobalMem / (1024.0 * 1024 * 1024);
+        d["multiprocessors"] = p.multiProcessorCount;
+        d["warp_size"] = p.warpSize;
+        return d;
+    });
+
+    py::class_<EngineConfig>(m, "EngineConfig")
+        .def(py::init<>())
+        .def_readwrite("batch", &EngineConfig::batch)
+        .def_readwrite("max_seq", &EngineConfig::max_seq)
+        .def_readwrite("page_size", &EngineConfig::page_size)
+        .def_readwrite("gen_cap", &EngineConfig::gen_cap)
+        .def_readwrite("device", &EngineConfig::device)
+        .def_readwrite("use_graph", &EngineConfig::use_graph);
+
+    py::class_<Engine>(m, "Engine")
+        .def(py::init<const std::string&, const EngineConfig&>(),
+             py::arg("gguf_path"), py::arg("config"))
+        .def("reset", &Engine::reset)
+        .def("prefill",
+             [](Engine& e, py::array_t<int32_t, py::array::c_style> ids) {
+                 if (ids.ndim() != 2)
+                     throw std::runtime_error("ids must be [batch][len]");
+                 const int len = (int)ids.shape(1);
+                 std::vector<int32_t> v(ids.data(), ids.data() + ids.size());
+                 py::gil_scoped_release rel;
+                 e.prefill(v, len);
+             })
+        .def("decode", [](Engine& e, int n) {
+                 py::gil_scoped_release rel;
+                 e.decode(n);
+             })
+        .def("gen_tokens", &Engine::gen_tokens)
+        .def("logits",
+             [](Engine& e, int slot) {
+                 auto v = e.logits(slot);
+                 return py::array_t<float>((py::ssize_t)v.size(), v.data());
+             })
+        .def("set_cur_token", &Engine::set_cur_token)
+        .def("cur_token", &Engine::cur_token)
+        .def("n_past", &Engine::n_past)
+        .def("vram_bytes", &Engine::vram_bytes)
+        .def("last_decode_ms", &Engine::last_decode_ms)
+        .def_property_readonly("meta", [](Engine& e) {
+            const auto& mm = e.meta();
+            py::dict d;
+            d["name"] = mm.name;
+            d["vocab"] = mm.vocab;
+            d["hidden"] = mm.hidden;
+            d["layers"] = mm.layers;
+            d["heads"] = mm.heads;
+            d["kv_heads"] = mm.kv_heads;
+            d["ffn"] = mm.ffn;
+            d["head_dim"] = mm.head_dim;
+            d["rope_theta"] = mm.rope_theta;
+            d["rms_eps"] = mm.rms_eps;
+            return d;
+        });
+
+    // ---- raw kernel entry points for numerics tests (tests/test_gpu_kernels.py)
+    m.def("test_gemv", [](py::array_t<uint8_t> qs, py::array_t<uint8_t> hdr,
+                          py::array_t<float> x, int dtype, int N, int K,
+                          int pre, py::array_t<float> gw) {
+        // x: [B][K] (or [B][2K] for PRE_SILU); returns y [B][N]
+        const int B = (int)x.shape(0);
+        const int xk = (int)x.shape(1);
+        py::array_t<float> y({B, N});
+        launch_gemv_test(qs.data(), hdr.data(), x.data(),
+                         gw.size() ? gw.data() : nullptr, y.mutable_data(),
+                         dtype, N, K, B, pre, qs.nbytes(), hdr.nbytes());
+        (void)xk;
+        return y;
+    });
+}

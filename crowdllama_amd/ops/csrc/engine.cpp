@@ -1,0 +1,483 @@
+#include "engine.h"
+
+#include <algorithm>
+#include <cmath>
+#include <cstring>
+#include <functional>
+#include <thread>
+
+#include "gguf.h"
+
+namespace cla {
+
+// kernel launchers (kernels.hip)
+void launch_gemv(const WTensor&, int pre, const float* xin, const float* gw,
+                 const float* res, float* y, int B, int ldy, float eps,
+                 hipStream_t);
+void launch_embed(const WTensor&, const int32_t* ids, float* x, int B,
+                  hipStream_t);
+void launch_rope_append(float* qkv, const float* inv_freq,
+                        const int32_t* page_table, uint16_t* kv_pool,
+                        const int32_t* n_past, int B, int NH, int NKV, int D,
+                        int page_size, int max_pages, int64_t page_stride,
+                        hipStream_t);
+void launch_attn_decode(const float* qkv, const int32_t* page_table,
+                        const uint16_t* kv_pool, const int32_t* n_past,
+                        float* part_o, float* part_ml, int B, int NH, int NKV,
+                        int D, int S, int page_size, int max_pages,
+                        int64_t page_stride, float scale, hipStream_t);
+void launch_attn_combine(const float* part_o, const float* part_ml,
+                         float* attn_out, int B, int NH, int S, int D,
+                         hipStream_t);
+void launch_argmax(const float* logits, float* pval, int32_t* pidx,
+                   int32_t* cur_ids, int32_t* n_past, int32_t* gen_tokens,
+                   int32_t* gen_count, int B, int V, int gen_cap, hipStream_t);
+
+namespace {
+
+enum { PRE_NONE = 0, PRE_RMS = 1, PRE_SILU = 2 };
+
+void parallel_for(int64_t n, const std::function<void(int64_t, int64_t)>& fn) {
+    const int nt = std::min<int64_t>(std::thread::hardware_concurrency(), 16);
+    if (nt <= 1 || n < 4) { fn(0, n); return; }
+    std::vector<std::thread> ts;
+    const int64_t per = (n + nt - 1) / nt;
+    for (int t = 0; t < nt; t++) {
+        const int64_t lo = t * per, hi = std::min<int64_t>(lo + per, n);
+        if (lo >= hi) break;
+        ts.emplace_back([&fn, lo, hi] { fn(lo, hi); });
+    }
+    for (auto& t : ts) t.join();
+}
+
+DT device_dtype(int32_t ggml_type) {
+    switch (ggml_type) {
+        case 0: return DT::F32;
+        case 1: return DT::F16;
+        case 30: return DT::BF16;
+        case 8: return DT::DQ8;
+        case 12: return DT::DQ4K;
+        case 14: return DT::DQ6K;
+        default:
+            throw std::runtime_error("unsupported ggml type " +
+                                     std::to_string(ggml_type));
+    }
+}
+
+// Repack disk-format rows into split qs/hdr device format (see common.h DT).
+void repack(int32_t ggml_type, const uint8_t* src, int64_t rows, int64_t k,
+            uint8_t* qs_out, uint8_t* hdr_out) {
+    const int64_t src_rb = ggml_row_bytes(ggml_type, k);
+    switch (ggml_type) {
+        case 0: case 1: case 30:  // float passthrough
+            std::memcpy(qs_out, src, src_rb * rows);
+            return;
+        case 12: {  // Q4_K: disk block 144B = {hdr 16B, qs 128B}
+            const int64_t nsb = k / 256;
+            parallel_for(rows, [&](int64_t lo, int64_t hi) {
+                for (int64_t r = lo; r < hi; r++) {
+                    const uint8_t* s = src + r * src_rb;
+                    uint8_t* q = qs_out + r * nsb * 128;
+                    uint8_t* h = hdr_out + r * nsb * 16;
+                    for (int64_t b = 0; b < nsb; b++) {
+                        std::memcpy(h + b * 16, s + b * 144, 16);
+                        std::memcpy(q + b * 128, s + b * 144 + 16, 128);
+                    }
+                }
+            });
+            return;
+        }
+        case 14: {  // Q6_K: disk 210B = {ql 128, qh 64, sc 16, d 2}
+            const int64_t nsb = k / 256;
+            parallel_for(rows, [&](int64_t lo, int64_t hi) {
+                for (int64_t r = lo; r < hi; r++) {
+                    const uint8_t* s = src + r * src_rb;
+                    int8_t* q = reinterpret_cast<int8_t*>(qs_out + r * k);
+                    uint8_t* h = hdr_out + r * nsb * 32;
+                    for (int64_t b = 0; b < nsb; b++) {
+                        const uint8_t* ql = s + b * 210;
+                        const uint8_t* qh = ql + 128;
+                        const uint8_t* sc = ql + 192;
+                        // hdr: {f16 d, u8 pad[2], i8 sc[16], pad}
+                        std::memcpy(h + b * 32, ql + 208, 2);
+                        std::memcpy(h + b * 32 + 4, sc, 16);
+                        int8_t* qb = q + b * 256;
+                        for (int half = 0; half < 2; half++) {
+                            const uint8_t* qlh = ql + half * 64;
+                            const uint8_t* qhh = qh + half * 32;
+                            for (int l = 0; l < 32; l++) {
+                                const int q1 = (qlh[l] & 0xF) | (((qhh[l] >> 0) & 3) << 4);
+                                const int q2 = (qlh[l + 32] & 0xF) | (((qhh[l] >> 2) & 3) << 4);
+                                const int q3 = (qlh[l] >> 4) | (((qhh[l] >> 4) & 3) << 4);
+                                const int q4 = (qlh[l + 32] >> 4) | (((qhh[l] >> 6) & 3) << 4);
+                                int8_t* o = qb + half * 128;
+                                o[l] = (int8_t)(q1 - 32);
+                                o[l + 32] = (int8_t)(q2 - 32);
+                                o[l + 64] = (int8_t)(q3 - 32);
+                                o[l + 96] = (int8_t)(q4 - 32);
+                            }
+                        }
+                    }
+                }
+            });
+            return;
+        }
+        case 8: {  // Q8_0: disk 34B = {f16 d, i8 qs[32]}
+            const int64_t nb = k / 32;
+            parallel_for(rows, [&](int64_t lo, int64_t hi) {
+                for (int64_t r = lo; r < hi; r++) {
+                    const uint8_t* s = src + r * src_rb;
+                    uint8_t* q = qs_out + r * k;
+                    uint8_t* h = hdr_out + r * nb * 2;
+                    for (int64_t b = 0; b < nb; b++) {
+                        std::memcpy(h + b * 2, s + b * 34, 2);
+                        std::memcpy(q + b * 32, s + b * 34 + 2, 32);
+                    }
+                }
+            });
+            return;
+        }
+        default:
+            throw std::runtime_error("repack: unsupported type");
+    }
+}
+
+}  // namespace
+
+Engine::Engine(const std::string& gguf_path, const EngineConfig& cfg)
+    : cfg_(cfg) {
+    HIP_CHECK(hipSetDevice(cfg_.device));
+    HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+    GGUFFile gf(gguf_path);
+    meta_.name = gf.meta_str("general.name", "unknown");
+    meta_.vocab = (int)gf.meta_int("llama.vocab_size", 0);
+    meta_.hidden = (int)gf.meta_int("llama.embedding_length", 0);
+    meta_.layers = (int)gf.meta_int("llama.block_count", 0);
+    meta_.heads = (int)gf.meta_int("llama.attention.head_count", 0);
+    meta_.kv_heads = (int)gf.meta_int("llama.attention.head_count_kv", meta_.heads);
+    meta_.ffn = (int)gf.meta_int("llama.feed_forward_length", 0);
+    meta_.max_ctx = (int)gf.meta_int("llama.context_length", 4096);
+    meta_.rope_theta = (float)gf.meta_float("llama.rope.freq_base", 10000.0);
+    meta_.rms_eps = (float)gf.meta_float("llama.attention.layer_norm_rms_epsilon", 1e-5);
+    meta_.head_dim = meta_.hidden / meta_.heads;
+    if (meta_.vocab == 0) {
+        auto& t = gf.tensor("token_embd.weight");
+        meta_.vocab = (int)t.shape[0];
+    }
+    load_weights(gf);
+    alloc_state();
+    reset();
+}
+
+Engine::~Engine() {
+    if (graph_exec_) hipGraphExecDestroy(graph_exec_);
+    for (void* p : allocs_) hipFree(p);
+    if (stream_) hipStreamDestroy(stream_);
+}
+
+WTensor Engine::upload_tensor(const GGUFFile& gf, const std::string& name) {
+    const GGUFTensor& t = gf.tensor(name);
+    const int64_t k = t.shape.back();
+    int64_t rows = 1;
+    for (size_t i = 0; i + 1 < t.shape.size(); i++) rows *= t.shape[i];
+    WTensor w;
+    w.dtype = device_dtype(t.ggml_type);
+    w.n = rows;
+    w.k = k;
+    const int64_t qs_bytes = dqs_row_bytes(w.dtype, k) * rows;
+    const int64_t hdr_bytes = dhdr_row_bytes(w.dtype, k) * rows;
+    std::vector<uint8_t> h_qs(qs_bytes), h_hdr(hdr_bytes ? hdr_bytes : 1);
+    repack(t.ggml_type, t.data, rows, k, h_qs.data(), h_hdr.data());
+    void* d_qs = nullptr;
+    HIP_CHECK(hipMalloc(&d_qs, qs_bytes));
+    HIP_CHECK(hipMemcpy(d_qs, h_qs.data(), qs_bytes, hipMemcpyHostToDevice));
+    allocs_.push_back(d_qs);
+    w.qs = d_qs;
+    vram_bytes_ += qs_bytes;
+    if (hdr_bytes) {
+        void* d_hdr = nullptr;
+        HIP_CHECK(hipMalloc(&d_hdr, hdr_bytes));
+        HIP_CHECK(hipMemcpy(d_hdr, h_hdr.data(), hdr_bytes, hipMemcpyHostToDevice));
+        allocs_.push_back(d_hdr);
+        w.hdr = d_hdr;
+        vram_bytes_ += hdr_bytes;
+    }
+    return w;
+}
+
+const float* Engine::upload_norm(const GGUFFile& gf, const std::string& name) {
+    const GGUFTensor& t = gf.tensor(name);
+    if (t.ggml_type != 0)
+        throw std::runtime_error(name + ": norm weights must be f32");
+    void* d = nullptr;
+    HIP_CHECK(hipMalloc(&d, t.nbytes));
+    HIP_CHECK(hipMemcpy(d, t.data, t.nbytes, hipMemcpyHostToDevice));
+    allocs_.push_back(d);
+    vram_bytes_ += t.nbytes;
+    return reinterpret_cast<const float*>(d);
+}
+
+Proj Engine::load_proj(const GGUFFile& gf, const std::vector<std::string>& names) {
+    Proj p;
+    // Merge row-blocks with identical dtype+k into one device tensor so a
+    // single GEMV launch covers them (q/k/v and gate/up stacking).
+    bool mergeable = names.size() > 1;
+    int32_t t0 = gf.tensor(names[0]).ggml_type;
+    int64_t k0 = gf.tensor(names[0]).shape.back();
+    for (auto& nm : names) {
+        const auto& t = gf.tensor(nm);
+        if (t.ggml_type != t0 || t.shape.back() != k0) mergeable = false;
+    }
+    if (mergeable) {
+        // concatenate repacked rows
+        DT dt = device_dtype(t0);
+        int64_t rows = 0;
+        for (auto& nm : names) rows += gf.tensor(nm).shape[0];
+        const int64_t qs_rb = dqs_row_bytes(dt, k0);
+        const int64_t hdr_rb = dhdr_row_bytes(dt, k0);
+        std::vector<uint8_t> h_qs(qs_rb * rows), h_hdr(hdr_rb * rows + 1);
+        int64_t roff = 0;
+        for (auto& nm : names) {
+            const auto& t = gf.tensor(nm);
+            repack(t0, t.data, t.shape[0], k0, h_qs.data() + roff * qs_rb,
+                   h_hdr.data() + roff * hdr_rb);
+            roff += t.shape[0];
+        }
+        WTensor w;
+        w.dtype = dt; w.n = rows; w.k = k0;
+        void* d_qs = nullptr;
+        HIP_CHECK(hipMalloc(&d_qs, h_qs.size()));
+        HIP_CHECK(hipMemcpy(d_qs, h_qs.data(), h_qs.size(), hipMemcpyHostToDevice));
+        allocs_.push_back(d_qs);
+        w.qs = d_qs;
+        vram_bytes_ += h_qs.size();
+        if (hdr_rb) {
+            void* d_hdr = nullptr;
+            HIP_CHECK(hipMalloc(&d_hdr, hdr_rb * rows));
+            HIP_CHECK(hipMemcpy(d_hdr, h_hdr.data(), hdr_rb * rows,
+                                hipMemcpyHostToDevice));
+            allocs_.push_back(d_hdr);
+            w.hdr = d_hdr;
+            vram_bytes_ += hdr_rb * rows;
+        }
+        p.parts.push_back({w, 0});
+        p.n_total = rows;
+        return p;
+    }
+    int64_t roff = 0;
+    for (auto& nm : names) {
+        WTensor w = upload_tensor(gf, nm);
+        p.parts.push_back({w, roff});
+        roff += w.n;
+    }
+    p.n_total = roff;
+    return p;
+}
+
+void Engine::load_weights(const GGUFFile& gf) {
+    embed_ = upload_tensor(gf, "token_embd.weight");
+    out_norm_ = upload_norm(gf, "output_norm.weight");
+    head_ = gf.has_tensor("output.weight") ? upload_tensor(gf, "output.weight")
+                                           : embed_;  // tied embeddings
+    layers_.resize(meta_.layers);
+    for (int i = 0; i < meta_.layers; i++) {
+        const std::string p = "blk." + std::to_string(i) + ".";
+        Layer& L = layers_[i];
+        L.attn_norm = upload_norm(gf, p + "attn_norm.weight");
+        L.ffn_norm = upload_norm(gf, p + "ffn_norm.weight");
+        L.qkv = load_proj(gf, {p + "attn_q.weight", p + "attn_k.weight",
+                               p + "attn_v.weight"});
+        L.o = load_proj(gf, {p + "attn_output.weight"});
+        L.gate_up = load_proj(gf, {p + "ffn_gate.weight", p + "ffn_up.weight"});
+        L.down = load_proj(gf, {p + "ffn_down.weight"});
+    }
+}
+
+void Engine::alloc_state() {
+    const int B = cfg_.batch, H = meta_.hidden, V = meta_.vocab;
+    const int NH = meta_.heads, NKV = meta_.kv_heads, D = meta_.head_dim;
+    const int F = meta_.ffn;
+    attn_splits_ = std::max(1, std::min(64, 512 / std::max(1, B * NKV)));
+    max_pages_ = (cfg_.max_seq + cfg_.page_size - 1) / cfg_.page_size;
+    page_stride_ = (int64_t)NKV * 2 * cfg_.page_size * D;
+
+    auto dalloc = [&](size_t bytes) {
+        void* p = nullptr;
+        HIP_CHECK(hipMalloc(&p, bytes));
+        HIP_CHECK(hipMemset(p, 0, bytes));
+        allocs_.push_back(p);
+        vram_bytes_ += bytes;
+        return p;
+    };
+    x_ = (float*)dalloc((size_t)B * H * 4);
+    qkv_ = (float*)dalloc((size_t)B * (NH + 2 * NKV) * D * 4);
+    attn_out_ = (float*)dalloc((size_t)B * NH * D * 4);
+    gu_ = (float*)dalloc((size_t)B * 2 * F * 4);
+    logits_ = (float*)dalloc((size_t)B * V * 4);
+    part_o_ = (float*)dalloc((size_t)B * NH * attn_splits_ * D * 4);
+    part_ml_ = (float*)dalloc((size_t)B * NH * attn_splits_ * 2 * 4);
+    amax_val_ = (float*)dalloc((size_t)B * 64 * 4);
+    amax_idx_ = (int32_t*)dalloc((size_t)B * 64 * 4);
+    cur_ids_ = (int32_t*)dalloc((size_t)B * 4);
+    n_past_ = (int32_t*)dalloc((size_t)B * 4);
+    gen_tokens_ = (int32_t*)dalloc((size_t)B * cfg_.gen_cap * 4);
+    gen_count_ = (int32_t*)dalloc((size_t)B * 4);
+    page_table_ = (int32_t*)dalloc((size_t)B * max_pages_ * 4);
+    kv_pool_ = (uint16_t*)dalloc((size_t)B * max_pages_ * page_stride_ * 2);
+    // rope frequency table
+    std::vector<float> invf(D / 2);
+    for (int i = 0; i < D / 2; i++)
+        invf[i] = std::pow(meta_.rope_theta, -2.0f * i / (float)D);
+    inv_freq_ = (float*)dalloc(invf.size() * 4);
+    HIP_CHECK(hipMemcpy(inv_freq_, invf.data(), invf.size() * 4,
+                        hipMemcpyHostToDevice));
+    // static page table: slot b owns pages [b*max_pages_, (b+1)*max_pages_)
+    std::vector<int32_t> pt((size_t)B * max_pages_);
+    for (int b = 0; b < B; b++)
+        for (int pg = 0; pg < max_pages_; pg++)
+            pt[(size_t)b * max_pages_ + pg] = b * max_pages_ + pg;
+    HIP_CHECK(hipMemcpy(page_table_, pt.data(), pt.size() * 4,
+                        hipMemcpyHostToDevice));
+}
+
+void Engine::reset() {
+    const int B = cfg_.batch;
+    HIP_CHECK(hipMemset(n_past_, 0, B * 4));
+    HIP_CHECK(hipMemset(gen_count_, 0, B * 4));
+    HIP_CHECK(hipMemset(cur_ids_, 0, B * 4));
+    HIP_CHECK(hipDeviceSynchronize());
+}
+
+void Engine::step(hipStream_t s) {
+    const int B = cfg_.batch;
+    const int NH = meta_.heads, NKV = meta_.kv_heads, D = meta_.head_dim;
+    const float scale = 1.0f / std::sqrt((float)D);
+    const float eps = meta_.rms_eps;
+
+    launch_embed(embed_, cur_ids_, x_, B, s);
+    for (auto& L : layers_) {
+        for (auto& pt : L.qkv.parts)
+            launch_gemv(pt.w, PRE_RMS, x_, L.attn_norm, nullptr,
+                        qkv_ + pt.row_off, B, (int)L.qkv.n_total, eps, s);
+        launch_rope_append(qkv_, inv_freq_, page_table_, kv_pool_, n_past_,
+                           B, NH, NKV, D, cfg_.page_size, max_pages_,
+                           page_stride_, s);
+        launch_attn_decode(qkv_, page_table_, kv_pool_, n_past_, part_o_,
+                           part_ml_, B, NH, NKV, D, attn_splits_,
+                           cfg_.page_size, max_pages_, page_stride_, scale, s);
+        launch_attn_combine(part_o_, part_ml_, attn_out_, B, NH, attn_splits_,
+                            D, s);
+        for (auto& pt : L.o.parts)
+            launch_gemv(pt.w, PRE_NONE, attn_out_, nullptr, x_ + pt.row_off,
+                        x_ + pt.row_off, B, (int)L.o.n_total, eps, s);
+        for (auto& pt : L.gate_up.parts)
+            launch_gemv(pt.w, PRE_RMS, x_, L.ffn_norm, nullptr,
+                        gu_ + pt.row_off, B, (int)L.gate_up.n_total, eps, s);
+        for (auto& pt : L.down.parts)
+            launch_gemv(pt.w, PRE_SILU, gu_, nullptr, x_ + pt.row_off,
+                        x_ + pt.row_off, B, (int)L.down.n_total, eps, s);
+    }
+    launch_gemv(head_, PRE_RMS, x_, out_norm_, nullptr, logits_, B,
+                meta_.vocab, eps, s);
+    launch_argmax(logits_, amax_val_, amax_idx_, cur_ids_, n_past_,
+                  gen_tokens_, gen_count_, B, meta_.vocab, cfg_.gen_cap, s);
+}
+
+void Engine::ensure_graph() {
+    if (graph_exec_ || !cfg_.use_graph) return;
+    // warm-up eager step is NOT run here; capture directly (all state is
+    // device-resident, shapes static).
+    hipGraph_t graph = nullptr;
+    HIP_CHECK(hipStreamBeginCapture(stream_, hipStreamCaptureModeThreadLocal));
+    step(stream_);
+    HIP_CHECK(hipStreamEndCapture(stream_, &graph));
+    HIP_CHECK(hipGraphInstantiate(&graph_exec_, graph, nullptr, nullptr, 0));
+    HIP_CHECK(hipGraphDestroy(graph));
+}
+
+void Engine::prefill(const std::vector<int32_t>& ids, int len) {
+    const int B = cfg_.batch;
+    if ((int)ids.size() != B * len)
+        throw std::runtime_error("prefill: ids must be batch*len");
+    ensure_graph();
+    std::vector<int32_t> col(B);
+    for (int t = 0; t < len; t++) {
+        for (int b = 0; b < B; b++) col[b] = ids[(size_t)b * len + t];
+        HIP_CHECK(hipMemcpyAsync(cur_ids_, col.data(), B * 4,
+                                 hipMemcpyHostToDevice, stream_));
+        if (graph_exec_) {
+            HIP_CHECK(hipGraphLaunch(graph_exec_, stream_));
+        } else {
+            step(stream_);
+        }
+    }
+    HIP_CHECK(hipStreamSynchronize(stream_));
+    // after the last prompt token, cur_ids holds the first generated token;
+    // reset the gen ring so it lands at index 0.
+    std::vector<int32_t> first(B);
+    HIP_CHECK(hipMemcpy(first.data(), cur_ids_, B * 4, hipMemcpyDeviceToHost));
+    std::vector<int32_t> zero(B, 1);
+    HIP_CHECK(hipMemcpy(gen_count_, zero.data(), B * 4, hipMemcpyHostToDevice));
+    for (int b = 0; b < B; b++)
+        HIP_CHECK(hipMemcpy(gen_tokens_ + (size_t)b * cfg_.gen_cap, &first[b],
+                            4, hipMemcpyHostToDevice));
+}
+
+void Engine::decode(int n_steps) {
+    ensure_graph();
+    hipEvent_t e0, e1;
+    HIP_CHECK(hipEventCreate(&e0));
+    HIP_CHECK(hipEventCreate(&e1));
+    HIP_CHECK(hipEventRecord(e0, stream_));
+    for (int i = 0; i < n_steps; i++) {
+        if (graph_exec_) {
+            HIP_CHECK(hipGraphLaunch(graph_exec_, stream_));
+        } else {
+            step(stream_);
+        }
+    }
+    HIP_CHECK(hipEventRecord(e1, stream_));
+    HIP_CHECK(hipStreamSynchronize(stream_));
+    float ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+    last_decode_ms_ = ms;
+    hipEventDestroy(e0);
+    hipEventDestroy(e1);
+}
+
+std::vector<int32_t> Engine::gen_tokens(int slot) {
+    int32_t count = 0;
+    HIP_CHECK(hipMemcpy(&count, gen_count_ + slot, 4, hipMemcpyDeviceToHost));
+    count = std::min(count, cfg_.gen_cap);
+    std::vector<int32_t> out(count);
+    if (count)
+        HIP_CHECK(hipMemcpy(out.data(), gen_tokens_ + (size_t)slot * cfg_.gen_cap,
+                            count * 4, hipMemcpyDeviceToHost));
+    return out;
+}
+
+std::vector<float> Engine::logits(int slot) {
+    std::vector<float> out(meta_.vocab);
+    HIP_CHECK(hipMemcpy(out.data(), logits_ + (size_t)slot * meta_.vocab,
+                        meta_.vocab * 4, hipMemcpyDeviceToHost));
+    return out;
+}
+
+void Engine::set_cur_token(int slot, int32_t id) {
+    HIP_CHECK(hipMemcpy(cur_ids_ + slot, &id, 4, hipMemcpyHostToDevice));
+}
+
+int32_t Engine::cur_token(int slot) {
+    int32_t id = 0;
+    HIP_CHECK(hipMemcpy(&id, cur_ids_ + slot, 4, hipMemcpyDeviceToHost));
+    return id;
+}
+
+std::vector<int32_t> Engine::n_past() {
+    std::vector<int32_t> out(cfg_.batch);
+    HIP_CHECK(hipMemcpy(out.data(), n_past_, cfg_.batch * 4,
+                        hipMemcpyDeviceToHost));
+    return out;
+}
+
+}  // namespace cla

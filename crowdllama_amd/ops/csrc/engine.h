@@ -1,0 +1,125 @@
+// MI355X-native decode engine: owns device weights, paged KV cache, and the
+// per-token decode step (hipGraph-captured). This is the worker-side compute
+// that replaces the reference's shell-out to Ollama (reference
+// pkg/crowdllama/api.go:45-160 -> Ollama -> llama.cpp; here it is first-party).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <memory>
+#include <string>
+#include <vector>
+
+#include "common.h"
+
+namespace cla {
+
+class GGUFFile;
+
+struct EngineConfig {
+    int batch = 1;          // decode slots
+    int max_seq = 4096;     // max positions per slot
+    int page_size = 64;     // KV page tokens
+    int gen_cap = 2048;     // generated-token ring per slot
+    int device = 0;
+    bool use_graph = true;
+};
+
+struct ModelMeta {
+    std::string name;
+    int vocab = 0, hidden = 0, layers = 0, heads = 0, kv_heads = 0;
+    int ffn = 0, head_dim = 0, max_ctx = 0;
+    float rope_theta = 10000.f, rms_eps = 1e-5f;
+};
+
+// One logical projection: a list of row-blocks (merged when dtypes match).
+struct Proj {
+    struct Part {
+        WTensor w;
+        int64_t row_off = 0;  // offset into the output vector
+    };
+    std::vector<Part> parts;
+    int64_t n_total = 0;
+};
+
+struct Layer {
+    const float* attn_norm = nullptr;
+    const float* ffn_norm = nullptr;
+    Proj qkv, o, gate_up, down;
+};
+
+class Engine {
+public:
+    Engine(const std::string& gguf_path, const EngineConfig& cfg);
+    ~Engine();
+    Engine(const Engine&) = delete;
+
+    const ModelMeta& meta() const { return meta_; }
+    const EngineConfig& config() const { return cfg_; }
+
+    // Reset all slots (empty KV).
+    void reset();
+    // Feed prompt tokens (same length for all slots in this call) through
+    // the decode path; afterwards each slot's first generated token is in
+    // gen_tokens[slot][0]. ids is [batch][len] row-major.
+    void prefill(const std::vector<int32_t>& ids, int len);
+    // Run n decode steps back-to-back (graph replays; one sync at the end).
+    void decode(int n_steps);
+    // Fetch generated tokens for a slot (gen_count entries).
+    std::vector<int32_t> gen_tokens(int slot);
+    // Copy a slot's current logits to host (for host-side sampling).
+    std::vector<float> logits(int slot);
+    // Override a slot's current input token (host-side sampling path).
+    void set_cur_token(int slot, int32_t id);
+    int32_t cur_token(int slot);
+    std::vector<int32_t> n_past();
+
+    size_t vram_bytes() const { return vram_bytes_; }
+    double last_decode_ms() const { return last_decode_ms_; }
+
+private:
+    void load_weights(const GGUFFile& gf);
+    Proj load_proj(const GGUFFile& gf, const std::vector<std::string>& names);
+    WTensor upload_tensor(const GGUFFile& gf, const std::string& name);
+    const float* upload_norm(const GGUFFile& gf, const std::string& name);
+    void alloc_state();
+    void step(hipStream_t stream);
+    void ensure_graph();
+
+    EngineConfig cfg_;
+    ModelMeta meta_;
+    hipStream_t stream_ = nullptr;
+
+    // weights
+    std::vector<Layer> layers_;
+    WTensor embed_, head_;
+    const float* out_norm_ = nullptr;
+    std::vector<void*> allocs_;
+    size_t vram_bytes_ = 0;
+
+    // state buffers (device)
+    float* x_ = nullptr;        // [B][h]
+    float* qkv_ = nullptr;      // [B][(NH+2KV)*D]
+    float* attn_out_ = nullptr; // [B][NH*D]
+    float* gu_ = nullptr;       // [B][2F]
+    float* logits_ = nullptr;   // [B][V]
+    float* part_o_ = nullptr;   // [B][NH][S][D]
+    float* part_ml_ = nullptr;  // [B][NH][S][2]
+    float* amax_val_ = nullptr; // [B][64]
+    int32_t* amax_idx_ = nullptr;
+    float* inv_freq_ = nullptr; // [D/2]
+    int32_t* cur_ids_ = nullptr;   // [B]
+    int32_t* n_past_ = nullptr;    // [B]
+    int32_t* gen_tokens_ = nullptr;  // [B][gen_cap]
+    int32_t* gen_count_ = nullptr;   // [B]
+    int32_t* page_table_ = nullptr;  // [B][max_pages]
+    uint16_t* kv_pool_ = nullptr;
+    int attn_splits_ = 16;
+    int max_pages_ = 0;
+    int64_t page_stride_ = 0;
+
+    hipGraphExec_t graph_exec_ = nullptr;
+    double last_decode_ms_ = 0.0;
+};
+
+}  // namespace cla

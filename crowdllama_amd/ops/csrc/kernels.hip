@@ -1,0 +1,612 @@
+// HIP/CDNA4 (gfx950) kernels for the crowdllama-amd decode engine.
+//
+// MI355X-native replacements for the GGML compute the reference delegates to
+// Ollama/llama.cpp: fused dequant-GEMV (Q4_K/Q6_K/Q8_0),
+// RMSNorm + SwiGLU fused into GEMV activation staging, RoPE + paged-KV
+// append, split-KV online-softmax decode attention, embedding gather,
+// on-device greedy sampling.
+//
+// Design notes (see /opt/skills guides):
+// - wave64 everywhere; block = 256 threads = 4 waves.
+// - GEMV: one wave per output row sweeping the row's quant payload in 16-B
+//   chunks (coalesced dwordx4), activations staged once per workgroup into
+//   LDS as f32; block headers ride the same-address broadcast path through
+//   L1/L2 (no cross-lane shuffles needed).
+// - Weights are repacked at upload into split qs/hdr arrays (common.h DT).
+// - Attention: each 16-lane quarter-wave owns a full 128-dim online-softmax
+//   accumulator; 4 positions in flight per wave; split-KV partials combined
+//   by a second small kernel.
+
+#include "common.h"
+
+namespace cla {
+
+// ------------------------------------------------------------------ utils
+
+__device__ __forceinline__ float f16_bits_to_f32(uint32_t h) {
+    __half_raw r;
+    r.x = static_cast<uint16_t>(h);
+    return __half2float(*reinterpret_cast<__half*>(&r));
+}
+
+__device__ __forceinline__ float bf16_bits_to_f32(uint32_t h) {
+    union { uint32_t u; float f; } v;
+    v.u = h << 16;
+    return v.f;
+}
+
+__device__ __forceinline__ uint16_t f32_to_bf16_bits(float f) {
+    union { uint32_t u; float f; } v;
+    v.f = f;
+    uint32_t r = (v.u + 0x7FFF + ((v.u >> 16) & 1)) >> 16;
+    return static_cast<uint16_t>(r);
+}
+
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+    return v;
+}
+
+enum Pre : int { PRE_NONE = 0, PRE_RMS = 1, PRE_SILU = 2 };
+
+// Weights per 16-B qs chunk for each device dtype.
+template <DT W> struct ChunkTraits;
+template <> struct ChunkTraits<DT::DQ4K> { static constexpr int W_PER_CHUNK = 32; };
+template <> struct ChunkTraits<DT::DQ6K> { static constexpr int W_PER_CHUNK = 16; };
+template <> struct ChunkTraits<DT::DQ8>  { static constexpr int W_PER_CHUNK = 16; };
+template <> struct ChunkTraits<DT::BF16> { static constexpr int W_PER_CHUNK = 8; };
+template <> struct ChunkTraits<DT::F16>  { static constexpr int W_PER_CHUNK = 8; };
+template <> struct ChunkTraits<DT::F32>  { static constexpr int W_PER_CHUNK = 4; };
+
+// Decode one 16-B chunk of a row into `w[]` weights starting at column
+// `k0` (and for DQ4K a second run of 16 at k0+32).
+template <DT W>
+__device__ __forceinline__ void decode_chunk(
+    const uint8_t* __restrict__ qs_row, const uint8_t* __restrict__ hdr_row,
+    int c, float* __restrict__ w, int* k0) {
+    const uint4 qv = reinterpret_cast<const uint4*>(qs_row)[c];
+    const uint32_t dw[4] = {qv.x, qv.y, qv.z, qv.w};
+    if constexpr (W == DT::DQ4K) {
+        const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
+        const uint4 hd = reinterpret_cast<const uint4*>(hdr_row)[sb];
+        const float d = f16_bits_to_f32(hd.x & 0xFFFF);
+        const float dmin = f16_bits_to_f32(hd.x >> 16);
+        const uint32_t sbytes[3] = {hd.y, hd.z, hd.w};
+        auto sbyte = [&](int i) -> uint32_t {
+            return (sbytes[i >> 2] >> ((i & 3) * 8)) & 0xFF;
+        };
+        const int j0 = 2 * q, j1 = 2 * q + 1;
+        uint32_t sc0, mn0, sc1, mn1;
+        if (q < 2) {
+            sc0 = sbyte(j0) & 63; mn0 = sbyte(j0 + 4) & 63;
+            sc1 = sbyte(j1) & 63; mn1 = sbyte(j1 + 4) & 63;
+        } else {
+            sc0 = (sbyte(j0 + 4) & 0xF) | ((sbyte(j0 - 4) >> 6) << 4);
+            mn0 = (sbyte(j0 + 4) >> 4) | ((sbyte(j0) >> 6) << 4);
+            sc1 = (sbyte(j1 + 4) & 0xF) | ((sbyte(j1 - 4) >> 6) << 4);
+            mn1 = (sbyte(j1 + 4) >> 4) | ((sbyte(j1) >> 6) << 4);
+        }
+        const float dl = d * (float)sc0, ml = dmin * (float)mn0;
+        const float dh = d * (float)sc1, mh = dmin * (float)mn1;
+        *k0 = sb * 256 + q * 64 + h * 16;
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            const uint32_t lo = dw[j] & 0x0F0F0F0Fu;
+            const uint32_t hi = (dw[j] >> 4) & 0x0F0F0F0Fu;
+            #pragma unroll
+            for (int t = 0; t < 4; t++) {
+                w[j * 4 + t] = dl * (float)((lo >> (8 * t)) & 0xFF) - ml;
+                w[16 + j * 4 + t] = dh * (float)((hi >> (8 * t)) & 0xFF) - mh;
+            }
+        }
+    } else if constexpr (W == DT::DQ6K) {
+        const int sb = c >> 4, s16 = c & 15;
+        const uint8_t* hb = hdr_row + sb * 32;
+        const float d = f16_bits_to_f32(*reinterpret_cast<const uint16_t*>(hb));
+        const float sc = d * (float)(reinterpret_cast<const int8_t*>(hb)[4 + s16]);
+        *k0 = c * 16;
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            #pragma unroll
+            for (int t = 0; t < 4; t++) {
+                const int8_t q8 = (int8_t)((dw[j] >> (8 * t)) & 0xFF);
+                w[j * 4 + t] = sc * (float)q8;
+            }
+        }
+    } else if constexpr (W == DT::DQ8) {
+        const float d = f16_bits_to_f32(
+            reinterpret_cast<const uint16_t*>(hdr_row)[c >> 1]);
+        *k0 = c * 16;
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            #pragma unroll
+            for (int t = 0; t < 4; t++) {
+                const int8_t q8 = (int8_t)((dw[j] >> (8 * t)) & 0xFF);
+                w[j * 4 + t] = d * (float)q8;
+            }
+        }
+    } else if constexpr (W == DT::BF16) {
+        *k0 = c * 8;
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            w[j * 2] = bf16_bits_to_f32(dw[j] & 0xFFFF);
+            w[j * 2 + 1] = bf16_bits_to_f32(dw[j] >> 16);
+        }
+    } else if constexpr (W == DT::F16) {
+        *k0 = c * 8;
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            w[j * 2] = f16_bits_to_f32(dw[j] & 0xFFFF);
+            w[j * 2 + 1] = f16_bits_to_f32(dw[j] >> 16);
+        }
+    } else {  // F32
+        *k0 = c * 4;
+        #pragma unroll
+        for (int j = 0; j < 4; j++)
+            w[j] = __uint_as_float(dw[j]);
+    }
+}
+
+// ------------------------------------------------------------------ GEMV
+
+// y[b][r] = sum_k W[r][k] * x[b][k] (+ res[b][r])
+// PRE_RMS:  x = rmsnorm(xin) * gw   (gamma)
+// PRE_SILU: x[k] = silu(xin[b][k]) * xin[b][K+k]   (xin is [B][2K])
+// One wave per row; x staged in LDS as f32 with per-16 partial sums.
+template <DT W, int P>
+__global__ __launch_bounds__(256) void k_gemv(
+    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
+    const float* __restrict__ xin, const float* __restrict__ gw,
+    const float* __restrict__ res, float* __restrict__ y,
+    int N, int K, int B, int ldy, float eps) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* x_lds = reinterpret_cast<float*>(smem);           // [B][K]
+    float* red = x_lds + (size_t)B * K;                      // [256]
+
+    const int tid = threadIdx.x;
+    // ---- stage activations ----
+    for (int b = 0; b < B; b++) {
+        float ss = 0.f;
+        if constexpr (P == PRE_SILU) {
+            const float* gub = xin + (size_t)b * 2 * K;
+            for (int k = tid; k < K; k += 256) {
+                const float g = gub[k], u = gub[K + k];
+                x_lds[(size_t)b * K + k] = (g / (1.f + __expf(-g))) * u;
+            }
+        } else {
+            const float* xb = xin + (size_t)b * K;
+            for (int k = tid; k < K; k += 256) {
+                const float v = xb[k];
+                x_lds[(size_t)b * K + k] = v;
+                if constexpr (P == PRE_RMS) ss += v * v;
+            }
+        }
+        if constexpr (P == PRE_RMS) {
+            red[tid] = ss;
+            __syncthreads();
+            #pragma unroll
+            for (int off = 128; off > 0; off >>= 1) {
+                if (tid < off) red[tid] += red[tid + off];
+                __syncthreads();
+            }
+            const float inv = rsqrtf(red[0] / (float)K + eps);
+            __syncthreads();
+            for (int k = tid; k < K; k += 256)
+                x_lds[(size_t)b * K + k] *= inv * gw[k];
+        }
+    }
+    __syncthreads();
+
+    // ---- per-wave row sweep ----
+    const int wave = tid >> 6, lane = tid & 63;
+    const int r = blockIdx.x * 4 + wave;
+    if (r >= N) return;
+    const int64_t qs_rb = dqs_row_bytes(W, K);
+    const int64_t hdr_rb = dhdr_row_bytes(W, K);
+    const uint8_t* qs_row = qs + (int64_t)r * qs_rb;
+    const uint8_t* hdr_row = hdr + (int64_t)r * hdr_rb;
+    const int n_chunks = (int)(qs_rb / 16);
+    constexpr int WPC = ChunkTraits<W>::W_PER_CHUNK;
+
+    float acc[2] = {0.f, 0.f};  // GEMV path supports B <= 2
+    for (int c = lane; c < n_chunks; c += 64) {
+        float w[WPC];
+        int k0;
+        decode_chunk<W>(qs_row, hdr_row, c, w, &k0);
+        #pragma unroll 2
+        for (int b = 0; b < B; b++) {
+            const float* xb = x_lds + (size_t)b * K;
+            float s = 0.f;
+            if constexpr (W == DT::DQ4K) {
+                // lo run at k0, hi run at k0+32; fold the min term via xsum16
+                #pragma unroll
+                for (int t = 0; t < 16; t++) s += w[t] * xb[k0 + t];
+                #pragma unroll
+                for (int t = 0; t < 16; t++) s += w[16 + t] * xb[k0 + 32 + t];
+            } else {
+                #pragma unroll
+                for (int t = 0; t < WPC; t++) s += w[t] * xb[k0 + t];
+            }
+            acc[b] += s;
+        }
+    }
+    #pragma unroll 2
+    for (int b = 0; b < B; b++) {
+        float v = wave_reduce_sum(acc[b]);
+        if (lane == 0) {
+            const int64_t idx = (int64_t)b * ldy + r;
+            y[idx] = v + (res ? res[idx] : 0.f);
+        }
+    }
+}
+
+// ------------------------------------------------------------- embedding
+
+template <DT W>
+__global__ __launch_bounds__(256) void k_embed(
+    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
+    const int32_t* __restrict__ ids, float* __restrict__ x, int K) {
+    const int b = blockIdx.x;
+    const int row = ids[b];
+    const int64_t qs_rb = dqs_row_bytes(W, K);
+    const uint8_t* qs_row = qs + (int64_t)row * qs_rb;
+    const uint8_t* hdr_row = hdr + (int64_t)row * dhdr_row_bytes(W, K);
+    const int n_chunks = (int)(qs_rb / 16);
+    constexpr int WPC = ChunkTraits<W>::W_PER_CHUNK;
+    float* xb = x + (size_t)b * K;
+    for (int c = threadIdx.x; c < n_chunks; c += 256) {
+        float w[WPC];
+        int k0;
+        decode_chunk<W>(qs_row, hdr_row, c, w, &k0);
+        if constexpr (W == DT::DQ4K) {
+            #pragma unroll
+            for (int t = 0; t < 16; t++) xb[k0 + t] = w[t];
+            #pragma unroll
+            for (int t = 0; t < 16; t++) xb[k0 + 32 + t] = w[16 + t];
+        } else {
+            #pragma unroll
+            for (int t = 0; t < WPC; t++) xb[k0 + t] = w[t];
+        }
+    }
+}
+
+// -------------------------------------------------------- RoPE + append
+
+// grid (KVH, B); block 256. Applies NORM-style RoPE (adjacent pairs) to the
+// G q-heads and 1 k-head of this kv-group in the qkv buffer (in place, f32),
+// then appends the roped k and raw v to the paged bf16 KV cache at position
+// n_past[b].
+__global__ __launch_bounds__(256) void k_rope_append(
+    float* __restrict__ qkv, const float* __restrict__ inv_freq,
+    const int32_t* __restrict__ page_table, uint16_t* __restrict__ kv_pool,
+    const int32_t* __restrict__ n_past,
+    int NH, int NKV, int D, int G, int page_size, int max_pages,
+    int64_t page_stride) {
+    const int kvh = blockIdx.x;
+    const int b = blockIdx.y;
+    const int pos = n_past[b];
+    float* qkv_b = qkv + (size_t)b * (NH + 2 * NKV) * D;
+    float* kh = qkv_b + (size_t)(NH + kvh) * D;
+    const float* vh = qkv_b + (size_t)(NH + NKV + kvh) * D;
+    const int half = D / 2;
+    // rope: G q-heads + 1 k-head
+    for (int idx = threadIdx.x; idx < (G + 1) * half; idx += 256) {
+        const int hsel = idx / half, i = idx % half;
+        float* p = (hsel < G) ? (qkv_b + (size_t)(kvh * G + hsel) * D) : kh;
+        const float ang = (float)pos * inv_freq[i];
+        float s, c;
+        __sincosf(ang, &s, &c);
+        const float x0 = p[2 * i], x1 = p[2 * i + 1];
+        p[2 * i] = x0 * c - x1 * s;
+        p[2 * i + 1] = x0 * s + x1 * c;
+    }
+    __syncthreads();
+    // append k (roped) and v to the cache page
+    const int page = page_table[(size_t)b * max_pages + pos / page_size];
+    const int off = pos % page_size;
+    // pool layout: [page][kvh][2][page_size][D] bf16
+    uint16_t* kdst = kv_pool + (int64_t)page * page_stride
+                     + ((int64_t)kvh * 2 + 0) * page_size * D + (int64_t)off * D;
+    uint16_t* vdst = kv_pool + (int64_t)page * page_stride
+                     + ((int64_t)kvh * 2 + 1) * page_size * D + (int64_t)off * D;
+    for (int d = threadIdx.x; d < D; d += 256) {
+        kdst[d] = f32_to_bf16_bits(kh[d]);
+        vdst[d] = f32_to_bf16_bits(vh[d]);
+    }
+}
+
+// ------------------------------------------------- decode attention (GQA)
+
+// grid (S, KVH, B); block 256 = 4 waves; each 16-lane quarter owns a full
+// (m, l, o[128]) online-softmax accumulator over its position subsequence,
+// for each of the G query heads of this kv group.
+template <int G>
+__global__ __launch_bounds__(256) void k_attn_decode(
+    const float* __restrict__ qkv, const int32_t* __restrict__ page_table,
+    const uint16_t* __restrict__ kv_pool, const int32_t* __restrict__ n_past,
+    float* __restrict__ part_o,   // [B][NH][S][D]
+    float* __restrict__ part_ml,  // [B][NH][S][2]
+    int NH, int NKV, int D, int S, int page_size, int max_pages,
+    int64_t page_stride, float scale) {
+    const int s = blockIdx.x, kvh = blockIdx.y, b = blockIdx.z;
+    const int len = n_past[b] + 1;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6, lane = tid & 63;
+    const int quarter = lane >> 4, qlane = lane & 15;
+    const int sub = wave * 4 + quarter;       // 0..15 position subsequence
+    const int d0 = qlane * 8;                 // this lane's 8-dim slice
+
+    int chunk = (len + S - 1) / S;
+    chunk = (chunk + 3) & ~3;
+    const int start = s * chunk;
+    const int end = min(start + chunk, len);
+
+    const float* qkv_b = qkv + (size_t)b * (NH + 2 * NKV) * D;
+    // q fragments for the G heads of this group
+    float qf[G][8];
+    #pragma unroll
+    for (int g = 0; g < G; g++) {
+        const float* qh = qkv_b + (size_t)(kvh * G + g) * D + d0;
+        #pragma unroll
+        for (int j = 0; j < 8; j++) qf[g][j] = qh[j] * scale;
+    }
+
+    float m[G], l[G], o[G][8];
+    #pragma unroll
+    for (int g = 0; g < G; g++) {
+        m[g] = -1e30f; l[g] = 0.f;
+        #pragma unroll
+        for (int j = 0; j < 8; j++) o[g][j] = 0.f;
+    }
+
+    for (int p = start + sub; p < end; p += 16) {
+        const int page = page_table[(size_t)b * max_pages + p / page_size];
+        const uint16_t* kp = kv_pool + (int64_t)page * page_stride
+                             + ((int64_t)kvh * 2 + 0) * page_size * D
+                             + (int64_t)(p % page_size) * D + d0;
+        const uint4 kq = *reinterpret_cast<const uint4*>(kp);
+        float kf[8];
+        kf[0] = bf16_bits_to_f32(kq.x & 0xFFFF); kf[1] = bf16_bits_to_f32(kq.x >> 16);
+        kf[2] = bf16_bits_to_f32(kq.y & 0xFFFF); kf[3] = bf16_bits_to_f32(kq.y >> 16);
+        kf[4] = bf16_bits_to_f32(kq.z & 0xFFFF); kf[5] = bf16_bits_to_f32(kq.z >> 16);
+        kf[6] = bf16_bits_to_f32(kq.w & 0xFFFF); kf[7] = bf16_bits_to_f32(kq.w >> 16);
+        float sc[G];
+        #pragma unroll
+        for (int g = 0; g < G; g++) {
+            float d = 0.f;
+            #pragma unroll
+            for (int j = 0; j < 8; j++) d += qf[g][j] * kf[j];
+            // reduce over the 16-lane quarter
+            #pragma unroll
+            for (int off = 1; off < 16; off <<= 1) d += __shfl_xor(d, off, 64);
+            sc[g] = d;
+        }
+        const uint16_t* vp = kv_pool + (int64_t)page * page_stride
+                             + ((int64_t)kvh * 2 + 1) * page_size * D
+                             + (int64_t)(p % page_size) * D + d0;
+        const uint4 vq = *reinterpret_cast<const uint4*>(vp);
+        float vf[8];
+        vf[0] = bf16_bits_to_f32(vq.x & 0xFFFF); vf[1] = bf16_bits_to_f32(vq.x >> 16);
+        vf[2] = bf16_bits_to_f32(vq.y & 0xFFFF); vf[3] = bf16_bits_to_f32(vq.y >> 16);
+        vf[4] = bf16_bits_to_f32(vq.z & 0xFFFF); vf[5] = bf16_bits_to_f32(vq.z >> 16);
+        vf[6] = bf16_bits_to_f32(vq.w & 0xFFFF); vf[7] = bf16_bits_to_f32(vq.w >> 16);
+        #pragma unroll
+        for (int g = 0; g < G; g++) {
+            const float mn = fmaxf(m[g], sc[g]);
+            const float alpha = __expf(m[g] - mn);
+            const float w = __expf(sc[g] - mn);
+            l[g] = l[g] * alpha + w;
+            #pragma unroll
+            for (int j = 0; j < 8; j++) o[g][j] = o[g][j] * alpha + w * vf[j];
+            m[g] = mn;
+        }
+    }
+
+    // combine the 16 sub-accumulators through LDS
+    __shared__ __attribute__((aligned(16))) float sm[16 * G];
+    __shared__ __attribute__((aligned(16))) float sl[16 * G];
+    __shared__ __attribute__((aligned(16))) float so[16 * G * 128];
+    #pragma unroll
+    for (int g = 0; g < G; g++) {
+        if (qlane == 0) {
+            sm[sub * G + g] = m[g];
+            sl[sub * G + g] = l[g];
+        }
+        #pragma unroll
+        for (int j = 0; j < 8; j++) so[(sub * G + g) * 128 + d0 + j] = o[g][j];
+    }
+    __syncthreads();
+    // threads 0..(G*128-1): one output dim each (G<=8 -> <=1024; block 256 loops)
+    for (int idx = tid; idx < G * 128; idx += 256) {
+        const int g = idx / 128, d = idx % 128;
+        float mstar = -1e30f;
+        #pragma unroll
+        for (int t = 0; t < 16; t++) mstar = fmaxf(mstar, sm[t * G + g]);
+        float lsum = 0.f, osum = 0.f;
+        #pragma unroll
+        for (int t = 0; t < 16; t++) {
+            const float e = __expf(sm[t * G + g] - mstar);
+            lsum += e * sl[t * G + g];
+            osum += e * so[(t * G + g) * 128 + d];
+        }
+        const int head = kvh * G + g;
+        part_o[(((size_t)b * NH + head) * S + s) * D + d] = osum;
+        if (d == 0) {
+            part_ml[(((size_t)b * NH + head) * S + s) * 2 + 0] = mstar;
+            part_ml[(((size_t)b * NH + head) * S + s) * 2 + 1] = lsum;
+        }
+    }
+}
+
+// grid (B*NH); block 128. Combine S split-KV partials into attn_out.
+__global__ __launch_bounds__(128) void k_attn_combine(
+    const float* __restrict__ part_o, const float* __restrict__ part_ml,
+    float* __restrict__ attn_out, int NH, int S, int D) {
+    const int bh = blockIdx.x;  // b * NH + head
+    const int d = threadIdx.x;
+    float mstar = -1e30f;
+    for (int s = 0; s < S; s++)
+        mstar = fmaxf(mstar, part_ml[((size_t)bh * S + s) * 2]);
+    float denom = 0.f, osum = 0.f;
+    for (int s = 0; s < S; s++) {
+        const float e = __expf(part_ml[((size_t)bh * S + s) * 2] - mstar);
+        denom += e * part_ml[((size_t)bh * S + s) * 2 + 1];
+        osum += e * part_o[((size_t)bh * S + s) * D + d];
+    }
+    attn_out[(size_t)bh * D + d] = osum / denom;
+}
+
+// ------------------------------------------------------ argmax sampling
+
+// stage 1: grid (NCHUNK, B); each block scans a slice of logits[b].
+__global__ __launch_bounds__(256) void k_argmax_part(
+    const float* __restrict__ logits, float* __restrict__ pval,
+    int32_t* __restrict__ pidx, int V, int nchunk) {
+    const int c = blockIdx.x, b = blockIdx.y;
+    const int per = (V + nchunk - 1) / nchunk;
+    const int lo = c * per, hi = min(lo + per, V);
+    float best = -1e30f;
+    int besti = 0;
+    for (int i = lo + (int)threadIdx.x; i < hi; i += 256) {
+        const float v = logits[(size_t)b * V + i];
+        if (v > best) { best = v; besti = i; }
+    }
+    __shared__ float sv[256];
+    __shared__ int si[256];
+    sv[threadIdx.x] = best;
+    si[threadIdx.x] = besti;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off && sv[threadIdx.x + off] > sv[threadIdx.x]) {
+            sv[threadIdx.x] = sv[threadIdx.x + off];
+            si[threadIdx.x] = si[threadIdx.x + off];
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        pval[(size_t)b * nchunk + c] = sv[0];
+        pidx[(size_t)b * nchunk + c] = si[0];
+    }
+}
+
+// stage 2 + state advance: grid B, block 64.
+__global__ __launch_bounds__(64) void k_argmax_final(
+    const float* __restrict__ pval, const int32_t* __restrict__ pidx,
+    int32_t* __restrict__ cur_ids, int32_t* __restrict__ n_past,
+    int32_t* __restrict__ gen_tokens, int32_t* __restrict__ gen_count,
+    int nchunk, int gen_cap) {
+    const int b = blockIdx.x;
+    const int t = threadIdx.x;
+    float v = (t < nchunk) ? pval[(size_t)b * nchunk + t] : -1e30f;
+    int i = (t < nchunk) ? pidx[(size_t)b * nchunk + t] : 0;
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        const float ov = __shfl_xor(v, off, 64);
+        const int oi = __shfl_xor(i, off, 64);
+        if (ov > v) { v = ov; i = oi; }
+    }
+    if (t == 0) {
+        cur_ids[b] = i;
+        const int gc = gen_count[b];
+        if (gc < gen_cap) gen_tokens[(size_t)b * gen_cap + gc] = i;
+        gen_count[b] = gc + 1;
+        n_past[b] = n_past[b] + 1;
+    }
+}
+
+// --------------------------------------------------------- launch stubs
+
+#define DISPATCH_DT(DTV, FN)                                          \
+    switch (DTV) {                                                    \
+        case DT::DQ4K: FN(DT::DQ4K); break;                           \
+        case DT::DQ6K: FN(DT::DQ6K); break;                           \
+        case DT::DQ8:  FN(DT::DQ8);  break;                           \
+        case DT::BF16: FN(DT::BF16); break;                           \
+        case DT::F16:  FN(DT::F16);  break;                           \
+        case DT::F32:  FN(DT::F32);  break;                           \
+        default: throw std::runtime_error("bad dtype");               \
+    }
+
+void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
+                 const float* res, float* y, int B, int ldy, float eps,
+                 hipStream_t stream) {
+    const int N = (int)w.n, K = (int)w.k;
+    if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
+    const size_t lds = (size_t)B * K * 4 + 256 * 4;
+    dim3 grid((N + 3) / 4), block(256);
+    #define GEMV_CASE(WT)                                                        \
+        do {                                                                     \
+            auto kern = (pre == PRE_RMS) ? k_gemv<WT, PRE_RMS>                   \
+                       : (pre == PRE_SILU) ? k_gemv<WT, PRE_SILU>                \
+                       : k_gemv<WT, PRE_NONE>;                                   \
+            hipLaunchKernelGGL(kern, grid, block, lds, stream,                   \
+                (const uint8_t*)w.qs, (const uint8_t*)w.hdr, xin, gw, res, y,    \
+                N, K, B, ldy, eps);                                                   \
+        } while (0)
+    DISPATCH_DT(w.dtype, GEMV_CASE);
+    #undef GEMV_CASE
+}
+
+void launch_embed(const WTensor& w, const int32_t* ids, float* x, int B,
+                  hipStream_t stream) {
+    #define EMBED_CASE(WT)                                                     \
+        hipLaunchKernelGGL(k_embed<WT>, dim3(B), dim3(256), 0, stream,         \
+            (const uint8_t*)w.qs, (const uint8_t*)w.hdr, ids, x, (int)w.k)
+    DISPATCH_DT(w.dtype, EMBED_CASE);
+    #undef EMBED_CASE
+}
+
+void launch_rope_append(float* qkv, const float* inv_freq,
+                        const int32_t* page_table, uint16_t* kv_pool,
+                        const int32_t* n_past, int B, int NH, int NKV, int D,
+                        int page_size, int max_pages, int64_t page_stride,
+                        hipStream_t stream) {
+    const int G = NH / NKV;
+    hipLaunchKernelGGL(k_rope_append, dim3(NKV, B), dim3(256), 0, stream,
+                       qkv, inv_freq, page_table, kv_pool, n_past,
+                       NH, NKV, D, G, page_size, max_pages, page_stride);
+}
+
+void launch_attn_decode(const float* qkv, const int32_t* page_table,
+                        const uint16_t* kv_pool, const int32_t* n_past,
+                        float* part_o, float* part_ml, int B, int NH, int NKV,
+                        int D, int S, int page_size, int max_pages,
+                        int64_t page_stride, float scale, hipStream_t stream) {
+    if (D != 128) throw std::runtime_error("attn kernel requires head_dim 128");
+    const int G = NH / NKV;
+    dim3 grid(S, NKV, B), block(256);
+    #define ATTN_CASE(GV)                                                       \
+        hipLaunchKernelGGL(k_attn_decode<GV>, grid, block, 0, stream,           \
+            qkv, page_table, kv_pool, n_past, part_o, part_ml,                  \
+            NH, NKV, D, S, page_size, max_pages, page_stride, scale)
+    switch (G) {
+        case 1: ATTN_CASE(1); break;
+        case 2: ATTN_CASE(2); break;
+        case 4: ATTN_CASE(4); break;
+        case 8: ATTN_CASE(8); break;
+        default: throw std::runtime_error("unsupported GQA ratio");
+    }
+    #undef ATTN_CASE
+}
+
+void launch_attn_combine(const float* part_o, const float* part_ml,
+                         float* attn_out, int B, int NH, int S, int D,
+                         hipStream_t stream) {
+    hipLaunchKernelGGL(k_attn_combine, dim3(B * NH), dim3(D), 0, stream,
+                       part_o, part_ml, attn_out, NH, S, D);
+}
+
+void launch_argmax(const float* logits, float* pval, int32_t* pidx,
+                   int32_t* cur_ids, int32_t* n_past, int32_t* gen_tokens,
+                   int32_t* gen_count, int B, int V, int gen_cap,
+                   hipStream_t stream) {
+    constexpr int NCHUNK = 64;
+    hipLaunchKernelGGL(k_argmax_part, dim3(NCHUNK, B), dim3(256), 0, stream,
+                       logits, pval, pidx, V, NCHUNK);
+    hipLaunchKernelGGL(k_argmax_final, dim3(B), dim3(64), 0, stream,
+                       pval, pidx, cur_ids, n_past, gen_tokens, gen_count,
+                       NCHUNK, gen_cap);
+}
+
+}  // namespace cla

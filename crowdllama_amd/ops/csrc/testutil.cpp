@@ -1,0 +1,42 @@
+// Test-only helpers: run single kernels on host-provided numpy buffers
+// (upload -> kernel -> download). Used by tests/test_gpu_kernels.py to diff
+// HIP kernels against the CPU reference codecs.
+#include "common.h"
+
+namespace cla {
+
+void launch_gemv(const WTensor&, int pre, const float* xin, const float* gw,
+                 const float* res, float* y, int B, int ldy, float eps,
+                 hipStream_t);
+
+void launch_gemv_test(const void* qs, const void* hdr, const float* x,
+                      const float* gw, float* y, int dtype, int N, int K,
+                      int B, int pre, size_t qs_bytes, size_t hdr_bytes) {
+    const DT dt = static_cast<DT>(dtype);
+    const size_t xn = (pre == 2) ? (size_t)B * 2 * K : (size_t)B * K;
+    void *d_qs = nullptr, *d_hdr = nullptr, *d_x = nullptr, *d_y = nullptr,
+         *d_gw = nullptr;
+    HIP_CHECK(hipMalloc(&d_qs, qs_bytes));
+    HIP_CHECK(hipMemcpy(d_qs, qs, qs_bytes, hipMemcpyHostToDevice));
+    if (hdr_bytes) {
+        HIP_CHECK(hipMalloc(&d_hdr, hdr_bytes));
+        HIP_CHECK(hipMemcpy(d_hdr, hdr, hdr_bytes, hipMemcpyHostToDevice));
+    }
+    HIP_CHECK(hipMalloc(&d_x, xn * 4));
+    HIP_CHECK(hipMemcpy(d_x, x, xn * 4, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMalloc(&d_y, (size_t)B * N * 4));
+    if (gw) {
+        HIP_CHECK(hipMalloc(&d_gw, (size_t)K * 4));
+        HIP_CHECK(hipMemcpy(d_gw, gw, (size_t)K * 4, hipMemcpyHostToDevice));
+    }
+    WTensor w;
+    w.dtype = dt; w.n = N; w.k = K; w.qs = d_qs; w.hdr = d_hdr;
+    launch_gemv(w, pre, (const float*)d_x, (const float*)d_gw, nullptr,
+                (float*)d_y, B, N, 1e-5f, nullptr);
+    HIP_CHECK(hipDeviceSynchronize());
+    HIP_CHECK(hipMemcpy(y, d_y, (size_t)B * N * 4, hipMemcpyDeviceToHost));
+    hipFree(d_qs); if (d_hdr) hipFree(d_hdr);
+    hipFree(d_x); hipFree(d_y); if (d_gw) hipFree(d_gw);
+}
+
+}  // namespace cla
