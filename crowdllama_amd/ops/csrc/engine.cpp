@@ -146,6 +146,10 @@ void repack(int32_t ggml_type, const uint8_t* src, int64_t rows, int64_t k,
 
 Engine::Engine(const std::string& gguf_path, const EngineConfig& cfg)
     : cfg_(cfg) {
+    if (cfg_.batch < 1 || cfg_.batch > 2)
+        throw std::runtime_error(
+            "decode batch must be 1 or 2 on the GEMV path "
+            "(batched MFMA path lands next)");
     HIP_CHECK(hipSetDevice(cfg_.device));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     GGUFFile gf(gguf_path);

@@ -540,6 +540,10 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
             auto kern = (pre == PRE_RMS) ? k_gemv<WT, PRE_RMS>                   \
                        : (pre == PRE_SILU) ? k_gemv<WT, PRE_SILU>                \
                        : k_gemv<WT, PRE_NONE>;                                   \
+            if (lds > 64 * 1024) {                                               \
+                (void)hipFuncSetAttribute((const void*)kern,                     \
+                    hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);       \
+            }                                                                    \
             hipLaunchKernelGGL(kern, grid, block, lds, stream,                   \
                 (const uint8_t*)w.qs, (const uint8_t*)w.hdr, xin, gw, res, y,    \
                 N, K, B, ldy, eps);                                                   \

@@ -1,0 +1,143 @@
+"""GPU kernel numerics: HIP fused dequant-GEMV vs the CPU reference codecs
+(plain fp32 numpy ground truth). Requires an MI355X (run via gpurun)."""
+
+import numpy as np
+import pytest
+
+from crowdllama_amd.quant import (
+    GGMLType, quantize_q4_k, quantize_q6_k, quantize_q8_0,
+    dequantize_q4_k, dequantize_q6_k, dequantize_q8_0,
+)
+
+pytestmark = pytest.mark.gpu
+
+# device DT codes (common.h)
+DT_F32, DT_F16, DT_BF16, DT_DQ4K, DT_DQ6K, DT_DQ8 = 0, 1, 2, 3, 4, 5
+PRE_NONE, PRE_RMS, PRE_SILU = 0, 1, 2
+
+
+@pytest.fixture(scope="module")
+def core():
+    from crowdllama_amd.ops import get_core
+    c = get_core()
+    if c.device_count() == 0:
+        pytest.skip("no GPU")
+    return c
+
+
+def _repack_q4k(raw, rows, k):
+    nsb = k // 256
+    blk = raw.reshape(rows, nsb, 144)
+    hdr = np.ascontiguousarray(blk[:, :, :16]).reshape(rows, -1)
+    qs = np.ascontiguousarray(blk[:, :, 16:]).reshape(rows, -1)
+    return qs, hdr
+
+
+def _repack_q6k(raw, rows, k):
+    nsb = k // 256
+    blk = raw.reshape(rows, nsb, 210)
+    ql = blk[:, :, 0:128]
+    qh = blk[:, :, 128:192]
+    sc = blk[:, :, 192:208]
+    d = blk[:, :, 208:210]
+    hdr = np.zeros((rows, nsb, 32), dtype=np.uint8)
+    hdr[:, :, 0:2] = d
+    hdr[:, :, 4:20] = sc
+    q = np.zeros((rows, nsb, 256), dtype=np.int8)
+    for half in range(2):
+        qlh = ql[:, :, half * 64:(half + 1) * 64]
+        qhh = qh[:, :, half * 32:(half + 1) * 32]
+        base = half * 128
+        q[:, :, base + 0:base + 32] = (((qlh[:, :, :32] & 0xF) | ((qhh & 3) << 4)).astype(np.int16) - 32).astype(np.int8)
+        q[:, :, base + 32:base + 64] = (((qlh[:, :, 32:] & 0xF) | (((qhh >> 2) & 3) << 4)).astype(np.int16) - 32).astype(np.int8)
+        q[:, :, base + 64:base + 96] = (((qlh[:, :, :32] >> 4) | (((qhh >> 4) & 3) << 4)).astype(np.int16) - 32).astype(np.int8)
+        q[:, :, base + 96:base + 128] = (((qlh[:, :, 32:] >> 4) | (((qhh >> 6) & 3) << 4)).astype(np.int16) - 32).astype(np.int8)
+    return q.view(np.uint8).reshape(rows, -1), hdr.reshape(rows, -1)
+
+
+def _repack_q8(raw, rows, k):
+    nb = k // 32
+    blk = raw.reshape(rows, nb, 34)
+    hdr = np.ascontiguousarray(blk[:, :, :2]).reshape(rows, -1)
+    qs = np.ascontiguousarray(blk[:, :, 2:]).reshape(rows, -1)
+    return qs, hdr
+
+
+CASES = {
+    "q4k": (DT_DQ4K, quantize_q4_k, dequantize_q4_k, _repack_q4k),
+    "q6k": (DT_DQ6K, quantize_q6_k, dequantize_q6_k, _repack_q6k),
+    "q8": (DT_DQ8, quantize_q8_0, dequantize_q8_0, _repack_q8),
+}
+
+
+@pytest.mark.parametrize("name", list(CASES))
+@pytest.mark.parametrize("B", [1, 2])
+def test_gemv_quant(core, name, B):
+    dt, quant, dequant, repack_fn = CASES[name]
+    rng = np.random.default_rng(42)
+    N, K = 64, 512
+    w = rng.standard_normal((N, K)).astype(np.float32) * 0.1
+    raw = quant(w)
+    wref = dequant(raw, K)          # exact values the GPU should use
+    x = rng.standard_normal((B, K)).astype(np.float32)
+    qs, hdr = repack_fn(raw.reshape(N, -1), N, K)
+    y = core.test_gemv(np.ascontiguousarray(qs), np.ascontiguousarray(hdr),
+                       x, dt, N, K, PRE_NONE, np.zeros(0, dtype=np.float32))
+    yref = x @ wref.reshape(N, K).T
+    np.testing.assert_allclose(y, yref, rtol=2e-4, atol=2e-4)
+
+
+@pytest.mark.parametrize("dt,conv", [
+    (DT_F32, lambda w: w.view(np.uint8)),
+    (DT_BF16, None),
+    (DT_F16, lambda w: w.astype(np.float16).view(np.uint8)),
+])
+def test_gemv_float(core, dt, conv):
+    from crowdllama_amd.quant import quantize
+    rng = np.random.default_rng(0)
+    N, K = 32, 256
+    w = rng.standard_normal((N, K)).astype(np.float32)
+    if dt == DT_BF16:
+        raw = quantize(w, GGMLType.BF16)
+        wref = ((raw.view(np.uint16).astype(np.uint32) << 16)
+                .view(np.float32)).reshape(N, K)
+    elif dt == DT_F16:
+        raw = conv(w)
+        wref = w.astype(np.float16).astype(np.float32)
+    else:
+        raw = conv(w)
+        wref = w
+    x = rng.standard_normal((1, K)).astype(np.float32)
+    y = core.test_gemv(np.ascontiguousarray(raw.reshape(N, -1)),
+                       np.zeros(0, dtype=np.uint8), x, dt, N, K, PRE_NONE,
+                       np.zeros(0, dtype=np.float32))
+    yref = x @ wref.T
+    np.testing.assert_allclose(y, yref, rtol=1e-4, atol=1e-4)
+
+
+def test_gemv_pre_rms(core):
+    rng = np.random.default_rng(1)
+    N, K = 32, 256
+    w = rng.standard_normal((N, K)).astype(np.float32)
+    gw = rng.standard_normal(K).astype(np.float32)
+    x = rng.standard_normal((1, K)).astype(np.float32)
+    y = core.test_gemv(np.ascontiguousarray(w.view(np.uint8)),
+                       np.zeros(0, dtype=np.uint8), x, DT_F32, N, K, PRE_RMS, gw)
+    eps = 1e-5
+    xn = x / np.sqrt((x * x).mean(axis=1, keepdims=True) + eps) * gw
+    yref = xn @ w.T
+    np.testing.assert_allclose(y, yref, rtol=1e-4, atol=1e-4)
+
+
+def test_gemv_pre_silu(core):
+    rng = np.random.default_rng(2)
+    N, K = 32, 256
+    w = rng.standard_normal((N, K)).astype(np.float32)
+    gu = rng.standard_normal((1, 2 * K)).astype(np.float32)
+    y = core.test_gemv(np.ascontiguousarray(w.view(np.uint8)),
+                       np.zeros(0, dtype=np.uint8), gu, DT_F32, N, K, PRE_SILU,
+                       np.zeros(0, dtype=np.float32))
+    g, u = gu[:, :K], gu[:, K:]
+    act = (g / (1 + np.exp(-g))) * u
+    yref = act @ w.T
+    np.testing.assert_allclose(y, yref, rtol=2e-4, atol=2e-4)
