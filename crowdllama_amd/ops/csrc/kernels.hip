@@ -319,23 +319,24 @@ __global__ __launch_bounds__(256) void k_rope_append(
 // ------------------------------------------------- decode attention (GQA)
 
 // grid (S, KVH, B); block 256 = 4 waves; each 16-lane quarter owns a full
-// (m, l, o[128]) online-softmax accumulator over its position subsequence,
-// for each of the G query heads of this kv group.
-template <int G>
+// (m, l, o[D]) online-softmax accumulator over its position subsequence,
+// for each of the G query heads of this kv group. D = 16*DPL (64 or 128).
+template <int G, int D>
 __global__ __launch_bounds__(256) void k_attn_decode(
     const float* __restrict__ qkv, const int32_t* __restrict__ page_table,
     const uint16_t* __restrict__ kv_pool, const int32_t* __restrict__ n_past,
     float* __restrict__ part_o,   // [B][NH][S][D]
     float* __restrict__ part_ml,  // [B][NH][S][2]
-    int NH, int NKV, int D, int S, int page_size, int max_pages,
+    int NH, int NKV, int S, int page_size, int max_pages,
     int64_t page_stride, float scale) {
+    constexpr int DPL = D / 16;   // dims per lane (bf16: 2*DPL bytes)
     const int s = blockIdx.x, kvh = blockIdx.y, b = blockIdx.z;
     const int len = n_past[b] + 1;
     const int tid = threadIdx.x;
     const int wave = tid >> 6, lane = tid & 63;
     const int quarter = lane >> 4, qlane = lane & 15;
     const int sub = wave * 4 + quarter;       // 0..15 position subsequence
-    const int d0 = qlane * 8;                 // this lane's 8-dim slice
+    const int d0 = qlane * DPL;               // this lane's dim slice
 
     int chunk = (len + S - 1) / S;
     chunk = (chunk + 3) & ~3;
@@ -343,21 +344,20 @@ __global__ __launch_bounds__(256) void k_attn_decode(
     const int end = min(start + chunk, len);
 
     const float* qkv_b = qkv + (size_t)b * (NH + 2 * NKV) * D;
-    // q fragments for the G heads of this group
-    float qf[G][8];
+    float qf[G][DPL];
     #pragma unroll
     for (int g = 0; g < G; g++) {
         const float* qh = qkv_b + (size_t)(kvh * G + g) * D + d0;
         #pragma unroll
-        for (int j = 0; j < 8; j++) qf[g][j] = qh[j] * scale;
+        for (int j = 0; j < DPL; j++) qf[g][j] = qh[j] * scale;
     }
 
-    float m[G], l[G], o[G][8];
+    float m[G], l[G], o[G][DPL];
     #pragma unroll
     for (int g = 0; g < G; g++) {
         m[g] = -1e30f; l[g] = 0.f;
         #pragma unroll
-        for (int j = 0; j < 8; j++) o[g][j] = 0.f;
+        for (int j = 0; j < DPL; j++) o[g][j] = 0.f;
     }
 
     for (int p = start + sub; p < end; p += 16) {
@@ -365,19 +365,22 @@ __global__ __launch_bounds__(256) void k_attn_decode(
         const uint16_t* kp = kv_pool + (int64_t)page * page_stride
                              + ((int64_t)kvh * 2 + 0) * page_size * D
                              + (int64_t)(p % page_size) * D + d0;
-        const uint4 kq = *reinterpret_cast<const uint4*>(kp);
-        float kf[8];
-        kf[0] = bf16_bits_to_f32(kq.x & 0xFFFF); kf[1] = bf16_bits_to_f32(kq.x >> 16);
-        kf[2] = bf16_bits_to_f32(kq.y & 0xFFFF); kf[3] = bf16_bits_to_f32(kq.y >> 16);
-        kf[4] = bf16_bits_to_f32(kq.z & 0xFFFF); kf[5] = bf16_bits_to_f32(kq.z >> 16);
-        kf[6] = bf16_bits_to_f32(kq.w & 0xFFFF); kf[7] = bf16_bits_to_f32(kq.w >> 16);
+        uint32_t kw[DPL / 2];
+        #pragma unroll
+        for (int j = 0; j < DPL / 2; j++)
+            kw[j] = reinterpret_cast<const uint32_t*>(kp)[j];
+        float kf[DPL];
+        #pragma unroll
+        for (int j = 0; j < DPL / 2; j++) {
+            kf[2 * j] = bf16_bits_to_f32(kw[j] & 0xFFFF);
+            kf[2 * j + 1] = bf16_bits_to_f32(kw[j] >> 16);
+        }
         float sc[G];
         #pragma unroll
         for (int g = 0; g < G; g++) {
             float d = 0.f;
             #pragma unroll
-            for (int j = 0; j < 8; j++) d += qf[g][j] * kf[j];
-            // reduce over the 16-lane quarter
+            for (int j = 0; j < DPL; j++) d += qf[g][j] * kf[j];
             #pragma unroll
             for (int off = 1; off < 16; off <<= 1) d += __shfl_xor(d, off, 64);
             sc[g] = d;
@@ -385,12 +388,16 @@ __global__ __launch_bounds__(256) void k_attn_decode(
         const uint16_t* vp = kv_pool + (int64_t)page * page_stride
                              + ((int64_t)kvh * 2 + 1) * page_size * D
                              + (int64_t)(p % page_size) * D + d0;
-        const uint4 vq = *reinterpret_cast<const uint4*>(vp);
-        float vf[8];
-        vf[0] = bf16_bits_to_f32(vq.x & 0xFFFF); vf[1] = bf16_bits_to_f32(vq.x >> 16);
-        vf[2] = bf16_bits_to_f32(vq.y & 0xFFFF); vf[3] = bf16_bits_to_f32(vq.y >> 16);
-        vf[4] = bf16_bits_to_f32(vq.z & 0xFFFF); vf[5] = bf16_bits_to_f32(vq.z >> 16);
-        vf[6] = bf16_bits_to_f32(vq.w & 0xFFFF); vf[7] = bf16_bits_to_f32(vq.w >> 16);
+        uint32_t vw[DPL / 2];
+        #pragma unroll
+        for (int j = 0; j < DPL / 2; j++)
+            vw[j] = reinterpret_cast<const uint32_t*>(vp)[j];
+        float vf[DPL];
+        #pragma unroll
+        for (int j = 0; j < DPL / 2; j++) {
+            vf[2 * j] = bf16_bits_to_f32(vw[j] & 0xFFFF);
+            vf[2 * j + 1] = bf16_bits_to_f32(vw[j] >> 16);
+        }
         #pragma unroll
         for (int g = 0; g < G; g++) {
             const float mn = fmaxf(m[g], sc[g]);
@@ -398,7 +405,7 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             const float w = __expf(sc[g] - mn);
             l[g] = l[g] * alpha + w;
             #pragma unroll
-            for (int j = 0; j < 8; j++) o[g][j] = o[g][j] * alpha + w * vf[j];
+            for (int j = 0; j < DPL; j++) o[g][j] = o[g][j] * alpha + w * vf[j];
             m[g] = mn;
         }
     }
@@ -406,7 +413,7 @@ __global__ __launch_bounds__(256) void k_attn_decode(
     // combine the 16 sub-accumulators through LDS
     __shared__ __attribute__((aligned(16))) float sm[16 * G];
     __shared__ __attribute__((aligned(16))) float sl[16 * G];
-    __shared__ __attribute__((aligned(16))) float so[16 * G * 128];
+    __shared__ __attribute__((aligned(16))) float so[16 * G * D];
     #pragma unroll
     for (int g = 0; g < G; g++) {
         if (qlane == 0) {
@@ -414,12 +421,11 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             sl[sub * G + g] = l[g];
         }
         #pragma unroll
-        for (int j = 0; j < 8; j++) so[(sub * G + g) * 128 + d0 + j] = o[g][j];
+        for (int j = 0; j < DPL; j++) so[(sub * G + g) * D + d0 + j] = o[g][j];
     }
     __syncthreads();
-    // threads 0..(G*128-1): one output dim each (G<=8 -> <=1024; block 256 loops)
-    for (int idx = tid; idx < G * 128; idx += 256) {
-        const int g = idx / 128, d = idx % 128;
+    for (int idx = tid; idx < G * D; idx += 256) {
+        const int g = idx / D, d = idx % D;
         float mstar = -1e30f;
         #pragma unroll
         for (int t = 0; t < 16; t++) mstar = fmaxf(mstar, sm[t * G + g]);
@@ -428,7 +434,7 @@ __global__ __launch_bounds__(256) void k_attn_decode(
         for (int t = 0; t < 16; t++) {
             const float e = __expf(sm[t * G + g] - mstar);
             lsum += e * sl[t * G + g];
-            osum += e * so[(t * G + g) * 128 + d];
+            osum += e * so[(t * G + g) * D + d];
         }
         const int head = kvh * G + g;
         part_o[(((size_t)b * NH + head) * S + s) * D + d] = osum;
@@ -577,20 +583,25 @@ void launch_attn_decode(const float* qkv, const int32_t* page_table,
                         float* part_o, float* part_ml, int B, int NH, int NKV,
                         int D, int S, int page_size, int max_pages,
                         int64_t page_stride, float scale, hipStream_t stream) {
-    if (D != 128) throw std::runtime_error("attn kernel requires head_dim 128");
     const int G = NH / NKV;
     dim3 grid(S, NKV, B), block(256);
-    #define ATTN_CASE(GV)                                                       \
-        hipLaunchKernelGGL(k_attn_decode<GV>, grid, block, 0, stream,           \
+    #define ATTN_CASE(GV, DV)                                                   \
+        hipLaunchKernelGGL((k_attn_decode<GV, DV>), grid, block, 0, stream,     \
             qkv, page_table, kv_pool, n_past, part_o, part_ml,                  \
-            NH, NKV, D, S, page_size, max_pages, page_stride, scale)
+            NH, NKV, S, page_size, max_pages, page_stride, scale)
+    #define ATTN_D(GV)                                                          \
+        do { if (D == 128) ATTN_CASE(GV, 128);                                  \
+             else if (D == 64) ATTN_CASE(GV, 64);                               \
+             else throw std::runtime_error("head_dim must be 64 or 128");       \
+        } while (0)
     switch (G) {
-        case 1: ATTN_CASE(1); break;
-        case 2: ATTN_CASE(2); break;
-        case 4: ATTN_CASE(4); break;
-        case 8: ATTN_CASE(8); break;
+        case 1: ATTN_D(1); break;
+        case 2: ATTN_D(2); break;
+        case 4: ATTN_D(4); break;
+        case 8: ATTN_D(8); break;
         default: throw std::runtime_error("unsupported GQA ratio");
     }
+    #undef ATTN_D
     #undef ATTN_CASE
 }
 
