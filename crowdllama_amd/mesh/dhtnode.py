@@ -1,0 +1,170 @@
+"""Rendezvous/bootstrap node — the MI355X mesh analog of the reference's
+Kademlia DHT server (pkg/dht/dht.go:25-437).
+
+The reference uses libp2p Kademlia purely as a rendezvous: workers
+`Provide` a namespace CID every second and consumers `FindProvidersAsync`
+it. This node implements those semantics directly over the
+mesh wire protocol: provider records with TTL expiry, peer-address lookup,
+connection stats logging, and eager eviction on disconnect
+(dht.go:370-383).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import time
+from dataclasses import dataclass, field
+
+from ..config import Config
+from ..logutil import new_app_logger
+from .wire import (PROTO_RENDEZVOUS, accept_protocol, read_json, write_json)
+
+
+@dataclass
+class ProviderRecord:
+    peer_id: str
+    addrs: list[str]
+    last_provide: float = field(default_factory=time.time)
+
+
+class DHTServer:
+    """Standalone bootstrap/rendezvous node (reference pkg/dht Server)."""
+
+    PROVIDER_TTL = 30.0  # records expire without re-provide (ref: 1 s loop)
+
+    def __init__(self, cfg: Config, peer_id: str):
+        self.cfg = cfg
+        self.peer_id = peer_id
+        self.log = new_app_logger("dht", cfg.verbose)
+        self.providers: dict[str, dict[str, ProviderRecord]] = {}  # ns -> id -> rec
+        self.peer_addrs: dict[str, list[str]] = {}
+        self.conn_count = 0
+        self.total_conns = 0
+        self._server: asyncio.base_events.Server | None = None
+        self._tasks: list[asyncio.Task] = []
+        self.port: int | None = None
+
+    async def start(self, host: str | None = None, port: int | None = None):
+        host = host or self.cfg.listen_host
+        port = self.cfg.dht_port if port is None else port
+        self._server = await asyncio.start_server(self._on_conn, host, port)
+        self.port = self._server.sockets[0].getsockname()[1]
+        self._tasks.append(asyncio.create_task(self._stats_loop()))
+        self._tasks.append(asyncio.create_task(self._expiry_loop()))
+        self.log.info("DHT server %s listening on %s:%d",
+                      self.peer_id, host, self.port)
+        return self.port
+
+    async def stop(self):
+        for t in self._tasks:
+            t.cancel()
+        for t in self._tasks:
+            try:
+                await t
+            except (asyncio.CancelledError, Exception):
+                pass
+        self._tasks.clear()
+        if self._server:
+            self._server.close()
+            await self._server.wait_closed()
+            self._server = None
+
+    # ------------------------------------------------------------ serving
+
+    async def _on_conn(self, reader: asyncio.StreamReader,
+                       writer: asyncio.StreamWriter):
+        self.conn_count += 1
+        self.total_conns += 1
+        try:
+            proto = await accept_protocol(reader)
+            if proto != PROTO_RENDEZVOUS:
+                self.log.warning("unknown protocol %r", proto)
+                return
+            while True:
+                try:
+                    msg = await read_json(reader, timeout=60.0)
+                except (asyncio.IncompleteReadError, asyncio.TimeoutError,
+                        ConnectionError):
+                    return
+                resp = self._handle(msg)
+                await write_json(writer, resp)
+        except Exception as e:  # noqa: BLE001 — per-conn isolation
+            self.log.debug("conn error: %s", e)
+        finally:
+            self.conn_count -= 1
+            writer.close()
+
+    def _handle(self, msg: dict) -> dict:
+        op = msg.get("op")
+        if op == "ping":
+            return {"ok": True, "peer_id": self.peer_id}
+        if op == "provide":
+            ns = msg.get("ns", "")
+            pid = msg.get("peer_id", "")
+            addrs = list(msg.get("addrs", []))
+            if not ns or not pid:
+                return {"ok": False, "error": "missing ns/peer_id"}
+            self.providers.setdefault(ns, {})[pid] = ProviderRecord(
+                pid, addrs)
+            self.peer_addrs[pid] = addrs
+            return {"ok": True}
+        if op == "find_providers":
+            ns = msg.get("ns", "")
+            limit = int(msg.get("limit", 10))  # ref: FindProvidersAsync(cid,10)
+            out = []
+            now = time.time()
+            for rec in list(self.providers.get(ns, {}).values()):
+                if now - rec.last_provide > self.PROVIDER_TTL:
+                    continue
+                out.append({"peer_id": rec.peer_id, "addrs": rec.addrs})
+                if len(out) >= limit:
+                    break
+            return {"ok": True, "providers": out}
+        if op == "find_peer":
+            pid = msg.get("peer_id", "")
+            addrs = self.peer_addrs.get(pid)
+            if addrs is None:
+                return {"ok": False, "error": "peer not found"}
+            return {"ok": True, "addrs": addrs}
+        if op == "remove":
+            pid = msg.get("peer_id", "")
+            for ns in self.providers.values():
+                ns.pop(pid, None)
+            self.peer_addrs.pop(pid, None)
+            return {"ok": True}
+        if op == "stats":
+            return {"ok": True, "stats": self.stats()}
+        return {"ok": False, "error": f"unknown op {op!r}"}
+
+    def stats(self) -> dict:
+        nprov = sum(len(v) for v in self.providers.values())
+        return {
+            "peer_id": self.peer_id,
+            "known_peers": len(self.peer_addrs),
+            "providers": nprov,
+            "namespaces": len(self.providers),
+            "active_conns": self.conn_count,
+            "total_conns": self.total_conns,
+        }
+
+    # ----------------------------------------------------------- bg loops
+
+    async def _stats_loop(self):
+        # reference: NAT/peer stats every 30 s / 15 s (dht.go:279-321)
+        while True:
+            await asyncio.sleep(self.cfg.intervals.nat_log)
+            s = self.stats()
+            self.log.info("stats: peers=%d providers=%d conns=%d/%d",
+                          s["known_peers"], s["providers"],
+                          s["active_conns"], s["total_conns"])
+
+    async def _expiry_loop(self):
+        while True:
+            await asyncio.sleep(self.PROVIDER_TTL / 2)
+            now = time.time()
+            for ns, recs in self.providers.items():
+                dead = [pid for pid, r in recs.items()
+                        if now - r.last_provide > self.PROVIDER_TTL]
+                for pid in dead:
+                    del recs[pid]
+                    self.log.debug("expired provider %s in %s", pid, ns)

@@ -1,0 +1,171 @@
+"""Discovery client (reference parity: internal/discovery/discovery.go).
+
+Wraps rendezvous RPCs (provide / find_providers / find_peer — the analog of
+DHT Provide/FindProvidersAsync/FindPeer) and the peer metadata protocol
+(RequestPeerMetadata, discovery.go:186-275: 5 s deadline, EOF-delimited
+Resource JSON)."""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+
+from .resource import Resource
+from .wire import (NAMESPACE, PROTO_METADATA, PROTO_RENDEZVOUS,
+                   open_protocol, read_json, write_json)
+
+
+def parse_addr(addr: str) -> tuple[str, int]:
+    host, _, port = addr.rpartition(":")
+    return host or "127.0.0.1", int(port)
+
+
+class RendezvousClient:
+    """Persistent connection to one bootstrap/rendezvous node."""
+
+    def __init__(self, addr: str, timeout: float = 5.0):
+        self.addr = addr
+        self.timeout = timeout
+        self._reader: asyncio.StreamReader | None = None
+        self._writer: asyncio.StreamWriter | None = None
+        self._lock = asyncio.Lock()
+
+    async def _ensure(self):
+        if self._writer is None or self._writer.is_closing():
+            host, port = parse_addr(self.addr)
+            self._reader, self._writer = await open_protocol(
+                host, port, PROTO_RENDEZVOUS, self.timeout)
+
+    async def call(self, msg: dict) -> dict:
+        async with self._lock:
+            try:
+                await self._ensure()
+                await write_json(self._writer, msg)
+                return await read_json(self._reader, self.timeout)
+            except Exception:
+                # one reconnect attempt per call
+                await self.close()
+                await self._ensure()
+                await write_json(self._writer, msg)
+                return await read_json(self._reader, self.timeout)
+
+    async def close(self):
+        if self._writer is not None:
+            try:
+                self._writer.close()
+            except Exception:
+                pass
+        self._reader = self._writer = None
+
+    async def ping(self) -> bool:
+        try:
+            r = await self.call({"op": "ping"})
+            return bool(r.get("ok"))
+        except Exception:
+            return False
+
+    async def provide(self, peer_id: str, addrs: list[str],
+                      ns: str = NAMESPACE) -> bool:
+        r = await self.call({"op": "provide", "ns": ns, "peer_id": peer_id,
+                             "addrs": addrs})
+        return bool(r.get("ok"))
+
+    async def find_providers(self, ns: str = NAMESPACE,
+                             limit: int = 10) -> list[dict]:
+        r = await self.call({"op": "find_providers", "ns": ns, "limit": limit})
+        return r.get("providers", []) if r.get("ok") else []
+
+    async def find_peer(self, peer_id: str) -> list[str] | None:
+        r = await self.call({"op": "find_peer", "peer_id": peer_id})
+        return r.get("addrs") if r.get("ok") else None
+
+
+class Discovery:
+    """Multi-bootstrap discovery + metadata fetch (reference Discovery)."""
+
+    def __init__(self, bootstrap_addrs: list[str],
+                 metadata_timeout: float = 5.0,
+                 metadata_max_age: float = 3600.0,
+                 log: logging.Logger | None = None):
+        self.clients = [RendezvousClient(a) for a in bootstrap_addrs]
+        self.metadata_timeout = metadata_timeout
+        self.metadata_max_age = metadata_max_age
+        self.log = log or logging.getLogger("discovery")
+
+    async def close(self):
+        for c in self.clients:
+            await c.close()
+
+    async def bootstrap_ok(self) -> bool:
+        for c in self.clients:
+            if await c.ping():
+                return True
+        return False
+
+    async def advertise(self, peer_id: str, addrs: list[str],
+                        ns: str = NAMESPACE) -> None:
+        for c in self.clients:
+            try:
+                await c.provide(peer_id, addrs, ns)
+            except Exception as e:
+                self.log.debug("advertise to %s failed: %s", c.addr, e)
+
+    async def find_providers(self, ns: str = NAMESPACE,
+                             limit: int = 10) -> list[dict]:
+        seen = {}
+        for c in self.clients:
+            try:
+                for p in await c.find_providers(ns, limit):
+                    seen[p["peer_id"]] = p
+            except Exception as e:
+                self.log.debug("find_providers on %s failed: %s", c.addr, e)
+        return list(seen.values())[:limit]
+
+    async def find_peer_addrs(self, peer_id: str) -> list[str] | None:
+        for c in self.clients:
+            try:
+                addrs = await c.find_peer(peer_id)
+                if addrs:
+                    return addrs
+            except Exception:
+                continue
+        return None
+
+    async def request_metadata(self, addrs: list[str]) -> Resource:
+        """Open a metadata stream and read EOF-delimited Resource JSON
+        (reference discovery.go:186-275)."""
+        last_err: Exception | None = None
+        for addr in addrs:
+            host, port = parse_addr(addr)
+            try:
+                reader, writer = await open_protocol(
+                    host, port, PROTO_METADATA, self.metadata_timeout)
+                try:
+                    data = await asyncio.wait_for(reader.read(64 * 1024),
+                                                  self.metadata_timeout)
+                    res = Resource.from_json(data)
+                    if res.age() > self.metadata_max_age:
+                        raise ValueError("metadata too stale")
+                    return res
+                finally:
+                    writer.close()
+            except Exception as e:  # noqa: BLE001
+                last_err = e
+        raise last_err or ConnectionError("no addresses")
+
+    async def discover_peers(self, limit: int = 10,
+                             ns: str = NAMESPACE) -> list[Resource]:
+        """FindProviders + per-provider metadata fetch
+        (reference DiscoverPeers, discovery.go:332-366)."""
+        out = []
+        for p in await self.find_providers(ns, limit):
+            try:
+                res = await self.request_metadata(p.get("addrs", []))
+                res.peer_id = p["peer_id"]
+                if not res.addrs:
+                    res.addrs = p.get("addrs", [])
+                out.append(res)
+            except Exception as e:  # noqa: BLE001
+                self.log.debug("metadata fetch from %s failed: %s",
+                               p.get("peer_id"), e)
+        return out

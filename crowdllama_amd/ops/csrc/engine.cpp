@@ -303,7 +303,10 @@ void Engine::alloc_state() {
     const int F = meta_.ffn;
     attn_splits_ = std::max(1, std::min(64, 512 / std::max(1, B * NKV)));
     max_pages_ = (cfg_.max_seq + cfg_.page_size - 1) / cfg_.page_size;
-    page_stride_ = (int64_t)NKV * 2 * cfg_.page_size * D;
+    // pool layout: [page][layer][kvh][2][page_size][D] bf16 — one pool, all
+    // layers; per-layer base pointer passed at launch.
+    layer_stride_ = (int64_t)NKV * 2 * cfg_.page_size * D;
+    page_stride_ = layer_stride_ * meta_.layers;
 
     auto dalloc = [&](size_t bytes) {
         void* p = nullptr;
@@ -359,14 +362,17 @@ void Engine::step(hipStream_t s) {
     const float eps = meta_.rms_eps;
 
     launch_embed(embed_, cur_ids_, x_, B, s);
+    int li = 0;
     for (auto& L : layers_) {
+        uint16_t* kv_layer = kv_pool_ + (int64_t)li * layer_stride_;
+        li++;
         for (auto& pt : L.qkv.parts)
             launch_gemv(pt.w, PRE_RMS, x_, L.attn_norm, nullptr,
                         qkv_ + pt.row_off, B, (int)L.qkv.n_total, eps, s);
-        launch_rope_append(qkv_, inv_freq_, page_table_, kv_pool_, n_past_,
+        launch_rope_append(qkv_, inv_freq_, page_table_, kv_layer, n_past_,
                            B, NH, NKV, D, cfg_.page_size, max_pages_,
                            page_stride_, s);
-        launch_attn_decode(qkv_, page_table_, kv_pool_, n_past_, part_o_,
+        launch_attn_decode(qkv_, page_table_, kv_layer, n_past_, part_o_,
                            part_ml_, B, NH, NKV, D, attn_splits_,
                            cfg_.page_size, max_pages_, page_stride_, scale, s);
         launch_attn_combine(part_o_, part_ml_, attn_out_, B, NH, attn_splits_,
@@ -404,10 +410,15 @@ void Engine::prefill(const std::vector<int32_t>& ids, int len) {
     if ((int)ids.size() != B * len)
         throw std::runtime_error("prefill: ids must be batch*len");
     ensure_graph();
-    std::vector<int32_t> col(B);
+    // Transposed staging buffer that stays alive until the final stream
+    // sync: hipMemcpyAsync from pageable memory may read the buffer late,
+    // so per-iteration reuse of one buffer races with in-flight copies.
+    std::vector<int32_t> cols((size_t)len * B);
+    for (int t = 0; t < len; t++)
+        for (int b = 0; b < B; b++)
+            cols[(size_t)t * B + b] = ids[(size_t)b * len + t];
     for (int t = 0; t < len; t++) {
-        for (int b = 0; b < B; b++) col[b] = ids[(size_t)b * len + t];
-        HIP_CHECK(hipMemcpyAsync(cur_ids_, col.data(), B * 4,
+        HIP_CHECK(hipMemcpyAsync(cur_ids_, cols.data() + (size_t)t * B, B * 4,
                                  hipMemcpyHostToDevice, stream_));
         if (graph_exec_) {
             HIP_CHECK(hipGraphLaunch(graph_exec_, stream_));

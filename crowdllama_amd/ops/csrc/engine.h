@@ -116,7 +116,8 @@ private:
     uint16_t* kv_pool_ = nullptr;
     int attn_splits_ = 16;
     int max_pages_ = 0;
-    int64_t page_stride_ = 0;
+    int64_t page_stride_ = 0;   // bytes-in-elements per page (all layers)
+    int64_t layer_stride_ = 0;  // per-layer offset within a page
 
     hipGraphExec_t graph_exec_ = nullptr;
     double last_decode_ms_ = 0.0;
