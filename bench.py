@@ -1,0 +1,146 @@
+#!/usr/bin/env python3
+"""Flagship bench: llama3-8b Q4_K_M decode on N MI355X GPUs as N DP workers.
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W`; for N>1
+launched via torch.distributed.run with one rank per GPU. One "step" = one
+decode iteration of the engine's whole batch (B tokens per step per GPU).
+Rank 0 prints ONE JSON line with the aggregate whole-job tokens/sec
+(BASELINE.json metric: "aggregate tokens/sec + p50 latency, llama3:8b Q4_K
+across 1/2/4/8 workers"); weights are random-init GGUF (no network for real
+checkpoints), prompts synthetic.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+
+def log(msg: str) -> None:
+    print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=256)
+    ap.add_argument("--warmup", type=int, default=32)
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--scheme", default="q4_k_m")
+    ap.add_argument("--batch", type=int, default=1,
+                    help="decode slots per GPU")
+    ap.add_argument("--prompt-len", type=int, default=128)
+    ap.add_argument("--max-seq", type=int, default=4096)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(args.gpus, world)
+
+    import torch
+    import torch.distributed as dist
+    distributed = world > 1
+    if distributed:
+        # DP replicas share no tensors; gloo carries the control barriers.
+        # (The TP=8 worker path is where RCCL over xGMI runs — see
+        # crowdllama_amd/parallel/.)
+        dist.init_process_group(backend="gloo")
+
+    def barrier():
+        if distributed:
+            dist.barrier()
+
+    from crowdllama_amd.models import get_preset, synth_path
+
+    cfg = get_preset(args.model)
+    # rank 0 generates the shared checkpoint; same /tmp on a 1-node job
+    if rank == 0:
+        t0 = time.time()
+        path = synth_path(args.model, scheme=args.scheme, mode="fast")
+        log(f"checkpoint ready in {time.time() - t0:.1f}s: {path}")
+    barrier()
+    path = synth_path(args.model, scheme=args.scheme, mode="fast")
+
+    from crowdllama_amd.ops import get_core
+    core = get_core()
+    if core.device_count() == 0:
+        log("no GPU visible — printing a null result")
+        print(json.dumps({"metric": "aggregate_tokens_per_sec", "value": None,
+                          "error": "no GPU"}))
+        return
+
+    ecfg = core.EngineConfig()
+    ecfg.batch = args.batch
+    ecfg.max_seq = args.max_seq
+    ecfg.device = local_rank if core.device_count() > local_rank else 0
+    t0 = time.time()
+    eng = core.Engine(path, ecfg)
+    log(f"rank {rank}: engine loaded in {time.time() - t0:.1f}s "
+        f"({eng.vram_bytes() / 1e9:.2f} GB VRAM)")
+
+    # synthetic prompt prefill (untimed)
+    rng = np.random.default_rng(1234 + rank)
+    prompts = rng.integers(3, cfg.vocab_size - 1,
+                           size=(args.batch, args.prompt_len)).astype(np.int32)
+    eng.prefill(prompts)
+
+    # warmup
+    if args.warmup > 0:
+        eng.decode(args.warmup)
+    barrier()
+    torch.cuda.synchronize() if torch.cuda.is_available() else None
+
+    t_start = time.perf_counter()
+    eng.decode(args.steps)  # one hipDeviceSynchronize inside after K replays
+    torch.cuda.synchronize() if torch.cuda.is_available() else None
+    elapsed = time.perf_counter() - t_start
+    barrier()
+
+    # MAX over ranks of elapsed
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    tokens_total = n_gpus * args.batch * args.steps
+    value = tokens_total / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+
+    if rank == 0:
+        result = {
+            "metric": "aggregate_tokens_per_sec",
+            "value": value,
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "p50_latency_ms": ms_per_step,  # B=1: step latency == token latency
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "q4_k_m+f32-activations",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "quant": args.scheme,
+                "global_batch": n_gpus * args.batch,
+                "seq_len": args.prompt_len,
+                "parallelism": f"dp{n_gpus}",
+                "engine_ms_per_step": eng.last_decode_ms() / args.steps,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

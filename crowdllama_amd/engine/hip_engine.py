@@ -1,0 +1,96 @@
+"""HIP-backed EngineBase: the worker's real compute backend (replaces the
+reference's WorkerAPIHandler -> Ollama HTTP shell-out, api.go:45-160)."""
+
+from __future__ import annotations
+
+import asyncio
+import threading
+import time
+from concurrent.futures import ThreadPoolExecutor
+
+from ..quant.gguf import GGUFReader
+from ..tokenizer import Tokenizer
+from .api import EngineBase, GenerateResult, RollingRate
+
+
+class HipEngine(EngineBase):
+    """One GGUF model resident on one MI355X, served request-at-a-time
+    (matching the reference's per-worker request granularity)."""
+
+    def __init__(self, model_name: str, gguf_path: str, device: int = 0,
+                 max_seq: int = 4096, use_graph: bool = True):
+        from ..ops import get_core
+        core = self._core = get_core()
+        if core.device_count() == 0:
+            raise RuntimeError("HipEngine requires a GPU (none visible); "
+                               "use MockEngine for CPU-only meshes")
+        self.model_name = model_name
+        cfg = core.EngineConfig()
+        cfg.batch = 1
+        cfg.max_seq = max_seq
+        cfg.device = device
+        cfg.use_graph = use_graph
+        self.eng = core.Engine(gguf_path, cfg)
+        with GGUFReader(gguf_path) as r:
+            self.tok = Tokenizer.from_gguf(r)
+        self._props = core.device_props(device)
+        self._rate = RollingRate()
+        self._lock = threading.Lock()
+        self._pool = ThreadPoolExecutor(max_workers=1,
+                                        thread_name_prefix="hipengine")
+        self._active = 0
+        self.max_seq = max_seq
+
+    # ------------------------------------------------------------ generate
+
+    def _generate_sync(self, prompt: str, max_new_tokens: int) -> GenerateResult:
+        import numpy as np
+        with self._lock:
+            t0 = time.monotonic_ns()
+            ids = self.tok.encode(prompt)[: self.max_seq - max_new_tokens - 1]
+            if not ids:
+                ids = [self.tok.bos_id]
+            self.eng.reset()
+            self.eng.prefill(np.asarray([ids], dtype=np.int32))
+            n_new = max(1, max_new_tokens)
+            if n_new > 1:
+                self.eng.decode(n_new - 1)
+            out = list(self.eng.gen_tokens(0))
+            if self.tok.eos_id in out:
+                out = out[: out.index(self.tok.eos_id)]
+                reason = "stop"
+            else:
+                reason = "length"
+            text = self.tok.decode(out)
+            dur = time.monotonic_ns() - t0
+            self._rate.add(len(out))
+            return GenerateResult(text=text, tokens_generated=len(out),
+                                  duration_ns=dur, done_reason=reason)
+
+    async def generate(self, prompt: str, max_new_tokens: int = 256,
+                       temperature: float = 0.0) -> GenerateResult:
+        loop = asyncio.get_running_loop()
+        self._active += 1
+        try:
+            return await loop.run_in_executor(
+                self._pool, self._generate_sync, prompt, max_new_tokens)
+        finally:
+            self._active -= 1
+
+    # ------------------------------------------------------------ metadata
+
+    def throughput(self) -> float:
+        r = self._rate.rate()
+        return r if r > 0 else 100.0  # pre-first-request estimate
+
+    def load(self) -> float:
+        return min(1.0, float(self._active))
+
+    def vram_gb(self) -> float:
+        return float(self._props["total_mem_gb"])
+
+    def gpu_model(self) -> str:
+        return str(self._props["name"])
+
+    async def close(self) -> None:
+        self._pool.shutdown(wait=False)

@@ -1,0 +1,225 @@
+"""Consumer HTTP gateway — Ollama-compatible API (reference parity:
+pkg/gateway/gateway.go).
+
+Serves POST /api/chat + GET /api/health on :9001 with request-logging
+middleware (gateway.go:87-154); handleChat: validate -> FindBestWorker ->
+503 if none -> RequestInference over the mesh inference protocol -> map PB
+response to Ollama-style JSON (gateway.go:168-231). Non-streaming, like the
+reference (gateway.go:243-293 is strictly one-shot). Background discovery
+pulls DHT providers into the peer manager (gateway.go:351-423)."""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+
+from aiohttp import web
+
+from ..config import Config
+from ..logutil import new_app_logger
+from . import pb
+from .discovery import parse_addr
+from .peer import Peer
+from .wire import PROTO_INFERENCE, open_protocol, read_frame, write_frame
+
+
+class Gateway:
+    def __init__(self, peer: Peer, cfg: Config):
+        self.peer = peer
+        self.cfg = cfg
+        self.log = new_app_logger("gateway", cfg.verbose)
+        self.app = web.Application(middlewares=[self._log_middleware])
+        self.app.router.add_post("/api/chat", self.handle_chat)
+        self.app.router.add_post("/api/generate", self.handle_generate)
+        self.app.router.add_get("/api/health", self.handle_health)
+        self.app.router.add_get("/api/tags", self.handle_tags)
+        self._runner: web.AppRunner | None = None
+        self.port: int | None = None
+        self._tasks: list[asyncio.Task] = []
+
+    # ----------------------------------------------------------- lifecycle
+
+    async def start(self, port: int | None = None) -> int:
+        self._runner = web.AppRunner(self.app, access_log=None)
+        await self._runner.setup()
+        site = web.TCPSite(self._runner, self.cfg.listen_host,
+                           self.cfg.gateway_port if port is None else port)
+        await site.start()
+        self.port = site._server.sockets[0].getsockname()[1]  # noqa: SLF001
+        self._tasks.append(asyncio.create_task(self._discovery_loop()))
+        self.log.info("gateway listening on :%d", self.port)
+        return self.port
+
+    async def stop(self) -> None:
+        for t in self._tasks:
+            t.cancel()
+        for t in self._tasks:
+            try:
+                await t
+            except (asyncio.CancelledError, Exception):
+                pass
+        self._tasks.clear()
+        if self._runner:
+            await self._runner.cleanup()
+            self._runner = None
+
+    @web.middleware
+    async def _log_middleware(self, request: web.Request, handler):
+        t0 = time.monotonic()
+        try:
+            resp = await handler(request)
+            return resp
+        finally:
+            self.log.info("%s %s %.1fms", request.method, request.path,
+                          (time.monotonic() - t0) * 1e3)
+
+    # ------------------------------------------------------------ discovery
+
+    async def _discovery_loop(self) -> None:
+        # gateway-side provider pull (reference gateway.go:351-423); unlike
+        # the reference this shares the peer's single PeerManager code path
+        # (deduplicate gateway-vs-peermanager discovery).
+        while True:
+            try:
+                for res in await self.peer.discovery.discover_peers():
+                    await self.peer.peer_manager.add_or_update_peer(res)
+            except Exception as e:  # noqa: BLE001
+                self.log.debug("gateway discovery failed: %s", e)
+            await asyncio.sleep(self.cfg.intervals.gateway_discovery)
+
+    # ------------------------------------------------------------- handlers
+
+    def find_best_worker(self, model: str):
+        return self.peer.peer_manager.find_best_worker(model)
+
+    async def request_inference(self, worker, model: str, prompt: str,
+                                stream: bool = False,
+                                timeout: float = 300.0) -> pb.GenerateResponse:
+        """Dial the worker's inference protocol, one PB request/response
+        (reference RequestInference, gateway.go:243-293)."""
+        addrs = list(worker.addrs)
+        if not addrs:
+            found = await self.peer.discovery.find_peer_addrs(worker.peer_id)
+            addrs = found or []
+        last: Exception | None = None
+        for addr in addrs:
+            host, port = parse_addr(addr)
+            try:
+                reader, writer = await open_protocol(host, port,
+                                                     PROTO_INFERENCE)
+                try:
+                    await write_frame(writer,
+                                      pb.request_message(model, prompt,
+                                                         stream).encode())
+                    frame = await read_frame(reader, timeout=timeout)
+                finally:
+                    writer.close()
+                resp = pb.BaseMessage.decode(frame).generate_response
+                if resp is None:
+                    raise ValueError("no GenerateResponse in reply")
+                return resp
+            except Exception as e:  # noqa: BLE001
+                last = e
+        raise last or ConnectionError("worker unreachable")
+
+    async def handle_chat(self, request: web.Request) -> web.Response:
+        try:
+            body = await request.json()
+        except json.JSONDecodeError:
+            return web.json_response({"error": "invalid JSON"}, status=400)
+        model = body.get("model", "")
+        messages = body.get("messages", [])
+        if not model or not messages:
+            return web.json_response(
+                {"error": "model and messages are required"}, status=400)
+        prompt = "\n".join(m.get("content", "") for m in messages
+                           if isinstance(m, dict))
+        worker = self.find_best_worker(model)
+        if worker is None:
+            return web.json_response(
+                {"error": f"no available worker for model {model}"},
+                status=503)  # gateway.go:192-199
+        try:
+            resp = await self.request_inference(worker, model, prompt,
+                                                bool(body.get("stream")))
+        except Exception as e:  # noqa: BLE001
+            self.log.warning("inference via %s failed: %s",
+                             worker.peer_id, e)
+            return web.json_response(
+                {"error": f"inference failed: {e}"}, status=500)
+        if resp.done_reason == "error":
+            return web.json_response({"error": resp.response}, status=500)
+        return web.json_response({
+            "model": resp.model,
+            "created_at": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
+            "message": {"role": "assistant", "content": resp.response},
+            "done": resp.done,
+            "done_reason": resp.done_reason or "stop",
+            "total_duration": resp.total_duration,
+            "worker_id": resp.worker_id,
+        })
+
+    async def handle_generate(self, request: web.Request) -> web.Response:
+        """Ollama /api/generate (prompt-in, response-out)."""
+        try:
+            body = await request.json()
+        except json.JSONDecodeError:
+            return web.json_response({"error": "invalid JSON"}, status=400)
+        model = body.get("model", "")
+        prompt = body.get("prompt", "")
+        if not model:
+            return web.json_response({"error": "model is required"},
+                                     status=400)
+        worker = self.find_best_worker(model)
+        if worker is None:
+            return web.json_response(
+                {"error": f"no available worker for model {model}"},
+                status=503)
+        try:
+            resp = await self.request_inference(worker, model, prompt,
+                                                bool(body.get("stream")))
+        except Exception as e:  # noqa: BLE001
+            return web.json_response({"error": f"inference failed: {e}"},
+                                     status=500)
+        return web.json_response({
+            "model": resp.model,
+            "created_at": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
+            "response": resp.response,
+            "done": resp.done,
+            "done_reason": resp.done_reason or "stop",
+            "total_duration": resp.total_duration,
+        })
+
+    async def handle_health(self, request: web.Request) -> web.Response:
+        """Full per-worker health map (reference gateway.go:426-461)."""
+        pm = self.peer.peer_manager
+        workers = []
+        for pid, pi in pm.peers.items():
+            r = pi.resource
+            workers.append({
+                "peer_id": pid,
+                "healthy": pi.is_healthy,
+                "worker_mode": r.worker_mode,
+                "gpu_model": r.gpu_model,
+                "supported_models": r.supported_models,
+                "tokens_throughput": r.tokens_throughput,
+                "vram_gb": r.vram_gb,
+                "load": r.load,
+                "last_seen": pi.last_seen,
+                "failed_attempts": pi.failed_attempts,
+            })
+        return web.json_response({
+            "status": "ok",
+            "peer_id": self.peer.peer_id,
+            "statistics": pm.get_peer_statistics(),
+            "workers": workers,
+        })
+
+    async def handle_tags(self, request: web.Request) -> web.Response:
+        """Ollama /api/tags analog: models available across the mesh."""
+        models = set()
+        for r in self.peer.peer_manager.get_healthy_peers():
+            models.update(r.supported_models)
+        return web.json_response(
+            {"models": [{"name": m, "model": m} for m in sorted(models)]})

@@ -1,0 +1,200 @@
+"""Unified peer — one struct serves both worker and consumer roles, flipped
+by worker_mode (reference parity: pkg/peer/peer.go:42-525).
+
+Serves the inference protocol (length-prefixed protobuf BaseMessage,
+5 s read deadline, errors stringified into GenerateResponse — peer.go:190-281)
+and the metadata protocol (Resource JSON then close, EOF-delimited —
+peer.go:284-316); runs the advertise loop (Provide every 1 s, peer.go:450-504)
+and metadata refresh (peer.go:361-389). Worker capabilities are real HIP
+device properties + measured throughput, not the reference's hardcoded
+"RTX 4090"/150 tok/s (peer.go:319-343)."""
+
+from __future__ import annotations
+
+import asyncio
+import time
+
+from ..config import Config
+from ..engine.api import EngineBase, RollingRate
+from ..keys import load_peer_id
+from ..logutil import new_app_logger
+from ..version import __version__, commit_hash
+from . import pb
+from .discovery import Discovery
+from .peermanager import PeerManager
+from .resource import Resource
+from .wire import (NAMESPACE, PROTO_INFERENCE, PROTO_METADATA,
+                   accept_protocol, read_frame, write_frame)
+
+
+class Peer:
+    def __init__(self, cfg: Config, worker_mode: bool,
+                 engines: dict[str, EngineBase] | None = None):
+        self.cfg = cfg
+        self.worker_mode = worker_mode
+        component = "worker" if worker_mode else "consumer"
+        self.peer_id, _ = load_peer_id(component, cfg.key_path)
+        self.log = new_app_logger(f"peer.{component}", cfg.verbose)
+        self.engines: dict[str, EngineBase] = engines or {}
+        self.discovery = Discovery(
+            cfg.bootstrap_peers,
+            metadata_timeout=cfg.intervals.metadata_timeout,
+            metadata_max_age=cfg.intervals.metadata_max_age,
+            log=self.log)
+        self.peer_manager = PeerManager(self.discovery, cfg.intervals,
+                                        log=self.log, self_id=self.peer_id)
+        self.rate = RollingRate()
+        self._active_requests = 0
+        self.resource = Resource(peer_id=self.peer_id,
+                                 worker_mode=worker_mode,
+                                 version=f"{__version__}+{commit_hash()}")
+        self._server: asyncio.base_events.Server | None = None
+        self._tasks: list[asyncio.Task] = []
+        self.port: int | None = None
+        self.requests_served = 0
+
+    # ----------------------------------------------------------- lifecycle
+
+    async def start(self) -> None:
+        self._server = await asyncio.start_server(
+            self._on_conn, self.cfg.listen_host, self.cfg.listen_port)
+        self.port = self._server.sockets[0].getsockname()[1]
+        self.update_metadata()
+        await self.peer_manager.start()
+        self._tasks = [
+            asyncio.create_task(self._advertise_loop()),
+            asyncio.create_task(self._metadata_update_loop()),
+        ]
+        self.log.info("peer %s (%s) listening on :%d, bootstrap=%s",
+                      self.peer_id, "worker" if self.worker_mode else
+                      "consumer", self.port, self.cfg.bootstrap_peers)
+
+    async def stop(self) -> None:
+        for t in self._tasks:
+            t.cancel()
+        for t in self._tasks:
+            try:
+                await t
+            except (asyncio.CancelledError, Exception):
+                pass
+        self._tasks.clear()
+        await self.peer_manager.stop()
+        await self.discovery.close()
+        if self._server:
+            self._server.close()
+            await self._server.wait_closed()
+        for e in self.engines.values():
+            await e.close()
+
+    @property
+    def addrs(self) -> list[str]:
+        host = "127.0.0.1" if self.cfg.listen_host in ("0.0.0.0", "::") \
+            else self.cfg.listen_host
+        return [f"{host}:{self.port}"]
+
+    # ------------------------------------------------------------ metadata
+
+    def update_metadata(self) -> None:
+        """Fill the advertised Resource from live engine state
+        (reference UpdateMetadata, peer.go:319-343 — but measured)."""
+        r = self.resource
+        if self.worker_mode and self.engines:
+            r.supported_models = sorted(self.engines.keys())
+            any_engine = next(iter(self.engines.values()))
+            r.gpu_model = any_engine.gpu_model()
+            r.vram_gb = any_engine.vram_gb()
+            measured = self.rate.rate()
+            r.tokens_throughput = measured if measured > 0 else \
+                max(any_engine.throughput(), 1.0)
+            r.load = min(1.0, self._active_requests / 4.0)
+        else:
+            r.supported_models = []
+            r.tokens_throughput = 0.0
+            r.load = 0.0
+        r.addrs = self.addrs
+        r.touch()
+
+    async def _metadata_update_loop(self) -> None:
+        while True:
+            await asyncio.sleep(self.cfg.intervals.metadata_update)
+            self.update_metadata()
+
+    async def _advertise_loop(self) -> None:
+        # reference: Provide namespace CID every 1 s (peer.go:450-504); also
+        # per-model namespaces (AdvertiseModel, discovery.go:144-166).
+        while True:
+            try:
+                await self.discovery.advertise(self.peer_id, self.addrs,
+                                               NAMESPACE)
+                if self.worker_mode:
+                    for model in self.engines:
+                        await self.discovery.advertise(
+                            self.peer_id, self.addrs,
+                            f"{NAMESPACE}/model/{model}")
+            except Exception as e:  # noqa: BLE001
+                self.log.debug("advertise failed: %s", e)
+            await asyncio.sleep(self.cfg.intervals.advertise)
+
+    # -------------------------------------------------------------- serving
+
+    async def _on_conn(self, reader: asyncio.StreamReader,
+                       writer: asyncio.StreamWriter) -> None:
+        try:
+            proto = await accept_protocol(reader)
+            if proto == PROTO_METADATA:
+                self.update_metadata()
+                writer.write(self.resource.to_json().encode("utf-8"))
+                await writer.drain()
+                writer.write_eof()
+                return
+            if proto == PROTO_INFERENCE:
+                await self._handle_inference(reader, writer)
+                return
+            self.log.warning("unknown protocol %r", proto)
+        except Exception as e:  # noqa: BLE001
+            self.log.debug("conn error: %s", e)
+        finally:
+            try:
+                writer.close()
+            except Exception:
+                pass
+
+    async def _handle_inference(self, reader, writer) -> None:
+        # read deadline parity: 5 s (peer.go:259-271)
+        frame = await read_frame(reader, timeout=5.0)
+        msg = pb.BaseMessage.decode(frame)
+        if not self.worker_mode:
+            resp = pb.response_message("", "Error: peer is not a worker",
+                                       self.peer_id, done_reason="error")
+            await write_frame(writer, resp.encode())
+            return
+        req = msg.generate_request
+        if req is None:
+            resp = pb.response_message("", "Error: no request in message",
+                                       self.peer_id, done_reason="error")
+            await write_frame(writer, resp.encode())
+            return
+        t0 = time.monotonic_ns()
+        self._active_requests += 1
+        try:
+            engine = self.engines.get(req.model)
+            if engine is None:
+                raise KeyError(f"model {req.model!r} not served here "
+                               f"(have {sorted(self.engines)})")
+            result = await engine.generate(req.prompt)
+            self.rate.add(result.tokens_generated)
+            self.requests_served += 1
+            resp = pb.response_message(
+                req.model, result.text, self.peer_id,
+                done_reason=result.done_reason,
+                total_duration_ns=time.monotonic_ns() - t0)
+        except Exception as e:  # noqa: BLE001 — reference stringifies errors
+            resp = pb.response_message(req.model, f"Error: {e}", self.peer_id,
+                                       done_reason="error",
+                                       total_duration_ns=time.monotonic_ns() - t0)
+        finally:
+            self._active_requests -= 1
+        await write_frame(writer, resp.encode())
+
+    def is_dht_connected(self) -> bool:
+        return bool(self.cfg.bootstrap_peers)

@@ -54,20 +54,23 @@ def _random_blocks(t: GGMLType, n_elems: int, rng: np.random.Generator,
     """Random quantized bytes for n_elems weights with constant scale d."""
     if t == GGMLType.Q4_K:
         nb = n_elems // QK_K
-        out = rng.integers(0, 256, size=(nb, Q4_K_BLOCK_BYTES), dtype=np.uint8)
+        out = np.frombuffer(rng.bytes(nb * Q4_K_BLOCK_BYTES),
+                            dtype=np.uint8).reshape(nb, -1).copy()
         d16 = np.float16(d); dmin16 = np.float16(d * 7.5)
         out[:, 0:2] = np.frombuffer(d16.tobytes(), dtype=np.uint8)
         out[:, 2:4] = np.frombuffer(dmin16.tobytes(), dtype=np.uint8)
         return out.reshape(-1)
     if t == GGMLType.Q6_K:
         nb = n_elems // QK_K
-        out = rng.integers(0, 256, size=(nb, Q6_K_BLOCK_BYTES), dtype=np.uint8)
+        out = np.frombuffer(rng.bytes(nb * Q6_K_BLOCK_BYTES),
+                            dtype=np.uint8).reshape(nb, -1).copy()
         d16 = np.float16(d)
         out[:, 208:210] = np.frombuffer(d16.tobytes(), dtype=np.uint8)
         return out.reshape(-1)
     if t == GGMLType.Q8_0:
         nb = n_elems // Q8_0_BLOCK
-        out = rng.integers(0, 256, size=(nb, Q8_0_BLOCK_BYTES), dtype=np.uint8)
+        out = np.frombuffer(rng.bytes(nb * Q8_0_BLOCK_BYTES),
+                            dtype=np.uint8).reshape(nb, -1).copy()
         d16 = np.float16(d)
         out[:, 0:2] = np.frombuffer(d16.tobytes(), dtype=np.uint8)
         return out.reshape(-1)

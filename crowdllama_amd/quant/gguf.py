@@ -150,7 +150,7 @@ class GGUFWriter:
                 if off > pos:
                     f.write(b"\x00" * (off - pos))
                     pos = off
-                f.write(raw.tobytes())
+                f.write(memoryview(raw))
                 pos += raw.nbytes
 
 
