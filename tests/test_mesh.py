@@ -1,0 +1,145 @@
+"""Mesh component tests on loopback: rendezvous server, discovery,
+peer manager scheduling/health (reference parity: dht_test.go +
+peermanager semantics)."""
+
+import asyncio
+import time
+
+import pytest
+
+from crowdllama_amd.config import Config
+from crowdllama_amd.mesh.dhtnode import DHTServer
+from crowdllama_amd.mesh.discovery import Discovery, RendezvousClient
+from crowdllama_amd.mesh.peermanager import PeerManager
+from crowdllama_amd.mesh.resource import Resource
+
+
+def run(coro):
+    return asyncio.run(coro)
+
+
+@pytest.fixture()
+def cfg():
+    c = Config(test_mode=True, listen_host="127.0.0.1")
+    return c
+
+
+def test_dht_server_start_stop(cfg):
+    async def go():
+        srv = DHTServer(cfg, "CLADHT")
+        port = await srv.start("127.0.0.1", 0)
+        assert port > 0
+        cli = RendezvousClient(f"127.0.0.1:{port}")
+        assert await cli.ping()
+        await cli.close()
+        await srv.stop()
+    run(go())
+
+
+def test_provide_and_find(cfg):
+    async def go():
+        srv = DHTServer(cfg, "CLADHT")
+        port = await srv.start("127.0.0.1", 0)
+        cli = RendezvousClient(f"127.0.0.1:{port}")
+        ok = await cli.provide("CLAWORKER1", ["127.0.0.1:5001"])
+        assert ok
+        provs = await cli.find_providers()
+        assert any(p["peer_id"] == "CLAWORKER1" for p in provs)
+        addrs = await cli.find_peer("CLAWORKER1")
+        assert addrs == ["127.0.0.1:5001"]
+        assert await cli.find_peer("CLANOBODY") is None
+        # model namespace
+        await cli.provide("CLAWORKER1", ["127.0.0.1:5001"],
+                          ns="crowdllama-ns/model/llama3-8b")
+        provs = await cli.find_providers(ns="crowdllama-ns/model/llama3-8b")
+        assert len(provs) == 1
+        await cli.close()
+        await srv.stop()
+    run(go())
+
+
+def test_provider_limit(cfg):
+    async def go():
+        srv = DHTServer(cfg, "CLADHT")
+        port = await srv.start("127.0.0.1", 0)
+        cli = RendezvousClient(f"127.0.0.1:{port}")
+        for i in range(15):
+            await cli.provide(f"CLAW{i}", [f"127.0.0.1:{5000 + i}"])
+        provs = await cli.find_providers(limit=10)
+        assert len(provs) == 10  # reference: FindProvidersAsync(cid, 10)
+        await cli.close()
+        await srv.stop()
+    run(go())
+
+
+def _mk_resource(pid, model="m1", thr=100.0, load=0.0, worker=True):
+    r = Resource(peer_id=pid, supported_models=[model],
+                 tokens_throughput=thr, load=load, worker_mode=worker,
+                 addrs=["127.0.0.1:1"])
+    r.touch()
+    return r
+
+
+def test_find_best_worker_scoring(cfg):
+    """Scheduler maximizes throughput/(1+load) (manager.go:338-387)."""
+    async def go():
+        disco = Discovery([], log=None)
+        pm = PeerManager(disco, cfg.intervals)
+        await pm.add_or_update_peer(_mk_resource("A", thr=100, load=0.0))
+        await pm.add_or_update_peer(_mk_resource("B", thr=300, load=2.0))
+        await pm.add_or_update_peer(_mk_resource("C", thr=150, load=0.2))
+        # scores: A=100, B=100, C=125 -> C
+        best = pm.find_best_worker("m1")
+        assert best.peer_id == "C"
+        # model filter
+        await pm.add_or_update_peer(_mk_resource("D", model="m2", thr=999))
+        assert pm.find_best_worker("m1").peer_id == "C"
+        assert pm.find_best_worker("m2").peer_id == "D"
+        # consumers never selected
+        await pm.add_or_update_peer(_mk_resource("E", thr=9999, worker=False))
+        assert pm.find_best_worker("m1").peer_id == "C"
+        # no worker for unknown model
+        assert pm.find_best_worker("nope") is None
+    run(go())
+
+
+def test_tombstones_prevent_readd(cfg):
+    async def go():
+        disco = Discovery([], log=None)
+        pm = PeerManager(disco, cfg.intervals)
+        await pm.add_or_update_peer(_mk_resource("A"))
+        await pm.remove_peer("A")
+        assert "A" not in pm.peers
+        await pm.add_or_update_peer(_mk_resource("A"))
+        assert "A" not in pm.peers  # tombstoned
+        # expire the tombstone manually and re-add
+        pm.recently_removed["A"] = time.time() - cfg.intervals.tombstone - 1
+        await pm.add_or_update_peer(_mk_resource("A"))
+        assert "A" in pm.peers
+    run(go())
+
+
+def test_stale_cleanup(cfg):
+    async def go():
+        disco = Discovery([], log=None)
+        pm = PeerManager(disco, cfg.intervals)
+        await pm.add_or_update_peer(_mk_resource("A"))
+        pm.peers["A"].last_seen = time.time() - cfg.intervals.stale_timeout - 1
+        await pm.start()
+        try:
+            deadline = time.time() + 10
+            while "A" in pm.peers and time.time() < deadline:
+                await asyncio.sleep(0.1)
+            assert "A" not in pm.peers
+        finally:
+            await pm.stop()
+    run(go())
+
+
+def test_self_never_added(cfg):
+    async def go():
+        disco = Discovery([], log=None)
+        pm = PeerManager(disco, cfg.intervals, self_id="ME")
+        await pm.add_or_update_peer(_mk_resource("ME"))
+        assert "ME" not in pm.peers
+    run(go())
