@@ -16,13 +16,9 @@ void launch_gemv(const WTensor&, int pre, const float* xin, const float* gw,
                  hipStream_t);
 void launch_embed(const WTensor&, const int32_t* ids, float* x, int B,
                   hipStream_t);
-void launch_rope_append(float* qkv, const float* inv_freq,
-                        const int32_t* page_table, uint16_t* kv_pool,
-                        const int32_t* n_past, int B, int NH, int NKV, int D,
-                        int page_size, int max_pages, int64_t page_stride,
-                        hipStream_t);
-void launch_attn_decode(const float* qkv, const int32_t* page_table,
-                        const uint16_t* kv_pool, const int32_t* n_past,
+void launch_attn_decode(const float* qkv, const float* inv_freq,
+                        const int32_t* page_table,
+                        uint16_t* kv_pool, const int32_t* n_past,
                         float* part_o, float* part_ml, int B, int NH, int NKV,
                         int D, int S, int page_size, int max_pages,
                         int64_t page_stride, float scale, hipStream_t);
@@ -301,7 +297,7 @@ void Engine::alloc_state() {
     const int B = cfg_.batch, H = meta_.hidden, V = meta_.vocab;
     const int NH = meta_.heads, NKV = meta_.kv_heads, D = meta_.head_dim;
     const int F = meta_.ffn;
-    attn_splits_ = std::max(1, std::min(64, 512 / std::max(1, B * NKV)));
+    attn_splits_ = std::max(1, std::min(32, 256 / std::max(1, B * NKV)));
     max_pages_ = (cfg_.max_seq + cfg_.page_size - 1) / cfg_.page_size;
     // pool layout: [page][layer][kvh][2][page_size][D] bf16 — one pool, all
     // layers; per-layer base pointer passed at launch.
@@ -369,11 +365,8 @@ void Engine::step(hipStream_t s) {
         for (auto& pt : L.qkv.parts)
             launch_gemv(pt.w, PRE_RMS, x_, L.attn_norm, nullptr,
                         qkv_ + pt.row_off, B, (int)L.qkv.n_total, eps, s);
-        launch_rope_append(qkv_, inv_freq_, page_table_, kv_layer, n_past_,
-                           B, NH, NKV, D, cfg_.page_size, max_pages_,
-                           page_stride_, s);
-        launch_attn_decode(qkv_, page_table_, kv_layer, n_past_, part_o_,
-                           part_ml_, B, NH, NKV, D, attn_splits_,
+        launch_attn_decode(qkv_, inv_freq_, page_table_, kv_layer, n_past_,
+                           part_o_, part_ml_, B, NH, NKV, D, attn_splits_,
                            cfg_.page_size, max_pages_, page_stride_, scale, s);
         launch_attn_combine(part_o_, part_ml_, attn_out_, B, NH, attn_splits_,
                             D, s);

@@ -65,7 +65,12 @@ template <DT W>
 __device__ __forceinline__ void decode_chunk(
     const uint8_t* __restrict__ qs_row, const uint8_t* __restrict__ hdr_row,
     int c, float* __restrict__ w, int* k0) {
-    const uint4 qv = reinterpret_cast<const uint4*>(qs_row)[c];
+    // weights stream through once per step: non-temporal (bypass L1/L2 keep)
+    typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
+    const u32x4 qv4 = __builtin_nontemporal_load(
+        reinterpret_cast<const u32x4*>(qs_row) + c);
+    uint4 qv;
+    qv.x = qv4.x; qv.y = qv4.y; qv.z = qv4.z; qv.w = qv4.w;
     const uint32_t dw[4] = {qv.x, qv.y, qv.z, qv.w};
     if constexpr (W == DT::DQ4K) {
         const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
@@ -165,21 +170,33 @@ __global__ __launch_bounds__(256) void k_gemv(
     float* red = x_lds + (size_t)B * K;                      // [256]
 
     const int tid = threadIdx.x;
-    // ---- stage activations ----
+    // ---- stage activations (vectorized float4: K is always %4==0) ----
+    const int K4 = K >> 2;
     for (int b = 0; b < B; b++) {
         float ss = 0.f;
+        float4* xl4 = reinterpret_cast<float4*>(x_lds + (size_t)b * K);
         if constexpr (P == PRE_SILU) {
-            const float* gub = xin + (size_t)b * 2 * K;
-            for (int k = tid; k < K; k += 256) {
-                const float g = gub[k], u = gub[K + k];
-                x_lds[(size_t)b * K + k] = (g / (1.f + __expf(-g))) * u;
+            const float4* g4 = reinterpret_cast<const float4*>(
+                xin + (size_t)b * 2 * K);
+            const float4* u4 = reinterpret_cast<const float4*>(
+                xin + (size_t)b * 2 * K + K);
+            for (int k = tid; k < K4; k += 256) {
+                const float4 g = g4[k], u = u4[k];
+                float4 o;
+                o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+                o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+                o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+                o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
+                xl4[k] = o;
             }
         } else {
-            const float* xb = xin + (size_t)b * K;
-            for (int k = tid; k < K; k += 256) {
-                const float v = xb[k];
-                x_lds[(size_t)b * K + k] = v;
-                if constexpr (P == PRE_RMS) ss += v * v;
+            const float4* x4 = reinterpret_cast<const float4*>(
+                xin + (size_t)b * K);
+            for (int k = tid; k < K4; k += 256) {
+                const float4 v = x4[k];
+                xl4[k] = v;
+                if constexpr (P == PRE_RMS)
+                    ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
             }
         }
         if constexpr (P == PRE_RMS) {
@@ -192,8 +209,15 @@ __global__ __launch_bounds__(256) void k_gemv(
             }
             const float inv = rsqrtf(red[0] / (float)K + eps);
             __syncthreads();
-            for (int k = tid; k < K; k += 256)
-                x_lds[(size_t)b * K + k] *= inv * gw[k];
+            float4* xl4 = reinterpret_cast<float4*>(x_lds + (size_t)b * K);
+            const float4* gw4 = reinterpret_cast<const float4*>(gw);
+            for (int k = tid; k < K4; k += 256) {
+                const float4 g = gw4[k];
+                float4 v = xl4[k];
+                v.x *= inv * g.x; v.y *= inv * g.y;
+                v.z *= inv * g.z; v.w *= inv * g.w;
+                xl4[k] = v;
+            }
         }
     }
     __syncthreads();
@@ -216,17 +240,28 @@ __global__ __launch_bounds__(256) void k_gemv(
         decode_chunk<W>(qs_row, hdr_row, c, w, &k0);
         #pragma unroll 2
         for (int b = 0; b < B; b++) {
-            const float* xb = x_lds + (size_t)b * K;
+            // k0 is a multiple of 16 weights for every dtype's chunk map,
+            // so float4 LDS reads (ds_read_b128) are aligned.
+            const float4* xb4 = reinterpret_cast<const float4*>(
+                x_lds + (size_t)b * K + k0);
             float s = 0.f;
             if constexpr (W == DT::DQ4K) {
-                // lo run at k0, hi run at k0+32; fold the min term via xsum16
                 #pragma unroll
-                for (int t = 0; t < 16; t++) s += w[t] * xb[k0 + t];
-                #pragma unroll
-                for (int t = 0; t < 16; t++) s += w[16 + t] * xb[k0 + 32 + t];
+                for (int t4 = 0; t4 < 4; t4++) {
+                    const float4 xl = xb4[t4];
+                    const float4 xh = xb4[8 + t4];
+                    s += w[t4 * 4 + 0] * xl.x + w[t4 * 4 + 1] * xl.y
+                       + w[t4 * 4 + 2] * xl.z + w[t4 * 4 + 3] * xl.w;
+                    s += w[16 + t4 * 4 + 0] * xh.x + w[16 + t4 * 4 + 1] * xh.y
+                       + w[16 + t4 * 4 + 2] * xh.z + w[16 + t4 * 4 + 3] * xh.w;
+                }
             } else {
                 #pragma unroll
-                for (int t = 0; t < WPC; t++) s += w[t] * xb[k0 + t];
+                for (int t4 = 0; t4 < WPC / 4; t4++) {
+                    const float4 xv = xb4[t4];
+                    s += w[t4 * 4 + 0] * xv.x + w[t4 * 4 + 1] * xv.y
+                       + w[t4 * 4 + 2] * xv.z + w[t4 * 4 + 3] * xv.w;
+                }
             }
             acc[b] += s;
         }
@@ -273,65 +308,26 @@ __global__ __launch_bounds__(256) void k_embed(
 
 // -------------------------------------------------------- RoPE + append
 
-// grid (KVH, B); block 256. Applies NORM-style RoPE (adjacent pairs) to the
-// G q-heads and 1 k-head of this kv-group in the qkv buffer (in place, f32),
-// then appends the roped k and raw v to the paged bf16 KV cache at position
-// n_past[b].
-__global__ __launch_bounds__(256) void k_rope_append(
-    float* __restrict__ qkv, const float* __restrict__ inv_freq,
-    const int32_t* __restrict__ page_table, uint16_t* __restrict__ kv_pool,
-    const int32_t* __restrict__ n_past,
-    int NH, int NKV, int D, int G, int page_size, int max_pages,
-    int64_t page_stride) {
-    const int kvh = blockIdx.x;
-    const int b = blockIdx.y;
-    const int pos = n_past[b];
-    float* qkv_b = qkv + (size_t)b * (NH + 2 * NKV) * D;
-    float* kh = qkv_b + (size_t)(NH + kvh) * D;
-    const float* vh = qkv_b + (size_t)(NH + NKV + kvh) * D;
-    const int half = D / 2;
-    // rope: G q-heads + 1 k-head
-    for (int idx = threadIdx.x; idx < (G + 1) * half; idx += 256) {
-        const int hsel = idx / half, i = idx % half;
-        float* p = (hsel < G) ? (qkv_b + (size_t)(kvh * G + hsel) * D) : kh;
-        const float ang = (float)pos * inv_freq[i];
-        float s, c;
-        __sincosf(ang, &s, &c);
-        const float x0 = p[2 * i], x1 = p[2 * i + 1];
-        p[2 * i] = x0 * c - x1 * s;
-        p[2 * i + 1] = x0 * s + x1 * c;
-    }
-    __syncthreads();
-    // append k (roped) and v to the cache page
-    const int page = page_table[(size_t)b * max_pages + pos / page_size];
-    const int off = pos % page_size;
-    // pool layout: [page][kvh][2][page_size][D] bf16
-    uint16_t* kdst = kv_pool + (int64_t)page * page_stride
-                     + ((int64_t)kvh * 2 + 0) * page_size * D + (int64_t)off * D;
-    uint16_t* vdst = kv_pool + (int64_t)page * page_stride
-                     + ((int64_t)kvh * 2 + 1) * page_size * D + (int64_t)off * D;
-    for (int d = threadIdx.x; d < D; d += 256) {
-        kdst[d] = f32_to_bf16_bits(kh[d]);
-        vdst[d] = f32_to_bf16_bits(vh[d]);
-    }
-}
-
-// ------------------------------------------------- decode attention (GQA)
-
 // grid (S, KVH, B); block 256 = 4 waves; each 16-lane quarter owns a full
 // (m, l, o[D]) online-softmax accumulator over its position subsequence,
 // for each of the G query heads of this kv group. D = 16*DPL (64 or 128).
+// RoPE is fused: every workgroup ropes its q fragment in-register; the
+// split that owns position n_past computes the current token's roped k (and
+// raw v) from the qkv buffer, uses it in the online softmax, and appends it
+// to the cache page for future steps. No separate rope/append kernel.
 template <int G, int D>
 __global__ __launch_bounds__(256) void k_attn_decode(
-    const float* __restrict__ qkv, const int32_t* __restrict__ page_table,
-    const uint16_t* __restrict__ kv_pool, const int32_t* __restrict__ n_past,
+    const float* __restrict__ qkv, const float* __restrict__ inv_freq,
+    const int32_t* __restrict__ page_table,
+    uint16_t* __restrict__ kv_pool, const int32_t* __restrict__ n_past,
     float* __restrict__ part_o,   // [B][NH][S][D]
     float* __restrict__ part_ml,  // [B][NH][S][2]
     int NH, int NKV, int S, int page_size, int max_pages,
     int64_t page_stride, float scale) {
     constexpr int DPL = D / 16;   // dims per lane (bf16: 2*DPL bytes)
     const int s = blockIdx.x, kvh = blockIdx.y, b = blockIdx.z;
-    const int len = n_past[b] + 1;
+    const int pos = n_past[b];
+    const int len = pos + 1;
     const int tid = threadIdx.x;
     const int wave = tid >> 6, lane = tid & 63;
     const int quarter = lane >> 4, qlane = lane & 15;
@@ -343,13 +339,27 @@ __global__ __launch_bounds__(256) void k_attn_decode(
     const int start = s * chunk;
     const int end = min(start + chunk, len);
 
+    // rope angles for this lane's dim pairs at the current position
+    float cs[DPL / 2], sn[DPL / 2];
+    #pragma unroll
+    for (int j = 0; j < DPL / 2; j++) {
+        const float ang = (float)pos * inv_freq[d0 / 2 + j];
+        __sincosf(ang, &sn[j], &cs[j]);
+    }
+
     const float* qkv_b = qkv + (size_t)b * (NH + 2 * NKV) * D;
     float qf[G][DPL];
     #pragma unroll
     for (int g = 0; g < G; g++) {
         const float* qh = qkv_b + (size_t)(kvh * G + g) * D + d0;
         #pragma unroll
-        for (int j = 0; j < DPL; j++) qf[g][j] = qh[j] * scale;
+        for (int j = 0; j < DPL; j++) qf[g][j] = qh[j];
+        #pragma unroll
+        for (int j = 0; j < DPL / 2; j++) {  // in-register RoPE
+            const float x0 = qf[g][2 * j], x1 = qf[g][2 * j + 1];
+            qf[g][2 * j] = (x0 * cs[j] - x1 * sn[j]) * scale;
+            qf[g][2 * j + 1] = (x0 * sn[j] + x1 * cs[j]) * scale;
+        }
     }
 
     float m[G], l[G], o[G][DPL];
@@ -360,7 +370,8 @@ __global__ __launch_bounds__(256) void k_attn_decode(
         for (int j = 0; j < DPL; j++) o[g][j] = 0.f;
     }
 
-    for (int p = start + sub; p < end; p += 16) {
+    const int cache_end = min(end, pos);  // cached positions only
+    for (int p = start + sub; p < cache_end; p += 16) {
         const int page = page_table[(size_t)b * max_pages + p / page_size];
         const uint16_t* kp = kv_pool + (int64_t)page * page_stride
                              + ((int64_t)kvh * 2 + 0) * page_size * D
@@ -410,6 +421,63 @@ __global__ __launch_bounds__(256) void k_attn_decode(
         }
     }
 
+    // current token: the quarter whose subsequence covers `pos` computes
+    // roped k / raw v from the qkv buffer, folds it into its accumulator,
+    // and appends both to the cache page.
+    if (pos >= start && pos < end && ((pos - start) & 15) == sub) {
+        const float* kh = qkv_b + (size_t)(NH + kvh) * D + d0;
+        const float* vh = qkv_b + (size_t)(NH + NKV + kvh) * D + d0;
+        float kf[DPL], vf[DPL];
+        #pragma unroll
+        for (int j = 0; j < DPL; j++) { kf[j] = kh[j]; vf[j] = vh[j]; }
+        #pragma unroll
+        for (int j = 0; j < DPL / 2; j++) {
+            const float x0 = kf[2 * j], x1 = kf[2 * j + 1];
+            kf[2 * j] = x0 * cs[j] - x1 * sn[j];
+            kf[2 * j + 1] = x0 * sn[j] + x1 * cs[j];
+        }
+        // append to cache (bf16)
+        const int page = page_table[(size_t)b * max_pages + pos / page_size];
+        uint16_t* kdst = kv_pool + (int64_t)page * page_stride
+                         + ((int64_t)kvh * 2 + 0) * page_size * D
+                         + (int64_t)(pos % page_size) * D + d0;
+        uint16_t* vdst = kv_pool + (int64_t)page * page_stride
+                         + ((int64_t)kvh * 2 + 1) * page_size * D
+                         + (int64_t)(pos % page_size) * D + d0;
+        #pragma unroll
+        for (int j = 0; j < DPL; j++) {
+            kdst[j] = f32_to_bf16_bits(kf[j]);
+            vdst[j] = f32_to_bf16_bits(vf[j]);
+        }
+        // fold into the online softmax (k/v as bf16-rounded, matching what
+        // future steps will read from the cache)
+        #pragma unroll
+        for (int j = 0; j < DPL; j++) {
+            kf[j] = bf16_bits_to_f32(kdst[j]);
+            vf[j] = bf16_bits_to_f32(vdst[j]);
+        }
+        float sc[G];
+        #pragma unroll
+        for (int g = 0; g < G; g++) {
+            float d = 0.f;
+            #pragma unroll
+            for (int j = 0; j < DPL; j++) d += qf[g][j] * kf[j];
+            #pragma unroll
+            for (int off = 1; off < 16; off <<= 1) d += __shfl_xor(d, off, 64);
+            sc[g] = d;
+        }
+        #pragma unroll
+        for (int g = 0; g < G; g++) {
+            const float mn = fmaxf(m[g], sc[g]);
+            const float alpha = __expf(m[g] - mn);
+            const float w = __expf(sc[g] - mn);
+            l[g] = l[g] * alpha + w;
+            #pragma unroll
+            for (int j = 0; j < DPL; j++) o[g][j] = o[g][j] * alpha + w * vf[j];
+            m[g] = mn;
+        }
+    }
+
     // combine the 16 sub-accumulators through LDS
     __shared__ __attribute__((aligned(16))) float sm[16 * G];
     __shared__ __attribute__((aligned(16))) float sl[16 * G];
@@ -446,20 +514,25 @@ __global__ __launch_bounds__(256) void k_attn_decode(
 }
 
 // grid (B*NH); block 128. Combine S split-KV partials into attn_out.
+// ml pairs staged to LDS first (one coalesced read), then the per-dim o
+// reduction issues S independent loads per thread.
 __global__ __launch_bounds__(128) void k_attn_combine(
     const float* __restrict__ part_o, const float* __restrict__ part_ml,
     float* __restrict__ attn_out, int NH, int S, int D) {
     const int bh = blockIdx.x;  // b * NH + head
     const int d = threadIdx.x;
+    __shared__ __attribute__((aligned(16))) float ml[2 * 64];  // S <= 64
+    for (int i = d; i < 2 * S; i += 128)
+        ml[i] = part_ml[(size_t)bh * S * 2 + i];
+    __syncthreads();
     float mstar = -1e30f;
+    for (int s = 0; s < S; s++) mstar = fmaxf(mstar, ml[2 * s]);
+    float denom = 0.f;
+    for (int s = 0; s < S; s++) denom += __expf(ml[2 * s] - mstar) * ml[2 * s + 1];
+    float osum = 0.f;
+    #pragma unroll 4
     for (int s = 0; s < S; s++)
-        mstar = fmaxf(mstar, part_ml[((size_t)bh * S + s) * 2]);
-    float denom = 0.f, osum = 0.f;
-    for (int s = 0; s < S; s++) {
-        const float e = __expf(part_ml[((size_t)bh * S + s) * 2] - mstar);
-        denom += e * part_ml[((size_t)bh * S + s) * 2 + 1];
-        osum += e * part_o[((size_t)bh * S + s) * D + d];
-    }
+        osum += __expf(ml[2 * s] - mstar) * part_o[((size_t)bh * S + s) * D + d];
     attn_out[(size_t)bh * D + d] = osum / denom;
 }
 
@@ -567,19 +640,9 @@ void launch_embed(const WTensor& w, const int32_t* ids, float* x, int B,
     #undef EMBED_CASE
 }
 
-void launch_rope_append(float* qkv, const float* inv_freq,
-                        const int32_t* page_table, uint16_t* kv_pool,
-                        const int32_t* n_past, int B, int NH, int NKV, int D,
-                        int page_size, int max_pages, int64_t page_stride,
-                        hipStream_t stream) {
-    const int G = NH / NKV;
-    hipLaunchKernelGGL(k_rope_append, dim3(NKV, B), dim3(256), 0, stream,
-                       qkv, inv_freq, page_table, kv_pool, n_past,
-                       NH, NKV, D, G, page_size, max_pages, page_stride);
-}
-
-void launch_attn_decode(const float* qkv, const int32_t* page_table,
-                        const uint16_t* kv_pool, const int32_t* n_past,
+void launch_attn_decode(const float* qkv, const float* inv_freq,
+                        const int32_t* page_table,
+                        uint16_t* kv_pool, const int32_t* n_past,
                         float* part_o, float* part_ml, int B, int NH, int NKV,
                         int D, int S, int page_size, int max_pages,
                         int64_t page_stride, float scale, hipStream_t stream) {
@@ -587,7 +650,7 @@ void launch_attn_decode(const float* qkv, const int32_t* page_table,
     dim3 grid(S, NKV, B), block(256);
     #define ATTN_CASE(GV, DV)                                                   \
         hipLaunchKernelGGL((k_attn_decode<GV, DV>), grid, block, 0, stream,     \
-            qkv, page_table, kv_pool, n_past, part_o, part_ml,                  \
+            qkv, inv_freq, page_table, kv_pool, n_past, part_o, part_ml,        \
             NH, NKV, S, page_size, max_pages, page_stride, scale)
     #define ATTN_D(GV)                                                          \
         do { if (D == 128) ATTN_CASE(GV, 128);                                  \
