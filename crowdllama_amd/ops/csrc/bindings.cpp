@@ -12,6 +12,7 @@ using namespace cla;
 namespace cla {
 void launch_gemv_test(const void*, const void*, const float*, const float*,
                       float*, int, int, int, int, int, size_t, size_t);
+void launch_mfma_probe_test(const uint16_t*, const uint16_t*, float*);
 }
 
 PYBIND11_MODULE(_core, m) {
@@ -85,6 +86,12 @@ PYBIND11_MODULE(_core, m) {
             d["rms_eps"] = mm.rms_eps;
             return d;
         });
+
+    m.def("test_mfma_probe", [](py::array_t<uint16_t> A, py::array_t<uint16_t> B) {
+        py::array_t<float> C({16, 16});
+        launch_mfma_probe_test(A.data(), B.data(), C.mutable_data());
+        return C;
+    });
 
     // ---- raw kernel entry points for numerics tests (tests/test_gpu_kernels.py)
     m.def("test_gemv", [](py::array_t<uint8_t> qs, py::array_t<uint8_t> hdr,

@@ -141,3 +141,24 @@ def test_gemv_pre_silu(core):
     act = (g / (1 + np.exp(-g))) * u
     yref = act @ w.T
     np.testing.assert_allclose(y, yref, rtol=2e-4, atol=2e-4)
+
+
+def test_mfma_fragment_layout(core):
+    """Verify the assumed v_mfma_f32_16x16x32_bf16 lane mappings with
+    asymmetric inputs (transpose-detecting, guide §5.4 rule 16)."""
+    rng = np.random.default_rng(9)
+    A = rng.standard_normal((16, 32)).astype(np.float32)
+    B = rng.standard_normal((32, 16)).astype(np.float32)
+
+    def to_bf16_bits(x):
+        u = x.view(np.uint32)
+        return ((u + 0x7FFF + ((u >> 16) & 1)) >> 16).astype(np.uint16)
+
+    def bf16_val(bits):
+        return (bits.astype(np.uint32) << 16).view(np.float32)
+
+    Ab, Bb = to_bf16_bits(A), to_bf16_bits(B)
+    C = core.test_mfma_probe(np.ascontiguousarray(Ab),
+                             np.ascontiguousarray(Bb))
+    Cref = bf16_val(Ab) @ bf16_val(Bb)
+    np.testing.assert_allclose(C, Cref, rtol=1e-5, atol=1e-5)
