@@ -13,6 +13,8 @@ namespace cla {
 void launch_gemv_test(const void*, const void*, const float*, const float*,
                       float*, int, int, int, int, int, size_t, size_t);
 void launch_mfma_probe_test(const uint16_t*, const uint16_t*, float*);
+double bench_gemv(const void*, const void*, int, int, int, int, int, size_t,
+                  size_t, int);
 }
 
 PYBIND11_MODULE(_core, m) {
@@ -86,6 +88,16 @@ PYBIND11_MODULE(_core, m) {
             d["rms_eps"] = mm.rms_eps;
             return d;
         });
+
+    m.def("bench_gemv", [](int dtype, int N, int K, int B, int pre, int iters) {
+        // random weight bytes (content irrelevant for timing)
+        const DT dt = static_cast<DT>(dtype);
+        const size_t qs_bytes = (size_t)N * dqs_row_bytes(dt, K);
+        const size_t hdr_bytes = (size_t)N * dhdr_row_bytes(dt, K);
+        std::vector<uint8_t> qs(qs_bytes, 1), hdr(hdr_bytes ? hdr_bytes : 1, 1);
+        return bench_gemv(qs.data(), hdr.data(), dtype, N, K, B, pre,
+                          qs_bytes, hdr_bytes, iters);
+    });
 
     m.def("test_mfma_probe", [](py::array_t<uint16_t> A, py::array_t<uint16_t> B) {
         py::array_t<float> C({16, 16});

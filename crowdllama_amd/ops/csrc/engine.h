@@ -23,6 +23,7 @@ struct EngineConfig {
     int gen_cap = 2048;     // generated-token ring per slot
     int device = 0;
     bool use_graph = true;
+    int prefill_chunk = 1024;  // prompt tokens per prefill GEMM pass
 };
 
 struct ModelMeta {
@@ -60,9 +61,11 @@ public:
     // Reset all slots (empty KV).
     void reset();
     // Feed prompt tokens (same length for all slots in this call) through
-    // the decode path; afterwards each slot's first generated token is in
-    // gen_tokens[slot][0]. ids is [batch][len] row-major.
+    // the GEMM prefill path; afterwards each slot's first generated token is
+    // in gen_tokens[slot][0]. ids is [batch][len] row-major.
     void prefill(const std::vector<int32_t>& ids, int len);
+    // Prefill one slot (chunked GEMM path); used for serving.
+    void prefill_slot(int slot, const std::vector<int32_t>& ids);
     // Run n decode steps back-to-back (graph replays; one sync at the end).
     void decode(int n_steps);
     // Fetch generated tokens for a slot (gen_count entries).
@@ -98,7 +101,18 @@ private:
     size_t vram_bytes_ = 0;
 
     // state buffers (device)
+    void prefill_chunk_pass(int slot, int pos0, int m);
     float* x_ = nullptr;        // [B][h]
+    float* xn_ = nullptr;       // [B][h] (GEMM decode path rmsnorm out)
+    float* act_ = nullptr;      // [B][F]
+    // prefill scratch ([Mchunk] rows)
+    float* xp_ = nullptr;
+    float* xnp_ = nullptr;
+    float* qkvp_ = nullptr;
+    float* attnp_ = nullptr;
+    float* gup_ = nullptr;
+    float* actp_ = nullptr;
+    int32_t* pids_ = nullptr;   // prefill token ids
     float* qkv_ = nullptr;      // [B][(NH+2KV)*D]
     float* attn_out_ = nullptr; // [B][NH*D]
     float* gu_ = nullptr;       // [B][2F]

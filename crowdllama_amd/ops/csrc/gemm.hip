@@ -1,0 +1,511 @@
+// MFMA dequant-GEMM for gfx950: C[M,N] = X[M,K] @ W[N,K]^T (+res), where W
+// rows are quantized (Q4_K/Q6_K/Q8_0) or bf16/f16/f32. Used for prompt
+// prefill and batched decode (the GEMV kernels cover decode B<=2).
+//
+// Structure (guide §5 canonical anatomy, correctness-first tier):
+// 128x128 tile, BK=64, 256 threads = 4 waves as 2x2, each wave a 64x64
+// sub-tile = 4x4 fragments of v_mfma_f32_16x16x32_bf16. Weight tiles are
+// dequantized on the fly into LDS bf16 during staging (16-B padded rows
+// against ds_read_b128 bank conflicts). Fragment lane maps verified on
+// hardware by mfma_probe.hip / tests/test_gpu_kernels.py.
+#include "common.h"
+
+namespace cla {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+namespace {
+
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int PAD = 8;                       // bf16 elems -> 16B row pad
+constexpr int LDW = BK + PAD;                // LDS row stride (elems)
+
+__device__ __forceinline__ uint16_t f32_to_bf16b(float f) {
+    union { uint32_t u; float f; } v;
+    v.f = f;
+    return (uint16_t)((v.u + 0x7FFF + ((v.u >> 16) & 1)) >> 16);
+}
+
+__device__ __forceinline__ float f16b_to_f32(uint32_t h) {
+    __half_raw r;
+    r.x = (uint16_t)h;
+    return __half2float(*reinterpret_cast<__half*>(&r));
+}
+
+// Dequantize this thread's 32-weight slice of the W tile into `out` bf16.
+// Thread t covers W row (tile_row = t>>1), k-halves h = t&1 within [kb,kb+BK).
+// Returns false if the row is out of range (caller zero-fills).
+template <DT W>
+__device__ __forceinline__ void stage_w_slice(
+    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
+    int64_t row, int K, int kb, int h, uint16_t out[32]) {
+    if constexpr (W == DT::DQ4K) {
+        const int sb = kb >> 8;
+        const int q = (kb & 255) >> 6;
+        const int p = 2 * q + h;
+        const uint8_t* qrow = qs + row * (K / 256) * 128;
+        const uint4* hrow = reinterpret_cast<const uint4*>(
+            hdr + row * (K / 256) * 16);
+        typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
+        const u32x4 qv = __builtin_nontemporal_load(
+            reinterpret_cast<const u32x4*>(qrow) + sb * 8 + p);
+        const uint32_t dw[4] = {qv.x, qv.y, qv.z, qv.w};
+        const uint4 hd = hrow[sb];
+        const float d = f16b_to_f32(hd.x & 0xFFFF);
+        const float dmin = f16b_to_f32(hd.x >> 16);
+        const uint32_t sbytes[3] = {hd.y, hd.z, hd.w};
+        auto sbyte = [&](int i) -> uint32_t {
+            return (sbytes[i >> 2] >> ((i & 3) * 8)) & 0xFF;
+        };
+        const int j0 = 2 * q, j1 = 2 * q + 1;
+        uint32_t sc0, mn0, sc1, mn1;
+        if (q < 2) {
+            sc0 = sbyte(j0) & 63; mn0 = sbyte(j0 + 4) & 63;
+            sc1 = sbyte(j1) & 63; mn1 = sbyte(j1 + 4) & 63;
+        } else {
+            sc0 = (sbyte(j0 + 4) & 0xF) | ((sbyte(j0 - 4) >> 6) << 4);
+            mn0 = (sbyte(j0 + 4) >> 4) | ((sbyte(j0) >> 6) << 4);
+            sc1 = (sbyte(j1 + 4) & 0xF) | ((sbyte(j1 - 4) >> 6) << 4);
+            mn1 = (sbyte(j1 + 4) >> 4) | ((sbyte(j1) >> 6) << 4);
+        }
+        const float dl = d * (float)sc0, ml = dmin * (float)mn0;
+        const float dh2 = d * (float)sc1, mh = dmin * (float)mn1;
+        // lo nibbles -> klocal h*16+t (sub-block 2q), hi -> 32+h*16+t
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            const uint32_t lo = dw[j] & 0x0F0F0F0Fu;
+            const uint32_t hi = (dw[j] >> 4) & 0x0F0F0F0Fu;
+            #pragma unroll
+            for (int t = 0; t < 4; t++) {
+                out[j * 4 + t] =
+                    f32_to_bf16b(dl * (float)((lo >> (8 * t)) & 0xFF) - ml);
+                out[16 + j * 4 + t] =
+                    f32_to_bf16b(dh2 * (float)((hi >> (8 * t)) & 0xFF) - mh);
+            }
+        }
+    } else if constexpr (W == DT::DQ6K || W == DT::DQ8) {
+        const int k0 = kb + h * 32;
+        const int8_t* qrow = reinterpret_cast<const int8_t*>(qs + row * K);
+        typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
+        const u32x4 q0 = __builtin_nontemporal_load(
+            reinterpret_cast<const u32x4*>(qrow + k0));
+        const u32x4 q1 = __builtin_nontemporal_load(
+            reinterpret_cast<const u32x4*>(qrow + k0) + 1);
+        const uint32_t dw[8] = {q0.x, q0.y, q0.z, q0.w, q1.x, q1.y, q1.z, q1.w};
+        float sc[2];
+        if constexpr (W == DT::DQ6K) {
+            const uint8_t* hb = hdr + row * (K / 256) * 32 + (k0 >> 8) * 32;
+            const float d = f16b_to_f32(*reinterpret_cast<const uint16_t*>(hb));
+            const int s16 = (k0 & 255) >> 4;
+            sc[0] = d * (float)(reinterpret_cast<const int8_t*>(hb)[4 + s16]);
+            sc[1] = d * (float)(reinterpret_cast<const int8_t*>(hb)[4 + s16 + 1]);
+        } else {
+            const uint16_t* drow = reinterpret_cast<const uint16_t*>(
+                hdr + row * (K / 32) * 2);
+            sc[0] = f16b_to_f32(drow[k0 >> 5]);
+            sc[1] = sc[0];  // 32 weights span exactly one Q8 block
+        }
+        #pragma unroll
+        for (int j = 0; j < 8; j++) {
+            const float s = sc[j >> 2];
+            #pragma unroll
+            for (int t = 0; t < 4; t++) {
+                const int8_t v = (int8_t)((dw[j] >> (8 * t)) & 0xFF);
+                out[j * 4 + t] = f32_to_bf16b(s * (float)v);
+            }
+        }
+    } else if constexpr (W == DT::BF16) {
+        const uint16_t* wrow = reinterpret_cast<const uint16_t*>(qs) + row * K
+                               + kb + h * 32;
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            const uint4 v = reinterpret_cast<const uint4*>(wrow)[j];
+            const uint32_t dw[4] = {v.x, v.y, v.z, v.w};
+            #pragma unroll
+            for (int t = 0; t < 4; t++) {
+                out[j * 8 + 2 * t] = (uint16_t)(dw[t] & 0xFFFF);
+                out[j * 8 + 2 * t + 1] = (uint16_t)(dw[t] >> 16);
+            }
+        }
+    } else if constexpr (W == DT::F16) {
+        const uint16_t* wrow = reinterpret_cast<const uint16_t*>(qs) + row * K
+                               + kb + h * 32;
+        #pragma unroll
+        for (int t = 0; t < 32; t++)
+            out[t] = f32_to_bf16b(f16b_to_f32(wrow[t]));
+    } else {  // F32
+        const float* wrow = reinterpret_cast<const float*>(qs) + row * K
+                            + kb + h * 32;
+        #pragma unroll
+        for (int t = 0; t < 32; t++) out[t] = f32_to_bf16b(wrow[t]);
+    }
+}
+
+}  // namespace
+
+template <DT W>
+__global__ __launch_bounds__(256) void k_gemm(
+    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
+    const float* __restrict__ X,     // [M][K] f32
+    const float* __restrict__ res,   // [M][ldc] or null (C col-offset applied)
+    float* __restrict__ C,           // [M][ldc]
+    int M, int N, int K, int ldc) {
+    __shared__ __attribute__((aligned(16))) uint16_t Xl[BM * LDW];
+    __shared__ __attribute__((aligned(16))) uint16_t Wl[BN * LDW];
+
+    const int tid = threadIdx.x;
+    const int bn = blockIdx.x, bm = blockIdx.y;
+    const int m0 = bm * BM, n0 = bn * BN;
+    const int wid = tid >> 6, lane = tid & 63;
+    const int wm = wid >> 1, wn = wid & 1;       // 2x2 wave grid
+    const int lrow = lane & 15, lk = lane >> 4;  // fragment lane coords
+
+    f32x4 acc[4][4];
+    #pragma unroll
+    for (int i = 0; i < 4; i++)
+        #pragma unroll
+        for (int j = 0; j < 4; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    const int srow = tid >> 1, sh = tid & 1;     // staging coords
+
+    for (int kb = 0; kb < K; kb += BK) {
+        // ---- stage X tile (f32 -> bf16) ----
+        {
+            uint16_t tmp[32];
+            const int gm = m0 + srow;
+            if (gm < M) {
+                const float4* src = reinterpret_cast<const float4*>(
+                    X + (size_t)gm * K + kb + sh * 32);
+                #pragma unroll
+                for (int j = 0; j < 8; j++) {
+                    const float4 v = src[j];
+                    tmp[j * 4 + 0] = f32_to_bf16b(v.x);
+                    tmp[j * 4 + 1] = f32_to_bf16b(v.y);
+                    tmp[j * 4 + 2] = f32_to_bf16b(v.z);
+                    tmp[j * 4 + 3] = f32_to_bf16b(v.w);
+                }
+            } else {
+                #pragma unroll
+                for (int j = 0; j < 32; j++) tmp[j] = 0;
+            }
+            uint4* dst = reinterpret_cast<uint4*>(
+                Xl + srow * LDW + sh * 32);
+            #pragma unroll
+            for (int j = 0; j < 4; j++)
+                dst[j] = reinterpret_cast<const uint4*>(tmp)[j];
+        }
+        // ---- stage + dequant W tile ----
+        {
+            uint16_t tmp[32];
+            const int64_t gn = (int64_t)n0 + srow;
+            if (gn < N) {
+                stage_w_slice<W>(qs, hdr, gn, K, kb, sh, tmp);
+            } else {
+                #pragma unroll
+                for (int j = 0; j < 32; j++) tmp[j] = 0;
+            }
+            if constexpr (W == DT::DQ4K) {
+                // slice covers klocal [h*16, h*16+16) and [32+h*16, ...)
+                uint4* d0 = reinterpret_cast<uint4*>(
+                    Wl + srow * LDW + sh * 16);
+                uint4* d1 = reinterpret_cast<uint4*>(
+                    Wl + srow * LDW + 32 + sh * 16);
+                d0[0] = reinterpret_cast<const uint4*>(tmp)[0];
+                d0[1] = reinterpret_cast<const uint4*>(tmp)[1];
+                d1[0] = reinterpret_cast<const uint4*>(tmp)[2];
+                d1[1] = reinterpret_cast<const uint4*>(tmp)[3];
+            } else {
+                uint4* dst = reinterpret_cast<uint4*>(
+                    Wl + srow * LDW + sh * 32);
+                #pragma unroll
+                for (int j = 0; j < 4; j++)
+                    dst[j] = reinterpret_cast<const uint4*>(tmp)[j];
+            }
+        }
+        __syncthreads();
+        // ---- MFMA over the tile ----
+        #pragma unroll
+        for (int ks = 0; ks < BK; ks += 32) {
+            bf16x8 a[4], b[4];
+            #pragma unroll
+            for (int i = 0; i < 4; i++) {
+                const int xr = wm * 64 + i * 16 + lrow;
+                a[i] = *reinterpret_cast<const bf16x8*>(
+                    Xl + xr * LDW + ks + lk * 8);
+            }
+            #pragma unroll
+            for (int j = 0; j < 4; j++) {
+                const int wr = wn * 64 + j * 16 + lrow;
+                b[j] = *reinterpret_cast<const bf16x8*>(
+                    Wl + wr * LDW + ks + lk * 8);
+            }
+            #pragma unroll
+            for (int i = 0; i < 4; i++)
+                #pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    // ---- epilogue: C[m][n] (+res) ----
+    #pragma unroll
+    for (int i = 0; i < 4; i++) {
+        #pragma unroll
+        for (int r = 0; r < 4; r++) {
+            const int m = m0 + wm * 64 + i * 16 + lk * 4 + r;
+            if (m >= M) continue;
+            #pragma unroll
+            for (int j = 0; j < 4; j++) {
+                const int n = n0 + wn * 64 + j * 16 + lrow;
+                if (n >= N) continue;
+                const size_t idx = (size_t)m * ldc + n;
+                C[idx] = acc[i][j][r] + (res ? res[idx] : 0.f);
+            }
+        }
+    }
+}
+
+// ---- row-wise elementwise kernels for the GEMM (prefill/batched) path ----
+
+// grid M; block 256. out[m] = rmsnorm(x[m]) * gw
+__global__ __launch_bounds__(256) void k_rmsnorm_rows(
+    const float* __restrict__ X, const float* __restrict__ gw,
+    float* __restrict__ out, int K, float eps) {
+    const int m = blockIdx.x;
+    const float4* x4 = reinterpret_cast<const float4*>(X + (size_t)m * K);
+    float4* o4 = reinterpret_cast<float4*>(out + (size_t)m * K);
+    const float4* g4 = reinterpret_cast<const float4*>(gw);
+    const int K4 = K >> 2;
+    float ss = 0.f;
+    for (int k = threadIdx.x; k < K4; k += 256) {
+        const float4 v = x4[k];
+        ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+    }
+    __shared__ float red[256];
+    red[threadIdx.x] = ss;
+    __syncthreads();
+    #pragma unroll
+    for (int off = 128; off > 0; off >>= 1) {
+        if ((int)threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+        __syncthreads();
+    }
+    const float inv = rsqrtf(red[0] / (float)K + eps);
+    for (int k = threadIdx.x; k < K4; k += 256) {
+        const float4 v = x4[k];
+        const float4 g = g4[k];
+        float4 o;
+        o.x = v.x * inv * g.x; o.y = v.y * inv * g.y;
+        o.z = v.z * inv * g.z; o.w = v.w * inv * g.w;
+        o4[k] = o;
+    }
+}
+
+// grid M; block 256. act[m][k] = silu(gu[m][k]) * gu[m][F+k]
+__global__ __launch_bounds__(256) void k_silu_rows(
+    const float* __restrict__ GU, float* __restrict__ out, int F) {
+    const int m = blockIdx.x;
+    const float4* g4 = reinterpret_cast<const float4*>(GU + (size_t)m * 2 * F);
+    const float4* u4 = reinterpret_cast<const float4*>(GU + (size_t)m * 2 * F + F);
+    float4* o4 = reinterpret_cast<float4*>(out + (size_t)m * F);
+    const int F4 = F >> 2;
+    for (int k = threadIdx.x; k < F4; k += 256) {
+        const float4 g = g4[k], u = u4[k];
+        float4 o;
+        o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+        o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+        o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+        o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
+        o4[k] = o;
+    }
+}
+
+// Prefill RoPE + KV append over M prompt rows of ONE slot.
+// grid (M, KVH); block 128. Position of row m is pos0 + m.
+__global__ __launch_bounds__(128) void k_rope_prefill(
+    float* __restrict__ qkv,          // [M][(NH+2*NKV)*D]
+    const float* __restrict__ inv_freq,
+    const int32_t* __restrict__ page_table, uint16_t* __restrict__ kv_pool,
+    int slot, int pos0, int NH, int NKV, int D, int G, int page_size,
+    int max_pages, int64_t page_stride) {
+    const int m = blockIdx.x, kvh = blockIdx.y;
+    const int pos = pos0 + m;
+    float* row = qkv + (size_t)m * (NH + 2 * NKV) * D;
+    float* kh = row + (size_t)(NH + kvh) * D;
+    const float* vh = row + (size_t)(NH + NKV + kvh) * D;
+    const int half = D / 2;
+    for (int idx = threadIdx.x; idx < (G + 1) * half; idx += 128) {
+        const int hsel = idx / half, i = idx % half;
+        float* p = (hsel < G) ? (row + (size_t)(kvh * G + hsel) * D) : kh;
+        float sn, cs;
+        __sincosf((float)pos * inv_freq[i], &sn, &cs);
+        const float x0 = p[2 * i], x1 = p[2 * i + 1];
+        p[2 * i] = x0 * cs - x1 * sn;
+        p[2 * i + 1] = x0 * sn + x1 * cs;
+    }
+    __syncthreads();
+    const int page = page_table[(size_t)slot * max_pages + pos / page_size];
+    uint16_t* kdst = kv_pool + (int64_t)page * page_stride
+                     + ((int64_t)kvh * 2 + 0) * page_size * D
+                     + (int64_t)(pos % page_size) * D;
+    uint16_t* vdst = kv_pool + (int64_t)page * page_stride
+                     + ((int64_t)kvh * 2 + 1) * page_size * D
+                     + (int64_t)(pos % page_size) * D;
+    for (int d = threadIdx.x; d < D; d += 128) {
+        union { uint32_t u; float f; } a, b;
+        a.f = kh[d];
+        b.f = vh[d];
+        kdst[d] = (uint16_t)((a.u + 0x7FFF + ((a.u >> 16) & 1)) >> 16);
+        vdst[d] = (uint16_t)((b.u + 0x7FFF + ((b.u >> 16) & 1)) >> 16);
+    }
+}
+
+// Causal prefill attention for ONE slot reading the paged cache.
+// grid (ceil(M/16), NH); block 256 = 4 waves x 4 quarters = 16 q-rows.
+// Each 16-lane quarter owns one q row's full (m,l,o[D]) accumulator and
+// sweeps cached positions 0..pos0+row (causal).
+template <int D>
+__global__ __launch_bounds__(256) void k_attn_prefill(
+    const float* __restrict__ qkv,    // [M][(NH+2*NKV)*D], q already roped
+    const int32_t* __restrict__ page_table,
+    const uint16_t* __restrict__ kv_pool,
+    float* __restrict__ attn_out,     // [M][NH*D]
+    int slot, int pos0, int M, int NH, int NKV, int page_size, int max_pages,
+    int64_t page_stride, float scale) {
+    constexpr int DPL = D / 16;
+    const int qt = blockIdx.x, head = blockIdx.y;
+    const int kvh = head / (NH / NKV);
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int quarter = (tid >> 6) * 4 + (lane >> 4);   // 0..15
+    const int qlane = lane & 15;
+    const int d0 = qlane * DPL;
+    const int m = qt * 16 + quarter;                    // q row in [0,M)
+    const bool active = m < M;
+    const int causal_len = active ? (pos0 + m + 1) : 0;
+
+    float qf[DPL];
+    #pragma unroll
+    for (int j = 0; j < DPL; j++) qf[j] = 0.f;
+    if (active) {
+        const float* qh = qkv + (size_t)m * (NH + 2 * NKV) * D
+                          + (size_t)head * D + d0;
+        #pragma unroll
+        for (int j = 0; j < DPL; j++) qf[j] = qh[j] * scale;
+    }
+
+    float mx = -1e30f, l = 0.f, o[DPL];
+    #pragma unroll
+    for (int j = 0; j < DPL; j++) o[j] = 0.f;
+
+    for (int p = 0; p < causal_len; p++) {
+        const int page = page_table[(size_t)slot * max_pages + p / page_size];
+        const uint16_t* kp = kv_pool + (int64_t)page * page_stride
+                             + ((int64_t)kvh * 2 + 0) * page_size * D
+                             + (int64_t)(p % page_size) * D + d0;
+        const uint16_t* vp = kv_pool + (int64_t)page * page_stride
+                             + ((int64_t)kvh * 2 + 1) * page_size * D
+                             + (int64_t)(p % page_size) * D + d0;
+        float dot = 0.f;
+        #pragma unroll
+        for (int j = 0; j < DPL / 2; j++) {
+            const uint32_t kw = reinterpret_cast<const uint32_t*>(kp)[j];
+            union { uint32_t u; float f; } lo, hi;
+            lo.u = (kw & 0xFFFF) << 16;
+            hi.u = (kw >> 16) << 16;
+            dot += qf[2 * j] * lo.f + qf[2 * j + 1] * hi.f;
+        }
+        #pragma unroll
+        for (int off = 1; off < 16; off <<= 1) dot += __shfl_xor(dot, off, 64);
+        const float mn = fmaxf(mx, dot);
+        const float alpha = __expf(mx - mn);
+        const float w = __expf(dot - mn);
+        l = l * alpha + w;
+        #pragma unroll
+        for (int j = 0; j < DPL / 2; j++) {
+            const uint32_t vw = reinterpret_cast<const uint32_t*>(vp)[j];
+            union { uint32_t u; float f; } lo, hi;
+            lo.u = (vw & 0xFFFF) << 16;
+            hi.u = (vw >> 16) << 16;
+            o[2 * j] = o[2 * j] * alpha + w * lo.f;
+            o[2 * j + 1] = o[2 * j + 1] * alpha + w * hi.f;
+        }
+        mx = mn;
+    }
+    if (active) {
+        float* dst = attn_out + (size_t)m * NH * D + (size_t)head * D + d0;
+        #pragma unroll
+        for (int j = 0; j < DPL; j++) dst[j] = o[j] / l;
+    }
+}
+
+// --------------------------------------------------------- launch stubs
+
+#define DISPATCH_DT_GEMM(DTV, FN)                                     \
+    switch (DTV) {                                                    \
+        case DT::DQ4K: FN(DT::DQ4K); break;                           \
+        case DT::DQ6K: FN(DT::DQ6K); break;                           \
+        case DT::DQ8:  FN(DT::DQ8);  break;                           \
+        case DT::BF16: FN(DT::BF16); break;                           \
+        case DT::F16:  FN(DT::F16);  break;                           \
+        case DT::F32:  FN(DT::F32);  break;                           \
+        default: throw std::runtime_error("bad dtype");               \
+    }
+
+void launch_gemm(const WTensor& w, const float* X, const float* res, float* C,
+                 int M, int ldc, hipStream_t stream) {
+    const int N = (int)w.n, K = (int)w.k;
+    if (K % BK != 0) throw std::runtime_error("gemm: K must be /64");
+    dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM), block(256);
+    #define GEMM_CASE(WT)                                                      \
+        hipLaunchKernelGGL(k_gemm<WT>, grid, block, 0, stream,                 \
+            (const uint8_t*)w.qs, (const uint8_t*)w.hdr, X, res, C, M, N, K,   \
+            ldc)
+    DISPATCH_DT_GEMM(w.dtype, GEMM_CASE);
+    #undef GEMM_CASE
+}
+
+void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
+                         int K, float eps, hipStream_t stream) {
+    hipLaunchKernelGGL(k_rmsnorm_rows, dim3(M), dim3(256), 0, stream,
+                       X, gw, out, K, eps);
+}
+
+void launch_silu_rows(const float* GU, float* out, int M, int F,
+                      hipStream_t stream) {
+    hipLaunchKernelGGL(k_silu_rows, dim3(M), dim3(256), 0, stream, GU, out, F);
+}
+
+void launch_rope_prefill(float* qkv, const float* inv_freq,
+                         const int32_t* page_table, uint16_t* kv_pool,
+                         int slot, int pos0, int M, int NH, int NKV, int D,
+                         int page_size, int max_pages, int64_t page_stride,
+                         hipStream_t stream) {
+    hipLaunchKernelGGL(k_rope_prefill, dim3(M, NKV), dim3(128), 0, stream,
+                       qkv, inv_freq, page_table, kv_pool, slot, pos0,
+                       NH, NKV, D, NH / NKV, page_size, max_pages,
+                       page_stride);
+}
+
+void launch_attn_prefill(const float* qkv, const int32_t* page_table,
+                         const uint16_t* kv_pool, float* attn_out, int slot,
+                         int pos0, int M, int NH, int NKV, int D,
+                         int page_size, int max_pages, int64_t page_stride,
+                         float scale, hipStream_t stream) {
+    dim3 grid((M + 15) / 16, NH), block(256);
+    if (D == 128) {
+        hipLaunchKernelGGL(k_attn_prefill<128>, grid, block, 0, stream,
+                           qkv, page_table, kv_pool, attn_out, slot, pos0, M,
+                           NH, NKV, page_size, max_pages, page_stride, scale);
+    } else if (D == 64) {
+        hipLaunchKernelGGL(k_attn_prefill<64>, grid, block, 0, stream,
+                           qkv, page_table, kv_pool, attn_out, slot, pos0, M,
+                           NH, NKV, page_size, max_pages, page_stride, scale);
+    } else {
+        throw std::runtime_error("head_dim must be 64 or 128");
+    }
+}
+
+}  // namespace cla

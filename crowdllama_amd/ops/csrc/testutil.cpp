@@ -39,4 +39,45 @@ void launch_gemv_test(const void* qs, const void* hdr, const float* x,
     hipFree(d_x); hipFree(d_y); if (d_gw) hipFree(d_gw);
 }
 
+double bench_gemv(const void* qs, const void* hdr, int dtype, int N, int K,
+                  int B, int pre, size_t qs_bytes, size_t hdr_bytes,
+                  int iters) {
+    const DT dt = static_cast<DT>(dtype);
+    const size_t xn = (pre == 2) ? (size_t)B * 2 * K : (size_t)B * K;
+    void *d_qs = nullptr, *d_hdr = nullptr, *d_x = nullptr, *d_y = nullptr,
+         *d_gw = nullptr;
+    HIP_CHECK(hipMalloc(&d_qs, qs_bytes));
+    HIP_CHECK(hipMemcpy(d_qs, qs, qs_bytes, hipMemcpyHostToDevice));
+    if (hdr_bytes) {
+        HIP_CHECK(hipMalloc(&d_hdr, hdr_bytes));
+        HIP_CHECK(hipMemcpy(d_hdr, hdr, hdr_bytes, hipMemcpyHostToDevice));
+    }
+    HIP_CHECK(hipMalloc(&d_x, xn * 4));
+    HIP_CHECK(hipMemset(d_x, 0, xn * 4));
+    HIP_CHECK(hipMalloc(&d_y, (size_t)B * N * 4));
+    HIP_CHECK(hipMalloc(&d_gw, (size_t)K * 4));
+    HIP_CHECK(hipMemset(d_gw, 0, (size_t)K * 4));
+    WTensor w;
+    w.dtype = dt; w.n = N; w.k = K; w.qs = d_qs; w.hdr = d_hdr;
+    for (int i = 0; i < 3; i++)
+        launch_gemv(w, pre, (const float*)d_x, (const float*)d_gw, nullptr,
+                    (float*)d_y, B, N, 1e-5f, nullptr);
+    HIP_CHECK(hipDeviceSynchronize());
+    hipEvent_t e0, e1;
+    HIP_CHECK(hipEventCreate(&e0));
+    HIP_CHECK(hipEventCreate(&e1));
+    HIP_CHECK(hipEventRecord(e0, nullptr));
+    for (int i = 0; i < iters; i++)
+        launch_gemv(w, pre, (const float*)d_x, (const float*)d_gw, nullptr,
+                    (float*)d_y, B, N, 1e-5f, nullptr);
+    HIP_CHECK(hipEventRecord(e1, nullptr));
+    HIP_CHECK(hipDeviceSynchronize());
+    float ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+    hipEventDestroy(e0); hipEventDestroy(e1);
+    hipFree(d_qs); if (d_hdr) hipFree(d_hdr);
+    hipFree(d_x); hipFree(d_y); hipFree(d_gw);
+    return ms / iters;
+}
+
 }  // namespace cla
