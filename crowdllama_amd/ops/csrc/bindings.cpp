@@ -15,6 +15,9 @@ void launch_gemv_test(const void*, const void*, const float*, const float*,
 void launch_mfma_probe_test(const uint16_t*, const uint16_t*, float*);
 double bench_gemv(const void*, const void*, int, int, int, int, int, size_t,
                   size_t, int);
+void launch_gemm_test(const void*, const void*, const float*, float*, int,
+                      int, int, int, size_t, size_t);
+double bench_gemm(int, int, int, int, int);
 }
 
 PYBIND11_MODULE(_core, m) {
@@ -88,6 +91,19 @@ PYBIND11_MODULE(_core, m) {
             d["rms_eps"] = mm.rms_eps;
             return d;
         });
+
+    m.def("test_gemm", [](py::array_t<uint8_t> qs, py::array_t<uint8_t> hdr,
+                          py::array_t<float, py::array::c_style> x, int dtype,
+                          int N, int K) {
+        const int M = (int)x.shape(0);
+        py::array_t<float> y({M, N});
+        launch_gemm_test(qs.data(), hdr.data(), x.data(), y.mutable_data(),
+                         dtype, M, N, K, qs.nbytes(), hdr.nbytes());
+        return y;
+    });
+    m.def("bench_gemm", [](int dtype, int M, int N, int K, int iters) {
+        return bench_gemm(dtype, M, N, K, iters);
+    });
 
     m.def("bench_gemv", [](int dtype, int N, int K, int B, int pre, int iters) {
         // random weight bytes (content irrelevant for timing)

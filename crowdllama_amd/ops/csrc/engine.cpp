@@ -28,6 +28,21 @@ void launch_attn_combine(const float* part_o, const float* part_ml,
 void launch_argmax(const float* logits, float* pval, int32_t* pidx,
                    int32_t* cur_ids, int32_t* n_past, int32_t* gen_tokens,
                    int32_t* gen_count, int B, int V, int gen_cap, hipStream_t);
+void launch_gemm(const WTensor&, const float* X, const float* res, float* C,
+                 int M, int ldc, hipStream_t);
+void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
+                         int K, float eps, hipStream_t);
+void launch_silu_rows(const float* GU, float* out, int M, int F, hipStream_t);
+void launch_rope_prefill(float* qkv, const float* inv_freq,
+                         const int32_t* page_table, uint16_t* kv_pool,
+                         int slot, int pos0, int M, int NH, int NKV, int D,
+                         int page_size, int max_pages, int64_t page_stride,
+                         hipStream_t);
+void launch_attn_prefill(const float* qkv, const int32_t* page_table,
+                         const uint16_t* kv_pool, float* attn_out, int slot,
+                         int pos0, int M, int NH, int NKV, int D,
+                         int page_size, int max_pages, int64_t page_stride,
+                         float scale, hipStream_t);
 
 namespace {
 
@@ -142,10 +157,8 @@ void repack(int32_t ggml_type, const uint8_t* src, int64_t rows, int64_t k,
 
 Engine::Engine(const std::string& gguf_path, const EngineConfig& cfg)
     : cfg_(cfg) {
-    if (cfg_.batch < 1 || cfg_.batch > 2)
-        throw std::runtime_error(
-            "decode batch must be 1 or 2 on the GEMV path "
-            "(batched MFMA path lands next)");
+    if (cfg_.batch < 1 || cfg_.batch > 64)
+        throw std::runtime_error("decode batch must be in [1, 64]");
     HIP_CHECK(hipSetDevice(cfg_.device));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     GGUFFile gf(gguf_path);
@@ -313,6 +326,19 @@ void Engine::alloc_state() {
         return p;
     };
     x_ = (float*)dalloc((size_t)B * H * 4);
+    xn_ = (float*)dalloc((size_t)B * H * 4);
+    act_ = (float*)dalloc((size_t)B * F * 4);
+    {   // prefill scratch
+        const int Mc = cfg_.prefill_chunk;
+        const int QKV = (NH + 2 * NKV) * D;
+        xp_ = (float*)dalloc((size_t)Mc * H * 4);
+        xnp_ = (float*)dalloc((size_t)Mc * H * 4);
+        qkvp_ = (float*)dalloc((size_t)Mc * QKV * 4);
+        attnp_ = (float*)dalloc((size_t)Mc * NH * D * 4);
+        gup_ = (float*)dalloc((size_t)Mc * 2 * F * 4);
+        actp_ = (float*)dalloc((size_t)Mc * F * 4);
+        pids_ = (int32_t*)dalloc((size_t)Mc * 4);
+    }
     qkv_ = (float*)dalloc((size_t)B * (NH + 2 * NKV) * D * 4);
     attn_out_ = (float*)dalloc((size_t)B * NH * D * 4);
     gu_ = (float*)dalloc((size_t)B * 2 * F * 4);
@@ -358,30 +384,57 @@ void Engine::step(hipStream_t s) {
     const float eps = meta_.rms_eps;
 
     launch_embed(embed_, cur_ids_, x_, B, s);
+    const bool gemv_path = B <= 2;
     int li = 0;
     for (auto& L : layers_) {
         uint16_t* kv_layer = kv_pool_ + (int64_t)li * layer_stride_;
         li++;
-        for (auto& pt : L.qkv.parts)
-            launch_gemv(pt.w, PRE_RMS, x_, L.attn_norm, nullptr,
-                        qkv_ + pt.row_off, B, (int)L.qkv.n_total, eps, s);
+        if (gemv_path) {
+            for (auto& pt : L.qkv.parts)
+                launch_gemv(pt.w, PRE_RMS, x_, L.attn_norm, nullptr,
+                            qkv_ + pt.row_off, B, (int)L.qkv.n_total, eps, s);
+        } else {
+            launch_rmsnorm_rows(x_, L.attn_norm, xn_, B, meta_.hidden, eps, s);
+            for (auto& pt : L.qkv.parts)
+                launch_gemm(pt.w, xn_, nullptr, qkv_ + pt.row_off, B,
+                            (int)L.qkv.n_total, s);
+        }
         launch_attn_decode(qkv_, inv_freq_, page_table_, kv_layer, n_past_,
                            part_o_, part_ml_, B, NH, NKV, D, attn_splits_,
                            cfg_.page_size, max_pages_, page_stride_, scale, s);
         launch_attn_combine(part_o_, part_ml_, attn_out_, B, NH, attn_splits_,
                             D, s);
-        for (auto& pt : L.o.parts)
-            launch_gemv(pt.w, PRE_NONE, attn_out_, nullptr, x_ + pt.row_off,
-                        x_ + pt.row_off, B, (int)L.o.n_total, eps, s);
-        for (auto& pt : L.gate_up.parts)
-            launch_gemv(pt.w, PRE_RMS, x_, L.ffn_norm, nullptr,
-                        gu_ + pt.row_off, B, (int)L.gate_up.n_total, eps, s);
-        for (auto& pt : L.down.parts)
-            launch_gemv(pt.w, PRE_SILU, gu_, nullptr, x_ + pt.row_off,
-                        x_ + pt.row_off, B, (int)L.down.n_total, eps, s);
+        if (gemv_path) {
+            for (auto& pt : L.o.parts)
+                launch_gemv(pt.w, PRE_NONE, attn_out_, nullptr, x_ + pt.row_off,
+                            x_ + pt.row_off, B, (int)L.o.n_total, eps, s);
+            for (auto& pt : L.gate_up.parts)
+                launch_gemv(pt.w, PRE_RMS, x_, L.ffn_norm, nullptr,
+                            gu_ + pt.row_off, B, (int)L.gate_up.n_total, eps, s);
+            for (auto& pt : L.down.parts)
+                launch_gemv(pt.w, PRE_SILU, gu_, nullptr, x_ + pt.row_off,
+                            x_ + pt.row_off, B, (int)L.down.n_total, eps, s);
+        } else {
+            for (auto& pt : L.o.parts)
+                launch_gemm(pt.w, attn_out_, x_ + pt.row_off, x_ + pt.row_off,
+                            B, (int)L.o.n_total, s);
+            launch_rmsnorm_rows(x_, L.ffn_norm, xn_, B, meta_.hidden, eps, s);
+            for (auto& pt : L.gate_up.parts)
+                launch_gemm(pt.w, xn_, nullptr, gu_ + pt.row_off, B,
+                            (int)L.gate_up.n_total, s);
+            launch_silu_rows(gu_, act_, B, meta_.ffn, s);
+            for (auto& pt : L.down.parts)
+                launch_gemm(pt.w, act_, x_ + pt.row_off, x_ + pt.row_off, B,
+                            (int)L.down.n_total, s);
+        }
     }
-    launch_gemv(head_, PRE_RMS, x_, out_norm_, nullptr, logits_, B,
-                meta_.vocab, eps, s);
+    if (gemv_path) {
+        launch_gemv(head_, PRE_RMS, x_, out_norm_, nullptr, logits_, B,
+                    meta_.vocab, eps, s);
+    } else {
+        launch_rmsnorm_rows(x_, out_norm_, xn_, B, meta_.hidden, eps, s);
+        launch_gemm(head_, xn_, nullptr, logits_, B, meta_.vocab, s);
+    }
     launch_argmax(logits_, amax_val_, amax_idx_, cur_ids_, n_past_,
                   gen_tokens_, gen_count_, B, meta_.vocab, cfg_.gen_cap, s);
 }
@@ -398,37 +451,84 @@ void Engine::ensure_graph() {
     HIP_CHECK(hipGraphDestroy(graph));
 }
 
+void Engine::prefill_chunk_pass(int slot, int pos0, int m) {
+    // One GEMM pass over m prompt rows (ids already in pids_).
+    hipStream_t s = stream_;
+    const int NH = meta_.heads, NKV = meta_.kv_heads, D = meta_.head_dim;
+    const float scale = 1.0f / std::sqrt((float)D);
+    const float eps = meta_.rms_eps;
+    launch_embed(embed_, pids_, xp_, m, s);
+    int li = 0;
+    for (auto& L : layers_) {
+        uint16_t* kv_layer = kv_pool_ + (int64_t)li * layer_stride_;
+        li++;
+        launch_rmsnorm_rows(xp_, L.attn_norm, xnp_, m, meta_.hidden, eps, s);
+        for (auto& pt : L.qkv.parts)
+            launch_gemm(pt.w, xnp_, nullptr, qkvp_ + pt.row_off, m,
+                        (int)L.qkv.n_total, s);
+        launch_rope_prefill(qkvp_, inv_freq_, page_table_, kv_layer, slot,
+                            pos0, m, NH, NKV, D, cfg_.page_size, max_pages_,
+                            page_stride_, s);
+        launch_attn_prefill(qkvp_, page_table_, kv_layer, attnp_, slot, pos0,
+                            m, NH, NKV, D, cfg_.page_size, max_pages_,
+                            page_stride_, scale, s);
+        for (auto& pt : L.o.parts)
+            launch_gemm(pt.w, attnp_, xp_ + pt.row_off, xp_ + pt.row_off, m,
+                        (int)L.o.n_total, s);
+        launch_rmsnorm_rows(xp_, L.ffn_norm, xnp_, m, meta_.hidden, eps, s);
+        for (auto& pt : L.gate_up.parts)
+            launch_gemm(pt.w, xnp_, nullptr, gup_ + pt.row_off, m,
+                        (int)L.gate_up.n_total, s);
+        launch_silu_rows(gup_, actp_, m, meta_.ffn, s);
+        for (auto& pt : L.down.parts)
+            launch_gemm(pt.w, actp_, xp_ + pt.row_off, xp_ + pt.row_off, m,
+                        (int)L.down.n_total, s);
+    }
+}
+
+void Engine::prefill_slot(int slot, const std::vector<int32_t>& ids) {
+    if (ids.empty())
+        throw std::runtime_error("prefill_slot: empty prompt");
+    const int len = (int)ids.size();
+    if (len >= cfg_.max_seq)
+        throw std::runtime_error("prompt longer than max_seq");
+    int pos0 = 0;  // prefill restarts the slot
+    int done = 0;
+    const float eps = meta_.rms_eps;
+    while (done < len) {
+        const int m = std::min(cfg_.prefill_chunk, len - done);
+        HIP_CHECK(hipMemcpyAsync(pids_, ids.data() + done, (size_t)m * 4,
+                                 hipMemcpyHostToDevice, stream_));
+        prefill_chunk_pass(slot, pos0, m);
+        HIP_CHECK(hipStreamSynchronize(stream_));
+        pos0 += m;
+        done += m;
+    }
+    // logits of the LAST prompt row -> slot's logits; sample + advance state
+    const int last = ((len - 1) % cfg_.prefill_chunk);
+    launch_gemv(head_, PRE_RMS, xp_ + (size_t)last * meta_.hidden, out_norm_,
+                nullptr, logits_ + (size_t)slot * meta_.vocab, 1,
+                meta_.vocab, eps, stream_);
+    HIP_CHECK(hipStreamSynchronize(stream_));
+    // n_past[slot] = len - 1; argmax's advance makes it len.
+    const int32_t npast = len - 1, zero = 0;
+    HIP_CHECK(hipMemcpy(n_past_ + slot, &npast, 4, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(gen_count_ + slot, &zero, 4, hipMemcpyHostToDevice));
+    launch_argmax(logits_ + (size_t)slot * meta_.vocab,
+                  amax_val_ + (size_t)slot * 64, amax_idx_ + (size_t)slot * 64,
+                  cur_ids_ + slot, n_past_ + slot,
+                  gen_tokens_ + (size_t)slot * cfg_.gen_cap,
+                  gen_count_ + slot, 1, meta_.vocab, cfg_.gen_cap, stream_);
+    HIP_CHECK(hipStreamSynchronize(stream_));
+}
+
 void Engine::prefill(const std::vector<int32_t>& ids, int len) {
     const int B = cfg_.batch;
     if ((int)ids.size() != B * len)
         throw std::runtime_error("prefill: ids must be batch*len");
-    ensure_graph();
-    // Transposed staging buffer that stays alive until the final stream
-    // sync: hipMemcpyAsync from pageable memory may read the buffer late,
-    // so per-iteration reuse of one buffer races with in-flight copies.
-    std::vector<int32_t> cols((size_t)len * B);
-    for (int t = 0; t < len; t++)
-        for (int b = 0; b < B; b++)
-            cols[(size_t)t * B + b] = ids[(size_t)b * len + t];
-    for (int t = 0; t < len; t++) {
-        HIP_CHECK(hipMemcpyAsync(cur_ids_, cols.data() + (size_t)t * B, B * 4,
-                                 hipMemcpyHostToDevice, stream_));
-        if (graph_exec_) {
-            HIP_CHECK(hipGraphLaunch(graph_exec_, stream_));
-        } else {
-            step(stream_);
-        }
-    }
-    HIP_CHECK(hipStreamSynchronize(stream_));
-    // after the last prompt token, cur_ids holds the first generated token;
-    // reset the gen ring so it lands at index 0.
-    std::vector<int32_t> first(B);
-    HIP_CHECK(hipMemcpy(first.data(), cur_ids_, B * 4, hipMemcpyDeviceToHost));
-    std::vector<int32_t> zero(B, 1);
-    HIP_CHECK(hipMemcpy(gen_count_, zero.data(), B * 4, hipMemcpyHostToDevice));
     for (int b = 0; b < B; b++)
-        HIP_CHECK(hipMemcpy(gen_tokens_ + (size_t)b * cfg_.gen_cap, &first[b],
-                            4, hipMemcpyHostToDevice));
+        prefill_slot(b, std::vector<int32_t>(ids.begin() + (size_t)b * len,
+                                             ids.begin() + (size_t)(b + 1) * len));
 }
 
 void Engine::decode(int n_steps) {

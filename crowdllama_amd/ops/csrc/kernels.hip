@@ -59,22 +59,47 @@ template <> struct ChunkTraits<DT::BF16> { static constexpr int W_PER_CHUNK = 8;
 template <> struct ChunkTraits<DT::F16>  { static constexpr int W_PER_CHUNK = 8; };
 template <> struct ChunkTraits<DT::F32>  { static constexpr int W_PER_CHUNK = 4; };
 
+typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
+
+// Raw per-chunk payload (issued early; decode consumes it later).
+template <DT W>
+struct ChunkRaw {
+    u32x4 qv;
+    uint4 hd;       // DQ4K header; DQ6K/DQ8 use scalar fields
+    float d0, d1;   // DQ6K: two eff scales; DQ8: block scale
+};
+
+template <DT W>
+__device__ __forceinline__ void load_chunk(
+    const uint8_t* __restrict__ qs_row, const uint8_t* __restrict__ hdr_row,
+    int c, ChunkRaw<W>* r) {
+    // weights stream through once per step: non-temporal (bypass L1/L2 keep)
+    r->qv = __builtin_nontemporal_load(
+        reinterpret_cast<const u32x4*>(qs_row) + c);
+    if constexpr (W == DT::DQ4K) {
+        r->hd = reinterpret_cast<const uint4*>(hdr_row)[c >> 3];
+    } else if constexpr (W == DT::DQ6K) {
+        const int sb = c >> 4, s16 = c & 15;
+        const uint8_t* hb = hdr_row + sb * 32;
+        const float d = f16_bits_to_f32(*reinterpret_cast<const uint16_t*>(hb));
+        r->d0 = d * (float)(reinterpret_cast<const int8_t*>(hb)[4 + s16]);
+    } else if constexpr (W == DT::DQ8) {
+        r->d0 = f16_bits_to_f32(
+            reinterpret_cast<const uint16_t*>(hdr_row)[c >> 1]);
+    }
+}
+
 // Decode one 16-B chunk of a row into `w[]` weights starting at column
 // `k0` (and for DQ4K a second run of 16 at k0+32).
 template <DT W>
-__device__ __forceinline__ void decode_chunk(
-    const uint8_t* __restrict__ qs_row, const uint8_t* __restrict__ hdr_row,
-    int c, float* __restrict__ w, int* k0) {
-    // weights stream through once per step: non-temporal (bypass L1/L2 keep)
-    typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
-    const u32x4 qv4 = __builtin_nontemporal_load(
-        reinterpret_cast<const u32x4*>(qs_row) + c);
+__device__ __forceinline__ void decode_chunk_raw(
+    const ChunkRaw<W>& r, int c, float* __restrict__ w, int* k0) {
     uint4 qv;
-    qv.x = qv4.x; qv.y = qv4.y; qv.z = qv4.z; qv.w = qv4.w;
+    qv.x = r.qv.x; qv.y = r.qv.y; qv.z = r.qv.z; qv.w = r.qv.w;
     const uint32_t dw[4] = {qv.x, qv.y, qv.z, qv.w};
     if constexpr (W == DT::DQ4K) {
         const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
-        const uint4 hd = reinterpret_cast<const uint4*>(hdr_row)[sb];
+        const uint4 hd = r.hd;
         const float d = f16_bits_to_f32(hd.x & 0xFFFF);
         const float dmin = f16_bits_to_f32(hd.x >> 16);
         const uint32_t sbytes[3] = {hd.y, hd.z, hd.w};
@@ -106,10 +131,7 @@ __device__ __forceinline__ void decode_chunk(
             }
         }
     } else if constexpr (W == DT::DQ6K) {
-        const int sb = c >> 4, s16 = c & 15;
-        const uint8_t* hb = hdr_row + sb * 32;
-        const float d = f16_bits_to_f32(*reinterpret_cast<const uint16_t*>(hb));
-        const float sc = d * (float)(reinterpret_cast<const int8_t*>(hb)[4 + s16]);
+        const float sc = r.d0;
         *k0 = c * 16;
         #pragma unroll
         for (int j = 0; j < 4; j++) {
@@ -120,8 +142,7 @@ __device__ __forceinline__ void decode_chunk(
             }
         }
     } else if constexpr (W == DT::DQ8) {
-        const float d = f16_bits_to_f32(
-            reinterpret_cast<const uint16_t*>(hdr_row)[c >> 1]);
+        const float d = r.d0;
         *k0 = c * 16;
         #pragma unroll
         for (int j = 0; j < 4; j++) {
@@ -234,10 +255,13 @@ __global__ __launch_bounds__(256) void k_gemv(
     constexpr int WPC = ChunkTraits<W>::W_PER_CHUNK;
 
     float acc[2] = {0.f, 0.f};  // GEMV path supports B <= 2
+    ChunkRaw<W> cur, nxt;
+    if (lane < n_chunks) load_chunk<W>(qs_row, hdr_row, lane, &cur);
     for (int c = lane; c < n_chunks; c += 64) {
+        if (c + 64 < n_chunks) load_chunk<W>(qs_row, hdr_row, c + 64, &nxt);
         float w[WPC];
         int k0;
-        decode_chunk<W>(qs_row, hdr_row, c, w, &k0);
+        decode_chunk_raw<W>(cur, c, w, &k0);
         #pragma unroll 2
         for (int b = 0; b < B; b++) {
             // k0 is a multiple of 16 weights for every dtype's chunk map,
@@ -265,6 +289,7 @@ __global__ __launch_bounds__(256) void k_gemv(
             }
             acc[b] += s;
         }
+        cur = nxt;
     }
     #pragma unroll 2
     for (int b = 0; b < B; b++) {
@@ -293,7 +318,9 @@ __global__ __launch_bounds__(256) void k_embed(
     for (int c = threadIdx.x; c < n_chunks; c += 256) {
         float w[WPC];
         int k0;
-        decode_chunk<W>(qs_row, hdr_row, c, w, &k0);
+        ChunkRaw<W> r;
+        load_chunk<W>(qs_row, hdr_row, c, &r);
+        decode_chunk_raw<W>(r, c, w, &k0);
         if constexpr (W == DT::DQ4K) {
             #pragma unroll
             for (int t = 0; t < 16; t++) xb[k0 + t] = w[t];

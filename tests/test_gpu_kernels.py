@@ -162,3 +162,47 @@ def test_mfma_fragment_layout(core):
                              np.ascontiguousarray(Bb))
     Cref = bf16_val(Ab) @ bf16_val(Bb)
     np.testing.assert_allclose(C, Cref, rtol=1e-5, atol=1e-5)
+
+
+@pytest.mark.parametrize("name", ["q4k", "q6k", "q8"])
+@pytest.mark.parametrize("M", [3, 16, 130])
+def test_gemm_quant(core, name, M):
+    """MFMA dequant-GEMM vs CPU reference (bf16-rounded operands)."""
+    dt, quant, dequant, repack_fn = CASES[name]
+    rng = np.random.default_rng(5)
+    N, K = 192, 512
+    w = rng.standard_normal((N, K)).astype(np.float32) * 0.1
+    raw = quant(w)
+    wref = dequant(raw, K).reshape(N, K)
+    x = rng.standard_normal((M, K)).astype(np.float32)
+    qs, hdr = repack_fn(raw.reshape(N, -1), N, K)
+    y = core.test_gemm(np.ascontiguousarray(qs), np.ascontiguousarray(hdr),
+                       x, dt, N, K)
+
+    def bf16(a):
+        u = np.ascontiguousarray(a, dtype=np.float32).view(np.uint32)
+        return (((u + 0x7FFF + ((u >> 16) & 1)) & 0xFFFF0000)).view(np.float32)
+
+    yref = bf16(x) @ bf16(wref).T
+    scale = np.abs(yref).max() + 1e-6
+    assert np.abs(y - yref).max() / scale < 5e-3
+
+
+def test_gemm_bf16(core):
+    from crowdllama_amd.quant import GGMLType, quantize
+    rng = np.random.default_rng(6)
+    M, N, K = 64, 256, 320
+    w = rng.standard_normal((N, K)).astype(np.float32)
+    raw = quantize(w, GGMLType.BF16)
+    x = rng.standard_normal((M, K)).astype(np.float32)
+    y = core.test_gemm(np.ascontiguousarray(raw.reshape(N, -1)),
+                       np.zeros(0, dtype=np.uint8), x, DT_BF16, N, K)
+
+    def bf16(a):
+        u = np.ascontiguousarray(a, dtype=np.float32).view(np.uint32)
+        return (((u + 0x7FFF + ((u >> 16) & 1)) & 0xFFFF0000)).view(np.float32)
+
+    wref = (raw.view(np.uint16).astype(np.uint32) << 16).view(np.float32).reshape(N, K)
+    yref = bf16(x) @ wref.T
+    scale = np.abs(yref).max() + 1e-6
+    assert np.abs(y - yref).max() / scale < 5e-3
