@@ -30,6 +30,7 @@ void launch_argmax(const float* logits, float* pval, int32_t* pidx,
                    int32_t* gen_count, int B, int V, int gen_cap, hipStream_t);
 void launch_gemm(const WTensor&, const float* X, const float* res, float* C,
                  int M, int ldc, hipStream_t);
+bool gemm_uses_splitk(int N, int K, int M);
 void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
                          int K, float eps, hipStream_t);
 void launch_silu_rows(const float* GU, float* out, int M, int F, hipStream_t);
@@ -327,11 +328,13 @@ void Engine::alloc_state() {
     };
     x_ = (float*)dalloc((size_t)B * H * 4);
     xn_ = (float*)dalloc((size_t)B * H * 4);
+    x2_ = (float*)dalloc((size_t)B * H * 4);
     act_ = (float*)dalloc((size_t)B * F * 4);
     {   // prefill scratch
         const int Mc = cfg_.prefill_chunk;
         const int QKV = (NH + 2 * NKV) * D;
         xp_ = (float*)dalloc((size_t)Mc * H * 4);
+        xp2_ = (float*)dalloc((size_t)Mc * H * 4);
         xnp_ = (float*)dalloc((size_t)Mc * H * 4);
         qkvp_ = (float*)dalloc((size_t)Mc * QKV * 4);
         attnp_ = (float*)dalloc((size_t)Mc * NH * D * 4);
@@ -377,6 +380,20 @@ void Engine::reset() {
     HIP_CHECK(hipDeviceSynchronize());
 }
 
+// Launch a projection through the GEMM path: pre-zero C when split-K
+// accumulation is in play (see launch_gemm), then one launch per part.
+static void gemm_proj(const Proj& p, const float* X, const float* res,
+                      float* C, int M, hipStream_t s) {
+    bool zero = false;
+    for (auto& pt : p.parts)
+        zero |= gemm_uses_splitk((int)pt.w.n, (int)pt.w.k, M);
+    if (zero)
+        HIP_CHECK(hipMemsetAsync(C, 0, (size_t)M * p.n_total * 4, s));
+    for (auto& pt : p.parts)
+        launch_gemm(pt.w, X, res ? res + pt.row_off : nullptr,
+                    C + pt.row_off, M, (int)p.n_total, s);
+}
+
 void Engine::step(hipStream_t s) {
     const int B = cfg_.batch;
     const int NH = meta_.heads, NKV = meta_.kv_heads, D = meta_.head_dim;
@@ -395,9 +412,7 @@ void Engine::step(hipStream_t s) {
                             qkv_ + pt.row_off, B, (int)L.qkv.n_total, eps, s);
         } else {
             launch_rmsnorm_rows(x_, L.attn_norm, xn_, B, meta_.hidden, eps, s);
-            for (auto& pt : L.qkv.parts)
-                launch_gemm(pt.w, xn_, nullptr, qkv_ + pt.row_off, B,
-                            (int)L.qkv.n_total, s);
+            gemm_proj(L.qkv, xn_, nullptr, qkv_, B, s);
         }
         launch_attn_decode(qkv_, inv_freq_, page_table_, kv_layer, n_past_,
                            part_o_, part_ml_, B, NH, NKV, D, attn_splits_,
@@ -415,17 +430,12 @@ void Engine::step(hipStream_t s) {
                 launch_gemv(pt.w, PRE_SILU, gu_, nullptr, x_ + pt.row_off,
                             x_ + pt.row_off, B, (int)L.down.n_total, eps, s);
         } else {
-            for (auto& pt : L.o.parts)
-                launch_gemm(pt.w, attn_out_, x_ + pt.row_off, x_ + pt.row_off,
-                            B, (int)L.o.n_total, s);
-            launch_rmsnorm_rows(x_, L.ffn_norm, xn_, B, meta_.hidden, eps, s);
-            for (auto& pt : L.gate_up.parts)
-                launch_gemm(pt.w, xn_, nullptr, gu_ + pt.row_off, B,
-                            (int)L.gate_up.n_total, s);
+            // residual ping-pong: x_ --(+attn)--> x2_ --(+ffn)--> x_
+            gemm_proj(L.o, attn_out_, x_, x2_, B, s);
+            launch_rmsnorm_rows(x2_, L.ffn_norm, xn_, B, meta_.hidden, eps, s);
+            gemm_proj(L.gate_up, xn_, nullptr, gu_, B, s);
             launch_silu_rows(gu_, act_, B, meta_.ffn, s);
-            for (auto& pt : L.down.parts)
-                launch_gemm(pt.w, act_, x_ + pt.row_off, x_ + pt.row_off, B,
-                            (int)L.down.n_total, s);
+            gemm_proj(L.down, act_, x2_, x_, B, s);
         }
     }
     if (gemv_path) {
@@ -433,7 +443,8 @@ void Engine::step(hipStream_t s) {
                     meta_.vocab, eps, s);
     } else {
         launch_rmsnorm_rows(x_, out_norm_, xn_, B, meta_.hidden, eps, s);
-        launch_gemm(head_, xn_, nullptr, logits_, B, meta_.vocab, s);
+        Proj hp; hp.parts.push_back({head_, 0}); hp.n_total = meta_.vocab;
+        gemm_proj(hp, xn_, nullptr, logits_, B, s);
     }
     launch_argmax(logits_, amax_val_, amax_idx_, cur_ids_, n_past_,
                   gen_tokens_, gen_count_, B, meta_.vocab, cfg_.gen_cap, s);
@@ -463,26 +474,18 @@ void Engine::prefill_chunk_pass(int slot, int pos0, int m) {
         uint16_t* kv_layer = kv_pool_ + (int64_t)li * layer_stride_;
         li++;
         launch_rmsnorm_rows(xp_, L.attn_norm, xnp_, m, meta_.hidden, eps, s);
-        for (auto& pt : L.qkv.parts)
-            launch_gemm(pt.w, xnp_, nullptr, qkvp_ + pt.row_off, m,
-                        (int)L.qkv.n_total, s);
+        gemm_proj(L.qkv, xnp_, nullptr, qkvp_, m, s);
         launch_rope_prefill(qkvp_, inv_freq_, page_table_, kv_layer, slot,
                             pos0, m, NH, NKV, D, cfg_.page_size, max_pages_,
                             page_stride_, s);
         launch_attn_prefill(qkvp_, page_table_, kv_layer, attnp_, slot, pos0,
                             m, NH, NKV, D, cfg_.page_size, max_pages_,
                             page_stride_, scale, s);
-        for (auto& pt : L.o.parts)
-            launch_gemm(pt.w, attnp_, xp_ + pt.row_off, xp_ + pt.row_off, m,
-                        (int)L.o.n_total, s);
-        launch_rmsnorm_rows(xp_, L.ffn_norm, xnp_, m, meta_.hidden, eps, s);
-        for (auto& pt : L.gate_up.parts)
-            launch_gemm(pt.w, xnp_, nullptr, gup_ + pt.row_off, m,
-                        (int)L.gate_up.n_total, s);
+        gemm_proj(L.o, attnp_, xp_, xp2_, m, s);
+        launch_rmsnorm_rows(xp2_, L.ffn_norm, xnp_, m, meta_.hidden, eps, s);
+        gemm_proj(L.gate_up, xnp_, nullptr, gup_, m, s);
         launch_silu_rows(gup_, actp_, m, meta_.ffn, s);
-        for (auto& pt : L.down.parts)
-            launch_gemm(pt.w, actp_, xp_ + pt.row_off, xp_ + pt.row_off, m,
-                        (int)L.down.n_total, s);
+        gemm_proj(L.down, actp_, xp2_, xp_, m, s);
     }
 }
 

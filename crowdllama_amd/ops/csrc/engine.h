@@ -104,9 +104,11 @@ private:
     void prefill_chunk_pass(int slot, int pos0, int m);
     float* x_ = nullptr;        // [B][h]
     float* xn_ = nullptr;       // [B][h] (GEMM decode path rmsnorm out)
+    float* x2_ = nullptr;       // [B][h] residual ping-pong (split-K GEMM)
     float* act_ = nullptr;      // [B][F]
     // prefill scratch ([Mchunk] rows)
     float* xp_ = nullptr;
+    float* xp2_ = nullptr;
     float* xnp_ = nullptr;
     float* qkvp_ = nullptr;
     float* attnp_ = nullptr;

@@ -144,59 +144,69 @@ __device__ __forceinline__ void stage_w_slice(
 
 }  // namespace
 
-template <DT W>
+template <DT W, int BM_>
 __global__ __launch_bounds__(256) void k_gemm(
     const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
     const float* __restrict__ X,     // [M][K] f32
     const float* __restrict__ res,   // [M][ldc] or null (C col-offset applied)
     float* __restrict__ C,           // [M][ldc]
-    int M, int N, int K, int ldc) {
-    __shared__ __attribute__((aligned(16))) uint16_t Xl[BM * LDW];
+    int M, int N, int K, int ldc, int k_chunk) {
+    // k_chunk: this block's K-range is [z*k_chunk, min((z+1)*k_chunk, K));
+    // splitk > 1 => partial results accumulated with atomicAdd (C pre-zeroed,
+    // residual folded in by the z==0 block).
+    constexpr int FM = BM_ / 32;                 // m-fragments per wave
+    __shared__ __attribute__((aligned(16))) uint16_t Xl[BM_ * LDW];
     __shared__ __attribute__((aligned(16))) uint16_t Wl[BN * LDW];
 
     const int tid = threadIdx.x;
-    const int bn = blockIdx.x, bm = blockIdx.y;
-    const int m0 = bm * BM, n0 = bn * BN;
+    const int bn = blockIdx.x, bm = blockIdx.y, bz = blockIdx.z;
+    const bool splitk = gridDim.z > 1;
+    const int m0 = bm * BM_, n0 = bn * BN;
+    const int kb_lo = bz * k_chunk;
+    const int kb_hi = min(kb_lo + k_chunk, K);
     const int wid = tid >> 6, lane = tid & 63;
     const int wm = wid >> 1, wn = wid & 1;       // 2x2 wave grid
     const int lrow = lane & 15, lk = lane >> 4;  // fragment lane coords
 
-    f32x4 acc[4][4];
+    f32x4 acc[FM][4];
     #pragma unroll
-    for (int i = 0; i < 4; i++)
+    for (int i = 0; i < FM; i++)
         #pragma unroll
         for (int j = 0; j < 4; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-    const int srow = tid >> 1, sh = tid & 1;     // staging coords
-
-    for (int kb = 0; kb < K; kb += BK) {
-        // ---- stage X tile (f32 -> bf16) ----
+    for (int kb = kb_lo; kb < kb_hi; kb += BK) {
+        // ---- stage X tile (f32 -> bf16), 8-elem units ----
         {
-            uint16_t tmp[32];
-            const int gm = m0 + srow;
-            if (gm < M) {
-                const float4* src = reinterpret_cast<const float4*>(
-                    X + (size_t)gm * K + kb + sh * 32);
-                #pragma unroll
-                for (int j = 0; j < 8; j++) {
-                    const float4 v = src[j];
-                    tmp[j * 4 + 0] = f32_to_bf16b(v.x);
-                    tmp[j * 4 + 1] = f32_to_bf16b(v.y);
-                    tmp[j * 4 + 2] = f32_to_bf16b(v.z);
-                    tmp[j * 4 + 3] = f32_to_bf16b(v.w);
-                }
-            } else {
-                #pragma unroll
-                for (int j = 0; j < 32; j++) tmp[j] = 0;
-            }
-            uint4* dst = reinterpret_cast<uint4*>(
-                Xl + srow * LDW + sh * 32);
+            constexpr int UNITS = BM_ * BK / 8;   // 8 bf16 per unit
             #pragma unroll
-            for (int j = 0; j < 4; j++)
-                dst[j] = reinterpret_cast<const uint4*>(tmp)[j];
+            for (int ui = 0; ui < (UNITS + 255) / 256; ui++) {
+                const int unit = tid + ui * 256;
+                if (UNITS < 256 && unit >= UNITS) break;
+                const int row = unit >> 3, seg = unit & 7;
+                const int gm = m0 + row;
+                uint16_t tmp[8];
+                if (gm < M) {
+                    const float4* src = reinterpret_cast<const float4*>(
+                        X + (size_t)gm * K + kb + seg * 8);
+                    #pragma unroll
+                    for (int j = 0; j < 2; j++) {
+                        const float4 v = src[j];
+                        tmp[j * 4 + 0] = f32_to_bf16b(v.x);
+                        tmp[j * 4 + 1] = f32_to_bf16b(v.y);
+                        tmp[j * 4 + 2] = f32_to_bf16b(v.z);
+                        tmp[j * 4 + 3] = f32_to_bf16b(v.w);
+                    }
+                } else {
+                    #pragma unroll
+                    for (int j = 0; j < 8; j++) tmp[j] = 0;
+                }
+                *reinterpret_cast<uint4*>(Xl + row * LDW + seg * 8) =
+                    *reinterpret_cast<const uint4*>(tmp);
+            }
         }
         // ---- stage + dequant W tile ----
         {
+            const int srow = tid >> 1, sh = tid & 1;
             uint16_t tmp[32];
             const int64_t gn = (int64_t)n0 + srow;
             if (gn < N) {
@@ -206,7 +216,6 @@ __global__ __launch_bounds__(256) void k_gemm(
                 for (int j = 0; j < 32; j++) tmp[j] = 0;
             }
             if constexpr (W == DT::DQ4K) {
-                // slice covers klocal [h*16, h*16+16) and [32+h*16, ...)
                 uint4* d0 = reinterpret_cast<uint4*>(
                     Wl + srow * LDW + sh * 16);
                 uint4* d1 = reinterpret_cast<uint4*>(
@@ -227,10 +236,10 @@ __global__ __launch_bounds__(256) void k_gemm(
         // ---- MFMA over the tile ----
         #pragma unroll
         for (int ks = 0; ks < BK; ks += 32) {
-            bf16x8 a[4], b[4];
+            bf16x8 a[FM], b[4];
             #pragma unroll
-            for (int i = 0; i < 4; i++) {
-                const int xr = wm * 64 + i * 16 + lrow;
+            for (int i = 0; i < FM; i++) {
+                const int xr = wm * (BM_ / 2) + i * 16 + lrow;
                 a[i] = *reinterpret_cast<const bf16x8*>(
                     Xl + xr * LDW + ks + lk * 8);
             }
@@ -241,7 +250,7 @@ __global__ __launch_bounds__(256) void k_gemm(
                     Wl + wr * LDW + ks + lk * 8);
             }
             #pragma unroll
-            for (int i = 0; i < 4; i++)
+            for (int i = 0; i < FM; i++)
                 #pragma unroll
                 for (int j = 0; j < 4; j++)
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -250,19 +259,24 @@ __global__ __launch_bounds__(256) void k_gemm(
         __syncthreads();
     }
 
-    // ---- epilogue: C[m][n] (+res) ----
+    // ---- epilogue ----
     #pragma unroll
-    for (int i = 0; i < 4; i++) {
+    for (int i = 0; i < FM; i++) {
         #pragma unroll
         for (int r = 0; r < 4; r++) {
-            const int m = m0 + wm * 64 + i * 16 + lk * 4 + r;
+            const int m = m0 + wm * (BM_ / 2) + i * 16 + lk * 4 + r;
             if (m >= M) continue;
             #pragma unroll
             for (int j = 0; j < 4; j++) {
                 const int n = n0 + wn * 64 + j * 16 + lrow;
                 if (n >= N) continue;
                 const size_t idx = (size_t)m * ldc + n;
-                C[idx] = acc[i][j][r] + (res ? res[idx] : 0.f);
+                const float rv = (res && (!splitk || bz == 0)) ? res[idx] : 0.f;
+                if (splitk) {
+                    atomicAdd(&C[idx], acc[i][j][r] + rv);
+                } else {
+                    C[idx] = acc[i][j][r] + rv;
+                }
             }
         }
     }
@@ -454,15 +468,44 @@ __global__ __launch_bounds__(256) void k_attn_prefill(
         default: throw std::runtime_error("bad dtype");               \
     }
 
+bool gemm_uses_splitk(int N, int K, int M) {
+    if (M > 32) return false;
+    const int n_tiles = (N + BN - 1) / BN;
+    int splitk = K / BK < 512 / n_tiles ? K / BK : 512 / n_tiles;
+    return splitk > 1;
+}
+
 void launch_gemm(const WTensor& w, const float* X, const float* res, float* C,
                  int M, int ldc, hipStream_t stream) {
     const int N = (int)w.n, K = (int)w.k;
     if (K % BK != 0) throw std::runtime_error("gemm: K must be /64");
-    dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM), block(256);
+    const bool small_m = M <= 32;
+    const int bm_tiles = small_m ? 1 : (M + BM - 1) / BM;
+    const int n_tiles = (N + BN - 1) / BN;
+    // split-K keeps the chip full when M is small (decode batches): target
+    // >=512 workgroups, bounded by the number of K-steps.
+    int splitk = 1;
+    if (small_m) {
+        splitk = K / BK < 512 / (n_tiles ? n_tiles : 1)
+                     ? K / BK
+                     : 512 / (n_tiles ? n_tiles : 1);
+        if (splitk < 1) splitk = 1;
+    }
+    const int k_chunk = ((K / BK + splitk - 1) / splitk) * BK;
+    dim3 grid(n_tiles, bm_tiles, splitk), block(256);
+    // splitk > 1 accumulates with atomicAdd: caller must pre-zero C
+    // (gemm_uses_splitk tells it whether that is needed).
     #define GEMM_CASE(WT)                                                      \
-        hipLaunchKernelGGL(k_gemm<WT>, grid, block, 0, stream,                 \
-            (const uint8_t*)w.qs, (const uint8_t*)w.hdr, X, res, C, M, N, K,   \
-            ldc)
+        do {                                                                   \
+            if (small_m)                                                       \
+                hipLaunchKernelGGL((k_gemm<WT, 32>), grid, block, 0, stream,   \
+                    (const uint8_t*)w.qs, (const uint8_t*)w.hdr, X, res, C,    \
+                    M, N, K, ldc, k_chunk);                                    \
+            else                                                               \
+                hipLaunchKernelGGL((k_gemm<WT, 128>), grid, block, 0, stream,  \
+                    (const uint8_t*)w.qs, (const uint8_t*)w.hdr, X, res, C,    \
+                    M, N, K, ldc, k_chunk);                                    \
+        } while (0)
     DISPATCH_DT_GEMM(w.dtype, GEMM_CASE);
     #undef GEMM_CASE
 }

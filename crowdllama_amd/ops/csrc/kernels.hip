@@ -50,6 +50,15 @@ __device__ __forceinline__ float wave_reduce_sum(float v) {
 
 enum Pre : int { PRE_NONE = 0, PRE_RMS = 1, PRE_SILU = 2 };
 
+// LDS address map for staged activations: +16B pad per 64 floats breaks the
+// systematic bank collision of 16/64-aligned chunk bases across a wave's
+// lanes (ds_read_b128 bank = dword % 64; unpadded, every lane's k0 % 64 is
+// one of {0,16,32,48} -> up-to-8-way conflicts per lane group).
+__device__ __forceinline__ int xpad(int k) { return k + ((k >> 6) << 2); }
+__device__ __forceinline__ constexpr int xpad_size(int k) {
+    return k + (k >> 4);
+}
+
 // Weights per 16-B qs chunk for each device dtype.
 template <DT W> struct ChunkTraits;
 template <> struct ChunkTraits<DT::DQ4K> { static constexpr int W_PER_CHUNK = 32; };
@@ -187,15 +196,16 @@ __global__ __launch_bounds__(256) void k_gemv(
     const float* __restrict__ res, float* __restrict__ y,
     int N, int K, int B, int ldy, float eps) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    float* x_lds = reinterpret_cast<float*>(smem);           // [B][K]
-    float* red = x_lds + (size_t)B * K;                      // [256]
+    float* x_lds = reinterpret_cast<float*>(smem);     // [B][xpad_size(K)]
+    const int KP = xpad_size(K);
+    float* red = x_lds + (size_t)B * KP;               // [256]
 
     const int tid = threadIdx.x;
     // ---- stage activations (vectorized float4: K is always %4==0) ----
     const int K4 = K >> 2;
     for (int b = 0; b < B; b++) {
         float ss = 0.f;
-        float4* xl4 = reinterpret_cast<float4*>(x_lds + (size_t)b * K);
+        float* xlb = x_lds + (size_t)b * KP;
         if constexpr (P == PRE_SILU) {
             const float4* g4 = reinterpret_cast<const float4*>(
                 xin + (size_t)b * 2 * K);
@@ -208,14 +218,14 @@ __global__ __launch_bounds__(256) void k_gemv(
                 o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
                 o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
                 o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
-                xl4[k] = o;
+                *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = o;
             }
         } else {
             const float4* x4 = reinterpret_cast<const float4*>(
                 xin + (size_t)b * K);
             for (int k = tid; k < K4; k += 256) {
                 const float4 v = x4[k];
-                xl4[k] = v;
+                *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = v;
                 if constexpr (P == PRE_RMS)
                     ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
             }
@@ -230,14 +240,14 @@ __global__ __launch_bounds__(256) void k_gemv(
             }
             const float inv = rsqrtf(red[0] / (float)K + eps);
             __syncthreads();
-            float4* xl4 = reinterpret_cast<float4*>(x_lds + (size_t)b * K);
             const float4* gw4 = reinterpret_cast<const float4*>(gw);
             for (int k = tid; k < K4; k += 256) {
                 const float4 g = gw4[k];
-                float4 v = xl4[k];
+                float4* vp = reinterpret_cast<float4*>(xlb + xpad(k * 4));
+                float4 v = *vp;
                 v.x *= inv * g.x; v.y *= inv * g.y;
                 v.z *= inv * g.z; v.w *= inv * g.w;
-                xl4[k] = v;
+                *vp = v;
             }
         }
     }
@@ -265,9 +275,10 @@ __global__ __launch_bounds__(256) void k_gemv(
         #pragma unroll 2
         for (int b = 0; b < B; b++) {
             // k0 is a multiple of 16 weights for every dtype's chunk map,
-            // so float4 LDS reads (ds_read_b128) are aligned.
+            // so float4 LDS reads (ds_read_b128) are aligned; the DQ4K hi
+            // run (k0+32) stays inside the same 64-float pad group.
             const float4* xb4 = reinterpret_cast<const float4*>(
-                x_lds + (size_t)b * K + k0);
+                x_lds + (size_t)b * KP + xpad(k0));
             float s = 0.f;
             if constexpr (W == DT::DQ4K) {
                 #pragma unroll
@@ -639,7 +650,7 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
                  hipStream_t stream) {
     const int N = (int)w.n, K = (int)w.k;
     if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
-    const size_t lds = (size_t)B * K * 4 + 256 * 4;
+    const size_t lds = (size_t)B * (K + (K >> 4)) * 4 + 256 * 4;
     dim3 grid((N + 3) / 4), block(256);
     #define GEMV_CASE(WT)                                                        \
         do {                                                                     \
