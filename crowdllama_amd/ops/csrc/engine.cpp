@@ -6,9 +6,19 @@
 #include <functional>
 #include <thread>
 
+#include <rccl/rccl.h>
+
 #include "gguf.h"
 
 namespace cla {
+
+#define NCCL_CHECK(expr)                                                      \
+    do {                                                                      \
+        ncclResult_t _r = (expr);                                             \
+        if (_r != ncclSuccess)                                                \
+            throw std::runtime_error(std::string("RCCL error: ") +            \
+                                     ncclGetErrorString(_r));                 \
+    } while (0)
 
 // kernel launchers (kernels.hip)
 void launch_gemv(const WTensor&, int pre, const float* xin, const float* gw,
@@ -154,6 +164,22 @@ void repack(int32_t ggml_type, const uint8_t* src, int64_t rows, int64_t k,
     }
 }
 
+// Slice columns [c0, c1) out of quantized rows (block-aligned: c0/c1 are
+// multiples of the format's block; used for tensor-parallel col shards).
+std::vector<uint8_t> slice_cols(int32_t ggml_type, const uint8_t* src,
+                                int64_t rows, int64_t K, int64_t c0,
+                                int64_t c1) {
+    const int64_t rb = ggml_row_bytes(ggml_type, K);
+    const int64_t b0 = ggml_row_bytes(ggml_type, c0);
+    const int64_t bl = ggml_row_bytes(ggml_type, c1 - c0);
+    std::vector<uint8_t> out((size_t)rows * bl);
+    parallel_for(rows, [&](int64_t lo, int64_t hi) {
+        for (int64_t r = lo; r < hi; r++)
+            std::memcpy(out.data() + r * bl, src + r * rb + b0, bl);
+    });
+    return out;
+}
+
 }  // namespace
 
 Engine::Engine(const std::string& gguf_path, const EngineConfig& cfg)
@@ -178,6 +204,25 @@ Engine::Engine(const std::string& gguf_path, const EngineConfig& cfg)
         auto& t = gf.tensor("token_embd.weight");
         meta_.vocab = (int)t.shape[0];
     }
+    const int tp = cfg_.tp_size;
+    if (tp > 1) {
+        if (meta_.kv_heads % tp || meta_.heads % tp || meta_.ffn % tp ||
+            meta_.vocab % tp)
+            throw std::runtime_error("model dims not divisible by tp_size");
+        if ((meta_.hidden / tp) % 256 || (meta_.ffn / tp) % 256)
+            throw std::runtime_error("tp column shards not 256-aligned");
+        if (cfg_.nccl_id.size() != sizeof(ncclUniqueId))
+            throw std::runtime_error("tp_size > 1 requires nccl_id bytes");
+        ncclUniqueId id;
+        std::memcpy(&id, cfg_.nccl_id.data(), sizeof(id));
+        ncclComm_t comm;
+        NCCL_CHECK(ncclCommInitRank(&comm, tp, id, cfg_.tp_rank));
+        comm_ = comm;
+    }
+    meta_.heads_l = meta_.heads / tp;
+    meta_.kv_heads_l = meta_.kv_heads / tp;
+    meta_.ffn_l = meta_.ffn / tp;
+    meta_.vocab_l = meta_.vocab / tp;
     load_weights(gf);
     alloc_state();
     reset();
@@ -185,38 +230,81 @@ Engine::Engine(const std::string& gguf_path, const EngineConfig& cfg)
 
 Engine::~Engine() {
     if (graph_exec_) hipGraphExecDestroy(graph_exec_);
+    if (comm_) ncclCommDestroy((ncclComm_t)comm_);
     for (void* p : allocs_) hipFree(p);
     if (stream_) hipStreamDestroy(stream_);
 }
 
 WTensor Engine::upload_tensor(const GGUFFile& gf, const std::string& name) {
-    const GGUFTensor& t = gf.tensor(name);
-    const int64_t k = t.shape.back();
-    int64_t rows = 1;
-    for (size_t i = 0; i + 1 < t.shape.size(); i++) rows *= t.shape[i];
+    return upload_shard(gf, name, -1, -1, -1, -1);
+}
+
+namespace {
+struct HostPack {
+    DT dtype;
+    int64_t rows, k;
+    std::vector<uint8_t> qs, hdr;
+};
+
+HostPack repack_host(const GGUFTensor& t, int64_t r0, int64_t r1, int64_t c0,
+                     int64_t c1) {
+    const int64_t k_full = t.shape.back();
+    int64_t rows_full = 1;
+    for (size_t i = 0; i + 1 < t.shape.size(); i++) rows_full *= t.shape[i];
+    if (r0 < 0) { r0 = 0; r1 = rows_full; }
+    if (c0 < 0) { c0 = 0; c1 = k_full; }
+    HostPack hp;
+    hp.rows = r1 - r0;
+    hp.k = c1 - c0;
+    hp.dtype = device_dtype(t.ggml_type);
+    const int64_t src_rb = ggml_row_bytes(t.ggml_type, k_full);
+    const uint8_t* src = t.data + r0 * src_rb;
+    std::vector<uint8_t> sliced;
+    if (c0 != 0 || c1 != k_full) {
+        sliced = slice_cols(t.ggml_type, src, hp.rows, k_full, c0, c1);
+        src = sliced.data();
+    }
+    hp.qs.resize(dqs_row_bytes(hp.dtype, hp.k) * hp.rows);
+    const int64_t hb = dhdr_row_bytes(hp.dtype, hp.k) * hp.rows;
+    hp.hdr.resize(hb ? hb : 1);
+    repack(t.ggml_type, src, hp.rows, hp.k, hp.qs.data(), hp.hdr.data());
+    return hp;
+}
+}  // namespace
+
+WTensor Engine::upload_pack(const void* qs, size_t qs_bytes, const void* hdr,
+                            size_t hdr_bytes, DT dtype, int64_t rows,
+                            int64_t k) {
     WTensor w;
-    w.dtype = device_dtype(t.ggml_type);
+    w.dtype = dtype;
     w.n = rows;
     w.k = k;
-    const int64_t qs_bytes = dqs_row_bytes(w.dtype, k) * rows;
-    const int64_t hdr_bytes = dhdr_row_bytes(w.dtype, k) * rows;
-    std::vector<uint8_t> h_qs(qs_bytes), h_hdr(hdr_bytes ? hdr_bytes : 1);
-    repack(t.ggml_type, t.data, rows, k, h_qs.data(), h_hdr.data());
     void* d_qs = nullptr;
     HIP_CHECK(hipMalloc(&d_qs, qs_bytes));
-    HIP_CHECK(hipMemcpy(d_qs, h_qs.data(), qs_bytes, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(d_qs, qs, qs_bytes, hipMemcpyHostToDevice));
     allocs_.push_back(d_qs);
     w.qs = d_qs;
     vram_bytes_ += qs_bytes;
     if (hdr_bytes) {
         void* d_hdr = nullptr;
         HIP_CHECK(hipMalloc(&d_hdr, hdr_bytes));
-        HIP_CHECK(hipMemcpy(d_hdr, h_hdr.data(), hdr_bytes, hipMemcpyHostToDevice));
+        HIP_CHECK(hipMemcpy(d_hdr, hdr, hdr_bytes, hipMemcpyHostToDevice));
         allocs_.push_back(d_hdr);
         w.hdr = d_hdr;
         vram_bytes_ += hdr_bytes;
     }
     return w;
+}
+
+// Upload a (possibly sharded) weight matrix: rows [r0,r1), cols [c0,c1);
+// -1 = full range. Column shards are block-aligned byte slices.
+WTensor Engine::upload_shard(const GGUFFile& gf, const std::string& name,
+                             int64_t r0, int64_t r1, int64_t c0, int64_t c1) {
+    HostPack hp = repack_host(gf.tensor(name), r0, r1, c0, c1);
+    return upload_pack(hp.qs.data(), hp.qs.size(),
+                       dhdr_row_bytes(hp.dtype, hp.k) ? hp.hdr.data() : nullptr,
+                       dhdr_row_bytes(hp.dtype, hp.k) * hp.rows,
+                       hp.dtype, hp.rows, hp.k);
 }
 
 const float* Engine::upload_norm(const GGUFFile& gf, const std::string& name) {
@@ -231,56 +319,52 @@ const float* Engine::upload_norm(const GGUFFile& gf, const std::string& name) {
     return reinterpret_cast<const float*>(d);
 }
 
-Proj Engine::load_proj(const GGUFFile& gf, const std::vector<std::string>& names) {
-    Proj p;
-    // Merge row-blocks with identical dtype+k into one device tensor so a
-    // single GEMV launch covers them (q/k/v and gate/up stacking).
-    bool mergeable = names.size() > 1;
-    int32_t t0 = gf.tensor(names[0]).ggml_type;
-    int64_t k0 = gf.tensor(names[0]).shape.back();
+// Load several row-blocks as one projection; each name contributes its
+// [rank] row shard when row_shard is set. Parts with identical dtype+K are
+// merged into one device tensor so a single kernel launch covers them.
+Proj Engine::load_proj(const GGUFFile& gf, const std::vector<std::string>& names,
+                       bool row_shard, int64_t c0, int64_t c1) {
+    const int tp = cfg_.tp_size, rank = cfg_.tp_rank;
+    std::vector<HostPack> packs;
     for (auto& nm : names) {
         const auto& t = gf.tensor(nm);
-        if (t.ggml_type != t0 || t.shape.back() != k0) mergeable = false;
+        int64_t rows = 1;
+        for (size_t i = 0; i + 1 < t.shape.size(); i++) rows *= t.shape[i];
+        int64_t pr0 = -1, pr1 = -1;
+        if (row_shard && tp > 1) {
+            const int64_t per = rows / tp;
+            pr0 = rank * per;
+            pr1 = pr0 + per;
+        }
+        packs.push_back(repack_host(t, pr0, pr1, c0, c1));
     }
+    Proj p;
+    bool mergeable = packs.size() > 1;
+    for (auto& hp : packs)
+        if (hp.dtype != packs[0].dtype || hp.k != packs[0].k) mergeable = false;
     if (mergeable) {
-        // concatenate repacked rows
-        DT dt = device_dtype(t0);
+        std::vector<uint8_t> qs, hdr;
         int64_t rows = 0;
-        for (auto& nm : names) rows += gf.tensor(nm).shape[0];
-        const int64_t qs_rb = dqs_row_bytes(dt, k0);
-        const int64_t hdr_rb = dhdr_row_bytes(dt, k0);
-        std::vector<uint8_t> h_qs(qs_rb * rows), h_hdr(hdr_rb * rows + 1);
-        int64_t roff = 0;
-        for (auto& nm : names) {
-            const auto& t = gf.tensor(nm);
-            repack(t0, t.data, t.shape[0], k0, h_qs.data() + roff * qs_rb,
-                   h_hdr.data() + roff * hdr_rb);
-            roff += t.shape[0];
+        for (auto& hp : packs) {
+            qs.insert(qs.end(), hp.qs.begin(), hp.qs.end());
+            if (dhdr_row_bytes(hp.dtype, hp.k))
+                hdr.insert(hdr.end(), hp.hdr.begin(),
+                           hp.hdr.begin() + dhdr_row_bytes(hp.dtype, hp.k) * hp.rows);
+            rows += hp.rows;
         }
-        WTensor w;
-        w.dtype = dt; w.n = rows; w.k = k0;
-        void* d_qs = nullptr;
-        HIP_CHECK(hipMalloc(&d_qs, h_qs.size()));
-        HIP_CHECK(hipMemcpy(d_qs, h_qs.data(), h_qs.size(), hipMemcpyHostToDevice));
-        allocs_.push_back(d_qs);
-        w.qs = d_qs;
-        vram_bytes_ += h_qs.size();
-        if (hdr_rb) {
-            void* d_hdr = nullptr;
-            HIP_CHECK(hipMalloc(&d_hdr, hdr_rb * rows));
-            HIP_CHECK(hipMemcpy(d_hdr, h_hdr.data(), hdr_rb * rows,
-                                hipMemcpyHostToDevice));
-            allocs_.push_back(d_hdr);
-            w.hdr = d_hdr;
-            vram_bytes_ += hdr_rb * rows;
-        }
+        WTensor w = upload_pack(qs.data(), qs.size(),
+                                hdr.empty() ? nullptr : hdr.data(), hdr.size(),
+                                packs[0].dtype, rows, packs[0].k);
         p.parts.push_back({w, 0});
         p.n_total = rows;
         return p;
     }
     int64_t roff = 0;
-    for (auto& nm : names) {
-        WTensor w = upload_tensor(gf, nm);
+    for (auto& hp : packs) {
+        WTensor w = upload_pack(hp.qs.data(), hp.qs.size(),
+                                dhdr_row_bytes(hp.dtype, hp.k) ? hp.hdr.data() : nullptr,
+                                dhdr_row_bytes(hp.dtype, hp.k) * hp.rows,
+                                hp.dtype, hp.rows, hp.k);
         p.parts.push_back({w, roff});
         roff += w.n;
     }
@@ -289,10 +373,20 @@ Proj Engine::load_proj(const GGUFFile& gf, const std::vector<std::string>& names
 }
 
 void Engine::load_weights(const GGUFFile& gf) {
-    embed_ = upload_tensor(gf, "token_embd.weight");
+    const int tp = cfg_.tp_size, rank = cfg_.tp_rank;
+    embed_ = upload_tensor(gf, "token_embd.weight");  // replicated
     out_norm_ = upload_norm(gf, "output_norm.weight");
-    head_ = gf.has_tensor("output.weight") ? upload_tensor(gf, "output.weight")
-                                           : embed_;  // tied embeddings
+    const std::string head_name = gf.has_tensor("output.weight")
+                                      ? "output.weight" : "token_embd.weight";
+    if (tp > 1) {
+        const int64_t per = meta_.vocab / tp;
+        head_ = upload_shard(gf, head_name, (int64_t)rank * per,
+                             (int64_t)(rank + 1) * per, -1, -1);
+    } else {
+        head_ = (head_name == "token_embd.weight")
+                    ? embed_ : upload_tensor(gf, head_name);
+    }
+    const int64_t h = meta_.hidden, f = meta_.ffn;
     layers_.resize(meta_.layers);
     for (int i = 0; i < meta_.layers; i++) {
         const std::string p = "blk." + std::to_string(i) + ".";
@@ -300,17 +394,28 @@ void Engine::load_weights(const GGUFFile& gf) {
         L.attn_norm = upload_norm(gf, p + "attn_norm.weight");
         L.ffn_norm = upload_norm(gf, p + "ffn_norm.weight");
         L.qkv = load_proj(gf, {p + "attn_q.weight", p + "attn_k.weight",
-                               p + "attn_v.weight"});
-        L.o = load_proj(gf, {p + "attn_output.weight"});
-        L.gate_up = load_proj(gf, {p + "ffn_gate.weight", p + "ffn_up.weight"});
-        L.down = load_proj(gf, {p + "ffn_down.weight"});
+                               p + "attn_v.weight"}, /*row_shard=*/true);
+        // o / down: column shards (local input dims), all-reduced after
+        if (tp > 1) {
+            L.o = load_proj(gf, {p + "attn_output.weight"}, false,
+                            (int64_t)rank * (h / tp),
+                            (int64_t)(rank + 1) * (h / tp));
+            L.down = load_proj(gf, {p + "ffn_down.weight"}, false,
+                               (int64_t)rank * (f / tp),
+                               (int64_t)(rank + 1) * (f / tp));
+        } else {
+            L.o = load_proj(gf, {p + "attn_output.weight"}, false);
+            L.down = load_proj(gf, {p + "ffn_down.weight"}, false);
+        }
+        L.gate_up = load_proj(gf, {p + "ffn_gate.weight", p + "ffn_up.weight"},
+                              /*row_shard=*/true);
     }
 }
 
 void Engine::alloc_state() {
     const int B = cfg_.batch, H = meta_.hidden, V = meta_.vocab;
-    const int NH = meta_.heads, NKV = meta_.kv_heads, D = meta_.head_dim;
-    const int F = meta_.ffn;
+    const int NH = meta_.heads_l, NKV = meta_.kv_heads_l, D = meta_.head_dim;
+    const int F = meta_.ffn_l;
     attn_splits_ = std::max(1, std::min(32, 256 / std::max(1, B * NKV)));
     max_pages_ = (cfg_.max_seq + cfg_.page_size - 1) / cfg_.page_size;
     // pool layout: [page][layer][kvh][2][page_size][D] bf16 — one pool, all
@@ -329,6 +434,7 @@ void Engine::alloc_state() {
     x_ = (float*)dalloc((size_t)B * H * 4);
     xn_ = (float*)dalloc((size_t)B * H * 4);
     x2_ = (float*)dalloc((size_t)B * H * 4);
+    tmp_h_ = (float*)dalloc((size_t)B * H * 4);
     act_ = (float*)dalloc((size_t)B * F * 4);
     {   // prefill scratch
         const int Mc = cfg_.prefill_chunk;
@@ -341,6 +447,7 @@ void Engine::alloc_state() {
         gup_ = (float*)dalloc((size_t)Mc * 2 * F * 4);
         actp_ = (float*)dalloc((size_t)Mc * F * 4);
         pids_ = (int32_t*)dalloc((size_t)Mc * 4);
+        tmp_hp_ = (float*)dalloc((size_t)Mc * H * 4);
     }
     qkv_ = (float*)dalloc((size_t)B * (NH + 2 * NKV) * D * 4);
     attn_out_ = (float*)dalloc((size_t)B * NH * D * 4);
@@ -396,9 +503,15 @@ static void gemm_proj(const Proj& p, const float* X, const float* res,
 
 void Engine::step(hipStream_t s) {
     const int B = cfg_.batch;
-    const int NH = meta_.heads, NKV = meta_.kv_heads, D = meta_.head_dim;
+    const int NH = meta_.heads_l, NKV = meta_.kv_heads_l, D = meta_.head_dim;
     const float scale = 1.0f / std::sqrt((float)D);
     const float eps = meta_.rms_eps;
+    const bool tp = cfg_.tp_size > 1;
+    const bool r0 = cfg_.tp_rank == 0;
+    auto allreduce = [&](const float* send, float* recv, size_t n) {
+        NCCL_CHECK(ncclAllReduce(send, recv, n, ncclFloat, ncclSum,
+                                 (ncclComm_t)comm_, s));
+    };
 
     launch_embed(embed_, cur_ids_, x_, B, s);
     const bool gemv_path = B <= 2;
@@ -419,7 +532,7 @@ void Engine::step(hipStream_t s) {
                            cfg_.page_size, max_pages_, page_stride_, scale, s);
         launch_attn_combine(part_o_, part_ml_, attn_out_, B, NH, attn_splits_,
                             D, s);
-        if (gemv_path) {
+        if (gemv_path && !tp) {
             for (auto& pt : L.o.parts)
                 launch_gemv(pt.w, PRE_NONE, attn_out_, nullptr, x_ + pt.row_off,
                             x_ + pt.row_off, B, (int)L.o.n_total, eps, s);
@@ -429,22 +542,62 @@ void Engine::step(hipStream_t s) {
             for (auto& pt : L.down.parts)
                 launch_gemv(pt.w, PRE_SILU, gu_, nullptr, x_ + pt.row_off,
                             x_ + pt.row_off, B, (int)L.down.n_total, eps, s);
+        } else if (gemv_path) {
+            // TP: local partial -> all-reduce; rank 0 folds the residual so
+            // the summed result is residual + sum(partials) on every rank.
+            for (auto& pt : L.o.parts)
+                launch_gemv(pt.w, PRE_NONE, attn_out_, nullptr,
+                            r0 ? x_ + pt.row_off : nullptr, tmp_h_ + pt.row_off,
+                            B, (int)L.o.n_total, eps, s);
+            allreduce(tmp_h_, x2_, (size_t)B * meta_.hidden);
+            for (auto& pt : L.gate_up.parts)
+                launch_gemv(pt.w, PRE_RMS, x2_, L.ffn_norm, nullptr,
+                            gu_ + pt.row_off, B, (int)L.gate_up.n_total, eps, s);
+            for (auto& pt : L.down.parts)
+                launch_gemv(pt.w, PRE_SILU, gu_, nullptr,
+                            r0 ? x2_ + pt.row_off : nullptr, tmp_h_ + pt.row_off,
+                            B, (int)L.down.n_total, eps, s);
+            allreduce(tmp_h_, x_, (size_t)B * meta_.hidden);
         } else {
             // residual ping-pong: x_ --(+attn)--> x2_ --(+ffn)--> x_
-            gemm_proj(L.o, attn_out_, x_, x2_, B, s);
+            if (tp) {
+                gemm_proj(L.o, attn_out_, r0 ? x_ : nullptr, tmp_h_, B, s);
+                allreduce(tmp_h_, x2_, (size_t)B * meta_.hidden);
+            } else {
+                gemm_proj(L.o, attn_out_, x_, x2_, B, s);
+            }
             launch_rmsnorm_rows(x2_, L.ffn_norm, xn_, B, meta_.hidden, eps, s);
             gemm_proj(L.gate_up, xn_, nullptr, gu_, B, s);
-            launch_silu_rows(gu_, act_, B, meta_.ffn, s);
-            gemm_proj(L.down, act_, x2_, x_, B, s);
+            launch_silu_rows(gu_, act_, B, meta_.ffn_l, s);
+            if (tp) {
+                gemm_proj(L.down, act_, r0 ? x2_ : nullptr, tmp_h_, B, s);
+                allreduce(tmp_h_, x_, (size_t)B * meta_.hidden);
+            } else {
+                gemm_proj(L.down, act_, x2_, x_, B, s);
+            }
         }
     }
+    const int64_t voff = (int64_t)cfg_.tp_rank * meta_.vocab_l;
     if (gemv_path) {
-        launch_gemv(head_, PRE_RMS, x_, out_norm_, nullptr, logits_, B,
+        launch_gemv(head_, PRE_RMS, x_, out_norm_, nullptr, logits_ + voff, B,
                     meta_.vocab, eps, s);
     } else {
         launch_rmsnorm_rows(x_, out_norm_, xn_, B, meta_.hidden, eps, s);
-        Proj hp; hp.parts.push_back({head_, 0}); hp.n_total = meta_.vocab;
-        gemm_proj(hp, xn_, nullptr, logits_, B, s);
+        Proj hp; hp.parts.push_back({head_, 0}); hp.n_total = meta_.vocab_l;
+        // write the local vocab slice at its global offset (ldc = full V)
+        bool zero = gemm_uses_splitk((int)head_.n, (int)head_.k, B);
+        if (zero)
+            HIP_CHECK(hipMemsetAsync(logits_, 0, (size_t)B * meta_.vocab * 4, s));
+        launch_gemm(head_, xn_, nullptr, logits_ + voff, B, meta_.vocab, s);
+    }
+    if (tp) {
+        // all-gather the vocab slices (in-place: every rank's send buffer is
+        // its slice of the same recv buffer), one row per batch slot
+        for (int b = 0; b < B; b++)
+            NCCL_CHECK(ncclAllGather(logits_ + (size_t)b * meta_.vocab + voff,
+                                     logits_ + (size_t)b * meta_.vocab,
+                                     meta_.vocab_l, ncclFloat,
+                                     (ncclComm_t)comm_, s));
     }
     launch_argmax(logits_, amax_val_, amax_idx_, cur_ids_, n_past_,
                   gen_tokens_, gen_count_, B, meta_.vocab, cfg_.gen_cap, s);
@@ -465,9 +618,11 @@ void Engine::ensure_graph() {
 void Engine::prefill_chunk_pass(int slot, int pos0, int m) {
     // One GEMM pass over m prompt rows (ids already in pids_).
     hipStream_t s = stream_;
-    const int NH = meta_.heads, NKV = meta_.kv_heads, D = meta_.head_dim;
+    const int NH = meta_.heads_l, NKV = meta_.kv_heads_l, D = meta_.head_dim;
     const float scale = 1.0f / std::sqrt((float)D);
     const float eps = meta_.rms_eps;
+    const bool tp = cfg_.tp_size > 1;
+    const bool r0 = cfg_.tp_rank == 0;
     launch_embed(embed_, pids_, xp_, m, s);
     int li = 0;
     for (auto& L : layers_) {
@@ -481,11 +636,23 @@ void Engine::prefill_chunk_pass(int slot, int pos0, int m) {
         launch_attn_prefill(qkvp_, page_table_, kv_layer, attnp_, slot, pos0,
                             m, NH, NKV, D, cfg_.page_size, max_pages_,
                             page_stride_, scale, s);
-        gemm_proj(L.o, attnp_, xp_, xp2_, m, s);
+        if (tp) {
+            gemm_proj(L.o, attnp_, r0 ? xp_ : nullptr, tmp_hp_, m, s);
+            NCCL_CHECK(ncclAllReduce(tmp_hp_, xp2_, (size_t)m * meta_.hidden,
+                                     ncclFloat, ncclSum, (ncclComm_t)comm_, s));
+        } else {
+            gemm_proj(L.o, attnp_, xp_, xp2_, m, s);
+        }
         launch_rmsnorm_rows(xp2_, L.ffn_norm, xnp_, m, meta_.hidden, eps, s);
         gemm_proj(L.gate_up, xnp_, nullptr, gup_, m, s);
-        launch_silu_rows(gup_, actp_, m, meta_.ffn, s);
-        gemm_proj(L.down, actp_, xp2_, xp_, m, s);
+        launch_silu_rows(gup_, actp_, m, meta_.ffn_l, s);
+        if (tp) {
+            gemm_proj(L.down, actp_, r0 ? xp2_ : nullptr, tmp_hp_, m, s);
+            NCCL_CHECK(ncclAllReduce(tmp_hp_, xp_, (size_t)m * meta_.hidden,
+                                     ncclFloat, ncclSum, (ncclComm_t)comm_, s));
+        } else {
+            gemm_proj(L.down, actp_, xp2_, xp_, m, s);
+        }
     }
 }
 
@@ -509,9 +676,15 @@ void Engine::prefill_slot(int slot, const std::vector<int32_t>& ids) {
     }
     // logits of the LAST prompt row -> slot's logits; sample + advance state
     const int last = ((len - 1) % cfg_.prefill_chunk);
+    const int64_t voff = (int64_t)cfg_.tp_rank * meta_.vocab_l;
     launch_gemv(head_, PRE_RMS, xp_ + (size_t)last * meta_.hidden, out_norm_,
-                nullptr, logits_ + (size_t)slot * meta_.vocab, 1,
+                nullptr, logits_ + (size_t)slot * meta_.vocab + voff, 1,
                 meta_.vocab, eps, stream_);
+    if (cfg_.tp_size > 1)
+        NCCL_CHECK(ncclAllGather(
+            logits_ + (size_t)slot * meta_.vocab + voff,
+            logits_ + (size_t)slot * meta_.vocab, meta_.vocab_l, ncclFloat,
+            (ncclComm_t)comm_, stream_));
     HIP_CHECK(hipStreamSynchronize(stream_));
     // n_past[slot] = len - 1; argmax's advance makes it len.
     const int32_t npast = len - 1, zero = 0;

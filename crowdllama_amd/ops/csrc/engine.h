@@ -24,6 +24,11 @@ struct EngineConfig {
     int device = 0;
     bool use_graph = true;
     int prefill_chunk = 1024;  // prompt tokens per prefill GEMM pass
+    // tensor parallelism over RCCL/xGMI (capability extension; the
+    // reference has no collectives at all — SURVEY.md §2.3)
+    int tp_rank = 0;
+    int tp_size = 1;
+    std::string nccl_id;       // ncclUniqueId bytes (rank 0 creates)
 };
 
 struct ModelMeta {
@@ -31,6 +36,8 @@ struct ModelMeta {
     int vocab = 0, hidden = 0, layers = 0, heads = 0, kv_heads = 0;
     int ffn = 0, head_dim = 0, max_ctx = 0;
     float rope_theta = 10000.f, rms_eps = 1e-5f;
+    // per-rank geometry under tensor parallelism (== globals at tp=1)
+    int heads_l = 0, kv_heads_l = 0, ffn_l = 0, vocab_l = 0;
 };
 
 // One logical projection: a list of row-blocks (merged when dtypes match).
@@ -82,8 +89,13 @@ public:
 
 private:
     void load_weights(const GGUFFile& gf);
-    Proj load_proj(const GGUFFile& gf, const std::vector<std::string>& names);
+    Proj load_proj(const GGUFFile& gf, const std::vector<std::string>& names,
+                   bool row_shard = false, int64_t c0 = -1, int64_t c1 = -1);
     WTensor upload_tensor(const GGUFFile& gf, const std::string& name);
+    WTensor upload_shard(const GGUFFile& gf, const std::string& name,
+                         int64_t r0, int64_t r1, int64_t c0, int64_t c1);
+    WTensor upload_pack(const void* qs, size_t qs_bytes, const void* hdr,
+                        size_t hdr_bytes, DT dtype, int64_t rows, int64_t k);
     const float* upload_norm(const GGUFFile& gf, const std::string& name);
     void alloc_state();
     void step(hipStream_t stream);
@@ -137,6 +149,9 @@ private:
 
     hipGraphExec_t graph_exec_ = nullptr;
     double last_decode_ms_ = 0.0;
+    void* comm_ = nullptr;      // ncclComm_t when tp_size > 1
+    float* tmp_h_ = nullptr;    // [B][hidden] all-reduce staging
+    float* tmp_hp_ = nullptr;   // [Mchunk][hidden] prefill all-reduce staging
 };
 
 }  // namespace cla

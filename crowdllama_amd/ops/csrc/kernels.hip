@@ -198,10 +198,28 @@ __global__ __launch_bounds__(256) void k_gemv(
     extern __shared__ __attribute__((aligned(16))) char smem[];
     float* x_lds = reinterpret_cast<float*>(smem);     // [B][xpad_size(K)]
     const int KP = xpad_size(K);
-    float* red = x_lds + (size_t)B * KP;               // [256]
+    float* red = x_lds + (size_t)B * KP;               // [8]
 
     const int tid = threadIdx.x;
-    // ---- stage activations (vectorized float4: K is always %4==0) ----
+    const int wave = tid >> 6, lane = tid & 63;
+    // 2 rows per wave: amortize the staging latency over more weight
+    // traffic and keep 2x the loads in flight.
+    const int r[2] = {blockIdx.x * 8 + wave * 2, blockIdx.x * 8 + wave * 2 + 1};
+    const int64_t qs_rb = dqs_row_bytes(W, K);
+    const int64_t hdr_rb = dhdr_row_bytes(W, K);
+    const int n_chunks = (int)(qs_rb / 16);
+    constexpr int WPC = ChunkTraits<W>::W_PER_CHUNK;
+
+    // issue the first weight chunks BEFORE staging: they have no
+    // dependency on x, so their HBM latency hides under the staging phase.
+    ChunkRaw<W> cur[2], nxt[2];
+    #pragma unroll
+    for (int i = 0; i < 2; i++)
+        if (r[i] < N && lane < n_chunks)
+            load_chunk<W>(qs + (int64_t)r[i] * qs_rb,
+                          hdr + (int64_t)r[i] * hdr_rb, lane, &cur[i]);
+
+    // ---- stage activations (vectorized float4; K always %4==0) ----
     const int K4 = K >> 2;
     for (int b = 0; b < B; b++) {
         float ss = 0.f;
@@ -231,15 +249,12 @@ __global__ __launch_bounds__(256) void k_gemv(
             }
         }
         if constexpr (P == PRE_RMS) {
-            red[tid] = ss;
+            // wave-level reduce + one cross-wave pass (2 barriers total)
+            const float ws = wave_reduce_sum(ss);
+            if (lane == 0) red[wave] = ws;
             __syncthreads();
-            #pragma unroll
-            for (int off = 128; off > 0; off >>= 1) {
-                if (tid < off) red[tid] += red[tid + off];
-                __syncthreads();
-            }
-            const float inv = rsqrtf(red[0] / (float)K + eps);
-            __syncthreads();
+            const float inv = rsqrtf(
+                (red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
             const float4* gw4 = reinterpret_cast<const float4*>(gw);
             for (int k = tid; k < K4; k += 256) {
                 const float4 g = gw4[k];
@@ -253,61 +268,58 @@ __global__ __launch_bounds__(256) void k_gemv(
     }
     __syncthreads();
 
-    // ---- per-wave row sweep ----
-    const int wave = tid >> 6, lane = tid & 63;
-    const int r = blockIdx.x * 4 + wave;
-    if (r >= N) return;
-    const int64_t qs_rb = dqs_row_bytes(W, K);
-    const int64_t hdr_rb = dhdr_row_bytes(W, K);
-    const uint8_t* qs_row = qs + (int64_t)r * qs_rb;
-    const uint8_t* hdr_row = hdr + (int64_t)r * hdr_rb;
-    const int n_chunks = (int)(qs_rb / 16);
-    constexpr int WPC = ChunkTraits<W>::W_PER_CHUNK;
-
-    float acc[2] = {0.f, 0.f};  // GEMV path supports B <= 2
-    ChunkRaw<W> cur, nxt;
-    if (lane < n_chunks) load_chunk<W>(qs_row, hdr_row, lane, &cur);
+    // ---- per-wave sweep over 2 rows ----
+    float acc[2][2] = {{0.f, 0.f}, {0.f, 0.f}};  // [row][b], B <= 2
     for (int c = lane; c < n_chunks; c += 64) {
-        if (c + 64 < n_chunks) load_chunk<W>(qs_row, hdr_row, c + 64, &nxt);
-        float w[WPC];
-        int k0;
-        decode_chunk_raw<W>(cur, c, w, &k0);
+        #pragma unroll
+        for (int i = 0; i < 2; i++)
+            if (r[i] < N && c + 64 < n_chunks)
+                load_chunk<W>(qs + (int64_t)r[i] * qs_rb,
+                              hdr + (int64_t)r[i] * hdr_rb, c + 64, &nxt[i]);
+        #pragma unroll
+        for (int i = 0; i < 2; i++) {
+            if (r[i] >= N) continue;
+            float w[WPC];
+            int k0;
+            decode_chunk_raw<W>(cur[i], c, w, &k0);
+            #pragma unroll 2
+            for (int b = 0; b < B; b++) {
+                const float4* xb4 = reinterpret_cast<const float4*>(
+                    x_lds + (size_t)b * KP + xpad(k0));
+                float s = 0.f;
+                if constexpr (W == DT::DQ4K) {
+                    #pragma unroll
+                    for (int t4 = 0; t4 < 4; t4++) {
+                        const float4 xl = xb4[t4];
+                        const float4 xh = xb4[8 + t4];
+                        s += w[t4 * 4 + 0] * xl.x + w[t4 * 4 + 1] * xl.y
+                           + w[t4 * 4 + 2] * xl.z + w[t4 * 4 + 3] * xl.w;
+                        s += w[16 + t4 * 4 + 0] * xh.x + w[16 + t4 * 4 + 1] * xh.y
+                           + w[16 + t4 * 4 + 2] * xh.z + w[16 + t4 * 4 + 3] * xh.w;
+                    }
+                } else {
+                    #pragma unroll
+                    for (int t4 = 0; t4 < WPC / 4; t4++) {
+                        const float4 xv = xb4[t4];
+                        s += w[t4 * 4 + 0] * xv.x + w[t4 * 4 + 1] * xv.y
+                           + w[t4 * 4 + 2] * xv.z + w[t4 * 4 + 3] * xv.w;
+                    }
+                }
+                acc[i][b] += s;
+            }
+            cur[i] = nxt[i];
+        }
+    }
+    #pragma unroll
+    for (int i = 0; i < 2; i++) {
+        if (r[i] >= N) continue;
         #pragma unroll 2
         for (int b = 0; b < B; b++) {
-            // k0 is a multiple of 16 weights for every dtype's chunk map,
-            // so float4 LDS reads (ds_read_b128) are aligned; the DQ4K hi
-            // run (k0+32) stays inside the same 64-float pad group.
-            const float4* xb4 = reinterpret_cast<const float4*>(
-                x_lds + (size_t)b * KP + xpad(k0));
-            float s = 0.f;
-            if constexpr (W == DT::DQ4K) {
-                #pragma unroll
-                for (int t4 = 0; t4 < 4; t4++) {
-                    const float4 xl = xb4[t4];
-                    const float4 xh = xb4[8 + t4];
-                    s += w[t4 * 4 + 0] * xl.x + w[t4 * 4 + 1] * xl.y
-                       + w[t4 * 4 + 2] * xl.z + w[t4 * 4 + 3] * xl.w;
-                    s += w[16 + t4 * 4 + 0] * xh.x + w[16 + t4 * 4 + 1] * xh.y
-                       + w[16 + t4 * 4 + 2] * xh.z + w[16 + t4 * 4 + 3] * xh.w;
-                }
-            } else {
-                #pragma unroll
-                for (int t4 = 0; t4 < WPC / 4; t4++) {
-                    const float4 xv = xb4[t4];
-                    s += w[t4 * 4 + 0] * xv.x + w[t4 * 4 + 1] * xv.y
-                       + w[t4 * 4 + 2] * xv.z + w[t4 * 4 + 3] * xv.w;
-                }
+            float v = wave_reduce_sum(acc[i][b]);
+            if (lane == 0) {
+                const int64_t idx = (int64_t)b * ldy + r[i];
+                y[idx] = v + (res ? res[idx] : 0.f);
             }
-            acc[b] += s;
-        }
-        cur = nxt;
-    }
-    #pragma unroll 2
-    for (int b = 0; b < B; b++) {
-        float v = wave_reduce_sum(acc[b]);
-        if (lane == 0) {
-            const int64_t idx = (int64_t)b * ldy + r;
-            y[idx] = v + (res ? res[idx] : 0.f);
         }
     }
 }
@@ -650,8 +662,8 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
                  hipStream_t stream) {
     const int N = (int)w.n, K = (int)w.k;
     if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
-    const size_t lds = (size_t)B * (K + (K >> 4)) * 4 + 256 * 4;
-    dim3 grid((N + 3) / 4), block(256);
+    const size_t lds = (size_t)B * (K + (K >> 4)) * 4 + 8 * 4;
+    dim3 grid((N + 7) / 8), block(256);
     #define GEMV_CASE(WT)                                                        \
         do {                                                                     \
             auto kern = (pre == PRE_RMS) ? k_gemv<WT, PRE_RMS>                   \

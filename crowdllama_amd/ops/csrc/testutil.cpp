@@ -96,6 +96,7 @@ void launch_gemm_test(const void* qs, const void* hdr, const float* x,
     HIP_CHECK(hipMalloc(&d_x, (size_t)M * K * 4));
     HIP_CHECK(hipMemcpy(d_x, x, (size_t)M * K * 4, hipMemcpyHostToDevice));
     HIP_CHECK(hipMalloc(&d_y, (size_t)M * N * 4));
+    HIP_CHECK(hipMemset(d_y, 0, (size_t)M * N * 4));  // split-K accumulates
     WTensor w;
     w.dtype = dt; w.n = N; w.k = K; w.qs = d_qs; w.hdr = d_hdr;
     launch_gemm(w, (const float*)d_x, nullptr, (float*)d_y, M, N, nullptr);
@@ -119,6 +120,7 @@ double bench_gemm(int dtype, int M, int N, int K, int iters) {
     HIP_CHECK(hipMalloc(&d_x, (size_t)M * K * 4));
     HIP_CHECK(hipMemset(d_x, 0, (size_t)M * K * 4));
     HIP_CHECK(hipMalloc(&d_y, (size_t)M * N * 4));
+    HIP_CHECK(hipMemset(d_y, 0, (size_t)M * N * 4));
     WTensor w;
     w.dtype = dt; w.n = N; w.k = K; w.qs = d_qs; w.hdr = d_hdr;
     for (int i = 0; i < 3; i++)
