@@ -204,7 +204,8 @@ __global__ __launch_bounds__(256) void k_gemv(
     const int wave = tid >> 6, lane = tid & 63;
     // 2 rows per wave: amortize the staging latency over more weight
     // traffic and keep 2x the loads in flight.
-    const int r[2] = {blockIdx.x * 8 + wave * 2, blockIdx.x * 8 + wave * 2 + 1};
+    const int rbase = (int)blockIdx.x * 8 + wave * 2;
+    const int r[2] = {rbase, rbase + 1};
     const int64_t qs_rb = dqs_row_bytes(W, K);
     const int64_t hdr_rb = dhdr_row_bytes(W, K);
     const int n_chunks = (int)(qs_rb / 16);
