@@ -68,7 +68,7 @@ def build(verbose: bool = True, force: bool = False) -> str:
         if verbose and err.strip():
             print(err, file=sys.stderr)
     link = [hipcc, f"--offload-arch={ARCH}", "-shared", "-fPIC", *objs,
-            "-o", so]
+            "-L/opt/rocm/lib", "-lrccl", "-o", so]
     if verbose:
         print("[ops.build]", " ".join(link), file=sys.stderr)
     r = subprocess.run(link, capture_output=True, text=True)

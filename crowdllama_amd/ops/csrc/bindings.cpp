@@ -4,6 +4,8 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <rccl/rccl.h>
+
 #include "engine.h"
 
 namespace py = pybind11;
@@ -22,6 +24,13 @@ double bench_gemm(int, int, int, int, int);
 
 PYBIND11_MODULE(_core, m) {
     m.doc() = "crowdllama-amd MI355X (gfx950) HIP inference engine";
+
+    m.def("nccl_unique_id", [] {
+        ncclUniqueId id;
+        if (ncclGetUniqueId(&id) != ncclSuccess)
+            throw std::runtime_error("ncclGetUniqueId failed");
+        return py::bytes(reinterpret_cast<const char*>(&id), sizeof(id));
+    });
 
     m.def("device_count", [] {
         int n = 0;
@@ -46,7 +55,13 @@ PYBIND11_MODULE(_core, m) {
         .def_readwrite("page_size", &EngineConfig::page_size)
         .def_readwrite("gen_cap", &EngineConfig::gen_cap)
         .def_readwrite("device", &EngineConfig::device)
-        .def_readwrite("use_graph", &EngineConfig::use_graph);
+        .def_readwrite("use_graph", &EngineConfig::use_graph)
+        .def_readwrite("prefill_chunk", &EngineConfig::prefill_chunk)
+        .def_readwrite("tp_rank", &EngineConfig::tp_rank)
+        .def_readwrite("tp_size", &EngineConfig::tp_size)
+        .def_property("nccl_id",
+            [](EngineConfig& c) { return py::bytes(c.nccl_id); },
+            [](EngineConfig& c, py::bytes b) { c.nccl_id = std::string(b); });
 
     py::class_<Engine>(m, "Engine")
         .def(py::init<const std::string&, const EngineConfig&>(),
@@ -64,6 +79,11 @@ PYBIND11_MODULE(_core, m) {
         .def("decode", [](Engine& e, int n) {
                  py::gil_scoped_release rel;
                  e.decode(n);
+             })
+        .def("prefill_slot",
+             [](Engine& e, int slot, std::vector<int32_t> ids) {
+                 py::gil_scoped_release rel;
+                 e.prefill_slot(slot, ids);
              })
         .def("gen_tokens", &Engine::gen_tokens)
         .def("logits",
