@@ -32,6 +32,12 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=32)
     ap.add_argument("--model", default="llama3-8b")
     ap.add_argument("--scheme", default="q4_k_m")
+    ap.add_argument("--mode", default="dp", choices=["dp", "tp"],
+                    help="dp: one replica worker per GPU (default; the "
+                         "driver's scaling bench). tp: ONE worker spanning "
+                         "all ranks via RCCL tensor parallelism "
+                         "(BASELINE config 4; e.g. --mode tp --model "
+                         "llama3-70b --scheme bf16)")
     ap.add_argument("--batch", type=int, default=1,
                     help="decode slots per GPU")
     ap.add_argument("--prompt-len", type=int, default=128)
@@ -79,6 +85,18 @@ def main() -> None:
     ecfg.batch = args.batch
     ecfg.max_seq = args.max_seq
     ecfg.device = local_rank if core.device_count() > local_rank else 0
+    if args.mode == "tp":
+        # one worker spanning all ranks: RCCL over xGMI
+        ecfg.tp_rank = rank
+        ecfg.tp_size = world
+        if world > 1:
+            if rank == 0:
+                nid = core.nccl_unique_id()
+                obj = [nid]
+            else:
+                obj = [None]
+            dist.broadcast_object_list(obj, src=0)
+            ecfg.nccl_id = obj[0]
     t0 = time.time()
     eng = core.Engine(path, ecfg)
     log(f"rank {rank}: engine loaded in {time.time() - t0:.1f}s "
@@ -108,7 +126,10 @@ def main() -> None:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    tokens_total = n_gpus * args.batch * args.steps
+    if args.mode == "tp":
+        tokens_total = args.batch * args.steps  # one worker, whole job
+    else:
+        tokens_total = n_gpus * args.batch * args.steps
     value = tokens_total / elapsed
     ms_per_step = elapsed / args.steps * 1e3
 
@@ -132,7 +153,7 @@ def main() -> None:
                 "quant": args.scheme,
                 "global_batch": n_gpus * args.batch,
                 "seq_len": args.prompt_len,
-                "parallelism": f"dp{n_gpus}",
+                "parallelism": (f"tp{world}" if args.mode == "tp" else f"dp{n_gpus}"),
                 "engine_ms_per_step": eng.last_decode_ms() / args.steps,
             },
         }

@@ -74,7 +74,19 @@ def _random_blocks(t: GGMLType, n_elems: int, rng: np.random.Generator,
         d16 = np.float16(d)
         out[:, 0:2] = np.frombuffer(d16.tobytes(), dtype=np.uint8)
         return out.reshape(-1)
-    if t in (GGMLType.F16, GGMLType.BF16, GGMLType.F32):
+    if t == GGMLType.BF16:
+        # random mantissa+sign with a fixed exponent chosen so the value
+        # magnitude ~ d (fast: no float RNG for 100GB-class tensors)
+        e = int(np.clip(127 + np.floor(np.log2(max(d, 1e-30)) + 0.5), 1, 254))
+        r = np.frombuffer(rng.bytes(n_elems * 2), dtype=np.uint16)
+        bits = (r & 0x8000) | (np.uint16(e) << 7) | (r & 0x7F)
+        return bits.view(np.uint8)
+    if t == GGMLType.F16:
+        e16 = int(np.clip(15 + np.floor(np.log2(max(d, 1e-30)) + 0.5), 1, 30))
+        r = np.frombuffer(rng.bytes(n_elems * 2), dtype=np.uint16)
+        bits = (r & 0x8000) | (np.uint16(e16) << 10) | (r & 0x3FF)
+        return bits.view(np.uint8)
+    if t == GGMLType.F32:
         vals = rng.standard_normal(n_elems, dtype=np.float32) * d
         return quantize(vals, t).reshape(-1)
     raise ValueError(f"fast mode unsupported for {t}")
