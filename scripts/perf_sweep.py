@@ -1,5 +1,5 @@
 """GPU perf sweep: gemv A/B, gemm shapes, engine bench at several batches."""
-import sys; sys.path.insert(0, ".")
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from crowdllama_amd.ops import get_core
 core = get_core()
 DQ4K, DQ6K, DQ8, BF16 = 3, 4, 5, 2

@@ -1,4 +1,4 @@
-import sys; sys.path.insert(0, ".")
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from crowdllama_amd.ops import get_core
 core = get_core()
 # a few iterations of the two hot shapes only (keep the PMC file small)
