@@ -189,7 +189,7 @@ __device__ __forceinline__ void decode_chunk_raw(
 // PRE_RMS:  x = rmsnorm(xin) * gw   (gamma)
 // PRE_SILU: x[k] = silu(xin[b][k]) * xin[b][K+k]   (xin is [B][2K])
 // One wave per row; x staged in LDS as f32 with per-16 partial sums.
-template <DT W, int P>
+template <DT W, int P, int RPW>
 __global__ __launch_bounds__(256) void k_gemv(
     const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
     const float* __restrict__ xin, const float* __restrict__ gw,
@@ -202,10 +202,13 @@ __global__ __launch_bounds__(256) void k_gemv(
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6, lane = tid & 63;
-    // 2 rows per wave: amortize the staging latency over more weight
-    // traffic and keep 2x the loads in flight.
-    const int rbase = (int)blockIdx.x * 8 + wave * 2;
-    const int r[2] = {rbase, rbase + 1};
+    // RPW rows per wave: amortizes activation staging over more weight
+    // traffic and keeps RPW+ chunk loads in flight per wave (big-N GEMVs
+    // use RPW=8; small-N use RPW=1 to keep the grid large).
+    const int rbase = (int)blockIdx.x * (4 * RPW) + wave * RPW;
+    int r[RPW];
+    #pragma unroll
+    for (int i = 0; i < RPW; i++) r[i] = rbase + i;
     const int64_t qs_rb = dqs_row_bytes(W, K);
     const int64_t hdr_rb = dhdr_row_bytes(W, K);
     const int n_chunks = (int)(qs_rb / 16);
@@ -213,9 +216,9 @@ __global__ __launch_bounds__(256) void k_gemv(
 
     // issue the first weight chunks BEFORE staging: they have no
     // dependency on x, so their HBM latency hides under the staging phase.
-    ChunkRaw<W> cur[2], nxt[2];
+    ChunkRaw<W> cur[RPW], nxt[RPW];
     #pragma unroll
-    for (int i = 0; i < 2; i++)
+    for (int i = 0; i < RPW; i++)
         if (r[i] < N && lane < n_chunks)
             load_chunk<W>(qs + (int64_t)r[i] * qs_rb,
                           hdr + (int64_t)r[i] * hdr_rb, lane, &cur[i]);
@@ -269,16 +272,18 @@ __global__ __launch_bounds__(256) void k_gemv(
     }
     __syncthreads();
 
-    // ---- per-wave sweep over 2 rows ----
-    float acc[2][2] = {{0.f, 0.f}, {0.f, 0.f}};  // [row][b], B <= 2
+    // ---- per-wave sweep over RPW rows ----
+    float acc[RPW][2];  // [row][b], B <= 2
+    #pragma unroll
+    for (int i = 0; i < RPW; i++) { acc[i][0] = 0.f; acc[i][1] = 0.f; }
     for (int c = lane; c < n_chunks; c += 64) {
         #pragma unroll
-        for (int i = 0; i < 2; i++)
+        for (int i = 0; i < RPW; i++)
             if (r[i] < N && c + 64 < n_chunks)
                 load_chunk<W>(qs + (int64_t)r[i] * qs_rb,
                               hdr + (int64_t)r[i] * hdr_rb, c + 64, &nxt[i]);
         #pragma unroll
-        for (int i = 0; i < 2; i++) {
+        for (int i = 0; i < RPW; i++) {
             if (r[i] >= N) continue;
             float w[WPC];
             int k0;
@@ -312,7 +317,7 @@ __global__ __launch_bounds__(256) void k_gemv(
         }
     }
     #pragma unroll
-    for (int i = 0; i < 2; i++) {
+    for (int i = 0; i < RPW; i++) {
         if (r[i] >= N) continue;
         #pragma unroll 2
         for (int b = 0; b < B; b++) {
@@ -664,22 +669,38 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
     const int N = (int)w.n, K = (int)w.k;
     if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
     const size_t lds = (size_t)B * (K + (K >> 4)) * 4 + 8 * 4;
-    dim3 grid((N + 7) / 8), block(256);
-    #define GEMV_CASE(WT)                                                        \
+    // rows per wave: keep the grid large for small N; amortize staging
+    // and deepen the load pipeline for huge-N (head) kernels
+    int rpw = 1;
+    if (N >= 32768) rpw = 8;
+    else if (N >= 16384) rpw = 4;
+    else if (N >= 8192) rpw = 2;
+    dim3 grid((N + 4 * rpw - 1) / (4 * rpw)), block(256);
+    #define GEMV_RPW(WT, RPWV)                                                   \
         do {                                                                     \
-            auto kern = (pre == PRE_RMS) ? k_gemv<WT, PRE_RMS>                   \
-                       : (pre == PRE_SILU) ? k_gemv<WT, PRE_SILU>                \
-                       : k_gemv<WT, PRE_NONE>;                                   \
+            auto kern = (pre == PRE_RMS) ? k_gemv<WT, PRE_RMS, RPWV>             \
+                       : (pre == PRE_SILU) ? k_gemv<WT, PRE_SILU, RPWV>          \
+                       : k_gemv<WT, PRE_NONE, RPWV>;                             \
             if (lds > 64 * 1024) {                                               \
                 (void)hipFuncSetAttribute((const void*)kern,                     \
                     hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);       \
             }                                                                    \
             hipLaunchKernelGGL(kern, grid, block, lds, stream,                   \
                 (const uint8_t*)w.qs, (const uint8_t*)w.hdr, xin, gw, res, y,    \
-                N, K, B, ldy, eps);                                                   \
+                N, K, B, ldy, eps);                                              \
+        } while (0)
+    #define GEMV_CASE(WT)                                                        \
+        do {                                                                     \
+            switch (rpw) {                                                       \
+                case 8: GEMV_RPW(WT, 8); break;                                  \
+                case 4: GEMV_RPW(WT, 4); break;                                  \
+                case 2: GEMV_RPW(WT, 2); break;                                  \
+                default: GEMV_RPW(WT, 1); break;                                 \
+            }                                                                    \
         } while (0)
     DISPATCH_DT(w.dtype, GEMV_CASE);
     #undef GEMV_CASE
+    #undef GEMV_RPW
 }
 
 void launch_embed(const WTensor& w, const int32_t* ids, float* x, int B,
