@@ -669,12 +669,10 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
     const int N = (int)w.n, K = (int)w.k;
     if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
     const size_t lds = (size_t)B * (K + (K >> 4)) * 4 + 8 * 4;
-    // rows per wave: keep the grid large for small N; amortize staging
-    // and deepen the load pipeline for huge-N (head) kernels
-    int rpw = 1;
-    if (N >= 32768) rpw = 8;
-    else if (N >= 16384) rpw = 4;
-    else if (N >= 8192) rpw = 2;
+    // rows per wave: 1 measured best across all shapes (RPW>1 raises
+    // register pressure past the occupancy cliff: head 187us@1 vs 481us@8);
+    // the template stays parameterized for future tuning.
+    const int rpw = 1;
     dim3 grid((N + 4 * rpw - 1) / (4 * rpw)), block(256);
     #define GEMV_RPW(WT, RPWV)                                                   \
         do {                                                                     \
