@@ -224,6 +224,9 @@ __global__ __launch_bounds__(256) void k_gemv(
                           hdr + (int64_t)r[i] * hdr_rb, lane, &cur[i]);
 
     // ---- stage activations (vectorized float4; K always %4==0) ----
+    // Two-phase batches: issue up to 4 independent global loads, then the
+    // LDS writes — hipcc otherwise emits load->vmcnt(0)->ds_write per
+    // iteration, serializing the staging into K/1024 memory round trips.
     const int K4 = K >> 2;
     for (int b = 0; b < B; b++) {
         float ss = 0.f;
@@ -233,23 +236,47 @@ __global__ __launch_bounds__(256) void k_gemv(
                 xin + (size_t)b * 2 * K);
             const float4* u4 = reinterpret_cast<const float4*>(
                 xin + (size_t)b * 2 * K + K);
-            for (int k = tid; k < K4; k += 256) {
-                const float4 g = g4[k], u = u4[k];
-                float4 o;
-                o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
-                o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
-                o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
-                o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
-                *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = o;
+            for (int k0 = tid; k0 < K4; k0 += 1024) {
+                float4 gs[4], us[4];
+                #pragma unroll
+                for (int j = 0; j < 4; j++) {
+                    const int k = k0 + j * 256;
+                    if (k < K4) { gs[j] = g4[k]; us[j] = u4[k]; }
+                }
+                #pragma unroll
+                for (int j = 0; j < 4; j++) {
+                    const int k = k0 + j * 256;
+                    if (k < K4) {
+                        const float4 g = gs[j], u = us[j];
+                        float4 o;
+                        o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+                        o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+                        o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+                        o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
+                        *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = o;
+                    }
+                }
             }
         } else {
             const float4* x4 = reinterpret_cast<const float4*>(
                 xin + (size_t)b * K);
-            for (int k = tid; k < K4; k += 256) {
-                const float4 v = x4[k];
-                *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = v;
-                if constexpr (P == PRE_RMS)
-                    ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+            for (int k0 = tid; k0 < K4; k0 += 1024) {
+                float4 vs[4];
+                #pragma unroll
+                for (int j = 0; j < 4; j++) {
+                    const int k = k0 + j * 256;
+                    if (k < K4) vs[j] = x4[k];
+                }
+                #pragma unroll
+                for (int j = 0; j < 4; j++) {
+                    const int k = k0 + j * 256;
+                    if (k < K4) {
+                        const float4 v = vs[j];
+                        *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = v;
+                        if constexpr (P == PRE_RMS)
+                            ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+                    }
+                }
             }
         }
         if constexpr (P == PRE_RMS) {
