@@ -216,12 +216,19 @@ __global__ __launch_bounds__(256) void k_gemv(
 
     // issue the first weight chunks BEFORE staging: they have no
     // dependency on x, so their HBM latency hides under the staging phase.
+    // Loads are unconditional with clamped indices — a branch around a
+    // load makes hipcc drain vmcnt(0) at reconvergence (guide §5 traps),
+    // killing the software pipeline.
     ChunkRaw<W> cur[RPW], nxt[RPW];
+    int rc[RPW];   // clamped row for addressing; r[] keeps validity
     #pragma unroll
     for (int i = 0; i < RPW; i++)
-        if (r[i] < N && lane < n_chunks)
-            load_chunk<W>(qs + (int64_t)r[i] * qs_rb,
-                          hdr + (int64_t)r[i] * hdr_rb, lane, &cur[i]);
+        rc[i] = r[i] < N ? r[i] : N - 1;
+    const int c0 = lane < n_chunks ? lane : 0;
+    #pragma unroll
+    for (int i = 0; i < RPW; i++)
+        load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
+                      hdr + (int64_t)rc[i] * hdr_rb, c0, &cur[i]);
 
     // ---- stage activations (vectorized float4; K always %4==0) ----
     // Two-phase batches: issue up to 4 independent global loads, then the
@@ -236,47 +243,55 @@ __global__ __launch_bounds__(256) void k_gemv(
                 xin + (size_t)b * 2 * K);
             const float4* u4 = reinterpret_cast<const float4*>(
                 xin + (size_t)b * 2 * K + K);
-            for (int k0 = tid; k0 < K4; k0 += 1024) {
-                float4 gs[4], us[4];
+            int k0 = tid;
+            for (; k0 + 768 < K4; k0 += 1024) {  // guard hoisted: no
+                float4 gs[4], us[4];             // per-element branches
                 #pragma unroll
                 for (int j = 0; j < 4; j++) {
-                    const int k = k0 + j * 256;
-                    if (k < K4) { gs[j] = g4[k]; us[j] = u4[k]; }
+                    gs[j] = g4[k0 + j * 256];
+                    us[j] = u4[k0 + j * 256];
                 }
                 #pragma unroll
                 for (int j = 0; j < 4; j++) {
-                    const int k = k0 + j * 256;
-                    if (k < K4) {
-                        const float4 g = gs[j], u = us[j];
-                        float4 o;
-                        o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
-                        o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
-                        o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
-                        o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
-                        *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = o;
-                    }
+                    const float4 g = gs[j], u = us[j];
+                    float4 o;
+                    o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+                    o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+                    o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+                    o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
+                    *reinterpret_cast<float4*>(xlb + xpad(k0 * 4 + j * 1024)) = o;
                 }
+            }
+            for (int k = k0; k < K4; k += 256) {
+                const float4 g = g4[k], u = u4[k];
+                float4 o;
+                o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+                o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+                o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+                o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
+                *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = o;
             }
         } else {
             const float4* x4 = reinterpret_cast<const float4*>(
                 xin + (size_t)b * K);
-            for (int k0 = tid; k0 < K4; k0 += 1024) {
+            int k0 = tid;
+            for (; k0 + 768 < K4; k0 += 1024) {  // guard hoisted
                 float4 vs[4];
                 #pragma unroll
-                for (int j = 0; j < 4; j++) {
-                    const int k = k0 + j * 256;
-                    if (k < K4) vs[j] = x4[k];
-                }
+                for (int j = 0; j < 4; j++) vs[j] = x4[k0 + j * 256];
                 #pragma unroll
                 for (int j = 0; j < 4; j++) {
-                    const int k = k0 + j * 256;
-                    if (k < K4) {
-                        const float4 v = vs[j];
-                        *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = v;
-                        if constexpr (P == PRE_RMS)
-                            ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
-                    }
+                    const float4 v = vs[j];
+                    *reinterpret_cast<float4*>(xlb + xpad(k0 * 4 + j * 1024)) = v;
+                    if constexpr (P == PRE_RMS)
+                        ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
                 }
+            }
+            for (int k = k0; k < K4; k += 256) {
+                const float4 v = x4[k];
+                *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = v;
+                if constexpr (P == PRE_RMS)
+                    ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
             }
         }
         if constexpr (P == PRE_RMS) {
@@ -287,7 +302,22 @@ __global__ __launch_bounds__(256) void k_gemv(
             const float inv = rsqrtf(
                 (red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
             const float4* gw4 = reinterpret_cast<const float4*>(gw);
-            for (int k = tid; k < K4; k += 256) {
+            int kg = tid;
+            for (; kg + 768 < K4; kg += 1024) {  // batched gw loads
+                float4 gs[4];
+                #pragma unroll
+                for (int j = 0; j < 4; j++) gs[j] = gw4[kg + j * 256];
+                #pragma unroll
+                for (int j = 0; j < 4; j++) {
+                    float4* vp = reinterpret_cast<float4*>(
+                        xlb + xpad(kg * 4 + j * 1024));
+                    float4 v = *vp;
+                    v.x *= inv * gs[j].x; v.y *= inv * gs[j].y;
+                    v.z *= inv * gs[j].z; v.w *= inv * gs[j].w;
+                    *vp = v;
+                }
+            }
+            for (int k = kg; k < K4; k += 256) {
                 const float4 g = gw4[k];
                 float4* vp = reinterpret_cast<float4*>(xlb + xpad(k * 4));
                 float4 v = *vp;
@@ -304,14 +334,13 @@ __global__ __launch_bounds__(256) void k_gemv(
     #pragma unroll
     for (int i = 0; i < RPW; i++) { acc[i][0] = 0.f; acc[i][1] = 0.f; }
     for (int c = lane; c < n_chunks; c += 64) {
+        const int cn = (c + 64 < n_chunks) ? c + 64 : c;  // clamped prefetch
         #pragma unroll
         for (int i = 0; i < RPW; i++)
-            if (r[i] < N && c + 64 < n_chunks)
-                load_chunk<W>(qs + (int64_t)r[i] * qs_rb,
-                              hdr + (int64_t)r[i] * hdr_rb, c + 64, &nxt[i]);
+            load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
+                          hdr + (int64_t)rc[i] * hdr_rb, cn, &nxt[i]);
         #pragma unroll
         for (int i = 0; i < RPW; i++) {
-            if (r[i] >= N) continue;
             float w[WPC];
             int k0;
             decode_chunk_raw<W>(cur[i], c, w, &k0);
