@@ -187,6 +187,12 @@ def main(argv: list[str] | None = None) -> int:
 
     sub.add_parser("version")
 
+    kg = sub.add_parser("keygen", help="create an identity key "
+                        "(reference utils/dhtcertgen parity)")
+    kg.add_argument("--component", default="dht",
+                    choices=["dht", "worker", "consumer"])
+    kg.add_argument("--out", default=None)
+
     ns = sub.add_parser("network-status")
     _add_common(ns)
 
@@ -202,6 +208,12 @@ def main(argv: list[str] | None = None) -> int:
         return 0
     if args.cmd == "network-status":
         asyncio.run(_run_network_status(args))
+        return 0
+    if args.cmd == "keygen":
+        from .keys import default_key_path, get_or_create_key, peer_id_from_key
+        path = args.out or default_key_path(args.component)
+        seed = get_or_create_key(path)
+        print(f"{path}: peer id {peer_id_from_key(seed)}")
         return 0
     return 1
 

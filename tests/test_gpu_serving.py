@@ -1,0 +1,77 @@
+"""End-to-end serving on GPU: DHT + worker with the real HIP engine +
+gateway, request through /api/chat (the full reference round trip of
+SURVEY.md §3.2 with first-party compute)."""
+
+import asyncio
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def has_gpu():
+    from crowdllama_amd.ops import get_core
+    c = get_core()
+    if c.device_count() == 0:
+        pytest.skip("no GPU")
+    return True
+
+
+def test_chat_through_mesh_hip_engine(has_gpu, tmp_path):
+    from crowdllama_amd.config import Config
+    from crowdllama_amd.engine.hip_engine import HipEngine
+    from crowdllama_amd.mesh.dhtnode import DHTServer
+    from crowdllama_amd.mesh.gateway import Gateway
+    from crowdllama_amd.mesh.peer import Peer
+    from crowdllama_amd.models import synth_path
+
+    path = synth_path("testllama", scheme="q4_k_m", mode="exact", seed=7)
+    engine = HipEngine("testllama", path, max_seq=256)
+
+    async def go():
+        import aiohttp
+        import time
+        def mk(c):
+            return Config(test_mode=True, listen_host="127.0.0.1",
+                          key_path=str(tmp_path / f"{c}.key"))
+        dht = DHTServer(mk("dht"), "CLADHT")
+        dht_port = await dht.start("127.0.0.1", 0)
+        boot = [f"127.0.0.1:{dht_port}"]
+        wcfg = mk("worker")
+        wcfg.bootstrap_peers = boot
+        worker = Peer(wcfg, worker_mode=True, engines={"testllama": engine})
+        await worker.start()
+        ccfg = mk("consumer")
+        ccfg.bootstrap_peers = boot
+        consumer = Peer(ccfg, worker_mode=False)
+        await consumer.start()
+        gw = Gateway(consumer, ccfg)
+        gw_port = await gw.start(port=0)
+        try:
+            deadline = time.time() + 20
+            while gw.find_best_worker("testllama") is None:
+                assert time.time() < deadline, "worker never discovered"
+                await asyncio.sleep(0.1)
+            # worker metadata reflects real device props
+            res = gw.find_best_worker("testllama")
+            assert "gfx" in res.gpu_model or "MI" in res.gpu_model.upper(), \
+                res.gpu_model
+            assert res.vram_gb > 100  # MI355X: 288 GB
+            async with aiohttp.ClientSession() as s:
+                async with s.post(f"http://127.0.0.1:{gw_port}/api/chat",
+                                  json={"model": "testllama",
+                                        "messages": [{"role": "user",
+                                                      "content": "abc"}]}) as r:
+                    body = await r.json()
+                    assert r.status == 200, body
+            assert body["done"] is True
+            assert body["worker_id"] == worker.peer_id
+            assert isinstance(body["message"]["content"], str)
+            assert body["total_duration"] > 0
+        finally:
+            await gw.stop()
+            await consumer.stop()
+            await worker.stop()
+            await dht.stop()
+    asyncio.run(go())
