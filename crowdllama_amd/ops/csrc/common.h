@@ -40,7 +40,9 @@ enum class DT : int32_t {
     F32 = 0,
     F16 = 1,
     BF16 = 2,
-    DQ4K = 3,   // qs: [nsb][128B] nibbles; hdr: [nsb][16B] {f16 d, f16 dmin, u8 sc[12]}
+    DQ4K = 3,   // qs: [nsb][128B] nibbles; hdr: [nsb][4 pairs][8B]
+                //   {f16 d, f16 dmin, u8 sc_lo, u8 mn_lo, u8 sc_hi, u8 mn_hi}
+                //   (6-bit scales pre-decoded at upload: branchless kernel)
     DQ6K = 4,   // qs: [K] int8 (q-32 applied); hdr: [nsb][32B] {f16 d, i8 sc[16], pad}
     DQ8 = 5,    // qs: [K] int8; hdr: [K/32] f16 d
 };
@@ -61,7 +63,7 @@ __host__ __device__ inline int64_t dqs_row_bytes(DT t, int64_t k) {
 __host__ __device__ inline int64_t dhdr_row_bytes(DT t, int64_t k) {
     switch (t) {
         case DT::F32: case DT::F16: case DT::BF16: return 0;
-        case DT::DQ4K: return k / QK_K * 16;
+        case DT::DQ4K: return k / QK_K * 32;
         case DT::DQ6K: return k / QK_K * 32;
         case DT::DQ8: return k / 32 * 2;
     }

@@ -94,16 +94,35 @@ void repack(int32_t ggml_type, const uint8_t* src, int64_t rows, int64_t k,
         case 0: case 1: case 30:  // float passthrough
             std::memcpy(qs_out, src, src_rb * rows);
             return;
-        case 12: {  // Q4_K: disk block 144B = {hdr 16B, qs 128B}
+        case 12: {  // Q4_K: disk 144B = {f16 d, f16 dmin, u8 sc6[12], qs 128}
+            // device hdr: per sub-block PAIR (2 chunks): {d, dmin, sc_lo,
+            // mn_lo, sc_hi, mn_hi} — 6-bit scales pre-decoded so the kernel
+            // does one aligned 8B load and no divergent unpacking.
             const int64_t nsb = k / 256;
             parallel_for(rows, [&](int64_t lo, int64_t hi) {
                 for (int64_t r = lo; r < hi; r++) {
                     const uint8_t* s = src + r * src_rb;
                     uint8_t* q = qs_out + r * nsb * 128;
-                    uint8_t* h = hdr_out + r * nsb * 16;
+                    uint8_t* h = hdr_out + r * nsb * 32;
                     for (int64_t b = 0; b < nsb; b++) {
-                        std::memcpy(h + b * 16, s + b * 144, 16);
-                        std::memcpy(q + b * 128, s + b * 144 + 16, 128);
+                        const uint8_t* blk = s + b * 144;
+                        const uint8_t* sc6 = blk + 4;
+                        uint8_t sc[8], mn[8];
+                        for (int j = 0; j < 4; j++) {
+                            sc[j] = sc6[j] & 63;
+                            mn[j] = sc6[j + 4] & 63;
+                            sc[j + 4] = (sc6[j + 8] & 0xF) | ((sc6[j] >> 6) << 4);
+                            mn[j + 4] = (sc6[j + 8] >> 4) | ((sc6[j + 4] >> 6) << 4);
+                        }
+                        for (int pr = 0; pr < 4; pr++) {
+                            uint8_t* e = h + b * 32 + pr * 8;
+                            std::memcpy(e, blk, 4);  // d, dmin (f16 each)
+                            e[4] = sc[2 * pr];
+                            e[5] = mn[2 * pr];
+                            e[6] = sc[2 * pr + 1];
+                            e[7] = mn[2 * pr + 1];
+                        }
+                        std::memcpy(q + b * 128, blk + 16, 128);
                     }
                 }
             });

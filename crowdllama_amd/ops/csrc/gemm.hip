@@ -45,32 +45,19 @@ __device__ __forceinline__ void stage_w_slice(
         const int q = (kb & 255) >> 6;
         const int p = 2 * q + h;
         const uint8_t* qrow = qs + row * (K / 256) * 128;
-        const uint4* hrow = reinterpret_cast<const uint4*>(
-            hdr + row * (K / 256) * 16);
+        const uint2* hrow = reinterpret_cast<const uint2*>(
+            hdr + row * (K / 256) * 32);
         typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
         const u32x4 qv = __builtin_nontemporal_load(
             reinterpret_cast<const u32x4*>(qrow) + sb * 8 + p);
         const uint32_t dw[4] = {qv.x, qv.y, qv.z, qv.w};
-        const uint4 hd = hrow[sb];
+        const uint2 hd = hrow[sb * 4 + p / 2];   // pre-decoded pair header
         const float d = f16b_to_f32(hd.x & 0xFFFF);
         const float dmin = f16b_to_f32(hd.x >> 16);
-        const uint32_t sbytes[3] = {hd.y, hd.z, hd.w};
-        auto sbyte = [&](int i) -> uint32_t {
-            return (sbytes[i >> 2] >> ((i & 3) * 8)) & 0xFF;
-        };
-        const int j0 = 2 * q, j1 = 2 * q + 1;
-        uint32_t sc0, mn0, sc1, mn1;
-        if (q < 2) {
-            sc0 = sbyte(j0) & 63; mn0 = sbyte(j0 + 4) & 63;
-            sc1 = sbyte(j1) & 63; mn1 = sbyte(j1 + 4) & 63;
-        } else {
-            sc0 = (sbyte(j0 + 4) & 0xF) | ((sbyte(j0 - 4) >> 6) << 4);
-            mn0 = (sbyte(j0 + 4) >> 4) | ((sbyte(j0) >> 6) << 4);
-            sc1 = (sbyte(j1 + 4) & 0xF) | ((sbyte(j1 - 4) >> 6) << 4);
-            mn1 = (sbyte(j1 + 4) >> 4) | ((sbyte(j1) >> 6) << 4);
-        }
-        const float dl = d * (float)sc0, ml = dmin * (float)mn0;
-        const float dh2 = d * (float)sc1, mh = dmin * (float)mn1;
+        const float dl = d * (float)(hd.y & 0xFF);
+        const float ml = dmin * (float)((hd.y >> 8) & 0xFF);
+        const float dh2 = d * (float)((hd.y >> 16) & 0xFF);
+        const float mh = dmin * (float)(hd.y >> 24);
         // lo nibbles -> klocal h*16+t (sub-block 2q), hi -> 32+h*16+t
         #pragma unroll
         for (int j = 0; j < 4; j++) {

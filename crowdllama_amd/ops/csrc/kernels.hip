@@ -74,7 +74,7 @@ typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
 template <DT W>
 struct ChunkRaw {
     u32x4 qv;
-    uint4 hd;       // DQ4K header; DQ6K/DQ8 use scalar fields
+    uint2 hd;       // DQ4K pair header {d, dmin, sc/mn x2}
     float d0, d1;   // DQ6K: two eff scales; DQ8: block scale
 };
 
@@ -86,7 +86,7 @@ __device__ __forceinline__ void load_chunk(
     r->qv = __builtin_nontemporal_load(
         reinterpret_cast<const u32x4*>(qs_row) + c);
     if constexpr (W == DT::DQ4K) {
-        r->hd = reinterpret_cast<const uint4*>(hdr_row)[c >> 3];
+        r->hd = reinterpret_cast<const uint2*>(hdr_row)[c >> 1];
     } else if constexpr (W == DT::DQ6K) {
         const int sb = c >> 4, s16 = c & 15;
         const uint8_t* hb = hdr_row + sb * 32;
@@ -108,26 +108,12 @@ __device__ __forceinline__ void decode_chunk_raw(
     const uint32_t dw[4] = {qv.x, qv.y, qv.z, qv.w};
     if constexpr (W == DT::DQ4K) {
         const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
-        const uint4 hd = r.hd;
-        const float d = f16_bits_to_f32(hd.x & 0xFFFF);
-        const float dmin = f16_bits_to_f32(hd.x >> 16);
-        const uint32_t sbytes[3] = {hd.y, hd.z, hd.w};
-        auto sbyte = [&](int i) -> uint32_t {
-            return (sbytes[i >> 2] >> ((i & 3) * 8)) & 0xFF;
-        };
-        const int j0 = 2 * q, j1 = 2 * q + 1;
-        uint32_t sc0, mn0, sc1, mn1;
-        if (q < 2) {
-            sc0 = sbyte(j0) & 63; mn0 = sbyte(j0 + 4) & 63;
-            sc1 = sbyte(j1) & 63; mn1 = sbyte(j1 + 4) & 63;
-        } else {
-            sc0 = (sbyte(j0 + 4) & 0xF) | ((sbyte(j0 - 4) >> 6) << 4);
-            mn0 = (sbyte(j0 + 4) >> 4) | ((sbyte(j0) >> 6) << 4);
-            sc1 = (sbyte(j1 + 4) & 0xF) | ((sbyte(j1 - 4) >> 6) << 4);
-            mn1 = (sbyte(j1 + 4) >> 4) | ((sbyte(j1) >> 6) << 4);
-        }
-        const float dl = d * (float)sc0, ml = dmin * (float)mn0;
-        const float dh = d * (float)sc1, mh = dmin * (float)mn1;
+        const float d = f16_bits_to_f32(r.hd.x & 0xFFFF);
+        const float dmin = f16_bits_to_f32(r.hd.x >> 16);
+        const float dl = d * (float)(r.hd.y & 0xFF);
+        const float ml = dmin * (float)((r.hd.y >> 8) & 0xFF);
+        const float dh = d * (float)((r.hd.y >> 16) & 0xFF);
+        const float mh = dmin * (float)(r.hd.y >> 24);
         *k0 = sb * 256 + q * 64 + h * 16;
         #pragma unroll
         for (int j = 0; j < 4; j++) {

@@ -26,11 +26,24 @@ def core():
 
 
 def _repack_q4k(raw, rows, k):
+    from crowdllama_amd.quant.kquants import _unpack_q4k_scales
     nsb = k // 256
     blk = raw.reshape(rows, nsb, 144)
-    hdr = np.ascontiguousarray(blk[:, :, :16]).reshape(rows, -1)
     qs = np.ascontiguousarray(blk[:, :, 16:]).reshape(rows, -1)
-    return qs, hdr
+    # device header: per sub-block pair {f16 d, f16 dmin, sc_lo, mn_lo,
+    # sc_hi, mn_hi} (scales pre-decoded)
+    s6 = blk[:, :, 4:16].reshape(-1, 12)
+    sc, mn = _unpack_q4k_scales(s6)
+    sc = sc.reshape(rows, nsb, 8)
+    mn = mn.reshape(rows, nsb, 8)
+    hdr = np.zeros((rows, nsb, 4, 8), dtype=np.uint8)
+    for pr in range(4):
+        hdr[:, :, pr, 0:4] = blk[:, :, 0:4]
+        hdr[:, :, pr, 4] = sc[:, :, 2 * pr]
+        hdr[:, :, pr, 5] = mn[:, :, 2 * pr]
+        hdr[:, :, pr, 6] = sc[:, :, 2 * pr + 1]
+        hdr[:, :, pr, 7] = mn[:, :, 2 * pr + 1]
+    return qs, hdr.reshape(rows, -1)
 
 
 def _repack_q6k(raw, rows, k):
