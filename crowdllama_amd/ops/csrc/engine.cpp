@@ -29,7 +29,8 @@ void launch_embed(const WTensor&, const int32_t* ids, float* x, int B,
 void launch_attn_decode(const float* qkv, const float* inv_freq,
                         const int32_t* page_table,
                         uint16_t* kv_pool, const int32_t* n_past,
-                        float* part_o, float* part_ml, int B, int NH, int NKV,
+                        float* part_o, float* part_ml, int* tickets,
+                        float* attn_out, int B, int NH, int NKV,
                         int D, int S, int page_size, int max_pages,
                         int64_t page_stride, float scale, hipStream_t);
 void launch_attn_combine(const float* part_o, const float* part_ml,
@@ -474,6 +475,8 @@ void Engine::alloc_state() {
     logits_ = (float*)dalloc((size_t)B * V * 4);
     part_o_ = (float*)dalloc((size_t)B * NH * attn_splits_ * D * 4);
     part_ml_ = (float*)dalloc((size_t)B * NH * attn_splits_ * 2 * 4);
+    attn_tickets_ = (int32_t*)dalloc((size_t)B * NKV * 4);  // zeroed once;
+    // the last arriver resets its counter each launch
     amax_val_ = (float*)dalloc((size_t)B * 64 * 4);
     amax_idx_ = (int32_t*)dalloc((size_t)B * 64 * 4);
     cur_ids_ = (int32_t*)dalloc((size_t)B * 4);
@@ -547,10 +550,9 @@ void Engine::step(hipStream_t s) {
             gemm_proj(L.qkv, xn_, nullptr, qkv_, B, s);
         }
         launch_attn_decode(qkv_, inv_freq_, page_table_, kv_layer, n_past_,
-                           part_o_, part_ml_, B, NH, NKV, D, attn_splits_,
-                           cfg_.page_size, max_pages_, page_stride_, scale, s);
-        launch_attn_combine(part_o_, part_ml_, attn_out_, B, NH, attn_splits_,
-                            D, s);
+                           part_o_, part_ml_, attn_tickets_, attn_out_, B, NH,
+                           NKV, D, attn_splits_, cfg_.page_size, max_pages_,
+                           page_stride_, scale, s);
         if (gemv_path && !tp) {
             for (auto& pt : L.o.parts)
                 launch_gemv(pt.w, PRE_NONE, attn_out_, nullptr, x_ + pt.row_off,

@@ -133,6 +133,7 @@ private:
     float* logits_ = nullptr;   // [B][V]
     float* part_o_ = nullptr;   // [B][NH][S][D]
     float* part_ml_ = nullptr;  // [B][NH][S][2]
+    int32_t* attn_tickets_ = nullptr;  // [B][NKV] fan-in counters
     float* amax_val_ = nullptr; // [B][64]
     int32_t* amax_idx_ = nullptr;
     float* inv_freq_ = nullptr; // [D/2]

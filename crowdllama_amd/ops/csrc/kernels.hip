@@ -315,16 +315,74 @@ __global__ __launch_bounds__(256) void k_gemv(
     }
     __syncthreads();
 
-    // ---- per-wave sweep over RPW rows ----
+    // ---- per-wave sweep over RPW rows, 2 chunks in flight per row ----
+    // (pair-unrolled: chunks c and c+64 decoded per iteration with the
+    // next pair prefetched — 2x the outstanding loads of a 1-deep ring,
+    // which is what keeps a 2-chunk/lane row, e.g. DQ4K K=4096, streaming)
     float acc[RPW][2];  // [row][b], B <= 2
     #pragma unroll
     for (int i = 0; i < RPW; i++) { acc[i][0] = 0.f; acc[i][1] = 0.f; }
-    for (int c = lane; c < n_chunks; c += 64) {
-        const int cn = (c + 64 < n_chunks) ? c + 64 : c;  // clamped prefetch
+    ChunkRaw<W> cur2[RPW], nxt2[RPW];
+    {
+        const int c1 = (lane + 64 < n_chunks) ? lane + 64 : c0;
         #pragma unroll
         for (int i = 0; i < RPW; i++)
             load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
+                          hdr + (int64_t)rc[i] * hdr_rb, c1, &cur2[i]);
+    }
+    int c = lane;
+    for (; c + 64 < n_chunks; c += 128) {
+        const int cn = (c + 128 < n_chunks) ? c + 128 : c;
+        const int cn2 = (c + 192 < n_chunks) ? c + 192 : c;
+        #pragma unroll
+        for (int i = 0; i < RPW; i++) {
+            load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
                           hdr + (int64_t)rc[i] * hdr_rb, cn, &nxt[i]);
+            load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
+                          hdr + (int64_t)rc[i] * hdr_rb, cn2, &nxt2[i]);
+        }
+        #pragma unroll
+        for (int i = 0; i < RPW; i++) {
+            #pragma unroll
+            for (int half = 0; half < 2; half++) {
+                float w[WPC];
+                int k0;
+                decode_chunk_raw<W>(half ? cur2[i] : cur[i],
+                                    half ? c + 64 : c, w, &k0);
+                #pragma unroll 2
+                for (int b = 0; b < B; b++) {
+                    const float4* xb4 = reinterpret_cast<const float4*>(
+                        x_lds + (size_t)b * KP + xpad(k0));
+                    float sdot = 0.f;
+                    if constexpr (W == DT::DQ4K) {
+                        #pragma unroll
+                        for (int t4 = 0; t4 < 4; t4++) {
+                            const float4 xl = xb4[t4];
+                            const float4 xh = xb4[8 + t4];
+                            sdot += w[t4 * 4 + 0] * xl.x + w[t4 * 4 + 1] * xl.y
+                               + w[t4 * 4 + 2] * xl.z + w[t4 * 4 + 3] * xl.w;
+                            sdot += w[16 + t4 * 4 + 0] * xh.x
+                               + w[16 + t4 * 4 + 1] * xh.y
+                               + w[16 + t4 * 4 + 2] * xh.z
+                               + w[16 + t4 * 4 + 3] * xh.w;
+                        }
+                    } else {
+                        #pragma unroll
+                        for (int t4 = 0; t4 < WPC / 4; t4++) {
+                            const float4 xv = xb4[t4];
+                            sdot += w[t4 * 4 + 0] * xv.x + w[t4 * 4 + 1] * xv.y
+                               + w[t4 * 4 + 2] * xv.z + w[t4 * 4 + 3] * xv.w;
+                        }
+                    }
+                    acc[i][b] += sdot;
+                }
+            }
+            cur[i] = nxt[i];
+            cur2[i] = nxt2[i];
+        }
+    }
+    // tail: at most one chunk left per lane
+    if (c < n_chunks) {
         #pragma unroll
         for (int i = 0; i < RPW; i++) {
             float w[WPC];
@@ -355,7 +413,6 @@ __global__ __launch_bounds__(256) void k_gemv(
                 }
                 acc[i][b] += s;
             }
-            cur[i] = nxt[i];
         }
     }
     #pragma unroll
@@ -420,6 +477,8 @@ __global__ __launch_bounds__(256) void k_attn_decode(
     uint16_t* __restrict__ kv_pool, const int32_t* __restrict__ n_past,
     float* __restrict__ part_o,   // [B][NH][S][D]
     float* __restrict__ part_ml,  // [B][NH][S][2]
+    int* __restrict__ tickets,    // [B][NKV] fan-in counters (0 on entry)
+    float* __restrict__ attn_out, // [B][NH*D]
     int NH, int NKV, int S, int page_size, int max_pages,
     int64_t page_stride, float scale) {
     constexpr int DPL = D / 16;   // dims per lane (bf16: 2*DPL bytes)
@@ -609,6 +668,47 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             part_ml[(((size_t)b * NH + head) * S + s) * 2 + 1] = lsum;
         }
     }
+
+    // ---- fan-in: the LAST split workgroup of (b, kvh) combines ----
+    // Guide §6 G16 recipe: every wave drains its stores, __syncthreads,
+    // one lane agent-scope release + asm vmcnt(0) (restates the wait the
+    // compiler may drop, G16 pitfall 12), relaxed agent ticket add; the
+    // last arriver acquires, then reads every split's partials with plain
+    // loads. Placement-independent (no XCD/dispatch assumptions).
+    __shared__ int is_last;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every storing wave
+    __syncthreads();
+    if (tid == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        const int t = __hip_atomic_fetch_add(
+            &tickets[(size_t)b * NKV + kvh], 1, __ATOMIC_RELAXED,
+            __HIP_MEMORY_SCOPE_AGENT);
+        is_last = (t == S - 1);
+        if (is_last) {
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+            // reset for the next launch (we are the only reader)
+            __hip_atomic_store(&tickets[(size_t)b * NKV + kvh], 0,
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        }
+    }
+    __syncthreads();
+    if (!is_last) return;
+    for (int idx = tid; idx < G * D; idx += 256) {
+        const int g = idx / D, d = idx % D;
+        const int head = kvh * G + g;
+        const float* ml = part_ml + ((size_t)b * NH + head) * S * 2;
+        float mg = -1e30f;
+        for (int t = 0; t < S; t++) mg = fmaxf(mg, ml[2 * t]);
+        float denom = 0.f, osum = 0.f;
+        #pragma unroll 4
+        for (int t = 0; t < S; t++) {
+            const float e = __expf(ml[2 * t] - mg);
+            denom += e * ml[2 * t + 1];
+            osum += e * part_o[(((size_t)b * NH + head) * S + t) * D + d];
+        }
+        attn_out[(size_t)b * NH * D + (size_t)head * D + d] = osum / denom;
+    }
 }
 
 // grid (B*NH); block 128. Combine S split-KV partials into attn_out.
@@ -755,7 +855,8 @@ void launch_embed(const WTensor& w, const int32_t* ids, float* x, int B,
 void launch_attn_decode(const float* qkv, const float* inv_freq,
                         const int32_t* page_table,
                         uint16_t* kv_pool, const int32_t* n_past,
-                        float* part_o, float* part_ml, int B, int NH, int NKV,
+                        float* part_o, float* part_ml, int* tickets,
+                        float* attn_out, int B, int NH, int NKV,
                         int D, int S, int page_size, int max_pages,
                         int64_t page_stride, float scale, hipStream_t stream) {
     const int G = NH / NKV;
@@ -763,7 +864,8 @@ void launch_attn_decode(const float* qkv, const float* inv_freq,
     #define ATTN_CASE(GV, DV)                                                   \
         hipLaunchKernelGGL((k_attn_decode<GV, DV>), grid, block, 0, stream,     \
             qkv, inv_freq, page_table, kv_pool, n_past, part_o, part_ml,        \
-            NH, NKV, S, page_size, max_pages, page_stride, scale)
+            tickets, attn_out, NH, NKV, S, page_size, max_pages, page_stride,   \
+            scale)
     #define ATTN_D(GV)                                                          \
         do { if (D == 128) ATTN_CASE(GV, 128);                                  \
              else if (D == 64) ATTN_CASE(GV, 64);                               \
