@@ -315,74 +315,16 @@ __global__ __launch_bounds__(256) void k_gemv(
     }
     __syncthreads();
 
-    // ---- per-wave sweep over RPW rows, 2 chunks in flight per row ----
-    // (pair-unrolled: chunks c and c+64 decoded per iteration with the
-    // next pair prefetched — 2x the outstanding loads of a 1-deep ring,
-    // which is what keeps a 2-chunk/lane row, e.g. DQ4K K=4096, streaming)
+    // ---- per-wave sweep over RPW rows ----
     float acc[RPW][2];  // [row][b], B <= 2
     #pragma unroll
     for (int i = 0; i < RPW; i++) { acc[i][0] = 0.f; acc[i][1] = 0.f; }
-    ChunkRaw<W> cur2[RPW], nxt2[RPW];
-    {
-        const int c1 = (lane + 64 < n_chunks) ? lane + 64 : c0;
+    for (int c = lane; c < n_chunks; c += 64) {
+        const int cn = (c + 64 < n_chunks) ? c + 64 : c;  // clamped prefetch
         #pragma unroll
         for (int i = 0; i < RPW; i++)
             load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
-                          hdr + (int64_t)rc[i] * hdr_rb, c1, &cur2[i]);
-    }
-    int c = lane;
-    for (; c + 64 < n_chunks; c += 128) {
-        const int cn = (c + 128 < n_chunks) ? c + 128 : c;
-        const int cn2 = (c + 192 < n_chunks) ? c + 192 : c;
-        #pragma unroll
-        for (int i = 0; i < RPW; i++) {
-            load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
                           hdr + (int64_t)rc[i] * hdr_rb, cn, &nxt[i]);
-            load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
-                          hdr + (int64_t)rc[i] * hdr_rb, cn2, &nxt2[i]);
-        }
-        #pragma unroll
-        for (int i = 0; i < RPW; i++) {
-            #pragma unroll
-            for (int half = 0; half < 2; half++) {
-                float w[WPC];
-                int k0;
-                decode_chunk_raw<W>(half ? cur2[i] : cur[i],
-                                    half ? c + 64 : c, w, &k0);
-                #pragma unroll 2
-                for (int b = 0; b < B; b++) {
-                    const float4* xb4 = reinterpret_cast<const float4*>(
-                        x_lds + (size_t)b * KP + xpad(k0));
-                    float sdot = 0.f;
-                    if constexpr (W == DT::DQ4K) {
-                        #pragma unroll
-                        for (int t4 = 0; t4 < 4; t4++) {
-                            const float4 xl = xb4[t4];
-                            const float4 xh = xb4[8 + t4];
-                            sdot += w[t4 * 4 + 0] * xl.x + w[t4 * 4 + 1] * xl.y
-                               + w[t4 * 4 + 2] * xl.z + w[t4 * 4 + 3] * xl.w;
-                            sdot += w[16 + t4 * 4 + 0] * xh.x
-                               + w[16 + t4 * 4 + 1] * xh.y
-                               + w[16 + t4 * 4 + 2] * xh.z
-                               + w[16 + t4 * 4 + 3] * xh.w;
-                        }
-                    } else {
-                        #pragma unroll
-                        for (int t4 = 0; t4 < WPC / 4; t4++) {
-                            const float4 xv = xb4[t4];
-                            sdot += w[t4 * 4 + 0] * xv.x + w[t4 * 4 + 1] * xv.y
-                               + w[t4 * 4 + 2] * xv.z + w[t4 * 4 + 3] * xv.w;
-                        }
-                    }
-                    acc[i][b] += sdot;
-                }
-            }
-            cur[i] = nxt[i];
-            cur2[i] = nxt2[i];
-        }
-    }
-    // tail: at most one chunk left per lane
-    if (c < n_chunks) {
         #pragma unroll
         for (int i = 0; i < RPW; i++) {
             float w[WPC];
@@ -392,27 +334,30 @@ __global__ __launch_bounds__(256) void k_gemv(
             for (int b = 0; b < B; b++) {
                 const float4* xb4 = reinterpret_cast<const float4*>(
                     x_lds + (size_t)b * KP + xpad(k0));
-                float s = 0.f;
+                float sdot = 0.f;
                 if constexpr (W == DT::DQ4K) {
                     #pragma unroll
                     for (int t4 = 0; t4 < 4; t4++) {
                         const float4 xl = xb4[t4];
                         const float4 xh = xb4[8 + t4];
-                        s += w[t4 * 4 + 0] * xl.x + w[t4 * 4 + 1] * xl.y
+                        sdot += w[t4 * 4 + 0] * xl.x + w[t4 * 4 + 1] * xl.y
                            + w[t4 * 4 + 2] * xl.z + w[t4 * 4 + 3] * xl.w;
-                        s += w[16 + t4 * 4 + 0] * xh.x + w[16 + t4 * 4 + 1] * xh.y
-                           + w[16 + t4 * 4 + 2] * xh.z + w[16 + t4 * 4 + 3] * xh.w;
+                        sdot += w[16 + t4 * 4 + 0] * xh.x
+                           + w[16 + t4 * 4 + 1] * xh.y
+                           + w[16 + t4 * 4 + 2] * xh.z
+                           + w[16 + t4 * 4 + 3] * xh.w;
                     }
                 } else {
                     #pragma unroll
                     for (int t4 = 0; t4 < WPC / 4; t4++) {
                         const float4 xv = xb4[t4];
-                        s += w[t4 * 4 + 0] * xv.x + w[t4 * 4 + 1] * xv.y
+                        sdot += w[t4 * 4 + 0] * xv.x + w[t4 * 4 + 1] * xv.y
                            + w[t4 * 4 + 2] * xv.z + w[t4 * 4 + 3] * xv.w;
                     }
                 }
-                acc[i][b] += s;
+                acc[i][b] += sdot;
             }
+            cur[i] = nxt[i];
         }
     }
     #pragma unroll
