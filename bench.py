@@ -102,11 +102,14 @@ def main() -> None:
     log(f"rank {rank}: engine loaded in {time.time() - t0:.1f}s "
         f"({eng.vram_bytes() / 1e9:.2f} GB VRAM)")
 
-    # synthetic prompt prefill (untimed)
+    # synthetic prompt prefill (untimed in the decode metric; measured for
+    # the p50-latency story)
     rng = np.random.default_rng(1234 + rank)
     prompts = rng.integers(3, cfg.vocab_size - 1,
                            size=(args.batch, args.prompt_len)).astype(np.int32)
+    t_pf = time.perf_counter()
     eng.prefill(prompts)
+    prefill_ms = (time.perf_counter() - t_pf) * 1e3
 
     # warmup
     if args.warmup > 0:
@@ -155,6 +158,7 @@ def main() -> None:
                 "seq_len": args.prompt_len,
                 "parallelism": (f"tp{world}" if args.mode == "tp" else f"dp{n_gpus}"),
                 "engine_ms_per_step": eng.last_decode_ms() / args.steps,
+                "prefill_ms": prefill_ms,
             },
         }
         print(json.dumps(result), flush=True)

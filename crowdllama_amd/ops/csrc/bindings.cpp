@@ -20,6 +20,8 @@ double bench_gemv(const void*, const void*, int, int, int, int, int, size_t,
 void launch_gemm_test(const void*, const void*, const float*, float*, int,
                       int, int, int, size_t, size_t);
 double bench_gemm(int, int, int, int, int);
+std::vector<uint8_t> slice_cols_test(int32_t, const uint8_t*, int64_t,
+                                     int64_t, int64_t, int64_t);
 }
 
 PYBIND11_MODULE(_core, m) {
@@ -123,6 +125,11 @@ PYBIND11_MODULE(_core, m) {
     });
     m.def("bench_gemm", [](int dtype, int M, int N, int K, int iters) {
         return bench_gemm(dtype, M, N, K, iters);
+    });
+    m.def("test_slice_cols", [](int ggml_type, py::array_t<uint8_t> raw,
+                                int rows, int k, int c0, int c1) {
+        auto out = slice_cols_test(ggml_type, raw.data(), rows, k, c0, c1);
+        return py::bytes(reinterpret_cast<const char*>(out.data()), out.size());
     });
 
     m.def("bench_gemv", [](int dtype, int N, int K, int B, int pre, int iters) {

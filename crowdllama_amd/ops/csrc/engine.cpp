@@ -202,6 +202,13 @@ std::vector<uint8_t> slice_cols(int32_t ggml_type, const uint8_t* src,
 
 }  // namespace
 
+// test-only export of the TP column-slice helper
+std::vector<uint8_t> slice_cols_test(int32_t t, const uint8_t* src,
+                                     int64_t rows, int64_t k, int64_t c0,
+                                     int64_t c1) {
+    return slice_cols(t, src, rows, k, c0, c1);
+}
+
 Engine::Engine(const std::string& gguf_path, const EngineConfig& cfg)
     : cfg_(cfg) {
     if (cfg_.batch < 1 || cfg_.batch > 64)

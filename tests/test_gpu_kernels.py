@@ -219,3 +219,23 @@ def test_gemm_bf16(core):
     yref = bf16(x) @ wref.T
     scale = np.abs(yref).max() + 1e-6
     assert np.abs(y - yref).max() / scale < 5e-3
+
+
+def test_tp_slice_cols(core):
+    """C++ column-slice helper (TP shards) vs numpy byte slicing."""
+    from crowdllama_amd.quant import quantize
+    rng = np.random.default_rng(11)
+    rows, K = 4, 1024
+    w = rng.standard_normal((rows, K)).astype(np.float32)
+    raw = quantize_q4_k(w).reshape(rows, -1)
+    got = core.test_slice_cols(12, np.ascontiguousarray(raw), rows, K,
+                               256, 768)
+    # expected: per-row bytes for superblocks 1..2 (144 B each)
+    want = raw[:, 144:432].tobytes()
+    assert got == want
+    # f32 slice
+    raw32 = w.view(np.uint8)
+    got = core.test_slice_cols(0, np.ascontiguousarray(raw32), rows, K,
+                               256, 512)
+    want = w[:, 256:512].astype(np.float32).tobytes()
+    assert got == want
