@@ -315,23 +315,14 @@ __global__ __launch_bounds__(256) void k_gemv(
     }
     __syncthreads();
 
-    // ---- per-wave sweep over RPW rows, prefetch distance 2 ----
-    // 3-buffer rotation (cur <- pre1 <- nxt): each iteration issues one new
-    // chunk and decodes one loaded two iterations ago, keeping 2 chunks'
-    // loads outstanding per row without unroll register blowup.
+    // ---- per-wave sweep over RPW rows (2-buffer, prefetch distance 1;
+    // measured best: deeper rotations and pair-unrolls regress on register
+    // pressure — see profiles/ notes) ----
     float acc[RPW][2];  // [row][b], B <= 2
     #pragma unroll
     for (int i = 0; i < RPW; i++) { acc[i][0] = 0.f; acc[i][1] = 0.f; }
-    ChunkRaw<W> pre1[RPW];
-    {
-        const int c1 = (lane + 64 < n_chunks) ? lane + 64 : c0;
-        #pragma unroll
-        for (int i = 0; i < RPW; i++)
-            load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
-                          hdr + (int64_t)rc[i] * hdr_rb, c1, &pre1[i]);
-    }
     for (int c = lane; c < n_chunks; c += 64) {
-        const int cn = (c + 128 < n_chunks) ? c + 128 : c;  // clamped
+        const int cn = (c + 64 < n_chunks) ? c + 64 : c;  // clamped
         #pragma unroll
         for (int i = 0; i < RPW; i++)
             load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
@@ -368,8 +359,7 @@ __global__ __launch_bounds__(256) void k_gemv(
                 }
                 acc[i][b] += sdot;
             }
-            cur[i] = pre1[i];
-            pre1[i] = nxt[i];
+            cur[i] = nxt[i];
         }
     }
     #pragma unroll
