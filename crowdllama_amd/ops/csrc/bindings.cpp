@@ -20,6 +20,7 @@ double bench_gemv(const void*, const void*, int, int, int, int, int, size_t,
 void launch_gemm_test(const void*, const void*, const float*, float*, int,
                       int, int, int, size_t, size_t);
 double bench_gemm(int, int, int, int, int);
+double bench_gemv_g(int, int, int, int, int);
 std::vector<uint8_t> slice_cols_test(int32_t, const uint8_t*, int64_t,
                                      int64_t, int64_t, int64_t);
 }
@@ -132,6 +133,9 @@ PYBIND11_MODULE(_core, m) {
         return py::bytes(reinterpret_cast<const char*>(out.data()), out.size());
     });
 
+    m.def("bench_gemv_g", [](int dtype, int N, int K, int B, int iters) {
+        return bench_gemv_g(dtype, N, K, B, iters);
+    });
     m.def("bench_gemv", [](int dtype, int N, int K, int B, int pre, int iters) {
         // random weight bytes (content irrelevant for timing)
         const DT dt = static_cast<DT>(dtype);

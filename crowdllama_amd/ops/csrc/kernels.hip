@@ -376,6 +376,78 @@ __global__ __launch_bounds__(256) void k_gemv(
     }
 }
 
+// Global-x GEMV: no LDS staging — the activation vector (<=57 KB) is
+// L1-resident per CU after first touch, so reading it directly unlocks
+// full occupancy (no 16 KB+ LDS budget per workgroup) and removes the
+// staging phase + barrier entirely (guide: don't LDS-stage what L1/L2
+// fits). xin must already be normalized/activated (k_rmsnorm_rows /
+// k_silu_rows run once per projection input).
+template <DT W>
+__global__ __launch_bounds__(256) void k_gemv_g(
+    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
+    const float* __restrict__ xin, const float* __restrict__ res,
+    float* __restrict__ y, int N, int K, int B, int ldy) {
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6, lane = tid & 63;
+    const int r = (int)blockIdx.x * 4 + wave;
+    const int rc = r < N ? r : N - 1;
+    const int64_t qs_rb = dqs_row_bytes(W, K);
+    const int64_t hdr_rb = dhdr_row_bytes(W, K);
+    const int n_chunks = (int)(qs_rb / 16);
+    constexpr int WPC = ChunkTraits<W>::W_PER_CHUNK;
+
+    ChunkRaw<W> cur, nxt;
+    const int c0 = lane < n_chunks ? lane : 0;
+    load_chunk<W>(qs + (int64_t)rc * qs_rb, hdr + (int64_t)rc * hdr_rb, c0,
+                  &cur);
+    float acc[2] = {0.f, 0.f};
+    for (int c = lane; c < n_chunks; c += 64) {
+        const int cn = (c + 64 < n_chunks) ? c + 64 : c;
+        load_chunk<W>(qs + (int64_t)rc * qs_rb, hdr + (int64_t)rc * hdr_rb,
+                      cn, &nxt);
+        float w[WPC];
+        int k0;
+        decode_chunk_raw<W>(cur, c, w, &k0);
+        #pragma unroll 2
+        for (int b = 0; b < B; b++) {
+            const float4* xb4 = reinterpret_cast<const float4*>(
+                xin + (size_t)b * K + k0);
+            float sdot = 0.f;
+            if constexpr (W == DT::DQ4K) {
+                #pragma unroll
+                for (int t4 = 0; t4 < 4; t4++) {
+                    const float4 xl = xb4[t4];
+                    const float4 xh = xb4[8 + t4];
+                    sdot += w[t4 * 4 + 0] * xl.x + w[t4 * 4 + 1] * xl.y
+                       + w[t4 * 4 + 2] * xl.z + w[t4 * 4 + 3] * xl.w;
+                    sdot += w[16 + t4 * 4 + 0] * xh.x + w[16 + t4 * 4 + 1] * xh.y
+                       + w[16 + t4 * 4 + 2] * xh.z + w[16 + t4 * 4 + 3] * xh.w;
+                }
+            } else {
+                #pragma unroll
+                for (int t4 = 0; t4 < WPC / 4; t4++) {
+                    const float4 xv = xb4[t4];
+                    sdot += w[t4 * 4 + 0] * xv.x + w[t4 * 4 + 1] * xv.y
+                       + w[t4 * 4 + 2] * xv.z + w[t4 * 4 + 3] * xv.w;
+                }
+            }
+            acc[b] += sdot;
+        }
+        cur = nxt;
+    }
+    if (r < N) {
+        #pragma unroll 2
+        for (int b = 0; b < B; b++) {
+            float v = wave_reduce_sum(acc[b]);
+            if (lane == 0) {
+                const int64_t idx = (int64_t)b * ldy + r;
+                y[idx] = v + (res ? res[idx] : 0.f);
+            }
+        }
+    }
+}
+
+
 // ------------------------------------------------------------- embedding
 
 template <DT W>
@@ -789,6 +861,20 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
     #undef GEMV_CASE
     #undef GEMV_RPW
 }
+
+void launch_gemv_g(const WTensor& w, const float* xin, const float* res,
+                   float* y, int B, int ldy, hipStream_t stream) {
+    const int N = (int)w.n, K = (int)w.k;
+    if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
+    dim3 grid((N + 3) / 4), block(256);
+    #define GEMVG_CASE(WT)                                                     \
+        hipLaunchKernelGGL(k_gemv_g<WT>, grid, block, 0, stream,               \
+            (const uint8_t*)w.qs, (const uint8_t*)w.hdr, xin, res, y,          \
+            N, K, B, ldy)
+    DISPATCH_DT(w.dtype, GEMVG_CASE);
+    #undef GEMVG_CASE
+}
+
 
 void launch_embed(const WTensor& w, const int32_t* ids, float* x, int B,
                   hipStream_t stream) {
