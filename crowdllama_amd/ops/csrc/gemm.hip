@@ -33,25 +33,64 @@ __device__ __forceinline__ float f16b_to_f32(uint32_t h) {
     return __half2float(*reinterpret_cast<__half*>(&r));
 }
 
+typedef unsigned int u32x4g __attribute__((ext_vector_type(4)));
+
+// Raw staged bytes for one thread's 32-weight W-tile slice, loaded one
+// K-step early so HBM latency hides under the previous tile's MFMA phase
+// (T14 split). Quant formats only; float formats load synchronously.
+struct WRaw {
+    u32x4g q0, q1;   // DQ4K uses q0 only
+    uint2 hd;        // DQ4K pre-decoded pair header
+    float sc0, sc1;  // DQ6K/DQ8 effective scales
+};
+
+template <DT W>
+__device__ __forceinline__ void load_w_raw(
+    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
+    int64_t row, int K, int kb, int h, WRaw* r) {
+    if constexpr (W == DT::DQ4K) {
+        const int sb = kb >> 8, q = (kb & 255) >> 6, p = 2 * q + h;
+        const uint8_t* qrow = qs + row * (K / 256) * 128;
+        r->q0 = __builtin_nontemporal_load(
+            reinterpret_cast<const u32x4g*>(qrow) + sb * 8 + p);
+        r->hd = reinterpret_cast<const uint2*>(
+            hdr + row * (K / 256) * 32)[sb * 4 + p / 2];
+    } else if constexpr (W == DT::DQ6K || W == DT::DQ8) {
+        const int k0 = kb + h * 32;
+        const int8_t* qrow = reinterpret_cast<const int8_t*>(qs + row * K);
+        r->q0 = __builtin_nontemporal_load(
+            reinterpret_cast<const u32x4g*>(qrow + k0));
+        r->q1 = __builtin_nontemporal_load(
+            reinterpret_cast<const u32x4g*>(qrow + k0) + 1);
+        if constexpr (W == DT::DQ6K) {
+            const uint8_t* hb = hdr + row * (K / 256) * 32 + (k0 >> 8) * 32;
+            const float d = f16b_to_f32(*reinterpret_cast<const uint16_t*>(hb));
+            const int s16 = (k0 & 255) >> 4;
+            r->sc0 = d * (float)(reinterpret_cast<const int8_t*>(hb)[4 + s16]);
+            r->sc1 = d * (float)(reinterpret_cast<const int8_t*>(hb)[4 + s16 + 1]);
+        } else {
+            const uint16_t* drow = reinterpret_cast<const uint16_t*>(
+                hdr + row * (K / 32) * 2);
+            r->sc0 = f16b_to_f32(drow[k0 >> 5]);
+            r->sc1 = r->sc0;
+        }
+    } else {
+        (void)qs; (void)hdr; (void)row; (void)K; (void)kb; (void)h; (void)r;
+    }
+}
+
 // Dequantize this thread's 32-weight slice of the W tile into `out` bf16.
 // Thread t covers W row (tile_row = t>>1), k-halves h = t&1 within [kb,kb+BK).
-// Returns false if the row is out of range (caller zero-fills).
 template <DT W>
 __device__ __forceinline__ void stage_w_slice(
     const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
-    int64_t row, int K, int kb, int h, uint16_t out[32]) {
+    const WRaw& rr, int64_t row, int K, int kb, int h, uint16_t out[32]) {
     if constexpr (W == DT::DQ4K) {
-        const int sb = kb >> 8;
         const int q = (kb & 255) >> 6;
         const int p = 2 * q + h;
-        const uint8_t* qrow = qs + row * (K / 256) * 128;
-        const uint2* hrow = reinterpret_cast<const uint2*>(
-            hdr + row * (K / 256) * 32);
-        typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
-        const u32x4 qv = __builtin_nontemporal_load(
-            reinterpret_cast<const u32x4*>(qrow) + sb * 8 + p);
-        const uint32_t dw[4] = {qv.x, qv.y, qv.z, qv.w};
-        const uint2 hd = hrow[sb * 4 + p / 2];   // pre-decoded pair header
+        (void)p;
+        const uint32_t dw[4] = {rr.q0.x, rr.q0.y, rr.q0.z, rr.q0.w};
+        const uint2 hd = rr.hd;
         const float d = f16b_to_f32(hd.x & 0xFFFF);
         const float dmin = f16b_to_f32(hd.x >> 16);
         const float dl = d * (float)(hd.y & 0xFF);
@@ -72,27 +111,9 @@ __device__ __forceinline__ void stage_w_slice(
             }
         }
     } else if constexpr (W == DT::DQ6K || W == DT::DQ8) {
-        const int k0 = kb + h * 32;
-        const int8_t* qrow = reinterpret_cast<const int8_t*>(qs + row * K);
-        typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
-        const u32x4 q0 = __builtin_nontemporal_load(
-            reinterpret_cast<const u32x4*>(qrow + k0));
-        const u32x4 q1 = __builtin_nontemporal_load(
-            reinterpret_cast<const u32x4*>(qrow + k0) + 1);
-        const uint32_t dw[8] = {q0.x, q0.y, q0.z, q0.w, q1.x, q1.y, q1.z, q1.w};
-        float sc[2];
-        if constexpr (W == DT::DQ6K) {
-            const uint8_t* hb = hdr + row * (K / 256) * 32 + (k0 >> 8) * 32;
-            const float d = f16b_to_f32(*reinterpret_cast<const uint16_t*>(hb));
-            const int s16 = (k0 & 255) >> 4;
-            sc[0] = d * (float)(reinterpret_cast<const int8_t*>(hb)[4 + s16]);
-            sc[1] = d * (float)(reinterpret_cast<const int8_t*>(hb)[4 + s16 + 1]);
-        } else {
-            const uint16_t* drow = reinterpret_cast<const uint16_t*>(
-                hdr + row * (K / 32) * 2);
-            sc[0] = f16b_to_f32(drow[k0 >> 5]);
-            sc[1] = sc[0];  // 32 weights span exactly one Q8 block
-        }
+        const uint32_t dw[8] = {rr.q0.x, rr.q0.y, rr.q0.z, rr.q0.w,
+                                rr.q1.x, rr.q1.y, rr.q1.z, rr.q1.w};
+        const float sc[2] = {rr.sc0, rr.sc1};
         #pragma unroll
         for (int j = 0; j < 8; j++) {
             const float s = sc[j >> 2];
@@ -127,6 +148,7 @@ __device__ __forceinline__ void stage_w_slice(
         #pragma unroll
         for (int t = 0; t < 32; t++) out[t] = f32_to_bf16b(wrow[t]);
     }
+    (void)qs; (void)hdr; (void)row; (void)K; (void)kb; (void)h;
 }
 
 }  // namespace
@@ -161,6 +183,12 @@ __global__ __launch_bounds__(256) void k_gemm(
         #pragma unroll
         for (int j = 0; j < 4; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+    const int srow0 = tid >> 1, sh0 = tid & 1;   // W staging coords
+    const int64_t gn_s = (int64_t)n0 + srow0;
+    const int64_t gn_c = gn_s < N ? gn_s : N - 1;  // clamped (branchless)
+    WRaw wr, wr_next;
+    load_w_raw<W>(qs, hdr, gn_c, K, kb_lo, sh0, &wr);
+
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
         // ---- stage X tile (f32 -> bf16), 8-elem units ----
         {
@@ -191,13 +219,12 @@ __global__ __launch_bounds__(256) void k_gemm(
                     *reinterpret_cast<const uint4*>(tmp);
             }
         }
-        // ---- stage + dequant W tile ----
+        // ---- stage + dequant W tile (quant raw bytes preloaded) ----
         {
-            const int srow = tid >> 1, sh = tid & 1;
+            const int srow = srow0, sh = sh0;
             uint16_t tmp[32];
-            const int64_t gn = (int64_t)n0 + srow;
-            if (gn < N) {
-                stage_w_slice<W>(qs, hdr, gn, K, kb, sh, tmp);
+            if (gn_s < N) {
+                stage_w_slice<W>(qs, hdr, wr, gn_c, K, kb, sh, tmp);
             } else {
                 #pragma unroll
                 for (int j = 0; j < 32; j++) tmp[j] = 0;
@@ -220,6 +247,12 @@ __global__ __launch_bounds__(256) void k_gemm(
             }
         }
         __syncthreads();
+        // T14: next W tile's loads issue before the MFMA phase — their
+        // HBM latency hides under the matrix work.
+        {
+            const int kbn = (kb + BK < kb_hi) ? kb + BK : kb;
+            load_w_raw<W>(qs, hdr, gn_c, K, kbn, sh0, &wr_next);
+        }
         // ---- MFMA over the tile ----
         #pragma unroll
         for (int ks = 0; ks < BK; ks += 32) {
@@ -244,6 +277,7 @@ __global__ __launch_bounds__(256) void k_gemm(
                         a[i], b[j], acc[i][j], 0, 0, 0);
         }
         __syncthreads();
+        wr = wr_next;
     }
 
     // ---- epilogue ----
