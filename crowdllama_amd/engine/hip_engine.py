@@ -9,7 +9,7 @@ import time
 from concurrent.futures import ThreadPoolExecutor
 
 from ..quant.gguf import GGUFReader
-from ..tokenizer import Tokenizer
+from ..tokenizer import NativeTokenizer, Tokenizer
 from .api import EngineBase, GenerateResult, RollingRate
 
 
@@ -32,7 +32,10 @@ class HipEngine(EngineBase):
         cfg.use_graph = use_graph
         self.eng = core.Engine(gguf_path, cfg)
         with GGUFReader(gguf_path) as r:
-            self.tok = Tokenizer.from_gguf(r)
+            try:
+                self.tok = NativeTokenizer.from_gguf(r)
+            except Exception:
+                self.tok = Tokenizer.from_gguf(r)
         self._props = core.device_props(device)
         self._rate = RollingRate()
         self._lock = threading.Lock()

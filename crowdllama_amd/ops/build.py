@@ -18,7 +18,7 @@ CSRC = os.path.join(OPS_DIR, "csrc")
 ARCH = os.environ.get("CLA_GFX_ARCH", "gfx950")
 
 SOURCES = ["gguf.cpp", "engine.cpp", "testutil.cpp", "bindings.cpp",
-           "kernels.hip", "mfma_probe.hip", "gemm.hip"]
+           "kernels.hip", "mfma_probe.hip", "gemm.hip", "tokenizer.cpp"]
 
 
 def _ext_suffix() -> str:

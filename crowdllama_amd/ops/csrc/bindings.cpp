@@ -7,6 +7,7 @@
 #include <rccl/rccl.h>
 
 #include "engine.h"
+#include "../csrc/tokenizer_api.h"
 
 namespace py = pybind11;
 using namespace cla;
@@ -50,6 +51,34 @@ PYBIND11_MODULE(_core, m) {
         d["warp_size"] = p.warpSize;
         return d;
     });
+
+    struct PyTokenizer {
+        Tokenizer* t;
+        PyTokenizer(std::vector<std::string> tokens,
+                    std::vector<std::string> merges, int bos, int eos)
+            : t(tokenizer_new(std::move(tokens), merges, bos, eos)) {}
+        ~PyTokenizer() { tokenizer_free(t); }
+        PyTokenizer(const PyTokenizer&) = delete;
+    };
+    py::class_<PyTokenizer>(m, "Tokenizer")
+        .def(py::init<std::vector<std::string>, std::vector<std::string>,
+                      int, int>(),
+             py::arg("tokens"), py::arg("merges") = std::vector<std::string>{},
+             py::arg("bos_id") = 1, py::arg("eos_id") = 2)
+        .def("encode",
+             [](const PyTokenizer& pt, const std::string& s, bool add_bos) {
+                 return tokenizer_encode(pt.t, s, add_bos);
+             }, py::arg("text"), py::arg("add_bos") = true)
+        .def("decode",
+             [](const PyTokenizer& pt, std::vector<int32_t> ids) {
+                 return py::bytes(tokenizer_decode(pt.t, ids));
+             })
+        .def_property_readonly("bos_id",
+             [](const PyTokenizer& pt) { return tokenizer_bos(pt.t); })
+        .def_property_readonly("eos_id",
+             [](const PyTokenizer& pt) { return tokenizer_eos(pt.t); })
+        .def("__len__",
+             [](const PyTokenizer& pt) { return tokenizer_size(pt.t); });
 
     py::class_<EngineConfig>(m, "EngineConfig")
         .def(py::init<>())

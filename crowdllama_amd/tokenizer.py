@@ -2,9 +2,14 @@
 tokenizer inside llama.cpp that Ollama uses — SURVEY.md §2.3 row
 "Tokenizer").
 
-Supports:
-- byte-fallback vocabularies (synthetic checkpoints: <0xNN> byte tokens),
-- GPT-2-style byte-level BPE when `tokenizer.ggml.merges` is present.
+Two implementations with identical behavior:
+- Tokenizer: pure-Python reference (CPU tests, fallback),
+- NativeTokenizer: the C++ implementation in the engine extension
+  (ops/csrc/tokenizer.cpp), used by HipEngine.
+
+Supports byte-fallback vocabularies (<0xNN> byte tokens, as the synthetic
+checkpoints emit) and GPT-2-style byte-level BPE when
+`tokenizer.ggml.merges` is present.
 """
 
 from __future__ import annotations
@@ -115,9 +120,7 @@ class Tokenizer:
             if not (0 <= i < len(self.tokens)):
                 continue
             tok = self.tokens[i]
-            if i in rev_byte.keys() or tok in ():
-                pass
-            if i in set(self.byte_tokens.values()):
+            if i in rev_byte:
                 out.append(rev_byte[i])
             elif u2b is not None:
                 for ch in tok:
@@ -131,3 +134,32 @@ class Tokenizer:
 
     def __len__(self) -> int:
         return len(self.tokens)
+
+
+class NativeTokenizer:
+    """C++ tokenizer from the engine extension (ops/csrc/tokenizer.cpp)."""
+
+    def __init__(self, tokens, merges=None, bos_id=1, eos_id=2):
+        from .ops import get_core
+        core = get_core()
+        self._t = core.Tokenizer(list(tokens), list(merges or []),
+                                 bos_id, eos_id)
+        self.bos_id = bos_id
+        self.eos_id = eos_id
+
+    @classmethod
+    def from_gguf(cls, reader) -> "NativeTokenizer":
+        md = reader.metadata
+        return cls(tokens=md.get("tokenizer.ggml.tokens", []),
+                   merges=md.get("tokenizer.ggml.merges"),
+                   bos_id=md.get("tokenizer.ggml.bos_token_id", 1),
+                   eos_id=md.get("tokenizer.ggml.eos_token_id", 2))
+
+    def encode(self, text: str, add_bos: bool = True) -> list[int]:
+        return list(self._t.encode(text, add_bos))
+
+    def decode(self, ids: list[int]) -> str:
+        return self._t.decode(list(ids)).decode("utf-8", errors="replace")
+
+    def __len__(self) -> int:
+        return len(self._t)
