@@ -42,6 +42,8 @@ void launch_argmax(const float* logits, float* pval, int32_t* pidx,
 void launch_gemm(const WTensor&, const float* X, const float* res, float* C,
                  int M, int ldc, hipStream_t);
 bool gemm_uses_splitk(int N, int K, int M);
+void launch_zero4(float* p0, int64_t n0, float* p1, int64_t n1, float* p2,
+                  int64_t n2, float* p3, int64_t n3, hipStream_t);
 void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
                          int K, float eps, hipStream_t);
 void launch_silu_rows(const float* GU, float* out, int M, int F, hipStream_t);
@@ -461,6 +463,7 @@ void Engine::alloc_state() {
     x_ = (float*)dalloc((size_t)B * H * 4);
     xn_ = (float*)dalloc((size_t)B * H * 4);
     x2_ = (float*)dalloc((size_t)B * H * 4);
+    x3_ = (float*)dalloc((size_t)B * H * 4);
     tmp_h_ = (float*)dalloc((size_t)B * H * 4);
     act_ = (float*)dalloc((size_t)B * F * 4);
     {   // prefill scratch
@@ -519,11 +522,12 @@ void Engine::reset() {
 // Launch a projection through the GEMM path: pre-zero C when split-K
 // accumulation is in play (see launch_gemm), then one launch per part.
 static void gemm_proj(const Proj& p, const float* X, const float* res,
-                      float* C, int M, hipStream_t s) {
+                      float* C, int M, hipStream_t s,
+                      bool pre_zeroed = false) {
     bool zero = false;
     for (auto& pt : p.parts)
         zero |= gemm_uses_splitk((int)pt.w.n, (int)pt.w.k, M);
-    if (zero)
+    if (zero && !pre_zeroed)
         HIP_CHECK(hipMemsetAsync(C, 0, (size_t)M * p.n_total * 4, s));
     for (auto& pt : p.parts)
         launch_gemm(pt.w, X, res ? res + pt.row_off : nullptr,
@@ -553,8 +557,17 @@ void Engine::step(hipStream_t s) {
                 launch_gemv(pt.w, PRE_RMS, x_, L.attn_norm, nullptr,
                             qkv_ + pt.row_off, B, (int)L.qkv.n_total, eps, s);
         } else {
-            launch_rmsnorm_rows(x_, L.attn_norm, xn_, B, meta_.hidden, eps, s);
-            gemm_proj(L.qkv, xn_, nullptr, qkv_, B, s);
+            // 3-buffer rotation: in --(+attn)--> mid --(+ffn)--> out, with
+            // one fused clear of every split-K destination per layer
+            // (replaces 3-4 ~5 us hipMemsetAsync dispatches).
+            float* lin = ((li - 1) & 1) ? x2_ : x_;  // li already advanced
+            float* lout = ((li - 1) & 1) ? x_ : x2_;
+            launch_zero4(qkv_, (int64_t)B * (NH + 2 * NKV) * D,
+                         gu_, (int64_t)B * 2 * meta_.ffn_l,
+                         x3_, (int64_t)B * meta_.hidden,
+                         lout, (int64_t)B * meta_.hidden, s);
+            launch_rmsnorm_rows(lin, L.attn_norm, xn_, B, meta_.hidden, eps, s);
+            gemm_proj(L.qkv, xn_, nullptr, qkv_, B, s, /*pre_zeroed=*/true);
         }
         launch_attn_decode(qkv_, inv_freq_, page_table_, kv_layer, n_past_,
                            part_o_, part_ml_, attn_tickets_, attn_out_, B, NH,
@@ -587,30 +600,35 @@ void Engine::step(hipStream_t s) {
                             B, (int)L.down.n_total, eps, s);
             allreduce(tmp_h_, x_, (size_t)B * meta_.hidden);
         } else {
-            // residual ping-pong: x_ --(+attn)--> x2_ --(+ffn)--> x_
+            float* lin = ((li - 1) & 1) ? x2_ : x_;
+            float* lout = ((li - 1) & 1) ? x_ : x2_;
             if (tp) {
-                gemm_proj(L.o, attn_out_, r0 ? x_ : nullptr, tmp_h_, B, s);
-                allreduce(tmp_h_, x2_, (size_t)B * meta_.hidden);
+                gemm_proj(L.o, attn_out_, r0 ? lin : nullptr, tmp_h_, B, s);
+                allreduce(tmp_h_, x3_, (size_t)B * meta_.hidden);
             } else {
-                gemm_proj(L.o, attn_out_, x_, x2_, B, s);
+                gemm_proj(L.o, attn_out_, lin, x3_, B, s, /*pre_zeroed=*/true);
             }
-            launch_rmsnorm_rows(x2_, L.ffn_norm, xn_, B, meta_.hidden, eps, s);
-            gemm_proj(L.gate_up, xn_, nullptr, gu_, B, s);
+            launch_rmsnorm_rows(x3_, L.ffn_norm, xn_, B, meta_.hidden, eps, s);
+            gemm_proj(L.gate_up, xn_, nullptr, gu_, B, s, /*pre_zeroed=*/true);
             launch_silu_rows(gu_, act_, B, meta_.ffn_l, s);
             if (tp) {
-                gemm_proj(L.down, act_, r0 ? x2_ : nullptr, tmp_h_, B, s);
-                allreduce(tmp_h_, x_, (size_t)B * meta_.hidden);
+                gemm_proj(L.down, act_, r0 ? x3_ : nullptr, tmp_h_, B, s);
+                allreduce(tmp_h_, lout, (size_t)B * meta_.hidden);
             } else {
-                gemm_proj(L.down, act_, x2_, x_, B, s);
+                gemm_proj(L.down, act_, x3_, lout, B, s, /*pre_zeroed=*/true);
             }
         }
     }
     const int64_t voff = (int64_t)cfg_.tp_rank * meta_.vocab_l;
+    // GEMM path: the final residual lands in x_ for even layer counts,
+    // x2_ for odd (3-buffer rotation); the GEMV path is in-place in x_.
+    float* xfinal = x_;
+    if (!gemv_path && (meta_.layers & 1)) xfinal = x2_;
     if (gemv_path) {
         launch_gemv(head_, PRE_RMS, x_, out_norm_, nullptr, logits_ + voff, B,
                     meta_.vocab, eps, s);
     } else {
-        launch_rmsnorm_rows(x_, out_norm_, xn_, B, meta_.hidden, eps, s);
+        launch_rmsnorm_rows(xfinal, out_norm_, xn_, B, meta_.hidden, eps, s);
         Proj hp; hp.parts.push_back({head_, 0}); hp.n_total = meta_.vocab_l;
         // write the local vocab slice at its global offset (ldc = full V)
         bool zero = gemm_uses_splitk((int)head_.n, (int)head_.k, B);

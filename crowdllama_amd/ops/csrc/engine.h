@@ -117,6 +117,7 @@ private:
     float* x_ = nullptr;        // [B][h]
     float* xn_ = nullptr;       // [B][h] (GEMM decode path rmsnorm out)
     float* x2_ = nullptr;       // [B][h] residual ping-pong (split-K GEMM)
+    float* x3_ = nullptr;       // [B][h] mid buffer (post-attention)
     float* act_ = nullptr;      // [B][F]
     // prefill scratch ([Mchunk] rows)
     float* xp_ = nullptr;

@@ -188,10 +188,40 @@ __global__ __launch_bounds__(256) void k_gemm(
     const int64_t gn_c = gn_s < N ? gn_s : N - 1;  // clamped (branchless)
     WRaw wr, wr_next;
     load_w_raw<W>(qs, hdr, gn_c, K, kb_lo, sh0, &wr);
+    // BM=32: the X tile is 8 floats/thread — prefetch it alongside W so the
+    // whole staging phase runs from registers (M<=32 decode batches have
+    // too few workgroups to hide latency with occupancy alone).
+    float4 xr0, xr1, xr0n, xr1n;
+    const int xrow = (BM_ == 32) ? (tid >> 3) : 0;
+    const int xseg = (BM_ == 32) ? (tid & 7) : 0;
+    const int xgm = m0 + xrow < M ? m0 + xrow : (M > 0 ? M - 1 : 0);
+    const bool xvalid0 = (m0 + xrow) < M;
+    if (BM_ == 32) {
+        const float4* src = reinterpret_cast<const float4*>(
+            X + (size_t)xgm * K + kb_lo + xseg * 8);
+        xr0 = src[0];
+        xr1 = src[1];
+    }
 
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
         // ---- stage X tile (f32 -> bf16), 8-elem units ----
-        {
+        if (BM_ == 32) {
+            uint16_t tmp[8];
+            const float4 v[2] = {xr0, xr1};
+            #pragma unroll
+            for (int j = 0; j < 2; j++) {
+                tmp[j * 4 + 0] = f32_to_bf16b(v[j].x);
+                tmp[j * 4 + 1] = f32_to_bf16b(v[j].y);
+                tmp[j * 4 + 2] = f32_to_bf16b(v[j].z);
+                tmp[j * 4 + 3] = f32_to_bf16b(v[j].w);
+            }
+            if (!xvalid0) {
+                #pragma unroll
+                for (int j = 0; j < 8; j++) tmp[j] = 0;
+            }
+            *reinterpret_cast<uint4*>(Xl + xrow * LDW + xseg * 8) =
+                *reinterpret_cast<const uint4*>(tmp);
+        } else {
             constexpr int UNITS = BM_ * BK / 8;   // 8 bf16 per unit
             #pragma unroll
             for (int ui = 0; ui < (UNITS + 255) / 256; ui++) {
@@ -252,6 +282,12 @@ __global__ __launch_bounds__(256) void k_gemm(
         {
             const int kbn = (kb + BK < kb_hi) ? kb + BK : kb;
             load_w_raw<W>(qs, hdr, gn_c, K, kbn, sh0, &wr_next);
+            if (BM_ == 32) {
+                const float4* src = reinterpret_cast<const float4*>(
+                    X + (size_t)xgm * K + kbn + xseg * 8);
+                xr0n = src[0];
+                xr1n = src[1];
+            }
         }
         // ---- MFMA over the tile ----
         #pragma unroll
@@ -278,6 +314,7 @@ __global__ __launch_bounds__(256) void k_gemm(
         }
         __syncthreads();
         wr = wr_next;
+        if (BM_ == 32) { xr0 = xr0n; xr1 = xr1n; }
     }
 
     // ---- epilogue ----
@@ -492,7 +529,7 @@ __global__ __launch_bounds__(256) void k_attn_prefill(
 bool gemm_uses_splitk(int N, int K, int M) {
     if (M > 32) return false;
     const int n_tiles = (N + BN - 1) / BN;
-    int splitk = K / BK < 512 / n_tiles ? K / BK : 512 / n_tiles;
+    int splitk = K / BK < 768 / n_tiles ? K / BK : 768 / n_tiles;
     return splitk > 1;
 }
 
@@ -507,9 +544,9 @@ void launch_gemm(const WTensor& w, const float* X, const float* res, float* C,
     // >=512 workgroups, bounded by the number of K-steps.
     int splitk = 1;
     if (small_m) {
-        splitk = K / BK < 512 / (n_tiles ? n_tiles : 1)
+        splitk = K / BK < 768 / (n_tiles ? n_tiles : 1)
                      ? K / BK
-                     : 512 / (n_tiles ? n_tiles : 1);
+                     : 768 / (n_tiles ? n_tiles : 1);
         if (splitk < 1) splitk = 1;
     }
     const int k_chunk = ((K / BK + splitk - 1) / splitk) * BK;

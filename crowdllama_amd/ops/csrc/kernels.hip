@@ -211,10 +211,14 @@ __global__ __launch_bounds__(256) void k_gemv(
     for (int i = 0; i < RPW; i++)
         rc[i] = r[i] < N ? r[i] : N - 1;
     const int c0 = lane < n_chunks ? lane : 0;
+    const int c1 = (lane + 64 < n_chunks) ? lane + 64 : c0;
     #pragma unroll
-    for (int i = 0; i < RPW; i++)
+    for (int i = 0; i < RPW; i++) {
         load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
                       hdr + (int64_t)rc[i] * hdr_rb, c0, &cur[i]);
+        load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
+                      hdr + (int64_t)rc[i] * hdr_rb, c1, &nxt[i]);
+    }
 
     // ---- stage activations (vectorized float4; K always %4==0) ----
     // Two-phase batches: issue up to 4 independent global loads, then the
@@ -322,16 +326,17 @@ __global__ __launch_bounds__(256) void k_gemv(
     #pragma unroll
     for (int i = 0; i < RPW; i++) { acc[i][0] = 0.f; acc[i][1] = 0.f; }
     for (int c = lane; c < n_chunks; c += 64) {
-        const int cn = (c + 64 < n_chunks) ? c + 64 : c;  // clamped
-        #pragma unroll
-        for (int i = 0; i < RPW; i++)
-            load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
-                          hdr + (int64_t)rc[i] * hdr_rb, cn, &nxt[i]);
         #pragma unroll
         for (int i = 0; i < RPW; i++) {
             float w[WPC];
             int k0;
             decode_chunk_raw<W>(cur[i], c, w, &k0);
+            // refill the just-consumed buffer with chunk c+128, then swap:
+            // both first chunks were issued before staging, so the sweep
+            // never sees a cold load.
+            const int cn = (c + 128 < n_chunks) ? c + 128 : c0;
+            load_chunk<W>(qs + (int64_t)rc[i] * qs_rb,
+                          hdr + (int64_t)rc[i] * hdr_rb, cn, &cur[i]);
             #pragma unroll 2
             for (int b = 0; b < B; b++) {
                 const float4* xb4 = reinterpret_cast<const float4*>(
@@ -359,7 +364,9 @@ __global__ __launch_bounds__(256) void k_gemv(
                 }
                 acc[i][b] += sdot;
             }
+            ChunkRaw<W> t = cur[i];
             cur[i] = nxt[i];
+            nxt[i] = t;
         }
     }
     #pragma unroll
@@ -809,6 +816,35 @@ __global__ __launch_bounds__(64) void k_argmax_final(
         gen_count[b] = gc + 1;
         n_past[b] = n_past[b] + 1;
     }
+}
+
+// Clear up to 4 device regions in one launch (replaces several
+// hipMemsetAsync nodes ahead of split-K GEMMs: each memset costs a
+// ~5 us dispatch at decode batch sizes).
+__global__ __launch_bounds__(256) void k_zero4(
+    float* __restrict__ p0, int n0, float* __restrict__ p1, int n1,
+    float* __restrict__ p2, int n2, float* __restrict__ p3, int n3) {
+    const int tid = blockIdx.x * 256 + threadIdx.x;
+    const int stride = gridDim.x * 256;
+    const float4 z = {0.f, 0.f, 0.f, 0.f};
+    for (int i = tid; i < n0 >> 2; i += stride)
+        reinterpret_cast<float4*>(p0)[i] = z;
+    for (int i = tid; i < n1 >> 2; i += stride)
+        reinterpret_cast<float4*>(p1)[i] = z;
+    for (int i = tid; i < n2 >> 2; i += stride)
+        reinterpret_cast<float4*>(p2)[i] = z;
+    for (int i = tid; i < n3 >> 2; i += stride)
+        reinterpret_cast<float4*>(p3)[i] = z;
+}
+
+void launch_zero4(float* p0, int64_t n0, float* p1, int64_t n1, float* p2,
+                  int64_t n2, float* p3, int64_t n3, hipStream_t stream) {
+    const int64_t total = (n0 + n1 + n2 + n3) >> 2;
+    int blocks = (int)((total + 255) / 256);
+    if (blocks > 2048) blocks = 2048;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k_zero4, dim3(blocks), dim3(256), 0, stream,
+                       p0, (int)n0, p1, (int)n1, p2, (int)n2, p3, (int)n3);
 }
 
 // --------------------------------------------------------- launch stubs
