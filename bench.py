@@ -149,7 +149,7 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no numbers
-            "dtype": "q4_k_m+f32-activations",
+            "dtype": args.scheme + "+f32-activations",
             "data": "synthetic",
             "config": {
                 "model": args.model,
