@@ -341,28 +341,33 @@ __global__ __launch_bounds__(256) void k_gemv(
             for (int b = 0; b < B; b++) {
                 const float4* xb4 = reinterpret_cast<const float4*>(
                     x_lds + (size_t)b * KP + xpad(k0));
-                float sdot = 0.f;
+                // 4 independent partials: a single serial accumulator is a
+                // 32-deep dependent v_fma chain (~4 cyc each) — PMC showed
+                // 34% of cycles in SQ_WAIT_INST_ANY issue stalls.
+                float p0 = 0.f, p1 = 0.f, p2 = 0.f, p3 = 0.f;
                 if constexpr (W == DT::DQ4K) {
                     #pragma unroll
                     for (int t4 = 0; t4 < 4; t4++) {
                         const float4 xl = xb4[t4];
                         const float4 xh = xb4[8 + t4];
-                        sdot += w[t4 * 4 + 0] * xl.x + w[t4 * 4 + 1] * xl.y
-                           + w[t4 * 4 + 2] * xl.z + w[t4 * 4 + 3] * xl.w;
-                        sdot += w[16 + t4 * 4 + 0] * xh.x
-                           + w[16 + t4 * 4 + 1] * xh.y
-                           + w[16 + t4 * 4 + 2] * xh.z
-                           + w[16 + t4 * 4 + 3] * xh.w;
+                        p0 += w[t4 * 4 + 0] * xl.x + w[t4 * 4 + 1] * xl.y;
+                        p1 += w[t4 * 4 + 2] * xl.z + w[t4 * 4 + 3] * xl.w;
+                        p2 += w[16 + t4 * 4 + 0] * xh.x
+                            + w[16 + t4 * 4 + 1] * xh.y;
+                        p3 += w[16 + t4 * 4 + 2] * xh.z
+                            + w[16 + t4 * 4 + 3] * xh.w;
                     }
                 } else {
                     #pragma unroll
                     for (int t4 = 0; t4 < WPC / 4; t4++) {
                         const float4 xv = xb4[t4];
-                        sdot += w[t4 * 4 + 0] * xv.x + w[t4 * 4 + 1] * xv.y
-                           + w[t4 * 4 + 2] * xv.z + w[t4 * 4 + 3] * xv.w;
+                        p0 += w[t4 * 4 + 0] * xv.x;
+                        p1 += w[t4 * 4 + 1] * xv.y;
+                        p2 += w[t4 * 4 + 2] * xv.z;
+                        p3 += w[t4 * 4 + 3] * xv.w;
                     }
                 }
-                acc[i][b] += sdot;
+                acc[i][b] += (p0 + p1) + (p2 + p3);
             }
             ChunkRaw<W> t = cur[i];
             cur[i] = nxt[i];
