@@ -25,8 +25,21 @@ def _softmax(x: np.ndarray, axis: int = -1) -> np.ndarray:
     return e / e.sum(axis=axis, keepdims=True)
 
 
+def _act_q8(x: np.ndarray) -> np.ndarray:
+    """Per-32 symmetric int8 activation quantization — exactly the engine's
+    k_gemv_q8 staging (xd = amax/127, rint round-half-even)."""
+    b = x.reshape(-1, 32).astype(np.float32)
+    amax = np.abs(b).max(axis=1, keepdims=True)
+    rinv = np.where(amax > 0, 127.0 / np.where(amax == 0, 1, amax), 0.0)
+    q = np.rint(b * rinv)
+    xd = amax / 127.0
+    return (q * xd).reshape(x.shape).astype(np.float32)
+
+
 class RefLlama:
-    def __init__(self, gguf_path: str, kv_bf16: bool = True):
+    def __init__(self, gguf_path: str, kv_bf16: bool = True,
+                 act_q8: bool = False):
+        self.act_q8 = act_q8
         self.r = GGUFReader(gguf_path)
         md = self.r.metadata
         self.n_layers = md["llama.block_count"]
@@ -48,6 +61,16 @@ class RefLlama:
         if name not in self._cache:
             self._cache[name] = self.r.tensor_f32(name)
         return self._cache[name]
+
+    def _is_quant(self, name: str) -> bool:
+        return int(self.r.tensors[name].ggml_type) in (8, 12, 14)
+
+    def mm(self, name: str, x: np.ndarray) -> np.ndarray:
+        """W[name] @ x with the engine's activation quantization when the
+        weight is a quantized dtype and act_q8 is on."""
+        if self.act_q8 and self._is_quant(name):
+            x = _act_q8(x)
+        return self.w(name) @ x
 
     def reset(self):
         self.kv = [[] for _ in range(self.n_layers)]
@@ -77,9 +100,9 @@ class RefLlama:
         for li in range(self.n_layers):
             p = f"blk.{li}."
             xn = self._rms(x, self.w(p + "attn_norm.weight"))
-            q = self.w(p + "attn_q.weight") @ xn
-            k = self.w(p + "attn_k.weight") @ xn
-            v = self.w(p + "attn_v.weight") @ xn
+            q = self.mm(p + "attn_q.weight", xn)
+            k = self.mm(p + "attn_k.weight", xn)
+            v = self.mm(p + "attn_v.weight", xn)
             pos = len(self.kv[li])
             q = self._rope(q.reshape(nh, hd), pos)
             k = self._rope(k.reshape(nkv, hd), pos)
@@ -95,16 +118,16 @@ class RefLlama:
                 scores = ks[:, kvh] @ q[head] / np.sqrt(hd)
                 w = _softmax(scores)
                 attn[head] = w @ vs[:, kvh]
-            x = x + self.w(p + "attn_output.weight") @ attn.reshape(-1)
+            x = x + self.mm(p + "attn_output.weight", attn.reshape(-1))
             xn = self._rms(x, self.w(p + "ffn_norm.weight"))
-            g = self.w(p + "ffn_gate.weight") @ xn
-            u = self.w(p + "ffn_up.weight") @ xn
+            g = self.mm(p + "ffn_gate.weight", xn)
+            u = self.mm(p + "ffn_up.weight", xn)
             act = (g / (1.0 + np.exp(-g))) * u
-            x = x + self.w(p + "ffn_down.weight") @ act
+            x = x + self.mm(p + "ffn_down.weight", act)
         xn = self._rms(x, self.w("output_norm.weight"))
         head_name = ("output.weight" if "output.weight" in self.r.tensors
                      else "token_embd.weight")
-        return self.w(head_name) @ xn
+        return self.mm(head_name, xn)
 
     def generate(self, prompt: list[int], n_new: int) -> list[int]:
         """Greedy generation; returns n_new tokens (incl. first post-prompt)."""

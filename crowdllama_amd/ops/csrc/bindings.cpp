@@ -22,6 +22,9 @@ void launch_gemm_test(const void*, const void*, const float*, float*, int,
                       int, int, int, size_t, size_t);
 double bench_gemm(int, int, int, int, int);
 double bench_gemv_g(int, int, int, int, int);
+void launch_gemv_q8_test(const void*, const void*, const float*, float*, int,
+                         int, int, int, size_t, size_t);
+double bench_gemv_q8(int, int, int, int, int);
 std::vector<uint8_t> slice_cols_test(int32_t, const uint8_t*, int64_t,
                                      int64_t, int64_t, int64_t);
 }
@@ -91,6 +94,7 @@ PYBIND11_MODULE(_core, m) {
         .def_readwrite("prefill_chunk", &EngineConfig::prefill_chunk)
         .def_readwrite("tp_rank", &EngineConfig::tp_rank)
         .def_readwrite("tp_size", &EngineConfig::tp_size)
+        .def_readwrite("act_q8", &EngineConfig::act_q8)
         .def_property("nccl_id",
             [](EngineConfig& c) { return py::bytes(c.nccl_id); },
             [](EngineConfig& c, py::bytes b) { c.nccl_id = std::string(b); });
@@ -164,6 +168,18 @@ PYBIND11_MODULE(_core, m) {
 
     m.def("bench_gemv_g", [](int dtype, int N, int K, int B, int iters) {
         return bench_gemv_g(dtype, N, K, B, iters);
+    });
+    m.def("bench_gemv_q8", [](int dtype, int N, int K, int B, int iters) {
+        return bench_gemv_q8(dtype, N, K, B, iters);
+    });
+    m.def("test_gemv_q8", [](py::array_t<uint8_t> qs, py::array_t<uint8_t> hdr,
+                             py::array_t<float, py::array::c_style> x,
+                             int dtype, int N, int K) {
+        const int B = (int)x.shape(0);
+        py::array_t<float> y({B, N});
+        launch_gemv_q8_test(qs.data(), hdr.data(), x.data(), y.mutable_data(),
+                            dtype, N, K, B, qs.nbytes(), hdr.nbytes());
+        return y;
     });
     m.def("bench_gemv", [](int dtype, int N, int K, int B, int pre, int iters) {
         // random weight bytes (content irrelevant for timing)

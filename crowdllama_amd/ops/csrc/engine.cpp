@@ -24,6 +24,9 @@ namespace cla {
 void launch_gemv(const WTensor&, int pre, const float* xin, const float* gw,
                  const float* res, float* y, int B, int ldy, float eps,
                  hipStream_t);
+void launch_gemv_q8(const WTensor&, int pre, const float* xin,
+                    const float* gw, const float* res, float* y, int B,
+                    int ldy, float eps, hipStream_t);
 void launch_embed(const WTensor&, const int32_t* ids, float* x, int B,
                   hipStream_t);
 void launch_attn_decode(const float* qkv, const float* inv_freq,
@@ -519,6 +522,10 @@ void Engine::reset() {
     HIP_CHECK(hipDeviceSynchronize());
 }
 
+static bool is_quant_dt(DT t) {
+    return t == DT::DQ4K || t == DT::DQ6K || t == DT::DQ8;
+}
+
 // Launch a projection through the GEMM path: pre-zero C when split-K
 // accumulation is in play (see launch_gemm), then one launch per part.
 static void gemm_proj(const Proj& p, const float* X, const float* res,
@@ -532,6 +539,16 @@ static void gemm_proj(const Proj& p, const float* X, const float* res,
     for (auto& pt : p.parts)
         launch_gemm(pt.w, X, res ? res + pt.row_off : nullptr,
                     C + pt.row_off, M, (int)p.n_total, s);
+}
+
+// GEMV dispatch honoring the act_q8 config for quantized weights.
+static void gemv_pick(bool act_q8, const WTensor& w, int pre,
+                      const float* xin, const float* gw, const float* res,
+                      float* y, int B, int ldy, float eps, hipStream_t s) {
+    if (act_q8 && is_quant_dt(w.dtype))
+        launch_gemv_q8(w, pre, xin, gw, res, y, B, ldy, eps, s);
+    else
+        launch_gemv(w, pre, xin, gw, res, y, B, ldy, eps, s);
 }
 
 void Engine::step(hipStream_t s) {
@@ -554,7 +571,7 @@ void Engine::step(hipStream_t s) {
         li++;
         if (gemv_path) {
             for (auto& pt : L.qkv.parts)
-                launch_gemv(pt.w, PRE_RMS, x_, L.attn_norm, nullptr,
+                gemv_pick(cfg_.act_q8, pt.w, PRE_RMS, x_, L.attn_norm, nullptr,
                             qkv_ + pt.row_off, B, (int)L.qkv.n_total, eps, s);
         } else {
             // 3-buffer rotation: in --(+attn)--> mid --(+ffn)--> out, with
@@ -575,27 +592,27 @@ void Engine::step(hipStream_t s) {
                            page_stride_, scale, s);
         if (gemv_path && !tp) {
             for (auto& pt : L.o.parts)
-                launch_gemv(pt.w, PRE_NONE, attn_out_, nullptr, x_ + pt.row_off,
+                gemv_pick(cfg_.act_q8, pt.w, PRE_NONE, attn_out_, nullptr, x_ + pt.row_off,
                             x_ + pt.row_off, B, (int)L.o.n_total, eps, s);
             for (auto& pt : L.gate_up.parts)
-                launch_gemv(pt.w, PRE_RMS, x_, L.ffn_norm, nullptr,
+                gemv_pick(cfg_.act_q8, pt.w, PRE_RMS, x_, L.ffn_norm, nullptr,
                             gu_ + pt.row_off, B, (int)L.gate_up.n_total, eps, s);
             for (auto& pt : L.down.parts)
-                launch_gemv(pt.w, PRE_SILU, gu_, nullptr, x_ + pt.row_off,
+                gemv_pick(cfg_.act_q8, pt.w, PRE_SILU, gu_, nullptr, x_ + pt.row_off,
                             x_ + pt.row_off, B, (int)L.down.n_total, eps, s);
         } else if (gemv_path) {
             // TP: local partial -> all-reduce; rank 0 folds the residual so
             // the summed result is residual + sum(partials) on every rank.
             for (auto& pt : L.o.parts)
-                launch_gemv(pt.w, PRE_NONE, attn_out_, nullptr,
+                gemv_pick(cfg_.act_q8, pt.w, PRE_NONE, attn_out_, nullptr,
                             r0 ? x_ + pt.row_off : nullptr, tmp_h_ + pt.row_off,
                             B, (int)L.o.n_total, eps, s);
             allreduce(tmp_h_, x2_, (size_t)B * meta_.hidden);
             for (auto& pt : L.gate_up.parts)
-                launch_gemv(pt.w, PRE_RMS, x2_, L.ffn_norm, nullptr,
+                gemv_pick(cfg_.act_q8, pt.w, PRE_RMS, x2_, L.ffn_norm, nullptr,
                             gu_ + pt.row_off, B, (int)L.gate_up.n_total, eps, s);
             for (auto& pt : L.down.parts)
-                launch_gemv(pt.w, PRE_SILU, gu_, nullptr,
+                gemv_pick(cfg_.act_q8, pt.w, PRE_SILU, gu_, nullptr,
                             r0 ? x2_ + pt.row_off : nullptr, tmp_h_ + pt.row_off,
                             B, (int)L.down.n_total, eps, s);
             allreduce(tmp_h_, x_, (size_t)B * meta_.hidden);
@@ -625,7 +642,7 @@ void Engine::step(hipStream_t s) {
     float* xfinal = x_;
     if (!gemv_path && (meta_.layers & 1)) xfinal = x2_;
     if (gemv_path) {
-        launch_gemv(head_, PRE_RMS, x_, out_norm_, nullptr, logits_ + voff, B,
+        gemv_pick(cfg_.act_q8, head_, PRE_RMS, x_, out_norm_, nullptr, logits_ + voff, B,
                     meta_.vocab, eps, s);
     } else {
         launch_rmsnorm_rows(xfinal, out_norm_, xn_, B, meta_.hidden, eps, s);
@@ -723,7 +740,7 @@ void Engine::prefill_slot(int slot, const std::vector<int32_t>& ids) {
     // logits of the LAST prompt row -> slot's logits; sample + advance state
     const int last = ((len - 1) % cfg_.prefill_chunk);
     const int64_t voff = (int64_t)cfg_.tp_rank * meta_.vocab_l;
-    launch_gemv(head_, PRE_RMS, xp_ + (size_t)last * meta_.hidden, out_norm_,
+    gemv_pick(cfg_.act_q8, head_, PRE_RMS, xp_ + (size_t)last * meta_.hidden, out_norm_,
                 nullptr, logits_ + (size_t)slot * meta_.vocab + voff, 1,
                 meta_.vocab, eps, stream_);
     if (cfg_.tp_size > 1)

@@ -28,6 +28,10 @@ struct EngineConfig {
     // reference has no collectives at all — SURVEY.md §2.3)
     int tp_rank = 0;
     int tp_size = 1;
+    // int8-quantized activations (per-32 symmetric) for quantized-weight
+    // GEMVs: 4 weights per v_dot4 instruction; semantics replicated by
+    // ref_numpy(act_q8=True)
+    bool act_q8 = true;
     std::string nccl_id;       // ncclUniqueId bytes (rank 0 creates)
 };
 

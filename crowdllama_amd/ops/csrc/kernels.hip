@@ -388,6 +388,212 @@ __global__ __launch_bounds__(256) void k_gemv(
     }
 }
 
+// Q8-activation GEMV for quantized weights: activations are quantized to
+// int8 per 32-block during staging (symmetric, xd = amax/127, rint), and
+// the dot product runs on v_dot4_i32_i8 — 4 weights per instruction and a
+// quarter of the LDS read traffic vs the f32 path. PMC showed the f32
+// path saturating the VALU issue port (~30% active x 6 waves/SIMD).
+// Semantics are exactly y = W . (rint(x/xd)*xd), replicated by the numpy
+// reference (ref_numpy act_q8) and the kernel tests.
+__device__ __forceinline__ int xpad8(int k) {  // byte index pad per 256B
+    return k + ((k >> 8) << 4);
+}
+__device__ __forceinline__ constexpr int xpad8_size(int k) {
+    return k + (k >> 4);
+}
+
+template <DT W, int P>
+__global__ __launch_bounds__(256) void k_gemv_q8(
+    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
+    const float* __restrict__ xin, const float* __restrict__ gw,
+    const float* __restrict__ res, float* __restrict__ y,
+    int N, int K, int B, int ldy, float eps) {
+    static_assert(W == DT::DQ4K || W == DT::DQ6K || W == DT::DQ8,
+                  "q8 path covers quantized weights only");
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    // per b: x8 [xpad8_size(K)] int8, xd [K/32] f32, s16f [K/16] f32
+    const int X8B = (xpad8_size(K) + 15) & ~15;
+    const int NB32 = K / 32, NB16 = K / 16;
+    int8_t* x8 = reinterpret_cast<int8_t*>(smem);               // [B][X8B]
+    float* xd = reinterpret_cast<float*>(smem + (size_t)B * X8B);  // [B][NB32]
+    float* s16f = xd + (size_t)B * NB32;                        // [B][NB16]
+    float* red = s16f + (size_t)B * NB16;                       // [8]
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6, lane = tid & 63;
+    const int r = (int)blockIdx.x * 4 + wave;
+    const int rc = r < N ? r : N - 1;
+    const int64_t qs_rb = dqs_row_bytes(W, K);
+    const int64_t hdr_rb = dhdr_row_bytes(W, K);
+    const int n_chunks = (int)(qs_rb / 16);
+
+    // issue first weight chunks before staging (as in the f32 path)
+    ChunkRaw<W> cur, nxt;
+    const int c0 = lane < n_chunks ? lane : 0;
+    const int c1 = (lane + 64 < n_chunks) ? lane + 64 : c0;
+    load_chunk<W>(qs + (int64_t)rc * qs_rb, hdr + (int64_t)rc * hdr_rb, c0,
+                  &cur);
+    load_chunk<W>(qs + (int64_t)rc * qs_rb, hdr + (int64_t)rc * hdr_rb, c1,
+                  &nxt);
+
+    // ---- stage + quantize activations ----
+    for (int b = 0; b < B; b++) {
+        float inv = 1.f;
+        if constexpr (P == PRE_RMS) {
+            const float4* x4 = reinterpret_cast<const float4*>(
+                xin + (size_t)b * K);
+            float ss = 0.f;
+            for (int k = tid; k < K >> 2; k += 256) {
+                const float4 v = x4[k];
+                ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+            }
+            const float ws = wave_reduce_sum(ss);
+            if (lane == 0) red[wave] = ws;
+            __syncthreads();
+            inv = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
+        }
+        // one 32-block per thread per round
+        for (int blk = tid; blk < NB32; blk += 256) {
+            float v[32];
+            if constexpr (P == PRE_SILU) {
+                const float4* g4 = reinterpret_cast<const float4*>(
+                    xin + (size_t)b * 2 * K + blk * 32);
+                const float4* u4 = reinterpret_cast<const float4*>(
+                    xin + (size_t)b * 2 * K + K + blk * 32);
+                #pragma unroll
+                for (int j = 0; j < 8; j++) {
+                    const float4 g = g4[j], u = u4[j];
+                    v[j * 4 + 0] = (g.x / (1.f + __expf(-g.x))) * u.x;
+                    v[j * 4 + 1] = (g.y / (1.f + __expf(-g.y))) * u.y;
+                    v[j * 4 + 2] = (g.z / (1.f + __expf(-g.z))) * u.z;
+                    v[j * 4 + 3] = (g.w / (1.f + __expf(-g.w))) * u.w;
+                }
+            } else {
+                const float4* x4 = reinterpret_cast<const float4*>(
+                    xin + (size_t)b * K + blk * 32);
+                #pragma unroll
+                for (int j = 0; j < 8; j++) {
+                    const float4 g = x4[j];
+                    v[j * 4 + 0] = g.x; v[j * 4 + 1] = g.y;
+                    v[j * 4 + 2] = g.z; v[j * 4 + 3] = g.w;
+                }
+                if constexpr (P == PRE_RMS) {
+                    const float4* gw4 = reinterpret_cast<const float4*>(
+                        gw + blk * 32);
+                    #pragma unroll
+                    for (int j = 0; j < 8; j++) {
+                        const float4 g = gw4[j];
+                        v[j * 4 + 0] *= inv * g.x; v[j * 4 + 1] *= inv * g.y;
+                        v[j * 4 + 2] *= inv * g.z; v[j * 4 + 3] *= inv * g.w;
+                    }
+                }
+            }
+            float amax = 0.f;
+            #pragma unroll
+            for (int j = 0; j < 32; j++) amax = fmaxf(amax, fabsf(v[j]));
+            const float scale = amax / 127.f;
+            const float rinv = amax > 0.f ? 127.f / amax : 0.f;
+            int s0 = 0, s1 = 0;
+            uint32_t packed[8];
+            #pragma unroll
+            for (int j = 0; j < 8; j++) {
+                int q0 = (int)rintf(v[j * 4 + 0] * rinv);
+                int q1 = (int)rintf(v[j * 4 + 1] * rinv);
+                int q2 = (int)rintf(v[j * 4 + 2] * rinv);
+                int q3 = (int)rintf(v[j * 4 + 3] * rinv);
+                if (j < 4) s0 += q0 + q1 + q2 + q3;
+                else s1 += q0 + q1 + q2 + q3;
+                packed[j] = (uint32_t)(q0 & 0xFF) | ((uint32_t)(q1 & 0xFF) << 8)
+                          | ((uint32_t)(q2 & 0xFF) << 16)
+                          | ((uint32_t)(q3 & 0xFF) << 24);
+            }
+            int8_t* dst = x8 + (size_t)b * X8B + xpad8(blk * 32);
+            #pragma unroll
+            for (int j = 0; j < 8; j += 4)
+                *reinterpret_cast<uint4*>(dst + j * 4) =
+                    *reinterpret_cast<const uint4*>(&packed[j]);
+            xd[(size_t)b * NB32 + blk] = scale;
+            s16f[(size_t)b * NB16 + blk * 2] = scale * (float)s0;
+            s16f[(size_t)b * NB16 + blk * 2 + 1] = scale * (float)s1;
+        }
+        if constexpr (P == PRE_RMS) __syncthreads();  // red[] reuse next b
+    }
+    __syncthreads();
+
+    // ---- per-wave row sweep (dot4 on int8) ----
+    float acc[2] = {0.f, 0.f};
+    for (int c = lane; c < n_chunks; c += 64) {
+        uint32_t dwv[4] = {cur.qv.x, cur.qv.y, cur.qv.z, cur.qv.w};
+        if constexpr (W == DT::DQ4K) {
+            const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
+            const float d = f16_bits_to_f32(cur.hd.x & 0xFFFF);
+            const float dmin = f16_bits_to_f32(cur.hd.x >> 16);
+            const float dl = d * (float)(cur.hd.y & 0xFF);
+            const float ml = dmin * (float)((cur.hd.y >> 8) & 0xFF);
+            const float dh = d * (float)((cur.hd.y >> 16) & 0xFF);
+            const float mh = dmin * (float)(cur.hd.y >> 24);
+            const int k0 = sb * 256 + q * 64 + h * 16;
+            const int cn = (c + 128 < n_chunks) ? c + 128 : c0;
+            load_chunk<W>(qs + (int64_t)rc * qs_rb,
+                          hdr + (int64_t)rc * hdr_rb, cn, &cur);
+            #pragma unroll 2
+            for (int b = 0; b < B; b++) {
+                const int8_t* x8b = x8 + (size_t)b * X8B;
+                const uint4 xl = *reinterpret_cast<const uint4*>(
+                    x8b + xpad8(k0));
+                const uint4 xh = *reinterpret_cast<const uint4*>(
+                    x8b + xpad8(k0 + 32));
+                int s1l = 0, s1h = 0;
+                s1l = __builtin_amdgcn_sdot4((int)(dwv[0] & 0x0F0F0F0Fu), (int)xl.x, s1l, false);
+                s1l = __builtin_amdgcn_sdot4((int)(dwv[1] & 0x0F0F0F0Fu), (int)xl.y, s1l, false);
+                s1l = __builtin_amdgcn_sdot4((int)(dwv[2] & 0x0F0F0F0Fu), (int)xl.z, s1l, false);
+                s1l = __builtin_amdgcn_sdot4((int)(dwv[3] & 0x0F0F0F0Fu), (int)xl.w, s1l, false);
+                s1h = __builtin_amdgcn_sdot4((int)((dwv[0] >> 4) & 0x0F0F0F0Fu), (int)xh.x, s1h, false);
+                s1h = __builtin_amdgcn_sdot4((int)((dwv[1] >> 4) & 0x0F0F0F0Fu), (int)xh.y, s1h, false);
+                s1h = __builtin_amdgcn_sdot4((int)((dwv[2] >> 4) & 0x0F0F0F0Fu), (int)xh.z, s1h, false);
+                s1h = __builtin_amdgcn_sdot4((int)((dwv[3] >> 4) & 0x0F0F0F0Fu), (int)xh.w, s1h, false);
+                const float xdl = xd[(size_t)b * NB32 + (k0 >> 5)];
+                const float xdh = xd[(size_t)b * NB32 + ((k0 + 32) >> 5)];
+                acc[b] += dl * xdl * (float)s1l - ml * s16f[(size_t)b * NB16 + (k0 >> 4)]
+                        + dh * xdh * (float)s1h - mh * s16f[(size_t)b * NB16 + ((k0 + 32) >> 4)];
+            }
+        } else {
+            // DQ6K / DQ8: 16 signed weights per chunk, one scale
+            const float sc = cur.d0;
+            const int k0 = c * 16;
+            const int cn = (c + 128 < n_chunks) ? c + 128 : c0;
+            load_chunk<W>(qs + (int64_t)rc * qs_rb,
+                          hdr + (int64_t)rc * hdr_rb, cn, &cur);
+            #pragma unroll 2
+            for (int b = 0; b < B; b++) {
+                const int8_t* x8b = x8 + (size_t)b * X8B;
+                const uint4 xv = *reinterpret_cast<const uint4*>(
+                    x8b + xpad8(k0));
+                int s1 = 0;
+                s1 = __builtin_amdgcn_sdot4((int)dwv[0], (int)xv.x, s1, false);
+                s1 = __builtin_amdgcn_sdot4((int)dwv[1], (int)xv.y, s1, false);
+                s1 = __builtin_amdgcn_sdot4((int)dwv[2], (int)xv.z, s1, false);
+                s1 = __builtin_amdgcn_sdot4((int)dwv[3], (int)xv.w, s1, false);
+                acc[b] += sc * xd[(size_t)b * NB32 + (k0 >> 5)] * (float)s1;
+            }
+        }
+        // rotate: cur was refilled with chunk c+128; nxt holds c+64
+        ChunkRaw<W> t = cur;
+        cur = nxt;
+        nxt = t;
+    }
+    if (r < N) {
+        #pragma unroll 2
+        for (int b = 0; b < B; b++) {
+            float v = wave_reduce_sum(acc[b]);
+            if (lane == 0) {
+                const int64_t idx = (int64_t)b * ldy + r;
+                y[idx] = v + (res ? res[idx] : 0.f);
+            }
+        }
+    }
+}
+
 // Global-x GEMV: no LDS staging — the activation vector (<=57 KB) is
 // L1-resident per CU after first touch, so reading it directly unlocks
 // full occupancy (no 16 KB+ LDS budget per workgroup) and removes the
@@ -864,6 +1070,33 @@ void launch_zero4(float* p0, int64_t n0, float* p1, int64_t n1, float* p2,
         case DT::F32:  FN(DT::F32);  break;                           \
         default: throw std::runtime_error("bad dtype");               \
     }
+
+void launch_gemv_q8(const WTensor& w, int pre, const float* xin,
+                    const float* gw, const float* res, float* y, int B,
+                    int ldy, float eps, hipStream_t stream) {
+    const int N = (int)w.n, K = (int)w.k;
+    if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
+    const int X8B = ((K + (K >> 4)) + 15) & ~15;
+    const size_t lds = (size_t)B * X8B + (size_t)B * (K / 32) * 4
+                     + (size_t)B * (K / 16) * 4 + 8 * 4;
+    dim3 grid((N + 3) / 4), block(256);
+    #define GEMVQ_CASE(WT)                                                     \
+        do {                                                                   \
+            auto kern = (pre == PRE_RMS) ? k_gemv_q8<WT, PRE_RMS>              \
+                       : (pre == PRE_SILU) ? k_gemv_q8<WT, PRE_SILU>           \
+                       : k_gemv_q8<WT, PRE_NONE>;                              \
+            hipLaunchKernelGGL(kern, grid, block, lds, stream,                 \
+                (const uint8_t*)w.qs, (const uint8_t*)w.hdr, xin, gw, res, y,  \
+                N, K, B, ldy, eps);                                            \
+        } while (0)
+    switch (w.dtype) {
+        case DT::DQ4K: GEMVQ_CASE(DT::DQ4K); break;
+        case DT::DQ6K: GEMVQ_CASE(DT::DQ6K); break;
+        case DT::DQ8:  GEMVQ_CASE(DT::DQ8);  break;
+        default: throw std::runtime_error("q8 gemv: quant dtypes only");
+    }
+    #undef GEMVQ_CASE
+}
 
 void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
                  const float* res, float* y, int B, int ldy, float eps,

@@ -12,6 +12,9 @@ void launch_gemm(const WTensor&, const float* X, const float* res, float* C,
                  int M, int ldc, hipStream_t);
 void launch_gemv_g(const WTensor&, const float* xin, const float* res,
                    float* y, int B, int ldy, hipStream_t);
+void launch_gemv_q8(const WTensor&, int pre, const float* xin,
+                    const float* gw, const float* res, float* y, int B,
+                    int ldy, float eps, hipStream_t);
 
 void launch_gemv_test(const void* qs, const void* hdr, const float* x,
                       const float* gw, float* y, int dtype, int N, int K,
@@ -134,6 +137,67 @@ double bench_gemm(int dtype, int M, int N, int K, int iters) {
     HIP_CHECK(hipEventRecord(e0, nullptr));
     for (int i = 0; i < iters; i++)
         launch_gemm(w, (const float*)d_x, nullptr, (float*)d_y, M, N, nullptr);
+    HIP_CHECK(hipEventRecord(e1, nullptr));
+    HIP_CHECK(hipDeviceSynchronize());
+    float ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+    hipEventDestroy(e0); hipEventDestroy(e1);
+    hipFree(d_qs); if (d_hdr) hipFree(d_hdr);
+    hipFree(d_x); hipFree(d_y);
+    return ms / iters;
+}
+
+void launch_gemv_q8_test(const void* qs, const void* hdr, const float* x,
+                         float* y, int dtype, int N, int K, int B,
+                         size_t qs_bytes, size_t hdr_bytes) {
+    const DT dt = static_cast<DT>(dtype);
+    void *d_qs, *d_hdr = nullptr, *d_x, *d_y;
+    HIP_CHECK(hipMalloc(&d_qs, qs_bytes));
+    HIP_CHECK(hipMemcpy(d_qs, qs, qs_bytes, hipMemcpyHostToDevice));
+    if (hdr_bytes) {
+        HIP_CHECK(hipMalloc(&d_hdr, hdr_bytes));
+        HIP_CHECK(hipMemcpy(d_hdr, hdr, hdr_bytes, hipMemcpyHostToDevice));
+    }
+    HIP_CHECK(hipMalloc(&d_x, (size_t)B * K * 4));
+    HIP_CHECK(hipMemcpy(d_x, x, (size_t)B * K * 4, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMalloc(&d_y, (size_t)B * N * 4));
+    WTensor w;
+    w.dtype = dt; w.n = N; w.k = K; w.qs = d_qs; w.hdr = d_hdr;
+    launch_gemv_q8(w, 0, (const float*)d_x, nullptr, nullptr, (float*)d_y,
+                   B, N, 1e-5f, nullptr);
+    HIP_CHECK(hipDeviceSynchronize());
+    HIP_CHECK(hipMemcpy(y, d_y, (size_t)B * N * 4, hipMemcpyDeviceToHost));
+    hipFree(d_qs); if (d_hdr) hipFree(d_hdr);
+    hipFree(d_x); hipFree(d_y);
+}
+
+double bench_gemv_q8(int dtype, int N, int K, int B, int iters) {
+    const DT dt = static_cast<DT>(dtype);
+    const size_t qs_bytes = (size_t)N * dqs_row_bytes(dt, K);
+    const size_t hdr_bytes = (size_t)N * dhdr_row_bytes(dt, K);
+    void *d_qs, *d_hdr = nullptr, *d_x, *d_y;
+    HIP_CHECK(hipMalloc(&d_qs, qs_bytes));
+    HIP_CHECK(hipMemset(d_qs, 1, qs_bytes));
+    if (hdr_bytes) {
+        HIP_CHECK(hipMalloc(&d_hdr, hdr_bytes));
+        HIP_CHECK(hipMemset(d_hdr, 1, hdr_bytes));
+    }
+    HIP_CHECK(hipMalloc(&d_x, (size_t)B * K * 4));
+    HIP_CHECK(hipMemset(d_x, 0, (size_t)B * K * 4));
+    HIP_CHECK(hipMalloc(&d_y, (size_t)B * N * 4));
+    WTensor w;
+    w.dtype = dt; w.n = N; w.k = K; w.qs = d_qs; w.hdr = d_hdr;
+    for (int i = 0; i < 3; i++)
+        launch_gemv_q8(w, 1, (const float*)d_x, (const float*)d_x, nullptr,
+                       (float*)d_y, B, N, 1e-5f, nullptr);
+    HIP_CHECK(hipDeviceSynchronize());
+    hipEvent_t e0, e1;
+    HIP_CHECK(hipEventCreate(&e0));
+    HIP_CHECK(hipEventCreate(&e1));
+    HIP_CHECK(hipEventRecord(e0, nullptr));
+    for (int i = 0; i < iters; i++)
+        launch_gemv_q8(w, 1, (const float*)d_x, (const float*)d_x, nullptr,
+                       (float*)d_y, B, N, 1e-5f, nullptr);
     HIP_CHECK(hipEventRecord(e1, nullptr));
     HIP_CHECK(hipDeviceSynchronize());
     float ms = 0;

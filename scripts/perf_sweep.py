@@ -15,6 +15,12 @@ for name, dt, N, K, pre in [
     ms = core.bench_gemv(dt, N, K, 1, pre, 50)
     bpr = K//256*128 + K//256*16 if dt == DQ4K else (K + K//256*32 if dt == DQ6K else K+K//16)
     print(f"  {name:30s} {ms*1000:8.1f}us {N*bpr/ms/1e6:7.0f} GB/s")
+print("== gemv q8 (int8-activation dot4 path) ==")
+for name, dt, N, K in [("qkv q4k", DQ4K, 6144, 4096), ("gateup q4k", DQ4K, 28672, 4096),
+                       ("down q4k", DQ4K, 4096, 14336), ("head q6k", DQ6K, 128256, 4096)]:
+    ms = core.bench_gemv_q8(dt, N, K, 1, 50)
+    bpr = K//256*128 + K//256*32 if dt == DQ4K else K + K//256*32
+    print(f"  {name:30s} {ms*1000:8.1f}us {N*bpr/ms/1e6:7.0f} GB/s")
 print("== gemm (M=512 prefill, M=16 batched decode) ==")
 for name, dt, M, N, K in [
     ("gateup q4k M512", DQ4K, 512, 28672, 4096),

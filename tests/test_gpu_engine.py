@@ -31,6 +31,7 @@ def test_engine_logits_vs_ref(core, tiny_gguf):
     cfg = core.EngineConfig()
     cfg.batch = 1
     cfg.max_seq = 128
+    cfg.act_q8 = False  # strict comparison against the f32 reference
     eng = core.Engine(tiny_gguf, cfg)
     prompt = [3, 17, 99, 250, 7]
     ids = np.array([prompt], dtype=np.int32)
@@ -54,6 +55,7 @@ def test_engine_greedy_matches_ref(core, tiny_gguf):
     cfg = core.EngineConfig()
     cfg.batch = 1
     cfg.max_seq = 128
+    cfg.act_q8 = False
     eng = core.Engine(tiny_gguf, cfg)
     prompt = [5, 10, 200]
     n_new = 8
@@ -108,6 +110,7 @@ def test_engine_long_context_page_crossing(core, tiny_gguf):
     cfg = core.EngineConfig()
     cfg.batch = 1
     cfg.max_seq = 256
+    cfg.act_q8 = False
     eng = core.Engine(tiny_gguf, cfg)
     rng = np.random.default_rng(3)
     prompt = rng.integers(3, 500, size=150).tolist()  # crosses 2+ pages
@@ -166,3 +169,27 @@ def test_engine_batched_gemm_path(core, tiny_gguf):
     e4.decode(5)
     for slot in range(4):
         assert list(e4.gen_tokens(slot)) == ref_tokens, slot
+
+
+def test_engine_act_q8_decode(core, tiny_gguf):
+    """Default serving path: int8-quantized activations for quantized
+    weights (v_dot4). Logits stay close to the f32 reference and the
+    engine's argmax lands in the reference's top-5."""
+    from crowdllama_amd.engine.ref_numpy import RefLlama
+    import numpy as np
+    cfg = core.EngineConfig()
+    cfg.batch = 1
+    cfg.max_seq = 128
+    assert cfg.act_q8 is True  # default on
+    eng = core.Engine(tiny_gguf, cfg)
+    prompt = [3, 17, 99, 250, 7]
+    eng.prefill(np.asarray([prompt], dtype=np.int32))
+    got = np.asarray(eng.logits(0))
+    ref = RefLlama(tiny_gguf, act_q8=True)
+    want = None
+    for t in prompt:
+        want = ref.step(t)
+    rel = np.abs(got - want).max() / (np.abs(want).max() + 1e-9)
+    assert rel < 2e-2, rel
+    top5 = np.argsort(want)[-5:]
+    assert int(np.argmax(got)) in top5

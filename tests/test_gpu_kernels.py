@@ -239,3 +239,25 @@ def test_tp_slice_cols(core):
                                256, 512)
     want = w[:, 256:512].astype(np.float32).tobytes()
     assert got == want
+
+
+@pytest.mark.parametrize("name", ["q4k", "q6k", "q8"])
+def test_gemv_q8_path(core, name):
+    """int8-activation GEMV vs exact numpy emulation of its semantics."""
+    dt, quant, dequant, repack_fn = CASES[name]
+    rng = np.random.default_rng(21)
+    N, K = 64, 512
+    w = rng.standard_normal((N, K)).astype(np.float32) * 0.1
+    raw = quant(w)
+    wref = dequant(raw, K).reshape(N, K)
+    x = rng.standard_normal((1, K)).astype(np.float32)
+    qs, hdr = repack_fn(raw.reshape(N, -1), N, K)
+    y = core.test_gemv_q8(np.ascontiguousarray(qs), np.ascontiguousarray(hdr),
+                          x, dt, N, K)
+    # emulate: per-32 symmetric int8 activations
+    b = x.reshape(-1, 32)
+    amax = np.abs(b).max(axis=1, keepdims=True)
+    rinv = np.where(amax > 0, 127.0 / np.where(amax == 0, 1, amax), 0.0)
+    xq = (np.rint(b * rinv) * (amax / 127.0)).reshape(1, K)
+    yref = xq @ wref.T
+    np.testing.assert_allclose(y, yref, rtol=2e-4, atol=2e-4)
