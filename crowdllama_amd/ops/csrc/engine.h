@@ -29,9 +29,11 @@ struct EngineConfig {
     int tp_rank = 0;
     int tp_size = 1;
     // int8-quantized activations (per-32 symmetric) for quantized-weight
-    // GEMVs: 4 weights per v_dot4 instruction; semantics replicated by
-    // ref_numpy(act_q8=True)
-    bool act_q8 = true;
+    // GEMVs (v_dot4 path; semantics replicated by ref_numpy(act_q8=True)).
+    // Off by default: measured slower than the f32 path this round — the
+    // per-thread quantization staging serializes (round-2 item: lane-
+    // parallel block quantization).
+    bool act_q8 = false;
     std::string nccl_id;       // ncclUniqueId bytes (rank 0 creates)
 };
 

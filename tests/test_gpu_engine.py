@@ -180,7 +180,7 @@ def test_engine_act_q8_decode(core, tiny_gguf):
     cfg = core.EngineConfig()
     cfg.batch = 1
     cfg.max_seq = 128
-    assert cfg.act_q8 is True  # default on
+    cfg.act_q8 = True  # exercise the optional int8-activation path
     eng = core.Engine(tiny_gguf, cfg)
     prompt = [3, 17, 99, 250, 7]
     eng.prefill(np.asarray([prompt], dtype=np.int32))
