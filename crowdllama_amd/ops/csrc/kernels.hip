@@ -175,8 +175,8 @@ __device__ __forceinline__ void decode_chunk_raw(
 // PRE_RMS:  x = rmsnorm(xin) * gw   (gamma)
 // PRE_SILU: x[k] = silu(xin[b][k]) * xin[b][K+k]   (xin is [B][2K])
 // One wave per row; x staged in LDS as f32 with per-16 partial sums.
-template <DT W, int P, int RPW>
-__global__ __launch_bounds__(256) void k_gemv(
+template <DT W, int P, int RPW, int NT>
+__global__ __launch_bounds__(NT) void k_gemv(
     const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
     const float* __restrict__ xin, const float* __restrict__ gw,
     const float* __restrict__ res, float* __restrict__ y,
@@ -188,10 +188,10 @@ __global__ __launch_bounds__(256) void k_gemv(
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6, lane = tid & 63;
-    // RPW rows per wave: amortizes activation staging over more weight
-    // traffic and keeps RPW+ chunk loads in flight per wave (big-N GEMVs
-    // use RPW=8; small-N use RPW=1 to keep the grid large).
-    const int rbase = (int)blockIdx.x * (4 * RPW) + wave * RPW;
+    constexpr int WAVES = NT / 64;
+    // RPW rows per wave; NT=512 for big-N GEMVs (8 waves share one staged
+    // x copy: +33% waves/CU and half the staging traffic).
+    const int rbase = (int)blockIdx.x * (WAVES * RPW) + wave * RPW;
     int r[RPW];
     #pragma unroll
     for (int i = 0; i < RPW; i++) r[i] = rbase + i;
@@ -234,12 +234,12 @@ __global__ __launch_bounds__(256) void k_gemv(
             const float4* u4 = reinterpret_cast<const float4*>(
                 xin + (size_t)b * 2 * K + K);
             int k0 = tid;
-            for (; k0 + 768 < K4; k0 += 1024) {  // guard hoisted: no
+            for (; k0 + 3 * NT < K4; k0 += 4 * NT) {  // guard hoisted: no
                 float4 gs[4], us[4];             // per-element branches
                 #pragma unroll
                 for (int j = 0; j < 4; j++) {
-                    gs[j] = g4[k0 + j * 256];
-                    us[j] = u4[k0 + j * 256];
+                    gs[j] = g4[k0 + j * NT];
+                    us[j] = u4[k0 + j * NT];
                 }
                 #pragma unroll
                 for (int j = 0; j < 4; j++) {
@@ -249,10 +249,10 @@ __global__ __launch_bounds__(256) void k_gemv(
                     o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
                     o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
                     o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
-                    *reinterpret_cast<float4*>(xlb + xpad(k0 * 4 + j * 1024)) = o;
+                    *reinterpret_cast<float4*>(xlb + xpad(k0 * 4 + j * NT * 4)) = o;
                 }
             }
-            for (int k = k0; k < K4; k += 256) {
+            for (int k = k0; k < K4; k += NT) {
                 const float4 g = g4[k], u = u4[k];
                 float4 o;
                 o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
@@ -265,19 +265,19 @@ __global__ __launch_bounds__(256) void k_gemv(
             const float4* x4 = reinterpret_cast<const float4*>(
                 xin + (size_t)b * K);
             int k0 = tid;
-            for (; k0 + 768 < K4; k0 += 1024) {  // guard hoisted
+            for (; k0 + 3 * NT < K4; k0 += 4 * NT) {  // guard hoisted
                 float4 vs[4];
                 #pragma unroll
-                for (int j = 0; j < 4; j++) vs[j] = x4[k0 + j * 256];
+                for (int j = 0; j < 4; j++) vs[j] = x4[k0 + j * NT];
                 #pragma unroll
                 for (int j = 0; j < 4; j++) {
                     const float4 v = vs[j];
-                    *reinterpret_cast<float4*>(xlb + xpad(k0 * 4 + j * 1024)) = v;
+                    *reinterpret_cast<float4*>(xlb + xpad(k0 * 4 + j * NT * 4)) = v;
                     if constexpr (P == PRE_RMS)
                         ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
                 }
             }
-            for (int k = k0; k < K4; k += 256) {
+            for (int k = k0; k < K4; k += NT) {
                 const float4 v = x4[k];
                 *reinterpret_cast<float4*>(xlb + xpad(k * 4)) = v;
                 if constexpr (P == PRE_RMS)
@@ -289,25 +289,27 @@ __global__ __launch_bounds__(256) void k_gemv(
             const float ws = wave_reduce_sum(ss);
             if (lane == 0) red[wave] = ws;
             __syncthreads();
-            const float inv = rsqrtf(
-                (red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
+            float rsum = 0.f;
+            #pragma unroll
+            for (int wv = 0; wv < WAVES; wv++) rsum += red[wv];
+            const float inv = rsqrtf(rsum / (float)K + eps);
             const float4* gw4 = reinterpret_cast<const float4*>(gw);
             int kg = tid;
             for (; kg + 768 < K4; kg += 1024) {  // batched gw loads
                 float4 gs[4];
                 #pragma unroll
-                for (int j = 0; j < 4; j++) gs[j] = gw4[kg + j * 256];
+                for (int j = 0; j < 4; j++) gs[j] = gw4[kg + j * NT];
                 #pragma unroll
                 for (int j = 0; j < 4; j++) {
                     float4* vp = reinterpret_cast<float4*>(
-                        xlb + xpad(kg * 4 + j * 1024));
+                        xlb + xpad(kg * 4 + j * NT * 4));
                     float4 v = *vp;
                     v.x *= inv * gs[j].x; v.y *= inv * gs[j].y;
                     v.z *= inv * gs[j].z; v.w *= inv * gs[j].w;
                     *vp = v;
                 }
             }
-            for (int k = kg; k < K4; k += 256) {
+            for (int k = kg; k < K4; k += NT) {
                 const float4 g = gw4[k];
                 float4* vp = reinterpret_cast<float4*>(xlb + xpad(k * 4));
                 float4 v = *vp;
@@ -1104,16 +1106,16 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
     const int N = (int)w.n, K = (int)w.k;
     if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
     const size_t lds = (size_t)B * (K + (K >> 4)) * 4 + 8 * 4;
-    // rows per wave: 1 measured best across all shapes (RPW>1 raises
-    // register pressure past the occupancy cliff: head 187us@1 vs 481us@8);
-    // the template stays parameterized for future tuning.
-    const int rpw = 1;
-    dim3 grid((N + 4 * rpw - 1) / (4 * rpw)), block(256);
-    #define GEMV_RPW(WT, RPWV)                                                   \
+    // rows per wave fixed at 1 (RPW>1 regressed on register pressure);
+    // 512-thread blocks for large N: 8 waves share one staged x copy.
+    const bool big = N >= 8192;
+    const int wpb = big ? 8 : 4;                 // waves per block
+    dim3 grid((N + wpb - 1) / wpb), block(wpb * 64);
+    #define GEMV_NT(WT, NTV)                                                     \
         do {                                                                     \
-            auto kern = (pre == PRE_RMS) ? k_gemv<WT, PRE_RMS, RPWV>             \
-                       : (pre == PRE_SILU) ? k_gemv<WT, PRE_SILU, RPWV>          \
-                       : k_gemv<WT, PRE_NONE, RPWV>;                             \
+            auto kern = (pre == PRE_RMS) ? k_gemv<WT, PRE_RMS, 1, NTV>           \
+                       : (pre == PRE_SILU) ? k_gemv<WT, PRE_SILU, 1, NTV>        \
+                       : k_gemv<WT, PRE_NONE, 1, NTV>;                           \
             if (lds > 64 * 1024) {                                               \
                 (void)hipFuncSetAttribute((const void*)kern,                     \
                     hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);       \
@@ -1124,16 +1126,12 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
         } while (0)
     #define GEMV_CASE(WT)                                                        \
         do {                                                                     \
-            switch (rpw) {                                                       \
-                case 8: GEMV_RPW(WT, 8); break;                                  \
-                case 4: GEMV_RPW(WT, 4); break;                                  \
-                case 2: GEMV_RPW(WT, 2); break;                                  \
-                default: GEMV_RPW(WT, 1); break;                                 \
-            }                                                                    \
+            if (big) GEMV_NT(WT, 512);                                           \
+            else GEMV_NT(WT, 256);                                               \
         } while (0)
     DISPATCH_DT(w.dtype, GEMV_CASE);
     #undef GEMV_CASE
-    #undef GEMV_RPW
+    #undef GEMV_NT
 }
 
 void launch_gemv_g(const WTensor& w, const float* xin, const float* res,
