@@ -1106,9 +1106,10 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
     const int N = (int)w.n, K = (int)w.k;
     if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
     const size_t lds = (size_t)B * (K + (K >> 4)) * 4 + 8 * 4;
-    // rows per wave fixed at 1 (RPW>1 regressed on register pressure);
-    // 512-thread blocks for large N: 8 waves share one staged x copy.
-    const bool big = N >= 8192;
+    // rows per wave fixed at 1, 256-thread blocks: the measured optimum
+    // (A/B-rejected: RPW>1, 3-buffer rotation, pair-unroll, 512-thread
+    // blocks, global-x/no-LDS, int8-dot default — see profiles/).
+    const bool big = false;
     const int wpb = big ? 8 : 4;                 // waves per block
     dim3 grid((N + wpb - 1) / wpb), block(wpb * 64);
     #define GEMV_NT(WT, NTV)                                                     \
