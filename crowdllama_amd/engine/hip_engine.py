@@ -46,7 +46,8 @@ class HipEngine(EngineBase):
 
     # ------------------------------------------------------------ generate
 
-    def _generate_sync(self, prompt: str, max_new_tokens: int) -> GenerateResult:
+    def _generate_sync(self, prompt: str, max_new_tokens: int,
+                       temperature: float = 0.0) -> GenerateResult:
         import numpy as np
         with self._lock:
             t0 = time.monotonic_ns()
@@ -56,9 +57,27 @@ class HipEngine(EngineBase):
             self.eng.reset()
             self.eng.prefill(np.asarray([ids], dtype=np.int32))
             n_new = max(1, max_new_tokens)
-            if n_new > 1:
-                self.eng.decode(n_new - 1)
-            out = list(self.eng.gen_tokens(0))
+            if temperature > 0.0:
+                # host sampling path: re-sample each step from the logits
+                from .sampling import sample
+                rng = np.random.default_rng()
+                out = []
+                tok_id = sample(self.eng.logits(0), temperature=temperature,
+                                rng=rng)
+                out.append(tok_id)
+                self.eng.set_cur_token(0, tok_id)
+                for _ in range(n_new - 1):
+                    if tok_id == self.tok.eos_id:
+                        break
+                    self.eng.decode(1)
+                    tok_id = sample(self.eng.logits(0),
+                                    temperature=temperature, rng=rng)
+                    out.append(tok_id)
+                    self.eng.set_cur_token(0, tok_id)
+            else:
+                if n_new > 1:
+                    self.eng.decode(n_new - 1)
+                out = list(self.eng.gen_tokens(0))
             if self.tok.eos_id in out:
                 out = out[: out.index(self.tok.eos_id)]
                 reason = "stop"
@@ -76,7 +95,8 @@ class HipEngine(EngineBase):
         self._active += 1
         try:
             return await loop.run_in_executor(
-                self._pool, self._generate_sync, prompt, max_new_tokens)
+                self._pool, self._generate_sync, prompt, max_new_tokens,
+                temperature)
         finally:
             self._active -= 1
 
