@@ -75,3 +75,19 @@ def test_chat_through_mesh_hip_engine(has_gpu, tmp_path):
             await worker.stop()
             await dht.stop()
     asyncio.run(go())
+
+
+def test_hip_engine_temperature_sampling(has_gpu):
+    """Host-sampling path: temperature>0 generates via logits + re-seeding
+    cur_token each step."""
+    import asyncio
+    from crowdllama_amd.engine.hip_engine import HipEngine
+    from crowdllama_amd.models import synth_path
+    path = synth_path("testllama", scheme="q4_k_m", mode="exact", seed=7)
+    eng = HipEngine("testllama", path, max_seq=128)
+    r_greedy = asyncio.run(eng.generate("abc", max_new_tokens=8))
+    r_sampled = asyncio.run(eng.generate("abc", max_new_tokens=8,
+                                         temperature=1.0))
+    assert r_greedy.tokens_generated >= 1
+    assert r_sampled.tokens_generated >= 1
+    assert isinstance(r_sampled.text, str)
