@@ -36,9 +36,6 @@ void launch_attn_decode(const float* qkv, const float* inv_freq,
                         float* attn_out, int B, int NH, int NKV,
                         int D, int S, int page_size, int max_pages,
                         int64_t page_stride, float scale, hipStream_t);
-void launch_attn_combine(const float* part_o, const float* part_ml,
-                         float* attn_out, int B, int NH, int S, int D,
-                         hipStream_t);
 void launch_argmax(const float* logits, float* pval, int32_t* pidx,
                    int32_t* cur_ids, int32_t* n_past, int32_t* gen_tokens,
                    int32_t* gen_count, int B, int V, int gen_cap, hipStream_t);
