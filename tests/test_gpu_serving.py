@@ -94,3 +94,65 @@ def test_hip_engine_temperature_sampling(has_gpu):
     assert r_sampled.done_reason in ("stop", "length")
     assert r_greedy.tokens_generated >= 0
     assert isinstance(r_sampled.text, str)
+
+
+def test_worker_serves_two_models_one_gpu(has_gpu, tmp_path):
+    """Mixed-fleet building block: one worker process serving two models
+    resident on one 288 GB GPU, model-aware routing picks each."""
+    import asyncio
+    import time
+    from crowdllama_amd.config import Config
+    from crowdllama_amd.engine.hip_engine import HipEngine
+    from crowdllama_amd.mesh.dhtnode import DHTServer
+    from crowdllama_amd.mesh.gateway import Gateway
+    from crowdllama_amd.mesh.peer import Peer
+    from crowdllama_amd.models import synth_path
+    from crowdllama_amd.models.presets import ModelConfig
+    from crowdllama_amd.models.synth import write_synthetic_gguf
+
+    p1 = synth_path("testllama", scheme="q4_k_m", mode="exact", seed=7)
+    cfg2 = ModelConfig("testllama2", vocab_size=512, hidden_size=256,
+                       n_layers=2, n_heads=2, n_kv_heads=1, ffn_hidden=512,
+                       rope_theta=10000.0, max_seq_len=512)
+    p2 = str(tmp_path / "m2.gguf")
+    write_synthetic_gguf(p2, cfg2, scheme="q8_0", mode="exact", seed=8)
+    engines = {"testllama": HipEngine("testllama", p1, max_seq=128),
+               "testllama2": HipEngine("testllama2", p2, max_seq=128)}
+
+    async def go():
+        import aiohttp
+        def mk(c):
+            return Config(test_mode=True, listen_host="127.0.0.1",
+                          key_path=str(tmp_path / f"{c}.key"))
+        dht = DHTServer(mk("dht"), "CLADHT")
+        port = await dht.start("127.0.0.1", 0)
+        wcfg = mk("worker")
+        wcfg.bootstrap_peers = [f"127.0.0.1:{port}"]
+        worker = Peer(wcfg, worker_mode=True, engines=engines)
+        await worker.start()
+        ccfg = mk("consumer")
+        ccfg.bootstrap_peers = [f"127.0.0.1:{port}"]
+        consumer = Peer(ccfg, worker_mode=False)
+        await consumer.start()
+        gw = Gateway(consumer, ccfg)
+        gport = await gw.start(port=0)
+        try:
+            deadline = time.time() + 20
+            while gw.find_best_worker("testllama2") is None:
+                assert time.time() < deadline
+                await asyncio.sleep(0.1)
+            async with aiohttp.ClientSession() as s:
+                for model in ("testllama", "testllama2"):
+                    async with s.post(f"http://127.0.0.1:{gport}/api/chat",
+                                      json={"model": model,
+                                            "messages": [{"role": "user",
+                                                          "content": "x"}]}) as r:
+                        body = await r.json()
+                        assert r.status == 200, body
+                        assert body["model"] == model
+        finally:
+            await gw.stop()
+            await consumer.stop()
+            await worker.stop()
+            await dht.stop()
+    asyncio.run(go())
