@@ -52,6 +52,7 @@ def build(verbose: bool = True, force: bool = False) -> str:
         obj = os.path.join(OPS_DIR, "build", src.rsplit(".", 1)[0] + ".o")
         objs.append(obj)
         cmd = [hipcc, f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
+               *(["-DCLA_PERF_PROBE_NOATOMIC"] if os.environ.get("CLA_PROBE") else []),
                "-I", CSRC, "-I", py_inc, "-I", pybind11.get_include(),
                "-c", os.path.join(CSRC, src), "-o", obj]
         if src.endswith(".hip"):
