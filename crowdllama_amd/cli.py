@@ -76,14 +76,20 @@ async def _run_start(args) -> None:
         if args.engine == "mock":
             engines = {m: MockEngine(m) for m in (models or ["tinyllama"])}
         else:
-            from .engine.hip_engine import HipEngine
             from .models import synth_path
             for i, m in enumerate(models or ["llama3-8b"]):
                 path = (args.model_path or synth_path(m, scheme=args.scheme))
-                log.info("loading %s from %s on device %d", m, path,
-                         args.device)
-                engines[m] = HipEngine(m, path, device=args.device,
-                                       max_seq=cfg.max_seq)
+                log.info("loading %s from %s on device %d (batch=%d)", m,
+                         path, args.device, args.batch)
+                if args.batch > 1:
+                    from .engine.batching import BatchingHipEngine
+                    engines[m] = BatchingHipEngine(
+                        m, path, device=args.device, batch=args.batch,
+                        max_seq=cfg.max_seq)
+                else:
+                    from .engine.hip_engine import HipEngine
+                    engines[m] = HipEngine(m, path, device=args.device,
+                                           max_seq=cfg.max_seq)
 
     peer = Peer(cfg, worker_mode=args.worker_mode, engines=engines)
     await peer.start()
@@ -180,6 +186,8 @@ def main(argv: list[str] | None = None) -> int:
     st.add_argument("--scheme", default="q4_k_m")
     st.add_argument("--engine", default="hip", choices=["hip", "mock"])
     st.add_argument("--device", type=int, default=0)
+    st.add_argument("--batch", type=int, default=1,
+                    help="decode slots for continuous batching (worker mode)")
 
     dh = sub.add_parser("dht", help="start a bootstrap/rendezvous node")
     _add_common(dh)

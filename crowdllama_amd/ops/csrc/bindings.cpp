@@ -103,6 +103,7 @@ PYBIND11_MODULE(_core, m) {
         .def(py::init<const std::string&, const EngineConfig&>(),
              py::arg("gguf_path"), py::arg("config"))
         .def("reset", &Engine::reset)
+        .def("reset_slot", &Engine::reset_slot)
         .def("prefill",
              [](Engine& e, py::array_t<int32_t, py::array::c_style> ids) {
                  if (ids.ndim() != 2)

@@ -548,6 +548,13 @@ static void gemv_pick(bool act_q8, const WTensor& w, int pre,
         launch_gemv(w, pre, xin, gw, res, y, B, ldy, eps, s);
 }
 
+void Engine::reset_slot(int slot) {
+    const int32_t zero = 0;
+    HIP_CHECK(hipMemcpy(n_past_ + slot, &zero, 4, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(gen_count_ + slot, &zero, 4, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(cur_ids_ + slot, &zero, 4, hipMemcpyHostToDevice));
+}
+
 void Engine::step(hipStream_t s) {
     const int B = cfg_.batch;
     const int NH = meta_.heads_l, NKV = meta_.kv_heads_l, D = meta_.head_dim;

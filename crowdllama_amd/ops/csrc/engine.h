@@ -73,6 +73,8 @@ public:
 
     // Reset all slots (empty KV).
     void reset();
+    // Reset one slot (serving: reclaim a finished/idle slot).
+    void reset_slot(int slot);
     // Feed prompt tokens (same length for all slots in this call) through
     // the GEMM prefill path; afterwards each slot's first generated token is
     // in gen_tokens[slot][0]. ids is [batch][len] row-major.

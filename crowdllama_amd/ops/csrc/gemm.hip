@@ -186,12 +186,8 @@ __global__ __launch_bounds__(256) void k_gemm(
     const int srow0 = tid >> 1, sh0 = tid & 1;   // W staging coords
     const int64_t gn_s = (int64_t)n0 + srow0;
     const int64_t gn_c = gn_s < N ? gn_s : N - 1;  // clamped (branchless)
-    WRaw wr, wr_next, wr_next2;
+    WRaw wr, wr_next;
     load_w_raw<W>(qs, hdr, gn_c, K, kb_lo, sh0, &wr);
-    {
-        const int kb1 = (kb_lo + BK < kb_hi) ? kb_lo + BK : kb_lo;
-        load_w_raw<W>(qs, hdr, gn_c, K, kb1, sh0, &wr_next);
-    }
     // BM=32: the X tile is 8 floats/thread — prefetch it alongside W so the
     // whole staging phase runs from registers (M<=32 decode batches have
     // too few workgroups to hide latency with occupancy alone).
@@ -285,8 +281,7 @@ __global__ __launch_bounds__(256) void k_gemm(
         // HBM latency hides under the matrix work.
         {
             const int kbn = (kb + BK < kb_hi) ? kb + BK : kb;
-            const int kbn2 = (kb + 2 * BK < kb_hi) ? kb + 2 * BK : kb;
-            load_w_raw<W>(qs, hdr, gn_c, K, kbn2, sh0, &wr_next2);
+            load_w_raw<W>(qs, hdr, gn_c, K, kbn, sh0, &wr_next);
             if (BM_ == 32) {
                 const float4* src = reinterpret_cast<const float4*>(
                     X + (size_t)xgm * K + kbn + xseg * 8);
@@ -319,7 +314,6 @@ __global__ __launch_bounds__(256) void k_gemm(
         }
         __syncthreads();
         wr = wr_next;
-        wr_next = wr_next2;
         if (BM_ == 32) { xr0 = xr0n; xr1 = xr1n; }
     }
 

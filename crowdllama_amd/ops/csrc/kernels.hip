@@ -722,8 +722,10 @@ __global__ __launch_bounds__(256) void k_attn_decode(
     int64_t page_stride, float scale) {
     constexpr int DPL = D / 16;   // dims per lane (bf16: 2*DPL bytes)
     const int s = blockIdx.x, kvh = blockIdx.y, b = blockIdx.z;
-    const int pos = n_past[b];
-    const int len = pos + 1;
+    const int cap = max_pages * page_size;
+    int pos = n_past[b];
+    if (pos >= cap) pos = cap - 1;   // serving safety: never index past the
+    const int len = pos + 1;         // page table (idle slots, see engine)
     const int tid = threadIdx.x;
     const int wave = tid >> 6, lane = tid & 63;
     const int quarter = lane >> 4, qlane = lane & 15;

@@ -156,3 +156,25 @@ def test_worker_serves_two_models_one_gpu(has_gpu, tmp_path):
             await worker.stop()
             await dht.stop()
     asyncio.run(go())
+
+
+def test_continuous_batching_engine(has_gpu):
+    """Concurrent requests share decode slots; results match the
+    single-request engine (greedy, independent KV)."""
+    import asyncio
+    from crowdllama_amd.engine.batching import BatchingHipEngine
+    from crowdllama_amd.engine.hip_engine import HipEngine
+    from crowdllama_amd.models import synth_path
+    path = synth_path("testllama", scheme="q4_k_m", mode="exact", seed=7)
+    single = HipEngine("testllama", path, max_seq=128)
+    batched = BatchingHipEngine("testllama", path, batch=2, max_seq=128)
+
+    async def go():
+        prompts = ["abc", "hello there", "xyz", "abc"]
+        want = [await single.generate(p, max_new_tokens=6) for p in prompts]
+        got = await asyncio.gather(
+            *[batched.generate(p, max_new_tokens=6) for p in prompts])
+        for w, g in zip(want, got):
+            assert g.text == w.text, (g.text, w.text)
+        await batched.close()
+    asyncio.run(go())
