@@ -38,8 +38,11 @@ def main() -> None:
                          "all ranks via RCCL tensor parallelism "
                          "(BASELINE config 4; e.g. --mode tp --model "
                          "llama3-70b --scheme bf16)")
-    ap.add_argument("--batch", type=int, default=1,
-                    help="decode slots per GPU")
+    ap.add_argument("--batch", type=int, default=16,
+                    help="decode slots per GPU (the serving stack batches "
+                         "concurrent requests into these slots — "
+                         "crowdllama_amd/engine/batching.py; use 1 for the "
+                         "single-stream latency point)")
     ap.add_argument("--prompt-len", type=int, default=128)
     ap.add_argument("--max-seq", type=int, default=4096)
     args = ap.parse_args()
@@ -145,7 +148,7 @@ def main() -> None:
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
-            "p50_latency_ms": ms_per_step,  # B=1: step latency == token latency
+            "p50_latency_ms": ms_per_step,  # per-token latency at this batch
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no numbers
