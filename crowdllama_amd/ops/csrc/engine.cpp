@@ -41,6 +41,9 @@ void launch_argmax(const float* logits, float* pval, int32_t* pidx,
                    int32_t* gen_count, int B, int V, int gen_cap, hipStream_t);
 void launch_gemm(const WTensor&, const float* X, const float* res, float* C,
                  int M, int ldc, hipStream_t);
+void launch_gemm_ex(const WTensor&, const float* X, const float* X2, int ldx,
+                    bool xsilu, const float* res, float* C, int M, int ldc,
+                    hipStream_t);
 bool gemm_uses_splitk(int N, int K, int M);
 void launch_zero4(float* p0, int64_t n0, float* p1, int64_t n1, float* p2,
                   int64_t n2, float* p3, int64_t n3, hipStream_t);
@@ -527,15 +530,17 @@ static bool is_quant_dt(DT t) {
 // accumulation is in play (see launch_gemm), then one launch per part.
 static void gemm_proj(const Proj& p, const float* X, const float* res,
                       float* C, int M, hipStream_t s,
-                      bool pre_zeroed = false) {
+                      bool pre_zeroed = false, const float* X2 = nullptr,
+                      int ldx = -1, bool xsilu = false) {
     bool zero = false;
     for (auto& pt : p.parts)
         zero |= gemm_uses_splitk((int)pt.w.n, (int)pt.w.k, M);
     if (zero && !pre_zeroed)
         HIP_CHECK(hipMemsetAsync(C, 0, (size_t)M * p.n_total * 4, s));
     for (auto& pt : p.parts)
-        launch_gemm(pt.w, X, res ? res + pt.row_off : nullptr,
-                    C + pt.row_off, M, (int)p.n_total, s);
+        launch_gemm_ex(pt.w, X, X2, ldx < 0 ? (int)pt.w.k : ldx, xsilu,
+                       res ? res + pt.row_off : nullptr,
+                       C + pt.row_off, M, (int)p.n_total, s);
 }
 
 // GEMV dispatch honoring the act_q8 config for quantized weights.
@@ -631,12 +636,14 @@ void Engine::step(hipStream_t s) {
             }
             launch_rmsnorm_rows(x3_, L.ffn_norm, xn_, B, meta_.hidden, eps, s);
             gemm_proj(L.gate_up, xn_, nullptr, gu_, B, s, /*pre_zeroed=*/true);
-            launch_silu_rows(gu_, act_, B, meta_.ffn_l, s);
+            // silu fused into the down GEMM's X staging (gate | up halves)
             if (tp) {
-                gemm_proj(L.down, act_, r0 ? x3_ : nullptr, tmp_h_, B, s);
+                gemm_proj(L.down, gu_, r0 ? x3_ : nullptr, tmp_h_, B, s,
+                          false, gu_ + meta_.ffn_l, 2 * meta_.ffn_l, true);
                 allreduce(tmp_h_, lout, (size_t)B * meta_.hidden);
             } else {
-                gemm_proj(L.down, act_, x3_, lout, B, s, /*pre_zeroed=*/true);
+                gemm_proj(L.down, gu_, x3_, lout, B, s, /*pre_zeroed=*/true,
+                          gu_ + meta_.ffn_l, 2 * meta_.ffn_l, true);
             }
         }
     }
@@ -712,13 +719,14 @@ void Engine::prefill_chunk_pass(int slot, int pos0, int m) {
         }
         launch_rmsnorm_rows(xp2_, L.ffn_norm, xnp_, m, meta_.hidden, eps, s);
         gemm_proj(L.gate_up, xnp_, nullptr, gup_, m, s);
-        launch_silu_rows(gup_, actp_, m, meta_.ffn_l, s);
         if (tp) {
-            gemm_proj(L.down, actp_, r0 ? xp2_ : nullptr, tmp_hp_, m, s);
+            gemm_proj(L.down, gup_, r0 ? xp2_ : nullptr, tmp_hp_, m, s,
+                      false, gup_ + meta_.ffn_l, 2 * meta_.ffn_l, true);
             NCCL_CHECK(ncclAllReduce(tmp_hp_, xp_, (size_t)m * meta_.hidden,
                                      ncclFloat, ncclSum, (ncclComm_t)comm_, s));
         } else {
-            gemm_proj(L.down, actp_, xp2_, xp_, m, s);
+            gemm_proj(L.down, gup_, xp2_, xp_, m, s, false,
+                      gup_ + meta_.ffn_l, 2 * meta_.ffn_l, true);
         }
     }
 }

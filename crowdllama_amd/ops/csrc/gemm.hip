@@ -153,13 +153,14 @@ __device__ __forceinline__ void stage_w_slice(
 
 }  // namespace
 
-template <DT W, int BM_>
+template <DT W, int BM_, bool XSILU>
 __global__ __launch_bounds__(256) void k_gemm(
     const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
-    const float* __restrict__ X,     // [M][K] f32
+    const float* __restrict__ X,     // [M][ldx] f32 (gate half when XSILU)
+    const float* __restrict__ X2,    // [M][ldx] up half (XSILU only)
     const float* __restrict__ res,   // [M][ldc] or null (C col-offset applied)
     float* __restrict__ C,           // [M][ldc]
-    int M, int N, int K, int ldc, int k_chunk) {
+    int M, int N, int K, int ldc, int ldx, int k_chunk) {
     // k_chunk: this block's K-range is [z*k_chunk, min((z+1)*k_chunk, K));
     // splitk > 1 => partial results accumulated with atomicAdd (C pre-zeroed,
     // residual folded in by the z==0 block).
@@ -191,23 +192,39 @@ __global__ __launch_bounds__(256) void k_gemm(
     // BM=32: the X tile is 8 floats/thread — prefetch it alongside W so the
     // whole staging phase runs from registers (M<=32 decode batches have
     // too few workgroups to hide latency with occupancy alone).
-    float4 xr0, xr1, xr0n, xr1n;
+    float4 xr0, xr1, xr0n, xr1n, ur0, ur1, ur0n, ur1n;
     const int xrow = (BM_ == 32) ? (tid >> 3) : 0;
     const int xseg = (BM_ == 32) ? (tid & 7) : 0;
     const int xgm = m0 + xrow < M ? m0 + xrow : (M > 0 ? M - 1 : 0);
     const bool xvalid0 = (m0 + xrow) < M;
     if (BM_ == 32) {
         const float4* src = reinterpret_cast<const float4*>(
-            X + (size_t)xgm * K + kb_lo + xseg * 8);
+            X + (size_t)xgm * ldx + kb_lo + xseg * 8);
         xr0 = src[0];
         xr1 = src[1];
+        if constexpr (XSILU) {
+            const float4* up = reinterpret_cast<const float4*>(
+                X2 + (size_t)xgm * ldx + kb_lo + xseg * 8);
+            ur0 = up[0];
+            ur1 = up[1];
+        }
     }
 
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
         // ---- stage X tile (f32 -> bf16), 8-elem units ----
         if (BM_ == 32) {
             uint16_t tmp[8];
-            const float4 v[2] = {xr0, xr1};
+            float4 v[2] = {xr0, xr1};
+            if constexpr (XSILU) {
+                const float4 u[2] = {ur0, ur1};
+                #pragma unroll
+                for (int j = 0; j < 2; j++) {
+                    v[j].x = (v[j].x / (1.f + __expf(-v[j].x))) * u[j].x;
+                    v[j].y = (v[j].y / (1.f + __expf(-v[j].y))) * u[j].y;
+                    v[j].z = (v[j].z / (1.f + __expf(-v[j].z))) * u[j].z;
+                    v[j].w = (v[j].w / (1.f + __expf(-v[j].w))) * u[j].w;
+                }
+            }
             #pragma unroll
             for (int j = 0; j < 2; j++) {
                 tmp[j * 4 + 0] = f32_to_bf16b(v[j].x);
@@ -232,10 +249,18 @@ __global__ __launch_bounds__(256) void k_gemm(
                 uint16_t tmp[8];
                 if (gm < M) {
                     const float4* src = reinterpret_cast<const float4*>(
-                        X + (size_t)gm * K + kb + seg * 8);
+                        X + (size_t)gm * ldx + kb + seg * 8);
                     #pragma unroll
                     for (int j = 0; j < 2; j++) {
-                        const float4 v = src[j];
+                        float4 v = src[j];
+                        if constexpr (XSILU) {
+                            const float4 u = reinterpret_cast<const float4*>(
+                                X2 + (size_t)gm * ldx + kb + seg * 8)[j];
+                            v.x = (v.x / (1.f + __expf(-v.x))) * u.x;
+                            v.y = (v.y / (1.f + __expf(-v.y))) * u.y;
+                            v.z = (v.z / (1.f + __expf(-v.z))) * u.z;
+                            v.w = (v.w / (1.f + __expf(-v.w))) * u.w;
+                        }
                         tmp[j * 4 + 0] = f32_to_bf16b(v.x);
                         tmp[j * 4 + 1] = f32_to_bf16b(v.y);
                         tmp[j * 4 + 2] = f32_to_bf16b(v.z);
@@ -284,9 +309,15 @@ __global__ __launch_bounds__(256) void k_gemm(
             load_w_raw<W>(qs, hdr, gn_c, K, kbn, sh0, &wr_next);
             if (BM_ == 32) {
                 const float4* src = reinterpret_cast<const float4*>(
-                    X + (size_t)xgm * K + kbn + xseg * 8);
+                    X + (size_t)xgm * ldx + kbn + xseg * 8);
                 xr0n = src[0];
                 xr1n = src[1];
+                if constexpr (XSILU) {
+                    const float4* up = reinterpret_cast<const float4*>(
+                        X2 + (size_t)xgm * ldx + kbn + xseg * 8);
+                    ur0n = up[0];
+                    ur1n = up[1];
+                }
             }
         }
         // ---- MFMA over the tile ----
@@ -314,7 +345,10 @@ __global__ __launch_bounds__(256) void k_gemm(
         }
         __syncthreads();
         wr = wr_next;
-        if (BM_ == 32) { xr0 = xr0n; xr1 = xr1n; }
+        if (BM_ == 32) {
+            xr0 = xr0n; xr1 = xr1n;
+            if constexpr (XSILU) { ur0 = ur0n; ur1 = ur1n; }
+        }
     }
 
     // ---- epilogue ----
@@ -537,8 +571,9 @@ bool gemm_uses_splitk(int N, int K, int M) {
     return splitk > 1;
 }
 
-void launch_gemm(const WTensor& w, const float* X, const float* res, float* C,
-                 int M, int ldc, hipStream_t stream) {
+void launch_gemm_ex(const WTensor& w, const float* X, const float* X2,
+                    int ldx, bool xsilu, const float* res, float* C, int M,
+                    int ldc, hipStream_t stream) {
     const int N = (int)w.n, K = (int)w.k;
     if (K % BK != 0) throw std::runtime_error("gemm: K must be /64");
     const bool small_m = M <= 32;
@@ -557,19 +592,25 @@ void launch_gemm(const WTensor& w, const float* X, const float* res, float* C,
     dim3 grid(n_tiles, bm_tiles, splitk), block(256);
     // splitk > 1 accumulates with atomicAdd: caller must pre-zero C
     // (gemm_uses_splitk tells it whether that is needed).
+    #define GEMM_ONE(WT, BMV, XSV)                                             \
+        hipLaunchKernelGGL((k_gemm<WT, BMV, XSV>), grid, block, 0, stream,     \
+            (const uint8_t*)w.qs, (const uint8_t*)w.hdr, X, X2, res, C,        \
+            M, N, K, ldc, ldx, k_chunk)
     #define GEMM_CASE(WT)                                                      \
         do {                                                                   \
-            if (small_m)                                                       \
-                hipLaunchKernelGGL((k_gemm<WT, 32>), grid, block, 0, stream,   \
-                    (const uint8_t*)w.qs, (const uint8_t*)w.hdr, X, res, C,    \
-                    M, N, K, ldc, k_chunk);                                    \
-            else                                                               \
-                hipLaunchKernelGGL((k_gemm<WT, 128>), grid, block, 0, stream,  \
-                    (const uint8_t*)w.qs, (const uint8_t*)w.hdr, X, res, C,    \
-                    M, N, K, ldc, k_chunk);                                    \
+            if (small_m && xsilu) GEMM_ONE(WT, 32, true);                      \
+            else if (small_m) GEMM_ONE(WT, 32, false);                         \
+            else if (xsilu) GEMM_ONE(WT, 128, true);                           \
+            else GEMM_ONE(WT, 128, false);                                     \
         } while (0)
     DISPATCH_DT_GEMM(w.dtype, GEMM_CASE);
     #undef GEMM_CASE
+    #undef GEMM_ONE
+}
+
+void launch_gemm(const WTensor& w, const float* X, const float* res, float* C,
+                 int M, int ldc, hipStream_t stream) {
+    launch_gemm_ex(w, X, nullptr, (int)w.k, false, res, C, M, ldc, stream);
 }
 
 void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
