@@ -47,6 +47,10 @@ void launch_gemm_ex(const WTensor&, const float* X, const float* X2, int ldx,
 bool gemm_uses_splitk(int N, int K, int M);
 void launch_zero4(float* p0, int64_t n0, float* p1, int64_t n1, float* p2,
                   int64_t n2, float* p3, int64_t n3, hipStream_t);
+void launch_layer_prep(const float* X, const float* gw, float* xn, int B,
+                       int K, float eps, float* p0, int64_t n0, float* p1,
+                       int64_t n1, float* p2, int64_t n2, float* p3,
+                       int64_t n3, hipStream_t);
 void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
                          int K, float eps, hipStream_t);
 void launch_silu_rows(const float* GU, float* out, int M, int F, hipStream_t);
@@ -588,11 +592,11 @@ void Engine::step(hipStream_t s) {
             // (replaces 3-4 ~5 us hipMemsetAsync dispatches).
             float* lin = ((li - 1) & 1) ? x2_ : x_;  // li already advanced
             float* lout = ((li - 1) & 1) ? x_ : x2_;
-            launch_zero4(qkv_, (int64_t)B * (NH + 2 * NKV) * D,
-                         gu_, (int64_t)B * 2 * meta_.ffn_l,
-                         x3_, (int64_t)B * meta_.hidden,
-                         lout, (int64_t)B * meta_.hidden, s);
-            launch_rmsnorm_rows(lin, L.attn_norm, xn_, B, meta_.hidden, eps, s);
+            launch_layer_prep(lin, L.attn_norm, xn_, B, meta_.hidden, eps,
+                              qkv_, (int64_t)B * (NH + 2 * NKV) * D,
+                              gu_, (int64_t)B * 2 * meta_.ffn_l,
+                              x3_, (int64_t)B * meta_.hidden,
+                              lout, (int64_t)B * meta_.hidden, s);
             gemm_proj(L.qkv, xn_, nullptr, qkv_, B, s, /*pre_zeroed=*/true);
         }
         launch_attn_decode(qkv_, inv_freq_, page_table_, kv_layer, n_past_,

@@ -1052,6 +1052,71 @@ __global__ __launch_bounds__(256) void k_zero4(
         reinterpret_cast<float4*>(p3)[i] = z;
 }
 
+// Layer-start prep: rmsnorm the residual rows AND clear the split-K
+// destinations in one launch (blocks 0..B-1 normalize one row each; the
+// rest zero the regions). Replaces a zero4 + rmsnorm_rows pair.
+__global__ __launch_bounds__(256) void k_layer_prep(
+    const float* __restrict__ X, const float* __restrict__ gw,
+    float* __restrict__ xn, int B, int K, float eps,
+    float* __restrict__ p0, int n0, float* __restrict__ p1, int n1,
+    float* __restrict__ p2, int n2, float* __restrict__ p3, int n3) {
+    const int blk = blockIdx.x;
+    if (blk < B) {
+        // rmsnorm row blk (same structure as k_rmsnorm_rows)
+        const float4* x4 = reinterpret_cast<const float4*>(X + (size_t)blk * K);
+        float4* o4 = reinterpret_cast<float4*>(xn + (size_t)blk * K);
+        const float4* g4 = reinterpret_cast<const float4*>(gw);
+        const int K4 = K >> 2;
+        float ss = 0.f;
+        for (int k = threadIdx.x; k < K4; k += 256) {
+            const float4 v = x4[k];
+            ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+        }
+        __shared__ float red[256];
+        red[threadIdx.x] = ss;
+        __syncthreads();
+        #pragma unroll
+        for (int off = 128; off > 0; off >>= 1) {
+            if ((int)threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+            __syncthreads();
+        }
+        const float inv = rsqrtf(red[0] / (float)K + eps);
+        for (int k = threadIdx.x; k < K4; k += 256) {
+            const float4 v = x4[k];
+            const float4 g = g4[k];
+            float4 o;
+            o.x = v.x * inv * g.x; o.y = v.y * inv * g.y;
+            o.z = v.z * inv * g.z; o.w = v.w * inv * g.w;
+            o4[k] = o;
+        }
+        return;
+    }
+    const int tid = (blk - B) * 256 + threadIdx.x;
+    const int stride = (gridDim.x - B) * 256;
+    const float4 z = {0.f, 0.f, 0.f, 0.f};
+    for (int i = tid; i < n0 >> 2; i += stride)
+        reinterpret_cast<float4*>(p0)[i] = z;
+    for (int i = tid; i < n1 >> 2; i += stride)
+        reinterpret_cast<float4*>(p1)[i] = z;
+    for (int i = tid; i < n2 >> 2; i += stride)
+        reinterpret_cast<float4*>(p2)[i] = z;
+    for (int i = tid; i < n3 >> 2; i += stride)
+        reinterpret_cast<float4*>(p3)[i] = z;
+}
+
+void launch_layer_prep(const float* X, const float* gw, float* xn, int B,
+                       int K, float eps, float* p0, int64_t n0, float* p1,
+                       int64_t n1, float* p2, int64_t n2, float* p3,
+                       int64_t n3, hipStream_t stream) {
+    const int64_t total = (n0 + n1 + n2 + n3) >> 2;
+    int zb = (int)((total + 255) / 256);
+    if (zb > 1024) zb = 1024;
+    if (zb < 1) zb = 1;
+    hipLaunchKernelGGL(k_layer_prep, dim3(B + zb), dim3(256), 0, stream,
+                       X, gw, xn, B, K, eps, p0, (int)n0, p1, (int)n1,
+                       p2, (int)n2, p3, (int)n3);
+}
+
 void launch_zero4(float* p0, int64_t n0, float* p1, int64_t n1, float* p2,
                   int64_t n2, float* p3, int64_t n3, hipStream_t stream) {
     const int64_t total = (n0 + n1 + n2 + n3) >> 2;
