@@ -66,13 +66,16 @@ class Gateway:
 
     @web.middleware
     async def _log_middleware(self, request: web.Request, handler):
+        # request logging with status + duration (reference gateway.go:107-135)
         t0 = time.monotonic()
+        status = 500
         try:
             resp = await handler(request)
+            status = resp.status
             return resp
         finally:
-            self.log.info("%s %s %.1fms", request.method, request.path,
-                          (time.monotonic() - t0) * 1e3)
+            self.log.info("%s %s %d %.1fms", request.method, request.path,
+                          status, (time.monotonic() - t0) * 1e3)
 
     # ------------------------------------------------------------ discovery
 

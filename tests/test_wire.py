@@ -107,3 +107,44 @@ def test_resource_invalid_json():
         Resource.from_json("not json")
     with pytest.raises(ValueError):
         Resource.from_json("[1,2,3]")
+
+
+def test_frame_fragmented_delivery():
+    """Frames reassemble across arbitrary TCP fragmentation."""
+    async def run():
+        reader = asyncio.StreamReader()
+        buf = bytearray()
+
+        class W:
+            def write(self, b):
+                buf.extend(b)
+            async def drain(self):
+                pass
+        payload = bytes(range(256)) * 5
+        await write_frame(W(), payload)
+        # feed one byte at a time
+        task = asyncio.ensure_future(read_frame(reader))
+        for i in range(len(buf)):
+            reader.feed_data(bytes(buf[i:i + 1]))
+            await asyncio.sleep(0)
+        got = await task
+        assert got == payload
+    asyncio.run(run())
+
+
+def test_two_frames_back_to_back():
+    async def run():
+        reader = asyncio.StreamReader()
+        buf = bytearray()
+
+        class W:
+            def write(self, b):
+                buf.extend(b)
+            async def drain(self):
+                pass
+        await write_frame(W(), b"first")
+        await write_frame(W(), b"second")
+        reader.feed_data(bytes(buf))
+        assert await read_frame(reader) == b"first"
+        assert await read_frame(reader) == b"second"
+    asyncio.run(run())
