@@ -9,6 +9,7 @@ from __future__ import annotations
 
 import asyncio
 import logging
+import time
 
 from .resource import Resource
 from .wire import (NAMESPACE, PROTO_METADATA, PROTO_RENDEZVOUS,
@@ -91,6 +92,10 @@ class Discovery:
         self.metadata_timeout = metadata_timeout
         self.metadata_max_age = metadata_max_age
         self.log = log or logging.getLogger("discovery")
+        # monotonic time of the last successful rendezvous round trip;
+        # Peer.is_dht_connected derives liveness from this (the reference
+        # checks the DHT routing table instead, peer.go:513-525)
+        self.last_success = 0.0
 
     async def close(self):
         for c in self.clients:
@@ -107,6 +112,7 @@ class Discovery:
         for c in self.clients:
             try:
                 await c.provide(peer_id, addrs, ns)
+                self.last_success = time.monotonic()
             except Exception as e:
                 self.log.debug("advertise to %s failed: %s", c.addr, e)
 
@@ -117,6 +123,7 @@ class Discovery:
             try:
                 for p in await c.find_providers(ns, limit):
                     seen[p["peer_id"]] = p
+                self.last_success = time.monotonic()
             except Exception as e:
                 self.log.debug("find_providers on %s failed: %s", c.addr, e)
         return list(seen.values())[:limit]

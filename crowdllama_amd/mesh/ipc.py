@@ -1,19 +1,27 @@
 """IPC server for desktop UI bridge (reference parity: pkg/ipc/ipc.go).
 
-Unix-domain socket at $CROWDLLAMA_SOCKET, chmod 0600, speaking
-length-prefixed protobuf BaseMessage only (the reference's JSON/PB sniffing
-dual protocol is deliberately dropped — SURVEY.md §7.4; message semantics
-kept: prompt -> UnifiedAPIHandler -> response)."""
+Unix-domain socket at $CROWDLLAMA_SOCKET, chmod 0600. Each message is
+sniffed by its first 4 bytes (reference ipc.go:187-240): a plausible
+big-endian length (< 10 MB) selects the length-prefixed protobuf
+BaseMessage path; anything else (JSON starts with '{' = 0x7B, which reads
+as an implausibly large length) selects the newline-delimited JSON path
+with message types ping/pong, initialize/initialize_status and
+prompt/response (ipc.go:26-35). Both paths route prompts through the same
+injected UnifiedAPIHandler (ipc.go:278-313,437-477)."""
 
 from __future__ import annotations
 
 import asyncio
+import json
 import os
+import struct
 
 from ..config import Config
 from ..logutil import new_app_logger
 from . import pb
-from .wire import read_frame, write_frame
+from .wire import MAX_FRAME, write_frame
+
+_MAX_JSON = 4096  # reference caps the JSON fallback read at 4 KB
 
 
 class IPCServer:
@@ -42,21 +50,78 @@ class IPCServer:
         if os.path.exists(self.path):
             os.unlink(self.path)
 
+    # ------------------------------------------------------------- PB path
+
+    async def _handle_pb(self, frame: bytes, writer) -> None:
+        msg = pb.BaseMessage.decode(frame)
+        try:
+            resp = await self.handler(msg)
+        except Exception as e:  # noqa: BLE001
+            resp = pb.response_message("", f"Error: {e}",
+                                       done_reason="error")
+        await write_frame(writer, resp.encode())
+
+    # ----------------------------------------------------------- JSON path
+
+    async def _handle_json(self, line: bytes, writer) -> None:
+        try:
+            msg = json.loads(line)
+            mtype = msg.get("type", "")
+        except (json.JSONDecodeError, AttributeError):
+            await self._send_json(writer, {"type": "error",
+                                           "error": "invalid JSON"})
+            return
+        if mtype == "ping":
+            await self._send_json(writer, {"type": "pong"})
+        elif mtype == "initialize":
+            await self._send_json(writer, {"type": "initialize_status",
+                                           "status": "ready"})
+        elif mtype == "prompt":
+            req = pb.request_message(msg.get("model", ""),
+                                     msg.get("content", msg.get("prompt", "")))
+            try:
+                resp = await self.handler(req)
+                gr = resp.generate_response
+                await self._send_json(writer, {
+                    "type": "response",
+                    "model": gr.model if gr else "",
+                    "content": gr.response if gr else "",
+                    "done": gr.done if gr else True,
+                })
+            except Exception as e:  # noqa: BLE001
+                await self._send_json(writer, {"type": "error",
+                                               "error": str(e)})
+        else:
+            await self._send_json(writer, {"type": "error",
+                                           "error": f"unknown type {mtype!r}"})
+
+    @staticmethod
+    async def _send_json(writer, obj: dict) -> None:
+        writer.write(json.dumps(obj).encode("utf-8") + b"\n")
+        await writer.drain()
+
+    # ----------------------------------------------------------- conn loop
+
     async def _on_conn(self, reader: asyncio.StreamReader,
                        writer: asyncio.StreamWriter) -> None:
         try:
             while True:
                 try:
-                    frame = await read_frame(reader)
+                    head = await reader.readexactly(4)
                 except (asyncio.IncompleteReadError, ConnectionError):
                     return
-                msg = pb.BaseMessage.decode(frame)
-                try:
-                    resp = await self.handler(msg)
-                except Exception as e:  # noqa: BLE001
-                    resp = pb.response_message("", f"Error: {e}",
-                                               done_reason="error")
-                await write_frame(writer, resp.encode())
+                (length,) = struct.unpack(">I", head)
+                if length <= MAX_FRAME:
+                    body = await reader.readexactly(length)
+                    await self._handle_pb(body, writer)
+                else:
+                    rest = await reader.readline()
+                    line = head + rest
+                    if len(line) > _MAX_JSON:
+                        await self._send_json(writer, {"type": "error",
+                                                       "error": "too large"})
+                        return
+                    await self._handle_json(line, writer)
         except Exception as e:  # noqa: BLE001
             self.log.debug("ipc conn error: %s", e)
         finally:

@@ -197,4 +197,10 @@ class Peer:
         await write_frame(writer, resp.encode())
 
     def is_dht_connected(self) -> bool:
-        return bool(self.cfg.bootstrap_peers)
+        """True while rendezvous round trips are succeeding (reference
+        IsDHTConnected checks the routing table, peer.go:513-525; here the
+        equivalent liveness signal is a recent successful provide/find)."""
+        if not self.cfg.bootstrap_peers:
+            return False
+        window = 3.0 * max(self.cfg.intervals.advertise, 1.0)
+        return (time.monotonic() - self.discovery.last_success) < window

@@ -63,3 +63,47 @@ def test_ipc_handler_error_stringified(tmp_path):
         finally:
             await srv.stop()
     asyncio.run(go())
+
+
+def test_ipc_json_fallback(tmp_path):
+    """Reference dual protocol (ipc.go:187-240): the same socket accepts
+    newline-delimited JSON — ping/pong, initialize, prompt/response."""
+    async def go():
+        sock = str(tmp_path / "cla.sock")
+
+        async def handler(msg):
+            req = msg.generate_request
+            return pb.response_message(req.model, f"echo:{req.prompt}", "w0")
+
+        cfg = Config(test_mode=True)
+        srv = IPCServer(cfg, sock, handler)
+        await srv.start()
+        try:
+            import json
+            reader, writer = await asyncio.open_unix_connection(sock)
+
+            async def rpc(obj):
+                writer.write(json.dumps(obj).encode() + b"\n")
+                await writer.drain()
+                return json.loads(await asyncio.wait_for(reader.readline(), 5))
+
+            assert (await rpc({"type": "ping"}))["type"] == "pong"
+            init = await rpc({"type": "initialize"})
+            assert init["type"] == "initialize_status"
+            resp = await rpc({"type": "prompt", "model": "m1",
+                              "content": "hi"})
+            assert resp["type"] == "response"
+            assert resp["content"] == "echo:hi"
+            bad = await rpc({"type": "nope"})
+            assert bad["type"] == "error"
+            # PB on the same socket, fresh connection
+            r2, w2 = await asyncio.open_unix_connection(sock)
+            await write_frame(w2, pb.request_message("m1", "pb msg").encode())
+            frame = await read_frame(r2, timeout=5.0)
+            assert pb.BaseMessage.decode(frame).generate_response.response == \
+                "echo:pb msg"
+            w2.close()
+            writer.close()
+        finally:
+            await srv.stop()
+    asyncio.run(go())
