@@ -32,6 +32,15 @@ class EngineBase:
                        temperature: float = 0.0) -> GenerateResult:
         raise NotImplementedError
 
+    async def generate_stream(self, prompt: str, max_new_tokens: int = 256,
+                              temperature: float = 0.0):
+        """Async iterator of GenerateResult chunks; `text` is the DELTA since
+        the previous chunk, and exactly the last chunk has done_reason set
+        (capability extension over the reference — SURVEY.md §2.2 notes the
+        reference carries `stream` but never streams, gateway.go:243-293).
+        Default: one final chunk from the non-streaming path."""
+        yield await self.generate(prompt, max_new_tokens, temperature)
+
     def throughput(self) -> float:
         """Measured rolling tokens/sec (advertised in metadata)."""
         return 0.0
@@ -71,6 +80,19 @@ class MockEngine(EngineBase):
             f"to: {prompt[:64]}")
         return GenerateResult(text=text, tokens_generated=len(text.split()),
                               duration_ns=int(self._delay * 1e9))
+
+    async def generate_stream(self, prompt: str, max_new_tokens: int = 256,
+                              temperature: float = 0.0):
+        """Word-at-a-time streaming (test seam for the wire/gateway path)."""
+        full = await self.generate(prompt, max_new_tokens, temperature)
+        words = full.text.split(" ")
+        for i, w in enumerate(words[:-1]):
+            yield GenerateResult(text=w + " ", tokens_generated=i + 1,
+                                 done_reason="")
+        yield GenerateResult(text=words[-1] if words else "",
+                             tokens_generated=full.tokens_generated,
+                             duration_ns=full.duration_ns,
+                             done_reason=full.done_reason)
 
     def throughput(self) -> float:
         return self._throughput

@@ -34,6 +34,15 @@ class _Req:
     max_new: int
     future: Future = field(default_factory=Future)
     t0: int = 0
+    # streaming: deltas are pushed onto `chunks` (an asyncio.Queue) via
+    # `loop.call_soon_threadsafe` from the batcher thread
+    chunks: object = None
+    loop: object = None
+    sent_text: str = ""
+
+    def push(self, item) -> None:
+        if self.chunks is not None and self.loop is not None:
+            self.loop.call_soon_threadsafe(self.chunks.put_nowait, item)
 
 
 class _Slot:
@@ -110,12 +119,26 @@ class BatchingHipEngine(EngineBase):
             if done:
                 text = self.tok.decode(toks)
                 self._rate.add(len(toks))
+                reason = "stop" if len(toks) < req.max_new else "length"
+                req.push(GenerateResult(
+                    text=text[len(req.sent_text):],
+                    tokens_generated=len(toks),
+                    duration_ns=time.monotonic_ns() - req.t0,
+                    done_reason=reason))
                 req.future.set_result(GenerateResult(
                     text=text, tokens_generated=len(toks),
                     duration_ns=time.monotonic_ns() - req.t0,
-                    done_reason="stop" if len(toks) < req.max_new else "length"))
+                    done_reason=reason))
                 slot.req = None
                 self._active -= 1
+            elif req.chunks is not None:
+                text = self.tok.decode(toks)
+                delta = text[len(req.sent_text):]
+                if delta:
+                    req.sent_text = text
+                    req.push(GenerateResult(text=delta,
+                                            tokens_generated=len(toks),
+                                            done_reason=""))
 
     def _loop(self) -> None:
         while not self._stop:
@@ -142,6 +165,27 @@ class BatchingHipEngine(EngineBase):
         self._queue.put(req)
         self._wake.set()
         return await asyncio.wrap_future(req.future)
+
+    async def generate_stream(self, prompt: str, max_new_tokens: int = 256,
+                              temperature: float = 0.0):
+        """Streaming under continuous batching: deltas are emitted as the
+        shared decode strides advance this request's slot."""
+        ids = self.tok.encode(prompt)
+        max_new = min(max_new_tokens or self.default_max_new,
+                      self.max_seq - len(ids) - 1)
+        if max_new < 1:
+            raise ValueError("prompt exceeds max_seq")
+        req = _Req(ids=ids or [self.tok.bos_id], max_new=max_new,
+                   t0=time.monotonic_ns(), chunks=asyncio.Queue(),
+                   loop=asyncio.get_running_loop())
+        self._queue.put(req)
+        self._wake.set()
+        while True:
+            chunk = await req.chunks.get()
+            yield chunk
+            if chunk.done_reason:
+                break
+        await asyncio.wrap_future(req.future)
 
     def throughput(self) -> float:
         r = self._rate.rate()

@@ -100,6 +100,85 @@ class HipEngine(EngineBase):
         finally:
             self._active -= 1
 
+    # ------------------------------------------------------------ streaming
+
+    STREAM_STRIDE = 8  # decode steps between emitted deltas (graph replays)
+
+    def _stream_sync(self, prompt: str, max_new_tokens: int,
+                     temperature: float, emit) -> None:
+        """Greedy streaming: decode in strides, emit text deltas. Runs on the
+        engine executor with the lock held for the whole sequence; `emit` is
+        thread-safe."""
+        import numpy as np
+        with self._lock:
+            t0 = time.monotonic_ns()
+            ids = self.tok.encode(prompt)[: self.max_seq - max_new_tokens - 1]
+            if not ids:
+                ids = [self.tok.bos_id]
+            self.eng.reset()
+            self.eng.prefill(np.asarray([ids], dtype=np.int32))
+            n_new = max(1, max_new_tokens)
+            sent = ""
+            done_reason = "length"
+            while True:
+                toks = list(self.eng.gen_tokens(0))
+                finished = False
+                if self.tok.eos_id in toks:
+                    toks = toks[: toks.index(self.tok.eos_id)]
+                    done_reason = "stop"
+                    finished = True
+                if len(toks) >= n_new:
+                    toks = toks[:n_new]
+                    finished = True
+                text = self.tok.decode(toks)
+                delta = text[len(sent):]
+                if finished:
+                    self._rate.add(len(toks))
+                    emit(GenerateResult(
+                        text=delta, tokens_generated=len(toks),
+                        duration_ns=time.monotonic_ns() - t0,
+                        done_reason=done_reason))
+                    return
+                if delta:
+                    sent = text
+                    emit(GenerateResult(text=delta,
+                                        tokens_generated=len(toks),
+                                        done_reason=""))
+                self.eng.decode(min(self.STREAM_STRIDE,
+                                    n_new - len(toks)))
+
+    async def generate_stream(self, prompt: str, max_new_tokens: int = 256,
+                              temperature: float = 0.0):
+        if temperature > 0.0:
+            # sampled path is host-driven; stream it as one final chunk
+            yield await self.generate(prompt, max_new_tokens, temperature)
+            return
+        loop = asyncio.get_running_loop()
+        q: asyncio.Queue = asyncio.Queue()
+
+        def emit(chunk):
+            loop.call_soon_threadsafe(q.put_nowait, chunk)
+
+        def job():
+            try:
+                self._stream_sync(prompt, max_new_tokens, temperature, emit)
+            except Exception as e:  # noqa: BLE001 — surface to the consumer
+                emit(e)
+
+        self._active += 1
+        fut = loop.run_in_executor(self._pool, job)
+        try:
+            while True:
+                chunk = await q.get()
+                if isinstance(chunk, Exception):
+                    raise chunk
+                yield chunk
+                if chunk.done_reason:
+                    break
+            await fut
+        finally:
+            self._active -= 1
+
     # ------------------------------------------------------------ metadata
 
     def throughput(self) -> float:

@@ -126,6 +126,71 @@ class Gateway:
                 last = e
         raise last or ConnectionError("worker unreachable")
 
+    async def request_inference_stream(self, worker, model: str, prompt: str,
+                                       timeout: float = 300.0):
+        """Async iterator of GenerateResponse chunks from a streaming worker
+        (extension; the reference is strictly one-shot, gateway.go:243-293)."""
+        addrs = list(worker.addrs)
+        if not addrs:
+            addrs = await self.peer.discovery.find_peer_addrs(
+                worker.peer_id) or []
+        last: Exception | None = None
+        for addr in addrs:
+            host, port = parse_addr(addr)
+            try:
+                reader, writer = await open_protocol(host, port,
+                                                     PROTO_INFERENCE)
+            except Exception as e:  # noqa: BLE001
+                last = e
+                continue
+            try:
+                await write_frame(writer,
+                                  pb.request_message(model, prompt,
+                                                     stream=True).encode())
+                while True:
+                    frame = await read_frame(reader, timeout=timeout)
+                    resp = pb.BaseMessage.decode(frame).generate_response
+                    if resp is None:
+                        raise ValueError("no GenerateResponse in reply")
+                    yield resp
+                    if resp.done:
+                        return
+            finally:
+                writer.close()
+        raise last or ConnectionError("worker unreachable")
+
+    async def _stream_ndjson(self, request: web.Request, worker, model: str,
+                             prompt: str, chat: bool) -> web.StreamResponse:
+        """Ollama-style streaming: one NDJSON object per chunk."""
+        resp = web.StreamResponse(
+            headers={"Content-Type": "application/x-ndjson"})
+        await resp.prepare(request)
+        try:
+            async for chunk in self.request_inference_stream(worker, model,
+                                                             prompt):
+                obj = {
+                    "model": chunk.model,
+                    "created_at": time.strftime("%Y-%m-%dT%H:%M:%SZ",
+                                                time.gmtime()),
+                    "done": chunk.done,
+                }
+                if chat:
+                    obj["message"] = {"role": "assistant",
+                                      "content": chunk.response}
+                else:
+                    obj["response"] = chunk.response
+                if chunk.done:
+                    obj["done_reason"] = chunk.done_reason or "stop"
+                    obj["total_duration"] = chunk.total_duration
+                    obj["worker_id"] = chunk.worker_id
+                await resp.write(json.dumps(obj).encode("utf-8") + b"\n")
+        except Exception as e:  # noqa: BLE001
+            await resp.write(json.dumps(
+                {"error": f"inference failed: {e}", "done": True}
+            ).encode("utf-8") + b"\n")
+        await resp.write_eof()
+        return resp
+
     async def handle_chat(self, request: web.Request) -> web.Response:
         try:
             body = await request.json()
@@ -143,9 +208,11 @@ class Gateway:
             return web.json_response(
                 {"error": f"no available worker for model {model}"},
                 status=503)  # gateway.go:192-199
+        if body.get("stream"):
+            return await self._stream_ndjson(request, worker, model, prompt,
+                                             chat=True)
         try:
-            resp = await self.request_inference(worker, model, prompt,
-                                                bool(body.get("stream")))
+            resp = await self.request_inference(worker, model, prompt)
         except Exception as e:  # noqa: BLE001
             self.log.warning("inference via %s failed: %s",
                              worker.peer_id, e)
@@ -179,9 +246,11 @@ class Gateway:
             return web.json_response(
                 {"error": f"no available worker for model {model}"},
                 status=503)
+        if body.get("stream"):
+            return await self._stream_ndjson(request, worker, model, prompt,
+                                             chat=False)
         try:
-            resp = await self.request_inference(worker, model, prompt,
-                                                bool(body.get("stream")))
+            resp = await self.request_inference(worker, model, prompt)
         except Exception as e:  # noqa: BLE001
             return web.json_response({"error": f"inference failed: {e}"},
                                      status=500)

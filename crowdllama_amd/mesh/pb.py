@@ -221,8 +221,11 @@ def request_message(model: str, prompt: str, stream: bool = False) -> BaseMessag
 
 def response_message(model: str, response: str, worker_id: str = "",
                      done_reason: str = "stop",
-                     total_duration_ns: int = 0) -> BaseMessage:
+                     total_duration_ns: int = 0,
+                     done: bool = True) -> BaseMessage:
+    # done=False frames are streamed chunks (capability extension over the
+    # reference, which carries `stream` but never streams — SURVEY.md §2.2)
     return BaseMessage(generate_response=GenerateResponse(
-        model=model, created_at=Timestamp.now(), response=response, done=True,
-        done_reason=done_reason, worker_id=worker_id,
+        model=model, created_at=Timestamp.now(), response=response, done=done,
+        done_reason=done_reason if done else "", worker_id=worker_id,
         total_duration=total_duration_ns))

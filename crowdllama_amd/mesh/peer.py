@@ -181,6 +181,23 @@ class Peer:
             if engine is None:
                 raise KeyError(f"model {req.model!r} not served here "
                                f"(have {sorted(self.engines)})")
+            if req.stream:
+                # streamed chunks: one done=False frame per text delta, then
+                # a final done=True frame (extension over the reference,
+                # which reserves `stream` but never streams — SURVEY.md §2.2)
+                total_tokens = 0
+                async for chunk in engine.generate_stream(req.prompt):
+                    total_tokens = chunk.tokens_generated
+                    final = bool(chunk.done_reason)
+                    resp = pb.response_message(
+                        req.model, chunk.text, self.peer_id,
+                        done_reason=chunk.done_reason, done=final,
+                        total_duration_ns=(time.monotonic_ns() - t0
+                                           if final else 0))
+                    await write_frame(writer, resp.encode())
+                self.rate.add(total_tokens)
+                self.requests_served += 1
+                return
             result = await engine.generate(req.prompt)
             self.rate.add(result.tokens_generated)
             self.requests_served += 1

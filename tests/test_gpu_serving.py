@@ -178,3 +178,110 @@ def test_continuous_batching_engine(has_gpu):
             assert g.text == w.text, (g.text, w.text)
         await batched.close()
     asyncio.run(go())
+
+
+def test_hip_engine_streaming_matches_generate(has_gpu):
+    """generate_stream deltas concatenate to exactly the non-streamed greedy
+    output (same engine, same KV discipline)."""
+    import asyncio
+    from crowdllama_amd.engine.hip_engine import HipEngine
+    from crowdllama_amd.models import synth_path
+    path = synth_path("testllama", scheme="q4_k_m", mode="exact", seed=7)
+    eng = HipEngine("testllama", path, max_seq=128)
+
+    async def go():
+        want = await eng.generate("abc def", max_new_tokens=12)
+        chunks = []
+        async for c in eng.generate_stream("abc def", max_new_tokens=12):
+            chunks.append(c)
+        got = "".join(c.text for c in chunks)
+        assert got == want.text, (got, want.text)
+        assert chunks[-1].done_reason in ("stop", "length")
+        assert all(not c.done_reason for c in chunks[:-1])
+    asyncio.run(go())
+
+
+def test_batching_engine_streaming(has_gpu):
+    """Streaming under continuous batching: two concurrent streamed requests
+    share decode strides; each stream reassembles its own text."""
+    import asyncio
+    from crowdllama_amd.engine.batching import BatchingHipEngine
+    from crowdllama_amd.engine.hip_engine import HipEngine
+    from crowdllama_amd.models import synth_path
+    path = synth_path("testllama", scheme="q4_k_m", mode="exact", seed=7)
+    single = HipEngine("testllama", path, max_seq=128)
+    batched = BatchingHipEngine("testllama", path, batch=2, max_seq=128)
+
+    async def collect(prompt):
+        out = []
+        async for c in batched.generate_stream(prompt, max_new_tokens=10):
+            out.append(c.text)
+        return "".join(out)
+
+    async def go():
+        w1 = await single.generate("abc", max_new_tokens=10)
+        w2 = await single.generate("hello there", max_new_tokens=10)
+        g1, g2 = await asyncio.gather(collect("abc"),
+                                      collect("hello there"))
+        assert g1 == w1.text, (g1, w1.text)
+        assert g2 == w2.text, (g2, w2.text)
+        await batched.close()
+    asyncio.run(go())
+
+
+def test_streaming_through_mesh_gpu(has_gpu, tmp_path):
+    """stream=true through DHT + worker(HIP) + gateway: NDJSON chunks
+    reassemble to the full response."""
+    import json as _json
+    import time
+    from crowdllama_amd.config import Config
+    from crowdllama_amd.engine.hip_engine import HipEngine
+    from crowdllama_amd.mesh.dhtnode import DHTServer
+    from crowdllama_amd.mesh.gateway import Gateway
+    from crowdllama_amd.mesh.peer import Peer
+    from crowdllama_amd.models import synth_path
+
+    path = synth_path("testllama", scheme="q4_k_m", mode="exact", seed=7)
+    engine = HipEngine("testllama", path, max_seq=128)
+
+    async def go():
+        import aiohttp
+        def mk(c):
+            return Config(test_mode=True, listen_host="127.0.0.1",
+                          key_path=str(tmp_path / f"{c}.key"))
+        dht = DHTServer(mk("dht"), "CLADHT")
+        dht_port = await dht.start("127.0.0.1", 0)
+        boot = [f"127.0.0.1:{dht_port}"]
+        wcfg = mk("worker")
+        wcfg.bootstrap_peers = boot
+        worker = Peer(wcfg, worker_mode=True, engines={"testllama": engine})
+        await worker.start()
+        ccfg = mk("consumer")
+        ccfg.bootstrap_peers = boot
+        consumer = Peer(ccfg, worker_mode=False)
+        await consumer.start()
+        gw = Gateway(consumer, ccfg)
+        gw_port = await gw.start(port=0)
+        try:
+            deadline = time.time() + 20
+            while gw.find_best_worker("testllama") is None:
+                assert time.time() < deadline
+                await asyncio.sleep(0.1)
+            want = await engine.generate("abc", max_new_tokens=12)
+            async with aiohttp.ClientSession() as s:
+                async with s.post(f"http://127.0.0.1:{gw_port}/api/chat",
+                                  json={"model": "testllama", "stream": True,
+                                        "messages": [{"role": "user",
+                                                      "content": "abc"}]}) as r:
+                    assert r.status == 200
+                    lines = [_json.loads(ln) async for ln in r.content
+                             if ln.strip()]
+            text = "".join(ln["message"]["content"] for ln in lines)
+            assert text == want.text, (text, want.text)
+            assert lines[-1]["done"] is True
+        finally:
+            await gw.stop()
+            await consumer.stop()
+            await worker.stop()
+            await dht.stop()
+    asyncio.run(go())

@@ -239,3 +239,91 @@ def test_worker_failure_and_eviction(mesh_cfg):
             await consumer.stop()
             await dht.stop()
     asyncio.run(go())
+
+
+def test_streaming_chat_ndjson(mesh_cfg):
+    """stream=true: gateway emits NDJSON chunks whose concatenated content
+    equals the non-streamed response, final line carries done/done_reason
+    (capability extension over the reference — SURVEY.md §2.2)."""
+    async def go():
+        import aiohttp
+        import json as _json
+        dht = DHTServer(mesh_cfg("dht"), "CLADHT")
+        dht_port = await dht.start("127.0.0.1", 0)
+        boot = [f"127.0.0.1:{dht_port}"]
+        wcfg = mesh_cfg("worker")
+        wcfg.bootstrap_peers = boot
+        worker = Peer(wcfg, worker_mode=True,
+                      engines={"tinyllama": MockEngine(
+                          "tinyllama", response="alpha beta gamma delta")})
+        await worker.start()
+        ccfg = mesh_cfg("consumer")
+        ccfg.bootstrap_peers = boot
+        consumer = Peer(ccfg, worker_mode=False)
+        await consumer.start()
+        gw = Gateway(consumer, ccfg)
+        gw_port = await gw.start(port=0)
+        try:
+            await _poll(lambda: gw.find_best_worker("tinyllama") is not None,
+                        desc="worker discovery")
+            async with aiohttp.ClientSession() as s:
+                async with s.post(f"http://127.0.0.1:{gw_port}/api/chat",
+                                  json={"model": "tinyllama", "stream": True,
+                                        "messages": [{"role": "user",
+                                                      "content": "hi"}]}) as r:
+                    assert r.status == 200
+                    assert "ndjson" in r.headers["Content-Type"]
+                    lines = [_json.loads(ln) async for ln in r.content
+                             if ln.strip()]
+            assert len(lines) > 1, "expected multiple streamed chunks"
+            text = "".join(ln["message"]["content"] for ln in lines)
+            assert text == "alpha beta gamma delta"
+            assert all(not ln["done"] for ln in lines[:-1])
+            assert lines[-1]["done"] is True
+            assert lines[-1]["done_reason"] == "stop"
+            assert lines[-1]["worker_id"] == worker.peer_id
+        finally:
+            await gw.stop()
+            await consumer.stop()
+            await worker.stop()
+            await dht.stop()
+    asyncio.run(go())
+
+
+def test_streaming_generate_ndjson(mesh_cfg):
+    """/api/generate stream=true uses the `response` field per chunk."""
+    async def go():
+        import aiohttp
+        import json as _json
+        dht = DHTServer(mesh_cfg("dht"), "CLADHT")
+        dht_port = await dht.start("127.0.0.1", 0)
+        boot = [f"127.0.0.1:{dht_port}"]
+        wcfg = mesh_cfg("worker")
+        wcfg.bootstrap_peers = boot
+        worker = Peer(wcfg, worker_mode=True,
+                      engines={"m": MockEngine("m", response="one two")})
+        await worker.start()
+        ccfg = mesh_cfg("consumer")
+        ccfg.bootstrap_peers = boot
+        consumer = Peer(ccfg, worker_mode=False)
+        await consumer.start()
+        gw = Gateway(consumer, ccfg)
+        gw_port = await gw.start(port=0)
+        try:
+            await _poll(lambda: gw.find_best_worker("m") is not None,
+                        desc="worker discovery")
+            async with aiohttp.ClientSession() as s:
+                async with s.post(f"http://127.0.0.1:{gw_port}/api/generate",
+                                  json={"model": "m", "prompt": "x",
+                                        "stream": True}) as r:
+                    assert r.status == 200
+                    lines = [_json.loads(ln) async for ln in r.content
+                             if ln.strip()]
+            assert "".join(ln["response"] for ln in lines) == "one two"
+            assert lines[-1]["done"] is True
+        finally:
+            await gw.stop()
+            await consumer.stop()
+            await worker.stop()
+            await dht.stop()
+    asyncio.run(go())
