@@ -564,10 +564,22 @@ __global__ __launch_bounds__(256) void k_attn_prefill(
         default: throw std::runtime_error("bad dtype");               \
     }
 
+// split-K workgroup target: >=2 WGs per CU keeps all 8 XCDs fed when M is
+// small; overridable for on-hardware sweeps (CLA_SPLITK_TARGET).
+static int splitk_target() {
+    static int t = [] {
+        const char* e = getenv("CLA_SPLITK_TARGET");
+        int v = e ? atoi(e) : 768;
+        return v > 0 ? v : 768;
+    }();
+    return t;
+}
+
 bool gemm_uses_splitk(int N, int K, int M) {
     if (M > 32) return false;
     const int n_tiles = (N + BN - 1) / BN;
-    int splitk = K / BK < 768 / n_tiles ? K / BK : 768 / n_tiles;
+    const int tgt = splitk_target();
+    int splitk = K / BK < tgt / n_tiles ? K / BK : tgt / n_tiles;
     return splitk > 1;
 }
 
@@ -583,9 +595,10 @@ void launch_gemm_ex(const WTensor& w, const float* X, const float* X2,
     // >=512 workgroups, bounded by the number of K-steps.
     int splitk = 1;
     if (small_m) {
-        splitk = K / BK < 768 / (n_tiles ? n_tiles : 1)
+        const int tgt = splitk_target();
+        splitk = K / BK < tgt / (n_tiles ? n_tiles : 1)
                      ? K / BK
-                     : 768 / (n_tiles ? n_tiles : 1);
+                     : tgt / (n_tiles ? n_tiles : 1);
         if (splitk < 1) splitk = 1;
     }
     const int k_chunk = ((K / BK + splitk - 1) / splitk) * BK;
