@@ -267,8 +267,15 @@ def test_streaming_through_mesh_gpu(has_gpu, tmp_path):
             while gw.find_best_worker("testllama") is None:
                 assert time.time() < deadline
                 await asyncio.sleep(0.1)
-            want = await engine.generate("abc", max_new_tokens=12)
             async with aiohttp.ClientSession() as s:
+                # non-streamed reference through the same mesh path (same
+                # worker-side token budget)
+                async with s.post(f"http://127.0.0.1:{gw_port}/api/chat",
+                                  json={"model": "testllama",
+                                        "messages": [{"role": "user",
+                                                      "content": "abc"}]}) as r:
+                    ref = await r.json()
+                    assert r.status == 200, ref
                 async with s.post(f"http://127.0.0.1:{gw_port}/api/chat",
                                   json={"model": "testllama", "stream": True,
                                         "messages": [{"role": "user",
@@ -277,7 +284,8 @@ def test_streaming_through_mesh_gpu(has_gpu, tmp_path):
                     lines = [_json.loads(ln) async for ln in r.content
                              if ln.strip()]
             text = "".join(ln["message"]["content"] for ln in lines)
-            assert text == want.text, (text, want.text)
+            want = ref["message"]["content"]
+            assert text == want, (text, want)
             assert lines[-1]["done"] is True
         finally:
             await gw.stop()
