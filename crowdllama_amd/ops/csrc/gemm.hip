@@ -164,7 +164,13 @@ __global__ __launch_bounds__(256) void k_gemm(
     // k_chunk: this block's K-range is [z*k_chunk, min((z+1)*k_chunk, K));
     // splitk > 1 => partial results accumulated with atomicAdd (C pre-zeroed,
     // residual folded in by the z==0 block).
-    constexpr int FM = BM_ / 32;                 // m-fragments per wave
+    // Wave grid: BM>=32 uses 2x2 (each wave BM/2 rows x 64 cols); BM=16 uses
+    // 1x4 (each wave all 16 rows x 32 cols) so no MFMA row is padding when
+    // the decode batch is <=16.
+    constexpr int WMW = (BM_ == 16) ? 1 : 2;     // waves tiling M
+    constexpr int WNW = 4 / WMW;                 // waves tiling N
+    constexpr int JF = BN / WNW / 16;            // b-fragments per wave
+    constexpr int FM = (BM_ / WMW) / 16;         // a-fragments per wave
     __shared__ __attribute__((aligned(16))) uint16_t Xl[BM_ * LDW];
     __shared__ __attribute__((aligned(16))) uint16_t Wl[BN * LDW];
 
@@ -175,14 +181,15 @@ __global__ __launch_bounds__(256) void k_gemm(
     const int kb_lo = bz * k_chunk;
     const int kb_hi = min(kb_lo + k_chunk, K);
     const int wid = tid >> 6, lane = tid & 63;
-    const int wm = wid >> 1, wn = wid & 1;       // 2x2 wave grid
+    const int wm = (WMW == 1) ? 0 : (wid >> 1);
+    const int wn = (WMW == 1) ? wid : (wid & 1);
     const int lrow = lane & 15, lk = lane >> 4;  // fragment lane coords
 
-    f32x4 acc[FM][4];
+    f32x4 acc[FM][JF];
     #pragma unroll
     for (int i = 0; i < FM; i++)
         #pragma unroll
-        for (int j = 0; j < 4; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+        for (int j = 0; j < JF; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
     const int srow0 = tid >> 1, sh0 = tid & 1;   // W staging coords
     const int64_t gn_s = (int64_t)n0 + srow0;
@@ -192,12 +199,13 @@ __global__ __launch_bounds__(256) void k_gemm(
     // BM=32: the X tile is 8 floats/thread — prefetch it alongside W so the
     // whole staging phase runs from registers (M<=32 decode batches have
     // too few workgroups to hide latency with occupancy alone).
+    constexpr bool SMALLM = (BM_ <= 32);
     float4 xr0, xr1, xr0n, xr1n, ur0, ur1, ur0n, ur1n;
-    const int xrow = (BM_ == 32) ? (tid >> 3) : 0;
-    const int xseg = (BM_ == 32) ? (tid & 7) : 0;
+    const int xrow = SMALLM ? (tid >> 3) : 0;    // 0..31 (BM=16: >=16 idle)
+    const int xseg = SMALLM ? (tid & 7) : 0;
     const int xgm = m0 + xrow < M ? m0 + xrow : (M > 0 ? M - 1 : 0);
-    const bool xvalid0 = (m0 + xrow) < M;
-    if (BM_ == 32) {
+    const bool xvalid0 = (m0 + xrow) < M && xrow < BM_;
+    if (SMALLM) {
         const float4* src = reinterpret_cast<const float4*>(
             X + (size_t)xgm * ldx + kb_lo + xseg * 8);
         xr0 = src[0];
@@ -212,7 +220,7 @@ __global__ __launch_bounds__(256) void k_gemm(
 
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
         // ---- stage X tile (f32 -> bf16), 8-elem units ----
-        if (BM_ == 32) {
+        if (SMALLM) {
             uint16_t tmp[8];
             float4 v[2] = {xr0, xr1};
             if constexpr (XSILU) {
@@ -236,8 +244,9 @@ __global__ __launch_bounds__(256) void k_gemm(
                 #pragma unroll
                 for (int j = 0; j < 8; j++) tmp[j] = 0;
             }
-            *reinterpret_cast<uint4*>(Xl + xrow * LDW + xseg * 8) =
-                *reinterpret_cast<const uint4*>(tmp);
+            if (xrow < BM_)
+                *reinterpret_cast<uint4*>(Xl + xrow * LDW + xseg * 8) =
+                    *reinterpret_cast<const uint4*>(tmp);
         } else {
             constexpr int UNITS = BM_ * BK / 8;   // 8 bf16 per unit
             #pragma unroll
@@ -307,7 +316,7 @@ __global__ __launch_bounds__(256) void k_gemm(
         {
             const int kbn = (kb + BK < kb_hi) ? kb + BK : kb;
             load_w_raw<W>(qs, hdr, gn_c, K, kbn, sh0, &wr_next);
-            if (BM_ == 32) {
+            if (SMALLM) {
                 const float4* src = reinterpret_cast<const float4*>(
                     X + (size_t)xgm * ldx + kbn + xseg * 8);
                 xr0n = src[0];
@@ -323,29 +332,29 @@ __global__ __launch_bounds__(256) void k_gemm(
         // ---- MFMA over the tile ----
         #pragma unroll
         for (int ks = 0; ks < BK; ks += 32) {
-            bf16x8 a[FM], b[4];
+            bf16x8 a[FM], b[JF];
             #pragma unroll
             for (int i = 0; i < FM; i++) {
-                const int xr = wm * (BM_ / 2) + i * 16 + lrow;
+                const int xr = wm * (BM_ / WMW) + i * 16 + lrow;
                 a[i] = *reinterpret_cast<const bf16x8*>(
                     Xl + xr * LDW + ks + lk * 8);
             }
             #pragma unroll
-            for (int j = 0; j < 4; j++) {
-                const int wr = wn * 64 + j * 16 + lrow;
+            for (int j = 0; j < JF; j++) {
+                const int wr = wn * (BN / WNW) + j * 16 + lrow;
                 b[j] = *reinterpret_cast<const bf16x8*>(
                     Wl + wr * LDW + ks + lk * 8);
             }
             #pragma unroll
             for (int i = 0; i < FM; i++)
                 #pragma unroll
-                for (int j = 0; j < 4; j++)
+                for (int j = 0; j < JF; j++)
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a[i], b[j], acc[i][j], 0, 0, 0);
         }
         __syncthreads();
         wr = wr_next;
-        if (BM_ == 32) {
+        if (SMALLM) {
             xr0 = xr0n; xr1 = xr1n;
             if constexpr (XSILU) { ur0 = ur0n; ur1 = ur1n; }
         }
@@ -356,11 +365,11 @@ __global__ __launch_bounds__(256) void k_gemm(
     for (int i = 0; i < FM; i++) {
         #pragma unroll
         for (int r = 0; r < 4; r++) {
-            const int m = m0 + wm * (BM_ / 2) + i * 16 + lk * 4 + r;
+            const int m = m0 + wm * (BM_ / WMW) + i * 16 + lk * 4 + r;
             if (m >= M) continue;
             #pragma unroll
-            for (int j = 0; j < 4; j++) {
-                const int n = n0 + wn * 64 + j * 16 + lrow;
+            for (int j = 0; j < JF; j++) {
+                const int n = n0 + wn * (BN / WNW) + j * 16 + lrow;
                 if (n >= N) continue;
                 const size_t idx = (size_t)m * ldc + n;
                 const float rv = (res && (!splitk || bz == 0)) ? res[idx] : 0.f;
@@ -611,7 +620,9 @@ void launch_gemm_ex(const WTensor& w, const float* X, const float* X2,
             M, N, K, ldc, ldx, k_chunk)
     #define GEMM_CASE(WT)                                                      \
         do {                                                                   \
-            if (small_m && xsilu) GEMM_ONE(WT, 32, true);                      \
+            if (M <= 16 && xsilu) GEMM_ONE(WT, 16, true);                      \
+            else if (M <= 16) GEMM_ONE(WT, 16, false);                         \
+            else if (small_m && xsilu) GEMM_ONE(WT, 32, true);                 \
             else if (small_m) GEMM_ONE(WT, 32, false);                         \
             else if (xsilu) GEMM_ONE(WT, 128, true);                           \
             else GEMM_ONE(WT, 128, false);                                     \
