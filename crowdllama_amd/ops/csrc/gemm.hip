@@ -171,8 +171,7 @@ __global__ __launch_bounds__(256) void k_gemm(
     constexpr int WNW = 4 / WMW;                 // waves tiling N
     constexpr int JF = BN / WNW / 16;            // b-fragments per wave
     constexpr int FM = (BM_ / WMW) / 16;         // a-fragments per wave
-    __shared__ __attribute__((aligned(16))) uint16_t Xl[BM_ == 16 ? 1
-                                                         : BM_ * LDW];
+    __shared__ __attribute__((aligned(16))) uint16_t Xl[BM_ * LDW];
     __shared__ __attribute__((aligned(16))) uint16_t Wl[BN * LDW];
 
     const int tid = threadIdx.x;
@@ -200,20 +199,12 @@ __global__ __launch_bounds__(256) void k_gemm(
     // BM=32: the X tile is 8 floats/thread — prefetch it alongside W so the
     // whole staging phase runs from registers (M<=32 decode batches have
     // too few workgroups to hide latency with occupancy alone).
-    constexpr bool SMALLM = (BM_ == 32);
-    // BM=16: a-fragments are lane-addressed straight from global X (L2-hot
-    // at decode batch sizes) — no X staging, no Xl barrier dependency.
-    constexpr bool XDIRECT = (BM_ == 16);
+    constexpr bool SMALLM = (BM_ <= 32);
     float4 xr0, xr1, xr0n, xr1n, ur0, ur1, ur0n, ur1n;
     const int xrow = SMALLM ? (tid >> 3) : 0;    // 0..31 (BM=16: >=16 idle)
     const int xseg = SMALLM ? (tid & 7) : 0;
     const int xgm = m0 + xrow < M ? m0 + xrow : (M > 0 ? M - 1 : 0);
     const bool xvalid0 = (m0 + xrow) < M && xrow < BM_;
-    // XDIRECT: this lane's a-fragment row (clamped; m >= M rows produce
-    // accumulator rows the epilogue discards)
-    const int xlm = (m0 + lrow) < M ? m0 + lrow : (M > 0 ? M - 1 : 0);
-    const float* xlp = XDIRECT ? X + (size_t)xlm * ldx : nullptr;
-    const float* xlp2 = (XDIRECT && XSILU) ? X2 + (size_t)xlm * ldx : nullptr;
     if (SMALLM) {
         const float4* src = reinterpret_cast<const float4*>(
             X + (size_t)xgm * ldx + kb_lo + xseg * 8);
@@ -229,9 +220,7 @@ __global__ __launch_bounds__(256) void k_gemm(
 
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
         // ---- stage X tile (f32 -> bf16), 8-elem units ----
-        if (XDIRECT) {
-            // nothing: a-fragments load from global in the MFMA phase
-        } else if (SMALLM) {
+        if (SMALLM) {
             uint16_t tmp[8];
             float4 v[2] = {xr0, xr1};
             if constexpr (XSILU) {
@@ -344,37 +333,11 @@ __global__ __launch_bounds__(256) void k_gemm(
         #pragma unroll
         for (int ks = 0; ks < BK; ks += 32) {
             bf16x8 a[FM], b[JF];
-            if constexpr (XDIRECT) {
-                float4 v0 = *reinterpret_cast<const float4*>(
-                    xlp + kb + ks + lk * 8);
-                float4 v1 = *reinterpret_cast<const float4*>(
-                    xlp + kb + ks + lk * 8 + 4);
-                if constexpr (XSILU) {
-                    const float4 u0 = *reinterpret_cast<const float4*>(
-                        xlp2 + kb + ks + lk * 8);
-                    const float4 u1 = *reinterpret_cast<const float4*>(
-                        xlp2 + kb + ks + lk * 8 + 4);
-                    v0.x = (v0.x / (1.f + __expf(-v0.x))) * u0.x;
-                    v0.y = (v0.y / (1.f + __expf(-v0.y))) * u0.y;
-                    v0.z = (v0.z / (1.f + __expf(-v0.z))) * u0.z;
-                    v0.w = (v0.w / (1.f + __expf(-v0.w))) * u0.w;
-                    v1.x = (v1.x / (1.f + __expf(-v1.x))) * u1.x;
-                    v1.y = (v1.y / (1.f + __expf(-v1.y))) * u1.y;
-                    v1.z = (v1.z / (1.f + __expf(-v1.z))) * u1.z;
-                    v1.w = (v1.w / (1.f + __expf(-v1.w))) * u1.w;
-                }
-                uint16_t t[8] = {f32_to_bf16b(v0.x), f32_to_bf16b(v0.y),
-                                 f32_to_bf16b(v0.z), f32_to_bf16b(v0.w),
-                                 f32_to_bf16b(v1.x), f32_to_bf16b(v1.y),
-                                 f32_to_bf16b(v1.z), f32_to_bf16b(v1.w)};
-                a[0] = *reinterpret_cast<const bf16x8*>(t);
-            } else {
-                #pragma unroll
-                for (int i = 0; i < FM; i++) {
-                    const int xr = wm * (BM_ / WMW) + i * 16 + lrow;
-                    a[i] = *reinterpret_cast<const bf16x8*>(
-                        Xl + xr * LDW + ks + lk * 8);
-                }
+            #pragma unroll
+            for (int i = 0; i < FM; i++) {
+                const int xr = wm * (BM_ / WMW) + i * 16 + lrow;
+                a[i] = *reinterpret_cast<const bf16x8*>(
+                    Xl + xr * LDW + ks + lk * 8);
             }
             #pragma unroll
             for (int j = 0; j < JF; j++) {
