@@ -171,8 +171,10 @@ __global__ __launch_bounds__(256) void k_gemm(
     constexpr int WNW = 4 / WMW;                 // waves tiling N
     constexpr int JF = BN / WNW / 16;            // b-fragments per wave
     constexpr int FM = (BM_ / WMW) / 16;         // a-fragments per wave
-    __shared__ __attribute__((aligned(16))) uint16_t Xl[BM_ * LDW];
-    __shared__ __attribute__((aligned(16))) uint16_t Wl[BN * LDW];
+    // double-buffered tiles: stage tile t+1 into buf[1-p] while the MFMA
+    // phase reads buf[p] — one barrier per tile, LDS writes overlap MFMA.
+    __shared__ __attribute__((aligned(16))) uint16_t Xl[2][BM_ * LDW];
+    __shared__ __attribute__((aligned(16))) uint16_t Wl[2][BN * LDW];
 
     const int tid = threadIdx.x;
     const int bn = blockIdx.x, bm = blockIdx.y, bz = blockIdx.z;
@@ -180,6 +182,9 @@ __global__ __launch_bounds__(256) void k_gemm(
     const int m0 = bm * BM_, n0 = bn * BN;
     const int kb_lo = bz * k_chunk;
     const int kb_hi = min(kb_lo + k_chunk, K);
+    // ceil-rounded split-K chunking can leave trailing z-blocks with an
+    // empty K-range; they contribute nothing and must not touch memory
+    if (kb_lo >= kb_hi) return;
     const int wid = tid >> 6, lane = tid & 63;
     const int wm = (WMW == 1) ? 0 : (wid >> 1);
     const int wn = (WMW == 1) ? wid : (wid & 1);
@@ -218,7 +223,7 @@ __global__ __launch_bounds__(256) void k_gemm(
         }
     }
 
-    for (int kb = kb_lo; kb < kb_hi; kb += BK) {
+    auto stage_tile = [&](int kb_s, int pb) {
         // ---- stage X tile (f32 -> bf16), 8-elem units ----
         if (SMALLM) {
             uint16_t tmp[8];
@@ -233,20 +238,22 @@ __global__ __launch_bounds__(256) void k_gemm(
                     v[j].w = (v[j].w / (1.f + __expf(-v[j].w))) * u[j].w;
                 }
             }
+            uint16_t tmp8[8];
             #pragma unroll
             for (int j = 0; j < 2; j++) {
-                tmp[j * 4 + 0] = f32_to_bf16b(v[j].x);
-                tmp[j * 4 + 1] = f32_to_bf16b(v[j].y);
-                tmp[j * 4 + 2] = f32_to_bf16b(v[j].z);
-                tmp[j * 4 + 3] = f32_to_bf16b(v[j].w);
+                tmp8[j * 4 + 0] = f32_to_bf16b(v[j].x);
+                tmp8[j * 4 + 1] = f32_to_bf16b(v[j].y);
+                tmp8[j * 4 + 2] = f32_to_bf16b(v[j].z);
+                tmp8[j * 4 + 3] = f32_to_bf16b(v[j].w);
             }
+            (void)tmp;
             if (!xvalid0) {
                 #pragma unroll
-                for (int j = 0; j < 8; j++) tmp[j] = 0;
+                for (int j = 0; j < 8; j++) tmp8[j] = 0;
             }
             if (xrow < BM_)
-                *reinterpret_cast<uint4*>(Xl + xrow * LDW + xseg * 8) =
-                    *reinterpret_cast<const uint4*>(tmp);
+                *reinterpret_cast<uint4*>(Xl[pb] + xrow * LDW + xseg * 8) =
+                    *reinterpret_cast<const uint4*>(tmp8);
         } else {
             constexpr int UNITS = BM_ * BK / 8;   // 8 bf16 per unit
             #pragma unroll
@@ -258,13 +265,13 @@ __global__ __launch_bounds__(256) void k_gemm(
                 uint16_t tmp[8];
                 if (gm < M) {
                     const float4* src = reinterpret_cast<const float4*>(
-                        X + (size_t)gm * ldx + kb + seg * 8);
+                        X + (size_t)gm * ldx + kb_s + seg * 8);
                     #pragma unroll
                     for (int j = 0; j < 2; j++) {
                         float4 v = src[j];
                         if constexpr (XSILU) {
                             const float4 u = reinterpret_cast<const float4*>(
-                                X2 + (size_t)gm * ldx + kb + seg * 8)[j];
+                                X2 + (size_t)gm * ldx + kb_s + seg * 8)[j];
                             v.x = (v.x / (1.f + __expf(-v.x))) * u.x;
                             v.y = (v.y / (1.f + __expf(-v.y))) * u.y;
                             v.z = (v.z / (1.f + __expf(-v.z))) * u.z;
@@ -279,7 +286,7 @@ __global__ __launch_bounds__(256) void k_gemm(
                     #pragma unroll
                     for (int j = 0; j < 8; j++) tmp[j] = 0;
                 }
-                *reinterpret_cast<uint4*>(Xl + row * LDW + seg * 8) =
+                *reinterpret_cast<uint4*>(Xl[pb] + row * LDW + seg * 8) =
                     *reinterpret_cast<const uint4*>(tmp);
             }
         }
@@ -288,48 +295,70 @@ __global__ __launch_bounds__(256) void k_gemm(
             const int srow = srow0, sh = sh0;
             uint16_t tmp[32];
             if (gn_s < N) {
-                stage_w_slice<W>(qs, hdr, wr, gn_c, K, kb, sh, tmp);
+                stage_w_slice<W>(qs, hdr, wr, gn_c, K, kb_s, sh, tmp);
             } else {
                 #pragma unroll
                 for (int j = 0; j < 32; j++) tmp[j] = 0;
             }
             if constexpr (W == DT::DQ4K) {
                 uint4* d0 = reinterpret_cast<uint4*>(
-                    Wl + srow * LDW + sh * 16);
+                    Wl[pb] + srow * LDW + sh * 16);
                 uint4* d1 = reinterpret_cast<uint4*>(
-                    Wl + srow * LDW + 32 + sh * 16);
+                    Wl[pb] + srow * LDW + 32 + sh * 16);
                 d0[0] = reinterpret_cast<const uint4*>(tmp)[0];
                 d0[1] = reinterpret_cast<const uint4*>(tmp)[1];
                 d1[0] = reinterpret_cast<const uint4*>(tmp)[2];
                 d1[1] = reinterpret_cast<const uint4*>(tmp)[3];
             } else {
                 uint4* dst = reinterpret_cast<uint4*>(
-                    Wl + srow * LDW + sh * 32);
+                    Wl[pb] + srow * LDW + sh * 32);
                 #pragma unroll
                 for (int j = 0; j < 4; j++)
                     dst[j] = reinterpret_cast<const uint4*>(tmp)[j];
             }
         }
-        __syncthreads();
-        // T14: next W tile's loads issue before the MFMA phase — their
-        // HBM latency hides under the matrix work.
-        {
-            const int kbn = (kb + BK < kb_hi) ? kb + BK : kb;
-            load_w_raw<W>(qs, hdr, gn_c, K, kbn, sh0, &wr_next);
-            if (SMALLM) {
-                const float4* src = reinterpret_cast<const float4*>(
-                    X + (size_t)xgm * ldx + kbn + xseg * 8);
-                xr0n = src[0];
-                xr1n = src[1];
-                if constexpr (XSILU) {
-                    const float4* up = reinterpret_cast<const float4*>(
-                        X2 + (size_t)xgm * ldx + kbn + xseg * 8);
-                    ur0n = up[0];
-                    ur1n = up[1];
-                }
+    };
+
+    auto load_next = [&](int kb_n) {
+        load_w_raw<W>(qs, hdr, gn_c, K, kb_n, sh0, &wr_next);
+        if (SMALLM) {
+            const float4* src = reinterpret_cast<const float4*>(
+                X + (size_t)xgm * ldx + kb_n + xseg * 8);
+            xr0n = src[0];
+            xr1n = src[1];
+            if constexpr (XSILU) {
+                const float4* up = reinterpret_cast<const float4*>(
+                    X2 + (size_t)xgm * ldx + kb_n + xseg * 8);
+                ur0n = up[0];
+                ur1n = up[1];
             }
         }
-        // ---- MFMA over the tile ----
+    };
+
+    // prologue: tile 0 into buf 0, then prefetch tile 1's raw bytes
+    stage_tile(kb_lo, 0);
+    __syncthreads();
+    {
+        const int kb1 = (kb_lo + BK < kb_hi) ? kb_lo + BK : kb_lo;
+        load_next(kb1);
+        wr = wr_next;
+        if (SMALLM) {
+            xr0 = xr0n; xr1 = xr1n;
+            if constexpr (XSILU) { ur0 = ur0n; ur1 = ur1n; }
+        }
+    }
+
+    int pb = 0;
+    for (int kb = kb_lo; kb < kb_hi; kb += BK) {
+        const bool has_next = (kb + BK) < kb_hi;
+        if (has_next) {
+            // raw loads for tile t+2 issue first: their HBM latency hides
+            // under this iteration's ds_writes + MFMA (T14)
+            const int kb2 = (kb + 2 * BK < kb_hi) ? kb + 2 * BK : kb;
+            load_next(kb2);
+            stage_tile(kb + BK, 1 - pb);   // overlaps MFMA below (other buf)
+        }
+        // ---- MFMA over the resident tile ----
         #pragma unroll
         for (int ks = 0; ks < BK; ks += 32) {
             bf16x8 a[FM], b[JF];
@@ -337,13 +366,13 @@ __global__ __launch_bounds__(256) void k_gemm(
             for (int i = 0; i < FM; i++) {
                 const int xr = wm * (BM_ / WMW) + i * 16 + lrow;
                 a[i] = *reinterpret_cast<const bf16x8*>(
-                    Xl + xr * LDW + ks + lk * 8);
+                    Xl[pb] + xr * LDW + ks + lk * 8);
             }
             #pragma unroll
             for (int j = 0; j < JF; j++) {
-                const int wr = wn * (BN / WNW) + j * 16 + lrow;
+                const int wrr = wn * (BN / WNW) + j * 16 + lrow;
                 b[j] = *reinterpret_cast<const bf16x8*>(
-                    Wl + wr * LDW + ks + lk * 8);
+                    Wl[pb] + wrr * LDW + ks + lk * 8);
             }
             #pragma unroll
             for (int i = 0; i < FM; i++)
@@ -358,6 +387,7 @@ __global__ __launch_bounds__(256) void k_gemm(
             xr0 = xr0n; xr1 = xr1n;
             if constexpr (XSILU) { ur0 = ur0n; ur1 = ur1n; }
         }
+        pb ^= 1;
     }
 
     // ---- epilogue ----
@@ -609,6 +639,10 @@ void launch_gemm_ex(const WTensor& w, const float* X, const float* X2,
                      ? K / BK
                      : tgt / (n_tiles ? n_tiles : 1);
         if (splitk < 1) splitk = 1;
+        // re-derive so ceil-rounding leaves no empty z-blocks
+        const int steps = K / BK;
+        const int chunks = (steps + splitk - 1) / splitk;
+        splitk = (steps + chunks - 1) / chunks;
     }
     const int k_chunk = ((K / BK + splitk - 1) / splitk) * BK;
     dim3 grid(n_tiles, bm_tiles, splitk), block(256);
