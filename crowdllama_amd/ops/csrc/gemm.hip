@@ -171,10 +171,14 @@ __global__ __launch_bounds__(256) void k_gemm(
     constexpr int WNW = 4 / WMW;                 // waves tiling N
     constexpr int JF = BN / WNW / 16;            // b-fragments per wave
     constexpr int FM = (BM_ / WMW) / 16;         // a-fragments per wave
-    // double-buffered tiles: stage tile t+1 into buf[1-p] while the MFMA
-    // phase reads buf[p] — one barrier per tile, LDS writes overlap MFMA.
-    __shared__ __attribute__((aligned(16))) uint16_t Xl[2][BM_ * LDW];
-    __shared__ __attribute__((aligned(16))) uint16_t Wl[2][BN * LDW];
+    // Decode tiles (BM<=32) double-buffer: stage tile t+1 into buf[1-p]
+    // while the MFMA phase reads buf[p] — one barrier per tile, LDS writes
+    // overlap MFMA. The BM=128 prefill tile stays single-buffered: its
+    // double LDS footprint (74 KB) would halve occupancy, which costs more
+    // than the extra barrier on MFMA-dense prefill tiles.
+    constexpr int NBUF = (BM_ <= 32) ? 2 : 1;
+    __shared__ __attribute__((aligned(16))) uint16_t Xl[NBUF][BM_ * LDW];
+    __shared__ __attribute__((aligned(16))) uint16_t Wl[NBUF][BN * LDW];
 
     const int tid = threadIdx.x;
     const int bn = blockIdx.x, bm = blockIdx.y, bz = blockIdx.z;
@@ -335,30 +339,7 @@ __global__ __launch_bounds__(256) void k_gemm(
         }
     };
 
-    // prologue: tile 0 into buf 0, then prefetch tile 1's raw bytes
-    stage_tile(kb_lo, 0);
-    __syncthreads();
-    {
-        const int kb1 = (kb_lo + BK < kb_hi) ? kb_lo + BK : kb_lo;
-        load_next(kb1);
-        wr = wr_next;
-        if (SMALLM) {
-            xr0 = xr0n; xr1 = xr1n;
-            if constexpr (XSILU) { ur0 = ur0n; ur1 = ur1n; }
-        }
-    }
-
-    int pb = 0;
-    for (int kb = kb_lo; kb < kb_hi; kb += BK) {
-        const bool has_next = (kb + BK) < kb_hi;
-        if (has_next) {
-            // raw loads for tile t+2 issue first: their HBM latency hides
-            // under this iteration's ds_writes + MFMA (T14)
-            const int kb2 = (kb + 2 * BK < kb_hi) ? kb + 2 * BK : kb;
-            load_next(kb2);
-            stage_tile(kb + BK, 1 - pb);   // overlaps MFMA below (other buf)
-        }
-        // ---- MFMA over the resident tile ----
+    auto mfma_tile = [&](int pb) {
         #pragma unroll
         for (int ks = 0; ks < BK; ks += 32) {
             bf16x8 a[FM], b[JF];
@@ -381,13 +362,56 @@ __global__ __launch_bounds__(256) void k_gemm(
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a[i], b[j], acc[i][j], 0, 0, 0);
         }
-        __syncthreads();
-        wr = wr_next;
-        if (SMALLM) {
-            xr0 = xr0n; xr1 = xr1n;
-            if constexpr (XSILU) { ur0 = ur0n; ur1 = ur1n; }
+    };
+
+    if constexpr (NBUF == 1) {
+        // single-buffer: stage -> barrier -> (prefetch t+1 raw) -> MFMA ->
+        // barrier, raw W/X registers one tile ahead (T14)
+        for (int kb = kb_lo; kb < kb_hi; kb += BK) {
+            stage_tile(kb, 0);
+            __syncthreads();
+            const int kbn = (kb + BK < kb_hi) ? kb + BK : kb;
+            load_next(kbn);
+            mfma_tile(0);
+            __syncthreads();
+            wr = wr_next;
+            if (SMALLM) {
+                xr0 = xr0n; xr1 = xr1n;
+                if constexpr (XSILU) { ur0 = ur0n; ur1 = ur1n; }
+            }
         }
-        pb ^= 1;
+    } else {
+        // prologue: tile 0 into buf 0, then prefetch tile 1's raw bytes
+        stage_tile(kb_lo, 0);
+        __syncthreads();
+        {
+            const int kb1 = (kb_lo + BK < kb_hi) ? kb_lo + BK : kb_lo;
+            load_next(kb1);
+            wr = wr_next;
+            if (SMALLM) {
+                xr0 = xr0n; xr1 = xr1n;
+                if constexpr (XSILU) { ur0 = ur0n; ur1 = ur1n; }
+            }
+        }
+        int pb = 0;
+        for (int kb = kb_lo; kb < kb_hi; kb += BK) {
+            const bool has_next = (kb + BK) < kb_hi;
+            if (has_next) {
+                // raw loads for tile t+2 issue first: their HBM latency
+                // hides under this iteration's ds_writes + MFMA (T14)
+                const int kb2 = (kb + 2 * BK < kb_hi) ? kb + 2 * BK : kb;
+                load_next(kb2);
+                stage_tile(kb + BK, 1 - pb);   // overlaps MFMA (other buf)
+            }
+            mfma_tile(pb);
+            __syncthreads();
+            wr = wr_next;
+            if (SMALLM) {
+                xr0 = xr0n; xr1 = xr1n;
+                if constexpr (XSILU) { ur0 = ur0n; ur1 = ur1n; }
+            }
+            pb ^= 1;
+        }
     }
 
     // ---- epilogue ----
