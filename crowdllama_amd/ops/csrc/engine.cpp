@@ -453,6 +453,10 @@ void Engine::alloc_state() {
     const int NH = meta_.heads_l, NKV = meta_.kv_heads_l, D = meta_.head_dim;
     const int F = meta_.ffn_l;
     attn_splits_ = std::max(1, std::min(32, 256 / std::max(1, B * NKV)));
+    if (const char* e = getenv("CLA_ATTN_SPLITS")) {   // on-HW sweeps
+        const int v = atoi(e);
+        if (v >= 1 && v <= 64) attn_splits_ = v;
+    }
     max_pages_ = (cfg_.max_seq + cfg_.page_size - 1) / cfg_.page_size;
     // pool layout: [page][layer][kvh][2][page_size][D] bf16 — one pool, all
     // layers; per-layer base pointer passed at launch.
