@@ -96,6 +96,29 @@ class Gateway:
     def find_best_worker(self, model: str):
         return self.peer.peer_manager.find_best_worker(model)
 
+    MAX_WORKER_ATTEMPTS = 3
+
+    async def infer_with_failover(self, model: str, prompt: str):
+        """Route to the best worker; on failure mark it and retry the
+        next-best (extension — the reference errors out after its single
+        best worker, gateway.go:200-214). Returns the PB response."""
+        pm = self.peer.peer_manager
+        tried: set[str] = set()
+        last: Exception | None = None
+        for _ in range(self.MAX_WORKER_ATTEMPTS):
+            worker = pm.find_best_worker(model, exclude=tried)
+            if worker is None:
+                break
+            try:
+                return await self.request_inference(worker, model, prompt)
+            except Exception as e:  # noqa: BLE001
+                last = e
+                tried.add(worker.peer_id)
+                pm.record_request_failure(worker.peer_id)
+                self.log.warning("worker %s failed (%s), trying next",
+                                 worker.peer_id, e)
+        raise last or ConnectionError(f"no available worker for {model}")
+
     async def request_inference(self, worker, model: str, prompt: str,
                                 stream: bool = False,
                                 timeout: float = 300.0) -> pb.GenerateResponse:
@@ -212,10 +235,8 @@ class Gateway:
             return await self._stream_ndjson(request, worker, model, prompt,
                                              chat=True)
         try:
-            resp = await self.request_inference(worker, model, prompt)
+            resp = await self.infer_with_failover(model, prompt)
         except Exception as e:  # noqa: BLE001
-            self.log.warning("inference via %s failed: %s",
-                             worker.peer_id, e)
             return web.json_response(
                 {"error": f"inference failed: {e}"}, status=500)
         if resp.done_reason == "error":
@@ -250,7 +271,7 @@ class Gateway:
             return await self._stream_ndjson(request, worker, model, prompt,
                                              chat=False)
         try:
-            resp = await self.request_inference(worker, model, prompt)
+            resp = await self.infer_with_failover(model, prompt)
         except Exception as e:  # noqa: BLE001
             return web.json_response({"error": f"inference failed: {e}"},
                                      status=500)

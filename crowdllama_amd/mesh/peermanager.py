@@ -78,12 +78,17 @@ class PeerManager:
 
     # ------------------------------------------------------------ scheduler
 
-    def find_best_worker(self, model: str) -> Resource | None:
+    def find_best_worker(self, model: str,
+                         exclude: set[str] | None = None) -> Resource | None:
         """Max of tokens_throughput/(1+load) over healthy workers supporting
-        the model (reference manager.go:338-387)."""
+        the model (reference manager.go:338-387). `exclude` skips peers that
+        already failed this request (gateway failover — extension, the
+        reference gives up after its single best worker)."""
         best, best_score = None, -1.0
         for pi in self.peers.values():
             r = pi.resource
+            if exclude and r.peer_id in exclude:
+                continue
             if not pi.is_healthy or not r.worker_mode:
                 continue
             if model and model not in r.supported_models:
@@ -92,6 +97,13 @@ class PeerManager:
             if score > best_score:
                 best, best_score = r, score
         return best
+
+    def record_request_failure(self, peer_id: str) -> None:
+        """Count an inference-path failure against the peer's health
+        (same counter the health checker uses, manager.go:536-622)."""
+        pi = self.peers.get(peer_id)
+        if pi is not None:
+            pi.failed_attempts += 1
 
     def get_peer_statistics(self) -> dict:
         healthy = sum(1 for p in self.peers.values() if p.is_healthy)

@@ -327,3 +327,52 @@ def test_streaming_generate_ndjson(mesh_cfg):
             await worker.stop()
             await dht.stop()
     asyncio.run(go())
+
+
+def test_gateway_failover_to_next_worker(mesh_cfg):
+    """If the best-scored worker is unreachable, the gateway retries the
+    next-best instead of erroring (extension over the reference, which gives
+    up after one worker — gateway.go:200-214)."""
+    async def go():
+        from crowdllama_amd.mesh.resource import Resource
+        dht = DHTServer(mesh_cfg("dht"), "CLADHT")
+        dht_port = await dht.start("127.0.0.1", 0)
+        boot = [f"127.0.0.1:{dht_port}"]
+        wcfg = mesh_cfg("worker")
+        wcfg.bootstrap_peers = boot
+        worker = Peer(wcfg, worker_mode=True,
+                      engines={"m": MockEngine("m", response="live-worker")})
+        await worker.start()
+        ccfg = mesh_cfg("consumer")
+        ccfg.bootstrap_peers = boot
+        consumer = Peer(ccfg, worker_mode=False)
+        await consumer.start()
+        gw = Gateway(consumer, ccfg)
+        gw_port = await gw.start(port=0)
+        try:
+            await _poll(lambda: gw.find_best_worker("m") is not None,
+                        desc="worker discovery")
+            # inject a phantom worker that outranks the real one but whose
+            # address refuses connections
+            ghost = Resource(peer_id="CLAGHOST", worker_mode=True,
+                             supported_models=["m"],
+                             tokens_throughput=1e9, load=0.0,
+                             addrs=["127.0.0.1:9"])  # discard port: refused
+            ghost.touch()
+            await consumer.peer_manager.add_or_update_peer(ghost)
+            best = gw.find_best_worker("m")
+            assert best.peer_id == "CLAGHOST"  # ghost outranks the real one
+            status, resp = await _http_json(
+                "POST", f"http://127.0.0.1:{gw_port}/api/chat",
+                {"model": "m", "messages": [{"role": "user", "content": "x"}]})
+            assert status == 200, resp
+            assert resp["message"]["content"] == "live-worker"
+            assert resp["worker_id"] == worker.peer_id
+            # the failure was counted against the ghost
+            assert consumer.peer_manager.peers["CLAGHOST"].failed_attempts >= 1
+        finally:
+            await gw.stop()
+            await consumer.stop()
+            await worker.stop()
+            await dht.stop()
+    asyncio.run(go())
