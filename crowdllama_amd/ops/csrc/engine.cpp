@@ -45,15 +45,12 @@ void launch_gemm_ex(const WTensor&, const float* X, const float* X2, int ldx,
                     bool xsilu, const float* res, float* C, int M, int ldc,
                     hipStream_t);
 bool gemm_uses_splitk(int N, int K, int M);
-void launch_zero4(float* p0, int64_t n0, float* p1, int64_t n1, float* p2,
-                  int64_t n2, float* p3, int64_t n3, hipStream_t);
 void launch_layer_prep(const float* X, const float* gw, float* xn, int B,
                        int K, float eps, float* p0, int64_t n0, float* p1,
                        int64_t n1, float* p2, int64_t n2, float* p3,
                        int64_t n3, hipStream_t);
 void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
                          int K, float eps, hipStream_t);
-void launch_silu_rows(const float* GU, float* out, int M, int F, hipStream_t);
 void launch_rope_prefill(float* qkv, const float* inv_freq,
                          const int32_t* page_table, uint16_t* kv_pool,
                          int slot, int pos0, int M, int NH, int NKV, int D,

@@ -476,24 +476,6 @@ __global__ __launch_bounds__(256) void k_rmsnorm_rows(
     }
 }
 
-// grid M; block 256. act[m][k] = silu(gu[m][k]) * gu[m][F+k]
-__global__ __launch_bounds__(256) void k_silu_rows(
-    const float* __restrict__ GU, float* __restrict__ out, int F) {
-    const int m = blockIdx.x;
-    const float4* g4 = reinterpret_cast<const float4*>(GU + (size_t)m * 2 * F);
-    const float4* u4 = reinterpret_cast<const float4*>(GU + (size_t)m * 2 * F + F);
-    float4* o4 = reinterpret_cast<float4*>(out + (size_t)m * F);
-    const int F4 = F >> 2;
-    for (int k = threadIdx.x; k < F4; k += 256) {
-        const float4 g = g4[k], u = u4[k];
-        float4 o;
-        o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
-        o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
-        o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
-        o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
-        o4[k] = o;
-    }
-}
 
 // Prefill RoPE + KV append over M prompt rows of ONE slot.
 // grid (M, KVH); block 128. Position of row m is pos0 + m.
@@ -703,10 +685,6 @@ void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
                        X, gw, out, K, eps);
 }
 
-void launch_silu_rows(const float* GU, float* out, int M, int F,
-                      hipStream_t stream) {
-    hipLaunchKernelGGL(k_silu_rows, dim3(M), dim3(256), 0, stream, GU, out, F);
-}
 
 void launch_rope_prefill(float* qkv, const float* inv_freq,
                          const int32_t* page_table, uint16_t* kv_pool,

@@ -1035,22 +1035,6 @@ __global__ __launch_bounds__(64) void k_argmax_final(
 
 // Clear up to 4 device regions in one launch (replaces several
 // hipMemsetAsync nodes ahead of split-K GEMMs: each memset costs a
-// ~5 us dispatch at decode batch sizes).
-__global__ __launch_bounds__(256) void k_zero4(
-    float* __restrict__ p0, int n0, float* __restrict__ p1, int n1,
-    float* __restrict__ p2, int n2, float* __restrict__ p3, int n3) {
-    const int tid = blockIdx.x * 256 + threadIdx.x;
-    const int stride = gridDim.x * 256;
-    const float4 z = {0.f, 0.f, 0.f, 0.f};
-    for (int i = tid; i < n0 >> 2; i += stride)
-        reinterpret_cast<float4*>(p0)[i] = z;
-    for (int i = tid; i < n1 >> 2; i += stride)
-        reinterpret_cast<float4*>(p1)[i] = z;
-    for (int i = tid; i < n2 >> 2; i += stride)
-        reinterpret_cast<float4*>(p2)[i] = z;
-    for (int i = tid; i < n3 >> 2; i += stride)
-        reinterpret_cast<float4*>(p3)[i] = z;
-}
 
 // Layer-start prep: rmsnorm the residual rows AND clear the split-K
 // destinations in one launch (blocks 0..B-1 normalize one row each; the
@@ -1117,15 +1101,6 @@ void launch_layer_prep(const float* X, const float* gw, float* xn, int B,
                        p2, (int)n2, p3, (int)n3);
 }
 
-void launch_zero4(float* p0, int64_t n0, float* p1, int64_t n1, float* p2,
-                  int64_t n2, float* p3, int64_t n3, hipStream_t stream) {
-    const int64_t total = (n0 + n1 + n2 + n3) >> 2;
-    int blocks = (int)((total + 255) / 256);
-    if (blocks > 2048) blocks = 2048;
-    if (blocks < 1) blocks = 1;
-    hipLaunchKernelGGL(k_zero4, dim3(blocks), dim3(256), 0, stream,
-                       p0, (int)n0, p1, (int)n1, p2, (int)n2, p3, (int)n3);
-}
 
 // --------------------------------------------------------- launch stubs
 
