@@ -154,7 +154,7 @@ __device__ __forceinline__ void stage_w_slice(
 }  // namespace
 
 template <DT W, int BM_, bool XSILU>
-__global__ __launch_bounds__(BM_ == 16 ? 128 : 256) void k_gemm(
+__global__ __launch_bounds__(256) void k_gemm(
     const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
     const float* __restrict__ X,     // [M][ldx] f32 (gate half when XSILU)
     const float* __restrict__ X2,    // [M][ldx] up half (XSILU only)
@@ -167,16 +167,9 @@ __global__ __launch_bounds__(BM_ == 16 ? 128 : 256) void k_gemm(
     // Wave grid: BM>=32 uses 2x2 (each wave BM/2 rows x 64 cols); BM=16 uses
     // 1x4 (each wave all 16 rows x 32 cols) so no MFMA row is padding when
     // the decode batch is <=16.
-    // BM=16 decode tiles run 2-wave workgroups over a 64-column tile:
-    // twice as many independent WGs at the same wave count, and each
-    // barrier synchronizes 2 waves instead of 4 (PMC showed 53 % wait
-    // cycles at 2 WGs/CU with 4-wave groups).
-    constexpr int NT_ = (BM_ == 16) ? 128 : 256;   // threads per WG
-    constexpr int BN_t = (BM_ == 16) ? 64 : BN;    // n-columns per tile
-    constexpr int NWAVE = NT_ / 64;
     constexpr int WMW = (BM_ == 16) ? 1 : 2;     // waves tiling M
-    constexpr int WNW = NWAVE / WMW;             // waves tiling N
-    constexpr int JF = BN_t / WNW / 16;          // b-fragments per wave
+    constexpr int WNW = 4 / WMW;                 // waves tiling N
+    constexpr int JF = BN / WNW / 16;            // b-fragments per wave
     constexpr int FM = (BM_ / WMW) / 16;         // a-fragments per wave
     // Decode tiles (BM<=32) double-buffer: stage tile t+1 into buf[1-p]
     // while the MFMA phase reads buf[p] — one barrier per tile, LDS writes
@@ -185,12 +178,12 @@ __global__ __launch_bounds__(BM_ == 16 ? 128 : 256) void k_gemm(
     // than the extra barrier on MFMA-dense prefill tiles.
     constexpr int NBUF = (BM_ <= 32) ? 2 : 1;
     __shared__ __attribute__((aligned(16))) uint16_t Xl[NBUF][BM_ * LDW];
-    __shared__ __attribute__((aligned(16))) uint16_t Wl[NBUF][BN_t * LDW];
+    __shared__ __attribute__((aligned(16))) uint16_t Wl[NBUF][BN * LDW];
 
     const int tid = threadIdx.x;
     const int bn = blockIdx.x, bm = blockIdx.y, bz = blockIdx.z;
     const bool splitk = gridDim.z > 1;
-    const int m0 = bm * BM_, n0 = bn * BN_t;
+    const int m0 = bm * BM_, n0 = bn * BN;
     const int kb_lo = bz * k_chunk;
     const int kb_hi = min(kb_lo + k_chunk, K);
     // ceil-rounded split-K chunking can leave trailing z-blocks with an
@@ -358,7 +351,7 @@ __global__ __launch_bounds__(BM_ == 16 ? 128 : 256) void k_gemm(
             }
             #pragma unroll
             for (int j = 0; j < JF; j++) {
-                const int wrr = wn * (BN_t / WNW) + j * 16 + lrow;
+                const int wrr = wn * (BN / WNW) + j * 16 + lrow;
                 b[j] = *reinterpret_cast<const bf16x8*>(
                     Wl[pb] + wrr * LDW + ks + lk * 8);
             }
@@ -430,7 +423,7 @@ __global__ __launch_bounds__(BM_ == 16 ? 128 : 256) void k_gemm(
             if (m >= M) continue;
             #pragma unroll
             for (int j = 0; j < JF; j++) {
-                const int n = n0 + wn * (BN_t / WNW) + j * 16 + lrow;
+                const int n = n0 + wn * (BN / WNW) + j * 16 + lrow;
                 if (n >= N) continue;
                 const size_t idx = (size_t)m * ldc + n;
                 const float rv = (res && (!splitk || bz == 0)) ? res[idx] : 0.f;
@@ -631,8 +624,7 @@ static int splitk_target() {
 
 bool gemm_uses_splitk(int N, int K, int M) {
     if (M > 32) return false;
-    const int bn = (M <= 16) ? 64 : 128;   // BM=16 tiles are 64 cols wide
-    const int n_tiles = (N + bn - 1) / bn;
+    const int n_tiles = (N + BN - 1) / BN;
     const int tgt = splitk_target();
     int splitk = K / BK < tgt / n_tiles ? K / BK : tgt / n_tiles;
     return splitk > 1;
@@ -644,10 +636,8 @@ void launch_gemm_ex(const WTensor& w, const float* X, const float* X2,
     const int N = (int)w.n, K = (int)w.k;
     if (K % BK != 0) throw std::runtime_error("gemm: K must be /64");
     const bool small_m = M <= 32;
-    const bool bm16 = M <= 16;             // 2-wave 64-col tile
-    const int bn = bm16 ? 64 : 128;
     const int bm_tiles = small_m ? 1 : (M + BM - 1) / BM;
-    const int n_tiles = (N + bn - 1) / bn;
+    const int n_tiles = (N + BN - 1) / BN;
     // split-K keeps the chip full when M is small (decode batches): target
     // >=512 workgroups, bounded by the number of K-steps.
     int splitk = 1;
@@ -663,7 +653,7 @@ void launch_gemm_ex(const WTensor& w, const float* X, const float* X2,
         splitk = (steps + chunks - 1) / chunks;
     }
     const int k_chunk = ((K / BK + splitk - 1) / splitk) * BK;
-    dim3 grid(n_tiles, bm_tiles, splitk), block(bm16 ? 128 : 256);
+    dim3 grid(n_tiles, bm_tiles, splitk), block(256);
     // splitk > 1 accumulates with atomicAdd: caller must pre-zero C
     // (gemm_uses_splitk tells it whether that is needed).
     #define GEMM_ONE(WT, BMV, XSV)                                             \
