@@ -100,10 +100,13 @@ class PeerManager:
 
     def record_request_failure(self, peer_id: str) -> None:
         """Count an inference-path failure against the peer's health
-        (same counter the health checker uses, manager.go:536-622)."""
+        (same counter and threshold the health checker uses,
+        manager.go:536-622); a recovered health check resets it."""
         pi = self.peers.get(peer_id)
         if pi is not None:
             pi.failed_attempts += 1
+            if pi.failed_attempts >= self.iv.max_failed_attempts:
+                pi.is_healthy = False
 
     def get_peer_statistics(self) -> dict:
         healthy = sum(1 for p in self.peers.values() if p.is_healthy)

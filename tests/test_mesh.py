@@ -143,3 +143,42 @@ def test_self_never_added(cfg):
         await pm.add_or_update_peer(_mk_resource("ME"))
         assert "ME" not in pm.peers
     run(go())
+
+
+def test_provider_ttl_expiry(cfg, monkeypatch):
+    """Provider records expire without re-provide (reference: the 1 s
+    advertise loop keeps records alive; dead peers age out)."""
+    async def go():
+        monkeypatch.setattr(DHTServer, "PROVIDER_TTL", 0.2)
+        srv = DHTServer(cfg, "CLADHT")
+        port = await srv.start("127.0.0.1", 0)
+        cl = RendezvousClient(f"127.0.0.1:{port}")
+        try:
+            await cl.provide("CLAW1", ["127.0.0.1:1111"], "ns")
+            assert len(await cl.find_providers("ns")) == 1
+            await asyncio.sleep(0.4)
+            assert await cl.find_providers("ns") == []
+        finally:
+            await cl.close()
+            await srv.stop()
+    run(go())
+
+
+def test_request_failure_marks_unhealthy(cfg):
+    """Gateway-reported request failures accumulate into the same counter
+    the health checker uses; max_failed_attempts flips is_healthy."""
+    async def go():
+        disc = Discovery([])
+        pm = PeerManager(disc, cfg.intervals, self_id="CLAME")
+        r = Resource(peer_id="CLAW1", worker_mode=True,
+                     supported_models=["m"], tokens_throughput=100.0)
+        r.touch()
+        await pm.add_or_update_peer(r)
+        assert pm.find_best_worker("m") is not None
+        for _ in range(cfg.intervals.max_failed_attempts - 1):
+            pm.record_request_failure("CLAW1")
+        assert pm.find_best_worker("m") is not None  # below threshold
+        pm.record_request_failure("CLAW1")
+        assert pm.find_best_worker("m") is None
+        assert pm.find_best_worker("m", exclude={"CLAW1"}) is None
+    run(go())
