@@ -769,15 +769,38 @@ __global__ __launch_bounds__(256) void k_attn_decode(
     }
 
     const int cache_end = min(end, pos);  // cached positions only
-    for (int p = start + sub; p < cache_end; p += 16) {
+    // software-pipelined scan: K and V of position p+16 are in flight while
+    // p's dot/softmax math runs (the un-prefetched loop measured ~0.4 TB/s
+    // — load->use serialization, not bandwidth). Prefetch addresses clamp
+    // to the current position so the loads stay branchless (counted
+    // s_waitcnt instead of a vmcnt(0) drain — see the GEMV notes).
+    auto kv_addr = [&](int p) {
         const int page = page_table[(size_t)b * max_pages + p / page_size];
-        const uint16_t* kp = kv_pool + (int64_t)page * page_stride
-                             + ((int64_t)kvh * 2 + 0) * page_size * D
-                             + (int64_t)(p % page_size) * D + d0;
-        uint32_t kw[DPL / 2];
+        return kv_pool + (int64_t)page * page_stride
+               + ((int64_t)kvh * 2) * page_size * D
+               + (int64_t)(p % page_size) * D + d0;
+    };
+    const int64_t voff = (int64_t)page_size * D;   // V plane within the slot
+    const int p0 = start + sub;
+    uint32_t kw[DPL / 2], vw[DPL / 2], kwn[DPL / 2], vwn[DPL / 2];
+    if (p0 < cache_end) {
+        const uint16_t* kp = kv_addr(p0);
         #pragma unroll
-        for (int j = 0; j < DPL / 2; j++)
+        for (int j = 0; j < DPL / 2; j++) {
             kw[j] = reinterpret_cast<const uint32_t*>(kp)[j];
+            vw[j] = reinterpret_cast<const uint32_t*>(kp + voff)[j];
+        }
+    }
+    for (int p = p0; p < cache_end; p += 16) {
+        {
+            const int pn = (p + 16 < cache_end) ? p + 16 : p;
+            const uint16_t* kp = kv_addr(pn);
+            #pragma unroll
+            for (int j = 0; j < DPL / 2; j++) {
+                kwn[j] = reinterpret_cast<const uint32_t*>(kp)[j];
+                vwn[j] = reinterpret_cast<const uint32_t*>(kp + voff)[j];
+            }
+        }
         float kf[DPL];
         #pragma unroll
         for (int j = 0; j < DPL / 2; j++) {
@@ -794,13 +817,6 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             for (int off = 1; off < 16; off <<= 1) d += __shfl_xor(d, off, 64);
             sc[g] = d;
         }
-        const uint16_t* vp = kv_pool + (int64_t)page * page_stride
-                             + ((int64_t)kvh * 2 + 1) * page_size * D
-                             + (int64_t)(p % page_size) * D + d0;
-        uint32_t vw[DPL / 2];
-        #pragma unroll
-        for (int j = 0; j < DPL / 2; j++)
-            vw[j] = reinterpret_cast<const uint32_t*>(vp)[j];
         float vf[DPL];
         #pragma unroll
         for (int j = 0; j < DPL / 2; j++) {
@@ -817,6 +833,8 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             for (int j = 0; j < DPL; j++) o[g][j] = o[g][j] * alpha + w * vf[j];
             m[g] = mn;
         }
+        #pragma unroll
+        for (int j = 0; j < DPL / 2; j++) { kw[j] = kwn[j]; vw[j] = vwn[j]; }
     }
 
     // current token: the quarter whose subsequence covers `pos` computes
