@@ -782,8 +782,7 @@ __global__ __launch_bounds__(256) void k_attn_decode(
     };
     const int64_t voff = (int64_t)page_size * D;   // V plane within the slot
     const int p0 = start + sub;
-    uint32_t kw[DPL / 2], vw[DPL / 2];
-    uint32_t kwn[2][DPL / 2], vwn[2][DPL / 2];     // distance-2 pipeline
+    uint32_t kw[DPL / 2], vw[DPL / 2], kwn[DPL / 2], vwn[DPL / 2];
     if (p0 < cache_end) {
         const uint16_t* kp = kv_addr(p0);
         #pragma unroll
@@ -791,25 +790,15 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             kw[j] = reinterpret_cast<const uint32_t*>(kp)[j];
             vw[j] = reinterpret_cast<const uint32_t*>(kp + voff)[j];
         }
-        const int p1 = (p0 + 16 < cache_end) ? p0 + 16 : p0;
-        const uint16_t* kp1 = kv_addr(p1);
-        #pragma unroll
-        for (int j = 0; j < DPL / 2; j++) {
-            kwn[0][j] = reinterpret_cast<const uint32_t*>(kp1)[j];
-            vwn[0][j] = reinterpret_cast<const uint32_t*>(kp1 + voff)[j];
-        }
     }
-    int pq = 0;   // queue slot holding position p+16
     for (int p = p0; p < cache_end; p += 16) {
         {
-            const int pn = (p + 32 < cache_end) ? p + 32
-                           : ((p + 16 < cache_end) ? p + 16 : p);
+            const int pn = (p + 16 < cache_end) ? p + 16 : p;
             const uint16_t* kp = kv_addr(pn);
             #pragma unroll
             for (int j = 0; j < DPL / 2; j++) {
-                kwn[1 - pq][j] = reinterpret_cast<const uint32_t*>(kp)[j];
-                vwn[1 - pq][j] =
-                    reinterpret_cast<const uint32_t*>(kp + voff)[j];
+                kwn[j] = reinterpret_cast<const uint32_t*>(kp)[j];
+                vwn[j] = reinterpret_cast<const uint32_t*>(kp + voff)[j];
             }
         }
         float kf[DPL];
@@ -845,11 +834,7 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             m[g] = mn;
         }
         #pragma unroll
-        for (int j = 0; j < DPL / 2; j++) {
-            kw[j] = kwn[pq][j];
-            vw[j] = vwn[pq][j];
-        }
-        pq = 1 - pq;
+        for (int j = 0; j < DPL / 2; j++) { kw[j] = kwn[j]; vw[j] = vwn[j]; }
     }
 
     // current token: the quarter whose subsequence covers `pos` computes
