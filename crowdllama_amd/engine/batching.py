@@ -19,7 +19,6 @@ import time
 from concurrent.futures import Future
 from dataclasses import dataclass, field
 
-import numpy as np
 
 from ..quant.gguf import GGUFReader
 from ..tokenizer import NativeTokenizer, Tokenizer

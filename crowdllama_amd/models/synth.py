@@ -27,7 +27,7 @@ import numpy as np
 from ..quant.gguf import GGUFWriter
 from ..quant.kquants import (
     GGMLType, QK_K, Q4_K_BLOCK_BYTES, Q6_K_BLOCK_BYTES, Q8_0_BLOCK,
-    Q8_0_BLOCK_BYTES, dequantize, quantize, row_bytes,
+    Q8_0_BLOCK_BYTES, dequantize, quantize,
 )
 from .presets import ModelConfig, get_preset
 
