@@ -37,6 +37,9 @@ class Gateway:
         self._runner: web.AppRunner | None = None
         self.port: int | None = None
         self._tasks: list[asyncio.Task] = []
+        # local in-flight requests per worker; advertised load is seconds
+        # stale, so the scheduler folds this in (find_best_worker extra_load)
+        self._inflight: dict[str, int] = {}
 
     # ----------------------------------------------------------- lifecycle
 
@@ -93,8 +96,11 @@ class Gateway:
 
     # ------------------------------------------------------------- handlers
 
-    def find_best_worker(self, model: str):
-        return self.peer.peer_manager.find_best_worker(model)
+    def find_best_worker(self, model: str,
+                         exclude: set[str] | None = None):
+        return self.peer.peer_manager.find_best_worker(
+            model, exclude=exclude,
+            extra_load={k: 0.25 * v for k, v in self._inflight.items() if v})
 
     MAX_WORKER_ATTEMPTS = 3
 
@@ -106,17 +112,21 @@ class Gateway:
         tried: set[str] = set()
         last: Exception | None = None
         for _ in range(self.MAX_WORKER_ATTEMPTS):
-            worker = pm.find_best_worker(model, exclude=tried)
+            worker = self.find_best_worker(model, exclude=tried)
             if worker is None:
                 break
+            pid = worker.peer_id
+            self._inflight[pid] = self._inflight.get(pid, 0) + 1
             try:
                 return await self.request_inference(worker, model, prompt)
             except Exception as e:  # noqa: BLE001
                 last = e
-                tried.add(worker.peer_id)
-                pm.record_request_failure(worker.peer_id)
+                tried.add(pid)
+                pm.record_request_failure(pid)
                 self.log.warning("worker %s failed (%s), trying next",
-                                 worker.peer_id, e)
+                                 pid, e)
+            finally:
+                self._inflight[pid] -= 1
         raise last or ConnectionError(f"no available worker for {model}")
 
     async def request_inference(self, worker, model: str, prompt: str,
@@ -188,6 +198,8 @@ class Gateway:
         resp = web.StreamResponse(
             headers={"Content-Type": "application/x-ndjson"})
         await resp.prepare(request)
+        pid = worker.peer_id
+        self._inflight[pid] = self._inflight.get(pid, 0) + 1
         try:
             async for chunk in self.request_inference_stream(worker, model,
                                                              prompt):
@@ -211,6 +223,8 @@ class Gateway:
             await resp.write(json.dumps(
                 {"error": f"inference failed: {e}", "done": True}
             ).encode("utf-8") + b"\n")
+        finally:
+            self._inflight[pid] -= 1
         await resp.write_eof()
         return resp
 

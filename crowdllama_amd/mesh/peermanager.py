@@ -79,11 +79,16 @@ class PeerManager:
     # ------------------------------------------------------------ scheduler
 
     def find_best_worker(self, model: str,
-                         exclude: set[str] | None = None) -> Resource | None:
+                         exclude: set[str] | None = None,
+                         extra_load: dict[str, float] | None = None
+                         ) -> Resource | None:
         """Max of tokens_throughput/(1+load) over healthy workers supporting
-        the model (reference manager.go:338-387). `exclude` skips peers that
-        already failed this request (gateway failover — extension, the
-        reference gives up after its single best worker)."""
+        the model (reference manager.go:338-387). Extensions over the
+        reference: `exclude` skips peers that already failed this request
+        (gateway failover), and `extra_load` adds the caller's own in-flight
+        request count per peer — advertised load is seconds stale, so
+        without it a burst of concurrent requests all tie-break onto the
+        same worker."""
         best, best_score = None, -1.0
         for pi in self.peers.values():
             r = pi.resource
@@ -93,7 +98,10 @@ class PeerManager:
                 continue
             if model and model not in r.supported_models:
                 continue
-            score = r.tokens_throughput / (1.0 + max(0.0, r.load))
+            load = max(0.0, r.load)
+            if extra_load:
+                load += max(0.0, extra_load.get(r.peer_id, 0.0))
+            score = r.tokens_throughput / (1.0 + load)
             if score > best_score:
                 best, best_score = r, score
         return best

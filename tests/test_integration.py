@@ -376,3 +376,50 @@ def test_gateway_failover_to_next_worker(mesh_cfg):
             await worker.stop()
             await dht.stop()
     asyncio.run(go())
+
+
+def test_concurrent_burst_spreads_across_workers(mesh_cfg):
+    """The gateway's local in-flight accounting spreads a concurrent burst
+    across equal workers (advertised load alone is seconds stale, so without
+    it every request in a burst tie-breaks onto one worker)."""
+    async def go():
+        import collections
+        dht = DHTServer(mesh_cfg("dht"), "CLADHT")
+        dht_port = await dht.start("127.0.0.1", 0)
+        boot = [f"127.0.0.1:{dht_port}"]
+        workers = []
+        for i in range(2):
+            wcfg = mesh_cfg(f"worker{i}")
+            wcfg.bootstrap_peers = boot
+            w = Peer(wcfg, worker_mode=True,
+                     engines={"m": MockEngine("m", delay=0.15)})
+            await w.start()
+            workers.append(w)
+        ccfg = mesh_cfg("consumer")
+        ccfg.bootstrap_peers = boot
+        consumer = Peer(ccfg, worker_mode=False)
+        await consumer.start()
+        gw = Gateway(consumer, ccfg)
+        gw_port = await gw.start(port=0)
+        try:
+            await _poll(lambda: len([r for r in
+                        consumer.peer_manager.get_healthy_peers()
+                        if r.worker_mode]) >= 2, desc="both workers")
+            served = collections.Counter()
+
+            async def one(i):
+                status, resp = await _http_json(
+                    "POST", f"http://127.0.0.1:{gw_port}/api/chat",
+                    {"model": "m",
+                     "messages": [{"role": "user", "content": f"r{i}"}]})
+                assert status == 200, resp
+                served[resp["worker_id"]] += 1
+            await asyncio.gather(*[one(i) for i in range(8)])
+            assert len(served) == 2, served  # both workers took traffic
+        finally:
+            await gw.stop()
+            await consumer.stop()
+            for w in workers:
+                await w.stop()
+            await dht.stop()
+    asyncio.run(go())
