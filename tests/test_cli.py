@@ -111,3 +111,69 @@ def test_cli_mesh_end_to_end(tmp_path):
                 p.wait(timeout=10)
             except subprocess.TimeoutExpired:
                 p.kill()
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(180)
+def test_cli_mesh_hip_engine(tmp_path):
+    """Full CLI path with the real HIP engine (+continuous batching) on one
+    GPU: `cla start --worker-mode --engine hip --batch 4`."""
+    from crowdllama_amd.models import synth_path
+    path = synth_path("testllama", scheme="q4_k_m", mode="exact", seed=7)
+    dht_port = _free_port()
+    gw_port = _free_port()
+    boot = f"127.0.0.1:{dht_port}"
+    procs = []
+    try:
+        procs.append(_spawn(["dht", "--port", str(dht_port), "--test-mode",
+                             "--key", str(tmp_path / "dht.key")],
+                            tmp_path, "dht"))
+        time.sleep(0.5)
+        procs.append(_spawn(["start", "--worker-mode", "--engine", "hip",
+                             "--models", "testllama", "--model-path", path,
+                             "--batch", "4", "--test-mode",
+                             "--bootstrap", boot,
+                             "--key", str(tmp_path / "w.key")],
+                            tmp_path, "worker"))
+        procs.append(_spawn(["start", "--test-mode", "--bootstrap", boot,
+                             "--port", str(gw_port),
+                             "--key", str(tmp_path / "c.key")],
+                            tmp_path, "consumer"))
+        deadline = time.time() + 120
+        found = False
+        while time.time() < deadline and not found:
+            for p in procs:
+                assert p.poll() is None, \
+                    (tmp_path / "worker.out").read_text()[-2000:]
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{gw_port}/api/health",
+                        timeout=2) as r:
+                    health = json.load(r)
+                found = any(w["healthy"] and
+                            "testllama" in w["supported_models"] and
+                            ("gfx" in w["gpu_model"] or
+                             "MI" in w["gpu_model"].upper())
+                            for w in health.get("workers", []))
+            except Exception:
+                pass
+            if not found:
+                time.sleep(0.3)
+        assert found, "HIP worker never appeared in gateway health"
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{gw_port}/api/chat",
+            data=json.dumps({"model": "testllama", "messages": [
+                {"role": "user", "content": "abc"}]}).encode(),
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=60) as r:
+            body = json.load(r)
+        assert body["done"] is True
+        assert isinstance(body["message"]["content"], str)
+    finally:
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                p.kill()
