@@ -635,17 +635,19 @@ void launch_gemm_ex(const WTensor& w, const float* X, const float* X2,
                     int ldc, hipStream_t stream) {
     const int N = (int)w.n, K = (int)w.k;
     if (K % BK != 0) throw std::runtime_error("gemm: K must be /64");
-    const bool small_m = M <= 32;
-    const int bm_tiles = small_m ? 1 : (M + BM - 1) / BM;
+    // decode-batch sizing: BM=32 tiles (BM=16 for B<=16) with split-K up
+    // to M=128 — a single 128-row tile grid is only n_tiles workgroups and
+    // leaves the chip mostly idle (B=64 measured 5x slower without this).
+    const bool small_m = M <= 128;
+    const int bm_tiles = small_m ? (M + 31) / 32 : (M + BM - 1) / BM;
     const int n_tiles = (N + BN - 1) / BN;
-    // split-K keeps the chip full when M is small (decode batches): target
-    // >=512 workgroups, bounded by the number of K-steps.
     int splitk = 1;
     if (small_m) {
         const int tgt = splitk_target();
-        splitk = K / BK < tgt / (n_tiles ? n_tiles : 1)
+        const int wgs = n_tiles * bm_tiles;
+        splitk = K / BK < tgt / (wgs ? wgs : 1)
                      ? K / BK
-                     : tgt / (n_tiles ? n_tiles : 1);
+                     : tgt / (wgs ? wgs : 1);
         if (splitk < 1) splitk = 1;
         // re-derive so ceil-rounding leaves no empty z-blocks
         const int steps = K / BK;
