@@ -72,3 +72,40 @@ def test_synthetic_exact_mode(tmp_path):
     with GGUFReader(path) as r:
         qw = r.tensor_f32("blk.0.attn_q.weight")
         assert np.isfinite(qw).all()
+
+
+def test_reader_rejects_bad_magic(tmp_path):
+    p = tmp_path / "bad.gguf"
+    p.write_bytes(b"NOPE" + b"\x00" * 64)
+    with pytest.raises(Exception):
+        GGUFReader(str(p))
+
+
+def test_reader_rejects_truncated(tmp_path):
+    path = str(tmp_path / "t.gguf")
+    w = GGUFWriter(path)
+    w.add("general.architecture", "llama")
+    w.add_tensor("a", (4, 32), GGMLType.F32,
+                 quantize(np.zeros((4, 32), dtype=np.float32),
+                          GGMLType.F32).reshape(4, -1))
+    w.write()
+    data = open(path, "rb").read()
+    trunc = str(tmp_path / "trunc.gguf")
+    open(trunc, "wb").write(data[: len(data) // 2])
+    with pytest.raises(Exception):
+        r = GGUFReader(trunc)
+        # if the header parsed, reading the tensor must still fail
+        r.tensor_data("a")
+
+
+def test_reader_unknown_tensor_errors(tmp_path):
+    path = str(tmp_path / "t.gguf")
+    w = GGUFWriter(path)
+    w.add("general.architecture", "llama")
+    w.add_tensor("a", (2, 32), GGMLType.F32,
+                 quantize(np.zeros((2, 32), dtype=np.float32),
+                          GGMLType.F32).reshape(2, -1))
+    w.write()
+    with GGUFReader(path) as r:
+        with pytest.raises(KeyError):
+            r.tensor_data("nope")
