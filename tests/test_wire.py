@@ -148,3 +148,33 @@ def test_two_frames_back_to_back():
         assert await read_frame(reader) == b"first"
         assert await read_frame(reader) == b"second"
     asyncio.run(run())
+
+
+def test_protocol_negotiation_loopback():
+    """open_protocol/accept_protocol handshake over a real socket, and an
+    unknown protocol is surfaced to the acceptor (reference: libp2p stream
+    protocol IDs, types.go:12-27)."""
+    from crowdllama_amd.mesh.wire import (PROTO_INFERENCE, accept_protocol,
+                                          open_protocol)
+
+    async def run():
+        seen = []
+
+        async def on_conn(reader, writer):
+            seen.append(await accept_protocol(reader))
+            writer.close()
+
+        server = await asyncio.start_server(on_conn, "127.0.0.1", 0)
+        port = server.sockets[0].getsockname()[1]
+        try:
+            _, w = await open_protocol("127.0.0.1", port, PROTO_INFERENCE)
+            w.close()
+            _, w = await open_protocol("127.0.0.1", port, "/bogus/9.9")
+            w.close()
+            await asyncio.sleep(0.2)
+            assert seen[0] == PROTO_INFERENCE
+            assert seen[1] == "/bogus/9.9"   # acceptor sees and can reject
+        finally:
+            server.close()
+            await server.wait_closed()
+    asyncio.run(run())
