@@ -1,6 +1,6 @@
 // MFMA dequant-GEMM for gfx950: C[M,N] = X[M,K] @ W[N,K]^T (+res), where W
 // rows are quantized (Q4_K/Q6_K/Q8_0) or bf16/f16/f32. Used for prompt
-// prefill and batched decode (the GEMV kernels cover decode B<=2).
+// prefill and batched decode (the GEMV kernels cover decode B=1).
 //
 // Structure (guide §5 canonical anatomy, correctness-first tier):
 // 128x128 tile, BK=64, 256 threads = 4 waves as 2x2, each wave a 64x64
