@@ -1,19 +1,24 @@
 """Version info (reference parity: pkg/version/version.go:9-47).
 
 The reference injects Version/CommitHash/BuildDate via ldflags; here the
-commit hash is resolved lazily from git when available so advertised peer
+stamp comes from CROWDLLAMA_VERSION / CROWDLLAMA_COMMIT env (set by the
+Dockerfiles' build args) with a lazy git fallback, so advertised peer
 metadata carries it (reference stamps CommitHash into metadata at
 pkg/peer/peer.go:335).
 """
 
 import functools
+import os
 import subprocess
 
-__version__ = "0.1.0"
+__version__ = os.environ.get("CROWDLLAMA_VERSION", "0.1.0")
 
 
 @functools.lru_cache(maxsize=1)
 def commit_hash() -> str:
+    env = os.environ.get("CROWDLLAMA_COMMIT")
+    if env and env != "unknown":
+        return env
     try:
         out = subprocess.run(
             ["git", "rev-parse", "--short", "HEAD"],
