@@ -182,3 +182,50 @@ def test_request_failure_marks_unhealthy(cfg):
         assert pm.find_best_worker("m") is None
         assert pm.find_best_worker("m", exclude={"CLAW1"}) is None
     run(go())
+
+
+def test_health_check_linear_backoff(cfg):
+    """Failed health checks push the next check out linearly
+    (failed_attempts * backoff_base — reference manager.go:544-548)."""
+    async def go():
+        disc = Discovery([])
+        pm = PeerManager(disc, cfg.intervals, self_id="CLAME")
+        r = Resource(peer_id="CLAW1", worker_mode=True,
+                     supported_models=["m"], tokens_throughput=10.0,
+                     addrs=["127.0.0.1:9"])  # refuses connections
+        r.touch()
+        await pm.add_or_update_peer(r)
+        pi = pm.peers["CLAW1"]
+        t0 = time.time()
+        # emulate two failing health passes (the loop body's except path)
+        for expect_fails in (1, 2):
+            try:
+                await disc.request_metadata(pi.resource.addrs)
+            except Exception:
+                pi.failed_attempts += 1
+                pi.next_health_check = (time.time() +
+                                        pi.failed_attempts *
+                                        cfg.intervals.backoff_base)
+            assert pi.failed_attempts == expect_fails
+        # backoff grows with the failure count
+        assert pi.next_health_check >= t0 + 2 * cfg.intervals.backoff_base
+    run(go())
+
+
+def test_rendezvous_rejects_malformed(cfg):
+    """The rendezvous server answers malformed ops with an error instead of
+    dying (and keeps serving)."""
+    async def go():
+        srv = DHTServer(cfg, "CLADHT")
+        port = await srv.start("127.0.0.1", 0)
+        cl = RendezvousClient(f"127.0.0.1:{port}")
+        try:
+            r = await cl.call({"op": "bogus"})
+            assert r.get("ok") is False
+            r = await cl.call({"op": "provide"})  # missing fields
+            assert r.get("ok") is False
+            assert await cl.ping()  # still alive
+        finally:
+            await cl.close()
+            await srv.stop()
+    run(go())
