@@ -230,7 +230,6 @@ __global__ __launch_bounds__(256) void k_gemm(
     auto stage_tile = [&](int kb_s, int pb) {
         // ---- stage X tile (f32 -> bf16), 8-elem units ----
         if (SMALLM) {
-            uint16_t tmp[8];
             float4 v[2] = {xr0, xr1};
             if constexpr (XSILU) {
                 const float4 u[2] = {ur0, ur1};
@@ -250,7 +249,6 @@ __global__ __launch_bounds__(256) void k_gemm(
                 tmp8[j * 4 + 2] = f32_to_bf16b(v[j].z);
                 tmp8[j * 4 + 3] = f32_to_bf16b(v[j].w);
             }
-            (void)tmp;
             if (!xvalid0) {
                 #pragma unroll
                 for (int j = 0; j < 8; j++) tmp8[j] = 0;
