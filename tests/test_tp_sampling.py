@@ -77,3 +77,37 @@ def test_sample_top_p():
     seen = {sample(logits, temperature=1.0, top_p=0.9, rng=rng)
             for _ in range(200)}
     assert seen <= {0, 1}
+
+
+def test_rolling_rate_window():
+    """RollingRate reports tokens/sec over its sliding window (replaces the
+    reference's hardcoded 150 tok/s advertisement, peer.go:323)."""
+    import time as _t
+    from crowdllama_amd.engine.api import RollingRate
+    rr = RollingRate(window=0.4)
+    assert rr.rate() == 0.0
+    rr.add(100)
+    _t.sleep(0.05)
+    rr.add(100)
+    r = rr.rate()
+    assert r > 100.0, r  # 200 tokens over well under a second
+    _t.sleep(0.6)
+    rr.add(0)  # trigger pruning
+    assert rr.rate() < 10.0  # old events fell out of the window
+
+
+def test_mock_engine_stream_matches_generate():
+    """The default streaming seam reassembles to the one-shot result."""
+    import asyncio
+    from crowdllama_amd.engine.api import MockEngine
+
+    async def go():
+        eng = MockEngine("m", response="alpha beta gamma")
+        full = await eng.generate("x")
+        parts = []
+        async for c in eng.generate_stream("x"):
+            parts.append(c)
+        assert "".join(p.text for p in parts) == full.text
+        assert parts[-1].done_reason == "stop"
+        assert all(not p.done_reason for p in parts[:-1])
+    asyncio.run(go())
