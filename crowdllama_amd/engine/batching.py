@@ -141,13 +141,26 @@ class BatchingHipEngine(EngineBase):
 
     def _loop(self) -> None:
         while not self._stop:
-            self._admit()
-            if self._active == 0:
-                self._wake.wait(timeout=0.05)
-                self._wake.clear()
-                continue
-            self.eng.decode(DECODE_STRIDE)
-            self._harvest()
+            try:
+                self._admit()
+                if self._active == 0:
+                    self._wake.wait(timeout=0.05)
+                    self._wake.clear()
+                    continue
+                self.eng.decode(DECODE_STRIDE)
+                self._harvest()
+            except Exception as e:  # noqa: BLE001
+                # fatal engine error: fail every waiting request instead of
+                # leaving their futures hanging, then keep serving new ones
+                for slot in self._slots:
+                    req, slot.req = slot.req, None
+                    if req is None:
+                        continue
+                    self._active -= 1
+                    req.push(GenerateResult(text="", done_reason="error"))
+                    if not req.future.done():
+                        req.future.set_exception(RuntimeError(
+                            f"batch engine error: {e}"))
 
     # --------------------------------------------------------------- api
 
