@@ -4,6 +4,7 @@ rank per GPU."""
 
 import json
 import os
+import socket
 import subprocess
 import sys
 
@@ -12,15 +13,26 @@ import pytest
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
 def test_bench_rendezvous_world2(tmp_path):
     """bench.py under torch.distributed.run with 2 CPU ranks: ranks must
     rendezvous, barrier, and exit cleanly (no GPU -> null result line)."""
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
+    # dynamic port: a fixed one collides with orphaned rendezvous stores
+    # from interrupted earlier runs and hangs the whole suite
+    port = str(_free_port())
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29517", "bench.py", "--gpus", "2",
+         "--master-port", port, "bench.py", "--gpus", "2",
          "--steps", "2", "--warmup", "0", "--model", "testllama",
          "--scheme", "q8_0"],
         cwd=REPO, env=env, capture_output=True, text=True, timeout=300)
@@ -37,8 +49,8 @@ def test_gloo_allreduce_max():
     import torch.distributed as dist
     if dist.is_initialized():
         pytest.skip("process group already active")
-    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-    os.environ.setdefault("MASTER_PORT", "29518")
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(_free_port())
     dist.init_process_group("gloo", rank=0, world_size=1)
     try:
         t = torch.tensor([1.5], dtype=torch.float64)
