@@ -178,3 +178,19 @@ def test_protocol_negotiation_loopback():
             server.close()
             await server.wait_closed()
     asyncio.run(run())
+
+
+def test_unknown_fields_skipped():
+    """proto3 forward compatibility: decoders skip unknown field numbers so
+    newer peers can extend the schema without breaking old ones."""
+    from crowdllama_amd.mesh.pb import _enc_varint
+
+    base = pb.request_message("m1", "hello").encode()
+    # splice unknown fields into the inner GenerateRequest payload:
+    # re-encode by appending to the OUTER message instead (same skip path)
+    extra = (bytes([99 << 3 | 0]) + _enc_varint(12345)        # field 99 varint
+             + bytes([(100 << 3) | 2]) + _enc_varint(3) + b"xyz")  # field 100 bytes
+    back = pb.BaseMessage.decode(base + extra)
+    assert back.generate_request is not None
+    assert back.generate_request.model == "m1"
+    assert back.generate_request.prompt == "hello"
