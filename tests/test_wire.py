@@ -188,8 +188,8 @@ def test_unknown_fields_skipped():
     base = pb.request_message("m1", "hello").encode()
     # splice unknown fields into the inner GenerateRequest payload:
     # re-encode by appending to the OUTER message instead (same skip path)
-    extra = (bytes([99 << 3 | 0]) + _enc_varint(12345)        # field 99 varint
-             + bytes([(100 << 3) | 2]) + _enc_varint(3) + b"xyz")  # field 100 bytes
+    extra = (_enc_varint(99 << 3 | 0) + _enc_varint(12345)           # varint
+             + _enc_varint((100 << 3) | 2) + _enc_varint(3) + b"xyz")  # bytes
     back = pb.BaseMessage.decode(base + extra)
     assert back.generate_request is not None
     assert back.generate_request.model == "m1"
