@@ -17,7 +17,10 @@ class Intervals:
     discovery: float = 10.0        # manager.go:69 DiscoveryInterval
     advertise: float = 1.0         # peer.go:455 1 s Provide loop
     metadata_update: float = 30.0  # peer.go:29 metadataUpdateInterval
-    metadata_publish: float = 5.0  # main.go:266 5 s PublishMetadata ticker
+    # main.go:266 5 s PublishMetadata ticker — subsumed here: metadata is
+    # served inline over the metadata protocol (no CID publish step), so
+    # this timer only exists for config parity.
+    metadata_publish: float = 5.0
     health_check: float = 20.0     # manager.go:87 HealthCheckInterval
     stale_timeout: float = 60.0    # manager.go:85 PeerStaleTimeout
     cleanup: float = 20.0          # manager.go:522 cleanup loop
