@@ -60,6 +60,14 @@ _reg(ModelConfig("mistral-7b", vocab_size=32000, hidden_size=4096, n_layers=32,
 _reg(ModelConfig("llama3-70b", vocab_size=128256, hidden_size=8192, n_layers=80,
                  n_heads=64, n_kv_heads=8, ffn_hidden=28672,
                  rope_theta=500000.0, max_seq_len=8192))
+# the reference's hardcoded advertised list also names llama-2 models
+# (peer.go:322); MHA (n_kv_heads == n_heads, G=1) is a supported layout
+_reg(ModelConfig("llama-2-7b", vocab_size=32000, hidden_size=4096,
+                 n_layers=32, n_heads=32, n_kv_heads=32, ffn_hidden=11008,
+                 rope_theta=10000.0, max_seq_len=4096))
+_reg(ModelConfig("llama-2-13b", vocab_size=32000, hidden_size=5120,
+                 n_layers=40, n_heads=40, n_kv_heads=40, ffn_hidden=13824,
+                 rope_theta=10000.0, max_seq_len=4096))
 # tiny model for tests (fast CPU generation + load)
 _reg(ModelConfig("testllama", vocab_size=512, hidden_size=256, n_layers=2,
                  n_heads=4, n_kv_heads=2, ffn_hidden=512,
@@ -70,7 +78,6 @@ ALIASES = {
     "llama3:8b": "llama3-8b",
     "llama3:70b": "llama3-70b",
     "mistral:7b": "mistral-7b",
-    "llama-2-7b": "llama3-8b",
 }
 
 
