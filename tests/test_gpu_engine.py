@@ -193,3 +193,27 @@ def test_engine_act_q8_decode(core, tiny_gguf):
     assert rel < 2e-2, rel
     top5 = np.argsort(want)[-5:]
     assert int(np.argmax(got)) in top5
+
+
+def test_engine_mha_g1(core, tmp_path):
+    """MHA layout (n_kv_heads == n_heads, GQA ratio G=1 — the llama-2
+    presets) matches the numpy reference exactly (greedy tokens). Verified
+    on hardware via scripts/g1_check.py before landing."""
+    import numpy as np
+    from crowdllama_amd.engine.ref_numpy import RefLlama
+    from crowdllama_amd.models.presets import ModelConfig
+    from crowdllama_amd.models.synth import write_synthetic_gguf
+
+    cfg = ModelConfig("mha", vocab_size=256, hidden_size=256, n_layers=2,
+                      n_heads=4, n_kv_heads=4, ffn_hidden=512,
+                      rope_theta=10000.0, max_seq_len=256)
+    path = str(tmp_path / "mha.gguf")
+    write_synthetic_gguf(path, cfg, scheme="q4_k_m", mode="exact", seed=3)
+    ec = core.EngineConfig()
+    ec.batch = 1
+    ec.max_seq = 128
+    eng = core.Engine(path, ec)
+    want = RefLlama(path).generate([5, 9, 2, 7], 8)
+    eng.prefill(np.asarray([[5, 9, 2, 7]], dtype=np.int32))
+    eng.decode(7)
+    assert list(eng.gen_tokens(0)) == want
