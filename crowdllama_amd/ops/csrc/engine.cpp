@@ -475,7 +475,6 @@ void Engine::alloc_state() {
     x2_ = (float*)dalloc((size_t)B * H * 4);
     x3_ = (float*)dalloc((size_t)B * H * 4);
     tmp_h_ = (float*)dalloc((size_t)B * H * 4);
-    act_ = (float*)dalloc((size_t)B * F * 4);
     {   // prefill scratch
         const int Mc = cfg_.prefill_chunk;
         const int QKV = (NH + 2 * NKV) * D;
@@ -485,7 +484,6 @@ void Engine::alloc_state() {
         qkvp_ = (float*)dalloc((size_t)Mc * QKV * 4);
         attnp_ = (float*)dalloc((size_t)Mc * NH * D * 4);
         gup_ = (float*)dalloc((size_t)Mc * 2 * F * 4);
-        actp_ = (float*)dalloc((size_t)Mc * F * 4);
         pids_ = (int32_t*)dalloc((size_t)Mc * 4);
         tmp_hp_ = (float*)dalloc((size_t)Mc * H * 4);
     }

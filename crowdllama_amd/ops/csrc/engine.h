@@ -126,7 +126,6 @@ private:
     float* xn_ = nullptr;       // [B][h] (GEMM decode path rmsnorm out)
     float* x2_ = nullptr;       // [B][h] residual ping-pong (split-K GEMM)
     float* x3_ = nullptr;       // [B][h] mid buffer (post-attention)
-    float* act_ = nullptr;      // [B][F]
     // prefill scratch ([Mchunk] rows)
     float* xp_ = nullptr;
     float* xp2_ = nullptr;
@@ -134,7 +133,6 @@ private:
     float* qkvp_ = nullptr;
     float* attnp_ = nullptr;
     float* gup_ = nullptr;
-    float* actp_ = nullptr;
     int32_t* pids_ = nullptr;   // prefill token ids
     float* qkv_ = nullptr;      // [B][(NH+2KV)*D]
     float* attn_out_ = nullptr; // [B][NH*D]
