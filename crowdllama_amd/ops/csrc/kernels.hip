@@ -13,9 +13,10 @@
 //   LDS as f32; block headers ride the same-address broadcast path through
 //   L1/L2 (no cross-lane shuffles needed).
 // - Weights are repacked at upload into split qs/hdr arrays (common.h DT).
-// - Attention: each 16-lane quarter-wave owns a full 128-dim online-softmax
-//   accumulator; 4 positions in flight per wave; split-KV partials combined
-//   by a second small kernel.
+// - Attention: each 16-lane quarter-wave owns a full online-softmax
+//   accumulator (D 64/128); 4 positions in flight per wave with the next
+//   position's K/V prefetched; split-KV partials combined in-kernel by the
+//   last-arriving split (agent-scope ticket + release/acquire fences).
 
 #include "common.h"
 
