@@ -149,14 +149,14 @@ def test_provider_ttl_expiry(cfg, monkeypatch):
     """Provider records expire without re-provide (reference: the 1 s
     advertise loop keeps records alive; dead peers age out)."""
     async def go():
-        monkeypatch.setattr(DHTServer, "PROVIDER_TTL", 0.2)
+        monkeypatch.setattr(DHTServer, "PROVIDER_TTL", 0.6)
         srv = DHTServer(cfg, "CLADHT")
         port = await srv.start("127.0.0.1", 0)
         cl = RendezvousClient(f"127.0.0.1:{port}")
         try:
             await cl.provide("CLAW1", ["127.0.0.1:1111"], "ns")
             assert len(await cl.find_providers("ns")) == 1
-            await asyncio.sleep(0.4)
+            await asyncio.sleep(1.3)
             assert await cl.find_providers("ns") == []
         finally:
             await cl.close()
