@@ -45,26 +45,48 @@ def test_cli_keygen(tmp_path):
     assert "peer id CLA" in out.stdout
 
 
-@pytest.mark.timeout(120)
-def test_cli_mesh_end_to_end(tmp_path):
+def _start_mesh(tmp_path, attempt):
+    """Spawn dht+worker+consumer on freshly probed ports. Returns
+    (procs, gw_port, boot) or Nones if a process lost the port race."""
     dht_port = _free_port()
     gw_port = _free_port()
     boot = f"127.0.0.1:{dht_port}"
-    procs = []
+    procs = [_spawn(["dht", "--port", str(dht_port), "--test-mode",
+                     "--key", str(tmp_path / f"dht{attempt}.key")],
+                    tmp_path, "dht")]
+    time.sleep(0.5)
+    procs.append(_spawn(["start", "--worker-mode", "--engine", "mock",
+                         "--models", "m1", "--test-mode",
+                         "--bootstrap", boot,
+                         "--key", str(tmp_path / f"w{attempt}.key")],
+                        tmp_path, "worker"))
+    procs.append(_spawn(["start", "--test-mode", "--bootstrap", boot,
+                         "--port", str(gw_port),
+                         "--key", str(tmp_path / f"c{attempt}.key")],
+                        tmp_path, "consumer"))
+    time.sleep(0.5)
+    if any(p.poll() is not None for p in procs):
+        # a process lost the probe->bind port race; caller retries
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                p.kill()
+        return None, None, None
+    return procs, gw_port, boot
+
+
+@pytest.mark.timeout(180)
+def test_cli_mesh_end_to_end(tmp_path):
+    procs = None
+    for attempt in range(3):
+        procs, gw_port, boot = _start_mesh(tmp_path, attempt)
+        if procs is not None:
+            break
+    assert procs is not None, "could not bind mesh ports in 3 attempts"
     try:
-        procs.append(_spawn(["dht", "--port", str(dht_port), "--test-mode",
-                             "--key", str(tmp_path / "dht.key")],
-                            tmp_path, "dht"))
-        time.sleep(0.5)
-        procs.append(_spawn(["start", "--worker-mode", "--engine", "mock",
-                             "--models", "m1", "--test-mode",
-                             "--bootstrap", boot,
-                             "--key", str(tmp_path / "w.key")],
-                            tmp_path, "worker"))
-        procs.append(_spawn(["start", "--test-mode", "--bootstrap", boot,
-                             "--port", str(gw_port),
-                             "--key", str(tmp_path / "c.key")],
-                            tmp_path, "consumer"))
         # poll health until the worker is discovered
         deadline = time.time() + 60
         found = False
