@@ -142,25 +142,39 @@ def test_cli_mesh_hip_engine(tmp_path):
     GPU: `cla start --worker-mode --engine hip --batch 4`."""
     from crowdllama_amd.models import synth_path
     path = synth_path("testllama", scheme="q4_k_m", mode="exact", seed=7)
-    dht_port = _free_port()
-    gw_port = _free_port()
-    boot = f"127.0.0.1:{dht_port}"
-    procs = []
-    try:
-        procs.append(_spawn(["dht", "--port", str(dht_port), "--test-mode",
-                             "--key", str(tmp_path / "dht.key")],
-                            tmp_path, "dht"))
+    procs = None
+    for attempt in range(3):
+        dht_port = _free_port()
+        gw_port = _free_port()
+        boot = f"127.0.0.1:{dht_port}"
+        procs = [_spawn(["dht", "--port", str(dht_port), "--test-mode",
+                         "--key", str(tmp_path / f"dht{attempt}.key")],
+                        tmp_path, "dht")]
         time.sleep(0.5)
         procs.append(_spawn(["start", "--worker-mode", "--engine", "hip",
                              "--models", "testllama", "--model-path", path,
                              "--batch", "4", "--test-mode",
                              "--bootstrap", boot,
-                             "--key", str(tmp_path / "w.key")],
+                             "--key", str(tmp_path / f"w{attempt}.key")],
                             tmp_path, "worker"))
         procs.append(_spawn(["start", "--test-mode", "--bootstrap", boot,
                              "--port", str(gw_port),
-                             "--key", str(tmp_path / "c.key")],
+                             "--key", str(tmp_path / f"c{attempt}.key")],
                             tmp_path, "consumer"))
+        time.sleep(0.5)
+        if any(p.poll() is not None for p in procs):  # port race: retry
+            for p in procs:
+                p.terminate()
+            for p in procs:
+                try:
+                    p.wait(timeout=10)
+                except subprocess.TimeoutExpired:
+                    p.kill()
+            procs = None
+            continue
+        break
+    assert procs is not None, "could not bind mesh ports in 3 attempts"
+    try:
         deadline = time.time() + 120
         found = False
         while time.time() < deadline and not found:
