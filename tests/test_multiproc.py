@@ -26,18 +26,24 @@ def test_bench_rendezvous_world2(tmp_path):
     rendezvous, barrier, and exit cleanly (no GPU -> null result line)."""
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
-    # dynamic port: a fixed one collides with orphaned rendezvous stores
-    # from interrupted earlier runs and hangs the whole suite
-    port = str(_free_port())
-    r = subprocess.run(
-        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", port, "bench.py", "--gpus", "2",
-         "--steps", "2", "--warmup", "0", "--model", "testllama",
-         "--scheme", "q8_0"],
-        cwd=REPO, env=env, capture_output=True, text=True, timeout=300)
+    # dynamic port (a fixed one collides with orphaned rendezvous stores
+    # from interrupted runs); retry once — the probe->bind window can still
+    # lose a race with other processes on a busy box
+    lines = []
+    r = None
+    for _ in range(2):
+        port = str(_free_port())
+        r = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", port, "bench.py", "--gpus", "2",
+             "--steps", "2", "--warmup", "0", "--model", "testllama",
+             "--scheme", "q8_0"],
+            cwd=REPO, env=env, capture_output=True, text=True, timeout=300)
+        lines = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
+        if r.returncode == 0 and lines:
+            break
     assert r.returncode == 0, r.stderr[-2000:]
-    lines = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
     assert lines, f"no JSON output: {r.stdout[-500:]} {r.stderr[-500:]}"
     d = json.loads(lines[-1])
     assert d["value"] is None and "no GPU" in d.get("error", "")
