@@ -160,6 +160,15 @@ async def _run_network_status(args) -> None:
     ok = await disco.bootstrap_ok()
     print(f"bootstrap reachable: {ok}")
     if ok:
+        for c in disco.clients:
+            try:
+                st = await c.stats()
+                if st:
+                    print(f"dht {c.addr}: peers={st['known_peers']} "
+                          f"providers={st['providers']} "
+                          f"conns={st['active_conns']}/{st['total_conns']}")
+            except Exception:  # noqa: BLE001
+                pass
         peers = await disco.discover_peers()
         print(f"discovered {len(peers)} peers:")
         for r in peers:

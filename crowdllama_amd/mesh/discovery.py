@@ -76,6 +76,11 @@ class RendezvousClient:
         r = await self.call({"op": "find_providers", "ns": ns, "limit": limit})
         return r.get("providers", []) if r.get("ok") else []
 
+    async def stats(self) -> dict | None:
+        "Rendezvous-server statistics (dht stats op)."
+        r = await self.call({"op": "stats"})
+        return r.get("stats") if r.get("ok") else None
+
     async def find_peer(self, peer_id: str) -> list[str] | None:
         r = await self.call({"op": "find_peer", "peer_id": peer_id})
         return r.get("addrs") if r.get("ok") else None
