@@ -80,8 +80,12 @@ def main() -> None:
     core = get_core()
     if core.device_count() == 0:
         log("no GPU visible — printing a null result")
-        print(json.dumps({"metric": "aggregate_tokens_per_sec", "value": None,
-                          "error": "no GPU"}))
+        if rank == 0:  # contract: exactly one JSON line, from rank 0
+            print(json.dumps({"metric": "aggregate_tokens_per_sec",
+                              "value": None, "error": "no GPU"}),
+                  flush=True)
+        if distributed:
+            dist.destroy_process_group()
         return
 
     ecfg = core.EngineConfig()
