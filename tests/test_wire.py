@@ -194,3 +194,13 @@ def test_unknown_fields_skipped():
     assert back.generate_request is not None
     assert back.generate_request.model == "m1"
     assert back.generate_request.prompt == "hello"
+
+
+def test_read_frame_timeout():
+    """read_frame enforces its deadline on a silent peer (the worker's 5 s
+    inference read deadline — reference peer.go:259-271)."""
+    async def run():
+        reader = asyncio.StreamReader()  # nothing ever fed
+        with pytest.raises(asyncio.TimeoutError):
+            await read_frame(reader, timeout=0.2)
+    asyncio.run(run())
