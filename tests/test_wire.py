@@ -204,3 +204,15 @@ def test_read_frame_timeout():
         with pytest.raises(asyncio.TimeoutError):
             await read_frame(reader, timeout=0.2)
     asyncio.run(run())
+
+
+def test_resource_defaults_isolated():
+    """Mutable defaults must not be shared between Resource instances
+    (reference types_test.go checks field defaults)."""
+    a = Resource(peer_id="A")
+    b = Resource(peer_id="B")
+    a.supported_models.append("m")
+    a.addrs.append("x")
+    assert b.supported_models == []
+    assert b.addrs == []
+    assert a.age() >= 0.0
