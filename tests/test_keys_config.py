@@ -68,3 +68,18 @@ def test_config_from_env(monkeypatch):
     assert cfg.verbose is True
     assert cfg.bootstrap_peers == ["10.0.0.1:9000", "10.0.0.2:9000"]
     assert cfg.test_mode is True
+
+
+def test_test_mode_shrinks_every_timer():
+    """Guard: every interval field must actually shrink (or stay equal for
+    counts) in test mode — catches fields added later but forgotten in
+    Intervals.test_mode() (reference CROWDLLAMA_TEST_MODE semantics)."""
+    from dataclasses import fields
+    from crowdllama_amd.config import Intervals
+    prod, test = Intervals(), Intervals.test_mode()
+    for f in fields(Intervals):
+        p, t = getattr(prod, f.name), getattr(test, f.name)
+        if f.name == "max_failed_attempts":
+            assert t == p  # a count, not a timer
+        else:
+            assert t <= p, f"{f.name}: test {t} > prod {p}"
