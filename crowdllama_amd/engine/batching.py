@@ -61,19 +61,30 @@ class BatchingHipEngine(EngineBase):
         cfg = core.EngineConfig()
         cfg.batch = batch
         cfg.max_seq = max_seq
+        # size the generated-token ring from max_seq so a request's max_new
+        # can never exceed it (round-1 advisor: a request with
+        # max_new > gen_cap saturated the ring and decoded forever)
+        cfg.gen_cap = max_seq
         cfg.device = device
         self.eng = core.Engine(gguf_path, cfg)
+        self.gen_cap = cfg.gen_cap
         with GGUFReader(gguf_path) as r:
             try:
                 self.tok = NativeTokenizer.from_gguf(r)
             except Exception:
+                import logging
+                logging.getLogger("crowdllama_amd.engine").warning(
+                    "native tokenizer unavailable for %s; falling back to "
+                    "the Python tokenizer", gguf_path, exc_info=True)
                 self.tok = Tokenizer.from_gguf(r)
         self._props = core.device_props(device)
         self.batch = batch
         self.max_seq = max_seq
         self.default_max_new = max_new
         self._rate = RollingRate()
-        self._queue: "queue.Queue[_Req]" = queue.Queue()
+        # bounded admission queue: reject instead of piling unbounded work
+        # behind a saturated engine (gateway failover retries elsewhere)
+        self._queue: "queue.Queue[_Req]" = queue.Queue(maxsize=8 * batch)
         self._slots = [_Slot() for _ in range(batch)]
         self._wake = threading.Event()
         self._stop = False
@@ -102,9 +113,9 @@ class BatchingHipEngine(EngineBase):
         for i, slot in enumerate(self._slots):
             req = slot.req
             if req is None:
-                # keep idle slots at position 0 so their (discarded) decode
-                # work never runs off the end of the KV page table
-                self.eng.reset_slot(i)
+                # idle slots are parked (set_slot_active False) when their
+                # request completes: the decode step no longer advances them,
+                # so no per-stride reset is needed
                 continue
             toks = list(self.eng.gen_tokens(i))
             done = False
@@ -130,6 +141,8 @@ class BatchingHipEngine(EngineBase):
                     done_reason=reason))
                 slot.req = None
                 self._active -= 1
+                self.eng.set_slot_active(i, False)
+                self.eng.reset_slot(i)
             elif req.chunks is not None:
                 text = self.tok.decode(toks)
                 delta = text[len(req.sent_text):]
@@ -164,34 +177,40 @@ class BatchingHipEngine(EngineBase):
 
     # --------------------------------------------------------------- api
 
+    def _make_req(self, prompt: str, max_new_tokens: int, **kw) -> _Req:
+        ids = self.tok.encode(prompt)
+        # clamp to both the sequence budget and the engine's generated-token
+        # ring (gen_cap): tokens past gen_cap are never recorded, so a
+        # max_new above it would otherwise decode forever
+        max_new = min(max_new_tokens or self.default_max_new,
+                      self.max_seq - len(ids) - 1, self.gen_cap)
+        if max_new < 1:
+            raise ValueError("prompt exceeds max_seq")
+        return _Req(ids=ids or [self.tok.bos_id], max_new=max_new,
+                    t0=time.monotonic_ns(), **kw)
+
+    def _enqueue(self, req: _Req) -> None:
+        try:
+            self._queue.put_nowait(req)
+        except queue.Full:
+            raise RuntimeError("engine request queue full") from None
+        self._wake.set()
+
     async def generate(self, prompt: str, max_new_tokens: int = 256,
                        temperature: float = 0.0) -> GenerateResult:
         # batched mode is greedy (per-slot host sampling would serialize)
-        ids = self.tok.encode(prompt)
-        max_new = min(max_new_tokens or self.default_max_new,
-                      self.max_seq - len(ids) - 1)
-        if max_new < 1:
-            raise ValueError("prompt exceeds max_seq")
-        req = _Req(ids=ids or [self.tok.bos_id], max_new=max_new,
-                   t0=time.monotonic_ns())
-        self._queue.put(req)
-        self._wake.set()
+        req = self._make_req(prompt, max_new_tokens)
+        self._enqueue(req)
         return await asyncio.wrap_future(req.future)
 
     async def generate_stream(self, prompt: str, max_new_tokens: int = 256,
                               temperature: float = 0.0):
         """Streaming under continuous batching: deltas are emitted as the
         shared decode strides advance this request's slot."""
-        ids = self.tok.encode(prompt)
-        max_new = min(max_new_tokens or self.default_max_new,
-                      self.max_seq - len(ids) - 1)
-        if max_new < 1:
-            raise ValueError("prompt exceeds max_seq")
-        req = _Req(ids=ids or [self.tok.bos_id], max_new=max_new,
-                   t0=time.monotonic_ns(), chunks=asyncio.Queue(),
-                   loop=asyncio.get_running_loop())
-        self._queue.put(req)
-        self._wake.set()
+        req = self._make_req(prompt, max_new_tokens,
+                             chunks=asyncio.Queue(),
+                             loop=asyncio.get_running_loop())
+        self._enqueue(req)
         while True:
             chunk = await req.chunks.get()
             yield chunk

@@ -28,13 +28,19 @@ class HipEngine(EngineBase):
         cfg = core.EngineConfig()
         cfg.batch = 1
         cfg.max_seq = max_seq
+        cfg.gen_cap = max_seq  # ring >= any admissible max_new
         cfg.device = device
         cfg.use_graph = use_graph
         self.eng = core.Engine(gguf_path, cfg)
+        self.gen_cap = cfg.gen_cap
         with GGUFReader(gguf_path) as r:
             try:
                 self.tok = NativeTokenizer.from_gguf(r)
             except Exception:
+                import logging
+                logging.getLogger("crowdllama_amd.engine").warning(
+                    "native tokenizer unavailable for %s; falling back to "
+                    "the Python tokenizer", gguf_path, exc_info=True)
                 self.tok = Tokenizer.from_gguf(r)
         self._props = core.device_props(device)
         self._rate = RollingRate()
@@ -46,14 +52,29 @@ class HipEngine(EngineBase):
 
     # ------------------------------------------------------------ generate
 
+    def _clamp_prompt(self, prompt: str, max_new_tokens: int) -> list:
+        """Validate the (prompt, max_new) budget against max_seq; reject
+        out-of-budget requests instead of silently truncating the prompt
+        from the wrong end (round-1 advisor finding)."""
+        if max_new_tokens >= self.max_seq - 1:
+            raise ValueError(
+                f"max_new_tokens={max_new_tokens} leaves no room for a "
+                f"prompt within max_seq={self.max_seq}")
+        ids = self.tok.encode(prompt)
+        budget = self.max_seq - max_new_tokens - 1
+        if len(ids) > budget:
+            raise ValueError(
+                f"prompt ({len(ids)} tokens) + max_new_tokens"
+                f"({max_new_tokens}) exceeds max_seq={self.max_seq}")
+        return ids or [self.tok.bos_id]
+
     def _generate_sync(self, prompt: str, max_new_tokens: int,
                        temperature: float = 0.0) -> GenerateResult:
         import numpy as np
         with self._lock:
             t0 = time.monotonic_ns()
-            ids = self.tok.encode(prompt)[: self.max_seq - max_new_tokens - 1]
-            if not ids:
-                ids = [self.tok.bos_id]
+            max_new_tokens = min(max_new_tokens, self.gen_cap)
+            ids = self._clamp_prompt(prompt, max_new_tokens)
             self.eng.reset()
             self.eng.prefill(np.asarray([ids], dtype=np.int32))
             n_new = max(1, max_new_tokens)
@@ -112,9 +133,8 @@ class HipEngine(EngineBase):
         import numpy as np
         with self._lock:
             t0 = time.monotonic_ns()
-            ids = self.tok.encode(prompt)[: self.max_seq - max_new_tokens - 1]
-            if not ids:
-                ids = [self.tok.bos_id]
+            max_new_tokens = min(max_new_tokens, self.gen_cap)
+            ids = self._clamp_prompt(prompt, max_new_tokens)
             self.eng.reset()
             self.eng.prefill(np.asarray([ids], dtype=np.int32))
             n_new = max(1, max_new_tokens)

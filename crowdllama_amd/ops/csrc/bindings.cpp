@@ -104,6 +104,7 @@ PYBIND11_MODULE(_core, m) {
              py::arg("gguf_path"), py::arg("config"))
         .def("reset", &Engine::reset)
         .def("reset_slot", &Engine::reset_slot)
+        .def("set_slot_active", &Engine::set_slot_active)
         .def("prefill",
              [](Engine& e, py::array_t<int32_t, py::array::c_style> ids) {
                  if (ids.ndim() != 2)

@@ -2,6 +2,7 @@
 
 #include <algorithm>
 #include <cmath>
+#include <cstdio>
 #include <cstring>
 #include <functional>
 #include <thread>
@@ -38,7 +39,8 @@ void launch_attn_decode(const float* qkv, const float* inv_freq,
                         int64_t page_stride, float scale, hipStream_t);
 void launch_argmax(const float* logits, float* pval, int32_t* pidx,
                    int32_t* cur_ids, int32_t* n_past, int32_t* gen_tokens,
-                   int32_t* gen_count, int B, int V, int gen_cap, hipStream_t);
+                   int32_t* gen_count, const uint8_t* slot_active, int B,
+                   int V, int gen_cap, hipStream_t);
 void launch_gemm(const WTensor&, const float* X, const float* res, float* C,
                  int M, int ldc, hipStream_t);
 void launch_gemm_ex(const WTensor&, const float* X, const float* X2, int ldx,
@@ -51,6 +53,8 @@ void launch_layer_prep(const float* X, const float* gw, float* xn, int B,
                        int64_t n3, hipStream_t);
 void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
                          int K, float eps, hipStream_t);
+void launch_scatter_logits(const float* src, float* dst, int B, int Vl,
+                           int TP, hipStream_t);
 void launch_rope_prefill(float* qkv, const float* inv_freq,
                          const int32_t* page_table, uint16_t* kv_pool,
                          int slot, int pos0, int M, int NH, int NKV, int D,
@@ -491,6 +495,10 @@ void Engine::alloc_state() {
     attn_out_ = (float*)dalloc((size_t)B * NH * D * 4);
     gu_ = (float*)dalloc((size_t)B * 2 * F * 4);
     logits_ = (float*)dalloc((size_t)B * V * 4);
+    if (cfg_.tp_size > 1) {   // compact local slice + rank-major gather buf
+        logits_tp_ = (float*)dalloc((size_t)B * meta_.vocab_l * 4);
+        gather_tp_ = (float*)dalloc((size_t)B * V * 4);
+    }
     part_o_ = (float*)dalloc((size_t)B * NH * attn_splits_ * D * 4);
     part_ml_ = (float*)dalloc((size_t)B * NH * attn_splits_ * 2 * 4);
     attn_tickets_ = (int32_t*)dalloc((size_t)B * NKV * 4);  // zeroed once;
@@ -502,6 +510,8 @@ void Engine::alloc_state() {
     gen_tokens_ = (int32_t*)dalloc((size_t)B * cfg_.gen_cap * 4);
     gen_count_ = (int32_t*)dalloc((size_t)B * 4);
     page_table_ = (int32_t*)dalloc((size_t)B * max_pages_ * 4);
+    slot_active_ = (uint8_t*)dalloc((size_t)B);
+    HIP_CHECK(hipMemset(slot_active_, 1, B));  // all-active by default
     kv_pool_ = (uint16_t*)dalloc((size_t)B * max_pages_ * page_stride_ * 2);
     // rope frequency table
     std::vector<float> invf(D / 2);
@@ -524,6 +534,7 @@ void Engine::reset() {
     HIP_CHECK(hipMemset(n_past_, 0, B * 4));
     HIP_CHECK(hipMemset(gen_count_, 0, B * 4));
     HIP_CHECK(hipMemset(cur_ids_, 0, B * 4));
+    HIP_CHECK(hipMemset(slot_active_, 1, B));
     HIP_CHECK(hipDeviceSynchronize());
 }
 
@@ -559,10 +570,19 @@ static void gemv_pick(bool act_q8, const WTensor& w, int pre,
 }
 
 void Engine::reset_slot(int slot) {
-    const int32_t zero = 0;
-    HIP_CHECK(hipMemcpy(n_past_ + slot, &zero, 4, hipMemcpyHostToDevice));
-    HIP_CHECK(hipMemcpy(gen_count_ + slot, &zero, 4, hipMemcpyHostToDevice));
-    HIP_CHECK(hipMemcpy(cur_ids_ + slot, &zero, 4, hipMemcpyHostToDevice));
+    // async on the engine stream: ordered behind any in-flight decode work,
+    // no host serialization (round-1 advisor: per-stride sync memcpys here
+    // serialized the batcher thread against the stream).
+    HIP_CHECK(hipMemsetAsync(n_past_ + slot, 0, 4, stream_));
+    HIP_CHECK(hipMemsetAsync(gen_count_ + slot, 0, 4, stream_));
+    HIP_CHECK(hipMemsetAsync(cur_ids_ + slot, 0, 4, stream_));
+}
+
+// Park / unpark a slot: idle slots do not advance n_past during shared
+// decode steps (k_argmax_final gate), so the attention scan of an idle
+// slot stays at length 1 instead of creeping with every stride.
+void Engine::set_slot_active(int slot, bool active) {
+    HIP_CHECK(hipMemsetAsync(slot_active_ + slot, active ? 1 : 0, 1, stream_));
 }
 
 void Engine::step(hipStream_t s) {
@@ -654,46 +674,67 @@ void Engine::step(hipStream_t s) {
             }
         }
     }
-    const int64_t voff = (int64_t)cfg_.tp_rank * meta_.vocab_l;
     // GEMM path: the final residual lands in x_ for even layer counts,
     // x2_ for odd (3-buffer rotation); the GEMV path is in-place in x_.
     float* xfinal = x_;
     if (!gemv_path && (meta_.layers & 1)) xfinal = x2_;
+    // TP: each rank computes its compact [B][vocab_l] slice, ONE all-gather
+    // of B*vocab_l moves every slice, then a scatter kernel restores the
+    // [B][V] layout (round-1 review: B separate per-row collectives).
+    float* lg = tp ? logits_tp_ : logits_;
+    const int ldl = tp ? meta_.vocab_l : meta_.vocab;
     if (gemv_path) {
-        gemv_pick(cfg_.act_q8, head_, PRE_RMS, x_, out_norm_, nullptr, logits_ + voff, B,
-                    meta_.vocab, eps, s);
+        gemv_pick(cfg_.act_q8, head_, PRE_RMS, x_, out_norm_, nullptr, lg, B,
+                    ldl, eps, s);
     } else {
         launch_rmsnorm_rows(xfinal, out_norm_, xn_, B, meta_.hidden, eps, s);
-        Proj hp; hp.parts.push_back({head_, 0}); hp.n_total = meta_.vocab_l;
-        // write the local vocab slice at its global offset (ldc = full V)
         bool zero = gemm_uses_splitk((int)head_.n, (int)head_.k, B);
         if (zero)
-            HIP_CHECK(hipMemsetAsync(logits_, 0, (size_t)B * meta_.vocab * 4, s));
-        launch_gemm(head_, xn_, nullptr, logits_ + voff, B, meta_.vocab, s);
+            HIP_CHECK(hipMemsetAsync(lg, 0, (size_t)B * ldl * 4, s));
+        launch_gemm(head_, xn_, nullptr, lg, B, ldl, s);
     }
     if (tp) {
-        // all-gather the vocab slices (in-place: every rank's send buffer is
-        // its slice of the same recv buffer), one row per batch slot
-        for (int b = 0; b < B; b++)
-            NCCL_CHECK(ncclAllGather(logits_ + (size_t)b * meta_.vocab + voff,
-                                     logits_ + (size_t)b * meta_.vocab,
-                                     meta_.vocab_l, ncclFloat,
-                                     (ncclComm_t)comm_, s));
+        NCCL_CHECK(ncclAllGather(logits_tp_, gather_tp_,
+                                 (size_t)B * meta_.vocab_l, ncclFloat,
+                                 (ncclComm_t)comm_, s));
+        launch_scatter_logits(gather_tp_, logits_, B, meta_.vocab_l,
+                              cfg_.tp_size, s);
     }
     launch_argmax(logits_, amax_val_, amax_idx_, cur_ids_, n_past_,
-                  gen_tokens_, gen_count_, B, meta_.vocab, cfg_.gen_cap, s);
+                  gen_tokens_, gen_count_, slot_active_, B, meta_.vocab,
+                  cfg_.gen_cap, s);
 }
 
 void Engine::ensure_graph() {
     if (graph_exec_ || !cfg_.use_graph) return;
     // warm-up eager step is NOT run here; capture directly (all state is
-    // device-resident, shapes static).
+    // device-resident, shapes static). Capture of in-graph RCCL collectives
+    // (tp>1) is the finicky case SURVEY §7.3 flags: on any capture failure
+    // fall back to eager stepping instead of dying.
     hipGraph_t graph = nullptr;
-    HIP_CHECK(hipStreamBeginCapture(stream_, hipStreamCaptureModeThreadLocal));
-    step(stream_);
-    HIP_CHECK(hipStreamEndCapture(stream_, &graph));
-    HIP_CHECK(hipGraphInstantiate(&graph_exec_, graph, nullptr, nullptr, 0));
-    HIP_CHECK(hipGraphDestroy(graph));
+    try {
+        HIP_CHECK(hipStreamBeginCapture(stream_,
+                                        hipStreamCaptureModeThreadLocal));
+        step(stream_);
+        HIP_CHECK(hipStreamEndCapture(stream_, &graph));
+        HIP_CHECK(hipGraphInstantiate(&graph_exec_, graph, nullptr,
+                                      nullptr, 0));
+        HIP_CHECK(hipGraphDestroy(graph));
+    } catch (const std::exception& e) {
+        if (graph) hipGraphDestroy(graph);
+        hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+        hipStreamIsCapturing(stream_, &st);
+        if (st != hipStreamCaptureStatusNone) {
+            hipGraph_t dead = nullptr;
+            hipStreamEndCapture(stream_, &dead);  // discard partial capture
+            if (dead) hipGraphDestroy(dead);
+        }
+        graph_exec_ = nullptr;
+        cfg_.use_graph = false;
+        fprintf(stderr,
+                "[cla engine] hipGraph capture failed (%s); decoding eagerly\n",
+                e.what());
+    }
 }
 
 void Engine::prefill_chunk_pass(int slot, int pos0, int m) {
@@ -772,11 +813,14 @@ void Engine::prefill_slot(int slot, const std::vector<int32_t>& ids) {
     const int32_t npast = len - 1, zero = 0;
     HIP_CHECK(hipMemcpy(n_past_ + slot, &npast, 4, hipMemcpyHostToDevice));
     HIP_CHECK(hipMemcpy(gen_count_ + slot, &zero, 4, hipMemcpyHostToDevice));
+    // prefill implies the slot is (re)activated before its first argmax
+    HIP_CHECK(hipMemsetAsync(slot_active_ + slot, 1, 1, stream_));
     launch_argmax(logits_ + (size_t)slot * meta_.vocab,
                   amax_val_ + (size_t)slot * 64, amax_idx_ + (size_t)slot * 64,
                   cur_ids_ + slot, n_past_ + slot,
                   gen_tokens_ + (size_t)slot * cfg_.gen_cap,
-                  gen_count_ + slot, 1, meta_.vocab, cfg_.gen_cap, stream_);
+                  gen_count_ + slot, slot_active_ + slot, 1, meta_.vocab,
+                  cfg_.gen_cap, stream_);
     HIP_CHECK(hipStreamSynchronize(stream_));
 }
 

@@ -75,6 +75,8 @@ public:
     void reset();
     // Reset one slot (serving: reclaim a finished/idle slot).
     void reset_slot(int slot);
+    // Park / unpark a slot: parked slots do not advance during decode.
+    void set_slot_active(int slot, bool active);
     // Feed prompt tokens (same length for all slots in this call) through
     // the GEMM prefill path; afterwards each slot's first generated token is
     // in gen_tokens[slot][0]. ids is [batch][len] row-major.
@@ -149,6 +151,7 @@ private:
     int32_t* gen_tokens_ = nullptr;  // [B][gen_cap]
     int32_t* gen_count_ = nullptr;   // [B]
     int32_t* page_table_ = nullptr;  // [B][max_pages]
+    uint8_t* slot_active_ = nullptr; // [B] decode-advance gate
     uint16_t* kv_pool_ = nullptr;
     int attn_splits_ = 16;
     int max_pages_ = 0;
@@ -160,6 +163,8 @@ private:
     void* comm_ = nullptr;      // ncclComm_t when tp_size > 1
     float* tmp_h_ = nullptr;    // [B][hidden] all-reduce staging
     float* tmp_hp_ = nullptr;   // [Mchunk][hidden] prefill all-reduce staging
+    float* logits_tp_ = nullptr;  // [B][vocab_l] local slice (tp>1)
+    float* gather_tp_ = nullptr;  // [tp][B][vocab_l] all-gather recv (tp>1)
 };
 
 }  // namespace cla

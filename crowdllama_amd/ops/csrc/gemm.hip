@@ -620,12 +620,29 @@ static int splitk_target() {
     return t;
 }
 
-bool gemm_uses_splitk(int N, int K, int M) {
-    if (M > 32) return false;
+// Single source of truth for the split-K factor: launch_gemm_ex and
+// gemm_uses_splitk (which tells the caller whether C must be pre-zeroed
+// for the atomicAdd accumulation) MUST agree exactly — a mismatch for any
+// M range means accumulating into stale scratch (round-1 advisor finding:
+// M=33..128 prompts silently corrupted activations).
+static int gemm_splitk_factor(int N, int K, int M) {
+    const bool small_m = M <= 128;
+    if (!small_m) return 1;
+    const int bm_tiles = (M + 31) / 32;
     const int n_tiles = (N + BN - 1) / BN;
     const int tgt = splitk_target();
-    int splitk = K / BK < tgt / n_tiles ? K / BK : tgt / n_tiles;
-    return splitk > 1;
+    const int wgs = n_tiles * bm_tiles;
+    int splitk = K / BK < tgt / (wgs ? wgs : 1) ? K / BK : tgt / (wgs ? wgs : 1);
+    if (splitk < 1) splitk = 1;
+    // re-derive so ceil-rounding leaves no empty z-blocks
+    const int steps = K / BK;
+    const int chunks = (steps + splitk - 1) / splitk;
+    splitk = (steps + chunks - 1) / chunks;
+    return splitk;
+}
+
+bool gemm_uses_splitk(int N, int K, int M) {
+    return gemm_splitk_factor(N, K, M) > 1;
 }
 
 void launch_gemm_ex(const WTensor& w, const float* X, const float* X2,
@@ -639,19 +656,7 @@ void launch_gemm_ex(const WTensor& w, const float* X, const float* X2,
     const bool small_m = M <= 128;
     const int bm_tiles = small_m ? (M + 31) / 32 : (M + BM - 1) / BM;
     const int n_tiles = (N + BN - 1) / BN;
-    int splitk = 1;
-    if (small_m) {
-        const int tgt = splitk_target();
-        const int wgs = n_tiles * bm_tiles;
-        splitk = K / BK < tgt / (wgs ? wgs : 1)
-                     ? K / BK
-                     : tgt / (wgs ? wgs : 1);
-        if (splitk < 1) splitk = 1;
-        // re-derive so ceil-rounding leaves no empty z-blocks
-        const int steps = K / BK;
-        const int chunks = (steps + splitk - 1) / splitk;
-        splitk = (steps + chunks - 1) / chunks;
-    }
+    const int splitk = gemm_splitk_factor(N, K, M);
     const int k_chunk = ((K / BK + splitk - 1) / splitk) * BK;
     dim3 grid(n_tiles, bm_tiles, splitk), block(256);
     // splitk > 1 accumulates with atomicAdd: caller must pre-zero C

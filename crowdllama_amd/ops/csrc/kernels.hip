@@ -455,69 +455,57 @@ __global__ __launch_bounds__(256) void k_gemv_q8(
             __syncthreads();
             inv = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
         }
-        // one 32-block per thread per round
-        for (int blk = tid; blk < NB32; blk += 256) {
-            float v[32];
+        // lane-parallel block quantization: 8 lanes per 32-block, one
+        // float4 each, amax/sum via 8-lane shuffles. The round-1 per-thread
+        // version serialized ~120 dependent ops per block on at most half
+        // the threads (PERF_NOTES lever 2).
+        const int jl = tid & 7;   // lane within the block's 8-lane group
+        for (int blk = tid >> 3; blk < NB32; blk += 32) {
+            float4 v;
             if constexpr (P == PRE_SILU) {
-                const float4* g4 = reinterpret_cast<const float4*>(
-                    xin + (size_t)b * 2 * K + blk * 32);
-                const float4* u4 = reinterpret_cast<const float4*>(
-                    xin + (size_t)b * 2 * K + K + blk * 32);
-                #pragma unroll
-                for (int j = 0; j < 8; j++) {
-                    const float4 g = g4[j], u = u4[j];
-                    v[j * 4 + 0] = (g.x / (1.f + __expf(-g.x))) * u.x;
-                    v[j * 4 + 1] = (g.y / (1.f + __expf(-g.y))) * u.y;
-                    v[j * 4 + 2] = (g.z / (1.f + __expf(-g.z))) * u.z;
-                    v[j * 4 + 3] = (g.w / (1.f + __expf(-g.w))) * u.w;
-                }
+                const float4 g = reinterpret_cast<const float4*>(
+                    xin + (size_t)b * 2 * K + blk * 32)[jl];
+                const float4 u = reinterpret_cast<const float4*>(
+                    xin + (size_t)b * 2 * K + K + blk * 32)[jl];
+                v.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+                v.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+                v.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+                v.w = (g.w / (1.f + __expf(-g.w))) * u.w;
             } else {
-                const float4* x4 = reinterpret_cast<const float4*>(
-                    xin + (size_t)b * K + blk * 32);
-                #pragma unroll
-                for (int j = 0; j < 8; j++) {
-                    const float4 g = x4[j];
-                    v[j * 4 + 0] = g.x; v[j * 4 + 1] = g.y;
-                    v[j * 4 + 2] = g.z; v[j * 4 + 3] = g.w;
-                }
+                v = reinterpret_cast<const float4*>(
+                    xin + (size_t)b * K + blk * 32)[jl];
                 if constexpr (P == PRE_RMS) {
-                    const float4* gw4 = reinterpret_cast<const float4*>(
-                        gw + blk * 32);
-                    #pragma unroll
-                    for (int j = 0; j < 8; j++) {
-                        const float4 g = gw4[j];
-                        v[j * 4 + 0] *= inv * g.x; v[j * 4 + 1] *= inv * g.y;
-                        v[j * 4 + 2] *= inv * g.z; v[j * 4 + 3] *= inv * g.w;
-                    }
+                    const float4 g = reinterpret_cast<const float4*>(
+                        gw + blk * 32)[jl];
+                    v.x *= inv * g.x; v.y *= inv * g.y;
+                    v.z *= inv * g.z; v.w *= inv * g.w;
                 }
             }
-            float amax = 0.f;
+            float amax = fmaxf(fmaxf(fabsf(v.x), fabsf(v.y)),
+                               fmaxf(fabsf(v.z), fabsf(v.w)));
             #pragma unroll
-            for (int j = 0; j < 32; j++) amax = fmaxf(amax, fabsf(v[j]));
+            for (int off = 1; off < 8; off <<= 1)
+                amax = fmaxf(amax, __shfl_xor(amax, off, 64));
             const float scale = amax / 127.f;
             const float rinv = amax > 0.f ? 127.f / amax : 0.f;
-            int s0 = 0, s1 = 0;
-            uint32_t packed[8];
-            #pragma unroll
-            for (int j = 0; j < 8; j++) {
-                int q0 = (int)rintf(v[j * 4 + 0] * rinv);
-                int q1 = (int)rintf(v[j * 4 + 1] * rinv);
-                int q2 = (int)rintf(v[j * 4 + 2] * rinv);
-                int q3 = (int)rintf(v[j * 4 + 3] * rinv);
-                if (j < 4) s0 += q0 + q1 + q2 + q3;
-                else s1 += q0 + q1 + q2 + q3;
-                packed[j] = (uint32_t)(q0 & 0xFF) | ((uint32_t)(q1 & 0xFF) << 8)
-                          | ((uint32_t)(q2 & 0xFF) << 16)
-                          | ((uint32_t)(q3 & 0xFF) << 24);
+            const int q0 = (int)rintf(v.x * rinv);
+            const int q1 = (int)rintf(v.y * rinv);
+            const int q2 = (int)rintf(v.z * rinv);
+            const int q3 = (int)rintf(v.w * rinv);
+            const uint32_t packed =
+                (uint32_t)(q0 & 0xFF) | ((uint32_t)(q1 & 0xFF) << 8) |
+                ((uint32_t)(q2 & 0xFF) << 16) | ((uint32_t)(q3 & 0xFF) << 24);
+            int s = q0 + q1 + q2 + q3;
+            s += __shfl_xor(s, 1, 64);
+            s += __shfl_xor(s, 2, 64);  // lanes {0..3}/{4..7} hold 16-sums
+            *reinterpret_cast<uint32_t*>(
+                x8 + (size_t)b * X8B + xpad8(blk * 32) + jl * 4) = packed;
+            if (jl == 0) {
+                xd[(size_t)b * NB32 + blk] = scale;
+                s16f[(size_t)b * NB16 + blk * 2] = scale * (float)s;
+            } else if (jl == 4) {
+                s16f[(size_t)b * NB16 + blk * 2 + 1] = scale * (float)s;
             }
-            int8_t* dst = x8 + (size_t)b * X8B + xpad8(blk * 32);
-            #pragma unroll
-            for (int j = 0; j < 8; j += 4)
-                *reinterpret_cast<uint4*>(dst + j * 4) =
-                    *reinterpret_cast<const uint4*>(&packed[j]);
-            xd[(size_t)b * NB32 + blk] = scale;
-            s16f[(size_t)b * NB16 + blk * 2] = scale * (float)s0;
-            s16f[(size_t)b * NB16 + blk * 2 + 1] = scale * (float)s1;
         }
         if constexpr (P == PRE_RMS) __syncthreads();  // red[] reuse next b
     }
@@ -1027,13 +1015,16 @@ __global__ __launch_bounds__(256) void k_argmax_part(
     }
 }
 
-// stage 2 + state advance: grid B, block 64.
+// stage 2 + state advance: grid B, block 64. Idle slots (slot_active[b]==0)
+// do not advance: their n_past stays pinned so the shared decode step never
+// scans garbage KV for them (serving: slots are parked between requests).
 __global__ __launch_bounds__(64) void k_argmax_final(
     const float* __restrict__ pval, const int32_t* __restrict__ pidx,
     int32_t* __restrict__ cur_ids, int32_t* __restrict__ n_past,
     int32_t* __restrict__ gen_tokens, int32_t* __restrict__ gen_count,
-    int nchunk, int gen_cap) {
+    const uint8_t* __restrict__ slot_active, int nchunk, int gen_cap) {
     const int b = blockIdx.x;
+    if (!slot_active[b]) return;
     const int t = threadIdx.x;
     float v = (t < nchunk) ? pval[(size_t)b * nchunk + t] : -1e30f;
     int i = (t < nchunk) ? pidx[(size_t)b * nchunk + t] : 0;
@@ -1050,6 +1041,28 @@ __global__ __launch_bounds__(64) void k_argmax_final(
         gen_count[b] = gc + 1;
         n_past[b] = n_past[b] + 1;
     }
+}
+
+// TP logits epilogue: undo the rank-major ncclAllGather layout in one
+// launch. src is [TP][B][Vl] (each rank's compact [B][Vl] slice gathered in
+// rank order), dst is [B][TP*Vl] global logits. One collective of B*Vl per
+// step instead of the round-1 per-batch-row loop (B collectives of Vl).
+__global__ __launch_bounds__(256) void k_scatter_logits(
+    const float* __restrict__ src, float* __restrict__ dst, int B, int Vl) {
+    const int r = blockIdx.x, b = blockIdx.y;
+    const int TP = gridDim.x;
+    const float4* s = reinterpret_cast<const float4*>(
+        src + ((size_t)r * B + b) * Vl);
+    float4* d = reinterpret_cast<float4*>(
+        dst + ((size_t)b * TP + r) * Vl);
+    const int n4 = Vl >> 2;
+    for (int i = threadIdx.x; i < n4; i += 256) d[i] = s[i];
+}
+
+void launch_scatter_logits(const float* src, float* dst, int B, int Vl,
+                           int TP, hipStream_t stream) {
+    hipLaunchKernelGGL(k_scatter_logits, dim3(TP, B), dim3(256), 0, stream,
+                       src, dst, B, Vl);
 }
 
 // Clear up to 4 device regions in one launch (replaces several
@@ -1258,14 +1271,14 @@ void launch_attn_combine(const float* part_o, const float* part_ml,
 
 void launch_argmax(const float* logits, float* pval, int32_t* pidx,
                    int32_t* cur_ids, int32_t* n_past, int32_t* gen_tokens,
-                   int32_t* gen_count, int B, int V, int gen_cap,
-                   hipStream_t stream) {
+                   int32_t* gen_count, const uint8_t* slot_active, int B,
+                   int V, int gen_cap, hipStream_t stream) {
     constexpr int NCHUNK = 64;
     hipLaunchKernelGGL(k_argmax_part, dim3(NCHUNK, B), dim3(256), 0, stream,
                        logits, pval, pidx, V, NCHUNK);
     hipLaunchKernelGGL(k_argmax_final, dim3(B), dim3(64), 0, stream,
                        pval, pidx, cur_ids, n_past, gen_tokens, gen_count,
-                       NCHUNK, gen_cap);
+                       slot_active, NCHUNK, gen_cap);
 }
 
 }  // namespace cla

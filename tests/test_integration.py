@@ -228,8 +228,11 @@ def test_worker_failure_and_eviction(mesh_cfg):
             await worker.stop()  # worker dies
             # eventually evicted (stale timeout 20 s in test mode; health
             # check failures accelerate) -> no worker for model
+            # stale-eviction needs >= 20 s (test-mode stale timeout) plus
+            # loop periods; give headroom for a loaded CI box (the 40 s
+            # deadline was flaky when the whole suite ran back-to-back)
             await _poll(lambda: gw.find_best_worker("m") is None,
-                        timeout=40.0, desc="worker eviction")
+                        timeout=90.0, desc="worker eviction")
             status, _ = await _http_json(
                 "POST", f"http://127.0.0.1:{gw_port}/api/chat",
                 {"model": "m", "messages": [{"role": "user", "content": "x"}]})
