@@ -45,6 +45,9 @@ def main() -> None:
                          "single-stream latency point)")
     ap.add_argument("--prompt-len", type=int, default=128)
     ap.add_argument("--max-seq", type=int, default=4096)
+    ap.add_argument("--act-q8", action="store_true",
+                    help="int8-quantized activations on the quantized-weight "
+                         "GEMV path (B=1 v_dot4 path A/B)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -91,6 +94,7 @@ def main() -> None:
     ecfg = core.EngineConfig()
     ecfg.batch = args.batch
     ecfg.max_seq = args.max_seq
+    ecfg.act_q8 = args.act_q8
     ecfg.device = local_rank if core.device_count() > local_rank else 0
     if args.mode == "tp":
         # one worker spanning all ranks: RCCL over xGMI

@@ -125,6 +125,63 @@ def test_engine_long_context_page_crossing(core, tiny_gguf):
     assert int(np.argmax(got)) == int(np.argmax(want))
 
 
+def test_engine_midrange_prompt_splitk(core, tiny_gguf):
+    """Prompt length in 33..128: prefill GEMM chunks hit BM=32 tiles WITH
+    split-K enabled — the range where round 1's gemm_uses_splitk /
+    launch_gemm_ex divergence skipped the C pre-zero and accumulated into
+    stale scratch (advisor finding; fixed by gemm_splitk_factor)."""
+    from crowdllama_amd.engine.ref_numpy import RefLlama
+    import numpy as np
+    cfg = core.EngineConfig()
+    cfg.batch = 1
+    cfg.max_seq = 256
+    cfg.act_q8 = False
+    eng = core.Engine(tiny_gguf, cfg)
+    rng = np.random.default_rng(11)
+    prompt = rng.integers(3, 500, size=70).tolist()
+    eng.prefill(np.asarray([prompt], dtype=np.int32))
+    got = np.asarray(eng.logits(0))
+    ref = RefLlama(tiny_gguf)
+    want = None
+    for t in prompt:
+        want = ref.step(t)
+    rel = np.abs(got - want).max() / (np.abs(want).max() + 1e-9)
+    assert rel < 1e-2, rel
+    assert int(np.argmax(got)) == int(np.argmax(want))
+
+    # back-to-back prefills must not leak split-K partials between passes
+    eng.reset()
+    eng.prefill(np.asarray([prompt], dtype=np.int32))
+    again = np.asarray(eng.logits(0))
+    np.testing.assert_allclose(again, got, rtol=0, atol=0)
+
+
+def test_engine_slot_parking(core, tiny_gguf):
+    """Parked slots do not advance during shared decode steps; unparked
+    slots decode identically to a fresh engine (serving slot lifecycle)."""
+    import numpy as np
+    prompt = [3, 17, 99]
+    cfg = core.EngineConfig()
+    cfg.batch = 2
+    cfg.max_seq = 64
+    eng = core.Engine(tiny_gguf, cfg)
+    eng.prefill(np.asarray([prompt, prompt], dtype=np.int32))
+    eng.set_slot_active(1, False)
+    eng.decode(5)
+    active_toks = list(eng.gen_tokens(0))
+    parked = eng.n_past()
+    assert parked[1] == len(prompt), "parked slot advanced during decode"
+    assert len(active_toks) == 6  # prefill token + 5 decode steps
+
+    cfg1 = core.EngineConfig()
+    cfg1.batch = 1
+    cfg1.max_seq = 64
+    e1 = core.Engine(tiny_gguf, cfg1)
+    e1.prefill(np.asarray([prompt], dtype=np.int32))
+    e1.decode(5)
+    assert active_toks == list(e1.gen_tokens(0))
+
+
 def test_engine_bf16_scheme(core, tmp_path_factory):
     """bf16 weights (the 70B TP dtype) through the same engine."""
     from crowdllama_amd.engine.ref_numpy import RefLlama
