@@ -52,29 +52,24 @@ class HipEngine(EngineBase):
 
     # ------------------------------------------------------------ generate
 
-    def _clamp_prompt(self, prompt: str, max_new_tokens: int) -> list:
-        """Validate the (prompt, max_new) budget against max_seq; reject
-        out-of-budget requests instead of silently truncating the prompt
-        from the wrong end (round-1 advisor finding)."""
-        if max_new_tokens >= self.max_seq - 1:
-            raise ValueError(
-                f"max_new_tokens={max_new_tokens} leaves no room for a "
-                f"prompt within max_seq={self.max_seq}")
+    def _budget(self, prompt: str, max_new_tokens: int):
+        """Clamp max_new to the sequence/ring budget (matching
+        BatchingHipEngine); reject only when the PROMPT itself does not fit
+        — never truncate it silently (round-1 advisor finding)."""
         ids = self.tok.encode(prompt)
-        budget = self.max_seq - max_new_tokens - 1
-        if len(ids) > budget:
+        max_new = min(max_new_tokens, self.max_seq - len(ids) - 1,
+                      self.gen_cap)
+        if max_new < 1:
             raise ValueError(
-                f"prompt ({len(ids)} tokens) + max_new_tokens"
-                f"({max_new_tokens}) exceeds max_seq={self.max_seq}")
-        return ids or [self.tok.bos_id]
+                f"prompt ({len(ids)} tokens) exceeds max_seq={self.max_seq}")
+        return ids or [self.tok.bos_id], max_new
 
     def _generate_sync(self, prompt: str, max_new_tokens: int,
                        temperature: float = 0.0) -> GenerateResult:
         import numpy as np
         with self._lock:
             t0 = time.monotonic_ns()
-            max_new_tokens = min(max_new_tokens, self.gen_cap)
-            ids = self._clamp_prompt(prompt, max_new_tokens)
+            ids, max_new_tokens = self._budget(prompt, max_new_tokens)
             self.eng.reset()
             self.eng.prefill(np.asarray([ids], dtype=np.int32))
             n_new = max(1, max_new_tokens)
@@ -133,8 +128,7 @@ class HipEngine(EngineBase):
         import numpy as np
         with self._lock:
             t0 = time.monotonic_ns()
-            max_new_tokens = min(max_new_tokens, self.gen_cap)
-            ids = self._clamp_prompt(prompt, max_new_tokens)
+            ids, max_new_tokens = self._budget(prompt, max_new_tokens)
             self.eng.reset()
             self.eng.prefill(np.asarray([ids], dtype=np.int32))
             n_new = max(1, max_new_tokens)

@@ -27,6 +27,7 @@ void launch_gemv_q8_test(const void*, const void*, const float*, float*, int,
 double bench_gemv_q8(int, int, int, int, int);
 std::vector<uint8_t> slice_cols_test(int32_t, const uint8_t*, int64_t,
                                      int64_t, int64_t, int64_t);
+std::vector<float> test_rccl_graph_1rank(const std::vector<float>&);
 }
 
 PYBIND11_MODULE(_core, m) {
@@ -150,6 +151,11 @@ PYBIND11_MODULE(_core, m) {
             return d;
         });
 
+    m.def("test_rccl_graph_1rank", [](py::array_t<float> in) {
+        std::vector<float> v(in.data(), in.data() + in.size());
+        auto out = test_rccl_graph_1rank(v);
+        return py::array_t<float>((py::ssize_t)out.size(), out.data());
+    });
     m.def("test_gemm", [](py::array_t<uint8_t> qs, py::array_t<uint8_t> hdr,
                           py::array_t<float, py::array::c_style> x, int dtype,
                           int N, int K) {

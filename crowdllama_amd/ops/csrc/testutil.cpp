@@ -244,3 +244,51 @@ double bench_gemv_g(int dtype, int N, int K, int B, int iters) {
 }
 
 }  // namespace cla
+
+#include <rccl/rccl.h>
+#include <vector>
+
+namespace cla {
+
+// 1-rank RCCL communicator + hipGraph-captured ncclAllReduce/ncclAllGather:
+// validates the graph-capture-of-collectives machinery (SURVEY §7.3 risk)
+// inside a 1-GPU lease. (TP>=2 on one device is impossible: RCCL rejects
+// duplicate devices per communicator — see tests/test_gpu_tp.py.)
+// Returns the buffer after 2 graph replays of (allreduce; allgather-self):
+// with 1 rank both are identity-ish copies, so out == in numerically.
+std::vector<float> test_rccl_graph_1rank(const std::vector<float>& in) {
+    const size_t n = in.size();
+    ncclUniqueId id;
+    if (ncclGetUniqueId(&id) != ncclSuccess)
+        throw std::runtime_error("ncclGetUniqueId failed");
+    ncclComm_t comm;
+    if (ncclCommInitRank(&comm, 1, id, 0) != ncclSuccess)
+        throw std::runtime_error("ncclCommInitRank(1) failed");
+    hipStream_t s;
+    HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+    float *d_a = nullptr, *d_b = nullptr;
+    HIP_CHECK(hipMalloc(&d_a, n * 4));
+    HIP_CHECK(hipMalloc(&d_b, n * 4));
+    HIP_CHECK(hipMemcpy(d_a, in.data(), n * 4, hipMemcpyHostToDevice));
+    hipGraph_t graph = nullptr;
+    hipGraphExec_t exec = nullptr;
+    HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
+    if (ncclAllReduce(d_a, d_b, n, ncclFloat, ncclSum, comm, s) != ncclSuccess)
+        throw std::runtime_error("captured ncclAllReduce failed");
+    if (ncclAllGather(d_b, d_a, n, ncclFloat, comm, s) != ncclSuccess)
+        throw std::runtime_error("captured ncclAllGather failed");
+    HIP_CHECK(hipStreamEndCapture(s, &graph));
+    HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
+    HIP_CHECK(hipGraphDestroy(graph));
+    for (int i = 0; i < 2; i++) HIP_CHECK(hipGraphLaunch(exec, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    std::vector<float> out(n);
+    HIP_CHECK(hipMemcpy(out.data(), d_a, n * 4, hipMemcpyDeviceToHost));
+    hipGraphExecDestroy(exec);
+    hipFree(d_a); hipFree(d_b);
+    hipStreamDestroy(s);
+    ncclCommDestroy(comm);
+    return out;
+}
+
+}  // namespace cla

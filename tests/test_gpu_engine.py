@@ -150,10 +150,12 @@ def test_engine_midrange_prompt_splitk(core, tiny_gguf):
     assert int(np.argmax(got)) == int(np.argmax(want))
 
     # back-to-back prefills must not leak split-K partials between passes
+    # (tolerance, not bitwise: atomicAdd accumulation order varies per run)
     eng.reset()
     eng.prefill(np.asarray([prompt], dtype=np.int32))
     again = np.asarray(eng.logits(0))
-    np.testing.assert_allclose(again, got, rtol=0, atol=0)
+    rel2 = np.abs(again - got).max() / (np.abs(got).max() + 1e-9)
+    assert rel2 < 1e-4, rel2
 
 
 def test_engine_slot_parking(core, tiny_gguf):

@@ -1,8 +1,13 @@
-"""Tensor-parallel engine on hardware, inside a 1-GPU lease: TP=2 with BOTH
-ranks on device 0 (RCCL supports multiple ranks per device). Exercises the
-real collectives (ncclAllReduce after o/down, the batched vocab
-ncclAllGather + scatter) and hipGraph capture of a step containing them —
-the paths round 1 shipped unexecuted.
+"""Tensor-parallel engine on hardware: TP=2 with one rank per GPU.
+
+Measured fact (MI355X, RCCL 2.27.7 / ROCm 7.2): RCCL REJECTS two ranks of
+one communicator on the same device — ncclCommInitRank fails with
+"Duplicate GPU detected : rank 0 and rank 1 both on CUDA device"
+(init.cc:1164) → ncclInvalidUsage. So TP=2 inside a 1-GPU lease is
+impossible at the RCCL layer; on a single-GPU box these tests skip and the
+graph-captured-collective risk is covered by test_rccl_graph_capture_1rank
+below (1-rank communicator: same ncclAllReduce/ncclAllGather call sites and
+hipGraph capture machinery, degenerate exchange).
 
 Reference parity note: the reference has no collectives at all (SURVEY.md
 §2.3 — its only parallelism is DP request scatter, manager.go:338); TP over
@@ -37,7 +42,7 @@ def _rank_main(rank, path, nccl_id, use_graph, conn):
         cfg = core.EngineConfig()
         cfg.batch = 2
         cfg.max_seq = 128
-        cfg.device = 0          # both ranks share the single leased GPU
+        cfg.device = rank % max(1, core.device_count())
         cfg.tp_rank = rank
         cfg.tp_size = 2
         cfg.nccl_id = nccl_id
@@ -69,6 +74,9 @@ def _run_tp2(tp_gguf, use_graph):
     core = get_core()
     if core.device_count() == 0:
         pytest.skip("no GPU")
+    if core.device_count() < 2:
+        pytest.skip("TP=2 needs 2 GPUs: RCCL rejects two ranks on one "
+                    "device (Duplicate GPU detected, ncclInvalidUsage)")
     nccl_id = core.nccl_unique_id()
     ctx = mp.get_context("spawn")
     results = {}
@@ -138,6 +146,21 @@ def test_tp2_matches_tp1(tp_gguf):
                 break
             match += 1
         assert match >= 4, f"slot {s}: {toks0[s]} vs {want_toks[s]}"
+
+
+def test_rccl_graph_capture_1rank():
+    """RCCL collectives inside hipGraph capture, replayed twice — the exact
+    machinery the TP decode step uses (same call sites: ncclAllReduce then
+    ncclAllGather on the engine stream). Runs in a 1-GPU lease via a 1-rank
+    communicator."""
+    from crowdllama_amd.ops import get_core
+    core = get_core()
+    if core.device_count() == 0:
+        pytest.skip("no GPU")
+    rng = np.random.default_rng(5)
+    x = rng.standard_normal(4096).astype(np.float32)
+    out = np.asarray(core.test_rccl_graph_1rank(x))
+    np.testing.assert_allclose(out, x, rtol=0, atol=0)
 
 
 def test_tp2_eager_matches_graph(tp_gguf):
