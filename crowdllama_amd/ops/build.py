@@ -18,7 +18,8 @@ CSRC = os.path.join(OPS_DIR, "csrc")
 ARCH = os.environ.get("CLA_GFX_ARCH", "gfx950")
 
 SOURCES = ["gguf.cpp", "engine.cpp", "testutil.cpp", "bindings.cpp",
-           "kernels.hip", "mfma_probe.hip", "gemm.hip", "tokenizer.cpp"]
+           "kernels.hip", "mfma_probe.hip", "gemm.hip", "gemm_i8.hip",
+           "tokenizer.cpp"]
 
 
 def _ext_suffix() -> str:

@@ -28,6 +28,9 @@ double bench_gemv_q8(int, int, int, int, int);
 std::vector<uint8_t> slice_cols_test(int32_t, const uint8_t*, int64_t,
                                      int64_t, int64_t, int64_t);
 std::vector<float> test_rccl_graph_1rank(const std::vector<float>&);
+void launch_gemm_i8_test(const void*, const void*, const float*, float*,
+                         int, int, int, int, size_t, size_t);
+void launch_mfma_probe_i8_test(const int8_t*, const int8_t*, int32_t*);
 }
 
 PYBIND11_MODULE(_core, m) {
@@ -155,6 +158,22 @@ PYBIND11_MODULE(_core, m) {
         std::vector<float> v(in.data(), in.data() + in.size());
         auto out = test_rccl_graph_1rank(v);
         return py::array_t<float>((py::ssize_t)out.size(), out.data());
+    });
+    m.def("test_gemm_i8", [](py::array_t<uint8_t> qs, py::array_t<uint8_t> hdr,
+                             py::array_t<float, py::array::c_style> x,
+                             int dtype, int N, int K) {
+        const int M = (int)x.shape(0);
+        py::array_t<float> y({M, N});
+        launch_gemm_i8_test(qs.data(), hdr.data(), x.data(),
+                            y.mutable_data(), dtype, M, N, K, qs.nbytes(),
+                            hdr.nbytes());
+        return y;
+    });
+    m.def("test_mfma_probe_i8", [](py::array_t<int8_t> A,
+                                   py::array_t<int8_t> B) {
+        py::array_t<int32_t> C({16, 16});
+        launch_mfma_probe_i8_test(A.data(), B.data(), C.mutable_data());
+        return C;
     });
     m.def("test_gemm", [](py::array_t<uint8_t> qs, py::array_t<uint8_t> hdr,
                           py::array_t<float, py::array::c_style> x, int dtype,

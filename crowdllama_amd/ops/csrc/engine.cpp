@@ -55,6 +55,13 @@ void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
                          int K, float eps, hipStream_t);
 void launch_scatter_logits(const float* src, float* dst, int B, int Vl,
                            int TP, hipStream_t);
+// gemm_i8.hip: int8-activation MFMA path for quantized weights
+void launch_quant_rows(const float* X, int8_t* xq, float* xs, float* xsum,
+                       int M, int K, int ldx, int mode, hipStream_t);
+bool gemm_i8_supported(DT dtype, int M, int K);
+void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xs,
+                    const float* xsum, int ldxq, const float* res, float* C,
+                    int M, int ldc, hipStream_t);
 void launch_rope_prefill(float* qkv, const float* inv_freq,
                          const int32_t* page_table, uint16_t* kv_pool,
                          int slot, int pos0, int M, int NH, int NKV, int D,
@@ -513,6 +520,15 @@ void Engine::alloc_state() {
     slot_active_ = (uint8_t*)dalloc((size_t)B);
     HIP_CHECK(hipMemset(slot_active_, 1, B));  // all-active by default
     kv_pool_ = (uint16_t*)dalloc((size_t)B * max_pages_ * page_stride_ * 2);
+    {   // i8-GEMM activation-quant scratch, sized for the widest input of
+        // any projection at the bigger of decode batch / prefill chunk
+        const size_t maxK = std::max<size_t>(
+            {(size_t)H, (size_t)F, (size_t)NH * D});
+        const size_t maxM = std::max<size_t>(B, cfg_.prefill_chunk);
+        xq_ = (int8_t*)dalloc(maxM * maxK);
+        xs_ = (float*)dalloc(maxM * (maxK / 32) * 4);
+        xsum_ = (float*)dalloc(maxM * (maxK / 32) * 4);
+    }
     // rope frequency table
     std::vector<float> invf(D / 2);
     for (int i = 0; i < D / 2; i++)
@@ -542,21 +558,48 @@ static bool is_quant_dt(DT t) {
     return t == DT::DQ4K || t == DT::DQ6K || t == DT::DQ8;
 }
 
+// Activation-quant scratch for the i8 GEMM path (engine-owned buffers).
+struct QBufs {
+    int8_t* xq = nullptr;
+    float* xs = nullptr;
+    float* xsum = nullptr;
+};
+
 // Launch a projection through the GEMM path: pre-zero C when split-K
 // accumulation is in play (see launch_gemm), then one launch per part.
+// With qb set (cfg.act_q8), quantized-weight parts run the int8-MFMA
+// kernel (gemm_i8.hip): activations are block-quantized once here (silu
+// fused into the quantizer for the down projection), float parts and
+// unsupported shapes fall back to the bf16-staging kernel.
 static void gemm_proj(const Proj& p, const float* X, const float* res,
                       float* C, int M, hipStream_t s,
                       bool pre_zeroed = false, const float* X2 = nullptr,
-                      int ldx = -1, bool xsilu = false) {
+                      int ldx = -1, bool xsilu = false,
+                      const QBufs* qb = nullptr) {
     bool zero = false;
     for (auto& pt : p.parts)
         zero |= gemm_uses_splitk((int)pt.w.n, (int)pt.w.k, M);
     if (zero && !pre_zeroed)
         HIP_CHECK(hipMemsetAsync(C, 0, (size_t)M * p.n_total * 4, s));
-    for (auto& pt : p.parts)
-        launch_gemm_ex(pt.w, X, X2, ldx < 0 ? (int)pt.w.k : ldx, xsilu,
-                       res ? res + pt.row_off : nullptr,
-                       C + pt.row_off, M, (int)p.n_total, s);
+    const int Kq = (int)p.parts[0].w.k;
+    bool any_i8 = false;
+    if (qb)
+        for (auto& pt : p.parts)
+            any_i8 |= gemm_i8_supported(pt.w.dtype, M, (int)pt.w.k);
+    if (any_i8)
+        launch_quant_rows(X, qb->xq, qb->xs, qb->xsum, M, Kq,
+                          ldx < 0 ? Kq : ldx, xsilu ? 1 : 0, s);
+    for (auto& pt : p.parts) {
+        if (qb && gemm_i8_supported(pt.w.dtype, M, (int)pt.w.k)) {
+            launch_gemm_i8(pt.w, qb->xq, qb->xs, qb->xsum, Kq,
+                           res ? res + pt.row_off : nullptr,
+                           C + pt.row_off, M, (int)p.n_total, s);
+        } else {
+            launch_gemm_ex(pt.w, X, X2, ldx < 0 ? (int)pt.w.k : ldx, xsilu,
+                           res ? res + pt.row_off : nullptr,
+                           C + pt.row_off, M, (int)p.n_total, s);
+        }
+    }
 }
 
 // GEMV dispatch honoring the act_q8 config for quantized weights.
@@ -601,6 +644,8 @@ void Engine::step(hipStream_t s) {
     // B=1 only: at B=2 the split-K MFMA GEMM already beats two GEMV
     // passes (measured 340 tok/s GEMV vs ~440 GEMM at B=2)
     const bool gemv_path = B <= 1;
+    const QBufs qbufs{xq_, xs_, xsum_};
+    const QBufs* qb = cfg_.act_q8 ? &qbufs : nullptr;
     int li = 0;
     for (auto& L : layers_) {
         uint16_t* kv_layer = kv_pool_ + (int64_t)li * layer_stride_;
@@ -620,7 +665,8 @@ void Engine::step(hipStream_t s) {
                               gu_, (int64_t)B * 2 * meta_.ffn_l,
                               x3_, (int64_t)B * meta_.hidden,
                               lout, (int64_t)B * meta_.hidden, s);
-            gemm_proj(L.qkv, xn_, nullptr, qkv_, B, s, /*pre_zeroed=*/true);
+            gemm_proj(L.qkv, xn_, nullptr, qkv_, B, s, /*pre_zeroed=*/true,
+                      nullptr, -1, false, qb);
         }
         launch_attn_decode(qkv_, inv_freq_, page_table_, kv_layer, n_past_,
                            part_o_, part_ml_, attn_tickets_, attn_out_, B, NH,
@@ -656,21 +702,25 @@ void Engine::step(hipStream_t s) {
             float* lin = ((li - 1) & 1) ? x2_ : x_;
             float* lout = ((li - 1) & 1) ? x_ : x2_;
             if (tp) {
-                gemm_proj(L.o, attn_out_, r0 ? lin : nullptr, tmp_h_, B, s);
+                gemm_proj(L.o, attn_out_, r0 ? lin : nullptr, tmp_h_, B, s,
+                          false, nullptr, -1, false, qb);
                 allreduce(tmp_h_, x3_, (size_t)B * meta_.hidden);
             } else {
-                gemm_proj(L.o, attn_out_, lin, x3_, B, s, /*pre_zeroed=*/true);
+                gemm_proj(L.o, attn_out_, lin, x3_, B, s, /*pre_zeroed=*/true,
+                          nullptr, -1, false, qb);
             }
             launch_rmsnorm_rows(x3_, L.ffn_norm, xn_, B, meta_.hidden, eps, s);
-            gemm_proj(L.gate_up, xn_, nullptr, gu_, B, s, /*pre_zeroed=*/true);
-            // silu fused into the down GEMM's X staging (gate | up halves)
+            gemm_proj(L.gate_up, xn_, nullptr, gu_, B, s, /*pre_zeroed=*/true,
+                      nullptr, -1, false, qb);
+            // silu fused into the down GEMM's X staging (gate | up halves;
+            // i8 path: fused into the activation quantizer instead)
             if (tp) {
                 gemm_proj(L.down, gu_, r0 ? x3_ : nullptr, tmp_h_, B, s,
-                          false, gu_ + meta_.ffn_l, 2 * meta_.ffn_l, true);
+                          false, gu_ + meta_.ffn_l, 2 * meta_.ffn_l, true, qb);
                 allreduce(tmp_h_, lout, (size_t)B * meta_.hidden);
             } else {
                 gemm_proj(L.down, gu_, x3_, lout, B, s, /*pre_zeroed=*/true,
-                          gu_ + meta_.ffn_l, 2 * meta_.ffn_l, true);
+                          gu_ + meta_.ffn_l, 2 * meta_.ffn_l, true, qb);
             }
         }
     }
@@ -691,7 +741,14 @@ void Engine::step(hipStream_t s) {
         bool zero = gemm_uses_splitk((int)head_.n, (int)head_.k, B);
         if (zero)
             HIP_CHECK(hipMemsetAsync(lg, 0, (size_t)B * ldl * 4, s));
-        launch_gemm(head_, xn_, nullptr, lg, B, ldl, s);
+        if (qb && gemm_i8_supported(head_.dtype, B, (int)head_.k)) {
+            launch_quant_rows(xn_, xq_, xs_, xsum_, B, (int)head_.k,
+                              (int)head_.k, 0, s);
+            launch_gemm_i8(head_, xq_, xs_, xsum_, (int)head_.k, nullptr,
+                           lg, B, ldl, s);
+        } else {
+            launch_gemm(head_, xn_, nullptr, lg, B, ldl, s);
+        }
     }
     if (tp) {
         NCCL_CHECK(ncclAllGather(logits_tp_, gather_tp_,
@@ -745,13 +802,18 @@ void Engine::prefill_chunk_pass(int slot, int pos0, int m) {
     const float eps = meta_.rms_eps;
     const bool tp = cfg_.tp_size > 1;
     const bool r0 = cfg_.tp_rank == 0;
+    const QBufs qbufs{xq_, xs_, xsum_};
+    // i8 path engages automatically for chunks <= 128 rows (same act_q8
+    // quantization semantics as the decode step)
+    const QBufs* qb = cfg_.act_q8 ? &qbufs : nullptr;
     launch_embed(embed_, pids_, xp_, m, s);
     int li = 0;
     for (auto& L : layers_) {
         uint16_t* kv_layer = kv_pool_ + (int64_t)li * layer_stride_;
         li++;
         launch_rmsnorm_rows(xp_, L.attn_norm, xnp_, m, meta_.hidden, eps, s);
-        gemm_proj(L.qkv, xnp_, nullptr, qkvp_, m, s);
+        gemm_proj(L.qkv, xnp_, nullptr, qkvp_, m, s, false, nullptr, -1,
+                  false, qb);
         launch_rope_prefill(qkvp_, inv_freq_, page_table_, kv_layer, slot,
                             pos0, m, NH, NKV, D, cfg_.page_size, max_pages_,
                             page_stride_, s);
@@ -759,22 +821,25 @@ void Engine::prefill_chunk_pass(int slot, int pos0, int m) {
                             m, NH, NKV, D, cfg_.page_size, max_pages_,
                             page_stride_, scale, s);
         if (tp) {
-            gemm_proj(L.o, attnp_, r0 ? xp_ : nullptr, tmp_hp_, m, s);
+            gemm_proj(L.o, attnp_, r0 ? xp_ : nullptr, tmp_hp_, m, s,
+                      false, nullptr, -1, false, qb);
             NCCL_CHECK(ncclAllReduce(tmp_hp_, xp2_, (size_t)m * meta_.hidden,
                                      ncclFloat, ncclSum, (ncclComm_t)comm_, s));
         } else {
-            gemm_proj(L.o, attnp_, xp_, xp2_, m, s);
+            gemm_proj(L.o, attnp_, xp_, xp2_, m, s, false, nullptr, -1,
+                      false, qb);
         }
         launch_rmsnorm_rows(xp2_, L.ffn_norm, xnp_, m, meta_.hidden, eps, s);
-        gemm_proj(L.gate_up, xnp_, nullptr, gup_, m, s);
+        gemm_proj(L.gate_up, xnp_, nullptr, gup_, m, s, false, nullptr, -1,
+                  false, qb);
         if (tp) {
             gemm_proj(L.down, gup_, r0 ? xp2_ : nullptr, tmp_hp_, m, s,
-                      false, gup_ + meta_.ffn_l, 2 * meta_.ffn_l, true);
+                      false, gup_ + meta_.ffn_l, 2 * meta_.ffn_l, true, qb);
             NCCL_CHECK(ncclAllReduce(tmp_hp_, xp_, (size_t)m * meta_.hidden,
                                      ncclFloat, ncclSum, (ncclComm_t)comm_, s));
         } else {
             gemm_proj(L.down, gup_, xp2_, xp_, m, s, false,
-                      gup_ + meta_.ffn_l, 2 * meta_.ffn_l, true);
+                      gup_ + meta_.ffn_l, 2 * meta_.ffn_l, true, qb);
         }
     }
 }

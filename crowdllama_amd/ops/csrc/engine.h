@@ -165,6 +165,9 @@ private:
     float* tmp_hp_ = nullptr;   // [Mchunk][hidden] prefill all-reduce staging
     float* logits_tp_ = nullptr;  // [B][vocab_l] local slice (tp>1)
     float* gather_tp_ = nullptr;  // [tp][B][vocab_l] all-gather recv (tp>1)
+    int8_t* xq_ = nullptr;        // [maxM][maxK] i8-GEMM activation quant
+    float* xs_ = nullptr;         // [maxM][maxK/32] block scales
+    float* xsum_ = nullptr;       // [maxM][maxK/32] dx*sum(qx)
 };
 
 }  // namespace cla

@@ -625,7 +625,7 @@ static int splitk_target() {
 // for the atomicAdd accumulation) MUST agree exactly — a mismatch for any
 // M range means accumulating into stale scratch (round-1 advisor finding:
 // M=33..128 prompts silently corrupted activations).
-static int gemm_splitk_factor(int N, int K, int M) {
+int gemm_splitk_factor(int N, int K, int M) {
     const bool small_m = M <= 128;
     if (!small_m) return 1;
     const int bm_tiles = (M + 31) / 32;

@@ -261,3 +261,38 @@ def test_gemv_q8_path(core, name):
     xq = (np.rint(b * rinv) * (amax / 127.0)).reshape(1, K)
     yref = xq @ wref.T
     np.testing.assert_allclose(y, yref, rtol=2e-4, atol=2e-4)
+
+
+def test_mfma_probe_i8(core):
+    """v_mfma_i32_16x16x32_i8 lane maps (assumed bf16-analogous) vs numpy
+    int32 math on asymmetric inputs — gates the i8 GEMM fragment layout."""
+    rng = np.random.default_rng(9)
+    A = rng.integers(-128, 128, size=(16, 32)).astype(np.int8)
+    B = rng.integers(-128, 128, size=(32, 16)).astype(np.int8)
+    C = np.asarray(core.test_mfma_probe_i8(A, B))
+    want = A.astype(np.int32) @ B.astype(np.int32)
+    np.testing.assert_array_equal(C, want)
+
+
+@pytest.mark.parametrize("name", ["q4k", "q8"])
+@pytest.mark.parametrize("M", [1, 7, 16, 24, 64, 128])
+def test_gemm_i8_path(core, name, M):
+    """int8-activation MFMA GEMM (batched decode path) vs exact numpy
+    emulation of its semantics: per-32 rint-quantized activations times the
+    exactly-dequantized weights (the i32 dot itself is exact)."""
+    dt, quant, dequant, repack_fn = CASES[name]
+    rng = np.random.default_rng(77 + M)
+    N, K = 192, 512  # exercises the N%128 edge tile too
+    w = rng.standard_normal((N, K)).astype(np.float32) * 0.1
+    raw = quant(w)
+    wref = dequant(raw, K).reshape(N, K)
+    x = rng.standard_normal((M, K)).astype(np.float32)
+    qs, hdr = repack_fn(raw.reshape(N, -1), N, K)
+    y = core.test_gemm_i8(np.ascontiguousarray(qs),
+                          np.ascontiguousarray(hdr), x, dt, N, K)
+    b = x.reshape(M, -1, 32)
+    amax = np.abs(b).max(axis=2, keepdims=True)
+    rinv = np.where(amax > 0, 127.0 / np.where(amax == 0, 1, amax), 0.0)
+    xq = (np.rint(b * rinv) * (amax / 127.0)).reshape(M, K)
+    yref = xq @ wref.T
+    np.testing.assert_allclose(y, yref, rtol=3e-4, atol=3e-4)
