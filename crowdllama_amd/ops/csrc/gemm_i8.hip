@@ -23,6 +23,7 @@
 namespace cla {
 
 typedef int v4i __attribute__((ext_vector_type(4)));
+typedef unsigned int u32x4v __attribute__((ext_vector_type(4)));
 
 int gemm_splitk_factor(int N, int K, int M);  // gemm.hip (shared contract)
 
@@ -40,9 +41,9 @@ __device__ __forceinline__ float f16b2f(uint32_t h) {
 // Raw bytes one thread stages per tile (loaded a tile ahead, T14 split).
 template <DT W>
 struct WRawI8 {
-    uint4 q0, q1;   // Q4K uses q0 only (16 B = 32 nibbles)
+    u32x4v q0, q1;  // Q4K uses q0 only (16 B = 32 nibbles)
     uint2 hd;       // Q4K pair header
-    float d0, d1;   // Q8: two block scales
+    float d0;       // Q8: block scale
 };
 
 template <DT W>
@@ -53,16 +54,16 @@ __device__ __forceinline__ void load_w_raw_i8(
         const int sb = kb >> 8, q = (kb & 255) >> 6, p = 2 * q + h;
         const uint8_t* qrow = qs + row * (K / 256) * 128;
         r->q0 = __builtin_nontemporal_load(
-            reinterpret_cast<const uint4*>(qrow) + sb * 8 + p);
+            reinterpret_cast<const u32x4v*>(qrow) + sb * 8 + p);
         r->hd = reinterpret_cast<const uint2*>(
             hdr + row * (K / 256) * 32)[sb * 4 + q];
     } else {  // DQ8
         const int k0 = kb + h * 32;
         const uint8_t* qrow = qs + row * K;
         r->q0 = __builtin_nontemporal_load(
-            reinterpret_cast<const uint4*>(qrow + k0));
+            reinterpret_cast<const u32x4v*>(qrow + k0));
         r->q1 = __builtin_nontemporal_load(
-            reinterpret_cast<const uint4*>(qrow + k0) + 1);
+            reinterpret_cast<const u32x4v*>(qrow + k0) + 1);
         const uint16_t* drow = reinterpret_cast<const uint16_t*>(
             hdr + row * (K / 32) * 2);
         r->d0 = f16b2f(drow[k0 >> 5]);  // both 16B chunks share this block
@@ -166,8 +167,8 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
             }
         } else {
             int8_t* wrow = Wq[pb] + srow * LDQ;
-            *reinterpret_cast<uint4*>(wrow + sh * 32) = wr.q0;
-            *reinterpret_cast<uint4*>(wrow + sh * 32 + 16) = wr.q1;
+            *reinterpret_cast<u32x4v*>(wrow + sh * 32) = wr.q0;
+            *reinterpret_cast<u32x4v*>(wrow + sh * 32 + 16) = wr.q1;
             // h selects one 32-weight half = exactly one 32-block (k0 =
             // kb+h*32); d0 is that block's scale, the min term is 0
             Wsc[pb][sh][srow] = {wr.d0, 0.f};
