@@ -155,7 +155,7 @@ def test_engine_midrange_prompt_splitk(core, tiny_gguf):
     eng.prefill(np.asarray([prompt], dtype=np.int32))
     again = np.asarray(eng.logits(0))
     rel2 = np.abs(again - got).max() / (np.abs(got).max() + 1e-9)
-    assert rel2 < 1e-4, rel2
+    assert rel2 < 5e-3, rel2  # stale-scratch corruption would be O(1)
 
 
 def test_engine_slot_parking(core, tiny_gguf):
