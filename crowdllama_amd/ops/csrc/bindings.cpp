@@ -29,7 +29,7 @@ std::vector<uint8_t> slice_cols_test(int32_t, const uint8_t*, int64_t,
                                      int64_t, int64_t, int64_t);
 std::vector<float> test_rccl_graph_1rank(const std::vector<float>&);
 void launch_gemm_i8_test(const void*, const void*, const float*, float*,
-                         int, int, int, int, size_t, size_t);
+                         int, int, int, int, size_t, size_t, int);
 void launch_mfma_probe_i8_test(const int8_t*, const int8_t*, int32_t*);
 }
 
@@ -161,14 +161,15 @@ PYBIND11_MODULE(_core, m) {
     });
     m.def("test_gemm_i8", [](py::array_t<uint8_t> qs, py::array_t<uint8_t> hdr,
                              py::array_t<float, py::array::c_style> x,
-                             int dtype, int N, int K) {
+                             int dtype, int N, int K, int force_splitk) {
         const int M = (int)x.shape(0);
         py::array_t<float> y({M, N});
         launch_gemm_i8_test(qs.data(), hdr.data(), x.data(),
                             y.mutable_data(), dtype, M, N, K, qs.nbytes(),
-                            hdr.nbytes());
+                            hdr.nbytes(), force_splitk);
         return y;
-    });
+    }, py::arg("qs"), py::arg("hdr"), py::arg("x"), py::arg("dtype"),
+       py::arg("N"), py::arg("K"), py::arg("force_splitk") = 0);
     m.def("test_mfma_probe_i8", [](py::array_t<int8_t> A,
                                    py::array_t<int8_t> B) {
         py::array_t<int32_t> C({16, 16});

@@ -61,7 +61,7 @@ void launch_quant_rows(const float* X, int8_t* xq, float* xs, float* xsum,
 bool gemm_i8_supported(DT dtype, int M, int K);
 void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xs,
                     const float* xsum, int ldxq, const float* res, float* C,
-                    int M, int ldc, hipStream_t);
+                    int M, int ldc, hipStream_t, int force_splitk = 0);
 void launch_rope_prefill(float* qkv, const float* inv_freq,
                          const int32_t* page_table, uint16_t* kv_pool,
                          int slot, int pos0, int M, int NH, int NKV, int D,

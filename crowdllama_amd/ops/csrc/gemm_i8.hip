@@ -146,42 +146,47 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         }
     };
 
-    // ---- scale prefetch (global -> regs, one tile ahead) ----
+    // ---- scale prefetch (global -> regs, two tiles ahead) ----
     // W headers: per fragment j, the (col, q-group) pair header. Q4K: uint2
     // {d,dmin | sc/mn x2} covers BOTH K=32 halves. Q8: one u32 = two f16 d.
-    uint2 hd_c[JF], hd_n[JF];
-    float4 dx_c[2], dx_n[2], sm_c[2], sm_n[2];
+    // Parity-indexed slots: tile t uses slot t&1; the slot is refilled with
+    // tile t+2's scales right after t consumes it (never overwrites t+1's).
+    uint2 hd_s[2][JF];
+    float4 dx_s[2][2], sm_s[2][2];
     int64_t hcol[JF];
     #pragma unroll
     for (int j = 0; j < JF; j++) {
         const int64_t gn = (int64_t)n0 + wn * (BN / WNW) + j * 16 + lrow;
         hcol[j] = gn < N ? gn : N - 1;
     }
-    auto load_scales = [&](int kb, uint2* hd, float4* dx, float4* sm) {
+    // Exactly this many VMEM instructions per wave per load_scales call
+    // (every wave issues the same count — vmcnt literals depend on it).
+    constexpr int RSC = JF + 4;
+    auto load_scales = [&](int kb, int slot) {
         #pragma unroll
         for (int j = 0; j < JF; j++) {
             if constexpr (W == DT::DQ4K) {
                 const int sb = kb >> 8, q = (kb & 255) >> 6;
-                hd[j] = *reinterpret_cast<const uint2*>(
+                hd_s[slot][j] = *reinterpret_cast<const uint2*>(
                     hdr + hcol[j] * hdr_rb + sb * 32 + q * 8);
             } else {
-                hd[j].x = *reinterpret_cast<const uint32_t*>(
+                hd_s[slot][j].x = *reinterpret_cast<const uint32_t*>(
                     hdr + hcol[j] * hdr_rb + (kb >> 5) * 2);
-                hd[j].y = 0;
+                hd_s[slot][j].y = 0;
             }
         }
         const int kg = kb >> 5;
         const int mrow = m0 + wm * 16 + lk * 4;
         #pragma unroll
         for (int k2 = 0; k2 < 2; k2++) {
-            dx[k2] = *reinterpret_cast<const float4*>(
+            dx_s[slot][k2] = *reinterpret_cast<const float4*>(
                 xsT + (size_t)(kg + k2) * M4 + mrow);
-            sm[k2] = *reinterpret_cast<const float4*>(
+            sm_s[slot][k2] = *reinterpret_cast<const float4*>(
                 xsumT + (size_t)(kg + k2) * M4 + mrow);
         }
     };
 
-    auto mfma_tile = [&](int pb) {
+    auto mfma_tile = [&](int pb, int slot) {
         // raw W fragment bytes: one b64 per j serves both K=32 halves (Q4K)
         long rawj[JF][(W == DT::DQ4K) ? 1 : 2];
         #pragma unroll
@@ -201,7 +206,7 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         for (int kb2 = 0; kb2 < 2; kb2++) {
             const long a = *reinterpret_cast<const long*>(
                 Xl[pb] + (wm * 16 + lrow) * LDXQ + kb2 * 32 + lk * 8);
-            const float4 dx4 = dx_c[kb2], sm4 = sm_c[kb2];
+            const float4 dx4 = dx_s[slot][kb2], sm4 = sm_s[slot][kb2];
             #pragma unroll
             for (int j = 0; j < JF; j++) {
                 long b;
@@ -210,14 +215,14 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
                     b = (kb2 == 0)
                             ? (rawj[j][0] & 0x0F0F0F0F0F0F0F0FLL)
                             : ((rawj[j][0] >> 4) & 0x0F0F0F0F0F0F0F0FLL);
-                    const float dd = f16b2f(hd_c[j].x & 0xFFFF);
-                    const float dmin = f16b2f(hd_c[j].x >> 16);
-                    const uint32_t y = hd_c[j].y;
+                    const float dd = f16b2f(hd_s[slot][j].x & 0xFFFF);
+                    const float dmin = f16b2f(hd_s[slot][j].x >> 16);
+                    const uint32_t y = hd_s[slot][j].y;
                     d = dd * (float)((y >> (16 * kb2)) & 0xFF);
                     m = dmin * (float)((y >> (16 * kb2 + 8)) & 0xFF);
                 } else {
                     b = rawj[j][kb2];
-                    d = f16b2f((hd_c[j].x >> (16 * kb2)) & 0xFFFF);
+                    d = f16b2f((hd_s[slot][j].x >> (16 * kb2)) & 0xFFFF);
                     m = 0.f;
                 }
                 v4i c = {0, 0, 0, 0};
@@ -230,55 +235,43 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         }
     };
 
-    auto rot_scales = [&] {
-        #pragma unroll
-        for (int j = 0; j < JF; j++) hd_c[j] = hd_n[j];
-        #pragma unroll
-        for (int k2 = 0; k2 < 2; k2++) {
-            dx_c[k2] = dx_n[k2];
-            sm_c[k2] = sm_n[k2];
-        }
-    };
-
     // ---- prologue: tile 0 staged, tile 1 in flight ----
+    const int kb_last = kb_hi - BK;          // all tiles full (K%BK==0)
+    auto clamp_kb = [&](int kb) { return kb <= kb_last ? kb : kb_last; };
     load_x(kb_lo);
-    load_scales(kb_lo, hd_c, dx_c, sm_c);
+    load_scales(kb_lo, 0);
     issue_w_glds(kb_lo, 0);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     write_x(0);
-    const int kb1 = (kb_lo + BK < kb_hi) ? kb_lo + BK : kb_lo;
-    load_x(kb1);
-    load_scales(kb1, hd_n, dx_n, sm_n);
-    issue_w_glds(kb1, 1);
+    load_x(clamp_kb(kb_lo + BK));
+    issue_w_glds(clamp_kb(kb_lo + BK), 1);
+    load_scales(clamp_kb(kb_lo + BK), 1);
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
 
-    int pb = 0;
+    // steady state at iteration t (buffer pb, scale slot t&1):
+    //   buf[pb] landed for every wave; buf[1-pb] DMA in flight;
+    //   xregs hold X(t+1); scale slots hold t (t&1) and t+1 (1-(t&1)).
+    // Tail iterations clamp their t+2 prefetches to the last tile (the
+    // re-staged bytes are never read) so the wait/barrier pattern stays
+    // uniform with no divergent branches around loads.
+    int pb = 0, slot = 0;
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
-        const bool has1 = (kb + BK) < kb_hi;
-        const bool has2 = (kb + 2 * BK) < kb_hi;
-        if (has1) write_x(1 - pb);           // X of tile t+1 (regs ready)
-        mfma_tile(pb);
-        if (has2) {
-            const int kb2 = kb + 2 * BK;
-            load_x(kb2);
-            load_scales(kb2, hd_n, dx_n, sm_n);
-        }
+        write_x(1 - pb);                     // X of tile t+1 (regs ready)
+        mfma_tile(pb, slot);
+        if (kb + BK >= kb_hi) break;         // last tile: no more staging
+        const int kb2 = clamp_kb(kb + 2 * BK);
+        load_x(kb2);
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();        // buf[pb] fully consumed
-        if (has2) issue_w_glds(kb + 2 * BK, pb);
-        if (has1) {
-            // own t+1 glds (and all older reg loads) landed; only the
-            // NGL t+2 glds may stay in flight across this barrier
-            if (has2)
-                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NGL) : "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-            __builtin_amdgcn_s_barrier();    // everyone's t+1 landed
-            if (has2) rot_scales();
-            else rot_scales();
-        }
+        issue_w_glds(kb2, pb);
+        load_scales(kb2, slot);              // slot t&1 just consumed
+        // retire everything older than [t+2 glds + t+2 scale loads]:
+        // exactly the t+1 glds (and the older t+2 xregs loads)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NGL + RSC) : "memory");
+        __builtin_amdgcn_s_barrier();        // everyone's t+1 landed
         pb ^= 1;
+        slot ^= 1;
     }
 
     // ---- epilogue (same contract as gemm.hip k_gemm) ----
@@ -373,14 +366,15 @@ bool gemm_i8_supported(DT dtype, int M, int K) {
 
 void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xs,
                     const float* xsum, int ldxq, const float* res, float* C,
-                    int M, int ldc, hipStream_t stream) {
+                    int M, int ldc, hipStream_t stream, int force_splitk) {
     const int N = (int)w.n, K = (int)w.k;
     if (!gemm_i8_supported(w.dtype, M, K))
         throw std::runtime_error("gemm_i8: unsupported dtype/shape");
     const bool bm16 = M <= 16;
     const int bm_tiles = bm16 ? 1 : (M + 31) / 32;
     const int n_tiles = (N + BN - 1) / BN;
-    const int splitk = gemm_splitk_factor(N, K, M);
+    const int splitk =
+        force_splitk > 0 ? force_splitk : gemm_splitk_factor(N, K, M);
     const int k_chunk = ((K / BK + splitk - 1) / splitk) * BK;
     dim3 grid(n_tiles, bm_tiles, splitk), block(256);
     #define GI8_ONE(WT, BMV)                                                   \
@@ -440,7 +434,8 @@ void launch_mfma_probe_i8_test(const int8_t* A, const int8_t* B, int32_t* C) {
 // Full path: quantize X rows, run the i8 GEMM (host buffers in/out).
 void launch_gemm_i8_test(const void* qs, const void* hdr, const float* x,
                          float* y, int dtype, int M, int N, int K,
-                         size_t qs_bytes, size_t hdr_bytes) {
+                         size_t qs_bytes, size_t hdr_bytes,
+                         int force_splitk) {
     const DT dt = static_cast<DT>(dtype);
     const int M4 = (M + 3) & ~3;
     void *d_qs = nullptr, *d_hdr = nullptr;
@@ -462,9 +457,11 @@ void launch_gemm_i8_test(const void* qs, const void* hdr, const float* x,
     launch_quant_rows(d_x, d_xq, d_xs, d_xsum, M, K, K, 0, nullptr);
     WTensor w;
     w.dtype = dt; w.n = N; w.k = K; w.qs = d_qs; w.hdr = d_hdr;
-    if (gemm_splitk_factor(N, K, M) > 1)
+    const int sk = force_splitk > 0 ? force_splitk : gemm_splitk_factor(N, K, M);
+    if (sk > 1)
         HIP_CHECK(hipMemset(d_y, 0, (size_t)M * N * 4));
-    launch_gemm_i8(w, d_xq, d_xs, d_xsum, K, nullptr, d_y, M, N, nullptr);
+    launch_gemm_i8(w, d_xq, d_xs, d_xsum, K, nullptr, d_y, M, N, nullptr,
+                   force_splitk);
     HIP_CHECK(hipDeviceSynchronize());
     HIP_CHECK(hipMemcpy(y, d_y, (size_t)M * N * 4, hipMemcpyDeviceToHost));
     (void)hipFree(d_qs); (void)hipFree(d_hdr); (void)hipFree(d_x);

@@ -275,21 +275,27 @@ def test_mfma_probe_i8(core):
 
 
 @pytest.mark.parametrize("name", ["q4k", "q8"])
-@pytest.mark.parametrize("M", [1, 7, 16, 24, 64, 128])
-def test_gemm_i8_path(core, name, M):
+@pytest.mark.parametrize("M,K,sk", [(1, 512, 0), (7, 512, 0), (16, 512, 0),
+                                    (16, 2048, 2), (24, 512, 0),
+                                    (64, 2048, 4), (128, 512, 0),
+                                    (16, 4096, 1)])
+def test_gemm_i8_path(core, name, M, K, sk):
     """int8-activation MFMA GEMM (batched decode path) vs exact numpy
     emulation of its semantics: per-32 rint-quantized activations times the
-    exactly-dequantized weights (the i32 dot itself is exact)."""
+    exactly-dequantized weights (the i32 dot itself is exact). The forced
+    small split-K cases run deep multi-tile k-chunks — the DMA/scale
+    software pipeline the auto split would collapse at these test sizes."""
     dt, quant, dequant, repack_fn = CASES[name]
     rng = np.random.default_rng(77 + M)
-    N, K = 192, 512  # exercises the N%128 edge tile too
+    N = 192  # exercises the N%128 edge tile too
     w = rng.standard_normal((N, K)).astype(np.float32) * 0.1
     raw = quant(w)
     wref = dequant(raw, K).reshape(N, K)
     x = rng.standard_normal((M, K)).astype(np.float32)
     qs, hdr = repack_fn(raw.reshape(N, -1), N, K)
     y = core.test_gemm_i8(np.ascontiguousarray(qs),
-                          np.ascontiguousarray(hdr), x, dt, N, K)
+                          np.ascontiguousarray(hdr), x, dt, N, K,
+                          force_splitk=sk)
     b = x.reshape(M, -1, 32)
     amax = np.abs(b).max(axis=2, keepdims=True)
     rinv = np.where(amax > 0, 127.0 / np.where(amax == 0, 1, amax), 0.0)
