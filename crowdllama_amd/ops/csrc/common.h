@@ -70,6 +70,18 @@ __host__ __device__ inline int64_t dhdr_row_bytes(DT t, int64_t k) {
     return 0;
 }
 
+// Transposed header copy for the i8 GEMM's LDS-DMA scale staging
+// (gemm_i8.hip): headers grouped so one k-window's headers for ALL rows
+// are contiguous. DQ4K: [K/64 q-groups][N][8B pair header];
+// DQ8: [K/32][N][f16 d]. Bytes per row (total = n * this):
+__host__ __device__ inline int64_t dhdr2_row_bytes(DT t, int64_t k) {
+    switch (t) {
+        case DT::DQ4K: return k / 64 * 8;
+        case DT::DQ8: return k / 32 * 2;
+        default: return 0;
+    }
+}
+
 // A weight matrix on device: N rows of K quantized columns.
 struct WTensor {
     DT dtype = DT::F32;
@@ -77,6 +89,7 @@ struct WTensor {
     int64_t k = 0;   // cols (input features)
     const void* qs = nullptr;
     const void* hdr = nullptr;
+    const void* hdr2 = nullptr;  // transposed headers (quant GEMM dtypes)
 };
 
 }  // namespace cla

@@ -56,12 +56,12 @@ void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
 void launch_scatter_logits(const float* src, float* dst, int B, int Vl,
                            int TP, hipStream_t);
 // gemm_i8.hip: int8-activation MFMA path for quantized weights
-void launch_quant_rows(const float* X, int8_t* xq, float* xs, float* xsum,
-                       int M, int K, int ldx, int mode, hipStream_t);
+void launch_quant_rows(const float* X, int8_t* xq, float* xsc, int M,
+                       int K, int ldx, int mode, hipStream_t);
 bool gemm_i8_supported(DT dtype, int M, int K);
-void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xs,
-                    const float* xsum, int ldxq, const float* res, float* C,
-                    int M, int ldc, hipStream_t, int force_splitk = 0);
+void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
+                    int ldxq, const float* res, float* C, int M, int ldc,
+                    hipStream_t, int force_splitk = 0);
 void launch_rope_prefill(float* qkv, const float* inv_freq,
                          const int32_t* page_table, uint16_t* kv_pool,
                          int slot, int pos0, int M, int NH, int NKV, int D,
@@ -337,6 +337,40 @@ WTensor Engine::upload_pack(const void* qs, size_t qs_bytes, const void* hdr,
         w.hdr = d_hdr;
         vram_bytes_ += hdr_bytes;
     }
+    // transposed header copy for the i8 GEMM's glds scale staging: one
+    // k-window's headers for all rows contiguous (see common.h)
+    const int64_t h2_rb = dhdr2_row_bytes(dtype, k);
+    if (h2_rb) {
+        // +1 KB slack: edge-tile glds lanes past the last row read (and
+        // discard) up to a wave's width beyond the array
+        std::vector<uint8_t> h2((size_t)h2_rb * rows + 1024, 0);
+        const uint8_t* hsrc = reinterpret_cast<const uint8_t*>(hdr);
+        const int64_t hrb = dhdr_row_bytes(dtype, k);
+        if (dtype == DT::DQ4K) {
+            const int64_t ng = k / 64;   // q-groups; 8B pair header each
+            parallel_for(rows, [&](int64_t lo, int64_t hi) {
+                for (int64_t r = lo; r < hi; r++)
+                    for (int64_t g = 0; g < ng; g++)
+                        std::memcpy(h2.data() + (g * rows + r) * 8,
+                                    hsrc + r * hrb + g * 8, 8);
+            });
+        } else {  // DQ8: [K/32][rows] f16 d
+            const int64_t nb = k / 32;
+            parallel_for(rows, [&](int64_t lo, int64_t hi) {
+                for (int64_t r = lo; r < hi; r++)
+                    for (int64_t b = 0; b < nb; b++)
+                        std::memcpy(h2.data() + (b * rows + r) * 2,
+                                    hsrc + r * hrb + b * 2, 2);
+            });
+        }
+        void* d_h2 = nullptr;
+        HIP_CHECK(hipMalloc(&d_h2, h2.size()));
+        HIP_CHECK(hipMemcpy(d_h2, h2.data(), h2.size(),
+                            hipMemcpyHostToDevice));
+        allocs_.push_back(d_h2);
+        w.hdr2 = d_h2;
+        vram_bytes_ += h2.size();
+    }
     return w;
 }
 
@@ -526,8 +560,7 @@ void Engine::alloc_state() {
             {(size_t)H, (size_t)F, (size_t)NH * D});
         const size_t maxM = std::max<size_t>(B, cfg_.prefill_chunk);
         xq_ = (int8_t*)dalloc(maxM * maxK);
-        xs_ = (float*)dalloc(maxM * (maxK / 32) * 4);
-        xsum_ = (float*)dalloc(maxM * (maxK / 32) * 4);
+        xsc_ = (float*)dalloc((maxM + 4) * (maxK / 32) * 2 * 4);
     }
     // rope frequency table
     std::vector<float> invf(D / 2);
@@ -561,8 +594,7 @@ static bool is_quant_dt(DT t) {
 // Activation-quant scratch for the i8 GEMM path (engine-owned buffers).
 struct QBufs {
     int8_t* xq = nullptr;
-    float* xs = nullptr;
-    float* xsum = nullptr;
+    float* xsc = nullptr;   // interleaved [K/32][2][M4] scales/sums
 };
 
 // Launch a projection through the GEMM path: pre-zero C when split-K
@@ -587,11 +619,11 @@ static void gemm_proj(const Proj& p, const float* X, const float* res,
         for (auto& pt : p.parts)
             any_i8 |= gemm_i8_supported(pt.w.dtype, M, (int)pt.w.k);
     if (any_i8)
-        launch_quant_rows(X, qb->xq, qb->xs, qb->xsum, M, Kq,
+        launch_quant_rows(X, qb->xq, qb->xsc, M, Kq,
                           ldx < 0 ? Kq : ldx, xsilu ? 1 : 0, s);
     for (auto& pt : p.parts) {
         if (qb && gemm_i8_supported(pt.w.dtype, M, (int)pt.w.k)) {
-            launch_gemm_i8(pt.w, qb->xq, qb->xs, qb->xsum, Kq,
+            launch_gemm_i8(pt.w, qb->xq, qb->xsc, Kq,
                            res ? res + pt.row_off : nullptr,
                            C + pt.row_off, M, (int)p.n_total, s);
         } else {
@@ -644,7 +676,7 @@ void Engine::step(hipStream_t s) {
     // B=1 only: at B=2 the split-K MFMA GEMM already beats two GEMV
     // passes (measured 340 tok/s GEMV vs ~440 GEMM at B=2)
     const bool gemv_path = B <= 1;
-    const QBufs qbufs{xq_, xs_, xsum_};
+    const QBufs qbufs{xq_, xsc_};
     const QBufs* qb = cfg_.act_q8 ? &qbufs : nullptr;
     int li = 0;
     for (auto& L : layers_) {
@@ -742,9 +774,9 @@ void Engine::step(hipStream_t s) {
         if (zero)
             HIP_CHECK(hipMemsetAsync(lg, 0, (size_t)B * ldl * 4, s));
         if (qb && gemm_i8_supported(head_.dtype, B, (int)head_.k)) {
-            launch_quant_rows(xn_, xq_, xs_, xsum_, B, (int)head_.k,
+            launch_quant_rows(xn_, xq_, xsc_, B, (int)head_.k,
                               (int)head_.k, 0, s);
-            launch_gemm_i8(head_, xq_, xs_, xsum_, (int)head_.k, nullptr,
+            launch_gemm_i8(head_, xq_, xsc_, (int)head_.k, nullptr,
                            lg, B, ldl, s);
         } else {
             launch_gemm(head_, xn_, nullptr, lg, B, ldl, s);
@@ -802,7 +834,7 @@ void Engine::prefill_chunk_pass(int slot, int pos0, int m) {
     const float eps = meta_.rms_eps;
     const bool tp = cfg_.tp_size > 1;
     const bool r0 = cfg_.tp_rank == 0;
-    const QBufs qbufs{xq_, xs_, xsum_};
+    const QBufs qbufs{xq_, xsc_};
     // i8 path engages automatically for chunks <= 128 rows (same act_q8
     // quantization semantics as the decode step)
     const QBufs* qb = cfg_.act_q8 ? &qbufs : nullptr;

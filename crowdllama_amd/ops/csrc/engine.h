@@ -166,8 +166,7 @@ private:
     float* logits_tp_ = nullptr;  // [B][vocab_l] local slice (tp>1)
     float* gather_tp_ = nullptr;  // [tp][B][vocab_l] all-gather recv (tp>1)
     int8_t* xq_ = nullptr;        // [maxM][maxK] i8-GEMM activation quant
-    float* xs_ = nullptr;         // [maxM][maxK/32] block scales
-    float* xsum_ = nullptr;       // [maxM][maxK/32] dx*sum(qx)
+    float* xsc_ = nullptr;        // [maxK/32][2][M4] scales + dx*sum(qx)
 };
 
 }  // namespace cla

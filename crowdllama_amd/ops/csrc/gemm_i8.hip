@@ -4,28 +4,32 @@
 // pre-quantized to int8 per 32-block (k_quant_rows below; same semantics as
 // the act_q8 GEMV / ref_numpy(act_q8=True): x ~ rint(x/xd)*xd).
 //
-// Design (v2, after the round-2 register-staged version measured the same
-// ~1.1 TB/s wall as the bf16 kernel — the bottleneck was staging latency,
-// not dequant VALU):
-// - W stages as RAW quant bytes via `global_load_lds` DMA (nt policy): no
-//   VGPR round-trip, no unpack in staging, 0.5 B/weight of LDS for Q4_K.
-//   Nibble unpack happens at fragment read (2-4 VALU per 8 weights), and
-//   one 8-byte LDS read serves both K=32 halves of a BK=64 tile.
-// - raw `s_barrier` + counted `s_waitcnt vmcnt(N)` keep the DMA pipeline
-//   across barriers (the guide's 8-phase discipline; __syncthreads would
-//   drain vmcnt(0) while glds is in flight).
+// Design (v3 — ALL k-loop operands stream by `global_load_lds` DMA):
+// Round-2 history: the register-staged v1 and the mixed glds/register v2
+// both plateaued at the bf16 kernel's ~1.1 TB/s because of two .s-level
+// traps the CDNA guide documents: a second __shared__ object, and any
+// ordinary VGPR-destination load inside a glds loop, each make hipcc wait
+// vmcnt(0) mid-pipeline (confirmed in this kernel's disassembly). v3:
+// - ONE __shared__ array; W raw bytes, the X int8 tile, W block headers
+//   (via hdr2, an upload-time transposed header copy — common.h), and the
+//   interleaved activation scales ALL arrive by glds (nt on the weight
+//   stream), so the k-loop contains no compiler-counted loads at all.
+// - raw `s_barrier` + counted `s_waitcnt vmcnt(N)` with an exact per-tile
+//   glds count (uniform across waves) keep one full tile of DMA in flight
+//   across every barrier.
 // - v_mfma_i32_16x16x32_i8 computes the exact i32 dot; per-32 scales are
 //   applied at a K=32 drain:
 //     y += d_w*d_x*dot_i32 - m_w*(d_x*sum qx)   (Q4_K; Q8_0 has m_w=0)
-//   Block headers and activation scales are read from global (L1/L2
-//   broadcast path), prefetched one tile ahead into registers.
-// - activation scales live TRANSPOSED (xsT/xsumT [K/32][M4], M4 = M
-//   rounded to 4) so a drain reads its 4 slots as one float4.
+//   Q4_K weights stay as raw nibbles in LDS (0.5 B/weight): one 8-byte
+//   read serves both K=32 halves (lo/hi nibble planes of the same bytes).
 //
 // Replaces (functionally) llama.cpp's MMQ path for batched decode — the
 // compute the reference delegates to Ollama; designed for
 // CDNA4 wave64/XCD/LDS-DMA geometry, not ported.
 #include "common.h"
+
+#include <cstring>
+#include <vector>
 
 namespace cla {
 
@@ -36,7 +40,6 @@ int gemm_splitk_factor(int N, int K, int M);  // gemm.hip (shared contract)
 namespace {
 
 constexpr int BN = 128, BK = 64;
-constexpr int LDXQ = BK + 16;   // X LDS row stride (conflict-free b64)
 
 __device__ __forceinline__ float f16b2f(uint32_t h) {
     __half_raw r;
@@ -50,12 +53,14 @@ __device__ __forceinline__ float f16b2f(uint32_t h) {
 // all 16 M-rows x 32 cols), BM=32 -> 2x2 (16 rows x 64 cols per wave).
 // Split-K accumulates into pre-zeroed C via atomicAdd (identical contract
 // to gemm.hip's k_gemm: gemm_uses_splitk tells the caller to pre-zero).
+//
+// xsc is the interleaved activation-scale array [K/32][2][M4]:
+// [kg][0][m] = block scale dx, [kg][1][m] = dx*sum(qx). M4 = M round-to-4.
 template <DT W, int BM_>
 __global__ __launch_bounds__(256) void k_gemm_i8(
-    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
+    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr2,
     const int8_t* __restrict__ xq,    // [M][ldxq] int8 (pre-quantized)
-    const float* __restrict__ xsT,    // [ldxq/32][M4] block scales
-    const float* __restrict__ xsumT,  // [ldxq/32][M4] dx*sum(qx) per block
+    const float* __restrict__ xsc,    // [ldxq/32][2][M4]
     const float* __restrict__ res,    // [M][ldc] or null
     float* __restrict__ C,            // [M][ldc]
     int M, int N, int K, int ldc, int ldxq, int k_chunk) {
@@ -63,9 +68,24 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     constexpr int WNW = 4 / WMW;
     constexpr int JF = BN / WNW / 16;        // b fragments per wave (2 or 4)
     constexpr int RAWB = (W == DT::DQ4K) ? BK / 2 : BK;  // raw bytes/row
-    constexpr int NGL = (BN / 4) * RAWB / 1024;          // glds per wave
-    __shared__ __attribute__((aligned(16))) int8_t Wr[2][BN * RAWB];
-    __shared__ __attribute__((aligned(16))) int8_t Xl[2][BM_ * LDXQ];
+    constexpr int NGW = (BN / 4) * RAWB / 1024;  // W glds per wave (1 or 2)
+    constexpr int NGX = BM_ * BK / 1024;         // X glds per wave (1 or 2)
+    constexpr int NGH = (W == DT::DQ4K) ? 1 : 2; // hdr2 glds per wave
+    constexpr int NGS = 1;                       // xsc glds per wave
+    constexpr int NGLT = NGW + NGX + NGH + NGS;  // per wave per tile
+    // --- single LDS array (a second __shared__ object would make hipcc
+    // drain vmcnt(0) before every ds_read — guide §5 trap (a)) ---
+    // map (per buffer):
+    //   [0, WSZ)                      W raw bytes, row-major RAWB/row
+    //   [WSZ, +4*XSZ)                 per-wave X int8 copies, [BM_][64]
+    //   [.., +4*1024) (Q4K) / +4*512 (Q8)  per-wave hdr copies
+    //   [.., +4*256)                  per-wave xsc copies [kg2][2][16] f32
+    constexpr int WSZ = BN * RAWB;
+    constexpr int XSZ = BM_ * BK;
+    constexpr int HSZ = (W == DT::DQ4K) ? 1024 : 512;
+    constexpr int SSZ = 256;
+    constexpr int BUFSZ = WSZ + 4 * (XSZ + HSZ + SSZ);
+    __shared__ __attribute__((aligned(16))) int8_t lds[2][BUFSZ];
 
     const int tid = threadIdx.x;
     const int bn = blockIdx.x, bm = blockIdx.y, bz = blockIdx.z;
@@ -79,7 +99,6 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     const int wn = (WMW == 1) ? wid : (wid & 1);
     const int lrow = lane & 15, lk = lane >> 4;
     const int M4 = (M + 3) & ~3;
-    const int nb32 = ldxq / 32;
 
     float facc[JF][4];
     #pragma unroll
@@ -87,16 +106,16 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         #pragma unroll
         for (int r = 0; r < 4; r++) facc[j][r] = 0.f;
 
-    // ---- W DMA addressing: wave wid owns rows [wid*32, wid*32+32) ----
-    // Q4K: 32 B/row -> lane covers row wid*32+(l>>1), 16B half (l&1).
-    // Q8:  64 B/row, two glds: glds g covers rows wid*32+g*16+(l>>2),
-    //      16B quarter (l&3).
     const int64_t qs_rb = dqs_row_bytes(W, K);
-    const int64_t hdr_rb = dhdr_row_bytes(W, K);
-    int64_t wrow_g[NGL];     // this lane's global W row per glds
-    int wboff[NGL];          // byte offset of this lane's 16B within the row
+    const int64_t h2_rb = dhdr2_row_bytes(W, K);
+    (void)h2_rb;
+
+    // ---- per-lane global source addresses (row parts precomputed) ----
+    // W: wave wid owns rows [wid*32, wid*32+32).
+    int64_t wrow_q[2];   // clamped global row per W glds
+    int wboff[2];
     #pragma unroll
-    for (int g = 0; g < NGL; g++) {
+    for (int g = 0; g < NGW; g++) {
         int rl;
         if constexpr (W == DT::DQ4K) {
             rl = wid * 32 + (lane >> 1);
@@ -106,107 +125,126 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
             wboff[g] = (lane & 3) * 16;
         }
         const int64_t gn = (int64_t)n0 + rl;
-        wrow_g[g] = gn < N ? gn : N - 1;
+        wrow_q[g] = (gn < N ? gn : N - 1) * qs_rb;
+    }
+    // X: each wave DMAs its own copy of the whole [BM_][64] tile.
+    // glds gx covers rows gx*16 + (lane>>2), 16B quarter lane&3.
+    int64_t xrow_off[2];
+    #pragma unroll
+    for (int g = 0; g < NGX; g++) {
+        const int xr = (NGX == 1) ? (lane >> 2) : (g * 16 + (lane >> 2));
+        const int gm = (m0 + xr < M) ? m0 + xr : (M > 0 ? M - 1 : 0);
+        xrow_off[g] = (int64_t)gm * ldxq + (lane & 3) * 16;
+    }
+    const int xzrow = (NGX == 1) ? (lane >> 2) : 0;  // guard rows >= M later
+    (void)xzrow;
+    // hdr2: per-wave copy of the k-window's headers for all BN cols.
+    // Q4K: 128 cols x 8B = 1KB, lane takes cols {2l, 2l+1} (16B).
+    // Q8: 2 x (128 cols x 2B f16) = 2 x 256B dword-glds, lane 4B.
+    // xsc: 4 x 64B chunks = 256B dword-glds: chunk c = (kg2, dx|sum),
+    //      lane l: chunk l>>4, slot (l&15) of this wave's 16 M-rows.
+
+    int8_t* lds_w[2];
+    int8_t* lds_x[2];
+    int8_t* lds_h[2];
+    int8_t* lds_s[2];
+    #pragma unroll
+    for (int b = 0; b < 2; b++) {
+        lds_w[b] = &lds[b][0];
+        lds_x[b] = &lds[b][0] + WSZ + wid * XSZ;
+        lds_h[b] = &lds[b][0] + WSZ + 4 * XSZ + wid * HSZ;
+        lds_s[b] = &lds[b][0] + WSZ + 4 * (XSZ + HSZ) + wid * SSZ;
     }
 
-    auto issue_w_glds = [&](int kb, int pb) {
-        // per-row byte offset of the BK window's raw bytes
+    auto issue_tile = [&](int kb, int pb) {
+        // W raw (nt: streamed once per step)
         #pragma unroll
-        for (int g = 0; g < NGL; g++) {
+        for (int g = 0; g < NGW; g++) {
             int64_t off;
             if constexpr (W == DT::DQ4K) {
                 const int sb = kb >> 8, q = (kb & 255) >> 6;
-                off = wrow_g[g] * qs_rb + sb * 128 + q * 32 + wboff[g];
+                off = wrow_q[g] + sb * 128 + q * 32 + wboff[g];
             } else {
-                off = wrow_g[g] * qs_rb + kb + wboff[g];
+                off = wrow_q[g] + kb + wboff[g];
             }
             __builtin_amdgcn_global_load_lds(
                 reinterpret_cast<const uint32_t*>(qs + off),
                 reinterpret_cast<uint32_t*>(
-                    Wr[pb] + wid * (NGL * 1024) + g * 1024),
-                16, 0, 2 /* nt: streamed once */);
+                    lds_w[pb] + wid * (NGW * 1024) +
+                    g * 1024),
+                16, 0, 2);
         }
-    };
-
-    // ---- X register staging (tiny: BM_*64 B/tile) ----
-    const int xrow = tid >> 2, xseg = tid & 3;
-    const bool xwave = xrow < BM_;           // wave-uniform (wave 0 / 0..1)
-    const int xgm = (m0 + xrow < M) ? m0 + xrow : (M > 0 ? M - 1 : 0);
-    const bool xvalid = (m0 + xrow) < M && xrow < BM_;
-    uint4 xregs = {0, 0, 0, 0};
-    auto load_x = [&](int kb) {
-        if (xwave)
-            xregs = *reinterpret_cast<const uint4*>(
-                xq + (size_t)xgm * ldxq + kb + xseg * 16);
-    };
-    auto write_x = [&](int pb) {
-        if (xwave) {
-            uint4 v = xvalid ? xregs : uint4{0, 0, 0, 0};
-            *reinterpret_cast<uint4*>(Xl[pb] + xrow * LDXQ + xseg * 16) = v;
-        }
-    };
-
-    // ---- scale prefetch (global -> regs, two tiles ahead) ----
-    // W headers: per fragment j, the (col, q-group) pair header. Q4K: uint2
-    // {d,dmin | sc/mn x2} covers BOTH K=32 halves. Q8: one u32 = two f16 d.
-    // Parity-indexed slots: tile t uses slot t&1; the slot is refilled with
-    // tile t+2's scales right after t consumes it (never overwrites t+1's).
-    uint2 hd_s[2][JF];
-    float4 dx_s[2][2], sm_s[2][2];
-    int64_t hcol[JF];
-    #pragma unroll
-    for (int j = 0; j < JF; j++) {
-        const int64_t gn = (int64_t)n0 + wn * (BN / WNW) + j * 16 + lrow;
-        hcol[j] = gn < N ? gn : N - 1;
-    }
-    // Exactly this many VMEM instructions per wave per load_scales call
-    // (every wave issues the same count — vmcnt literals depend on it).
-    constexpr int RSC = JF + 4;
-    auto load_scales = [&](int kb, int slot) {
+        // X tile copy for this wave
         #pragma unroll
-        for (int j = 0; j < JF; j++) {
-            if constexpr (W == DT::DQ4K) {
-                const int sb = kb >> 8, q = (kb & 255) >> 6;
-                hd_s[slot][j] = *reinterpret_cast<const uint2*>(
-                    hdr + hcol[j] * hdr_rb + sb * 32 + q * 8);
-            } else {
-                hd_s[slot][j].x = *reinterpret_cast<const uint32_t*>(
-                    hdr + hcol[j] * hdr_rb + (kb >> 5) * 2);
-                hd_s[slot][j].y = 0;
+        for (int g = 0; g < NGX; g++) {
+            __builtin_amdgcn_global_load_lds(
+                reinterpret_cast<const uint32_t*>(xq + xrow_off[g] + kb),
+                reinterpret_cast<uint32_t*>(lds_x[pb] + g * 1024), 16, 0, 0);
+        }
+        // headers
+        if constexpr (W == DT::DQ4K) {
+            const int grp = kb >> 6;  // q-group index
+            __builtin_amdgcn_global_load_lds(
+                reinterpret_cast<const uint32_t*>(
+                    hdr2 + ((int64_t)grp * N + n0 + 2 * lane) * 8),
+                reinterpret_cast<uint32_t*>(lds_h[pb]), 16, 0, 0);
+        } else {
+            const int kg = kb >> 5;
+            #pragma unroll
+            for (int g = 0; g < 2; g++) {
+                __builtin_amdgcn_global_load_lds(
+                    reinterpret_cast<const uint32_t*>(
+                        hdr2 + ((int64_t)(kg + g) * N + n0) * 2 + lane * 4),
+                    reinterpret_cast<uint32_t*>(lds_h[pb] + g * 256),
+                    4, 0, 0);
             }
         }
-        const int kg = kb >> 5;
-        const int mrow = m0 + wm * 16 + lk * 4;
-        #pragma unroll
-        for (int k2 = 0; k2 < 2; k2++) {
-            dx_s[slot][k2] = *reinterpret_cast<const float4*>(
-                xsT + (size_t)(kg + k2) * M4 + mrow);
-            sm_s[slot][k2] = *reinterpret_cast<const float4*>(
-                xsumT + (size_t)(kg + k2) * M4 + mrow);
+        // activation scales: chunk (kg2 = c>>1, arr = c&1), 16 slots each.
+        // Slot index clamps to M4-1 (partial BM tiles): the duplicated
+        // values land in LDS slots whose output rows are discarded anyway.
+        {
+            const int kg = kb >> 5;
+            const int c = lane >> 4;
+            int srow = m0 + wm * 16 + (lane & 15);
+            if (srow > M4 - 1) srow = M4 - 1;
+            const int64_t soff =
+                ((int64_t)(kg + (c >> 1)) * 2 + (c & 1)) * M4 + srow;
+            __builtin_amdgcn_global_load_lds(
+                reinterpret_cast<const uint32_t*>(xsc + soff),
+                reinterpret_cast<uint32_t*>(lds_s[pb]), 4, 0, 0);
         }
     };
 
-    auto mfma_tile = [&](int pb, int slot) {
+    auto mfma_tile = [&](int pb) {
         // raw W fragment bytes: one b64 per j serves both K=32 halves (Q4K)
         long rawj[JF][(W == DT::DQ4K) ? 1 : 2];
+        uint2 hdj[JF];
         #pragma unroll
         for (int j = 0; j < JF; j++) {
             const int r = wn * (BN / WNW) + j * 16 + lrow;
-            if constexpr (W == DT::DQ4K) {
-                rawj[j][0] = *reinterpret_cast<const long*>(
-                    Wr[pb] + r * RAWB + lk * 8);
-            } else {
-                rawj[j][0] = *reinterpret_cast<const long*>(
-                    Wr[pb] + r * RAWB + lk * 8);
+            rawj[j][0] = *reinterpret_cast<const long*>(
+                lds_w[pb] + r * RAWB + lk * 8);
+            if constexpr (W == DT::DQ8) {
                 rawj[j][1] = *reinterpret_cast<const long*>(
-                    Wr[pb] + r * RAWB + 32 + lk * 8);
+                    lds_w[pb] + r * RAWB + 32 + lk * 8);
+                hdj[j].x = (uint32_t) reinterpret_cast<const uint16_t*>(
+                               lds_h[pb])[r]
+                           | ((uint32_t) reinterpret_cast<const uint16_t*>(
+                                  lds_h[pb] + 256)[r] << 16);
+                hdj[j].y = 0;
+            } else {
+                hdj[j] = reinterpret_cast<const uint2*>(lds_h[pb])[r];
             }
         }
+        const float* sw = reinterpret_cast<const float*>(lds_s[pb]);
         #pragma unroll
         for (int kb2 = 0; kb2 < 2; kb2++) {
             const long a = *reinterpret_cast<const long*>(
-                Xl[pb] + (wm * 16 + lrow) * LDXQ + kb2 * 32 + lk * 8);
-            const float4 dx4 = dx_s[slot][kb2], sm4 = sm_s[slot][kb2];
+                lds_x[pb] + (wm * 16 + lrow) * BK + kb2 * 32 + lk * 8);
+            const float4 dx4 = *reinterpret_cast<const float4*>(
+                sw + (kb2 * 2 + 0) * 16 + lk * 4);
+            const float4 sm4 = *reinterpret_cast<const float4*>(
+                sw + (kb2 * 2 + 1) * 16 + lk * 4);
             #pragma unroll
             for (int j = 0; j < JF; j++) {
                 long b;
@@ -215,14 +253,14 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
                     b = (kb2 == 0)
                             ? (rawj[j][0] & 0x0F0F0F0F0F0F0F0FLL)
                             : ((rawj[j][0] >> 4) & 0x0F0F0F0F0F0F0F0FLL);
-                    const float dd = f16b2f(hd_s[slot][j].x & 0xFFFF);
-                    const float dmin = f16b2f(hd_s[slot][j].x >> 16);
-                    const uint32_t y = hd_s[slot][j].y;
+                    const float dd = f16b2f(hdj[j].x & 0xFFFF);
+                    const float dmin = f16b2f(hdj[j].x >> 16);
+                    const uint32_t y = hdj[j].y;
                     d = dd * (float)((y >> (16 * kb2)) & 0xFF);
                     m = dmin * (float)((y >> (16 * kb2 + 8)) & 0xFF);
                 } else {
                     b = rawj[j][kb2];
-                    d = f16b2f((hd_s[slot][j].x >> (16 * kb2)) & 0xFFFF);
+                    d = f16b2f((hdj[j].x >> (16 * kb2)) & 0xFFFF);
                     m = 0.f;
                 }
                 v4i c = {0, 0, 0, 0};
@@ -235,43 +273,26 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         }
     };
 
-    // ---- prologue: tile 0 staged, tile 1 in flight ----
+    // ---- DMA pipeline: one tile in flight across every barrier ----
     const int kb_last = kb_hi - BK;          // all tiles full (K%BK==0)
-    auto clamp_kb = [&](int kb) { return kb <= kb_last ? kb : kb_last; };
-    load_x(kb_lo);
-    load_scales(kb_lo, 0);
-    issue_w_glds(kb_lo, 0);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    write_x(0);
-    load_x(clamp_kb(kb_lo + BK));
-    issue_w_glds(clamp_kb(kb_lo + BK), 1);
-    load_scales(clamp_kb(kb_lo + BK), 1);
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    issue_tile(kb_lo, 0);
+    issue_tile(kb_lo + BK <= kb_last ? kb_lo + BK : kb_last, 1);
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NGLT) : "memory");  // tile 0 in
     __builtin_amdgcn_s_barrier();
 
-    // steady state at iteration t (buffer pb, scale slot t&1):
-    //   buf[pb] landed for every wave; buf[1-pb] DMA in flight;
-    //   xregs hold X(t+1); scale slots hold t (t&1) and t+1 (1-(t&1)).
-    // Tail iterations clamp their t+2 prefetches to the last tile (the
-    // re-staged bytes are never read) so the wait/barrier pattern stays
-    // uniform with no divergent branches around loads.
-    int pb = 0, slot = 0;
+    int pb = 0;
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
-        write_x(1 - pb);                     // X of tile t+1 (regs ready)
-        mfma_tile(pb, slot);
-        if (kb + BK >= kb_hi) break;         // last tile: no more staging
-        const int kb2 = clamp_kb(kb + 2 * BK);
-        load_x(kb2);
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma_tile(pb);
+        if (kb + BK >= kb_hi) break;         // last tile: nothing to stage
         __builtin_amdgcn_s_barrier();        // buf[pb] fully consumed
-        issue_w_glds(kb2, pb);
-        load_scales(kb2, slot);              // slot t&1 just consumed
-        // retire everything older than [t+2 glds + t+2 scale loads]:
-        // exactly the t+1 glds (and the older t+2 xregs loads)
-        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NGL + RSC) : "memory");
-        __builtin_amdgcn_s_barrier();        // everyone's t+1 landed
+        {   // tile t+2 (clamped at the tail; re-staged bytes never read)
+            const int kb2 = kb + 2 * BK;
+            issue_tile(kb2 <= kb_last ? kb2 : kb_last, pb);
+        }
+        // everyone's t+1 landed once own vmcnt hits the t+2 count
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NGLT) : "memory");
+        __builtin_amdgcn_s_barrier();
         pb ^= 1;
-        slot ^= 1;
     }
 
     // ---- epilogue (same contract as gemm.hip k_gemm) ----
@@ -298,12 +319,12 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
 // Per-row, per-32-block symmetric int8 (xd = amax/127, rint) — identical
 // semantics to the act_q8 GEMV staging / ref_numpy(act_q8=True). mode 1
 // applies silu(gate)*up first (X is [M][2K]: gate | up halves).
-// 8 lanes per block; scales/sums written TRANSPOSED ([K/32][M4]) so the
-// GEMM drain reads 4 slots as one float4. Grid (ceil(K/32/32), M).
+// 8 lanes per block; scale and dx*sum(qx) go to the interleaved transposed
+// array xsc[K/32][2][M4] the GEMM's DMA staging expects.
+// Grid (ceil(K/32/32), M).
 __global__ __launch_bounds__(256) void k_quant_rows(
     const float* __restrict__ X, int8_t* __restrict__ xq,
-    float* __restrict__ xsT, float* __restrict__ xsumT,
-    int K, int ldx, int mode, int M4) {
+    float* __restrict__ xsc, int K, int ldx, int mode, int M4) {
     const int m = blockIdx.y;
     const int tid = threadIdx.x;
     const int jl = tid & 7;
@@ -343,20 +364,20 @@ __global__ __launch_bounds__(256) void k_quant_rows(
     *reinterpret_cast<uint32_t*>(xq + (size_t)m * K + blk * 32 + jl * 4) =
         packed;
     if (jl == 0) {
-        xsT[(size_t)blk * M4 + m] = scale;
-        xsumT[(size_t)blk * M4 + m] = scale * (float)s;
+        xsc[((size_t)blk * 2 + 0) * M4 + m] = scale;
+        xsc[((size_t)blk * 2 + 1) * M4 + m] = scale * (float)s;
     }
 }
 
 // --------------------------------------------------------- launch stubs
 
-void launch_quant_rows(const float* X, int8_t* xq, float* xs, float* xsum,
-                       int M, int K, int ldx, int mode, hipStream_t stream) {
+void launch_quant_rows(const float* X, int8_t* xq, float* xsc, int M, int K,
+                       int ldx, int mode, hipStream_t stream) {
     if (K % 32) throw std::runtime_error("quant_rows: K must be /32");
     const int M4 = (M + 3) & ~3;
     dim3 grid((K / 32 + 31) / 32, M), block(256);
     hipLaunchKernelGGL(k_quant_rows, grid, block, 0, stream,
-                       X, xq, xs, xsum, K, ldx, mode, M4);
+                       X, xq, xsc, K, ldx, mode, M4);
 }
 
 bool gemm_i8_supported(DT dtype, int M, int K) {
@@ -364,11 +385,11 @@ bool gemm_i8_supported(DT dtype, int M, int K) {
            K % BK == 0 && K % 256 == 0;
 }
 
-void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xs,
-                    const float* xsum, int ldxq, const float* res, float* C,
-                    int M, int ldc, hipStream_t stream, int force_splitk) {
+void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
+                    int ldxq, const float* res, float* C, int M, int ldc,
+                    hipStream_t stream, int force_splitk) {
     const int N = (int)w.n, K = (int)w.k;
-    if (!gemm_i8_supported(w.dtype, M, K))
+    if (!gemm_i8_supported(w.dtype, M, K) || !w.hdr2)
         throw std::runtime_error("gemm_i8: unsupported dtype/shape");
     const bool bm16 = M <= 16;
     const int bm_tiles = bm16 ? 1 : (M + 31) / 32;
@@ -379,7 +400,7 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xs,
     dim3 grid(n_tiles, bm_tiles, splitk), block(256);
     #define GI8_ONE(WT, BMV)                                                   \
         hipLaunchKernelGGL((k_gemm_i8<WT, BMV>), grid, block, 0, stream,       \
-            (const uint8_t*)w.qs, (const uint8_t*)w.hdr, xq, xs, xsum,         \
+            (const uint8_t*)w.qs, (const uint8_t*)w.hdr2, xq, xsc,             \
             res, C, M, N, K, ldc, ldxq, k_chunk)
     switch (w.dtype) {
         case DT::DQ4K:
@@ -431,42 +452,61 @@ void launch_mfma_probe_i8_test(const int8_t* A, const int8_t* B, int32_t* C) {
 }
 
 // -------------------------------------------------------- test entry point
-// Full path: quantize X rows, run the i8 GEMM (host buffers in/out).
+// Full path: build hdr2 host-side, quantize X rows, run the i8 GEMM.
 void launch_gemm_i8_test(const void* qs, const void* hdr, const float* x,
                          float* y, int dtype, int M, int N, int K,
                          size_t qs_bytes, size_t hdr_bytes,
                          int force_splitk) {
     const DT dt = static_cast<DT>(dtype);
     const int M4 = (M + 3) & ~3;
-    void *d_qs = nullptr, *d_hdr = nullptr;
+    void *d_qs = nullptr, *d_h2 = nullptr;
     HIP_CHECK(hipMalloc(&d_qs, qs_bytes));
     HIP_CHECK(hipMemcpy(d_qs, qs, qs_bytes, hipMemcpyHostToDevice));
-    HIP_CHECK(hipMalloc(&d_hdr, hdr_bytes));
-    HIP_CHECK(hipMemcpy(d_hdr, hdr, hdr_bytes, hipMemcpyHostToDevice));
+    // transpose headers -> hdr2 (same as Engine::upload_pack)
+    {
+        const int64_t h2_rb = dhdr2_row_bytes(dt, K);
+        const int64_t hrb = dhdr_row_bytes(dt, K);
+        std::vector<uint8_t> h2((size_t)h2_rb * N + 1024, 0);
+        const uint8_t* hsrc = reinterpret_cast<const uint8_t*>(hdr);
+        (void)hdr_bytes;
+        if (dt == DT::DQ4K) {
+            for (int64_t r = 0; r < N; r++)
+                for (int64_t g = 0; g < K / 64; g++)
+                    std::memcpy(h2.data() + (g * N + r) * 8,
+                                hsrc + r * hrb + g * 8, 8);
+        } else {
+            for (int64_t r = 0; r < N; r++)
+                for (int64_t b = 0; b < K / 32; b++)
+                    std::memcpy(h2.data() + (b * N + r) * 2,
+                                hsrc + r * hrb + b * 2, 2);
+        }
+        HIP_CHECK(hipMalloc(&d_h2, h2.size()));
+        HIP_CHECK(hipMemcpy(d_h2, h2.data(), h2.size(),
+                            hipMemcpyHostToDevice));
+    }
     float* d_x = nullptr;
     HIP_CHECK(hipMalloc((void**)&d_x, (size_t)M * K * 4));
     HIP_CHECK(hipMemcpy(d_x, x, (size_t)M * K * 4, hipMemcpyHostToDevice));
     int8_t* d_xq = nullptr;
-    float *d_xs = nullptr, *d_xsum = nullptr, *d_y = nullptr;
+    float *d_xsc = nullptr, *d_y = nullptr;
     HIP_CHECK(hipMalloc((void**)&d_xq, (size_t)M * K));
-    HIP_CHECK(hipMalloc((void**)&d_xs, (size_t)M4 * (K / 32) * 4));
-    HIP_CHECK(hipMalloc((void**)&d_xsum, (size_t)M4 * (K / 32) * 4));
+    HIP_CHECK(hipMalloc((void**)&d_xsc, (size_t)M4 * (K / 32) * 2 * 4));
     HIP_CHECK(hipMalloc((void**)&d_y, (size_t)M * N * 4));
-    HIP_CHECK(hipMemset(d_xs, 0, (size_t)M4 * (K / 32) * 4));
-    HIP_CHECK(hipMemset(d_xsum, 0, (size_t)M4 * (K / 32) * 4));
-    launch_quant_rows(d_x, d_xq, d_xs, d_xsum, M, K, K, 0, nullptr);
+    HIP_CHECK(hipMemset(d_xsc, 0, (size_t)M4 * (K / 32) * 2 * 4));
+    launch_quant_rows(d_x, d_xq, d_xsc, M, K, K, 0, nullptr);
     WTensor w;
-    w.dtype = dt; w.n = N; w.k = K; w.qs = d_qs; w.hdr = d_hdr;
-    const int sk = force_splitk > 0 ? force_splitk : gemm_splitk_factor(N, K, M);
+    w.dtype = dt; w.n = N; w.k = K; w.qs = d_qs; w.hdr = nullptr;
+    w.hdr2 = d_h2;
+    const int sk = force_splitk > 0 ? force_splitk
+                                    : gemm_splitk_factor(N, K, M);
     if (sk > 1)
         HIP_CHECK(hipMemset(d_y, 0, (size_t)M * N * 4));
-    launch_gemm_i8(w, d_xq, d_xs, d_xsum, K, nullptr, d_y, M, N, nullptr,
+    launch_gemm_i8(w, d_xq, d_xsc, K, nullptr, d_y, M, N, nullptr,
                    force_splitk);
     HIP_CHECK(hipDeviceSynchronize());
     HIP_CHECK(hipMemcpy(y, d_y, (size_t)M * N * 4, hipMemcpyDeviceToHost));
-    (void)hipFree(d_qs); (void)hipFree(d_hdr); (void)hipFree(d_x);
-    (void)hipFree(d_xq); (void)hipFree(d_xs); (void)hipFree(d_xsum);
-    (void)hipFree(d_y);
+    (void)hipFree(d_qs); (void)hipFree(d_h2); (void)hipFree(d_x);
+    (void)hipFree(d_xq); (void)hipFree(d_xsc); (void)hipFree(d_y);
 }
 
 }  // namespace cla
