@@ -47,6 +47,48 @@ __device__ __forceinline__ float f16b2f(uint32_t h) {
     return __half2float(*reinterpret_cast<__half*>(&r));
 }
 
+// LDS-DMA issued through inline asm (guide §5.7 glds16_asm recipe): the
+// builtin form is tracked by hipcc's waitcnt pass, which inserts a
+// conservative vmcnt(0) before the loop-header ds_reads whenever a glds is
+// outstanding across the backedge — draining the prefetched tile every
+// iteration (confirmed in this kernel's .s). Inside asm the DMA is
+// invisible; ALL vmcnt accounting is done by this kernel's own counted
+// waits. M0 (the LDS destination base) is written and restored in the same
+// statement; `lds_off` must be wave-uniform (readfirstlane at the caller).
+__device__ __forceinline__ void glds16(const void* gsrc, unsigned lds_off) {
+    unsigned keep;
+    asm volatile(
+        "s_mov_b32 %0, m0\n\t"
+        "s_mov_b32 m0, %2\n\t"
+        "s_nop 0\n\t"
+        "global_load_lds_dwordx4 %1, off\n\t"
+        "s_mov_b32 m0, %0"
+        : "=&s"(keep) : "v"(gsrc), "s"(lds_off) : "memory");
+}
+
+__device__ __forceinline__ void glds16_nt(const void* gsrc,
+                                          unsigned lds_off) {
+    unsigned keep;
+    asm volatile(
+        "s_mov_b32 %0, m0\n\t"
+        "s_mov_b32 m0, %2\n\t"
+        "s_nop 0\n\t"
+        "global_load_lds_dwordx4 %1, off nt\n\t"
+        "s_mov_b32 m0, %0"
+        : "=&s"(keep) : "v"(gsrc), "s"(lds_off) : "memory");
+}
+
+__device__ __forceinline__ void glds4(const void* gsrc, unsigned lds_off) {
+    unsigned keep;
+    asm volatile(
+        "s_mov_b32 %0, m0\n\t"
+        "s_mov_b32 m0, %2\n\t"
+        "s_nop 0\n\t"
+        "global_load_lds_dword %1, off\n\t"
+        "s_mov_b32 m0, %0"
+        : "=&s"(keep) : "v"(gsrc), "s"(lds_off) : "memory");
+}
+
 }  // namespace
 
 // BM_ in {16, 32}. 256 threads = 4 waves; BM=16 -> 1x4 wave grid (each wave
@@ -144,17 +186,12 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     // xsc: 4 x 64B chunks = 256B dword-glds: chunk c = (kg2, dx|sum),
     //      lane l: chunk l>>4, slot (l&15) of this wave's 16 M-rows.
 
-    int8_t* lds_w[2];
-    int8_t* lds_x[2];
-    int8_t* lds_h[2];
-    int8_t* lds_s[2];
-    #pragma unroll
-    for (int b = 0; b < 2; b++) {
-        lds_w[b] = &lds[b][0];
-        lds_x[b] = &lds[b][0] + WSZ + wid * XSZ;
-        lds_h[b] = &lds[b][0] + WSZ + 4 * XSZ + wid * HSZ;
-        lds_s[b] = &lds[b][0] + WSZ + 4 * (XSZ + HSZ) + wid * SSZ;
-    }
+    // integer LDS offsets only — pointer-typed locals into __shared__
+    // decay to GENERIC address space and the reads compile to flat_load
+    // (caught in this kernel's v3 disassembly: zero ds_read in the loop)
+    const int xo = WSZ + wid * XSZ;
+    const int ho = WSZ + 4 * XSZ + wid * HSZ;
+    const int so = WSZ + 4 * (XSZ + HSZ) + wid * SSZ;
 
     auto issue_tile = [&](int kb, int pb) {
         // W raw (nt: streamed once per step)
@@ -167,36 +204,30 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
             } else {
                 off = wrow_q[g] + kb + wboff[g];
             }
-            __builtin_amdgcn_global_load_lds(
-                reinterpret_cast<const uint32_t*>(qs + off),
-                reinterpret_cast<uint32_t*>(
-                    lds_w[pb] + wid * (NGW * 1024) +
-                    g * 1024),
-                16, 0, 2);
+            glds16_nt(qs + off,
+                      __builtin_amdgcn_readfirstlane((unsigned)(size_t)
+                          &lds[pb][wid * (NGW * 1024) + g * 1024]));
         }
         // X tile copy for this wave
         #pragma unroll
         for (int g = 0; g < NGX; g++) {
-            __builtin_amdgcn_global_load_lds(
-                reinterpret_cast<const uint32_t*>(xq + xrow_off[g] + kb),
-                reinterpret_cast<uint32_t*>(lds_x[pb] + g * 1024), 16, 0, 0);
+            glds16(xq + xrow_off[g] + kb,
+                   __builtin_amdgcn_readfirstlane((unsigned)(size_t)
+                       &lds[pb][xo + g * 1024]));
         }
         // headers
         if constexpr (W == DT::DQ4K) {
             const int grp = kb >> 6;  // q-group index
-            __builtin_amdgcn_global_load_lds(
-                reinterpret_cast<const uint32_t*>(
-                    hdr2 + ((int64_t)grp * N + n0 + 2 * lane) * 8),
-                reinterpret_cast<uint32_t*>(lds_h[pb]), 16, 0, 0);
+            glds16(hdr2 + ((int64_t)grp * N + n0 + 2 * lane) * 8,
+                   __builtin_amdgcn_readfirstlane((unsigned)(size_t)
+                       &lds[pb][ho]));
         } else {
             const int kg = kb >> 5;
             #pragma unroll
             for (int g = 0; g < 2; g++) {
-                __builtin_amdgcn_global_load_lds(
-                    reinterpret_cast<const uint32_t*>(
-                        hdr2 + ((int64_t)(kg + g) * N + n0) * 2 + lane * 4),
-                    reinterpret_cast<uint32_t*>(lds_h[pb] + g * 256),
-                    4, 0, 0);
+                glds4(hdr2 + ((int64_t)(kg + g) * N + n0) * 2 + lane * 4,
+                      __builtin_amdgcn_readfirstlane((unsigned)(size_t)
+                          &lds[pb][ho + g * 256]));
             }
         }
         // activation scales: chunk (kg2 = c>>1, arr = c&1), 16 slots each.
@@ -209,9 +240,9 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
             if (srow > M4 - 1) srow = M4 - 1;
             const int64_t soff =
                 ((int64_t)(kg + (c >> 1)) * 2 + (c & 1)) * M4 + srow;
-            __builtin_amdgcn_global_load_lds(
-                reinterpret_cast<const uint32_t*>(xsc + soff),
-                reinterpret_cast<uint32_t*>(lds_s[pb]), 4, 0, 0);
+            glds4(xsc + soff,
+                  __builtin_amdgcn_readfirstlane((unsigned)(size_t)
+                      &lds[pb][so]));
         }
     };
 
@@ -223,28 +254,28 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         for (int j = 0; j < JF; j++) {
             const int r = wn * (BN / WNW) + j * 16 + lrow;
             rawj[j][0] = *reinterpret_cast<const long*>(
-                lds_w[pb] + r * RAWB + lk * 8);
+                &lds[pb][r * RAWB + lk * 8]);
             if constexpr (W == DT::DQ8) {
                 rawj[j][1] = *reinterpret_cast<const long*>(
-                    lds_w[pb] + r * RAWB + 32 + lk * 8);
-                hdj[j].x = (uint32_t) reinterpret_cast<const uint16_t*>(
-                               lds_h[pb])[r]
-                           | ((uint32_t) reinterpret_cast<const uint16_t*>(
-                                  lds_h[pb] + 256)[r] << 16);
+                    &lds[pb][r * RAWB + 32 + lk * 8]);
+                hdj[j].x = (uint32_t)*reinterpret_cast<const uint16_t*>(
+                               &lds[pb][ho + r * 2])
+                           | ((uint32_t)*reinterpret_cast<const uint16_t*>(
+                                  &lds[pb][ho + 256 + r * 2]) << 16);
                 hdj[j].y = 0;
             } else {
-                hdj[j] = reinterpret_cast<const uint2*>(lds_h[pb])[r];
+                hdj[j] = *reinterpret_cast<const uint2*>(
+                    &lds[pb][ho + r * 8]);
             }
         }
-        const float* sw = reinterpret_cast<const float*>(lds_s[pb]);
         #pragma unroll
         for (int kb2 = 0; kb2 < 2; kb2++) {
             const long a = *reinterpret_cast<const long*>(
-                lds_x[pb] + (wm * 16 + lrow) * BK + kb2 * 32 + lk * 8);
+                &lds[pb][xo + (wm * 16 + lrow) * BK + kb2 * 32 + lk * 8]);
             const float4 dx4 = *reinterpret_cast<const float4*>(
-                sw + (kb2 * 2 + 0) * 16 + lk * 4);
+                &lds[pb][so + ((kb2 * 2 + 0) * 16 + lk * 4) * 4]);
             const float4 sm4 = *reinterpret_cast<const float4*>(
-                sw + (kb2 * 2 + 1) * 16 + lk * 4);
+                &lds[pb][so + ((kb2 * 2 + 1) * 16 + lk * 4) * 4]);
             #pragma unroll
             for (int j = 0; j < JF; j++) {
                 long b;
