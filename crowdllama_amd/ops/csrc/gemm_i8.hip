@@ -127,7 +127,11 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     constexpr int HSZ = (W == DT::DQ4K) ? 1024 : 512;
     constexpr int SSZ = 256;
     constexpr int BUFSZ = WSZ + 4 * (XSZ + HSZ + SSZ);
-    __shared__ __attribute__((aligned(16))) int8_t lds[2][BUFSZ];
+    // 3-deep DMA ring: two tiles stay in flight across every barrier
+    // (guide 'Pipelining across barriers': 3-buf span +83% over 2-buf in
+    // the ~1-block/CU latency regime; here it trades 6 -> 4 WGs/CU of LDS)
+    constexpr int NBUF = 3;
+    __shared__ __attribute__((aligned(16))) int8_t lds[NBUF][BUFSZ];
 
     const int tid = threadIdx.x;
     const int bn = blockIdx.x, bm = blockIdx.y, bz = blockIdx.z;
@@ -304,26 +308,27 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         }
     };
 
-    // ---- DMA pipeline: one tile in flight across every barrier ----
+    // ---- DMA pipeline: two tiles in flight across every barrier ----
     const int kb_last = kb_hi - BK;          // all tiles full (K%BK==0)
+    auto clampkb = [&](int kb) { return kb <= kb_last ? kb : kb_last; };
     issue_tile(kb_lo, 0);
-    issue_tile(kb_lo + BK <= kb_last ? kb_lo + BK : kb_last, 1);
-    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NGLT) : "memory");  // tile 0 in
-    __builtin_amdgcn_s_barrier();
+    issue_tile(clampkb(kb_lo + BK), 1);
+    issue_tile(clampkb(kb_lo + 2 * BK), 2);
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * NGLT) : "memory");
+    __builtin_amdgcn_s_barrier();            // tile 0 landed everywhere
 
     int pb = 0;
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
         mfma_tile(pb);
         if (kb + BK >= kb_hi) break;         // last tile: nothing to stage
         __builtin_amdgcn_s_barrier();        // buf[pb] fully consumed
-        {   // tile t+2 (clamped at the tail; re-staged bytes never read)
-            const int kb2 = kb + 2 * BK;
-            issue_tile(kb2 <= kb_last ? kb2 : kb_last, pb);
-        }
-        // everyone's t+1 landed once own vmcnt hits the t+2 count
-        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NGLT) : "memory");
+        // tile t+3 overwrites buf[pb] (clamped at the tail; the re-staged
+        // bytes are never read)
+        issue_tile(clampkb(kb + 3 * BK), pb);
+        // everyone's t+1 landed once own vmcnt leaves t+2/t+3 in flight
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * NGLT) : "memory");
         __builtin_amdgcn_s_barrier();
-        pb ^= 1;
+        pb = (pb == NBUF - 1) ? 0 : pb + 1;
     }
 
     // ---- epilogue (same contract as gemm.hip k_gemm) ----
