@@ -98,7 +98,7 @@ __device__ __forceinline__ void glds4(const void* gsrc, unsigned lds_off) {
 //
 // xsc is the interleaved activation-scale array [K/32][2][M4]:
 // [kg][0][m] = block scale dx, [kg][1][m] = dx*sum(qx). M4 = M round-to-4.
-template <DT W, int BM_>
+template <DT W, int BM_, int NBUF>
 __global__ __launch_bounds__(256) void k_gemm_i8(
     const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr2,
     const int8_t* __restrict__ xq,    // [M][ldxq] int8 (pre-quantized)
@@ -127,10 +127,8 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     constexpr int HSZ = (W == DT::DQ4K) ? 1024 : 512;
     constexpr int SSZ = 256;
     constexpr int BUFSZ = WSZ + 4 * (XSZ + HSZ + SSZ);
-    // 3-deep DMA ring: two tiles stay in flight across every barrier
-    // (guide 'Pipelining across barriers': 3-buf span +83% over 2-buf in
-    // the ~1-block/CU latency regime; here it trades 6 -> 4 WGs/CU of LDS)
-    constexpr int NBUF = 3;
+    // NBUF-deep DMA ring: NBUF-1 tiles stay in flight across every barrier
+    // (3-buf trades LDS occupancy for latency cover — env CLA_I8_RING A/B)
     __shared__ __attribute__((aligned(16))) int8_t lds[NBUF][BUFSZ];
 
     const int tid = threadIdx.x;
@@ -311,10 +309,10 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     // ---- DMA pipeline: two tiles in flight across every barrier ----
     const int kb_last = kb_hi - BK;          // all tiles full (K%BK==0)
     auto clampkb = [&](int kb) { return kb <= kb_last ? kb : kb_last; };
-    issue_tile(kb_lo, 0);
-    issue_tile(clampkb(kb_lo + BK), 1);
-    issue_tile(clampkb(kb_lo + 2 * BK), 2);
-    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * NGLT) : "memory");
+    #pragma unroll
+    for (int b = 0; b < NBUF; b++)
+        issue_tile(clampkb(kb_lo + b * BK), b);
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"((NBUF - 1) * NGLT) : "memory");
     __builtin_amdgcn_s_barrier();            // tile 0 landed everywhere
 
     int pb = 0;
@@ -322,11 +320,12 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         mfma_tile(pb);
         if (kb + BK >= kb_hi) break;         // last tile: nothing to stage
         __builtin_amdgcn_s_barrier();        // buf[pb] fully consumed
-        // tile t+3 overwrites buf[pb] (clamped at the tail; the re-staged
+        // tile t+NBUF overwrites buf[pb] (clamped at the tail; re-staged
         // bytes are never read)
-        issue_tile(clampkb(kb + 3 * BK), pb);
-        // everyone's t+1 landed once own vmcnt leaves t+2/t+3 in flight
-        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * NGLT) : "memory");
+        issue_tile(clampkb(kb + NBUF * BK), pb);
+        // everyone's t+1 landed once own vmcnt leaves the newer tiles
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"((NBUF - 1) * NGLT)
+                     : "memory");
         __builtin_amdgcn_s_barrier();
         pb = (pb == NBUF - 1) ? 0 : pb + 1;
     }
@@ -434,19 +433,28 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
         force_splitk > 0 ? force_splitk : gemm_splitk_factor(N, K, M);
     const int k_chunk = ((K / BK + splitk - 1) / splitk) * BK;
     dim3 grid(n_tiles, bm_tiles, splitk), block(256);
-    #define GI8_ONE(WT, BMV)                                                   \
-        hipLaunchKernelGGL((k_gemm_i8<WT, BMV>), grid, block, 0, stream,       \
+    static const int ring = [] {   // DMA ring depth A/B knob
+        const char* e = getenv("CLA_I8_RING");
+        const int v = e ? atoi(e) : 2;
+        return (v == 3) ? 3 : 2;
+    }();
+    #define GI8_ONE(WT, BMV, NB)                                               \
+        hipLaunchKernelGGL((k_gemm_i8<WT, BMV, NB>), grid, block, 0, stream,   \
             (const uint8_t*)w.qs, (const uint8_t*)w.hdr2, xq, xsc,             \
             res, C, M, N, K, ldc, ldxq, k_chunk)
+    #define GI8_RING(WT, BMV)                                                  \
+        do { if (ring == 3) GI8_ONE(WT, BMV, 3); else GI8_ONE(WT, BMV, 2); }   \
+        while (0)
     switch (w.dtype) {
         case DT::DQ4K:
-            if (bm16) GI8_ONE(DT::DQ4K, 16); else GI8_ONE(DT::DQ4K, 32);
+            if (bm16) GI8_RING(DT::DQ4K, 16); else GI8_RING(DT::DQ4K, 32);
             break;
         case DT::DQ8:
-            if (bm16) GI8_ONE(DT::DQ8, 16); else GI8_ONE(DT::DQ8, 32);
+            if (bm16) GI8_RING(DT::DQ8, 16); else GI8_RING(DT::DQ8, 32);
             break;
         default: throw std::runtime_error("gemm_i8: quant dtypes only");
     }
+    #undef GI8_RING
     #undef GI8_ONE
 }
 
