@@ -70,15 +70,76 @@ __host__ __device__ inline int64_t dhdr_row_bytes(DT t, int64_t k) {
     return 0;
 }
 
-// Transposed header copy for the i8 GEMM's LDS-DMA scale staging
-// (gemm_i8.hip): headers grouped so one k-window's headers for ALL rows
-// are contiguous. DQ4K: [K/64 q-groups][N][8B pair header];
-// DQ8: [K/32][N][f16 d]. Bytes per row (total = n * this):
+// Transposed, PRE-DECODED header copy for the i8 GEMM's LDS-DMA scale
+// staging (gemm_i8.hip): one k-window's scales for all rows contiguous,
+// already in f32 so the drain does no header math.
+// DQ4K: [K/32][N][{f32 d*sc, f32 dmin*mn}]; DQ8: [K/32][N][f32 d].
+// Bytes per row (total = n * this):
 __host__ __device__ inline int64_t dhdr2_row_bytes(DT t, int64_t k) {
     switch (t) {
-        case DT::DQ4K: return k / 64 * 8;
-        case DT::DQ8: return k / 32 * 2;
+        case DT::DQ4K: return k / 32 * 8;
+        case DT::DQ8: return k / 32 * 4;
         default: return 0;
+    }
+}
+
+// host-side f16 bits -> f32 (GGUF block scales)
+inline float f16_bits_to_f32_host(uint16_t h) {
+    const uint32_t s = (uint32_t)(h >> 15) & 1u;
+    const uint32_t e = (uint32_t)(h >> 10) & 31u;
+    const uint32_t m = (uint32_t)h & 1023u;
+    uint32_t out;
+    if (e == 0) {
+        if (m == 0) {
+            out = s << 31;
+        } else {
+            int ex = -1;
+            uint32_t mm = m;
+            while (!(mm & 1024u)) { mm <<= 1; ex--; }
+            out = (s << 31) | ((uint32_t)(127 - 15 + 1 + ex) << 23) |
+                  ((mm & 1023u) << 13);
+        }
+    } else if (e == 31) {
+        out = (s << 31) | 0x7F800000u | (m << 13);
+    } else {
+        out = (s << 31) | ((e - 15 + 127) << 23) | (m << 13);
+    }
+    union { uint32_t u; float f; } v;
+    v.u = out;
+    return v.f;
+}
+
+// Decode one row-range of hdr into the hdr2 layout above (shared by
+// Engine::upload_pack and the gemm_i8 test harness).
+inline void build_hdr2_rows(DT t, const uint8_t* hdr, int64_t hrb,
+                            int64_t rows_total, int64_t k, int64_t r_lo,
+                            int64_t r_hi, uint8_t* out) {
+    const int64_t nb = k / 32;
+    if (t == DT::DQ4K) {
+        for (int64_t r = r_lo; r < r_hi; r++) {
+            const uint8_t* h = hdr + r * hrb;
+            for (int64_t kg = 0; kg < nb; kg++) {
+                const uint8_t* e = h + (kg >> 3) * 32 + ((kg & 7) >> 1) * 8;
+                const float d = f16_bits_to_f32_host(
+                    *reinterpret_cast<const uint16_t*>(e));
+                const float dmin = f16_bits_to_f32_host(
+                    *reinterpret_cast<const uint16_t*>(e + 2));
+                const int half = (int)(kg & 1);
+                float* dst = reinterpret_cast<float*>(
+                    out + ((size_t)kg * rows_total + r) * 8);
+                dst[0] = d * (float)e[4 + 2 * half];
+                dst[1] = dmin * (float)e[5 + 2 * half];
+            }
+        }
+    } else if (t == DT::DQ8) {
+        for (int64_t r = r_lo; r < r_hi; r++) {
+            const uint16_t* h = reinterpret_cast<const uint16_t*>(
+                hdr + r * hrb);
+            for (int64_t b = 0; b < nb; b++)
+                *reinterpret_cast<float*>(
+                    out + ((size_t)b * rows_total + r) * 4) =
+                    f16_bits_to_f32_host(h[b]);
+        }
     }
 }
 

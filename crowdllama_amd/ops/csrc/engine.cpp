@@ -337,8 +337,8 @@ WTensor Engine::upload_pack(const void* qs, size_t qs_bytes, const void* hdr,
         w.hdr = d_hdr;
         vram_bytes_ += hdr_bytes;
     }
-    // transposed header copy for the i8 GEMM's glds scale staging: one
-    // k-window's headers for all rows contiguous (see common.h)
+    // transposed, pre-decoded f32 header copy for the i8 GEMM's glds
+    // scale staging (see common.h build_hdr2_rows)
     const int64_t h2_rb = dhdr2_row_bytes(dtype, k);
     if (h2_rb) {
         // +1 KB slack: edge-tile glds lanes past the last row read (and
@@ -346,23 +346,9 @@ WTensor Engine::upload_pack(const void* qs, size_t qs_bytes, const void* hdr,
         std::vector<uint8_t> h2((size_t)h2_rb * rows + 1024, 0);
         const uint8_t* hsrc = reinterpret_cast<const uint8_t*>(hdr);
         const int64_t hrb = dhdr_row_bytes(dtype, k);
-        if (dtype == DT::DQ4K) {
-            const int64_t ng = k / 64;   // q-groups; 8B pair header each
-            parallel_for(rows, [&](int64_t lo, int64_t hi) {
-                for (int64_t r = lo; r < hi; r++)
-                    for (int64_t g = 0; g < ng; g++)
-                        std::memcpy(h2.data() + (g * rows + r) * 8,
-                                    hsrc + r * hrb + g * 8, 8);
-            });
-        } else {  // DQ8: [K/32][rows] f16 d
-            const int64_t nb = k / 32;
-            parallel_for(rows, [&](int64_t lo, int64_t hi) {
-                for (int64_t r = lo; r < hi; r++)
-                    for (int64_t b = 0; b < nb; b++)
-                        std::memcpy(h2.data() + (b * rows + r) * 2,
-                                    hsrc + r * hrb + b * 2, 2);
-            });
-        }
+        parallel_for(rows, [&](int64_t lo, int64_t hi) {
+            build_hdr2_rows(dtype, hsrc, hrb, rows, k, lo, hi, h2.data());
+        });
         void* d_h2 = nullptr;
         HIP_CHECK(hipMalloc(&d_h2, h2.size()));
         HIP_CHECK(hipMemcpy(d_h2, h2.data(), h2.size(),

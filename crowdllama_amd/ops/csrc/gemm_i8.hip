@@ -98,7 +98,7 @@ __device__ __forceinline__ void glds4(const void* gsrc, unsigned lds_off) {
 //
 // xsc is the interleaved activation-scale array [K/32][2][M4]:
 // [kg][0][m] = block scale dx, [kg][1][m] = dx*sum(qx). M4 = M round-to-4.
-template <DT W, int BM_, int NBUF>
+template <DT W, int BM_>
 __global__ __launch_bounds__(256) void k_gemm_i8(
     const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr2,
     const int8_t* __restrict__ xq,    // [M][ldxq] int8 (pre-quantized)
@@ -112,7 +112,7 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     constexpr int RAWB = (W == DT::DQ4K) ? BK / 2 : BK;  // raw bytes/row
     constexpr int NGW = (BN / 4) * RAWB / 1024;  // W glds per wave (1 or 2)
     constexpr int NGX = BM_ * BK / 1024;         // X glds per wave (1 or 2)
-    constexpr int NGH = (W == DT::DQ4K) ? 1 : 2; // hdr2 glds per wave
+    constexpr int NGH = (W == DT::DQ4K) ? JF : JF / 2;  // hdr2 glds/wave
     constexpr int NGS = 1;                       // xsc glds per wave
     constexpr int NGLT = NGW + NGX + NGH + NGS;  // per wave per tile
     // --- single LDS array (a second __shared__ object would make hipcc
@@ -124,11 +124,11 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     //   [.., +4*256)                  per-wave xsc copies [kg2][2][16] f32
     constexpr int WSZ = BN * RAWB;
     constexpr int XSZ = BM_ * BK;
-    constexpr int HSZ = (W == DT::DQ4K) ? 1024 : 512;
+    constexpr int HSZ = (W == DT::DQ4K) ? JF * 256 : JF * 128;
     constexpr int SSZ = 256;
     constexpr int BUFSZ = WSZ + 4 * (XSZ + HSZ + SSZ);
-    // NBUF-deep DMA ring: NBUF-1 tiles stay in flight across every barrier
-    // (3-buf trades LDS occupancy for latency cover — env CLA_I8_RING A/B)
+    // 3-deep DMA ring (required by the single-barrier pipeline below)
+    constexpr int NBUF = 3;
     __shared__ __attribute__((aligned(16))) int8_t lds[NBUF][BUFSZ];
 
     const int tid = threadIdx.x;
@@ -217,19 +217,34 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
                    __builtin_amdgcn_readfirstlane((unsigned)(size_t)
                        &lds[pb][xo + g * 1024]));
         }
-        // headers
-        if constexpr (W == DT::DQ4K) {
-            const int grp = kb >> 6;  // q-group index
-            glds16(hdr2 + ((int64_t)grp * N + n0 + 2 * lane) * 8,
-                   __builtin_amdgcn_readfirstlane((unsigned)(size_t)
-                       &lds[pb][ho]));
-        } else {
+        // headers: hdr2 is pre-decoded f32 ({d*sc, dmin*mn} pairs for
+        // Q4K, d for Q8) laid out [K/32][N]; each wave stages only its own
+        // fragment columns
+        {
             const int kg = kb >> 5;
-            #pragma unroll
-            for (int g = 0; g < 2; g++) {
-                glds4(hdr2 + ((int64_t)(kg + g) * N + n0) * 2 + lane * 4,
-                      __builtin_amdgcn_readfirstlane((unsigned)(size_t)
-                          &lds[pb][ho + g * 256]));
+            const int cb = n0 + wn * (BN / WNW);   // this wave's first col
+            if constexpr (W == DT::DQ4K) {
+                #pragma unroll
+                for (int kb2 = 0; kb2 < 2; kb2++)
+                    #pragma unroll
+                    for (int gg = 0; gg < JF / 2; gg++)
+                        glds4(hdr2 + ((int64_t)(kg + kb2) * N + cb +
+                                      gg * 32 + (lane >> 1)) * 8 +
+                                  (lane & 1) * 4,
+                              __builtin_amdgcn_readfirstlane(
+                                  (unsigned)(size_t)&lds[pb][
+                                      ho + kb2 * (JF * 128) + gg * 256]));
+            } else if constexpr (JF == 2) {  // Q8 BM16: one glds, both kb2
+                glds4(hdr2 + ((int64_t)(kg + (lane >> 5)) * N + cb +
+                              (lane & 31)) * 4,
+                      __builtin_amdgcn_readfirstlane(
+                          (unsigned)(size_t)&lds[pb][ho]));
+            } else {                         // Q8 BM32: one glds per kb2
+                #pragma unroll
+                for (int kb2 = 0; kb2 < 2; kb2++)
+                    glds4(hdr2 + ((int64_t)(kg + kb2) * N + cb + lane) * 4,
+                          __builtin_amdgcn_readfirstlane(
+                              (unsigned)(size_t)&lds[pb][ho + kb2 * 256]));
             }
         }
         // activation scales: chunk (kg2 = c>>1, arr = c&1), 16 slots each.
@@ -251,24 +266,14 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     auto mfma_tile = [&](int pb) {
         // raw W fragment bytes: one b64 per j serves both K=32 halves (Q4K)
         long rawj[JF][(W == DT::DQ4K) ? 1 : 2];
-        uint2 hdj[JF];
         #pragma unroll
         for (int j = 0; j < JF; j++) {
             const int r = wn * (BN / WNW) + j * 16 + lrow;
             rawj[j][0] = *reinterpret_cast<const long*>(
                 &lds[pb][r * RAWB + lk * 8]);
-            if constexpr (W == DT::DQ8) {
+            if constexpr (W == DT::DQ8)
                 rawj[j][1] = *reinterpret_cast<const long*>(
                     &lds[pb][r * RAWB + 32 + lk * 8]);
-                hdj[j].x = (uint32_t)*reinterpret_cast<const uint16_t*>(
-                               &lds[pb][ho + r * 2])
-                           | ((uint32_t)*reinterpret_cast<const uint16_t*>(
-                                  &lds[pb][ho + 256 + r * 2]) << 16);
-                hdj[j].y = 0;
-            } else {
-                hdj[j] = *reinterpret_cast<const uint2*>(
-                    &lds[pb][ho + r * 8]);
-            }
         }
         #pragma unroll
         for (int kb2 = 0; kb2 < 2; kb2++) {
@@ -280,20 +285,21 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
                 &lds[pb][so + ((kb2 * 2 + 1) * 16 + lk * 4) * 4]);
             #pragma unroll
             for (int j = 0; j < JF; j++) {
+                const int cl = j * 16 + lrow;   // wave-local column
                 long b;
                 float d, m;
                 if constexpr (W == DT::DQ4K) {
                     b = (kb2 == 0)
                             ? (rawj[j][0] & 0x0F0F0F0F0F0F0F0FLL)
                             : ((rawj[j][0] >> 4) & 0x0F0F0F0F0F0F0F0FLL);
-                    const float dd = f16b2f(hdj[j].x & 0xFFFF);
-                    const float dmin = f16b2f(hdj[j].x >> 16);
-                    const uint32_t y = hdj[j].y;
-                    d = dd * (float)((y >> (16 * kb2)) & 0xFF);
-                    m = dmin * (float)((y >> (16 * kb2 + 8)) & 0xFF);
+                    const float2 dm = *reinterpret_cast<const float2*>(
+                        &lds[pb][ho + kb2 * (JF * 128) + cl * 8]);
+                    d = dm.x;
+                    m = dm.y;
                 } else {
                     b = rawj[j][kb2];
-                    d = f16b2f((hdj[j].x >> (16 * kb2)) & 0xFFFF);
+                    d = *reinterpret_cast<const float*>(
+                        &lds[pb][ho + kb2 * (JF * 64) + cl * 4]);
                     m = 0.f;
                 }
                 v4i c = {0, 0, 0, 0};
@@ -306,27 +312,24 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         }
     };
 
-    // ---- DMA pipeline: two tiles in flight across every barrier ----
+    // ---- DMA pipeline: ONE barrier per tile, two tiles in flight ----
+    // At iteration t: wait own tile t (vmcnt leaves t+1), barrier (now
+    // everyone's t landed AND everyone finished reading t-1, so buffer
+    // (t+2)%3 is free), issue t+2 into it, compute t. The issue for t+2
+    // happens before mfma(t), so its DMA has a full tile of work to hide
+    // under, and the single barrier serves both release and acquire.
     const int kb_last = kb_hi - BK;          // all tiles full (K%BK==0)
-    auto clampkb = [&](int kb) { return kb <= kb_last ? kb : kb_last; };
-    #pragma unroll
-    for (int b = 0; b < NBUF; b++)
-        issue_tile(clampkb(kb_lo + b * BK), b);
-    asm volatile("s_waitcnt vmcnt(%0)" ::"i"((NBUF - 1) * NGLT) : "memory");
-    __builtin_amdgcn_s_barrier();            // tile 0 landed everywhere
-
+    issue_tile(kb_lo, 0);
+    issue_tile(kb_lo + BK <= kb_last ? kb_lo + BK : kb_lo, 1);
     int pb = 0;
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
-        mfma_tile(pb);
-        if (kb + BK >= kb_hi) break;         // last tile: nothing to stage
-        __builtin_amdgcn_s_barrier();        // buf[pb] fully consumed
-        // tile t+NBUF overwrites buf[pb] (clamped at the tail; re-staged
-        // bytes are never read)
-        issue_tile(clampkb(kb + NBUF * BK), pb);
-        // everyone's t+1 landed once own vmcnt leaves the newer tiles
-        asm volatile("s_waitcnt vmcnt(%0)" ::"i"((NBUF - 1) * NGLT)
-                     : "memory");
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NGLT) : "memory");
         __builtin_amdgcn_s_barrier();
+        if (kb + 2 * BK <= kb_last) {
+            const int nb = pb + 2 >= NBUF ? pb + 2 - NBUF : pb + 2;
+            issue_tile(kb + 2 * BK, nb);
+        }
+        mfma_tile(pb);
         pb = (pb == NBUF - 1) ? 0 : pb + 1;
     }
 
@@ -433,28 +436,19 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
         force_splitk > 0 ? force_splitk : gemm_splitk_factor(N, K, M);
     const int k_chunk = ((K / BK + splitk - 1) / splitk) * BK;
     dim3 grid(n_tiles, bm_tiles, splitk), block(256);
-    static const int ring = [] {   // DMA ring depth A/B knob
-        const char* e = getenv("CLA_I8_RING");
-        const int v = e ? atoi(e) : 2;
-        return (v == 3) ? 3 : 2;
-    }();
-    #define GI8_ONE(WT, BMV, NB)                                               \
-        hipLaunchKernelGGL((k_gemm_i8<WT, BMV, NB>), grid, block, 0, stream,   \
+    #define GI8_ONE(WT, BMV)                                                   \
+        hipLaunchKernelGGL((k_gemm_i8<WT, BMV>), grid, block, 0, stream,       \
             (const uint8_t*)w.qs, (const uint8_t*)w.hdr2, xq, xsc,             \
             res, C, M, N, K, ldc, ldxq, k_chunk)
-    #define GI8_RING(WT, BMV)                                                  \
-        do { if (ring == 3) GI8_ONE(WT, BMV, 3); else GI8_ONE(WT, BMV, 2); }   \
-        while (0)
     switch (w.dtype) {
         case DT::DQ4K:
-            if (bm16) GI8_RING(DT::DQ4K, 16); else GI8_RING(DT::DQ4K, 32);
+            if (bm16) GI8_ONE(DT::DQ4K, 16); else GI8_ONE(DT::DQ4K, 32);
             break;
         case DT::DQ8:
-            if (bm16) GI8_RING(DT::DQ8, 16); else GI8_RING(DT::DQ8, 32);
+            if (bm16) GI8_ONE(DT::DQ8, 16); else GI8_ONE(DT::DQ8, 32);
             break;
         default: throw std::runtime_error("gemm_i8: quant dtypes only");
     }
-    #undef GI8_RING
     #undef GI8_ONE
 }
 
@@ -506,24 +500,14 @@ void launch_gemm_i8_test(const void* qs, const void* hdr, const float* x,
     void *d_qs = nullptr, *d_h2 = nullptr;
     HIP_CHECK(hipMalloc(&d_qs, qs_bytes));
     HIP_CHECK(hipMemcpy(d_qs, qs, qs_bytes, hipMemcpyHostToDevice));
-    // transpose headers -> hdr2 (same as Engine::upload_pack)
+    // transposed pre-decoded headers (same builder Engine::upload_pack uses)
     {
         const int64_t h2_rb = dhdr2_row_bytes(dt, K);
         const int64_t hrb = dhdr_row_bytes(dt, K);
         std::vector<uint8_t> h2((size_t)h2_rb * N + 1024, 0);
-        const uint8_t* hsrc = reinterpret_cast<const uint8_t*>(hdr);
         (void)hdr_bytes;
-        if (dt == DT::DQ4K) {
-            for (int64_t r = 0; r < N; r++)
-                for (int64_t g = 0; g < K / 64; g++)
-                    std::memcpy(h2.data() + (g * N + r) * 8,
-                                hsrc + r * hrb + g * 8, 8);
-        } else {
-            for (int64_t r = 0; r < N; r++)
-                for (int64_t b = 0; b < K / 32; b++)
-                    std::memcpy(h2.data() + (b * N + r) * 2,
-                                hsrc + r * hrb + b * 2, 2);
-        }
+        build_hdr2_rows(dt, reinterpret_cast<const uint8_t*>(hdr), hrb, N,
+                        K, 0, N, h2.data());
         HIP_CHECK(hipMalloc(&d_h2, h2.size()));
         HIP_CHECK(hipMemcpy(d_h2, h2.data(), h2.size(),
                             hipMemcpyHostToDevice));
