@@ -143,6 +143,57 @@ inline void build_hdr2_rows(DT t, const uint8_t* hdr, int64_t hrb,
     }
 }
 
+// i8-GEMM tile geometry (gemm_i8.hip) — shared with the upload-time
+// tiled repack below.
+constexpr int I8G_BN = 128;   // rows (output features) per tile
+constexpr int I8G_BK = 64;    // k per tile
+
+// Raw quant bytes per row per BK window.
+__host__ __device__ inline int64_t i8g_rawb(DT t) {
+    return t == DT::DQ4K ? I8G_BK / 2 : I8G_BK;
+}
+
+// GEMM-tiled weight copy: [ceil(N/128)][K/64][128 rows][RAWB] — each
+// (n-block, k-window) tile is contiguous (= the kernel's LDS image), so
+// the weight DMA reads whole cachelines. The row-major qs layout reads
+// only 32 B per 128 B line at BK=64 windows (4x HBM over-fetch — the
+// round-2 ~1.5 TB/s wall on both GEMM kernels). Rows past N are zeros.
+__host__ __device__ inline int64_t dqs2_bytes(DT t, int64_t n, int64_t k) {
+    if (t != DT::DQ4K && t != DT::DQ8) return 0;
+    const int64_t nb = (n + I8G_BN - 1) / I8G_BN;
+    return nb * (k / I8G_BK) * I8G_BN * i8g_rawb(t);
+}
+
+// Build the tiled copy for a row range of the (already repacked,
+// row-major) qs buffer.
+inline void build_qs2_rows(DT t, const uint8_t* qs, int64_t qs_rb,
+                           int64_t n, int64_t k, int64_t r_lo, int64_t r_hi,
+                           uint8_t* out) {
+    const int64_t rawb = i8g_rawb(t);
+    const int64_t ktiles = k / I8G_BK;
+    const int64_t tile_bytes = I8G_BN * rawb;
+    for (int64_t r = r_lo; r < r_hi; r++) {
+        const int64_t nb = r / I8G_BN, rl = r % I8G_BN;
+        for (int64_t kt = 0; kt < ktiles; kt++) {
+            uint8_t* dst = out + (nb * ktiles + kt) * tile_bytes + rl * rawb;
+            if (r >= n) {
+                for (int64_t b = 0; b < rawb; b++) dst[b] = 0;
+                continue;
+            }
+            const int64_t kb = kt * I8G_BK;
+            const uint8_t* src;
+            if (t == DT::DQ4K) {
+                // qs row layout: [nsb][128B]; BK window = q-group of 64 =
+                // 32 contiguous bytes at sb*128 + q*32
+                src = qs + r * qs_rb + (kb >> 8) * 128 + ((kb & 255) >> 6) * 32;
+            } else {
+                src = qs + r * qs_rb + kb;
+            }
+            for (int64_t b = 0; b < rawb; b++) dst[b] = src[b];
+        }
+    }
+}
+
 // A weight matrix on device: N rows of K quantized columns.
 struct WTensor {
     DT dtype = DT::F32;
@@ -150,7 +201,8 @@ struct WTensor {
     int64_t k = 0;   // cols (input features)
     const void* qs = nullptr;
     const void* hdr = nullptr;
-    const void* hdr2 = nullptr;  // transposed headers (quant GEMM dtypes)
+    const void* hdr2 = nullptr;  // transposed pre-decoded headers (i8 GEMM)
+    const void* qs2 = nullptr;   // GEMM-tiled weight copy (i8 GEMM)
 };
 
 }  // namespace cla

@@ -357,6 +357,25 @@ WTensor Engine::upload_pack(const void* qs, size_t qs_bytes, const void* hdr,
         w.hdr2 = d_h2;
         vram_bytes_ += h2.size();
     }
+    // GEMM-tiled weight copy (see common.h: the row-major layout
+    // over-fetches 4x at BK windows; this one DMAs whole cachelines)
+    const int64_t q2_bytes = dqs2_bytes(dtype, rows, k);
+    if (q2_bytes) {
+        std::vector<uint8_t> q2((size_t)q2_bytes);
+        const uint8_t* qsrc = reinterpret_cast<const uint8_t*>(qs);
+        const int64_t qrb = dqs_row_bytes(dtype, k);
+        const int64_t rows_pad = (rows + I8G_BN - 1) / I8G_BN * I8G_BN;
+        parallel_for(rows_pad, [&](int64_t lo, int64_t hi) {
+            build_qs2_rows(dtype, qsrc, qrb, rows, k, lo, hi, q2.data());
+        });
+        void* d_q2 = nullptr;
+        HIP_CHECK(hipMalloc(&d_q2, q2.size()));
+        HIP_CHECK(hipMemcpy(d_q2, q2.data(), q2.size(),
+                            hipMemcpyHostToDevice));
+        allocs_.push_back(d_q2);
+        w.qs2 = d_q2;
+        vram_bytes_ += q2.size();
+    }
     return w;
 }
 

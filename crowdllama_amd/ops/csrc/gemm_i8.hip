@@ -100,7 +100,7 @@ __device__ __forceinline__ void glds4(const void* gsrc, unsigned lds_off) {
 // [kg][0][m] = block scale dx, [kg][1][m] = dx*sum(qx). M4 = M round-to-4.
 template <DT W, int BM_>
 __global__ __launch_bounds__(256) void k_gemm_i8(
-    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr2,
+    const uint8_t* __restrict__ qs2, const uint8_t* __restrict__ hdr2,
     const int8_t* __restrict__ xq,    // [M][ldxq] int8 (pre-quantized)
     const float* __restrict__ xsc,    // [ldxq/32][2][M4]
     const float* __restrict__ res,    // [M][ldc] or null
@@ -150,27 +150,13 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         #pragma unroll
         for (int r = 0; r < 4; r++) facc[j][r] = 0.f;
 
-    const int64_t qs_rb = dqs_row_bytes(W, K);
-    const int64_t h2_rb = dhdr2_row_bytes(W, K);
-    (void)h2_rb;
-
-    // ---- per-lane global source addresses (row parts precomputed) ----
-    // W: wave wid owns rows [wid*32, wid*32+32).
-    int64_t wrow_q[2];   // clamped global row per W glds
-    int wboff[2];
-    #pragma unroll
-    for (int g = 0; g < NGW; g++) {
-        int rl;
-        if constexpr (W == DT::DQ4K) {
-            rl = wid * 32 + (lane >> 1);
-            wboff[g] = (lane & 1) * 16;
-        } else {
-            rl = wid * 32 + g * 16 + (lane >> 2);
-            wboff[g] = (lane & 3) * 16;
-        }
-        const int64_t gn = (int64_t)n0 + rl;
-        wrow_q[g] = (gn < N ? gn : N - 1) * qs_rb;
-    }
+    // ---- W source: the GEMM-tiled qs2 copy (common.h) — each (n-block,
+    // k-window) tile is contiguous and identical to the LDS image, so a
+    // wave's DMA covers whole cachelines (the row-major layout read only
+    // 32 of every 128 B here: the round-2 bandwidth wall).
+    const int64_t ktiles = K / BK;
+    const uint8_t* wtile0 = qs2 + (int64_t)bn * ktiles * (BN * RAWB);
+    const int wlocal = wid * (NGW * 1024) + lane * 16;  // == LDS offset
     // X: each wave DMAs its own copy of the whole [BM_][64] tile.
     // glds gx covers rows gx*16 + (lane>>2), 16B quarter lane&3.
     int64_t xrow_off[2];
@@ -196,19 +182,14 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     const int so = WSZ + 4 * (XSZ + HSZ) + wid * SSZ;
 
     auto issue_tile = [&](int kb, int pb) {
-        // W raw (nt: streamed once per step)
-        #pragma unroll
-        for (int g = 0; g < NGW; g++) {
-            int64_t off;
-            if constexpr (W == DT::DQ4K) {
-                const int sb = kb >> 8, q = (kb & 255) >> 6;
-                off = wrow_q[g] + sb * 128 + q * 32 + wboff[g];
-            } else {
-                off = wrow_q[g] + kb + wboff[g];
-            }
-            glds16_nt(qs + off,
-                      __builtin_amdgcn_readfirstlane((unsigned)(size_t)
-                          &lds[pb][wid * (NGW * 1024) + g * 1024]));
+        // W raw (nt: streamed once per step); source offset == LDS offset
+        {
+            const uint8_t* wt = wtile0 + (int64_t)(kb / BK) * (BN * RAWB);
+            #pragma unroll
+            for (int g = 0; g < NGW; g++)
+                glds16_nt(wt + wlocal + g * 1024,
+                          __builtin_amdgcn_readfirstlane((unsigned)(size_t)
+                              &lds[pb][wid * (NGW * 1024) + g * 1024]));
         }
         // X tile copy for this wave
         #pragma unroll
@@ -427,7 +408,7 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
                     int ldxq, const float* res, float* C, int M, int ldc,
                     hipStream_t stream, int force_splitk) {
     const int N = (int)w.n, K = (int)w.k;
-    if (!gemm_i8_supported(w.dtype, M, K) || !w.hdr2)
+    if (!gemm_i8_supported(w.dtype, M, K) || !w.hdr2 || !w.qs2)
         throw std::runtime_error("gemm_i8: unsupported dtype/shape");
     const bool bm16 = M <= 16;
     const int bm_tiles = bm16 ? 1 : (M + 31) / 32;
@@ -438,7 +419,7 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
     dim3 grid(n_tiles, bm_tiles, splitk), block(256);
     #define GI8_ONE(WT, BMV)                                                   \
         hipLaunchKernelGGL((k_gemm_i8<WT, BMV>), grid, block, 0, stream,       \
-            (const uint8_t*)w.qs, (const uint8_t*)w.hdr2, xq, xsc,             \
+            (const uint8_t*)w.qs2, (const uint8_t*)w.hdr2, xq, xsc,            \
             res, C, M, N, K, ldc, ldxq, k_chunk)
     switch (w.dtype) {
         case DT::DQ4K:
@@ -497,9 +478,18 @@ void launch_gemm_i8_test(const void* qs, const void* hdr, const float* x,
                          int force_splitk) {
     const DT dt = static_cast<DT>(dtype);
     const int M4 = (M + 3) & ~3;
-    void *d_qs = nullptr, *d_h2 = nullptr;
+    void *d_qs = nullptr, *d_h2 = nullptr, *d_q2 = nullptr;
     HIP_CHECK(hipMalloc(&d_qs, qs_bytes));
     HIP_CHECK(hipMemcpy(d_qs, qs, qs_bytes, hipMemcpyHostToDevice));
+    {   // GEMM-tiled weight copy
+        std::vector<uint8_t> q2((size_t)dqs2_bytes(dt, N, K));
+        build_qs2_rows(dt, reinterpret_cast<const uint8_t*>(qs),
+                       dqs_row_bytes(dt, K), N, K, 0,
+                       (N + I8G_BN - 1) / I8G_BN * I8G_BN, q2.data());
+        HIP_CHECK(hipMalloc(&d_q2, q2.size()));
+        HIP_CHECK(hipMemcpy(d_q2, q2.data(), q2.size(),
+                            hipMemcpyHostToDevice));
+    }
     // transposed pre-decoded headers (same builder Engine::upload_pack uses)
     {
         const int64_t h2_rb = dhdr2_row_bytes(dt, K);
@@ -524,7 +514,7 @@ void launch_gemm_i8_test(const void* qs, const void* hdr, const float* x,
     launch_quant_rows(d_x, d_xq, d_xsc, M, K, K, 0, nullptr);
     WTensor w;
     w.dtype = dt; w.n = N; w.k = K; w.qs = d_qs; w.hdr = nullptr;
-    w.hdr2 = d_h2;
+    w.hdr2 = d_h2; w.qs2 = d_q2;
     const int sk = force_splitk > 0 ? force_splitk
                                     : gemm_splitk_factor(N, K, M);
     if (sk > 1)
@@ -533,7 +523,8 @@ void launch_gemm_i8_test(const void* qs, const void* hdr, const float* x,
                    force_splitk);
     HIP_CHECK(hipDeviceSynchronize());
     HIP_CHECK(hipMemcpy(y, d_y, (size_t)M * N * 4, hipMemcpyDeviceToHost));
-    (void)hipFree(d_qs); (void)hipFree(d_h2); (void)hipFree(d_x);
+    (void)hipFree(d_qs); (void)hipFree(d_h2); (void)hipFree(d_q2);
+    (void)hipFree(d_x);
     (void)hipFree(d_xq); (void)hipFree(d_xsc); (void)hipFree(d_y);
 }
 
