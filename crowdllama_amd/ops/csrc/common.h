@@ -194,6 +194,41 @@ inline void build_qs2_rows(DT t, const uint8_t* qs, int64_t qs_rb,
     }
 }
 
+#ifdef __HIPCC__
+// Fused activation block-quantizer (8 lanes per 32-block): emits the int8
+// row and the interleaved transposed scales ([K/32][2][M4]) the i8 GEMM's
+// DMA staging expects (gemm_i8.hip). Call with jl = tid&7 and the block's
+// 4 values in v; all 8 lanes of the group must call together.
+__device__ __forceinline__ void quant_block_emit(
+    float4 v, int jl, int blk, int m, int K, int M4,
+    int8_t* __restrict__ xq, float* __restrict__ xsc) {
+    float amax = fmaxf(fmaxf(fabsf(v.x), fabsf(v.y)),
+                       fmaxf(fabsf(v.z), fabsf(v.w)));
+    #pragma unroll
+    for (int off = 1; off < 8; off <<= 1)
+        amax = fmaxf(amax, __shfl_xor(amax, off, 64));
+    const float scale = amax / 127.f;
+    const float rinv = amax > 0.f ? 127.f / amax : 0.f;
+    const int q0 = (int)rintf(v.x * rinv);
+    const int q1 = (int)rintf(v.y * rinv);
+    const int q2 = (int)rintf(v.z * rinv);
+    const int q3 = (int)rintf(v.w * rinv);
+    const uint32_t packed =
+        (uint32_t)(q0 & 0xFF) | ((uint32_t)(q1 & 0xFF) << 8) |
+        ((uint32_t)(q2 & 0xFF) << 16) | ((uint32_t)(q3 & 0xFF) << 24);
+    int sq = q0 + q1 + q2 + q3;
+    #pragma unroll
+    for (int off = 1; off < 8; off <<= 1) sq += __shfl_xor(sq, off, 64);
+    *reinterpret_cast<uint32_t*>(xq + (size_t)m * K + blk * 32 + jl * 4) =
+        packed;
+    if (jl == 0) {
+        xsc[((size_t)blk * 2 + 0) * M4 + m] = scale;
+        xsc[((size_t)blk * 2 + 1) * M4 + m] = scale * (float)sq;
+    }
+}
+
+#endif  // __HIPCC__
+
 // A weight matrix on device: N rows of K quantized columns.
 struct WTensor {
     DT dtype = DT::F32;

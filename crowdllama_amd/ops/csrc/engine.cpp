@@ -34,9 +34,10 @@ void launch_attn_decode(const float* qkv, const float* inv_freq,
                         const int32_t* page_table,
                         uint16_t* kv_pool, const int32_t* n_past,
                         float* part_o, float* part_ml, int* tickets,
-                        float* attn_out, int B, int NH, int NKV,
-                        int D, int S, int page_size, int max_pages,
-                        int64_t page_stride, float scale, hipStream_t);
+                        float* attn_out, int8_t* xq, float* xsc, int B,
+                        int NH, int NKV, int D, int S, int page_size,
+                        int max_pages, int64_t page_stride, float scale,
+                        hipStream_t);
 void launch_argmax(const float* logits, float* pval, int32_t* pidx,
                    int32_t* cur_ids, int32_t* n_past, int32_t* gen_tokens,
                    int32_t* gen_count, const uint8_t* slot_active, int B,
@@ -50,9 +51,12 @@ bool gemm_uses_splitk(int N, int K, int M);
 void launch_layer_prep(const float* X, const float* gw, float* xn, int B,
                        int K, float eps, float* p0, int64_t n0, float* p1,
                        int64_t n1, float* p2, int64_t n2, float* p3,
-                       int64_t n3, hipStream_t);
+                       int64_t n3, int8_t* xq, float* xsc, hipStream_t);
 void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
                          int K, float eps, hipStream_t);
+void launch_rmsnorm_rows_q(const float* X, const float* gw, float* out,
+                           int M, int K, float eps, int8_t* xq, float* xsc,
+                           hipStream_t);
 void launch_scatter_logits(const float* src, float* dst, int B, int Vl,
                            int TP, hipStream_t);
 // gemm_i8.hip: int8-activation MFMA path for quantized weights
@@ -608,11 +612,18 @@ struct QBufs {
 // kernel (gemm_i8.hip): activations are block-quantized once here (silu
 // fused into the quantizer for the down projection), float parts and
 // unsupported shapes fall back to the bf16-staging kernel.
+static bool proj_i8_ok(const Proj& p, int M) {
+    bool ok = false;
+    for (auto& pt : p.parts)
+        ok |= gemm_i8_supported(pt.w.dtype, M, (int)pt.w.k);
+    return ok;
+}
+
 static void gemm_proj(const Proj& p, const float* X, const float* res,
                       float* C, int M, hipStream_t s,
                       bool pre_zeroed = false, const float* X2 = nullptr,
                       int ldx = -1, bool xsilu = false,
-                      const QBufs* qb = nullptr) {
+                      const QBufs* qb = nullptr, bool pre_quant = false) {
     bool zero = false;
     for (auto& pt : p.parts)
         zero |= gemm_uses_splitk((int)pt.w.n, (int)pt.w.k, M);
@@ -623,7 +634,7 @@ static void gemm_proj(const Proj& p, const float* X, const float* res,
     if (qb)
         for (auto& pt : p.parts)
             any_i8 |= gemm_i8_supported(pt.w.dtype, M, (int)pt.w.k);
-    if (any_i8)
+    if (any_i8 && !pre_quant)
         launch_quant_rows(X, qb->xq, qb->xsc, M, Kq,
                           ldx < 0 ? Kq : ldx, xsilu ? 1 : 0, s);
     for (auto& pt : p.parts) {
@@ -697,16 +708,20 @@ void Engine::step(hipStream_t s) {
             // (replaces 3-4 ~5 us hipMemsetAsync dispatches).
             float* lin = ((li - 1) & 1) ? x2_ : x_;  // li already advanced
             float* lout = ((li - 1) & 1) ? x_ : x2_;
+            const bool fq = qb && proj_i8_ok(L.qkv, B);
             launch_layer_prep(lin, L.attn_norm, xn_, B, meta_.hidden, eps,
                               qkv_, (int64_t)B * (NH + 2 * NKV) * D,
                               gu_, (int64_t)B * 2 * meta_.ffn_l,
                               x3_, (int64_t)B * meta_.hidden,
-                              lout, (int64_t)B * meta_.hidden, s);
+                              lout, (int64_t)B * meta_.hidden,
+                              fq ? xq_ : nullptr, fq ? xsc_ : nullptr, s);
             gemm_proj(L.qkv, xn_, nullptr, qkv_, B, s, /*pre_zeroed=*/true,
-                      nullptr, -1, false, qb);
+                      nullptr, -1, false, qb, /*pre_quant=*/fq);
         }
+        const bool fqo = !gemv_path && qb && proj_i8_ok(L.o, B);
         launch_attn_decode(qkv_, inv_freq_, page_table_, kv_layer, n_past_,
-                           part_o_, part_ml_, attn_tickets_, attn_out_, B, NH,
+                           part_o_, part_ml_, attn_tickets_, attn_out_,
+                           fqo ? xq_ : nullptr, fqo ? xsc_ : nullptr, B, NH,
                            NKV, D, attn_splits_, cfg_.page_size, max_pages_,
                            page_stride_, scale, s);
         if (gemv_path && !tp) {
@@ -740,15 +755,21 @@ void Engine::step(hipStream_t s) {
             float* lout = ((li - 1) & 1) ? x_ : x2_;
             if (tp) {
                 gemm_proj(L.o, attn_out_, r0 ? lin : nullptr, tmp_h_, B, s,
-                          false, nullptr, -1, false, qb);
-                allreduce(tmp_h_, x3_, (size_t)B * meta_.hidden);
+                          false, nullptr, -1, false, qb, fqo);
             } else {
                 gemm_proj(L.o, attn_out_, lin, x3_, B, s, /*pre_zeroed=*/true,
-                          nullptr, -1, false, qb);
+                          nullptr, -1, false, qb, fqo);
             }
-            launch_rmsnorm_rows(x3_, L.ffn_norm, xn_, B, meta_.hidden, eps, s);
+            if (tp) allreduce(tmp_h_, x3_, (size_t)B * meta_.hidden);
+            const bool fqg = qb && proj_i8_ok(L.gate_up, B);
+            if (fqg)
+                launch_rmsnorm_rows_q(x3_, L.ffn_norm, xn_, B, meta_.hidden,
+                                      eps, xq_, xsc_, s);
+            else
+                launch_rmsnorm_rows(x3_, L.ffn_norm, xn_, B, meta_.hidden,
+                                    eps, s);
             gemm_proj(L.gate_up, xn_, nullptr, gu_, B, s, /*pre_zeroed=*/true,
-                      nullptr, -1, false, qb);
+                      nullptr, -1, false, qb, fqg);
             // silu fused into the down GEMM's X staging (gate | up halves;
             // i8 path: fused into the activation quantizer instead)
             if (tp) {
@@ -774,13 +795,19 @@ void Engine::step(hipStream_t s) {
         gemv_pick(cfg_.act_q8, head_, PRE_RMS, x_, out_norm_, nullptr, lg, B,
                     ldl, eps, s);
     } else {
-        launch_rmsnorm_rows(xfinal, out_norm_, xn_, B, meta_.hidden, eps, s);
+        const bool fqh = qb && gemm_i8_supported(head_.dtype, B,
+                                                 (int)head_.k) &&
+                         head_.hdr2 && head_.qs2;
+        if (fqh)
+            launch_rmsnorm_rows_q(xfinal, out_norm_, xn_, B, meta_.hidden,
+                                  eps, xq_, xsc_, s);
+        else
+            launch_rmsnorm_rows(xfinal, out_norm_, xn_, B, meta_.hidden,
+                                eps, s);
         bool zero = gemm_uses_splitk((int)head_.n, (int)head_.k, B);
         if (zero)
             HIP_CHECK(hipMemsetAsync(lg, 0, (size_t)B * ldl * 4, s));
-        if (qb && gemm_i8_supported(head_.dtype, B, (int)head_.k)) {
-            launch_quant_rows(xn_, xq_, xsc_, B, (int)head_.k,
-                              (int)head_.k, 0, s);
+        if (fqh) {
             launch_gemm_i8(head_, xq_, xsc_, (int)head_.k, nullptr,
                            lg, B, ldl, s);
         } else {

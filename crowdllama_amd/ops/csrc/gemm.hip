@@ -441,10 +441,12 @@ __global__ __launch_bounds__(256) void k_gemm(
 
 // ---- row-wise elementwise kernels for the GEMM (prefill/batched) path ----
 
-// grid M; block 256. out[m] = rmsnorm(x[m]) * gw
+// grid M; block 256. out[m] = rmsnorm(x[m]) * gw; optionally emits the
+// fused i8-GEMM activation quantization (xq non-null — common.h helper).
 __global__ __launch_bounds__(256) void k_rmsnorm_rows(
     const float* __restrict__ X, const float* __restrict__ gw,
-    float* __restrict__ out, int K, float eps) {
+    float* __restrict__ out, int K, float eps,
+    int8_t* __restrict__ xq, float* __restrict__ xsc, int M4) {
     const int m = blockIdx.x;
     const float4* x4 = reinterpret_cast<const float4*>(X + (size_t)m * K);
     float4* o4 = reinterpret_cast<float4*>(out + (size_t)m * K);
@@ -464,13 +466,27 @@ __global__ __launch_bounds__(256) void k_rmsnorm_rows(
         __syncthreads();
     }
     const float inv = rsqrtf(red[0] / (float)K + eps);
-    for (int k = threadIdx.x; k < K4; k += 256) {
-        const float4 v = x4[k];
-        const float4 g = g4[k];
-        float4 o;
-        o.x = v.x * inv * g.x; o.y = v.y * inv * g.y;
-        o.z = v.z * inv * g.z; o.w = v.w * inv * g.w;
-        o4[k] = o;
+    if (xq) {
+        const int jl = threadIdx.x & 7;
+        for (int b32 = threadIdx.x >> 3; b32 < K / 32; b32 += 32) {
+            const int k = b32 * 8 + jl;
+            const float4 v = x4[k];
+            const float4 g = g4[k];
+            float4 o;
+            o.x = v.x * inv * g.x; o.y = v.y * inv * g.y;
+            o.z = v.z * inv * g.z; o.w = v.w * inv * g.w;
+            o4[k] = o;
+            quant_block_emit(o, jl, b32, m, K, M4, xq, xsc);
+        }
+    } else {
+        for (int k = threadIdx.x; k < K4; k += 256) {
+            const float4 v = x4[k];
+            const float4 g = g4[k];
+            float4 o;
+            o.x = v.x * inv * g.x; o.y = v.y * inv * g.y;
+            o.z = v.z * inv * g.z; o.w = v.w * inv * g.w;
+            o4[k] = o;
+        }
     }
 }
 
@@ -687,7 +703,17 @@ void launch_gemm(const WTensor& w, const float* X, const float* res, float* C,
 void launch_rmsnorm_rows(const float* X, const float* gw, float* out, int M,
                          int K, float eps, hipStream_t stream) {
     hipLaunchKernelGGL(k_rmsnorm_rows, dim3(M), dim3(256), 0, stream,
-                       X, gw, out, K, eps);
+                       X, gw, out, K, eps, (int8_t*)nullptr,
+                       (float*)nullptr, 0);
+}
+
+// fused-quant variant: also emits xq/xsc for the i8 GEMM consumer
+void launch_rmsnorm_rows_q(const float* X, const float* gw, float* out,
+                           int M, int K, float eps, int8_t* xq, float* xsc,
+                           hipStream_t stream) {
+    const int M4 = (M + 3) & ~3;
+    hipLaunchKernelGGL(k_rmsnorm_rows, dim3(M), dim3(256), 0, stream,
+                       X, gw, out, K, eps, xq, xsc, M4);
 }
 
 

@@ -707,6 +707,7 @@ __global__ __launch_bounds__(256) void k_attn_decode(
     float* __restrict__ part_ml,  // [B][NH][S][2]
     int* __restrict__ tickets,    // [B][NKV] fan-in counters (0 on entry)
     float* __restrict__ attn_out, // [B][NH*D]
+    int8_t* __restrict__ xq, float* __restrict__ xsc, int M4,  // fused quant
     int NH, int NKV, int S, int page_size, int max_pages,
     int64_t page_stride, float scale) {
     constexpr int DPL = D / 16;   // dims per lane (bf16: 2*DPL bytes)
@@ -942,6 +943,37 @@ __global__ __launch_bounds__(256) void k_attn_decode(
     }
     __syncthreads();
     if (!is_last) return;
+    if (xq) {
+        // combine + fused i8-GEMM activation quantization (8 lanes per
+        // 32-block; D%32==0 so a block never crosses heads)
+        const int jl = tid & 7;
+        for (int b32 = tid >> 3; b32 < G * D / 32; b32 += 32) {
+            const int g = (b32 * 32) / D;
+            const int head = kvh * G + g;
+            const float* ml = part_ml + ((size_t)b * NH + head) * S * 2;
+            float mg = -1e30f;
+            for (int t = 0; t < S; t++) mg = fmaxf(mg, ml[2 * t]);
+            float denom = 0.f;
+            for (int t = 0; t < S; t++)
+                denom += __expf(ml[2 * t] - mg) * ml[2 * t + 1];
+            float4 v;
+            float* vp = &v.x;
+            #pragma unroll
+            for (int j = 0; j < 4; j++) {
+                const int d = (b32 * 32 + jl * 4 + j) % D;
+                float osum = 0.f;
+                #pragma unroll 4
+                for (int t = 0; t < S; t++)
+                    osum += __expf(ml[2 * t] - mg) *
+                            part_o[(((size_t)b * NH + head) * S + t) * D + d];
+                vp[j] = osum / denom;
+                attn_out[(size_t)b * NH * D + (size_t)head * D + d] = vp[j];
+            }
+            const int kg = kvh * (G * D / 32) + b32;
+            quant_block_emit(v, jl, kg, b, NH * D, M4, xq, xsc);
+        }
+        return;
+    }
     for (int idx = tid; idx < G * D; idx += 256) {
         const int g = idx / D, d = idx % D;
         const int head = kvh * G + g;
@@ -1065,6 +1097,7 @@ void launch_scatter_logits(const float* src, float* dst, int B, int Vl,
                        src, dst, B, Vl);
 }
 
+
 // Clear up to 4 device regions in one launch (replaces several
 // hipMemsetAsync nodes ahead of split-K GEMMs: each memset costs a
 
@@ -1075,10 +1108,13 @@ __global__ __launch_bounds__(256) void k_layer_prep(
     const float* __restrict__ X, const float* __restrict__ gw,
     float* __restrict__ xn, int B, int K, float eps,
     float* __restrict__ p0, int n0, float* __restrict__ p1, int n1,
-    float* __restrict__ p2, int n2, float* __restrict__ p3, int n3) {
+    float* __restrict__ p2, int n2, float* __restrict__ p3, int n3,
+    int8_t* __restrict__ xq, float* __restrict__ xsc, int M4) {
     const int blk = blockIdx.x;
     if (blk < B) {
-        // rmsnorm row blk (same structure as k_rmsnorm_rows)
+        // rmsnorm row blk (same structure as k_rmsnorm_rows), with the
+        // i8-GEMM activation quantization fused (xq non-null): saves one
+        // kernel launch + a full re-read of xn per projection input
         const float4* x4 = reinterpret_cast<const float4*>(X + (size_t)blk * K);
         float4* o4 = reinterpret_cast<float4*>(xn + (size_t)blk * K);
         const float4* g4 = reinterpret_cast<const float4*>(gw);
@@ -1097,13 +1133,27 @@ __global__ __launch_bounds__(256) void k_layer_prep(
             __syncthreads();
         }
         const float inv = rsqrtf(red[0] / (float)K + eps);
-        for (int k = threadIdx.x; k < K4; k += 256) {
-            const float4 v = x4[k];
-            const float4 g = g4[k];
-            float4 o;
-            o.x = v.x * inv * g.x; o.y = v.y * inv * g.y;
-            o.z = v.z * inv * g.z; o.w = v.w * inv * g.w;
-            o4[k] = o;
+        if (xq) {
+            const int jl = threadIdx.x & 7;
+            for (int b32 = threadIdx.x >> 3; b32 < K / 32; b32 += 32) {
+                const int k = b32 * 8 + jl;
+                const float4 v = x4[k];
+                const float4 g = g4[k];
+                float4 o;
+                o.x = v.x * inv * g.x; o.y = v.y * inv * g.y;
+                o.z = v.z * inv * g.z; o.w = v.w * inv * g.w;
+                o4[k] = o;
+                quant_block_emit(o, jl, b32, blk, K, M4, xq, xsc);
+            }
+        } else {
+            for (int k = threadIdx.x; k < K4; k += 256) {
+                const float4 v = x4[k];
+                const float4 g = g4[k];
+                float4 o;
+                o.x = v.x * inv * g.x; o.y = v.y * inv * g.y;
+                o.z = v.z * inv * g.z; o.w = v.w * inv * g.w;
+                o4[k] = o;
+            }
         }
         return;
     }
@@ -1123,14 +1173,16 @@ __global__ __launch_bounds__(256) void k_layer_prep(
 void launch_layer_prep(const float* X, const float* gw, float* xn, int B,
                        int K, float eps, float* p0, int64_t n0, float* p1,
                        int64_t n1, float* p2, int64_t n2, float* p3,
-                       int64_t n3, hipStream_t stream) {
+                       int64_t n3, int8_t* xq, float* xsc,
+                       hipStream_t stream) {
     const int64_t total = (n0 + n1 + n2 + n3) >> 2;
     int zb = (int)((total + 255) / 256);
     if (zb > 1024) zb = 1024;
     if (zb < 1) zb = 1;
+    const int M4 = (B + 3) & ~3;
     hipLaunchKernelGGL(k_layer_prep, dim3(B + zb), dim3(256), 0, stream,
                        X, gw, xn, B, K, eps, p0, (int)n0, p1, (int)n1,
-                       p2, (int)n2, p3, (int)n3);
+                       p2, (int)n2, p3, (int)n3, xq, xsc, M4);
 }
 
 
@@ -1236,16 +1288,18 @@ void launch_attn_decode(const float* qkv, const float* inv_freq,
                         const int32_t* page_table,
                         uint16_t* kv_pool, const int32_t* n_past,
                         float* part_o, float* part_ml, int* tickets,
-                        float* attn_out, int B, int NH, int NKV,
-                        int D, int S, int page_size, int max_pages,
-                        int64_t page_stride, float scale, hipStream_t stream) {
+                        float* attn_out, int8_t* xq, float* xsc, int B,
+                        int NH, int NKV, int D, int S, int page_size,
+                        int max_pages, int64_t page_stride, float scale,
+                        hipStream_t stream) {
     const int G = NH / NKV;
+    const int M4 = (B + 3) & ~3;
     dim3 grid(S, NKV, B), block(256);
     #define ATTN_CASE(GV, DV)                                                   \
         hipLaunchKernelGGL((k_attn_decode<GV, DV>), grid, block, 0, stream,     \
             qkv, inv_freq, page_table, kv_pool, n_past, part_o, part_ml,        \
-            tickets, attn_out, NH, NKV, S, page_size, max_pages, page_stride,   \
-            scale)
+            tickets, attn_out, xq, xsc, M4, NH, NKV, S, page_size, max_pages,   \
+            page_stride, scale)
     #define ATTN_D(GV)                                                          \
         do { if (D == 128) ATTN_CASE(GV, 128);                                  \
              else if (D == 64) ATTN_CASE(GV, 64);                               \
