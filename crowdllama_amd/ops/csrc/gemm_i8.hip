@@ -127,8 +127,10 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     constexpr int HSZ = (W == DT::DQ4K) ? JF * 256 : JF * 128;
     constexpr int SSZ = 256;
     constexpr int BUFSZ = WSZ + 4 * (XSZ + HSZ + SSZ);
-    // 3-deep DMA ring (required by the single-barrier pipeline below)
-    constexpr int NBUF = 3;
+    // 4-deep DMA ring: the single-barrier pipeline below keeps NBUF-2
+    // tiles fully in flight, so each tile's DMA has ~2 iterations of
+    // compute to hide under (PMC: 1-deep exposed ~38% parked)
+    constexpr int NBUF = 4;
     __shared__ __attribute__((aligned(16))) int8_t lds[NBUF][BUFSZ];
 
     const int tid = threadIdx.x;
@@ -300,15 +302,17 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     // happens before mfma(t), so its DMA has a full tile of work to hide
     // under, and the single barrier serves both release and acquire.
     const int kb_last = kb_hi - BK;          // all tiles full (K%BK==0)
-    issue_tile(kb_lo, 0);
-    issue_tile(kb_lo + BK <= kb_last ? kb_lo + BK : kb_lo, 1);
+    #pragma unroll
+    for (int b = 0; b < NBUF - 1; b++)
+        issue_tile(kb_lo + b * BK <= kb_last ? kb_lo + b * BK : kb_lo, b);
     int pb = 0;
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
-        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NGLT) : "memory");
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"((NBUF - 2) * NGLT)
+                     : "memory");
         __builtin_amdgcn_s_barrier();
-        if (kb + 2 * BK <= kb_last) {
-            const int nb = pb + 2 >= NBUF ? pb + 2 - NBUF : pb + 2;
-            issue_tile(kb + 2 * BK, nb);
+        if (kb + (NBUF - 1) * BK <= kb_last) {
+            const int nb = pb + NBUF - 1 >= NBUF ? pb - 1 : pb + NBUF - 1;
+            issue_tile(kb + (NBUF - 1) * BK, nb);
         }
         mfma_tile(pb);
         pb = (pb == NBUF - 1) ? 0 : pb + 1;
