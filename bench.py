@@ -45,9 +45,12 @@ def main() -> None:
                          "single-stream latency point)")
     ap.add_argument("--prompt-len", type=int, default=128)
     ap.add_argument("--max-seq", type=int, default=4096)
-    ap.add_argument("--act-q8", action="store_true",
-                    help="int8-quantized activations on the quantized-weight "
-                         "GEMV path (B=1 v_dot4 path A/B)")
+    ap.add_argument("--act-q8", dest="act_q8", action="store_true",
+                    default=True,
+                    help="int8-quantized activations for quantized-weight "
+                         "GEMMs (default on; the i8 MFMA batched path)")
+    ap.add_argument("--no-act-q8", dest="act_q8", action="store_false",
+                    help="force the f32-activation bf16-staging GEMM path")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -160,7 +163,12 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no numbers
-            "dtype": args.scheme + "+f32-activations",
+            # act_q8: per-32-block int8 activations for quantized-weight
+            # GEMMs — the same activation-quantization scheme llama.cpp's
+            # MMQ path (the reference's compute backend) uses for these
+            # weight formats; --no-act-q8 forces f32 activations
+            "dtype": args.scheme + ("+int8-activations" if args.act_q8
+                                    else "+f32-activations"),
             "data": "synthetic",
             "config": {
                 "model": args.model,

@@ -99,6 +99,7 @@ PYBIND11_MODULE(_core, m) {
         .def_readwrite("tp_rank", &EngineConfig::tp_rank)
         .def_readwrite("tp_size", &EngineConfig::tp_size)
         .def_readwrite("act_q8", &EngineConfig::act_q8)
+        .def_readwrite("gemv_q8", &EngineConfig::gemv_q8)
         .def_property("nccl_id",
             [](EngineConfig& c) { return py::bytes(c.nccl_id); },
             [](EngineConfig& c, py::bytes b) { c.nccl_id = std::string(b); });

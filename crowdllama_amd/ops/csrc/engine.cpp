@@ -700,7 +700,7 @@ void Engine::step(hipStream_t s) {
         li++;
         if (gemv_path) {
             for (auto& pt : L.qkv.parts)
-                gemv_pick(cfg_.act_q8, pt.w, PRE_RMS, x_, L.attn_norm, nullptr,
+                gemv_pick(cfg_.gemv_q8, pt.w, PRE_RMS, x_, L.attn_norm, nullptr,
                             qkv_ + pt.row_off, B, (int)L.qkv.n_total, eps, s);
         } else {
             // 3-buffer rotation: in --(+attn)--> mid --(+ffn)--> out, with
@@ -726,27 +726,27 @@ void Engine::step(hipStream_t s) {
                            page_stride_, scale, s);
         if (gemv_path && !tp) {
             for (auto& pt : L.o.parts)
-                gemv_pick(cfg_.act_q8, pt.w, PRE_NONE, attn_out_, nullptr, x_ + pt.row_off,
+                gemv_pick(cfg_.gemv_q8, pt.w, PRE_NONE, attn_out_, nullptr, x_ + pt.row_off,
                             x_ + pt.row_off, B, (int)L.o.n_total, eps, s);
             for (auto& pt : L.gate_up.parts)
-                gemv_pick(cfg_.act_q8, pt.w, PRE_RMS, x_, L.ffn_norm, nullptr,
+                gemv_pick(cfg_.gemv_q8, pt.w, PRE_RMS, x_, L.ffn_norm, nullptr,
                             gu_ + pt.row_off, B, (int)L.gate_up.n_total, eps, s);
             for (auto& pt : L.down.parts)
-                gemv_pick(cfg_.act_q8, pt.w, PRE_SILU, gu_, nullptr, x_ + pt.row_off,
+                gemv_pick(cfg_.gemv_q8, pt.w, PRE_SILU, gu_, nullptr, x_ + pt.row_off,
                             x_ + pt.row_off, B, (int)L.down.n_total, eps, s);
         } else if (gemv_path) {
             // TP: local partial -> all-reduce; rank 0 folds the residual so
             // the summed result is residual + sum(partials) on every rank.
             for (auto& pt : L.o.parts)
-                gemv_pick(cfg_.act_q8, pt.w, PRE_NONE, attn_out_, nullptr,
+                gemv_pick(cfg_.gemv_q8, pt.w, PRE_NONE, attn_out_, nullptr,
                             r0 ? x_ + pt.row_off : nullptr, tmp_h_ + pt.row_off,
                             B, (int)L.o.n_total, eps, s);
             allreduce(tmp_h_, x2_, (size_t)B * meta_.hidden);
             for (auto& pt : L.gate_up.parts)
-                gemv_pick(cfg_.act_q8, pt.w, PRE_RMS, x2_, L.ffn_norm, nullptr,
+                gemv_pick(cfg_.gemv_q8, pt.w, PRE_RMS, x2_, L.ffn_norm, nullptr,
                             gu_ + pt.row_off, B, (int)L.gate_up.n_total, eps, s);
             for (auto& pt : L.down.parts)
-                gemv_pick(cfg_.act_q8, pt.w, PRE_SILU, gu_, nullptr,
+                gemv_pick(cfg_.gemv_q8, pt.w, PRE_SILU, gu_, nullptr,
                             r0 ? x2_ + pt.row_off : nullptr, tmp_h_ + pt.row_off,
                             B, (int)L.down.n_total, eps, s);
             allreduce(tmp_h_, x_, (size_t)B * meta_.hidden);
@@ -792,7 +792,7 @@ void Engine::step(hipStream_t s) {
     float* lg = tp ? logits_tp_ : logits_;
     const int ldl = tp ? meta_.vocab_l : meta_.vocab;
     if (gemv_path) {
-        gemv_pick(cfg_.act_q8, head_, PRE_RMS, x_, out_norm_, nullptr, lg, B,
+        gemv_pick(cfg_.gemv_q8, head_, PRE_RMS, x_, out_norm_, nullptr, lg, B,
                     ldl, eps, s);
     } else {
         const bool fqh = qb && gemm_i8_supported(head_.dtype, B,
@@ -929,7 +929,7 @@ void Engine::prefill_slot(int slot, const std::vector<int32_t>& ids) {
     // logits of the LAST prompt row -> slot's logits; sample + advance state
     const int last = ((len - 1) % cfg_.prefill_chunk);
     const int64_t voff = (int64_t)cfg_.tp_rank * meta_.vocab_l;
-    gemv_pick(cfg_.act_q8, head_, PRE_RMS, xp_ + (size_t)last * meta_.hidden, out_norm_,
+    gemv_pick(cfg_.gemv_q8, head_, PRE_RMS, xp_ + (size_t)last * meta_.hidden, out_norm_,
                 nullptr, logits_ + (size_t)slot * meta_.vocab + voff, 1,
                 meta_.vocab, eps, stream_);
     if (cfg_.tp_size > 1)

@@ -28,12 +28,16 @@ struct EngineConfig {
     // reference has no collectives at all — SURVEY.md §2.3)
     int tp_rank = 0;
     int tp_size = 1;
-    // int8-quantized activations (per-32 symmetric) for quantized-weight
-    // GEMVs (v_dot4 path; semantics replicated by ref_numpy(act_q8=True)).
-    // Off by default: measured slower than the f32 path this round — the
-    // per-thread quantization staging serializes (round-2 item: lane-
-    // parallel block quantization).
-    bool act_q8 = false;
+    // int8-quantized activations (per-32 symmetric; semantics replicated
+    // by ref_numpy(act_q8=True)) for quantized-weight GEMMs — the
+    // v_mfma_i32 batched-decode path (gemm_i8.hip). Default ON: measured
+    // +33% at B=16 over the bf16-staging GEMM (round 2). Exact-f32
+    // comparisons (tests) set this false.
+    bool act_q8 = true;
+    // int8-activation GEMV (v_dot4) for the B=1 decode path. Off: measured
+    // slower than the f32 GEMV (273 vs 292 tok/s llama3-8b B=1) even with
+    // lane-parallel staging — B=1 is bandwidth-, not issue-bound.
+    bool gemv_q8 = false;
     std::string nccl_id;       // ncclUniqueId bytes (rank 0 creates)
 };
 

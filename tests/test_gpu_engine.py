@@ -80,7 +80,7 @@ def test_engine_batch2(core, tiny_gguf):
     cfg = core.EngineConfig()
     cfg.batch = 2
     cfg.max_seq = 128
-    eng = core.Engine(tiny_gguf, cfg)
+    eng = core.Engine(tiny_gguf, cfg)  # act_q8 default: the serving path
     prompts = np.array([[3, 17, 99], [3, 17, 99]], dtype=np.int32)
     eng.prefill(prompts)
     eng.decode(5)
@@ -166,6 +166,7 @@ def test_engine_slot_parking(core, tiny_gguf):
     cfg = core.EngineConfig()
     cfg.batch = 2
     cfg.max_seq = 64
+    cfg.act_q8 = False  # compare against the B=1 f32 GEMV path below
     eng = core.Engine(tiny_gguf, cfg)
     eng.prefill(np.asarray([prompt, prompt], dtype=np.int32))
     eng.set_slot_active(1, False)
@@ -178,6 +179,7 @@ def test_engine_slot_parking(core, tiny_gguf):
     cfg1 = core.EngineConfig()
     cfg1.batch = 1
     cfg1.max_seq = 64
+    cfg1.act_q8 = False
     e1 = core.Engine(tiny_gguf, cfg1)
     e1.prefill(np.asarray([prompt], dtype=np.int32))
     e1.decode(5)
@@ -215,6 +217,7 @@ def test_engine_batched_gemm_path(core, tiny_gguf):
     cfg1 = core.EngineConfig()
     cfg1.batch = 1
     cfg1.max_seq = 64
+    cfg1.act_q8 = False  # f32-path equivalence (i8 has its own tests)
     e1 = core.Engine(tiny_gguf, cfg1)
     e1.prefill(np.asarray([prompt], dtype=np.int32))
     e1.decode(5)
@@ -223,6 +226,7 @@ def test_engine_batched_gemm_path(core, tiny_gguf):
     cfg4 = core.EngineConfig()
     cfg4.batch = 4
     cfg4.max_seq = 64
+    cfg4.act_q8 = False
     e4 = core.Engine(tiny_gguf, cfg4)
     e4.prefill(np.asarray([prompt] * 4, dtype=np.int32))
     e4.decode(5)
@@ -302,6 +306,7 @@ def test_engine_mha_g1(core, tmp_path):
     ec = core.EngineConfig()
     ec.batch = 1
     ec.max_seq = 128
+    ec.act_q8 = False  # exact-greedy vs the f32 numpy reference
     eng = core.Engine(path, ec)
     want = RefLlama(path).generate([5, 9, 2, 7], 8)
     eng.prefill(np.asarray([[5, 9, 2, 7]], dtype=np.int32))
