@@ -155,8 +155,10 @@ async def _run_dht(args) -> None:
 
 async def _run_network_status(args) -> None:
     cfg = _mk_config(args)
+    from .keys import load_identity
     from .mesh.discovery import Discovery
-    disco = Discovery(cfg.bootstrap_peers)
+    disco = Discovery(cfg.bootstrap_peers,
+                      load_identity("consumer", cfg.key_path))
     ok = await disco.bootstrap_ok()
     print(f"bootstrap reachable: {ok}")
     if ok:

@@ -16,8 +16,9 @@ import time
 from dataclasses import dataclass, field
 
 from ..config import Config
+from ..keys import load_identity
 from ..logutil import new_app_logger
-from .wire import (PROTO_RENDEZVOUS, accept_protocol, read_json, write_json)
+from .wire import PROTO_RENDEZVOUS, secure_accept
 
 
 @dataclass
@@ -32,9 +33,14 @@ class DHTServer:
 
     PROVIDER_TTL = 30.0  # records expire without re-provide (ref: 1 s loop)
 
-    def __init__(self, cfg: Config, peer_id: str):
+    def __init__(self, cfg: Config, peer_id: str = ""):
         self.cfg = cfg
-        self.peer_id = peer_id
+        # real cryptographic identity from the keyfile (the legacy second
+        # argument is kept for compatibility but the advertised id is
+        # always derived from the key — ids are now verifiable hashes of
+        # ed25519 public keys)
+        self.identity = load_identity("dht", cfg.key_path)
+        self.peer_id = self.identity.peer_id
         self.log = new_app_logger("dht", cfg.verbose)
         self.providers: dict[str, dict[str, ProviderRecord]] = {}  # ns -> id -> rec
         self.peer_addrs: dict[str, list[str]] = {}
@@ -76,25 +82,27 @@ class DHTServer:
         self.conn_count += 1
         self.total_conns += 1
         try:
-            proto = await accept_protocol(reader)
+            # authenticated, encrypted channel: the caller's peer_id is
+            # cryptographically verified before any op is served
+            ss, proto = await secure_accept(reader, writer, self.identity)
             if proto != PROTO_RENDEZVOUS:
                 self.log.warning("unknown protocol %r", proto)
                 return
             while True:
                 try:
-                    msg = await read_json(reader, timeout=60.0)
+                    msg = await ss.read_json(timeout=60.0)
                 except (asyncio.IncompleteReadError, asyncio.TimeoutError,
-                        ConnectionError):
+                        ConnectionError, ValueError):
                     return
-                resp = self._handle(msg)
-                await write_json(writer, resp)
+                resp = self._handle(msg, ss.peer_id)
+                await ss.write_json(resp)
         except Exception as e:  # noqa: BLE001 — per-conn isolation
             self.log.debug("conn error: %s", e)
         finally:
             self.conn_count -= 1
             writer.close()
 
-    def _handle(self, msg: dict) -> dict:
+    def _handle(self, msg: dict, caller_id: str = "") -> dict:
         op = msg.get("op")
         if op == "ping":
             return {"ok": True, "peer_id": self.peer_id}
@@ -104,6 +112,11 @@ class DHTServer:
             addrs = list(msg.get("addrs", []))
             if not ns or not pid:
                 return {"ok": False, "error": "missing ns/peer_id"}
+            if caller_id and pid != caller_id:
+                # a peer may only advertise ITS OWN (proven) identity
+                return {"ok": False,
+                        "error": f"peer_id {pid} does not match "
+                                 f"authenticated identity"}
             self.providers.setdefault(ns, {})[pid] = ProviderRecord(
                 pid, addrs)
             self.peer_addrs[pid] = addrs
@@ -128,6 +141,8 @@ class DHTServer:
             return {"ok": True, "addrs": addrs}
         if op == "remove":
             pid = msg.get("peer_id", "")
+            if caller_id and pid != caller_id:
+                return {"ok": False, "error": "cannot remove another peer"}
             for ns in self.providers.values():
                 ns.pop(pid, None)
             self.peer_addrs.pop(pid, None)

@@ -13,7 +13,7 @@ import time
 
 from .resource import Resource
 from .wire import (NAMESPACE, PROTO_METADATA, PROTO_RENDEZVOUS,
-                   open_protocol, read_json, write_json)
+                   SecureStream, secure_open)
 
 
 def parse_addr(addr: str) -> tuple[str, int]:
@@ -22,41 +22,38 @@ def parse_addr(addr: str) -> tuple[str, int]:
 
 
 class RendezvousClient:
-    """Persistent connection to one bootstrap/rendezvous node."""
+    """Persistent, encrypted connection to one bootstrap node."""
 
-    def __init__(self, addr: str, timeout: float = 5.0):
+    def __init__(self, addr: str, identity, timeout: float = 5.0):
         self.addr = addr
+        self.identity = identity
         self.timeout = timeout
-        self._reader: asyncio.StreamReader | None = None
-        self._writer: asyncio.StreamWriter | None = None
+        self._ss: SecureStream | None = None
         self._lock = asyncio.Lock()
 
     async def _ensure(self):
-        if self._writer is None or self._writer.is_closing():
+        if self._ss is None or self._ss.is_closing():
             host, port = parse_addr(self.addr)
-            self._reader, self._writer = await open_protocol(
-                host, port, PROTO_RENDEZVOUS, self.timeout)
+            self._ss = await secure_open(host, port, PROTO_RENDEZVOUS,
+                                         self.identity, self.timeout)
 
     async def call(self, msg: dict) -> dict:
         async with self._lock:
             try:
                 await self._ensure()
-                await write_json(self._writer, msg)
-                return await read_json(self._reader, self.timeout)
+                await self._ss.write_json(msg)
+                return await self._ss.read_json(self.timeout)
             except Exception:
                 # one reconnect attempt per call
                 await self.close()
                 await self._ensure()
-                await write_json(self._writer, msg)
-                return await read_json(self._reader, self.timeout)
+                await self._ss.write_json(msg)
+                return await self._ss.read_json(self.timeout)
 
     async def close(self):
-        if self._writer is not None:
-            try:
-                self._writer.close()
-            except Exception:
-                pass
-        self._reader = self._writer = None
+        if self._ss is not None:
+            self._ss.close()
+        self._ss = None
 
     async def ping(self) -> bool:
         try:
@@ -89,11 +86,13 @@ class RendezvousClient:
 class Discovery:
     """Multi-bootstrap discovery + metadata fetch (reference Discovery)."""
 
-    def __init__(self, bootstrap_addrs: list[str],
+    def __init__(self, bootstrap_addrs: list[str], identity,
                  metadata_timeout: float = 5.0,
                  metadata_max_age: float = 3600.0,
                  log: logging.Logger | None = None):
-        self.clients = [RendezvousClient(a) for a in bootstrap_addrs]
+        self.identity = identity
+        self.clients = [RendezvousClient(a, identity)
+                        for a in bootstrap_addrs]
         self.metadata_timeout = metadata_timeout
         self.metadata_max_age = metadata_max_age
         self.log = log or logging.getLogger("discovery")
@@ -143,24 +142,32 @@ class Discovery:
                 continue
         return None
 
-    async def request_metadata(self, addrs: list[str]) -> Resource:
-        """Open a metadata stream and read EOF-delimited Resource JSON
-        (reference discovery.go:186-275)."""
+    async def request_metadata(self, addrs: list[str],
+                               expected_peer_id: str | None = None
+                               ) -> Resource:
+        """Open a metadata stream and read one signed Resource frame
+        (reference discovery.go:186-275 reads EOF-delimited JSON over a
+        libp2p-secured stream; here the channel handshake authenticates
+        the peer AND the record carries its own signature)."""
         last_err: Exception | None = None
         for addr in addrs:
             host, port = parse_addr(addr)
             try:
-                reader, writer = await open_protocol(
-                    host, port, PROTO_METADATA, self.metadata_timeout)
+                ss = await secure_open(host, port, PROTO_METADATA,
+                                       self.identity, self.metadata_timeout,
+                                       expected_peer_id=expected_peer_id)
                 try:
-                    data = await asyncio.wait_for(reader.read(64 * 1024),
-                                                  self.metadata_timeout)
+                    data = await ss.read_frame(self.metadata_timeout)
                     res = Resource.from_json(data)
+                    if not res.verify():
+                        raise ValueError("resource signature invalid")
+                    if res.peer_id != ss.peer_id:
+                        raise ValueError("resource peer_id mismatch")
                     if res.age() > self.metadata_max_age:
                         raise ValueError("metadata too stale")
                     return res
                 finally:
-                    writer.close()
+                    ss.close()
             except Exception as e:  # noqa: BLE001
                 last_err = e
         raise last_err or ConnectionError("no addresses")
@@ -172,8 +179,8 @@ class Discovery:
         out = []
         for p in await self.find_providers(ns, limit):
             try:
-                res = await self.request_metadata(p.get("addrs", []))
-                res.peer_id = p["peer_id"]
+                res = await self.request_metadata(
+                    p.get("addrs", []), expected_peer_id=p["peer_id"])
                 if not res.addrs:
                     res.addrs = p.get("addrs", [])
                 out.append(res)

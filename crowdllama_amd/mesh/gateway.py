@@ -21,7 +21,7 @@ from ..logutil import new_app_logger
 from . import pb
 from .discovery import parse_addr
 from .peer import Peer
-from .wire import PROTO_INFERENCE, open_protocol, read_frame, write_frame
+from .wire import PROTO_INFERENCE, secure_open
 
 
 class Gateway:
@@ -142,15 +142,17 @@ class Gateway:
         for addr in addrs:
             host, port = parse_addr(addr)
             try:
-                reader, writer = await open_protocol(host, port,
-                                                     PROTO_INFERENCE)
+                # authenticated dial: fails unless the responder PROVES it
+                # owns worker.peer_id (wire.py secure_open)
+                ss = await secure_open(host, port, PROTO_INFERENCE,
+                                       self.peer.identity,
+                                       expected_peer_id=worker.peer_id)
                 try:
-                    await write_frame(writer,
-                                      pb.request_message(model, prompt,
-                                                         stream).encode())
-                    frame = await read_frame(reader, timeout=timeout)
+                    await ss.write_frame(
+                        pb.request_message(model, prompt, stream).encode())
+                    frame = await ss.read_frame(timeout=timeout)
                 finally:
-                    writer.close()
+                    ss.close()
                 resp = pb.BaseMessage.decode(frame).generate_response
                 if resp is None:
                     raise ValueError("no GenerateResponse in reply")
@@ -171,17 +173,17 @@ class Gateway:
         for addr in addrs:
             host, port = parse_addr(addr)
             try:
-                reader, writer = await open_protocol(host, port,
-                                                     PROTO_INFERENCE)
+                ss = await secure_open(host, port, PROTO_INFERENCE,
+                                       self.peer.identity,
+                                       expected_peer_id=worker.peer_id)
             except Exception as e:  # noqa: BLE001
                 last = e
                 continue
             try:
-                await write_frame(writer,
-                                  pb.request_message(model, prompt,
-                                                     stream=True).encode())
+                await ss.write_frame(
+                    pb.request_message(model, prompt, stream=True).encode())
                 while True:
-                    frame = await read_frame(reader, timeout=timeout)
+                    frame = await ss.read_frame(timeout=timeout)
                     resp = pb.BaseMessage.decode(frame).generate_response
                     if resp is None:
                         raise ValueError("no GenerateResponse in reply")
@@ -189,7 +191,7 @@ class Gateway:
                     if resp.done:
                         return
             finally:
-                writer.close()
+                ss.close()
         raise last or ConnectionError("worker unreachable")
 
     async def _stream_ndjson(self, request: web.Request, worker, model: str,

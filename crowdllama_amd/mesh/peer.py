@@ -16,15 +16,14 @@ import time
 
 from ..config import Config
 from ..engine.api import EngineBase, RollingRate
-from ..keys import load_peer_id
+from ..keys import load_identity
 from ..logutil import new_app_logger
 from ..version import __version__, commit_hash
 from . import pb
 from .discovery import Discovery
 from .peermanager import PeerManager
 from .resource import Resource
-from .wire import (NAMESPACE, PROTO_INFERENCE, PROTO_METADATA,
-                   accept_protocol, read_frame, write_frame)
+from .wire import NAMESPACE, PROTO_INFERENCE, PROTO_METADATA, secure_accept
 
 
 class Peer:
@@ -33,11 +32,12 @@ class Peer:
         self.cfg = cfg
         self.worker_mode = worker_mode
         component = "worker" if worker_mode else "consumer"
-        self.peer_id, _ = load_peer_id(component, cfg.key_path)
+        self.identity = load_identity(component, cfg.key_path)
+        self.peer_id = self.identity.peer_id
         self.log = new_app_logger(f"peer.{component}", cfg.verbose)
         self.engines: dict[str, EngineBase] = engines or {}
         self.discovery = Discovery(
-            cfg.bootstrap_peers,
+            cfg.bootstrap_peers, self.identity,
             metadata_timeout=cfg.intervals.metadata_timeout,
             metadata_max_age=cfg.intervals.metadata_max_age,
             log=self.log)
@@ -113,6 +113,9 @@ class Peer:
             r.load = 0.0
         r.addrs = self.addrs
         r.touch()
+        # signed record: receivers verify the signature and that peer_id
+        # is the hash of this public key (mesh/resource.py verify())
+        r.sign(self.identity.seed, self.identity.pub)
 
     async def _metadata_update_loop(self) -> None:
         while True:
@@ -140,15 +143,16 @@ class Peer:
     async def _on_conn(self, reader: asyncio.StreamReader,
                        writer: asyncio.StreamWriter) -> None:
         try:
-            proto = await accept_protocol(reader)
+            # every mesh stream is encrypted + mutually authenticated
+            # (wire.py secure_accept; the reference gets this from libp2p
+            # noise/TLS — discovery.go:48-84)
+            ss, proto = await secure_accept(reader, writer, self.identity)
             if proto == PROTO_METADATA:
                 self.update_metadata()
-                writer.write(self.resource.to_json().encode("utf-8"))
-                await writer.drain()
-                writer.write_eof()
+                await ss.write_frame(self.resource.to_json().encode("utf-8"))
                 return
             if proto == PROTO_INFERENCE:
-                await self._handle_inference(reader, writer)
+                await self._handle_inference(ss)
                 return
             self.log.warning("unknown protocol %r", proto)
         except Exception as e:  # noqa: BLE001
@@ -159,20 +163,20 @@ class Peer:
             except Exception:
                 pass
 
-    async def _handle_inference(self, reader, writer) -> None:
+    async def _handle_inference(self, ss) -> None:
         # read deadline parity: 5 s (peer.go:259-271)
-        frame = await read_frame(reader, timeout=5.0)
+        frame = await ss.read_frame(timeout=5.0)
         msg = pb.BaseMessage.decode(frame)
         if not self.worker_mode:
             resp = pb.response_message("", "Error: peer is not a worker",
                                        self.peer_id, done_reason="error")
-            await write_frame(writer, resp.encode())
+            await ss.write_frame(resp.encode())
             return
         req = msg.generate_request
         if req is None:
             resp = pb.response_message("", "Error: no request in message",
                                        self.peer_id, done_reason="error")
-            await write_frame(writer, resp.encode())
+            await ss.write_frame(resp.encode())
             return
         t0 = time.monotonic_ns()
         self._active_requests += 1
@@ -194,7 +198,7 @@ class Peer:
                         done_reason=chunk.done_reason, done=final,
                         total_duration_ns=(time.monotonic_ns() - t0
                                            if final else 0))
-                    await write_frame(writer, resp.encode())
+                    await ss.write_frame(resp.encode())
                 self.rate.add(total_tokens)
                 self.requests_served += 1
                 return
@@ -211,7 +215,7 @@ class Peer:
                                        total_duration_ns=time.monotonic_ns() - t0)
         finally:
             self._active_requests -= 1
-        await write_frame(writer, resp.encode())
+        await ss.write_frame(resp.encode())
 
     def is_dht_connected(self) -> bool:
         """True while rendezvous round trips are succeeding (reference

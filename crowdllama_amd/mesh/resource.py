@@ -11,6 +11,8 @@ import json
 import time
 from dataclasses import dataclass, field
 
+from . import crypto
+
 
 @dataclass
 class Resource:
@@ -24,12 +26,20 @@ class Resource:
     version: str = ""
     worker_mode: bool = False
     addrs: list[str] = field(default_factory=list)  # "host:port"
+    # record authentication: pubkey is the advertising
+    # peer's ed25519 public key (hex); sig signs the canonical payload.
+    # Receivers verify sig AND that peer_id == hash(pubkey), so a record
+    # cannot be forged for another peer's id. (The reference relies on
+    # libp2p stream authentication instead; records here also travel via
+    # rendezvous, so they carry their own proof.)
+    pubkey: str = ""
+    sig: str = ""
 
     def touch(self) -> None:
         self.last_updated = time.time()
 
-    def to_json(self) -> str:
-        return json.dumps({
+    def _payload(self) -> dict:
+        return {
             "peer_id": self.peer_id,
             "supported_models": self.supported_models,
             "tokens_throughput": self.tokens_throughput,
@@ -40,7 +50,32 @@ class Resource:
             "version": self.version,
             "worker_mode": self.worker_mode,
             "addrs": self.addrs,
-        })
+        }
+
+    def _signing_bytes(self) -> bytes:
+        return json.dumps(self._payload(), sort_keys=True,
+                          separators=(",", ":")).encode("utf-8")
+
+    def sign(self, seed: bytes, pub: bytes) -> None:
+        self.pubkey = pub.hex()
+        self.sig = crypto.ed25519_sign(seed, self._signing_bytes()).hex()
+
+    def verify(self) -> bool:
+        """True iff the record is signed by the key that owns peer_id."""
+        try:
+            pub = bytes.fromhex(self.pubkey)
+            sig = bytes.fromhex(self.sig)
+        except ValueError:
+            return False
+        if crypto.peer_id_from_pub(pub) != self.peer_id:
+            return False
+        return crypto.ed25519_verify(pub, self._signing_bytes(), sig)
+
+    def to_json(self) -> str:
+        d = self._payload()
+        d["pubkey"] = self.pubkey
+        d["sig"] = self.sig
+        return json.dumps(d)
 
     @classmethod
     def from_json(cls, data: str | bytes) -> "Resource":
@@ -58,6 +93,8 @@ class Resource:
         r.version = d.get("version", "")
         r.worker_mode = bool(d.get("worker_mode", False))
         r.addrs = list(d.get("addrs", []))
+        r.pubkey = d.get("pubkey", "")
+        r.sig = d.get("sig", "")
         return r
 
     def age(self) -> float:

@@ -42,7 +42,7 @@ def test_cli_keygen(tmp_path):
                          capture_output=True, text=True)
     assert out.returncode == 0
     assert key.exists()
-    assert "peer id CLA" in out.stdout
+    assert "peer id cla" in out.stdout
 
 
 def _start_mesh(tmp_path, attempt):

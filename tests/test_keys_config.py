@@ -26,7 +26,7 @@ def test_stable_peer_id(tmp_path):
     pid1, _ = load_peer_id("worker", path)
     pid2, _ = load_peer_id("worker", path)
     assert pid1 == pid2
-    assert pid1.startswith("CLA")
+    assert pid1.startswith("cla")  # hash of the ed25519 public key
 
 
 def test_distinct_keys_distinct_ids():

@@ -9,7 +9,10 @@ import pytest
 
 from crowdllama_amd.config import Config
 from crowdllama_amd.mesh.dhtnode import DHTServer
+from crowdllama_amd.keys import identity_from_seed
 from crowdllama_amd.mesh.discovery import Discovery, RendezvousClient
+
+_TEST_ID = identity_from_seed(b"t" * 32)  # throwaway mesh-test identity
 from crowdllama_amd.mesh.peermanager import PeerManager
 from crowdllama_amd.mesh.resource import Resource
 
@@ -29,7 +32,7 @@ def test_dht_server_start_stop(cfg):
         srv = DHTServer(cfg, "CLADHT")
         port = await srv.start("127.0.0.1", 0)
         assert port > 0
-        cli = RendezvousClient(f"127.0.0.1:{port}")
+        cli = RendezvousClient(f"127.0.0.1:{port}", _TEST_ID)
         assert await cli.ping()
         await cli.close()
         await srv.stop()
@@ -40,16 +43,19 @@ def test_provide_and_find(cfg):
     async def go():
         srv = DHTServer(cfg, "CLADHT")
         port = await srv.start("127.0.0.1", 0)
-        cli = RendezvousClient(f"127.0.0.1:{port}")
-        ok = await cli.provide("CLAWORKER1", ["127.0.0.1:5001"])
+        cli = RendezvousClient(f"127.0.0.1:{port}", _TEST_ID)
+        me = _TEST_ID.peer_id
+        # a peer may only advertise its own authenticated identity
+        assert not await cli.provide("CLAFORGED", ["127.0.0.1:9"])
+        ok = await cli.provide(me, ["127.0.0.1:5001"])
         assert ok
         provs = await cli.find_providers()
-        assert any(p["peer_id"] == "CLAWORKER1" for p in provs)
-        addrs = await cli.find_peer("CLAWORKER1")
+        assert any(p["peer_id"] == me for p in provs)
+        addrs = await cli.find_peer(me)
         assert addrs == ["127.0.0.1:5001"]
         assert await cli.find_peer("CLANOBODY") is None
         # model namespace
-        await cli.provide("CLAWORKER1", ["127.0.0.1:5001"],
+        await cli.provide(me, ["127.0.0.1:5001"],
                           ns="crowdllama-ns/model/llama3-8b")
         provs = await cli.find_providers(ns="crowdllama-ns/model/llama3-8b")
         assert len(provs) == 1
@@ -62,12 +68,16 @@ def test_provider_limit(cfg):
     async def go():
         srv = DHTServer(cfg, "CLADHT")
         port = await srv.start("127.0.0.1", 0)
-        cli = RendezvousClient(f"127.0.0.1:{port}")
-        for i in range(15):
-            await cli.provide(f"CLAW{i}", [f"127.0.0.1:{5000 + i}"])
+        ids = [identity_from_seed(bytes([i]) * 32) for i in range(15)]
+        clis = [RendezvousClient(f"127.0.0.1:{port}", ident)
+                for ident in ids]
+        for i, cli in enumerate(clis):
+            await cli.provide(ids[i].peer_id, [f"127.0.0.1:{5000 + i}"])
+        cli = clis[0]
         provs = await cli.find_providers(limit=10)
         assert len(provs) == 10  # reference: FindProvidersAsync(cid, 10)
-        await cli.close()
+        for c in clis:
+            await c.close()
         await srv.stop()
     run(go())
 
@@ -83,7 +93,7 @@ def _mk_resource(pid, model="m1", thr=100.0, load=0.0, worker=True):
 def test_find_best_worker_scoring(cfg):
     """Scheduler maximizes throughput/(1+load) (manager.go:338-387)."""
     async def go():
-        disco = Discovery([], log=None)
+        disco = Discovery([], _TEST_ID, log=None)
         pm = PeerManager(disco, cfg.intervals)
         await pm.add_or_update_peer(_mk_resource("A", thr=100, load=0.0))
         await pm.add_or_update_peer(_mk_resource("B", thr=300, load=2.0))
@@ -105,7 +115,7 @@ def test_find_best_worker_scoring(cfg):
 
 def test_tombstones_prevent_readd(cfg):
     async def go():
-        disco = Discovery([], log=None)
+        disco = Discovery([], _TEST_ID, log=None)
         pm = PeerManager(disco, cfg.intervals)
         await pm.add_or_update_peer(_mk_resource("A"))
         await pm.remove_peer("A")
@@ -121,7 +131,7 @@ def test_tombstones_prevent_readd(cfg):
 
 def test_stale_cleanup(cfg):
     async def go():
-        disco = Discovery([], log=None)
+        disco = Discovery([], _TEST_ID, log=None)
         pm = PeerManager(disco, cfg.intervals)
         await pm.add_or_update_peer(_mk_resource("A"))
         pm.peers["A"].last_seen = time.time() - cfg.intervals.stale_timeout - 1
@@ -138,7 +148,7 @@ def test_stale_cleanup(cfg):
 
 def test_self_never_added(cfg):
     async def go():
-        disco = Discovery([], log=None)
+        disco = Discovery([], _TEST_ID, log=None)
         pm = PeerManager(disco, cfg.intervals, self_id="ME")
         await pm.add_or_update_peer(_mk_resource("ME"))
         assert "ME" not in pm.peers
@@ -152,9 +162,9 @@ def test_provider_ttl_expiry(cfg, monkeypatch):
         monkeypatch.setattr(DHTServer, "PROVIDER_TTL", 0.6)
         srv = DHTServer(cfg, "CLADHT")
         port = await srv.start("127.0.0.1", 0)
-        cl = RendezvousClient(f"127.0.0.1:{port}")
+        cl = RendezvousClient(f"127.0.0.1:{port}", _TEST_ID)
         try:
-            await cl.provide("CLAW1", ["127.0.0.1:1111"], "ns")
+            await cl.provide(_TEST_ID.peer_id, ["127.0.0.1:1111"], "ns")
             assert len(await cl.find_providers("ns")) == 1
             await asyncio.sleep(1.3)
             assert await cl.find_providers("ns") == []
@@ -168,7 +178,7 @@ def test_request_failure_marks_unhealthy(cfg):
     """Gateway-reported request failures accumulate into the same counter
     the health checker uses; max_failed_attempts flips is_healthy."""
     async def go():
-        disc = Discovery([])
+        disc = Discovery([], _TEST_ID)
         pm = PeerManager(disc, cfg.intervals, self_id="CLAME")
         r = Resource(peer_id="CLAW1", worker_mode=True,
                      supported_models=["m"], tokens_throughput=100.0)
@@ -188,7 +198,7 @@ def test_health_check_linear_backoff(cfg):
     """Failed health checks push the next check out linearly
     (failed_attempts * backoff_base — reference manager.go:544-548)."""
     async def go():
-        disc = Discovery([])
+        disc = Discovery([], _TEST_ID)
         pm = PeerManager(disc, cfg.intervals, self_id="CLAME")
         r = Resource(peer_id="CLAW1", worker_mode=True,
                      supported_models=["m"], tokens_throughput=10.0,
@@ -218,7 +228,7 @@ def test_rendezvous_rejects_malformed(cfg):
     async def go():
         srv = DHTServer(cfg, "CLADHT")
         port = await srv.start("127.0.0.1", 0)
-        cl = RendezvousClient(f"127.0.0.1:{port}")
+        cl = RendezvousClient(f"127.0.0.1:{port}", _TEST_ID)
         try:
             r = await cl.call({"op": "bogus"})
             assert r.get("ok") is False
