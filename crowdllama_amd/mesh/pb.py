@@ -10,7 +10,8 @@ protobuf, decodable by any proto library given the schema:
     message GenerateResponse { string model=1; Timestamp created_at=2;
                                string response=3; bool done=4;
                                string done_reason=5; string worker_id=6;
-                               int64 total_duration=7; }
+                               int64 total_duration=7;
+                               int64 eval_count=8 /*extension*/; }
     message BaseMessage      { oneof msg { GenerateRequest generate_request=1;
                                            GenerateResponse generate_response=2; } }
 """
@@ -158,6 +159,8 @@ class GenerateResponse:
     done_reason: str = ""
     worker_id: str = ""
     total_duration: int = 0  # nanoseconds, reference parity (api.go:84)
+    eval_count: int = 0      # generated tokens (extension; Ollama exposes
+    #                          the same field name in its chat responses)
 
     def encode(self) -> bytes:
         return (_enc_str(1, self.model)
@@ -166,7 +169,8 @@ class GenerateResponse:
                 + _enc_bool(4, self.done)
                 + _enc_str(5, self.done_reason)
                 + _enc_str(6, self.worker_id)
-                + _enc_int(7, self.total_duration))
+                + _enc_int(7, self.total_duration)
+                + _enc_int(8, self.eval_count))
 
     @classmethod
     def decode(cls, buf: bytes) -> "GenerateResponse":
@@ -186,6 +190,8 @@ class GenerateResponse:
                 m.worker_id = v.decode("utf-8")
             elif fn == 7 and wire == 0:
                 m.total_duration = _signed(v)
+            elif fn == 8 and wire == 0:
+                m.eval_count = _signed(v)
         return m
 
 
@@ -222,10 +228,10 @@ def request_message(model: str, prompt: str, stream: bool = False) -> BaseMessag
 def response_message(model: str, response: str, worker_id: str = "",
                      done_reason: str = "stop",
                      total_duration_ns: int = 0,
-                     done: bool = True) -> BaseMessage:
+                     done: bool = True, eval_count: int = 0) -> BaseMessage:
     # done=False frames are streamed chunks (capability extension over the
     # reference, which carries `stream` but never streams — SURVEY.md §2.2)
     return BaseMessage(generate_response=GenerateResponse(
         model=model, created_at=Timestamp.now(), response=response, done=done,
         done_reason=done_reason if done else "", worker_id=worker_id,
-        total_duration=total_duration_ns))
+        total_duration=total_duration_ns, eval_count=eval_count))

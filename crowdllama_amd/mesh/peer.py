@@ -152,8 +152,13 @@ class Peer:
                 await ss.write_frame(self.resource.to_json().encode("utf-8"))
                 return
             if proto == PROTO_INFERENCE:
+                # serve multiple sequential requests per connection
+                # (keep-alive extension: the gateway pools authenticated
+                # streams so the handshake amortizes across requests; the
+                # reference opens one libp2p stream per request)
                 await self._handle_inference(ss)
-                return
+                while True:
+                    await self._handle_inference(ss, idle_timeout=120.0)
             self.log.warning("unknown protocol %r", proto)
         except Exception as e:  # noqa: BLE001
             self.log.debug("conn error: %s", e)
@@ -163,9 +168,10 @@ class Peer:
             except Exception:
                 pass
 
-    async def _handle_inference(self, ss) -> None:
-        # read deadline parity: 5 s (peer.go:259-271)
-        frame = await ss.read_frame(timeout=5.0)
+    async def _handle_inference(self, ss, idle_timeout: float = 5.0) -> None:
+        # first-request read deadline parity: 5 s (peer.go:259-271);
+        # keep-alive waits longer between pooled requests
+        frame = await ss.read_frame(timeout=idle_timeout)
         msg = pb.BaseMessage.decode(frame)
         if not self.worker_mode:
             resp = pb.response_message("", "Error: peer is not a worker",
@@ -197,7 +203,8 @@ class Peer:
                         req.model, chunk.text, self.peer_id,
                         done_reason=chunk.done_reason, done=final,
                         total_duration_ns=(time.monotonic_ns() - t0
-                                           if final else 0))
+                                           if final else 0),
+                        eval_count=chunk.tokens_generated)
                     await ss.write_frame(resp.encode())
                 self.rate.add(total_tokens)
                 self.requests_served += 1
@@ -208,7 +215,8 @@ class Peer:
             resp = pb.response_message(
                 req.model, result.text, self.peer_id,
                 done_reason=result.done_reason,
-                total_duration_ns=time.monotonic_ns() - t0)
+                total_duration_ns=time.monotonic_ns() - t0,
+                eval_count=result.tokens_generated)
         except Exception as e:  # noqa: BLE001 — reference stringifies errors
             resp = pb.response_message(req.model, f"Error: {e}", self.peer_id,
                                        done_reason="error",
