@@ -48,6 +48,7 @@ class DHTServer:
         self.total_conns = 0
         self._server: asyncio.base_events.Server | None = None
         self._tasks: list[asyncio.Task] = []
+        self._conn_writers: set = set()
         self.port: int | None = None
 
     async def start(self, host: str | None = None, port: int | None = None):
@@ -74,6 +75,15 @@ class DHTServer:
             self._server.close()
             await self._server.wait_closed()
             self._server = None
+        # asyncio's Server.close() only stops ACCEPTING; live connection
+        # handlers would keep answering old clients from this (stopped)
+        # node's state — close them so clients fail over/reconnect
+        for w in list(self._conn_writers):
+            try:
+                w.close()
+            except Exception:
+                pass
+        self._conn_writers.clear()
 
     # ------------------------------------------------------------ serving
 
@@ -81,6 +91,7 @@ class DHTServer:
                        writer: asyncio.StreamWriter):
         self.conn_count += 1
         self.total_conns += 1
+        self._conn_writers.add(writer)
         try:
             # authenticated, encrypted channel: the caller's peer_id is
             # cryptographically verified before any op is served
@@ -100,6 +111,7 @@ class DHTServer:
             self.log.debug("conn error: %s", e)
         finally:
             self.conn_count -= 1
+            self._conn_writers.discard(writer)
             writer.close()
 
     def _handle(self, msg: dict, caller_id: str = "") -> dict:

@@ -91,8 +91,14 @@ class Discovery:
                  metadata_max_age: float = 3600.0,
                  log: logging.Logger | None = None):
         self.identity = identity
-        self.clients = [RendezvousClient(a, identity)
+        self.clients = [RendezvousClient(a, identity,
+                                         timeout=metadata_timeout)
                         for a in bootstrap_addrs]
+        # gossip fallback (de-SPOF): a callable returning live peer addrs
+        # to query when every bootstrap node is down (Peer wires this to
+        # its peermanager; reference parity: every libp2p peer is a DHT
+        # server, pkg/dht/dht.go:106-112)
+        self.fallback_addrs = None
         self.metadata_timeout = metadata_timeout
         self.metadata_max_age = metadata_max_age
         self.log = log or logging.getLogger("discovery")
@@ -123,13 +129,29 @@ class Discovery:
     async def find_providers(self, ns: str = NAMESPACE,
                              limit: int = 10) -> list[dict]:
         seen = {}
+        ok = False
         for c in self.clients:
             try:
                 for p in await c.find_providers(ns, limit):
                     seen[p["peer_id"]] = p
+                ok = True
                 self.last_success = time.monotonic()
             except Exception as e:
                 self.log.debug("find_providers on %s failed: %s", c.addr, e)
+        if not ok and self.fallback_addrs is not None:
+            # bootstrap outage: ask known live peers (gossip, read-only)
+            for addr in self.fallback_addrs()[:5]:
+                try:
+                    cli = RendezvousClient(addr, self.identity,
+                                           timeout=self.metadata_timeout)
+                    try:
+                        for p in await cli.find_providers(ns, limit):
+                            seen[p["peer_id"]] = p
+                    finally:
+                        await cli.close()
+                    self.last_success = time.monotonic()
+                except Exception as e:  # noqa: BLE001
+                    self.log.debug("gossip find on %s failed: %s", addr, e)
         return list(seen.values())[:limit]
 
     async def find_peer_addrs(self, peer_id: str) -> list[str] | None:

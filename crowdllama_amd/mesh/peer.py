@@ -23,7 +23,8 @@ from . import pb
 from .discovery import Discovery
 from .peermanager import PeerManager
 from .resource import Resource
-from .wire import NAMESPACE, PROTO_INFERENCE, PROTO_METADATA, secure_accept
+from .wire import (NAMESPACE, PROTO_INFERENCE, PROTO_METADATA,
+                   PROTO_RENDEZVOUS, secure_accept)
 
 
 class Peer:
@@ -43,12 +44,16 @@ class Peer:
             log=self.log)
         self.peer_manager = PeerManager(self.discovery, cfg.intervals,
                                         log=self.log, self_id=self.peer_id)
+        # gossip fallback: when every bootstrap node is unreachable,
+        # discovery re-resolves providers through known healthy peers
+        self.discovery.fallback_addrs = self._gossip_addrs
         self.rate = RollingRate()
         self._active_requests = 0
         self.resource = Resource(peer_id=self.peer_id,
                                  worker_mode=worker_mode,
                                  version=f"{__version__}+{commit_hash()}")
         self._server: asyncio.base_events.Server | None = None
+        self._conn_writers: set = set()
         self._tasks: list[asyncio.Task] = []
         self.port: int | None = None
         self.requests_served = 0
@@ -83,6 +88,12 @@ class Peer:
         if self._server:
             self._server.close()
             await self._server.wait_closed()
+        for w in list(self._conn_writers):   # see DHTServer.stop
+            try:
+                w.close()
+            except Exception:
+                pass
+        self._conn_writers.clear()
         for e in self.engines.values():
             await e.close()
 
@@ -142,6 +153,7 @@ class Peer:
 
     async def _on_conn(self, reader: asyncio.StreamReader,
                        writer: asyncio.StreamWriter) -> None:
+        self._conn_writers.add(writer)
         try:
             # every mesh stream is encrypted + mutually authenticated
             # (wire.py secure_accept; the reference gets this from libp2p
@@ -150,6 +162,14 @@ class Peer:
             if proto == PROTO_METADATA:
                 self.update_metadata()
                 await ss.write_frame(self.resource.to_json().encode("utf-8"))
+                return
+            if proto == PROTO_RENDEZVOUS:
+                # read-only rendezvous served from this peer's OWN registry
+                # (gossip fallback): the mesh survives the bootstrap node
+                # dying — consumers re-resolve providers from any live peer
+                # (reference parity: every libp2p peer runs the DHT in
+                # ModeServer, pkg/dht/dht.go:106-112 / discovery.go:92-141)
+                await self._handle_rendezvous(ss)
                 return
             if proto == PROTO_INFERENCE:
                 # serve multiple sequential requests per connection
@@ -163,10 +183,62 @@ class Peer:
         except Exception as e:  # noqa: BLE001
             self.log.debug("conn error: %s", e)
         finally:
+            self._conn_writers.discard(writer)
             try:
                 writer.close()
             except Exception:
                 pass
+
+    async def _handle_rendezvous(self, ss) -> None:
+        import asyncio as _aio
+        while True:
+            try:
+                msg = await ss.read_json(timeout=60.0)
+            except (_aio.IncompleteReadError, _aio.TimeoutError,
+                    ConnectionError, ValueError):
+                return
+            op = msg.get("op")
+            if op == "ping":
+                await ss.write_json({"ok": True, "peer_id": self.peer_id})
+            elif op == "find_providers":
+                ns = msg.get("ns", "")
+                limit = int(msg.get("limit", 10))
+                out = []
+                # this peer itself is a provider of its namespaces
+                if self.worker_mode and (
+                        ns == NAMESPACE or any(
+                            ns == f"{NAMESPACE}/model/{m}"
+                            for m in self.engines)):
+                    out.append({"peer_id": self.peer_id,
+                                "addrs": self.addrs})
+                for pid, pi in self.peer_manager.peers.items():
+                    if len(out) >= limit:
+                        break
+                    r = pi.resource
+                    if not pi.is_healthy or not r.worker_mode:
+                        continue
+                    if ns != NAMESPACE and not any(
+                            ns == f"{NAMESPACE}/model/{m}"
+                            for m in r.supported_models):
+                        continue
+                    out.append({"peer_id": pid, "addrs": r.addrs})
+                await ss.write_json({"ok": True, "providers": out[:limit]})
+            elif op == "find_peer":
+                pid = msg.get("peer_id", "")
+                if pid == self.peer_id:
+                    await ss.write_json({"ok": True, "addrs": self.addrs})
+                    continue
+                pi = self.peer_manager.peers.get(pid)
+                if pi is None:
+                    await ss.write_json({"ok": False,
+                                         "error": "peer not found"})
+                else:
+                    await ss.write_json({"ok": True,
+                                         "addrs": pi.resource.addrs})
+            else:
+                # provide/remove are bootstrap-node ops; gossip is read-only
+                await ss.write_json({"ok": False,
+                                     "error": f"unsupported op {op!r}"})
 
     async def _handle_inference(self, ss, idle_timeout: float = 5.0) -> None:
         # first-request read deadline parity: 5 s (peer.go:259-271);
@@ -224,6 +296,13 @@ class Peer:
         finally:
             self._active_requests -= 1
         await ss.write_frame(resp.encode())
+
+    def _gossip_addrs(self) -> list[str]:
+        out = []
+        for pi in self.peer_manager.peers.values():
+            if pi.is_healthy and pi.resource.addrs:
+                out.extend(pi.resource.addrs)
+        return out
 
     def is_dht_connected(self) -> bool:
         """True while rendezvous round trips are succeeding (reference

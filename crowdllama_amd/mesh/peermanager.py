@@ -167,10 +167,11 @@ class PeerManager:
                 if now < pi.next_health_check:
                     continue  # linear backoff window
                 try:
+                    # authenticated probe: the metadata fetch fails unless
+                    # the responder proves it IS pid (wire.py secure_open)
                     res = await self.discovery.request_metadata(
-                        pi.resource.addrs)
+                        pi.resource.addrs, expected_peer_id=pid)
                     pi.resource = res if res.peer_id else pi.resource
-                    pi.resource.peer_id = pid
                     pi.last_seen = now
                     pi.failed_attempts = 0
                     pi.is_healthy = True

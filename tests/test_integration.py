@@ -426,3 +426,71 @@ def test_concurrent_burst_spreads_across_workers(mesh_cfg):
                 await w.stop()
             await dht.stop()
     asyncio.run(go())
+
+
+def test_mesh_survives_bootstrap_death(mesh_cfg):
+    """De-SPOF: kill the rendezvous node mid-run; the
+    gateway keeps serving (health probes are direct and discovery falls
+    back to gossip through known peers); restart a bootstrap on the same
+    port and advertisement resumes. Reference bar: every libp2p peer runs
+    the DHT in ModeServer (pkg/dht/dht.go:106-112), so losing one
+    bootstrap peer never partitions the reference mesh either."""
+    async def go():
+        dht = DHTServer(mesh_cfg("dht"), "CLADHT")
+        dht_port = await dht.start("127.0.0.1", 0)
+        boot = [f"127.0.0.1:{dht_port}"]
+        wcfg = mesh_cfg("worker")
+        wcfg.bootstrap_peers = boot
+        worker = Peer(wcfg, worker_mode=True,
+                      engines={"m": MockEngine("m")})
+        await worker.start()
+        ccfg = mesh_cfg("consumer")
+        ccfg.bootstrap_peers = boot
+        consumer = Peer(ccfg, worker_mode=False)
+        await consumer.start()
+        gw = Gateway(consumer, ccfg)
+        gw_port = await gw.start(port=0)
+        try:
+            await _poll(lambda: gw.find_best_worker("m") is not None,
+                        desc="worker discovery")
+            await dht.stop()  # bootstrap node dies
+
+            # serving continues: direct health probes keep the worker
+            # entry alive and /api/chat still routes
+            for _ in range(3):
+                status, body = await _http_json(
+                    "POST", f"http://127.0.0.1:{gw_port}/api/chat",
+                    {"model": "m",
+                     "messages": [{"role": "user", "content": "x"}]})
+                assert status == 200, body
+                await asyncio.sleep(0.5)
+
+            # gossip fallback: a fresh discovery round (bootstrap dead)
+            # still finds the worker via the consumer's known peers
+            provs = await consumer.discovery.find_providers()
+            assert any(p["peer_id"] == worker.peer_id for p in provs), provs
+
+            # bootstrap restart on the same port: advertising resumes
+            dht2 = DHTServer(mesh_cfg("dht"), "CLADHT")
+            await dht2.start("127.0.0.1", dht_port)
+            try:
+                deadline = time.time() + 60
+                readvertised = False
+                while time.time() < deadline and not readvertised:
+                    st = dht2.stats()
+                    readvertised = st["providers"] > 0
+                    if not readvertised:
+                        await asyncio.sleep(0.3)
+                assert readvertised, "worker never re-advertised"
+                status, _ = await _http_json(
+                    "POST", f"http://127.0.0.1:{gw_port}/api/chat",
+                    {"model": "m",
+                     "messages": [{"role": "user", "content": "y"}]})
+                assert status == 200
+            finally:
+                await dht2.stop()
+        finally:
+            await gw.stop()
+            await consumer.stop()
+            await worker.stop()
+    asyncio.run(go())
