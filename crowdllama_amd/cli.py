@@ -168,7 +168,11 @@ async def _run_network_status(args) -> None:
                 if st:
                     print(f"dht {c.addr}: peers={st['known_peers']} "
                           f"providers={st['providers']} "
-                          f"conns={st['active_conns']}/{st['total_conns']}")
+                          f"conns={st['active_conns']}/{st['total_conns']} "
+                          f"nat={st.get('nat', {})}")
+                obs = await c.observed_addr()
+                if obs:
+                    print(f"observed address (via {c.addr}): {obs}")
             except Exception:  # noqa: BLE001
                 pass
         peers = await disco.discover_peers()

@@ -44,6 +44,12 @@ class DHTServer:
         self.log = new_app_logger("dht", cfg.verbose)
         self.providers: dict[str, dict[str, ProviderRecord]] = {}  # ns -> id -> rec
         self.peer_addrs: dict[str, list[str]] = {}
+        # reachability classification per peer (reference NAT stats,
+        # pkg/dht/dht.go:279-309,346-395): the server compares each
+        # provider's OBSERVED source address with its advertised addrs.
+        # "direct" = advertised host matches what we see; "translated" =
+        # mismatch (NAT/proxy in the path); "loopback" = local testing.
+        self.peer_reachability: dict[str, str] = {}
         self.conn_count = 0
         self.total_conns = 0
         self._server: asyncio.base_events.Server | None = None
@@ -99,13 +105,15 @@ class DHTServer:
             if proto != PROTO_RENDEZVOUS:
                 self.log.warning("unknown protocol %r", proto)
                 return
+            peername = writer.get_extra_info("peername") or ("?", 0)
+            observed = f"{peername[0]}:{peername[1]}"
             while True:
                 try:
                     msg = await ss.read_json(timeout=60.0)
                 except (asyncio.IncompleteReadError, asyncio.TimeoutError,
                         ConnectionError, ValueError):
                     return
-                resp = self._handle(msg, ss.peer_id)
+                resp = self._handle(msg, ss.peer_id, observed)
                 await ss.write_json(resp)
         except Exception as e:  # noqa: BLE001 — per-conn isolation
             self.log.debug("conn error: %s", e)
@@ -114,10 +122,14 @@ class DHTServer:
             self._conn_writers.discard(writer)
             writer.close()
 
-    def _handle(self, msg: dict, caller_id: str = "") -> dict:
+    def _handle(self, msg: dict, caller_id: str = "",
+                observed: str = "") -> dict:
         op = msg.get("op")
         if op == "ping":
-            return {"ok": True, "peer_id": self.peer_id}
+            # observed_addr lets the caller classify its own reachability
+            # (the analog of libp2p's identify/observed-address)
+            return {"ok": True, "peer_id": self.peer_id,
+                    "observed_addr": observed}
         if op == "provide":
             ns = msg.get("ns", "")
             pid = msg.get("peer_id", "")
@@ -132,6 +144,7 @@ class DHTServer:
             self.providers.setdefault(ns, {})[pid] = ProviderRecord(
                 pid, addrs)
             self.peer_addrs[pid] = addrs
+            self.peer_reachability[pid] = self._classify(observed, addrs)
             return {"ok": True}
         if op == "find_providers":
             ns = msg.get("ns", "")
@@ -163,6 +176,20 @@ class DHTServer:
             return {"ok": True, "stats": self.stats()}
         return {"ok": False, "error": f"unknown op {op!r}"}
 
+    @staticmethod
+    def _classify(observed: str, advertised: list[str]) -> str:
+        host = observed.rsplit(":", 1)[0]
+        if host in ("127.0.0.1", "::1", "localhost"):
+            return "loopback"
+        adv_hosts = {a.rsplit(":", 1)[0] for a in advertised}
+        return "direct" if host in adv_hosts else "translated"
+
+    def nat_stats(self) -> dict:
+        out = {"direct": 0, "translated": 0, "loopback": 0}
+        for v in self.peer_reachability.values():
+            out[v] = out.get(v, 0) + 1
+        return out
+
     def stats(self) -> dict:
         nprov = sum(len(v) for v in self.providers.values())
         return {
@@ -172,6 +199,7 @@ class DHTServer:
             "namespaces": len(self.providers),
             "active_conns": self.conn_count,
             "total_conns": self.total_conns,
+            "nat": self.nat_stats(),
         }
 
     # ----------------------------------------------------------- bg loops
@@ -181,9 +209,9 @@ class DHTServer:
         while True:
             await asyncio.sleep(self.cfg.intervals.nat_log)
             s = self.stats()
-            self.log.info("stats: peers=%d providers=%d conns=%d/%d",
-                          s["known_peers"], s["providers"],
-                          s["active_conns"], s["total_conns"])
+            self.log.info("stats: peers=%d providers=%d conns=%d/%d "
+                          "nat=%s", s["known_peers"], s["providers"],
+                          s["active_conns"], s["total_conns"], s["nat"])
 
     async def _expiry_loop(self):
         while True:

@@ -62,6 +62,15 @@ class RendezvousClient:
         except Exception:
             return False
 
+    async def observed_addr(self) -> str | None:
+        """This client's address as seen by the bootstrap node (the
+        analog of libp2p identify's observed address)."""
+        try:
+            r = await self.call({"op": "ping"})
+            return r.get("observed_addr") if r.get("ok") else None
+        except Exception:
+            return None
+
     async def provide(self, peer_id: str, addrs: list[str],
                       ns: str = NAMESPACE) -> bool:
         r = await self.call({"op": "provide", "ns": ns, "peer_id": peer_id,

@@ -304,6 +304,25 @@ class Peer:
                 out.extend(pi.resource.addrs)
         return out
 
+    async def reachability(self) -> str:
+        """Classify this peer's reachability from a bootstrap node's view
+        (reference NAT classification, pkg/dht/dht.go:346-395): "direct"
+        when the observed address matches an advertised one, "translated"
+        behind a NAT/proxy, "loopback" for local meshes, "unknown" when no
+        bootstrap answers. Hole punching / UPnP port mapping are a
+        documented descope (docs/PARITY.md): deploy workers with a
+        reachable address or a TCP reverse proxy."""
+        for c in self.discovery.clients:
+            obs = await c.observed_addr()
+            if not obs:
+                continue
+            host = obs.rsplit(":", 1)[0]
+            if host in ("127.0.0.1", "::1", "localhost"):
+                return "loopback"
+            adv = {a.rsplit(":", 1)[0] for a in self.addrs}
+            return "direct" if host in adv else "translated"
+        return "unknown"
+
     def is_dht_connected(self) -> bool:
         """True while rendezvous round trips are succeeding (reference
         IsDHTConnected checks the routing table, peer.go:513-525; here the

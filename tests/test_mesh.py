@@ -239,3 +239,29 @@ def test_rendezvous_rejects_malformed(cfg):
             await cl.close()
             await srv.stop()
     run(go())
+
+
+def test_nat_reachability_classification(cfg):
+    """Reachability classification (reference NAT stats parity,
+    pkg/dht/dht.go:279-309): the rendezvous node classifies providers by
+    comparing their observed source address with their advertised addrs,
+    and peers can self-classify via the observed_addr ping echo."""
+    async def go():
+        srv = DHTServer(cfg, "CLADHT")
+        port = await srv.start("127.0.0.1", 0)
+        cli = RendezvousClient(f"127.0.0.1:{port}", _TEST_ID)
+        try:
+            obs = await cli.observed_addr()
+            assert obs and obs.startswith("127.0.0.1:")
+            await cli.provide(_TEST_ID.peer_id, ["127.0.0.1:5001"])
+            st = (await cli.stats())
+            assert st["nat"]["loopback"] == 1  # loopback test mesh
+            # server-side classifier unit cases
+            assert DHTServer._classify("10.0.0.5:44", ["10.0.0.5:9"]) == \
+                "direct"
+            assert DHTServer._classify("198.51.100.7:44",
+                                       ["10.0.0.5:9"]) == "translated"
+        finally:
+            await cli.close()
+            await srv.stop()
+    run(go())
