@@ -44,6 +44,11 @@ def main() -> None:
                          "2's metric)")
     ap.add_argument("--concurrency", type=int, default=32,
                     help="serve mode: concurrent in-flight HTTP requests")
+    ap.add_argument("--models", default="",
+                    help="serve mode: comma-separated models for a mixed "
+                         "fleet (one worker per model on this GPU; "
+                         "requests round-robin across models — BASELINE "
+                         "config 5's model-aware routing)")
     ap.add_argument("--max-new", type=int, default=64,
                     help="serve mode: max_new_tokens per request")
     ap.add_argument("--batch", type=int, default=16,
@@ -209,9 +214,12 @@ def serve_bench(args) -> None:
 
     from crowdllama_amd.models import get_preset, synth_path
 
-    cfg_model = get_preset(args.model)
-    path = synth_path(args.model, scheme=args.scheme, mode="fast")
-    log(f"checkpoint ready: {path}")
+    models = [m for m in args.models.split(",") if m] or [args.model]
+    paths = {}
+    for m in models:
+        get_preset(m)  # validate
+        paths[m] = synth_path(m, scheme=args.scheme, mode="fast")
+        log(f"checkpoint ready: {paths[m]}")
 
     from crowdllama_amd.ops import get_core
     if get_core().device_count() == 0:
@@ -256,13 +264,13 @@ def serve_bench(args) -> None:
         gw_port = await gw.start(port=0)
         url = f"http://127.0.0.1:{gw_port}/api/chat"
 
-        # wait for discovery
-        deadline = time.time() + 60
-        while gw.find_best_worker(args.model) is None:
+        # wait for discovery of every model
+        deadline = time.time() + 120
+        while any(gw.find_best_worker(m) is None for m in models):
             if time.time() > deadline:
                 raise RuntimeError("worker never discovered")
             await asyncio.sleep(0.2)
-        log("worker discovered; running serve bench")
+        log("workers discovered; running serve bench")
 
         rng = np.random.default_rng(7)
         words = ["alpha", "beta", "gamma", "delta", "omega", "sigma",
@@ -275,10 +283,14 @@ def serve_bench(args) -> None:
         lat: list[float] = []
         toks = [0]
 
+        rr = [0]
+
         async def one_request(session):
+            model = models[rr[0] % len(models)]
+            rr[0] += 1
             t0 = time.perf_counter()
             async with session.post(url, json={
-                    "model": args.model,
+                    "model": model,
                     "messages": [{"role": "user", "content": mk_prompt()}],
                     "options": {"num_predict": args.max_new}}) as r:
                 body = await r.json()
@@ -307,7 +319,8 @@ def serve_bench(args) -> None:
 
         await gw.stop()
         await consumer.stop()
-        await worker.stop()
+        for w in workers:
+            await w.stop()
         await dht.stop()
 
         lat_ms = sorted(x * 1e3 for x in lat)
@@ -327,7 +340,7 @@ def serve_bench(args) -> None:
             "dtype": args.scheme + "+int8-activations",
             "data": "synthetic",
             "config": {
-                "model": args.model,
+                "model": ",".join(models),
                 "quant": args.scheme,
                 "path": "dht+worker+gateway /api/chat (loopback, "
                         "encrypted mesh streams)",
