@@ -246,16 +246,20 @@ def serve_bench(args) -> None:
         dht = DHTServer(mk("dht"))
         dht_port = await dht.start("127.0.0.1", 0)
         boot = [f"127.0.0.1:{dht_port}"]
-        # per-request token budget rides the engine default (the PB
-        # GenerateRequest is reference-schema: model/prompt/stream only)
-        engine = BatchingHipEngine(args.model, path, batch=args.batch,
-                                   max_seq=args.max_seq,
-                                   max_new=args.max_new)
-        wcfg = mk("worker")
-        wcfg.bootstrap_peers = boot
-        worker = Peer(wcfg, worker_mode=True,
-                      engines={args.model: engine})
-        await worker.start()
+        # one worker per model, all on this GPU (288 GB HBM holds a mixed
+        # fleet; BASELINE config 5 runs 2 models with model-aware routing).
+        # Per-request token budget rides the engine default max_new (the
+        # PB GenerateRequest is reference-schema: model/prompt/stream).
+        workers = []
+        for i, m in enumerate(models):
+            engine = BatchingHipEngine(m, paths[m], batch=args.batch,
+                                       max_seq=args.max_seq,
+                                       max_new=args.max_new)
+            wcfg = mk(f"worker{i}")
+            wcfg.bootstrap_peers = boot
+            w = Peer(wcfg, worker_mode=True, engines={m: engine})
+            await w.start()
+            workers.append(w)
         ccfg = mk("consumer")
         ccfg.bootstrap_peers = boot
         consumer = Peer(ccfg, worker_mode=False)
