@@ -834,8 +834,12 @@ void Engine::ensure_graph() {
     // fall back to eager stepping instead of dying.
     hipGraph_t graph = nullptr;
     try {
+        // Relaxed: other engines in the process (mixed-fleet serving
+        // runs one engine per model) may touch the null stream while this
+        // engine captures; ThreadLocal mode made their sync memcpys fail
+        // with "legacy stream would depend on a capturing stream".
         HIP_CHECK(hipStreamBeginCapture(stream_,
-                                        hipStreamCaptureModeThreadLocal));
+                                        hipStreamCaptureModeRelaxed));
         step(stream_);
         HIP_CHECK(hipStreamEndCapture(stream_, &graph));
         HIP_CHECK(hipGraphInstantiate(&graph_exec_, graph, nullptr,
