@@ -626,12 +626,13 @@ __global__ __launch_bounds__(256) void k_attn_prefill(
 // split-K workgroup target: >=2 WGs per CU keeps all 8 XCDs fed when M is
 // small; overridable for on-hardware sweeps (CLA_SPLITK_TARGET).
 static int splitk_target() {
-    // 512 (2 WGs/CU) measured best once the decode tiles were
-    // double-buffered (sweep in docs/PERF_NOTES.md; 768 won pre-pipeline)
+    // 768 (~3 WGs/CU) measured best for the round-2 i8 DMA kernel with
+    // tiled weights (sweep in docs/PERF_NOTES.md; 512 won for the round-1
+    // register-staged tiles)
     static int t = [] {
         const char* e = getenv("CLA_SPLITK_TARGET");
-        int v = e ? atoi(e) : 512;
-        return v > 0 ? v : 512;
+        int v = e ? atoi(e) : 768;
+        return v > 0 ? v : 768;
     }();
     return t;
 }
