@@ -503,10 +503,12 @@ void Engine::alloc_state() {
     const int B = cfg_.batch, H = meta_.hidden, V = meta_.vocab;
     const int NH = meta_.heads_l, NKV = meta_.kv_heads_l, D = meta_.head_dim;
     const int F = meta_.ffn_l;
-    // KV split count: 256/(B*NKV) — S=2 at B=16 measured best with the
-    // round-2 i8 GEMM step (ctx<=384: 4548 vs 4396 tok/s at S=4); B=8
-    // still gets S=4 for long-context scan parallelism
+    // KV split count: 256/(B*NKV) with a floor of 2 up to B=32 — S=2 at
+    // B=16 measured best with the round-2 i8 GEMM step (4548 vs 4396
+    // tok/s at S=4), B=32 regressed at S=1 (5549 vs 6260 at S=2), and
+    // B=64 has enough workgroups without splits
     attn_splits_ = std::max(1, std::min(32, 256 / std::max(1, B * NKV)));
+    if (attn_splits_ < 2 && B <= 32) attn_splits_ = 2;
     if (const char* e = getenv("CLA_ATTN_SPLITS")) {   // on-HW sweeps
         const int v = atoi(e);
         if (v >= 1 && v <= 64) attn_splits_ = v;
