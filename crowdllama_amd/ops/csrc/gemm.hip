@@ -625,16 +625,16 @@ __global__ __launch_bounds__(256) void k_attn_prefill(
 
 // split-K workgroup target: >=2 WGs per CU keeps all 8 XCDs fed when M is
 // small; overridable for on-hardware sweeps (CLA_SPLITK_TARGET).
-static int splitk_target() {
-    // 768 (~3 WGs/CU) measured best for the round-2 i8 DMA kernel with
-    // tiled weights (sweep in docs/PERF_NOTES.md; 512 won for the round-1
-    // register-staged tiles)
-    static int t = [] {
+// BM=16 decode tiles (M<=16) like deeper splits (768 ~ 3 WGs/CU measured
+// +4% at B=16 with the i8 DMA kernel); BM=32 tiles regressed at 768
+// (B=32: 5549 vs 6260 tok/s) and keep 512.
+static int splitk_target(int M) {
+    static int env = [] {
         const char* e = getenv("CLA_SPLITK_TARGET");
-        int v = e ? atoi(e) : 768;
-        return v > 0 ? v : 768;
+        return e ? atoi(e) : 0;
     }();
-    return t;
+    if (env > 0) return env;
+    return M <= 16 ? 768 : 512;
 }
 
 // Single source of truth for the split-K factor: launch_gemm_ex and
@@ -647,7 +647,7 @@ int gemm_splitk_factor(int N, int K, int M) {
     if (!small_m) return 1;
     const int bm_tiles = (M + 31) / 32;
     const int n_tiles = (N + BN - 1) / BN;
-    const int tgt = splitk_target();
+    const int tgt = splitk_target(M);
     const int wgs = n_tiles * bm_tiles;
     int splitk = K / BK < tgt / (wgs ? wgs : 1) ? K / BK : tgt / (wgs ? wgs : 1);
     if (splitk < 1) splitk = 1;
