@@ -91,8 +91,12 @@ __device__ __forceinline__ void glds4(const void* gsrc, unsigned lds_off) {
 
 }  // namespace
 
-// BM_ in {16, 32}. 256 threads = 4 waves; BM=16 -> 1x4 wave grid (each wave
-// all 16 M-rows x 32 cols), BM=32 -> 2x2 (16 rows x 64 cols per wave).
+// BM_ in {16, 32, 128}. 256 threads = 4 waves; BM=16 -> 1x4 wave grid
+// (each wave all 16 M-rows x 32 cols), BM=32 -> 2x2 (16 rows x 64 cols),
+// BM=128 -> 2x2 with FM=4 row fragments per wave (prefill tiles: one pass
+// over the W stream covers 128 activation rows instead of re-streaming W
+// per 32-row tile). BM<=32 keeps per-wave X copies; BM=128 stages ONE
+// shared X image cooperatively.
 // Split-K accumulates into pre-zeroed C via atomicAdd (identical contract
 // to gemm.hip's k_gemm: gemm_uses_splitk tells the caller to pre-zero).
 //
@@ -109,9 +113,12 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     constexpr int WMW = (BM_ == 16) ? 1 : 2;
     constexpr int WNW = 4 / WMW;
     constexpr int JF = BN / WNW / 16;        // b fragments per wave (2 or 4)
+    constexpr int FM = (BM_ / WMW) / 16;     // a fragments per wave (1 or 4)
+    constexpr bool XSHARED = BM_ > 32;       // one X image vs per-wave
     constexpr int RAWB = (W == DT::DQ4K) ? BK / 2 : BK;  // raw bytes/row
     constexpr int NGW = (BN / 4) * RAWB / 1024;  // W glds per wave (1 or 2)
-    constexpr int NGX = BM_ * BK / 1024;         // X glds per wave (1 or 2)
+    constexpr int NGX = XSHARED ? BM_ * BK / 4096
+                                : BM_ * BK / 1024;   // X glds per wave
     constexpr int NGH = (W == DT::DQ4K) ? JF : JF / 2;  // hdr2 glds/wave
     constexpr int NGS = 1;                       // xsc glds per wave
     constexpr int NGLT = NGW + NGX + NGH + NGS;  // per wave per tile
@@ -124,9 +131,10 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     //   [.., +4*256)                  per-wave xsc copies [kg2][2][16] f32
     constexpr int WSZ = BN * RAWB;
     constexpr int XSZ = BM_ * BK;
+    constexpr int XTOT = XSHARED ? XSZ : 4 * XSZ;
     constexpr int HSZ = (W == DT::DQ4K) ? JF * 256 : JF * 128;
-    constexpr int SSZ = 256;
-    constexpr int BUFSZ = WSZ + 4 * (XSZ + HSZ + SSZ);
+    constexpr int SSZ = FM * 256;
+    constexpr int BUFSZ = WSZ + XTOT + 4 * (HSZ + SSZ);
     // 4-deep DMA ring: the single-barrier pipeline below keeps NBUF-2
     // tiles fully in flight, so each tile's DMA has ~2 iterations of
     // compute to hide under (PMC: 1-deep exposed ~38% parked)
@@ -146,11 +154,13 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     const int lrow = lane & 15, lk = lane >> 4;
     const int M4 = (M + 3) & ~3;
 
-    float facc[JF][4];
+    float facc[FM][JF][4];
     #pragma unroll
-    for (int j = 0; j < JF; j++)
+    for (int i = 0; i < FM; i++)
         #pragma unroll
-        for (int r = 0; r < 4; r++) facc[j][r] = 0.f;
+        for (int j = 0; j < JF; j++)
+            #pragma unroll
+            for (int r = 0; r < 4; r++) facc[i][j][r] = 0.f;
 
     // ---- W source: the GEMM-tiled qs2 copy (common.h) — each (n-block,
     // k-window) tile is contiguous and identical to the LDS image, so a
@@ -159,17 +169,21 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     const int64_t ktiles = K / BK;
     const uint8_t* wtile0 = qs2 + (int64_t)bn * ktiles * (BN * RAWB);
     const int wlocal = wid * (NGW * 1024) + lane * 16;  // == LDS offset
-    // X: each wave DMAs its own copy of the whole [BM_][64] tile.
-    // glds gx covers rows gx*16 + (lane>>2), 16B quarter lane&3.
-    int64_t xrow_off[2];
+    // X staging rows. BM<=32: each wave DMAs its own full copy (glds gx
+    // covers rows gx*16 + lane>>2). BM=128: ONE shared image; wave wid
+    // DMAs rows [wid*32, wid*32+32) (glds gx covers wid*32 + gx*16 + ...).
+    int64_t xrow_off[NGX];
     #pragma unroll
     for (int g = 0; g < NGX; g++) {
-        const int xr = (NGX == 1) ? (lane >> 2) : (g * 16 + (lane >> 2));
+        int xr;
+        if constexpr (XSHARED) {
+            xr = wid * 32 + g * 16 + (lane >> 2);
+        } else {
+            xr = (NGX == 1) ? (lane >> 2) : (g * 16 + (lane >> 2));
+        }
         const int gm = (m0 + xr < M) ? m0 + xr : (M > 0 ? M - 1 : 0);
         xrow_off[g] = (int64_t)gm * ldxq + (lane & 3) * 16;
     }
-    const int xzrow = (NGX == 1) ? (lane >> 2) : 0;  // guard rows >= M later
-    (void)xzrow;
     // hdr2: per-wave copy of the k-window's headers for all BN cols.
     // Q4K: 128 cols x 8B = 1KB, lane takes cols {2l, 2l+1} (16B).
     // Q8: 2 x (128 cols x 2B f16) = 2 x 256B dword-glds, lane 4B.
@@ -179,9 +193,10 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     // integer LDS offsets only — pointer-typed locals into __shared__
     // decay to GENERIC address space and the reads compile to flat_load
     // (caught in this kernel's v3 disassembly: zero ds_read in the loop)
-    const int xo = WSZ + wid * XSZ;
-    const int ho = WSZ + 4 * XSZ + wid * HSZ;
-    const int so = WSZ + 4 * (XSZ + HSZ) + wid * SSZ;
+    const int xo = XSHARED ? WSZ : WSZ + wid * XSZ;      // read base
+    const int xow = XSHARED ? WSZ + wid * (XSZ / 4) : xo;  // DMA dest base
+    const int ho = WSZ + XTOT + wid * HSZ;
+    const int so = WSZ + XTOT + 4 * HSZ + wid * SSZ;
 
     auto issue_tile = [&](int kb, int pb) {
         // W raw (nt: streamed once per step); source offset == LDS offset
@@ -193,12 +208,12 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
                           __builtin_amdgcn_readfirstlane((unsigned)(size_t)
                               &lds[pb][wid * (NGW * 1024) + g * 1024]));
         }
-        // X tile copy for this wave
+        // X tile (copy per wave, or this wave's quarter of the shared one)
         #pragma unroll
         for (int g = 0; g < NGX; g++) {
             glds16(xq + xrow_off[g] + kb,
                    __builtin_amdgcn_readfirstlane((unsigned)(size_t)
-                       &lds[pb][xo + g * 1024]));
+                       &lds[pb][xow + g * 1024]));
         }
         // headers: hdr2 is pre-decoded f32 ({d*sc, dmin*mn} pairs for
         // Q4K, d for Q8) laid out [K/32][N]; each wave stages only its own
@@ -230,10 +245,11 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
                               (unsigned)(size_t)&lds[pb][ho + kb2 * 256]));
             }
         }
-        // activation scales: chunk (kg2 = c>>1, arr = c&1), 16 slots each.
-        // Slot index clamps to M4-1 (partial BM tiles): the duplicated
-        // values land in LDS slots whose output rows are discarded anyway.
-        {
+        // activation scales: per (kg2, arr) chunks of this wave's M rows.
+        // Slot indices clamp into the valid [0, M4) range (partial BM
+        // tiles): the duplicated values land in LDS slots whose output
+        // rows are discarded anyway.
+        if constexpr (FM == 1) {
             const int kg = kb >> 5;
             const int c = lane >> 4;
             int srow = m0 + wm * 16 + (lane & 15);
@@ -243,9 +259,23 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
             glds4(xsc + soff,
                   __builtin_amdgcn_readfirstlane((unsigned)(size_t)
                       &lds[pb][so]));
+        } else {  // BM=128: one glds16 (64 lanes x 16B = 1KB) covers all
+            // 4 chunks (kg2 x {dx,sum}) of this wave's 64 M rows: lane l
+            // -> chunk c = l>>4, slots ((l&15)*4 .. +4)
+            const int kg = kb >> 5;
+            const int c = lane >> 4;
+            int sbase = m0 + wm * (BM_ / WMW) + (lane & 15) * 4;
+            if (sbase > M4 - 4) sbase = M4 - 4;
+            if (sbase < 0) sbase = 0;
+            const int64_t soff =
+                ((int64_t)(kg + (c >> 1)) * 2 + (c & 1)) * M4 + sbase;
+            glds16(xsc + soff,
+                   __builtin_amdgcn_readfirstlane((unsigned)(size_t)
+                       &lds[pb][so]));
         }
     };
 
+    constexpr int SROW = (FM == 1) ? 16 : 64;   // xsc slots per wave
     auto mfma_tile = [&](int pb) {
         // raw W fragment bytes: one b64 per j serves both K=32 halves (Q4K)
         long rawj[JF][(W == DT::DQ4K) ? 1 : 2];
@@ -260,12 +290,19 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         }
         #pragma unroll
         for (int kb2 = 0; kb2 < 2; kb2++) {
-            const long a = *reinterpret_cast<const long*>(
-                &lds[pb][xo + (wm * 16 + lrow) * BK + kb2 * 32 + lk * 8]);
-            const float4 dx4 = *reinterpret_cast<const float4*>(
-                &lds[pb][so + ((kb2 * 2 + 0) * 16 + lk * 4) * 4]);
-            const float4 sm4 = *reinterpret_cast<const float4*>(
-                &lds[pb][so + ((kb2 * 2 + 1) * 16 + lk * 4) * 4]);
+            long a[FM];
+            float4 dx4[FM], sm4[FM];
+            #pragma unroll
+            for (int i = 0; i < FM; i++) {
+                const int xr = wm * (BM_ / WMW) + i * 16 + lrow;
+                a[i] = *reinterpret_cast<const long*>(
+                    &lds[pb][xo + xr * BK + kb2 * 32 + lk * 8]);
+                const int sl = i * 16 + lk * 4;   // wave-local slot base
+                dx4[i] = *reinterpret_cast<const float4*>(
+                    &lds[pb][so + ((kb2 * 2 + 0) * SROW + sl) * 4]);
+                sm4[i] = *reinterpret_cast<const float4*>(
+                    &lds[pb][so + ((kb2 * 2 + 1) * SROW + sl) * 4]);
+            }
             #pragma unroll
             for (int j = 0; j < JF; j++) {
                 const int cl = j * 16 + lrow;   // wave-local column
@@ -285,12 +322,16 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
                         &lds[pb][ho + kb2 * (JF * 64) + cl * 4]);
                     m = 0.f;
                 }
-                v4i c = {0, 0, 0, 0};
-                c = __builtin_amdgcn_mfma_i32_16x16x32_i8(a, b, c, 0, 0, 0);
-                facc[j][0] += d * dx4.x * (float)c[0] - m * sm4.x;
-                facc[j][1] += d * dx4.y * (float)c[1] - m * sm4.y;
-                facc[j][2] += d * dx4.z * (float)c[2] - m * sm4.z;
-                facc[j][3] += d * dx4.w * (float)c[3] - m * sm4.w;
+                #pragma unroll
+                for (int i = 0; i < FM; i++) {
+                    v4i c = {0, 0, 0, 0};
+                    c = __builtin_amdgcn_mfma_i32_16x16x32_i8(a[i], b, c,
+                                                              0, 0, 0);
+                    facc[i][j][0] += d * dx4[i].x * (float)c[0] - m * sm4[i].x;
+                    facc[i][j][1] += d * dx4[i].y * (float)c[1] - m * sm4[i].y;
+                    facc[i][j][2] += d * dx4[i].z * (float)c[2] - m * sm4[i].z;
+                    facc[i][j][3] += d * dx4[i].w * (float)c[3] - m * sm4[i].w;
+                }
             }
         }
     };
@@ -320,19 +361,23 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
 
     // ---- epilogue (same contract as gemm.hip k_gemm) ----
     #pragma unroll
-    for (int r = 0; r < 4; r++) {
-        const int m = m0 + wm * 16 + lk * 4 + r;
-        if (m >= M) continue;
+    for (int i = 0; i < FM; i++) {
         #pragma unroll
-        for (int j = 0; j < JF; j++) {
-            const int n = n0 + wn * (BN / WNW) + j * 16 + lrow;
-            if (n >= N) continue;
-            const size_t idx = (size_t)m * ldc + n;
-            const float rv = (res && (!splitk || bz == 0)) ? res[idx] : 0.f;
-            if (splitk) {
-                atomicAdd(&C[idx], facc[j][r] + rv);
-            } else {
-                C[idx] = facc[j][r] + rv;
+        for (int r = 0; r < 4; r++) {
+            const int m = m0 + wm * (BM_ / WMW) + i * 16 + lk * 4 + r;
+            if (m >= M) continue;
+            #pragma unroll
+            for (int j = 0; j < JF; j++) {
+                const int n = n0 + wn * (BN / WNW) + j * 16 + lrow;
+                if (n >= N) continue;
+                const size_t idx = (size_t)m * ldc + n;
+                const float rv =
+                    (res && (!splitk || bz == 0)) ? res[idx] : 0.f;
+                if (splitk) {
+                    atomicAdd(&C[idx], facc[i][j][r] + rv);
+                } else {
+                    C[idx] = facc[i][j][r] + rv;
+                }
             }
         }
     }
@@ -404,7 +449,7 @@ void launch_quant_rows(const float* X, int8_t* xq, float* xsc, int M, int K,
 }
 
 bool gemm_i8_supported(DT dtype, int M, int K) {
-    return (dtype == DT::DQ4K || dtype == DT::DQ8) && M <= 128 &&
+    return (dtype == DT::DQ4K || dtype == DT::DQ8) && M <= 1024 &&
            K % BK == 0 && K % 256 == 0;
 }
 
@@ -414,8 +459,8 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
     const int N = (int)w.n, K = (int)w.k;
     if (!gemm_i8_supported(w.dtype, M, K) || !w.hdr2 || !w.qs2)
         throw std::runtime_error("gemm_i8: unsupported dtype/shape");
-    const bool bm16 = M <= 16;
-    const int bm_tiles = bm16 ? 1 : (M + 31) / 32;
+    const int BMSEL = M <= 16 ? 16 : (M <= 32 ? 32 : 128);
+    const int bm_tiles = (M + BMSEL - 1) / BMSEL;
     const int n_tiles = (N + BN - 1) / BN;
     const int splitk =
         force_splitk > 0 ? force_splitk : gemm_splitk_factor(N, K, M);
@@ -425,15 +470,16 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
         hipLaunchKernelGGL((k_gemm_i8<WT, BMV>), grid, block, 0, stream,       \
             (const uint8_t*)w.qs2, (const uint8_t*)w.hdr2, xq, xsc,            \
             res, C, M, N, K, ldc, ldxq, k_chunk)
+    #define GI8_BM(WT)                                                         \
+        do { if (BMSEL == 16) GI8_ONE(WT, 16);                                 \
+             else if (BMSEL == 32) GI8_ONE(WT, 32);                            \
+             else GI8_ONE(WT, 128); } while (0)
     switch (w.dtype) {
-        case DT::DQ4K:
-            if (bm16) GI8_ONE(DT::DQ4K, 16); else GI8_ONE(DT::DQ4K, 32);
-            break;
-        case DT::DQ8:
-            if (bm16) GI8_ONE(DT::DQ8, 16); else GI8_ONE(DT::DQ8, 32);
-            break;
+        case DT::DQ4K: GI8_BM(DT::DQ4K); break;
+        case DT::DQ8: GI8_BM(DT::DQ8); break;
         default: throw std::runtime_error("gemm_i8: quant dtypes only");
     }
+    #undef GI8_BM
     #undef GI8_ONE
 }
 
