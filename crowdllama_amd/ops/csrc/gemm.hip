@@ -647,8 +647,8 @@ int gemm_splitk_factor(int N, int K, int M) {
     // are enough (bm x n) tiles to fill the chip without it
     const bool small_m = M <= 128;
     if (!small_m) return 1;
-    // BM tiling matches the launchers: 16/32 decode tiles, 128 above
-    const int bm_tiles = M <= 32 ? 1 : (M + 127) / 128;
+    // BM tiling matches the launchers: 16/32 decode tiles
+    const int bm_tiles = (M + 31) / 32;
     const int n_tiles = (N + BN - 1) / BN;
     const int tgt = splitk_target(M);
     const int wgs = n_tiles * bm_tiles;

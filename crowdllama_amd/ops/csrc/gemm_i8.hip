@@ -449,7 +449,13 @@ void launch_quant_rows(const float* X, int8_t* xq, float* xsc, int M, int K,
 }
 
 bool gemm_i8_supported(DT dtype, int M, int K) {
-    return (dtype == DT::DQ4K || dtype == DT::DQ8) && M <= 1024 &&
+    // M <= 128: BM=16/32 decode tiles. The BM=128 FM=4 variant exists
+    // (template above) but measured SLOWER than both the BM=32-with-
+    // re-read i8 path (B=64: 4458 vs 8338 tok/s) and the bf16 prefill
+    // kernel for 1024-row chunks (prefill 1541 vs 1382 ms at 16x1024) —
+    // its 20 KB/buffer LDS footprint halves occupancy. Kept unselected;
+    // see docs/PERF_NOTES.md round-2 notes.
+    return (dtype == DT::DQ4K || dtype == DT::DQ8) && M <= 128 &&
            K % BK == 0 && K % 256 == 0;
 }
 
@@ -459,7 +465,7 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
     const int N = (int)w.n, K = (int)w.k;
     if (!gemm_i8_supported(w.dtype, M, K) || !w.hdr2 || !w.qs2)
         throw std::runtime_error("gemm_i8: unsupported dtype/shape");
-    const int BMSEL = M <= 16 ? 16 : (M <= 32 ? 32 : 128);
+    const int BMSEL = M <= 16 ? 16 : (M <= 128 ? 32 : 128);
     const int bm_tiles = (M + BMSEL - 1) / BMSEL;
     const int n_tiles = (N + BN - 1) / BN;
     const int splitk =
