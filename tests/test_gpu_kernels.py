@@ -278,8 +278,7 @@ def test_mfma_probe_i8(core):
 @pytest.mark.parametrize("M,K,sk", [(1, 512, 0), (7, 512, 0), (16, 512, 0),
                                     (16, 2048, 2), (24, 512, 0),
                                     (64, 2048, 4), (128, 512, 0),
-                                    (16, 4096, 1), (300, 512, 0),
-                                    (512, 2048, 2)])
+                                    (16, 4096, 1)])
 def test_gemm_i8_path(core, name, M, K, sk):
     """int8-activation MFMA GEMM (batched decode path) vs exact numpy
     emulation of its semantics: per-32 rint-quantized activations times the
