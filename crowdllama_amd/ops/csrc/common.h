@@ -73,12 +73,14 @@ __host__ __device__ inline int64_t dhdr_row_bytes(DT t, int64_t k) {
 // Transposed, PRE-DECODED header copy for the i8 GEMM's LDS-DMA scale
 // staging (gemm_i8.hip): one k-window's scales for all rows contiguous,
 // already in f32 so the drain does no header math.
-// DQ4K: [K/32][N][{f32 d*sc, f32 dmin*mn}]; DQ8: [K/32][N][f32 d].
+// DQ4K: [K/32][N][{f32 d*sc, f32 dmin*mn}]; DQ8: [K/32][N][f32 d];
+// DQ6K: [K/16][N][f32 d*sc16] (per-16 scales -> masked-MFMA K=16 drains).
 // Bytes per row (total = n * this):
 __host__ __device__ inline int64_t dhdr2_row_bytes(DT t, int64_t k) {
     switch (t) {
         case DT::DQ4K: return k / 32 * 8;
         case DT::DQ8: return k / 32 * 4;
+        case DT::DQ6K: return k / 16 * 4;
         default: return 0;
     }
 }
@@ -140,6 +142,22 @@ inline void build_hdr2_rows(DT t, const uint8_t* hdr, int64_t hrb,
                     out + ((size_t)b * rows_total + r) * 4) =
                     f16_bits_to_f32_host(h[b]);
         }
+    } else if (t == DT::DQ6K) {
+        // device hdr per row: [nsb][32B]{f16 d, pad2, i8 sc[16], pad};
+        // hdr2 entry per k16 block s: f32 d*sc[s%16]
+        const int64_t n16 = k / 16;
+        for (int64_t r = r_lo; r < r_hi; r++) {
+            const uint8_t* h = hdr + r * hrb;
+            for (int64_t s = 0; s < n16; s++) {
+                const uint8_t* e = h + (s >> 4) * 32;
+                const float d = f16_bits_to_f32_host(
+                    *reinterpret_cast<const uint16_t*>(e));
+                const int8_t sc =
+                    reinterpret_cast<const int8_t*>(e)[4 + (s & 15)];
+                *reinterpret_cast<float*>(
+                    out + ((size_t)s * rows_total + r) * 4) = d * (float)sc;
+            }
+        }
     }
 }
 
@@ -159,7 +177,7 @@ __host__ __device__ inline int64_t i8g_rawb(DT t) {
 // only 32 B per 128 B line at BK=64 windows (4x HBM over-fetch — the
 // round-2 ~1.5 TB/s wall on both GEMM kernels). Rows past N are zeros.
 __host__ __device__ inline int64_t dqs2_bytes(DT t, int64_t n, int64_t k) {
-    if (t != DT::DQ4K && t != DT::DQ8) return 0;
+    if (t != DT::DQ4K && t != DT::DQ8 && t != DT::DQ6K) return 0;
     const int64_t nb = (n + I8G_BN - 1) / I8G_BN;
     return nb * (k / I8G_BK) * I8G_BN * i8g_rawb(t);
 }
@@ -186,7 +204,7 @@ inline void build_qs2_rows(DT t, const uint8_t* qs, int64_t qs_rb,
                 // qs row layout: [nsb][128B]; BK window = q-group of 64 =
                 // 32 contiguous bytes at sb*128 + q*32
                 src = qs + r * qs_rb + (kb >> 8) * 128 + ((kb & 255) >> 6) * 32;
-            } else {
+            } else {  // DQ8 / DQ6K: row-major int8
                 src = qs + r * qs_rb + kb;
             }
             for (int64_t b = 0; b < rawb; b++) dst[b] = src[b];

@@ -119,7 +119,9 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     constexpr int NGW = (BN / 4) * RAWB / 1024;  // W glds per wave (1 or 2)
     constexpr int NGX = XSHARED ? BM_ * BK / 4096
                                 : BM_ * BK / 1024;   // X glds per wave
-    constexpr int NGH = (W == DT::DQ4K) ? JF : JF / 2;  // hdr2 glds/wave
+    // hdr2 glds/wave: Q4K JF (f32 pairs per 32-block), Q8 JF/2 (f32 per
+    // 32-block), Q6K JF (f32 per 16-block: 4 k16 x JF*16 cols x 4B)
+    constexpr int NGH = (W == DT::DQ8) ? JF / 2 : JF;
     constexpr int NGS = 1;                       // xsc glds per wave
     constexpr int NGLT = NGW + NGX + NGH + NGS;  // per wave per tile
     // --- single LDS array (a second __shared__ object would make hipcc
@@ -132,13 +134,13 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     constexpr int WSZ = BN * RAWB;
     constexpr int XSZ = BM_ * BK;
     constexpr int XTOT = XSHARED ? XSZ : 4 * XSZ;
-    constexpr int HSZ = (W == DT::DQ4K) ? JF * 256 : JF * 128;
+    constexpr int HSZ = (W == DT::DQ8) ? JF * 128 : JF * 256;
     constexpr int SSZ = FM * 256;
     constexpr int BUFSZ = WSZ + XTOT + 4 * (HSZ + SSZ);
-    // 4-deep DMA ring: the single-barrier pipeline below keeps NBUF-2
-    // tiles fully in flight, so each tile's DMA has ~2 iterations of
-    // compute to hide under (PMC: 1-deep exposed ~38% parked)
-    constexpr int NBUF = 4;
+    // 3-deep DMA ring: 4 was latency-equivalent (PMC: the park is
+    // barrier/arrival, not DMA) and its LDS cost one WG/CU of residency
+    // (44 KB -> 3 WGs/CU; 34 KB -> 4, letting split-K target 1024)
+    constexpr int NBUF = 3;
     __shared__ __attribute__((aligned(16))) int8_t lds[NBUF][BUFSZ];
 
     const int tid = threadIdx.x;
@@ -232,6 +234,23 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
                               __builtin_amdgcn_readfirstlane(
                                   (unsigned)(size_t)&lds[pb][
                                       ho + kb2 * (JF * 128) + gg * 256]));
+            } else if constexpr (W == DT::DQ6K && JF == 2) {
+                // per-16 scales: 4 k16 blocks x 32 cols x f32 = 512B per
+                // wave = 2 glds4 (each: 2 k16 blocks x 32 cols)
+                const int kg16 = kb >> 4;
+                #pragma unroll
+                for (int g = 0; g < 2; g++)
+                    glds4(hdr2 + ((int64_t)(kg16 + g * 2 + (lane >> 5)) * N
+                                  + cb + (lane & 31)) * 4,
+                          __builtin_amdgcn_readfirstlane(
+                              (unsigned)(size_t)&lds[pb][ho + g * 256]));
+            } else if constexpr (W == DT::DQ6K) {  // JF == 4 (BM32/128)
+                const int kg16 = kb >> 4;
+                #pragma unroll
+                for (int g = 0; g < 4; g++)
+                    glds4(hdr2 + ((int64_t)(kg16 + g) * N + cb + lane) * 4,
+                          __builtin_amdgcn_readfirstlane(
+                              (unsigned)(size_t)&lds[pb][ho + g * 256]));
             } else if constexpr (JF == 2) {  // Q8 BM16: one glds, both kb2
                 glds4(hdr2 + ((int64_t)(kg + (lane >> 5)) * N + cb +
                               (lane & 31)) * 4,
@@ -279,12 +298,13 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     auto mfma_tile = [&](int pb) {
         // raw W fragment bytes: one b64 per j serves both K=32 halves (Q4K)
         long rawj[JF][(W == DT::DQ4K) ? 1 : 2];
+        (void)0;
         #pragma unroll
         for (int j = 0; j < JF; j++) {
             const int r = wn * (BN / WNW) + j * 16 + lrow;
             rawj[j][0] = *reinterpret_cast<const long*>(
                 &lds[pb][r * RAWB + lk * 8]);
-            if constexpr (W == DT::DQ8)
+            if constexpr (W == DT::DQ8 || W == DT::DQ6K)
                 rawj[j][1] = *reinterpret_cast<const long*>(
                     &lds[pb][r * RAWB + 32 + lk * 8]);
         }
@@ -308,6 +328,38 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
                 const int cl = j * 16 + lrow;   // wave-local column
                 long b;
                 float d, m;
+                if constexpr (W == DT::DQ6K) {
+                    // per-16 scales: split the K=32 MFMA into two
+                    // lane-masked halves (lanes lk<2 carry k16 block 0,
+                    // lk>=2 block 1 — each lane's 8 values lie entirely
+                    // within one k16 block) and scale each i32 result by
+                    // its block's d*sc16.
+                    const long braw = rawj[j][kb2];
+                    const long b_lo = (lk < 2) ? braw : 0;
+                    const long b_hi = (lk >= 2) ? braw : 0;
+                    const int k16base = kb2 * 2;   // k16 index within tile
+                    const float d0 = *reinterpret_cast<const float*>(
+                        &lds[pb][ho + (k16base + 0) * (JF * 64) + cl * 4]);
+                    const float d1 = *reinterpret_cast<const float*>(
+                        &lds[pb][ho + (k16base + 1) * (JF * 64) + cl * 4]);
+                    #pragma unroll
+                    for (int i = 0; i < FM; i++) {
+                        v4i c0 = {0, 0, 0, 0}, c1 = {0, 0, 0, 0};
+                        c0 = __builtin_amdgcn_mfma_i32_16x16x32_i8(
+                            a[i], b_lo, c0, 0, 0, 0);
+                        c1 = __builtin_amdgcn_mfma_i32_16x16x32_i8(
+                            a[i], b_hi, c1, 0, 0, 0);
+                        facc[i][j][0] += dx4[i].x *
+                            (d0 * (float)c0[0] + d1 * (float)c1[0]);
+                        facc[i][j][1] += dx4[i].y *
+                            (d0 * (float)c0[1] + d1 * (float)c1[1]);
+                        facc[i][j][2] += dx4[i].z *
+                            (d0 * (float)c0[2] + d1 * (float)c1[2]);
+                        facc[i][j][3] += dx4[i].w *
+                            (d0 * (float)c0[3] + d1 * (float)c1[3]);
+                    }
+                    continue;
+                }
                 if constexpr (W == DT::DQ4K) {
                     b = (kb2 == 0)
                             ? (rawj[j][0] & 0x0F0F0F0F0F0F0F0FLL)
@@ -455,8 +507,8 @@ bool gemm_i8_supported(DT dtype, int M, int K) {
     // kernel for 1024-row chunks (prefill 1541 vs 1382 ms at 16x1024) —
     // its 20 KB/buffer LDS footprint halves occupancy. Kept unselected;
     // see docs/PERF_NOTES.md round-2 notes.
-    return (dtype == DT::DQ4K || dtype == DT::DQ8) && M <= 128 &&
-           K % BK == 0 && K % 256 == 0;
+    return (dtype == DT::DQ4K || dtype == DT::DQ8 || dtype == DT::DQ6K) &&
+           M <= 128 && K % BK == 0 && K % 256 == 0;
 }
 
 void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
@@ -483,6 +535,7 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
     switch (w.dtype) {
         case DT::DQ4K: GI8_BM(DT::DQ4K); break;
         case DT::DQ8: GI8_BM(DT::DQ8); break;
+        case DT::DQ6K: GI8_BM(DT::DQ6K); break;
         default: throw std::runtime_error("gemm_i8: quant dtypes only");
     }
     #undef GI8_BM

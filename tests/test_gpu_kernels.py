@@ -274,7 +274,7 @@ def test_mfma_probe_i8(core):
     np.testing.assert_array_equal(C, want)
 
 
-@pytest.mark.parametrize("name", ["q4k", "q8"])
+@pytest.mark.parametrize("name", ["q4k", "q8", "q6k"])
 @pytest.mark.parametrize("M,K,sk", [(1, 512, 0), (7, 512, 0), (16, 512, 0),
                                     (16, 2048, 2), (24, 512, 0),
                                     (64, 2048, 4), (128, 512, 0),
