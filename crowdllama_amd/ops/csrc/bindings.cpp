@@ -155,6 +155,13 @@ PYBIND11_MODULE(_core, m) {
             return d;
         });
 
+    m.def("test_f16_decode", [](py::array_t<uint16_t> bits) {
+        // host f16->f32 decoder (common.h) — CPU-testable
+        py::array_t<float> out((py::ssize_t)bits.size());
+        for (py::ssize_t i = 0; i < bits.size(); i++)
+            out.mutable_data()[i] = f16_bits_to_f32_host(bits.data()[i]);
+        return out;
+    });
     m.def("test_rccl_graph_1rank", [](py::array_t<float> in) {
         std::vector<float> v(in.data(), in.data() + in.size());
         auto out = test_rccl_graph_1rank(v);

@@ -95,7 +95,12 @@ inline float f16_bits_to_f32_host(uint16_t h) {
         if (m == 0) {
             out = s << 31;
         } else {
-            int ex = -1;
+            // subnormal: value = (mm/1024) * 2^(-14-k) after k shifts
+            // normalize; biased exponent 113 - k (an off-by-one here
+            // HALVED subnormal scales — caught by Q6_K super-scales,
+            // which land subnormal for small-magnitude weights; pinned
+            // by test_f16_decode_exhaustive)
+            int ex = 0;
             uint32_t mm = m;
             while (!(mm & 1024u)) { mm <<= 1; ex--; }
             out = (s << 31) | ((uint32_t)(127 - 15 + 1 + ex) << 23) |

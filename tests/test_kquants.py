@@ -91,3 +91,25 @@ def test_q4_k_scale_pack_unpack():
     sc2, mn2 = _unpack_q4k_scales(s)
     np.testing.assert_array_equal(sc, sc2)
     np.testing.assert_array_equal(mn, mn2)
+
+
+def test_f16_decode_exhaustive():
+    """Host f16->f32 decoder (ops/csrc/common.h) vs numpy float16 over all
+    65536 bit patterns. Subnormals matter: Q6_K super-scales of
+    small-magnitude weight blocks land subnormal, and an off-by-one in
+    the normalization HALVED them (caught via the i8 GEMM Q6_K path)."""
+    import numpy as np
+    import pytest
+    try:
+        from crowdllama_amd.ops import get_core
+        core = get_core()
+    except Exception:
+        pytest.skip("ops extension unavailable")
+    bits = np.arange(65536, dtype=np.uint16)
+    got = np.asarray(core.test_f16_decode(bits))
+    want = bits.view(np.float16).astype(np.float32)
+    fin = np.isfinite(want)
+    np.testing.assert_array_equal(got[fin], want[fin])
+    assert np.all(np.isnan(got[np.isnan(want)]))
+    inf = np.isinf(want)
+    np.testing.assert_array_equal(got[inf], want[inf])
