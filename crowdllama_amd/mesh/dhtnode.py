@@ -79,17 +79,24 @@ class DHTServer:
         self._tasks.clear()
         if self._server:
             self._server.close()
-            await self._server.wait_closed()
-            self._server = None
         # asyncio's Server.close() only stops ACCEPTING; live connection
         # handlers would keep answering old clients from this (stopped)
-        # node's state — close them so clients fail over/reconnect
+        # node's state — close them so clients fail over/reconnect. This
+        # must happen BEFORE wait_closed(): on 3.10 wait_closed blocks
+        # until the last live handler detaches, so waiting first
+        # deadlocks against a peer holding a persistent connection.
         for w in list(self._conn_writers):
             try:
                 w.close()
             except Exception:
                 pass
         self._conn_writers.clear()
+        if self._server:
+            try:
+                await asyncio.wait_for(self._server.wait_closed(), 5)
+            except asyncio.TimeoutError:
+                pass
+            self._server = None
 
     # ------------------------------------------------------------ serving
 

@@ -45,7 +45,10 @@ class IPCServer:
     async def stop(self) -> None:
         if self._server:
             self._server.close()
-            await self._server.wait_closed()
+            try:   # bounded: 3.10 wait_closed blocks on live handlers
+                await asyncio.wait_for(self._server.wait_closed(), 5)
+            except asyncio.TimeoutError:
+                pass
             self._server = None
         if os.path.exists(self.path):
             os.unlink(self.path)

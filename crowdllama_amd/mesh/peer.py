@@ -87,13 +87,17 @@ class Peer:
         await self.discovery.close()
         if self._server:
             self._server.close()
-            await self._server.wait_closed()
-        for w in list(self._conn_writers):   # see DHTServer.stop
-            try:
-                w.close()
+        for w in list(self._conn_writers):   # see DHTServer.stop: close
+            try:                             # BEFORE wait_closed or 3.10
+                w.close()                    # deadlocks on live handlers
             except Exception:
                 pass
         self._conn_writers.clear()
+        if self._server:
+            try:
+                await asyncio.wait_for(self._server.wait_closed(), 5)
+            except asyncio.TimeoutError:
+                pass
         for e in self.engines.values():
             await e.close()
 
