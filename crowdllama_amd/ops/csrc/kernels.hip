@@ -585,6 +585,285 @@ __global__ __launch_bounds__(256) void k_gemv_q8(
     }
 }
 
+// ------------------------------------------------------ register-x GEMV
+//
+// Round-2 B=1 decode path (guide row: "GEMV / M<=16 decode weights: load
+// straight to VGPRs, deep unroll, late vmcnt"). Key facts this design
+// exploits:
+//  - For every quant dtype the chunk->k mapping makes lane l's chunk set
+//    (c = l*CPL .. l*CPL+CPL-1) cover EXACTLY x[l*64*(SEGF/64) ...): the
+//    whole activation stripe fits in 64 (or 32) VGPRs per lane, loaded
+//    ONCE per wave — no LDS staging, no barriers in the sweep, and the
+//    per-WG re-staging traffic of the LDS kernel (which at B=1 rivals
+//    the weight traffic itself: ~16 KB x N/4 rows) is gone.
+//  - Each wave streams ROWS: per row, each lane issues CPL consecutive
+//    dwordx4 weight loads (the wave covers the stripe's payload as one
+//    perfectly-coalesced clause) into an NB-slot software pipeline, so
+//    2-3 rows of weight loads are always in flight per lane.
+//  - Dequant uses v_cvt_f32_ubyte0..3 (extract+convert in one VALU op)
+//    and, for Q4_K, the regroup sum((d*sc*q - dmin*mn)*x) =
+//    d*sc*sum(q*x) - dmin*mn*sum(x) with sum(x) per 16-run precomputed
+//    once per lane — ~2.3 VALU/weight, under the HBM-bound time.
+//  - K > 4096 runs as gridDim.z stripes of 4096 (+ one 2048 tail launch)
+//    accumulating with atomicAdd into a res-prefilled y (k_fill_res) —
+//    the down projection's K=14336 and the 70B o-projection's K=8192.
+//    PRE_RMS needs the whole K in one stripe (the norm is global), so
+//    K > 4096 PRE_RMS falls back to the LDS kernel.
+// Reference parity: same mat-vec the reference delegates to llama.cpp
+// (SURVEY.md §2.3); numerics = plain f32 dot of dequantized weights.
+
+__global__ __launch_bounds__(256) void k_fill_res(
+    const float* __restrict__ res, float* __restrict__ y, int n) {
+    const int i = (int)blockIdx.x * 256 + threadIdx.x;
+    if (i < n) y[i] = res ? res[i] : 0.f;
+}
+
+template <DT W, int P, int SEGF>
+__global__ __launch_bounds__(256) void k_gemv_r(
+    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
+    const float* __restrict__ xin, const float* __restrict__ gw,
+    const float* __restrict__ res, float* __restrict__ y,
+    int N, int K, int kbeg0, int accum, float eps) {
+    static_assert(W == DT::DQ4K || W == DT::DQ6K || W == DT::DQ8,
+                  "register-x GEMV covers quantized weights only");
+    constexpr int CPL = (W == DT::DQ4K) ? SEGF / 32 : SEGF / 16;
+    constexpr int NB = (W == DT::DQ4K) ? 3 : 2;   // pipeline slots (VGPRs)
+    static_assert(CPL >= 1 && CPL <= 4, "stripe must be 2048 or 4096");
+
+    const int lane = threadIdx.x & 63;
+    const int wid = (int)blockIdx.x * 4 + (threadIdx.x >> 6);
+    const int nw = (int)gridDim.x * 4;
+    const int kbeg = kbeg0 + (int)blockIdx.z * (SEGF * 64);
+    const int64_t qs_rb = dqs_row_bytes(W, K);
+    const int64_t hdr_rb = dhdr_row_bytes(W, K);
+    int64_t qso, hdro;   // stripe byte offsets within a row
+    if constexpr (W == DT::DQ4K) { qso = kbeg / 2; hdro = (int64_t)(kbeg >> 8) * 32; }
+    else if constexpr (W == DT::DQ8) { qso = kbeg; hdro = (int64_t)(kbeg >> 5) * 2; }
+    else { qso = kbeg; hdro = (int64_t)(kbeg >> 8) * 32; }
+    const uint8_t* qsb = qs + qso;
+    const uint8_t* hdb = hdr + hdro;
+
+    // ---- load this lane's x segment into registers (once) ----
+    auto ldx = [&](int k) -> float4 {   // k global; 16B-aligned
+        if constexpr (P == PRE_SILU) {
+            const float4 g = *reinterpret_cast<const float4*>(xin + k);
+            const float4 u = *reinterpret_cast<const float4*>(xin + K + k);
+            float4 o;
+            o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+            o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+            o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+            o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
+            return o;
+        }
+        return *reinterpret_cast<const float4*>(xin + k);
+    };
+    float4 xr[SEGF / 4];
+    #pragma unroll
+    for (int j = 0; j < CPL; j++) {
+        const int c = lane * CPL + j;
+        if constexpr (W == DT::DQ4K) {
+            const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
+            const int k0 = kbeg + sb * 256 + q * 64 + h * 16;
+            #pragma unroll
+            for (int t = 0; t < 4; t++) {
+                xr[j * 8 + t] = ldx(k0 + t * 4);
+                xr[j * 8 + 4 + t] = ldx(k0 + 32 + t * 4);
+            }
+        } else {
+            const int k0 = kbeg + c * 16;
+            #pragma unroll
+            for (int t = 0; t < 4; t++) xr[j * 4 + t] = ldx(k0 + t * 4);
+        }
+    }
+    if constexpr (P == PRE_RMS) {   // single-stripe only: K == SEGF*64
+        __shared__ float red[4];
+        float ss = 0.f;
+        #pragma unroll
+        for (int t = 0; t < SEGF / 4; t++) {
+            const float4 v = xr[t];
+            ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+        }
+        ss = wave_reduce_sum(ss);
+        if (lane == 0) red[threadIdx.x >> 6] = ss;
+        __syncthreads();
+        const float inv =
+            rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
+        #pragma unroll
+        for (int j = 0; j < CPL; j++) {
+            const int c = lane * CPL + j;
+            #pragma unroll
+            for (int t = 0; t < 4; t++) {
+                int ka, kb, ia, ib;
+                if constexpr (W == DT::DQ4K) {
+                    const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
+                    const int k0 = kbeg + sb * 256 + q * 64 + h * 16;
+                    ka = k0 + t * 4; kb = k0 + 32 + t * 4;
+                    ia = j * 8 + t; ib = j * 8 + 4 + t;
+                } else {
+                    ka = kbeg + c * 16 + t * 4; ia = j * 4 + t;
+                    kb = ka; ib = ia;   // unused second slot
+                }
+                const float4 g = *reinterpret_cast<const float4*>(gw + ka);
+                float4 v = xr[ia];
+                v.x *= inv * g.x; v.y *= inv * g.y;
+                v.z *= inv * g.z; v.w *= inv * g.w;
+                xr[ia] = v;
+                if constexpr (W == DT::DQ4K) {
+                    const float4 g2 = *reinterpret_cast<const float4*>(gw + kb);
+                    float4 v2 = xr[ib];
+                    v2.x *= inv * g2.x; v2.y *= inv * g2.y;
+                    v2.z *= inv * g2.z; v2.w *= inv * g2.w;
+                    xr[ib] = v2;
+                }
+            }
+        }
+    }
+    // Q4_K: per-16-run x sums for the d*sc*sum(qx) - dmin*mn*sum(x) regroup
+    float sxl[CPL], sxh[CPL];
+    if constexpr (W == DT::DQ4K) {
+        #pragma unroll
+        for (int j = 0; j < CPL; j++) {
+            float a = 0.f, b = 0.f;
+            #pragma unroll
+            for (int t = 0; t < 4; t++) {
+                const float4 l = xr[j * 8 + t], h = xr[j * 8 + 4 + t];
+                a += l.x + l.y + l.z + l.w;
+                b += h.x + h.y + h.z + h.w;
+            }
+            sxl[j] = a; sxh[j] = b;
+        }
+    } else {
+        #pragma unroll
+        for (int j = 0; j < CPL; j++) { sxl[j] = 0.f; sxh[j] = 0.f; }
+    }
+
+    // rows this wave owns: wid, wid+nw, ... (< N)
+    const int rows_my = (N > wid) ? (N - wid + nw - 1) / nw : 0;
+
+    u32x4 qv[NB][CPL];
+    uint2 hd4[NB];      // DQ4K pair header
+    uint32_t hs[NB];    // DQ8 packed f16 d's / DQ6K packed sc bytes
+    uint32_t hdd[NB];   // DQ6K f16 d bits
+    uint32_t rv[NB];    // res[r] bits, prefetched with the row's weights:
+                        // a res load at the store join costs a vmcnt(0)
+                        // drain of the whole pipeline EVERY row (seen in
+                        // disassembly) even when res == nullptr. When res
+                        // is null we load y[r] (valid memory) and zero it
+                        // with a bit mask — bitwise, so garbage bits can
+                        // never produce a NaN, and no select on the load
+                        // path that SimplifyCFG would turn back into a
+                        // branch + sunk load.
+    const float* resl = res ? res : y;
+    const uint32_t rmask = res ? 0xFFFFFFFFu : 0u;
+
+    auto stage = [&](int slot, int i) {
+        const int ic = i < rows_my - 1 ? i : rows_my - 1;   // clamp: L2 hit
+        const int r = wid + ic * nw;
+        const uint8_t* qrow = qsb + (int64_t)r * qs_rb;
+        const uint8_t* hrow = hdb + (int64_t)r * hdr_rb;
+        rv[slot] = __float_as_uint(resl[r]);
+        #pragma unroll
+        for (int j = 0; j < CPL; j++)
+            qv[slot][j] = __builtin_nontemporal_load(
+                reinterpret_cast<const u32x4*>(qrow) + lane * CPL + j);
+        if constexpr (W == DT::DQ4K) {
+            hd4[slot] = *reinterpret_cast<const uint2*>(
+                hrow + ((lane * CPL) >> 1) * 8);
+        } else if constexpr (W == DT::DQ8) {
+            if constexpr (CPL == 4)
+                hs[slot] = *reinterpret_cast<const uint32_t*>(hrow + lane * 4);
+            else
+                hs[slot] = *reinterpret_cast<const uint16_t*>(hrow + lane * 2);
+        } else {   // DQ6K: 32 B/sb {f16 d; i8 sc[16]}
+            const int sb = (lane * CPL) >> 4, s0 = (lane * CPL) & 15;
+            const uint8_t* hb = hrow + (int64_t)sb * 32;
+            hdd[slot] = *reinterpret_cast<const uint16_t*>(hb);
+            if constexpr (CPL == 4)
+                hs[slot] = *reinterpret_cast<const uint32_t*>(hb + 4 + s0);
+            else
+                hs[slot] = *reinterpret_cast<const uint16_t*>(hb + 4 + s0);
+        }
+    };
+    // (float)((dw >> 8n) & 0xff) matches LLVM's CVT_F32_UBYTEn combine:
+    // one v_cvt_f32_ubyte{0..3} per weight, no separate extract.
+    auto dot16 = [&](uint32_t lo, const float4& x) -> float {
+        return (float)(lo & 0xFF) * x.x
+             + (float)((lo >> 8) & 0xFF) * x.y
+             + (float)((lo >> 16) & 0xFF) * x.z
+             + (float)(lo >> 24) * x.w;
+    };
+    auto sdot16 = [&](uint32_t dw, const float4& x) -> float {
+        return (float)(int)(int8_t)(dw & 0xFF) * x.x
+             + (float)(int)(int8_t)((dw >> 8) & 0xFF) * x.y
+             + (float)(int)(int8_t)((dw >> 16) & 0xFF) * x.z
+             + (float)(int)(int8_t)(dw >> 24) * x.w;
+    };
+    auto compute = [&](int slot, int i) {
+        float acc = 0.f;
+        if constexpr (W == DT::DQ4K) {
+            const uint2 hd = hd4[slot];
+            const float d = f16_bits_to_f32(hd.x & 0xFFFF);
+            const float dmin = f16_bits_to_f32(hd.x >> 16);
+            const float dl = d * (float)(hd.y & 0xFF);
+            const float ml = dmin * (float)((hd.y >> 8) & 0xFF);
+            const float dh = d * (float)((hd.y >> 16) & 0xFF);
+            const float mh = dmin * (float)(hd.y >> 24);
+            #pragma unroll
+            for (int j = 0; j < CPL; j++) {
+                float ql = 0.f, qh = 0.f;
+                #pragma unroll
+                for (int t = 0; t < 4; t++) {
+                    const uint32_t dw = qv[slot][j][t];
+                    ql += dot16(dw & 0x0F0F0F0Fu, xr[j * 8 + t]);
+                    qh += dot16((dw >> 4) & 0x0F0F0F0Fu, xr[j * 8 + 4 + t]);
+                }
+                acc += dl * ql - ml * sxl[j] + dh * qh - mh * sxh[j];
+            }
+        } else if constexpr (W == DT::DQ8) {
+            #pragma unroll
+            for (int j = 0; j < CPL; j++) {
+                float q = 0.f;
+                #pragma unroll
+                for (int t = 0; t < 4; t++)
+                    q += sdot16(qv[slot][j][t], xr[j * 4 + t]);
+                const float dj = f16_bits_to_f32(
+                    (hs[slot] >> (CPL == 4 ? (j >> 1) * 16 : 0)) & 0xFFFF);
+                acc += dj * q;
+            }
+        } else {   // DQ6K
+            const float d = f16_bits_to_f32(hdd[slot]);
+            #pragma unroll
+            for (int j = 0; j < CPL; j++) {
+                float q = 0.f;
+                #pragma unroll
+                for (int t = 0; t < 4; t++)
+                    q += sdot16(qv[slot][j][t], xr[j * 4 + t]);
+                const float dj =
+                    d * (float)(int)(int8_t)((hs[slot] >> (8 * j)) & 0xFF);
+                acc += dj * q;
+            }
+        }
+        const float v = wave_reduce_sum(acc);
+        if (i < rows_my && lane == 0) {
+            const int r = wid + i * nw;
+            if (accum) atomicAdd(y + r, v);
+            else y[r] = v + __uint_as_float(rv[slot] & rmask);
+        }
+    };
+
+    if (rows_my > 0) {
+        #pragma unroll
+        for (int s = 0; s < NB - 1; s++) stage(s, s);
+        for (int i0 = 0; i0 < rows_my; i0 += NB) {
+            #pragma unroll
+            for (int j = 0; j < NB; j++) {
+                stage((j + NB - 1) % NB, i0 + j + NB - 1);
+                compute(j, i0 + j);
+            }
+        }
+    }
+}
+
 // Global-x GEMV: no LDS staging — the activation vector (<=57 KB) is
 // L1-resident per CU after first touch, so reading it directly unlocks
 // full occupancy (no 16 KB+ LDS budget per workgroup) and removes the
@@ -1226,11 +1505,58 @@ void launch_gemv_q8(const WTensor& w, int pre, const float* xin,
     #undef GEMVQ_CASE
 }
 
+static int gemv_r_mode() {   // CLA_GEMV_R=0 -> legacy LDS kernel (A/B)
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("CLA_GEMV_R");
+        v = (e && e[0] == '0') ? 0 : 1;
+    }
+    return v;
+}
+
 void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
                  const float* res, float* y, int B, int ldy, float eps,
                  hipStream_t stream) {
     const int N = (int)w.n, K = (int)w.k;
     if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
+    const bool quant_w = (w.dtype == DT::DQ4K || w.dtype == DT::DQ6K ||
+                          w.dtype == DT::DQ8);
+    if (B == 1 && quant_w && gemv_r_mode() && K >= 2048 &&
+        K % 2048 == 0 && (pre != PRE_RMS || K <= 4096)) {
+        const int nwg = std::min((N + 3) / 4, 1024);
+        const int nfull = K / 4096, tail = K % 4096;   // tail: 0 or 2048
+        const bool multi = K > 4096;
+        #define GEMVR_SEG(WT, SEGF, GZ, KB, ACC, RES)                          \
+            do {                                                               \
+                auto kern = (pre == PRE_RMS) ? k_gemv_r<WT, PRE_RMS, SEGF>     \
+                           : (pre == PRE_SILU) ? k_gemv_r<WT, PRE_SILU, SEGF>  \
+                           : k_gemv_r<WT, PRE_NONE, SEGF>;                     \
+                hipLaunchKernelGGL(kern, dim3(nwg, 1, GZ), dim3(256), 0,       \
+                    stream, (const uint8_t*)w.qs, (const uint8_t*)w.hdr,       \
+                    xin, gw, RES, y, N, K, KB, ACC, eps);                      \
+            } while (0)
+        #define GEMVR_CASE(WT)                                                 \
+            do {                                                               \
+                if (!multi) {                                                  \
+                    if (K == 4096) GEMVR_SEG(WT, 64, 1, 0, 0, res);            \
+                    else GEMVR_SEG(WT, 32, 1, 0, 0, res);                      \
+                } else {                                                       \
+                    hipLaunchKernelGGL(k_fill_res, dim3((N + 255) / 256),      \
+                        dim3(256), 0, stream, res, y, N);                      \
+                    GEMVR_SEG(WT, 64, nfull, 0, 1, nullptr);                   \
+                    if (tail)                                                  \
+                        GEMVR_SEG(WT, 32, 1, nfull * 4096, 1, nullptr);        \
+                }                                                              \
+            } while (0)
+        switch (w.dtype) {
+            case DT::DQ4K: GEMVR_CASE(DT::DQ4K); return;
+            case DT::DQ6K: GEMVR_CASE(DT::DQ6K); return;
+            case DT::DQ8:  GEMVR_CASE(DT::DQ8);  return;
+            default: break;
+        }
+        #undef GEMVR_CASE
+        #undef GEMVR_SEG
+    }
     const size_t lds = (size_t)B * (K + (K >> 4)) * 4 + 8 * 4;
     // rows per wave fixed at 1, 256-thread blocks: the measured optimum
     // (A/B-rejected: RPW>1, 3-buffer rotation, pair-unroll, 512-thread

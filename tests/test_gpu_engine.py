@@ -76,6 +76,49 @@ def test_engine_greedy_matches_ref(core, tiny_gguf):
     assert match >= 4, f"got {got} want {want}"
 
 
+def test_engine_gemv_r_shapes(core, tmp_path_factory):
+    """B=1 logits parity on a model whose projections hit the register-x
+    GEMV (k_gemv_r: K=2048 half-stripe for qkv/o/gate_up/head, K=4096
+    full stripe for down) instead of the tiny-shape legacy kernel the
+    other tests cover — including the o-projection residual prefetch."""
+    from crowdllama_amd.engine.ref_numpy import RefLlama
+    from crowdllama_amd.models import write_synthetic_gguf
+    from crowdllama_amd.models.presets import ModelConfig
+    mc = ModelConfig("gemvr", vocab_size=512, hidden_size=2048, n_layers=2,
+                     n_heads=16, n_kv_heads=8, ffn_hidden=4096,
+                     rope_theta=10000.0, max_seq_len=512)
+    path = str(tmp_path_factory.mktemp("m") / "gemvr.gguf")
+    write_synthetic_gguf(path, mc, scheme="q4_k_m", mode="exact", seed=11)
+    cfg = core.EngineConfig()
+    cfg.batch = 1
+    cfg.max_seq = 128
+    cfg.act_q8 = False
+    eng = core.Engine(path, cfg)
+    prompt = [3, 17, 99, 250, 7]
+    eng.prefill(np.array([prompt], dtype=np.int32))
+    got = np.asarray(eng.logits(0))
+    ref = RefLlama(path)
+    logits = None
+    for t in prompt:
+        logits = ref.step(t)
+    denom = np.abs(logits).max() + 1e-6
+    rel = np.abs(got - logits).max() / denom
+    assert rel < 5e-3, f"max rel err {rel}"
+    assert int(np.argmax(got)) == int(np.argmax(logits))
+    # decode steps run every projection through k_gemv_r (the prefill
+    # above only exercised the head GEMV)
+    eng.decode(5)
+    got_ids = list(eng.gen_tokens(0))
+    want = ref.generate(prompt, 6)
+    assert got_ids[0] == want[0]
+    match = 0
+    for a, b in zip(got_ids, want):
+        if a != b:
+            break
+        match += 1
+    assert match >= 4, f"got {got_ids} want {want}"
+
+
 def test_engine_batch2(core, tiny_gguf):
     cfg = core.EngineConfig()
     cfg.batch = 2

@@ -156,6 +156,42 @@ def test_gemv_pre_silu(core):
     np.testing.assert_allclose(y, yref, rtol=2e-4, atol=2e-4)
 
 
+@pytest.mark.parametrize("name", list(CASES))
+@pytest.mark.parametrize("K,pre", [
+    (2048, PRE_NONE), (2048, PRE_RMS), (2048, PRE_SILU),
+    (4096, PRE_NONE), (4096, PRE_RMS), (4096, PRE_SILU),
+    (14336, PRE_NONE), (14336, PRE_SILU),   # striped: fill + atomicAdd
+    (6144, PRE_SILU),                       # 4096 stripe + 2048 tail
+])
+def test_gemv_r_path(core, name, K, pre):
+    """Register-x B=1 GEMV (k_gemv_r): every quant dtype on the stripe
+    shapes the engine actually launches (K=2048 CPL-half, K=4096 full,
+    K>4096 multi-stripe atomic accumulation). K>=2048 at B=1 selects the
+    new kernel inside launch_gemv."""
+    dt, quant, dequant, repack_fn = CASES[name]
+    rng = np.random.default_rng(K * 7 + pre)
+    N = 192
+    w = rng.standard_normal((N, K)).astype(np.float32) * 0.1
+    raw = quant(w)
+    wref = dequant(raw, K).reshape(N, K)
+    qs, hdr = repack_fn(raw.reshape(N, -1), N, K)
+    gw = rng.standard_normal(K).astype(np.float32)
+    if pre == PRE_SILU:
+        x = rng.standard_normal((1, 2 * K)).astype(np.float32)
+        g, u = x[:, :K], x[:, K:]
+        xe = (g / (1 + np.exp(-g))) * u
+    else:
+        x = rng.standard_normal((1, K)).astype(np.float32)
+        xe = x
+        if pre == PRE_RMS:
+            xe = x / np.sqrt((x * x).mean(axis=1, keepdims=True) + 1e-5) * gw
+    y = core.test_gemv(np.ascontiguousarray(qs), np.ascontiguousarray(hdr),
+                       x, dt, N, K, pre,
+                       gw if pre == PRE_RMS else np.zeros(0, dtype=np.float32))
+    yref = xe @ wref.T
+    np.testing.assert_allclose(y, yref, rtol=5e-4, atol=5e-4)
+
+
 def test_mfma_fragment_layout(core):
     """Verify the assumed v_mfma_f32_16x16x32_bf16 lane mappings with
     asymmetric inputs (transpose-detecting, guide §5.4 rule 16)."""
