@@ -22,6 +22,7 @@ void launch_gemm_test(const void*, const void*, const float*, float*, int,
                       int, int, int, size_t, size_t);
 double bench_gemm(int, int, int, int, int);
 double bench_gemv_g(int, int, int, int, int);
+double bench_membw(int nt, int mb, int wgs, int iters);
 void launch_gemv_q8_test(const void*, const void*, const float*, float*, int,
                          int, int, int, size_t, size_t);
 double bench_gemv_q8(int, int, int, int, int);
@@ -202,6 +203,10 @@ PYBIND11_MODULE(_core, m) {
         return py::bytes(reinterpret_cast<const char*>(out.data()), out.size());
     });
 
+    m.def("bench_membw", [](int nt, int mb, int wgs, int iters) {
+        return bench_membw(nt, mb, wgs, iters);
+    }, py::arg("nt") = 1, py::arg("mb") = 1024, py::arg("wgs") = 2048,
+       py::arg("iters") = 10);
     m.def("bench_gemv_g", [](int dtype, int N, int K, int B, int iters) {
         return bench_gemv_g(dtype, N, K, B, iters);
     });

@@ -676,7 +676,10 @@ __global__ __launch_bounds__(256) void k_gemv_r(
         }
     }
     if constexpr (P == PRE_RMS) {   // single-stripe only: K == SEGF*64
-        __shared__ float red[4];
+        // EVERY wave holds the full x in registers, so the norm is a pure
+        // wave-local reduce — a cross-wave LDS pass here would count each
+        // x^2 once per wave (a 4x mean = exactly-half outputs, caught by
+        // test_gemv_r_path + gemvr_diag on hardware).
         float ss = 0.f;
         #pragma unroll
         for (int t = 0; t < SEGF / 4; t++) {
@@ -684,10 +687,7 @@ __global__ __launch_bounds__(256) void k_gemv_r(
             ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
         }
         ss = wave_reduce_sum(ss);
-        if (lane == 0) red[threadIdx.x >> 6] = ss;
-        __syncthreads();
-        const float inv =
-            rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
+        const float inv = rsqrtf(ss / (float)K + eps);
         #pragma unroll
         for (int j = 0; j < CPL; j++) {
             const int c = lane * CPL + j;
@@ -737,8 +737,14 @@ __global__ __launch_bounds__(256) void k_gemv_r(
         for (int j = 0; j < CPL; j++) { sxl[j] = 0.f; sxh[j] = 0.f; }
     }
 
-    // rows this wave owns: wid, wid+nw, ... (< N)
-    const int rows_my = (N > wid) ? (N - wid + nw - 1) / nw : 0;
+    // rows this wave owns: the CONTIGUOUS block [wid*rpw, wid*rpw+rows_my).
+    // Contiguous (not nw-strided) so one wave's row stream walks adjacent
+    // DRAM pages, and sized >=4 rows by the launcher so the one-time x
+    // register-load phase amortizes (both measured: strided single-row
+    // waves ran the o-projection at 740 GB/s).
+    const int rpw = (N + nw - 1) / nw;
+    const int r0 = wid * rpw;
+    const int rows_my = (N > r0) ? ((N - r0) < rpw ? (N - r0) : rpw) : 0;
 
     u32x4 qv[NB][CPL];
     uint2 hd4[NB];      // DQ4K pair header
@@ -758,7 +764,7 @@ __global__ __launch_bounds__(256) void k_gemv_r(
 
     auto stage = [&](int slot, int i) {
         const int ic = i < rows_my - 1 ? i : rows_my - 1;   // clamp: L2 hit
-        const int r = wid + ic * nw;
+        const int r = r0 + ic;
         const uint8_t* qrow = qsb + (int64_t)r * qs_rb;
         const uint8_t* hrow = hdb + (int64_t)r * hdr_rb;
         rv[slot] = __float_as_uint(resl[r]);
@@ -845,7 +851,7 @@ __global__ __launch_bounds__(256) void k_gemv_r(
         }
         const float v = wave_reduce_sum(acc);
         if (i < rows_my && lane == 0) {
-            const int r = wid + i * nw;
+            const int r = r0 + i;
             if (accum) atomicAdd(y + r, v);
             else y[r] = v + __uint_as_float(rv[slot] & rmask);
         }
@@ -1523,7 +1529,11 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
                           w.dtype == DT::DQ8);
     if (B == 1 && quant_w && gemv_r_mode() && K >= 2048 &&
         K % 2048 == 0 && (pre != PRE_RMS || K <= 4096)) {
-        const int nwg = std::min((N + 3) / 4, 1024);
+        // >=8 rows/wave when N is large (amortize the x-load phase and
+        // keep each wave's weight stream long); floor of 256 WGs (one
+        // per CU) when N allows at >=4 rows/wave.
+        const int nwg = std::max(std::min((N + 31) / 32, 1024),
+                                 std::min((N + 15) / 16, 256));
         const int nfull = K / 4096, tail = K % 4096;   // tail: 0 or 2048
         const bool multi = K > 4096;
         #define GEMVR_SEG(WT, SEGF, GZ, KB, ACC, RES)                          \

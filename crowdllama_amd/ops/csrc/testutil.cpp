@@ -292,3 +292,61 @@ std::vector<float> test_rccl_graph_1rank(const std::vector<float>& in) {
 }
 
 }  // namespace cla
+
+namespace cla {
+
+// ---- HBM streaming-read calibration (what does THIS box achieve for a
+// pure contiguous dwordx4 stream per lane? baseline for GEMV roofline) ----
+typedef unsigned int bw_u32x4 __attribute__((ext_vector_type(4)));
+template <int NT>
+__global__ __launch_bounds__(256) void k_membw(
+    const bw_u32x4* __restrict__ p, uint32_t* __restrict__ sink, size_t n16) {
+    // each lane streams a contiguous block of DEPTH x 16B chunks per step
+    constexpr int DEPTH = 8;
+    const size_t tid = (size_t)blockIdx.x * 256 + threadIdx.x;
+    const size_t nthreads = (size_t)gridDim.x * 256;
+    uint32_t acc = 0;
+    for (size_t base = tid * DEPTH; base + DEPTH <= n16;
+         base += nthreads * DEPTH) {
+        bw_u32x4 v[DEPTH];
+        #pragma unroll
+        for (int j = 0; j < DEPTH; j++) {
+            if constexpr (NT) v[j] = __builtin_nontemporal_load(p + base + j);
+            else v[j] = p[base + j];
+        }
+        #pragma unroll
+        for (int j = 0; j < DEPTH; j++)
+            acc ^= v[j].x ^ v[j].y ^ v[j].z ^ v[j].w;
+    }
+    if (acc == 0xDEADBEEFu) *sink = acc;   // never true: keep loads alive
+}
+
+double bench_membw(int nt, int mb, int wgs, int iters) {
+    const size_t bytes = (size_t)mb << 20;
+    void* d = nullptr;
+    HIP_CHECK(hipMalloc(&d, bytes));
+    HIP_CHECK(hipMemset(d, 1, bytes));
+    uint32_t* sink = nullptr;
+    HIP_CHECK(hipMalloc(&sink, 4));
+    const size_t n16 = bytes / 16;
+    auto kern = nt ? k_membw<1> : k_membw<0>;
+    for (int i = 0; i < 2; i++)
+        hipLaunchKernelGGL(kern, dim3(wgs), dim3(256), 0, 0,
+                           (const bw_u32x4*)d, sink, n16);
+    HIP_CHECK(hipDeviceSynchronize());
+    hipEvent_t e0, e1;
+    HIP_CHECK(hipEventCreate(&e0)); HIP_CHECK(hipEventCreate(&e1));
+    HIP_CHECK(hipEventRecord(e0, 0));
+    for (int i = 0; i < iters; i++)
+        hipLaunchKernelGGL(kern, dim3(wgs), dim3(256), 0, 0,
+                           (const bw_u32x4*)d, sink, n16);
+    HIP_CHECK(hipEventRecord(e1, 0));
+    HIP_CHECK(hipEventSynchronize(e1));
+    float ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+    hipEventDestroy(e0); hipEventDestroy(e1);
+    hipFree(d); hipFree(sink);
+    return (double)bytes * iters / ((double)ms * 1e6);   // GB/s
+}
+
+}  // namespace cla
