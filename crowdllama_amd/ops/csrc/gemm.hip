@@ -51,16 +51,16 @@ __device__ __forceinline__ void load_w_raw(
     if constexpr (W == DT::DQ4K) {
         const int sb = kb >> 8, q = (kb & 255) >> 6, p = 2 * q + h;
         const uint8_t* qrow = qs + row * (K / 256) * 128;
-        r->q0 = __builtin_nontemporal_load(
+        r->q0 = *(
             reinterpret_cast<const u32x4g*>(qrow) + sb * 8 + p);
         r->hd = reinterpret_cast<const uint2*>(
             hdr + row * (K / 256) * 32)[sb * 4 + p / 2];
     } else if constexpr (W == DT::DQ6K || W == DT::DQ8) {
         const int k0 = kb + h * 32;
         const int8_t* qrow = reinterpret_cast<const int8_t*>(qs + row * K);
-        r->q0 = __builtin_nontemporal_load(
+        r->q0 = *(
             reinterpret_cast<const u32x4g*>(qrow + k0));
-        r->q1 = __builtin_nontemporal_load(
+        r->q1 = *(
             reinterpret_cast<const u32x4g*>(qrow + k0) + 1);
         if constexpr (W == DT::DQ6K) {
             const uint8_t* hb = hdr + row * (K / 256) * 32 + (k0 >> 8) * 32;
