@@ -1532,9 +1532,15 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
         K % 2048 == 0 && (pre != PRE_RMS || K <= 4096)) {
         // >=8 rows/wave when N is large (amortize the x-load phase and
         // keep each wave's weight stream long); floor of 256 WGs (one
-        // per CU) when N allows at >=4 rows/wave.
-        const int nwg = std::max(std::min((N + 31) / 32, 1024),
-                                 std::min((N + 15) / 16, 256));
+        // per CU) when N allows at >=4 rows/wave. CLA_GEMVR_WGS
+        // overrides for on-HW rows/wave sweeps.
+        int nwg = std::max(std::min((N + 31) / 32, 1024),
+                           std::min((N + 15) / 16, 256));
+        static int wgs_env = [] {
+            const char* e = getenv("CLA_GEMVR_WGS");
+            return e ? atoi(e) : 0;
+        }();
+        if (wgs_env > 0) nwg = std::min(wgs_env, (N + 3) / 4);
         const int nfull = K / 4096, tail = K % 4096;   // tail: 0 or 2048
         const bool multi = K > 4096;
         #define GEMVR_SEG(WT, SEGF, GZ, KB, ACC, RES)                          \
