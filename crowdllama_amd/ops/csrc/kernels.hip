@@ -643,80 +643,76 @@ __global__ __launch_bounds__(256) void k_gemv_r(
     else { qso = kbeg; hdro = (int64_t)(kbeg >> 8) * 32; }
     const uint8_t* qsb = qs + qso;
     const uint8_t* hdb = hdr + hdro;
+    constexpr int KL = SEGF * 64;            // stripe width in weights
 
-    // ---- load this lane's x segment into registers (once) ----
-    auto ldx = [&](int k) -> float4 {   // k global; 16B-aligned
-        if constexpr (P == PRE_SILU) {
-            const float4 g = *reinterpret_cast<const float4*>(xin + k);
-            const float4 u = *reinterpret_cast<const float4*>(xin + K + k);
-            float4 o;
-            o.x = (g.x / (1.f + __expf(-g.x))) * u.x;
-            o.y = (g.y / (1.f + __expf(-g.y))) * u.y;
-            o.z = (g.z / (1.f + __expf(-g.z))) * u.z;
-            o.w = (g.w / (1.f + __expf(-g.w))) * u.w;
-            return o;
+    // ---- cooperative, COALESCED x phase ----
+    // The per-lane chunk->k mapping is a 256 B lane-stride gather: loading
+    // x straight into the mapped registers issues 64 single-line requests
+    // per instruction (measured as a ~10-15 us per-wave ramp that only
+    // head-sized rows/wave amortized). Instead the WG loads the stripe
+    // lane-LINEARLY (perfectly coalesced), applies SILU / gw and the
+    // block-wide RMS reduce in that linear pass, bounces through LDS, and
+    // each lane then gathers its mapped segment with ds_reads.
+    __shared__ float4 xs4[KL / 4];
+    __shared__ float redw[8];
+    {
+        const int tid = threadIdx.x;
+        float ssp = 0.f;
+        #pragma unroll
+        for (int i = 0; i < KL / 4 / 256; i++) {
+            const int k4 = i * 256 + tid;          // float4 index in stripe
+            const int k = kbeg + k4 * 4;
+            float4 v;
+            if constexpr (P == PRE_SILU) {
+                const float4 g = *reinterpret_cast<const float4*>(xin + k);
+                const float4 u = *reinterpret_cast<const float4*>(xin + K + k);
+                v.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+                v.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+                v.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+                v.w = (g.w / (1.f + __expf(-g.w))) * u.w;
+            } else {
+                v = *reinterpret_cast<const float4*>(xin + k);
+            }
+            if constexpr (P == PRE_RMS) {
+                ssp += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+                const float4 g = *reinterpret_cast<const float4*>(gw + k);
+                v.x *= g.x; v.y *= g.y; v.z *= g.z; v.w *= g.w;
+            }
+            xs4[k4] = v;
         }
-        return *reinterpret_cast<const float4*>(xin + k);
-    };
+        if constexpr (P == PRE_RMS) {
+            ssp = wave_reduce_sum(ssp);
+            if ((tid & 63) == 0) redw[tid >> 6] = ssp;
+        }
+        __syncthreads();
+    }
+    float inv = 1.f;
+    if constexpr (P == PRE_RMS)   // single-stripe only: K == KL
+        inv = rsqrtf((redw[0] + redw[1] + redw[2] + redw[3]) / (float)K + eps);
+
+    // gather this lane's mapped segment out of LDS (xr holds x*gw*inv)
     float4 xr[SEGF / 4];
     #pragma unroll
     for (int j = 0; j < CPL; j++) {
         const int c = lane * CPL + j;
         if constexpr (W == DT::DQ4K) {
             const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
-            const int k0 = kbeg + sb * 256 + q * 64 + h * 16;
+            const int k0 = sb * 256 + q * 64 + h * 16;   // stripe-local
             #pragma unroll
             for (int t = 0; t < 4; t++) {
-                xr[j * 8 + t] = ldx(k0 + t * 4);
-                xr[j * 8 + 4 + t] = ldx(k0 + 32 + t * 4);
+                xr[j * 8 + t] = xs4[(k0 >> 2) + t];
+                xr[j * 8 + 4 + t] = xs4[((k0 + 32) >> 2) + t];
             }
         } else {
-            const int k0 = kbeg + c * 16;
+            const int k0 = c * 16;
             #pragma unroll
-            for (int t = 0; t < 4; t++) xr[j * 4 + t] = ldx(k0 + t * 4);
+            for (int t = 0; t < 4; t++) xr[j * 4 + t] = xs4[(k0 >> 2) + t];
         }
     }
-    if constexpr (P == PRE_RMS) {   // single-stripe only: K == SEGF*64
-        // EVERY wave holds the full x in registers, so the norm is a pure
-        // wave-local reduce — a cross-wave LDS pass here would count each
-        // x^2 once per wave (a 4x mean = exactly-half outputs, caught by
-        // test_gemv_r_path + gemvr_diag on hardware).
-        float ss = 0.f;
+    if constexpr (P == PRE_RMS) {
         #pragma unroll
         for (int t = 0; t < SEGF / 4; t++) {
-            const float4 v = xr[t];
-            ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
-        }
-        ss = wave_reduce_sum(ss);
-        const float inv = rsqrtf(ss / (float)K + eps);
-        #pragma unroll
-        for (int j = 0; j < CPL; j++) {
-            const int c = lane * CPL + j;
-            #pragma unroll
-            for (int t = 0; t < 4; t++) {
-                int ka, kb, ia, ib;
-                if constexpr (W == DT::DQ4K) {
-                    const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
-                    const int k0 = kbeg + sb * 256 + q * 64 + h * 16;
-                    ka = k0 + t * 4; kb = k0 + 32 + t * 4;
-                    ia = j * 8 + t; ib = j * 8 + 4 + t;
-                } else {
-                    ka = kbeg + c * 16 + t * 4; ia = j * 4 + t;
-                    kb = ka; ib = ia;   // unused second slot
-                }
-                const float4 g = *reinterpret_cast<const float4*>(gw + ka);
-                float4 v = xr[ia];
-                v.x *= inv * g.x; v.y *= inv * g.y;
-                v.z *= inv * g.z; v.w *= inv * g.w;
-                xr[ia] = v;
-                if constexpr (W == DT::DQ4K) {
-                    const float4 g2 = *reinterpret_cast<const float4*>(gw + kb);
-                    float4 v2 = xr[ib];
-                    v2.x *= inv * g2.x; v2.y *= inv * g2.y;
-                    v2.z *= inv * g2.z; v2.w *= inv * g2.w;
-                    xr[ib] = v2;
-                }
-            }
+            xr[t].x *= inv; xr[t].y *= inv; xr[t].z *= inv; xr[t].w *= inv;
         }
     }
     // Q4_K: per-16-run x sums for the d*sc*sum(qx) - dmin*mn*sum(x) regroup
