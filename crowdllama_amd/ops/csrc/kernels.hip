@@ -628,7 +628,12 @@ __global__ __launch_bounds__(256) void k_gemv_r(
     static_assert(W == DT::DQ4K || W == DT::DQ6K || W == DT::DQ8,
                   "register-x GEMV covers quantized weights only");
     constexpr int CPL = (W == DT::DQ4K) ? SEGF / 32 : SEGF / 16;
-    constexpr int NB = (W == DT::DQ4K) ? 3 : 2;   // pipeline slots (VGPRs)
+    // pipeline depth: NB-1 rows of weight loads in flight per lane. The
+    // binding constraint at 8 waves/CU is outstanding bytes (measured:
+    // ~2 us effective latency wants ~4 rows in flight), so go as deep as
+    // the register file allows: 5 slots for DQ4K (11 VGPRs/slot), 3 for
+    // DQ8/DQ6K (19 VGPRs/slot).
+    constexpr int NB = (W == DT::DQ4K) ? 5 : 3;
     static_assert(CPL >= 1 && CPL <= 4, "stripe must be 2048 or 4096");
 
     const int lane = threadIdx.x & 63;
