@@ -605,11 +605,11 @@ __global__ __launch_bounds__(256) void k_gemv_q8(
 //    and, for Q4_K, the regroup sum((d*sc*q - dmin*mn)*x) =
 //    d*sc*sum(q*x) - dmin*mn*sum(x) with sum(x) per 16-run precomputed
 //    once per lane — ~2.3 VALU/weight, under the HBM-bound time.
-//  - K > 4096 runs as gridDim.z stripes of 4096 (+ one 2048 tail launch)
-//    accumulating with atomicAdd into a res-prefilled y (k_fill_res) —
-//    the down projection's K=14336 and the 70B o-projection's K=8192.
-//    PRE_RMS needs the whole K in one stripe (the norm is global), so
-//    K > 4096 PRE_RMS falls back to the LDS kernel.
+//  - Selected for single-stripe shapes (K = 2048 or 4096) only. A
+//    gridDim.z-striped atomicAdd variant for K > 4096 (kept: accum/kbeg
+//    args + k_fill_res) measured SLOWER than the legacy LDS kernel on
+//    the down projection (~33 vs 23 us: x re-staging there is L2-served
+//    and the stripe launches don't amortize), so K > 4096 falls back.
 // Reference parity: same mat-vec the reference delegates to llama.cpp
 // (SURVEY.md §2.3); numerics = plain f32 dot of dequantized weights.
 
@@ -629,11 +629,14 @@ __global__ __launch_bounds__(256) void k_gemv_r(
                   "register-x GEMV covers quantized weights only");
     constexpr int CPL = (W == DT::DQ4K) ? SEGF / 32 : SEGF / 16;
     // pipeline depth: NB-1 rows of weight loads in flight per lane. The
-    // binding constraint at 8 waves/CU is outstanding bytes (measured:
-    // ~2 us effective latency wants ~4 rows in flight), so go as deep as
-    // the register file allows: 5 slots for DQ4K (11 VGPRs/slot), 3 for
-    // DQ8/DQ6K (19 VGPRs/slot).
-    constexpr int NB = (W == DT::DQ4K) ? 5 : 3;
+    // binding constraint at 8 waves/CU is outstanding bytes (~2 us
+    // effective latency wants ~4 rows in flight), so go as deep as the
+    // register file allows: 5 slots for DQ4K (11 VGPRs/slot) — except
+    // PRE_RMS, where the extra live state tipped occupancy and measured
+    // slower (qkv 8.4 -> 10.7 us) — and 3 for DQ8/DQ6K (19 VGPRs/slot;
+    // the Q6_K head hit 6.0 TB/s = 93% of the streaming ceiling there).
+    constexpr int NB =
+        (W == DT::DQ4K) ? (P == PRE_RMS ? 3 : 5) : 3;
     static_assert(CPL >= 1 && CPL <= 4, "stripe must be 2048 or 4096");
 
     const int lane = threadIdx.x & 63;
@@ -1529,42 +1532,34 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
     if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
     const bool quant_w = (w.dtype == DT::DQ4K || w.dtype == DT::DQ6K ||
                           w.dtype == DT::DQ8);
-    if (B == 1 && quant_w && gemv_r_mode() && K >= 2048 &&
-        K % 2048 == 0 && (pre != PRE_RMS || K <= 4096)) {
-        // >=8 rows/wave when N is large (amortize the x-load phase and
-        // keep each wave's weight stream long); floor of 256 WGs (one
-        // per CU) when N allows at >=4 rows/wave. CLA_GEMVR_WGS
-        // overrides for on-HW rows/wave sweeps.
-        int nwg = std::max(std::min((N + 31) / 32, 1024),
-                           std::min((N + 15) / 16, 256));
+    if (B == 1 && quant_w && gemv_r_mode() &&
+        (K == 2048 || K == 4096)) {
+        // Single-stripe shapes only: for K > 4096 (the down projection)
+        // a gridDim.z-striped atomic variant measured ~33 us vs the
+        // legacy LDS kernel's 23 us (x re-staging there is L2-served and
+        // cheap at B=1), so multi-stripe stays unselected.
+        // nwg: ~8 waves/CU saturates the stream for big N; smaller N
+        // can't afford the per-wave ramp at 512 WGs (rows/wave < 4).
+        int nwg = (N >= 16384) ? 512 : 256;
+        nwg = std::min(nwg, (N + 3) / 4);
         static int wgs_env = [] {
             const char* e = getenv("CLA_GEMVR_WGS");
             return e ? atoi(e) : 0;
         }();
         if (wgs_env > 0) nwg = std::min(wgs_env, (N + 3) / 4);
-        const int nfull = K / 4096, tail = K % 4096;   // tail: 0 or 2048
-        const bool multi = K > 4096;
-        #define GEMVR_SEG(WT, SEGF, GZ, KB, ACC, RES)                          \
+        #define GEMVR_SEG(WT, SEGF)                                            \
             do {                                                               \
                 auto kern = (pre == PRE_RMS) ? k_gemv_r<WT, PRE_RMS, SEGF>     \
                            : (pre == PRE_SILU) ? k_gemv_r<WT, PRE_SILU, SEGF>  \
                            : k_gemv_r<WT, PRE_NONE, SEGF>;                     \
-                hipLaunchKernelGGL(kern, dim3(nwg, 1, GZ), dim3(256), 0,       \
+                hipLaunchKernelGGL(kern, dim3(nwg, 1, 1), dim3(256), 0,        \
                     stream, (const uint8_t*)w.qs, (const uint8_t*)w.hdr,       \
-                    xin, gw, RES, y, N, K, KB, ACC, eps);                      \
+                    xin, gw, res, y, N, K, 0, 0, eps);                         \
             } while (0)
         #define GEMVR_CASE(WT)                                                 \
             do {                                                               \
-                if (!multi) {                                                  \
-                    if (K == 4096) GEMVR_SEG(WT, 64, 1, 0, 0, res);            \
-                    else GEMVR_SEG(WT, 32, 1, 0, 0, res);                      \
-                } else {                                                       \
-                    hipLaunchKernelGGL(k_fill_res, dim3((N + 255) / 256),      \
-                        dim3(256), 0, stream, res, y, N);                      \
-                    GEMVR_SEG(WT, 64, nfull, 0, 1, nullptr);                   \
-                    if (tail)                                                  \
-                        GEMVR_SEG(WT, 32, 1, nfull * 4096, 1, nullptr);        \
-                }                                                              \
+                if (K == 4096) GEMVR_SEG(WT, 64);                              \
+                else GEMVR_SEG(WT, 32);                                        \
             } while (0)
         switch (w.dtype) {
             case DT::DQ4K: GEMVR_CASE(DT::DQ4K); return;
