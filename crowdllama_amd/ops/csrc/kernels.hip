@@ -1525,6 +1525,28 @@ static int gemv_r_mode() {   // CLA_GEMV_R=0 -> legacy LDS kernel (A/B)
     return v;
 }
 
+// silu(g)*u in place over [B] rows of a [B][2K] buffer (writes [0:K)).
+__global__ __launch_bounds__(256) void k_silu_mul(
+    float* __restrict__ gu, int K, int64_t ld) {
+    const int b = (int)blockIdx.y;
+    const int i = ((int)blockIdx.x * 256 + threadIdx.x) * 4;
+    if (i >= K) return;
+    float4* gp = reinterpret_cast<float4*>(gu + (int64_t)b * ld + i);
+    const float4 u = *reinterpret_cast<const float4*>(
+        gu + (int64_t)b * ld + K + i);
+    float4 g = *gp;
+    g.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+    g.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+    g.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+    g.w = (g.w / (1.f + __expf(-g.w))) * u.w;
+    *gp = g;
+}
+
+void launch_silu_mul(float* gu, int B, int K, int64_t ld, hipStream_t stream) {
+    hipLaunchKernelGGL(k_silu_mul, dim3((K / 4 + 255) / 256, B), dim3(256),
+                       0, stream, gu, K, ld);
+}
+
 void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
                  const float* res, float* y, int B, int ldy, float eps,
                  hipStream_t stream) {
