@@ -25,7 +25,6 @@ namespace cla {
 void launch_gemv(const WTensor&, int pre, const float* xin, const float* gw,
                  const float* res, float* y, int B, int ldy, float eps,
                  hipStream_t);
-void launch_silu_mul(float* gu, int B, int K, int64_t ld, hipStream_t);
 void launch_gemv_q8(const WTensor&, int pre, const float* xin,
                     const float* gw, const float* res, float* y, int B,
                     int ldy, float eps, hipStream_t);
@@ -739,12 +738,8 @@ void Engine::step(hipStream_t s) {
             for (auto& pt : L.gate_up.parts)
                 gemv_pick(cfg_.gemv_q8, pt.w, PRE_RMS, x_, L.ffn_norm, nullptr,
                             gu_ + pt.row_off, B, (int)L.gate_up.n_total, eps, s);
-            // silu(g)*u computed ONCE in-place (writes gu_[0:F]): fused
-            // PRE_SILU staging re-evaluated 14M exps across the down
-            // GEMV's workgroups (23.3 vs 16.3 us on llama3-8b's down)
-            launch_silu_mul(gu_, B, meta_.ffn_l, 2 * (int64_t)meta_.ffn_l, s);
             for (auto& pt : L.down.parts)
-                gemv_pick(cfg_.gemv_q8, pt.w, PRE_NONE, gu_, nullptr, x_ + pt.row_off,
+                gemv_pick(cfg_.gemv_q8, pt.w, PRE_SILU, gu_, nullptr, x_ + pt.row_off,
                             x_ + pt.row_off, B, (int)L.down.n_total, eps, s);
         } else if (gemv_path) {
             // TP: local partial -> all-reduce; rank 0 folds the residual so
@@ -757,9 +752,8 @@ void Engine::step(hipStream_t s) {
             for (auto& pt : L.gate_up.parts)
                 gemv_pick(cfg_.gemv_q8, pt.w, PRE_RMS, x2_, L.ffn_norm, nullptr,
                             gu_ + pt.row_off, B, (int)L.gate_up.n_total, eps, s);
-            launch_silu_mul(gu_, B, meta_.ffn_l, 2 * (int64_t)meta_.ffn_l, s);
             for (auto& pt : L.down.parts)
-                gemv_pick(cfg_.gemv_q8, pt.w, PRE_NONE, gu_, nullptr,
+                gemv_pick(cfg_.gemv_q8, pt.w, PRE_SILU, gu_, nullptr,
                             r0 ? x2_ + pt.row_off : nullptr, tmp_h_ + pt.row_off,
                             B, (int)L.down.n_total, eps, s);
             allreduce(tmp_h_, x_, (size_t)B * meta_.hidden);

@@ -875,6 +875,175 @@ __global__ __launch_bounds__(256) void k_gemv_r(
     }
 }
 
+
+// LDS-x row-streaming GEMV for long-K quant mats (the down projection,
+// K = 14336): same deep row pipeline and coalesced weight clauses as
+// k_gemv_r, but the stripe's activations live in LDS (57 KB/WG -> 2
+// WGs/CU) instead of per-lane registers, chunks are assigned lane-
+// STRIDED (c = j*64 + lane: each weight instruction reads 1 KB
+// contiguous, and every header load is one uint2/u16 per j for any
+// chunks-per-lane), and SILU folds into the one coalesced x pass
+// (replacing a separate k_silu_mul launch + the legacy kernel's
+// per-WG re-evaluation of 14M exps).
+template <DT W, int P, int KLT>   // KLT: K in units of 2048
+__global__ __launch_bounds__(256) void k_gemv_rl(
+    const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
+    const float* __restrict__ xin, const float* __restrict__ res,
+    float* __restrict__ y, int N, int K) {
+    static_assert(P != PRE_RMS, "long-K path: norm-free projections only");
+    constexpr int KL = KLT * 2048;
+    constexpr int CPL = (W == DT::DQ4K) ? KLT : 2 * KLT;
+    constexpr int NB = (W == DT::DQ4K) ? 3 : 2;
+    const int lane = threadIdx.x & 63;
+    const int wid = (int)blockIdx.x * 4 + (threadIdx.x >> 6);
+    const int nw = (int)gridDim.x * 4;
+    const int64_t qs_rb = dqs_row_bytes(W, K);
+    const int64_t hdr_rb = dhdr_row_bytes(W, K);
+
+    __shared__ float4 xs4[KL / 4];
+    {
+        const int tid = threadIdx.x;
+        #pragma unroll
+        for (int i = 0; i < KL / 4 / 256; i++) {
+            const int k4 = i * 256 + tid;
+            const int k = k4 * 4;
+            float4 v;
+            if constexpr (P == PRE_SILU) {
+                const float4 g = *reinterpret_cast<const float4*>(xin + k);
+                const float4 u = *reinterpret_cast<const float4*>(xin + K + k);
+                v.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+                v.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+                v.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+                v.w = (g.w / (1.f + __expf(-g.w))) * u.w;
+            } else {
+                v = *reinterpret_cast<const float4*>(xin + k);
+            }
+            xs4[k4] = v;
+        }
+        __syncthreads();
+    }
+
+    // per-chunk x bases (stripe-local float4 index) + Q4_K per-run x sums
+    float sxl[CPL], sxh[CPL];
+    int xb[CPL];
+    #pragma unroll
+    for (int j = 0; j < CPL; j++) {
+        const int c = j * 64 + lane;
+        if constexpr (W == DT::DQ4K) {
+            const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
+            xb[j] = (sb * 256 + q * 64 + h * 16) >> 2;
+            float a = 0.f, b = 0.f;
+            #pragma unroll
+            for (int t = 0; t < 4; t++) {
+                const float4 l = xs4[xb[j] + t], hv = xs4[xb[j] + 8 + t];
+                a += l.x + l.y + l.z + l.w;
+                b += hv.x + hv.y + hv.z + hv.w;
+            }
+            sxl[j] = a; sxh[j] = b;
+        } else {
+            xb[j] = c * 4;
+            sxl[j] = 0.f; sxh[j] = 0.f;
+        }
+    }
+
+    const int rpw = (N + nw - 1) / nw;
+    const int r0 = wid * rpw;
+    const int rows_my = (N > r0) ? ((N - r0) < rpw ? (N - r0) : rpw) : 0;
+
+    u32x4 qv[NB][CPL];
+    uint2 hd4[NB][CPL];        // DQ4K pair per j
+    uint32_t hs[NB][CPL];      // DQ8 f16 d / DQ6K sc byte per j
+    uint32_t hdd[NB][CPL];     // DQ6K f16 d per j
+    uint32_t rv[NB];
+    const float* resl = res ? res : y;
+    const uint32_t rmask = res ? 0xFFFFFFFFu : 0u;
+
+    auto stage = [&](int slot, int i) {
+        const int ic = i < rows_my - 1 ? i : rows_my - 1;
+        const int r = r0 + ic;
+        const uint8_t* qrow = qs + (int64_t)r * qs_rb;
+        const uint8_t* hrow = hdr + (int64_t)r * hdr_rb;
+        rv[slot] = __float_as_uint(resl[r]);
+        #pragma unroll
+        for (int j = 0; j < CPL; j++) {
+            const int c = j * 64 + lane;
+            qv[slot][j] = *(reinterpret_cast<const u32x4*>(qrow) + c);
+            if constexpr (W == DT::DQ4K) {
+                hd4[slot][j] = *reinterpret_cast<const uint2*>(
+                    hrow + (int64_t)(c >> 1) * 8);
+            } else if constexpr (W == DT::DQ8) {
+                hs[slot][j] = *reinterpret_cast<const uint16_t*>(
+                    hrow + (int64_t)(c >> 1) * 2);
+            } else {   // DQ6K
+                const int sb = c >> 4;
+                const uint8_t* hb = hrow + (int64_t)sb * 32;
+                hdd[slot][j] = *reinterpret_cast<const uint16_t*>(hb);
+                hs[slot][j] = hb[4 + (lane & 15)];
+            }
+        }
+    };
+    auto dot16u = [&](uint32_t lo, const float4& x) -> float {
+        return (float)(lo & 0xFF) * x.x + (float)((lo >> 8) & 0xFF) * x.y
+             + (float)((lo >> 16) & 0xFF) * x.z + (float)(lo >> 24) * x.w;
+    };
+    auto dot16s = [&](uint32_t dw, const float4& x) -> float {
+        return (float)(int)(int8_t)(dw & 0xFF) * x.x
+             + (float)(int)(int8_t)((dw >> 8) & 0xFF) * x.y
+             + (float)(int)(int8_t)((dw >> 16) & 0xFF) * x.z
+             + (float)(int)(int8_t)(dw >> 24) * x.w;
+    };
+    auto compute = [&](int slot, int i) {
+        float acc = 0.f;
+        #pragma unroll
+        for (int j = 0; j < CPL; j++) {
+            if constexpr (W == DT::DQ4K) {
+                const uint2 hd = hd4[slot][j];
+                const float d = f16_bits_to_f32(hd.x & 0xFFFF);
+                const float dmin = f16_bits_to_f32(hd.x >> 16);
+                float ql = 0.f, qh = 0.f;
+                #pragma unroll
+                for (int t = 0; t < 4; t++) {
+                    const uint32_t dw = qv[slot][j][t];
+                    ql += dot16u(dw & 0x0F0F0F0Fu, xs4[xb[j] + t]);
+                    qh += dot16u((dw >> 4) & 0x0F0F0F0Fu, xs4[xb[j] + 8 + t]);
+                }
+                acc += (d * (float)(hd.y & 0xFF)) * ql
+                     - (dmin * (float)((hd.y >> 8) & 0xFF)) * sxl[j]
+                     + (d * (float)((hd.y >> 16) & 0xFF)) * qh
+                     - (dmin * (float)(hd.y >> 24)) * sxh[j];
+            } else {
+                float q = 0.f;
+                #pragma unroll
+                for (int t = 0; t < 4; t++)
+                    q += dot16s(qv[slot][j][t], xs4[xb[j] + t]);
+                if constexpr (W == DT::DQ8) {
+                    acc += f16_bits_to_f32(hs[slot][j]) * q;
+                } else {
+                    acc += f16_bits_to_f32(hdd[slot][j])
+                         * (float)(int)(int8_t)(hs[slot][j]) * q;
+                }
+            }
+        }
+        const float v = wave_reduce_sum(acc);
+        if (i < rows_my && lane == 0) {
+            const int r = r0 + i;
+            y[r] = v + __uint_as_float(rv[slot] & rmask);
+        }
+    };
+
+    if (rows_my > 0) {
+        #pragma unroll
+        for (int sl = 0; sl < NB - 1; sl++) stage(sl, sl);
+        for (int i0 = 0; i0 < rows_my; i0 += NB) {
+            #pragma unroll
+            for (int j = 0; j < NB; j++) {
+                stage((j + NB - 1) % NB, i0 + j + NB - 1);
+                compute(j, i0 + j);
+            }
+        }
+    }
+}
+
 // Global-x GEMV: no LDS staging — the activation vector (<=57 KB) is
 // L1-resident per CU after first touch, so reading it directly unlocks
 // full occupancy (no 16 KB+ LDS budget per workgroup) and removes the
@@ -1525,28 +1694,6 @@ static int gemv_r_mode() {   // CLA_GEMV_R=0 -> legacy LDS kernel (A/B)
     return v;
 }
 
-// silu(g)*u in place over [B] rows of a [B][2K] buffer (writes [0:K)).
-__global__ __launch_bounds__(256) void k_silu_mul(
-    float* __restrict__ gu, int K, int64_t ld) {
-    const int b = (int)blockIdx.y;
-    const int i = ((int)blockIdx.x * 256 + threadIdx.x) * 4;
-    if (i >= K) return;
-    float4* gp = reinterpret_cast<float4*>(gu + (int64_t)b * ld + i);
-    const float4 u = *reinterpret_cast<const float4*>(
-        gu + (int64_t)b * ld + K + i);
-    float4 g = *gp;
-    g.x = (g.x / (1.f + __expf(-g.x))) * u.x;
-    g.y = (g.y / (1.f + __expf(-g.y))) * u.y;
-    g.z = (g.z / (1.f + __expf(-g.z))) * u.z;
-    g.w = (g.w / (1.f + __expf(-g.w))) * u.w;
-    *gp = g;
-}
-
-void launch_silu_mul(float* gu, int B, int K, int64_t ld, hipStream_t stream) {
-    hipLaunchKernelGGL(k_silu_mul, dim3((K / 4 + 255) / 256, B), dim3(256),
-                       0, stream, gu, K, ld);
-}
-
 void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
                  const float* res, float* y, int B, int ldy, float eps,
                  hipStream_t stream) {
@@ -1591,6 +1738,27 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
         }
         #undef GEMVR_CASE
         #undef GEMVR_SEG
+    }
+    if (B == 1 && quant_w && gemv_r_mode() && K == 14336 &&
+        pre != PRE_RMS) {
+        // long-K (down projection): LDS-x row streaming, silu folded
+        // into the coalesced x pass (k_gemv_rl doc above)
+        const int nwg = std::min(256, (N + 3) / 4);
+        #define GEMVRL_CASE(WT)                                                \
+            do {                                                               \
+                auto kern = (pre == PRE_SILU) ? k_gemv_rl<WT, PRE_SILU, 7>     \
+                                              : k_gemv_rl<WT, PRE_NONE, 7>;    \
+                hipLaunchKernelGGL(kern, dim3(nwg), dim3(256), 0, stream,      \
+                    (const uint8_t*)w.qs, (const uint8_t*)w.hdr, xin, res, y,  \
+                    N, K);                                                     \
+            } while (0)
+        switch (w.dtype) {
+            case DT::DQ4K: GEMVRL_CASE(DT::DQ4K); return;
+            case DT::DQ6K: GEMVRL_CASE(DT::DQ6K); return;
+            case DT::DQ8:  GEMVRL_CASE(DT::DQ8);  return;
+            default: break;
+        }
+        #undef GEMVRL_CASE
     }
     const size_t lds = (size_t)B * (K + (K >> 4)) * 4 + 8 * 4;
     // rows per wave fixed at 1, 256-thread blocks: the measured optimum
