@@ -893,7 +893,7 @@ __global__ __launch_bounds__(256) void k_gemv_rl(
     static_assert(P != PRE_RMS, "long-K path: norm-free projections only");
     constexpr int KL = KLT * 2048;
     constexpr int CPL = (W == DT::DQ4K) ? KLT : 2 * KLT;
-    constexpr int NB = (W == DT::DQ4K) ? 3 : 2;
+    constexpr int NB = 2;
     const int lane = threadIdx.x & 63;
     const int wid = (int)blockIdx.x * 4 + (threadIdx.x >> 6);
     const int nw = (int)gridDim.x * 4;
@@ -996,6 +996,10 @@ __global__ __launch_bounds__(256) void k_gemv_rl(
         float acc = 0.f;
         #pragma unroll
         for (int j = 0; j < CPL; j++) {
+            // scheduling fence per chunk: without it the scheduler lifts
+            // ALL CPL chunks' LDS x reads to the top of the row compute
+            // (224 live floats -> 256 VGPR + AGPR spill, 1 wave/SIMD)
+            __builtin_amdgcn_sched_barrier(0);
             if constexpr (W == DT::DQ4K) {
                 const uint2 hd = hd4[slot][j];
                 const float d = f16_bits_to_f32(hd.x & 0xFFFF);
