@@ -91,7 +91,7 @@ __device__ __forceinline__ void glds4(const void* gsrc, unsigned lds_off) {
 
 }  // namespace
 
-// BM_ in {16, 32, 128}. 256 threads = 4 waves; BM=16 -> 1x4 wave grid
+// BM_ in {16, 32, 64, 128}. 256 threads = 4 waves; BM=16 -> 1x4 wave grid
 // (each wave all 16 M-rows x 32 cols), BM=32 -> 2x2 (16 rows x 64 cols),
 // BM=128 -> 2x2 with FM=4 row fragments per wave (prefill tiles: one pass
 // over the W stream covers 128 activation rows instead of re-streaming W
@@ -122,7 +122,7 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     // hdr2 glds/wave: Q4K JF (f32 pairs per 32-block), Q8 JF/2 (f32 per
     // 32-block), Q6K JF (f32 per 16-block: 4 k16 x JF*16 cols x 4B)
     constexpr int NGH = (W == DT::DQ8) ? JF / 2 : JF;
-    constexpr int NGS = 1;                       // xsc glds per wave
+    constexpr int NGS = (FM == 2) ? 2 : 1;       // xsc glds per wave
     constexpr int NGLT = NGW + NGX + NGH + NGS;  // per wave per tile
     // --- single LDS array (a second __shared__ object would make hipcc
     // drain vmcnt(0) before every ds_read — guide §5 trap (a)) ---
@@ -179,7 +179,7 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     for (int g = 0; g < NGX; g++) {
         int xr;
         if constexpr (XSHARED) {
-            xr = wid * 32 + g * 16 + (lane >> 2);
+            xr = wid * (BM_ / 4) + g * 16 + (lane >> 2);
         } else {
             xr = (NGX == 1) ? (lane >> 2) : (g * 16 + (lane >> 2));
         }
@@ -278,6 +278,22 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
             glds4(xsc + soff,
                   __builtin_amdgcn_readfirstlane((unsigned)(size_t)
                       &lds[pb][so]));
+        } else if constexpr (FM == 2) {
+            // BM=64: two glds4 (64 lanes x 4B), each covering 2 of the 4
+            // (kg2 x {dx,sum}) chunks for this wave's 32 M rows
+            const int kg = kb >> 5;
+            const int cw = lane >> 5;
+            int srow = m0 + wm * (BM_ / WMW) + (lane & 31);
+            if (srow > M4 - 1) srow = M4 - 1;
+            #pragma unroll
+            for (int g = 0; g < 2; g++) {
+                const int c = g * 2 + cw;
+                const int64_t soff =
+                    ((int64_t)(kg + (c >> 1)) * 2 + (c & 1)) * M4 + srow;
+                glds4(xsc + soff,
+                      __builtin_amdgcn_readfirstlane((unsigned)(size_t)
+                          &lds[pb][so + g * 256]));
+            }
         } else {  // BM=128: one glds16 (64 lanes x 16B = 1KB) covers all
             // 4 chunks (kg2 x {dx,sum}) of this wave's 64 M rows: lane l
             // -> chunk c = l>>4, slots ((l&15)*4 .. +4)
@@ -294,7 +310,7 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         }
     };
 
-    constexpr int SROW = (FM == 1) ? 16 : 64;   // xsc slots per wave
+    constexpr int SROW = BM_ / WMW;             // xsc slots per wave
     auto mfma_tile = [&](int pb) {
         // raw W fragment bytes: one b64 per j serves both K=32 halves (Q4K)
         long rawj[JF][(W == DT::DQ4K) ? 1 : 2];
@@ -517,7 +533,16 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
     const int N = (int)w.n, K = (int)w.k;
     if (!gemm_i8_supported(w.dtype, M, K) || !w.hdr2 || !w.qs2)
         throw std::runtime_error("gemm_i8: unsupported dtype/shape");
-    const int BMSEL = M <= 16 ? 16 : (M <= 128 ? 32 : 128);
+    int BMSEL = M <= 16 ? 16 : (M <= 128 ? 32 : 128);
+    // prefill tile A/B: CLA_I8_BM=64 selects the FM=2 tile for M>32
+    // (halves the per-output weight re-stream vs BM=32 at 14 KB/buffer
+    // LDS — BM=128's 20 KB halved occupancy and lost)
+    if (M > 32) {
+        if (const char* e = getenv("CLA_I8_BM")) {
+            const int v = atoi(e);
+            if (v == 16 || v == 32 || v == 64 || v == 128) BMSEL = v;
+        }
+    }
     const int bm_tiles = (M + BMSEL - 1) / BMSEL;
     const int n_tiles = (N + BN - 1) / BN;
     const int splitk =
@@ -531,6 +556,7 @@ void launch_gemm_i8(const WTensor& w, const int8_t* xq, const float* xsc,
     #define GI8_BM(WT)                                                         \
         do { if (BMSEL == 16) GI8_ONE(WT, 16);                                 \
              else if (BMSEL == 32) GI8_ONE(WT, 32);                            \
+             else if (BMSEL == 64) GI8_ONE(WT, 64);                            \
              else GI8_ONE(WT, 128); } while (0)
     switch (w.dtype) {
         case DT::DQ4K: GI8_BM(DT::DQ4K); break;

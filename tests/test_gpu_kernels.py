@@ -338,3 +338,33 @@ def test_gemm_i8_path(core, name, M, K, sk):
     xq = (np.rint(b * rinv) * (amax / 127.0)).reshape(M, K)
     yref = xq @ wref.T
     np.testing.assert_allclose(y, yref, rtol=3e-4, atol=3e-4)
+
+
+@pytest.mark.parametrize("name", ["q4k", "q8", "q6k"])
+@pytest.mark.parametrize("M,K,sk", [(64, 512, 0), (128, 2048, 2), (48, 512, 0)])
+def test_gemm_i8_bm64(core, name, M, K, sk):
+    """The BM=64 (FM=2) prefill tile, forced via CLA_I8_BM (read per
+    launch): same numpy act_q8 reference as test_gemm_i8_path. M=48
+    covers a partial second m-tile."""
+    import os
+    dt, quant, dequant, repack_fn = CASES[name]
+    rng = np.random.default_rng(640 + M)
+    N = 192
+    w = rng.standard_normal((N, K)).astype(np.float32) * 0.1
+    raw = quant(w)
+    wref = dequant(raw, K).reshape(N, K)
+    x = rng.standard_normal((M, K)).astype(np.float32)
+    qs, hdr = repack_fn(raw.reshape(N, -1), N, K)
+    os.environ["CLA_I8_BM"] = "64"
+    try:
+        y = core.test_gemm_i8(np.ascontiguousarray(qs),
+                              np.ascontiguousarray(hdr), x, dt, N, K,
+                              force_splitk=sk)
+    finally:
+        del os.environ["CLA_I8_BM"]
+    b = x.reshape(M, -1, 32)
+    amax = np.abs(b).max(axis=2, keepdims=True)
+    rinv = np.where(amax > 0, 127.0 / np.where(amax == 0, 1, amax), 0.0)
+    xq = (np.rint(b * rinv) * (amax / 127.0)).reshape(M, K)
+    yref = xq @ wref.T
+    np.testing.assert_allclose(y, yref, rtol=3e-4, atol=3e-4)
