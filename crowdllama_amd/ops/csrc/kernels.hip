@@ -1361,6 +1361,63 @@ __global__ __launch_bounds__(256) void k_attn_decode(
         for (int j = 0; j < DPL; j++) so[(sub * G + g) * D + d0 + j] = o[g][j];
     }
     __syncthreads();
+    if (S == 1) {
+        // Single split: this WG owns (b, kvh) outright — combine the 16
+        // sub-accumulators straight from LDS into attn_out (+ fused
+        // quant) and skip the whole fan-in: no part_o round trip, no
+        // vmcnt(0) drains, no ticket atomic, and crucially no
+        // agent-scope acquire (which invalidates the XCD's L2 on CDNA).
+        if (xq) {
+            const int jl = tid & 7;
+            for (int b32 = tid >> 3; b32 < G * D / 32; b32 += 32) {
+                const int g = (b32 * 32) / D;
+                const int head = kvh * G + g;
+                float mstar = -1e30f;
+                #pragma unroll
+                for (int t = 0; t < 16; t++)
+                    mstar = fmaxf(mstar, sm[t * G + g]);
+                float denom = 0.f;
+                #pragma unroll
+                for (int t = 0; t < 16; t++)
+                    denom += __expf(sm[t * G + g] - mstar) * sl[t * G + g];
+                float4 v;
+                float* vp = &v.x;
+                #pragma unroll
+                for (int j = 0; j < 4; j++) {
+                    const int d = (b32 * 32 + jl * 4 + j) % D;
+                    float osum = 0.f;
+                    #pragma unroll
+                    for (int t = 0; t < 16; t++)
+                        osum += __expf(sm[t * G + g] - mstar) *
+                                so[(t * G + g) * D + d];
+                    vp[j] = osum / denom;
+                    attn_out[(size_t)b * NH * D + (size_t)head * D + d] =
+                        vp[j];
+                }
+                quant_block_emit(v, jl, kvh * (G * D / 32) + b32, b,
+                                 NH * D, M4, xq, xsc);
+            }
+        } else {
+            for (int idx = tid; idx < G * D; idx += 256) {
+                const int g = idx / D, d = idx % D;
+                float mstar = -1e30f;
+                #pragma unroll
+                for (int t = 0; t < 16; t++)
+                    mstar = fmaxf(mstar, sm[t * G + g]);
+                float lsum = 0.f, osum = 0.f;
+                #pragma unroll
+                for (int t = 0; t < 16; t++) {
+                    const float e = __expf(sm[t * G + g] - mstar);
+                    lsum += e * sl[t * G + g];
+                    osum += e * so[(t * G + g) * D + d];
+                }
+                const int head = kvh * G + g;
+                attn_out[(size_t)b * NH * D + (size_t)head * D + d] =
+                    osum / lsum;
+            }
+        }
+        return;
+    }
     for (int idx = tid; idx < G * D; idx += 256) {
         const int g = idx / D, d = idx % D;
         float mstar = -1e30f;
