@@ -508,7 +508,10 @@ void Engine::alloc_state() {
     // tok/s at S=4), B=32 regressed at S=1 (5549 vs 6260 at S=2), and
     // B=64 has enough workgroups without splits
     attn_splits_ = std::max(1, std::min(32, 256 / std::max(1, B * NKV)));
-    if (attn_splits_ < 2 && B <= 32) attn_splits_ = 2;
+    // floor of 2 up to B=16 only: with the fence-free S=1 combine path,
+    // B=32 measured best at its natural S=1 (6732 vs 6498 tok/s) while
+    // B=16 still wants S=2 (4883 vs 4766) and long contexts more
+    if (attn_splits_ < 2 && B <= 16) attn_splits_ = 2;
     // B<=2: cap at 8 — at short contexts the in-kernel split combine
     // (ticket spin + fences) dominates; S=8 measured 399 vs 363 tok/s at
     // S=32 on llama3-8b B=1 (S=4 equal, S=2 worse: too few WGs)
