@@ -1235,7 +1235,10 @@ __global__ __launch_bounds__(256) void k_attn_decode(
     };
     const int64_t voff = (int64_t)page_size * D;   // V plane within the slot
     const int p0 = start + sub;
+    // depth-2 prefetch: K/V of p+16 AND p+32 in flight while p's math
+    // runs (depth 1 left the scan latency-exposed every other step)
     uint32_t kw[DPL / 2], vw[DPL / 2], kwn[DPL / 2], vwn[DPL / 2];
+    uint32_t kw2[DPL / 2], vw2[DPL / 2];
     if (p0 < cache_end) {
         const uint16_t* kp = kv_addr(p0);
         #pragma unroll
@@ -1243,15 +1246,22 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             kw[j] = reinterpret_cast<const uint32_t*>(kp)[j];
             vw[j] = reinterpret_cast<const uint32_t*>(kp + voff)[j];
         }
+        const int p1 = (p0 + 16 < cache_end) ? p0 + 16 : p0;
+        const uint16_t* kp1 = kv_addr(p1);
+        #pragma unroll
+        for (int j = 0; j < DPL / 2; j++) {
+            kwn[j] = reinterpret_cast<const uint32_t*>(kp1)[j];
+            vwn[j] = reinterpret_cast<const uint32_t*>(kp1 + voff)[j];
+        }
     }
     for (int p = p0; p < cache_end; p += 16) {
         {
-            const int pn = (p + 16 < cache_end) ? p + 16 : p;
+            const int pn = (p + 32 < cache_end) ? p + 32 : p;
             const uint16_t* kp = kv_addr(pn);
             #pragma unroll
             for (int j = 0; j < DPL / 2; j++) {
-                kwn[j] = reinterpret_cast<const uint32_t*>(kp)[j];
-                vwn[j] = reinterpret_cast<const uint32_t*>(kp + voff)[j];
+                kw2[j] = reinterpret_cast<const uint32_t*>(kp)[j];
+                vw2[j] = reinterpret_cast<const uint32_t*>(kp + voff)[j];
             }
         }
         float kf[DPL];
@@ -1287,7 +1297,10 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             m[g] = mn;
         }
         #pragma unroll
-        for (int j = 0; j < DPL / 2; j++) { kw[j] = kwn[j]; vw[j] = vwn[j]; }
+        for (int j = 0; j < DPL / 2; j++) {
+            kw[j] = kwn[j]; vw[j] = vwn[j];
+            kwn[j] = kw2[j]; vwn[j] = vw2[j];
+        }
     }
 
     // current token: the quarter whose subsequence covers `pos` computes
