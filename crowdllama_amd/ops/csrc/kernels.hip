@@ -1468,22 +1468,14 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             __HIP_MEMORY_SCOPE_AGENT);
         is_last = (t == S - 1);
         if (is_last) {
-            // reset for the next launch (we are the only reader). No
-            // acquire FENCE here: on CDNA an agent-scope acquire fence
-            // lowers to a full L2 invalidate of this XCD; instead the
-            // partial reads below are agent-scope relaxed atomic loads
-            // (L2-bypassing), which see the writers' released values
-            // without nuking the cache for every following kernel.
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+            // reset for the next launch (we are the only reader)
             __hip_atomic_store(&tickets[(size_t)b * NKV + kvh], 0,
                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         }
     }
     __syncthreads();
     if (!is_last) return;
-    auto ld_agent = [](const float* p) {
-        return __hip_atomic_load(p, __ATOMIC_RELAXED,
-                                 __HIP_MEMORY_SCOPE_AGENT);
-    };
     if (xq) {
         // combine + fused i8-GEMM activation quantization (8 lanes per
         // 32-block; D%32==0 so a block never crosses heads)
@@ -1493,11 +1485,10 @@ __global__ __launch_bounds__(256) void k_attn_decode(
             const int head = kvh * G + g;
             const float* ml = part_ml + ((size_t)b * NH + head) * S * 2;
             float mg = -1e30f;
-            for (int t = 0; t < S; t++) mg = fmaxf(mg, ld_agent(ml + 2 * t));
+            for (int t = 0; t < S; t++) mg = fmaxf(mg, ml[2 * t]);
             float denom = 0.f;
             for (int t = 0; t < S; t++)
-                denom += __expf(ld_agent(ml + 2 * t) - mg) *
-                         ld_agent(ml + 2 * t + 1);
+                denom += __expf(ml[2 * t] - mg) * ml[2 * t + 1];
             float4 v;
             float* vp = &v.x;
             #pragma unroll
@@ -1506,9 +1497,8 @@ __global__ __launch_bounds__(256) void k_attn_decode(
                 float osum = 0.f;
                 #pragma unroll 4
                 for (int t = 0; t < S; t++)
-                    osum += __expf(ld_agent(ml + 2 * t) - mg) *
-                            ld_agent(part_o +
-                                     (((size_t)b * NH + head) * S + t) * D + d);
+                    osum += __expf(ml[2 * t] - mg) *
+                            part_o[(((size_t)b * NH + head) * S + t) * D + d];
                 vp[j] = osum / denom;
                 attn_out[(size_t)b * NH * D + (size_t)head * D + d] = vp[j];
             }
@@ -1522,14 +1512,13 @@ __global__ __launch_bounds__(256) void k_attn_decode(
         const int head = kvh * G + g;
         const float* ml = part_ml + ((size_t)b * NH + head) * S * 2;
         float mg = -1e30f;
-        for (int t = 0; t < S; t++) mg = fmaxf(mg, ld_agent(ml + 2 * t));
+        for (int t = 0; t < S; t++) mg = fmaxf(mg, ml[2 * t]);
         float denom = 0.f, osum = 0.f;
         #pragma unroll 4
         for (int t = 0; t < S; t++) {
-            const float e = __expf(ld_agent(ml + 2 * t) - mg);
-            denom += e * ld_agent(ml + 2 * t + 1);
-            osum += e * ld_agent(part_o +
-                                 (((size_t)b * NH + head) * S + t) * D + d);
+            const float e = __expf(ml[2 * t] - mg);
+            denom += e * ml[2 * t + 1];
+            osum += e * part_o[(((size_t)b * NH + head) * S + t) * D + d];
         }
         attn_out[(size_t)b * NH * D + (size_t)head * D + d] = osum / denom;
     }
