@@ -119,6 +119,36 @@ def test_engine_gemv_r_shapes(core, tmp_path_factory):
     assert match >= 4, f"got {got_ids} want {want}"
 
 
+def test_engine_attn_single_split(core, tiny_gguf):
+    """The fence-free S=1 attention combine (no fan-in, no ticket, no
+    agent-scope L2 invalidate — the B>16 default) produces the same
+    logits as the reference. CLA_ATTN_SPLITS is read at engine init."""
+    import os
+    from crowdllama_amd.engine.ref_numpy import RefLlama
+    cfg = core.EngineConfig()
+    cfg.batch = 1
+    cfg.max_seq = 128
+    cfg.act_q8 = False
+    os.environ["CLA_ATTN_SPLITS"] = "1"
+    try:
+        eng = core.Engine(tiny_gguf, cfg)
+    finally:
+        del os.environ["CLA_ATTN_SPLITS"]
+    prompt = [3, 17, 99, 250, 7]
+    eng.prefill(np.array([prompt], dtype=np.int32))
+    eng.decode(4)
+    got = np.asarray(eng.logits(0))
+    ref = RefLlama(tiny_gguf)
+    logits = None
+    for t in prompt:
+        logits = ref.step(t)
+    for t in list(eng.gen_tokens(0))[:4]:
+        logits = ref.step(t)
+    denom = np.abs(logits).max() + 1e-6
+    rel = np.abs(got - logits).max() / denom
+    assert rel < 5e-3, f"max rel err {rel}"
+
+
 def test_engine_batch2(core, tiny_gguf):
     cfg = core.EngineConfig()
     cfg.batch = 2
