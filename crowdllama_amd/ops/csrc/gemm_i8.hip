@@ -144,29 +144,7 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     __shared__ __attribute__((aligned(16))) int8_t lds[NBUF][BUFSZ];
 
     const int tid = threadIdx.x;
-    // XCD-aware remap for multi-m-tile (prefill) launches: the observed
-    // dispatcher places linear block b on XCD b%8, so the m-blocks that
-    // share one (n, k-chunk) W tile land on 8 DIFFERENT XCDs and refetch
-    // it through MALL instead of hitting their own L2. The bijective
-    // swizzle (guide §5.4) gives each XCD a CONTIGUOUS run of tile ids
-    // ordered m-fastest, so one XCD's L2 serves a W tile to all of its
-    // sharers. Decode launches (gridDim.y == 1) keep the identity map.
-    int bn, bm, bz;
-    if (gridDim.y > 1) {
-        const int nwg = (int)(gridDim.x * gridDim.y * gridDim.z);
-        const int orig = (int)(blockIdx.x +
-                               gridDim.x * (blockIdx.y +
-                                            gridDim.y * blockIdx.z));
-        const int q = nwg / 8, r = nwg % 8, xcd = orig % 8;
-        const int wgid = (xcd < r ? xcd * (q + 1)
-                                  : r * (q + 1) + (xcd - r) * q) + orig / 8;
-        bm = wgid % (int)gridDim.y;
-        const int t = wgid / (int)gridDim.y;
-        bn = t % (int)gridDim.x;
-        bz = t / (int)gridDim.x;
-    } else {
-        bn = (int)blockIdx.x; bm = 0; bz = (int)blockIdx.z;
-    }
+    const int bn = blockIdx.x, bm = blockIdx.y, bz = blockIdx.z;
     const bool splitk = gridDim.z > 1;
     const int m0 = bm * BM_, n0 = bn * BN;
     const int kb_lo = bz * k_chunk;
