@@ -998,8 +998,12 @@ __global__ __launch_bounds__(256) void k_gemv_rl(
         for (int j = 0; j < CPL; j++) {
             // scheduling fence per chunk: without it the scheduler lifts
             // ALL CPL chunks' LDS x reads to the top of the row compute
-            // (224 live floats -> 256 VGPR + AGPR spill, 1 wave/SIMD)
-            __builtin_amdgcn_sched_barrier(0);
+            // (224 live floats -> 256 VGPR + AGPR spill, 1 wave/SIMD).
+            // Mask 0xF lets ALU/VALU/SALU/MFMA cross (FMA chains of
+            // adjacent chunks interleave for ILP) while pinning memory
+            // ops — a full sched_barrier(0) costs ~40% in the guide's
+            // GEMM ladder by defeating the scheduler outright.
+            __builtin_amdgcn_sched_barrier(0x000F);
             if constexpr (W == DT::DQ4K) {
                 const uint2 hd = hd4[slot][j];
                 const float d = f16_bits_to_f32(hd.x & 0xFFFF);
