@@ -59,6 +59,9 @@ class Config:
     test_mode: bool = False
     ipc_socket: str | None = None
     max_seq: int = 4096
+    # every peer runs an embedded rendezvous server (reference: libp2p
+    # DHT ModeServer on every peer)
+    peer_dht: bool = True
     intervals: Intervals = field(default_factory=Intervals)
 
     def __post_init__(self):

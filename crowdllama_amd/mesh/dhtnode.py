@@ -33,13 +33,14 @@ class DHTServer:
 
     PROVIDER_TTL = 30.0  # records expire without re-provide (ref: 1 s loop)
 
-    def __init__(self, cfg: Config, peer_id: str = ""):
+    def __init__(self, cfg: Config, peer_id: str = "", identity=None):
         self.cfg = cfg
         # real cryptographic identity from the keyfile (the legacy second
         # argument is kept for compatibility but the advertised id is
         # always derived from the key — ids are now verifiable hashes of
-        # ed25519 public keys)
-        self.identity = load_identity("dht", cfg.key_path)
+        # ed25519 public keys). An embedded per-peer
+        # server passes the peer's own identity instead.
+        self.identity = identity or load_identity("dht", cfg.key_path)
         self.peer_id = self.identity.peer_id
         self.log = new_app_logger("dht", cfg.verbose)
         self.providers: dict[str, dict[str, ProviderRecord]] = {}  # ns -> id -> rec

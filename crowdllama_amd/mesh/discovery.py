@@ -116,6 +116,21 @@ class Discovery:
         # checks the DHT routing table instead, peer.go:513-525)
         self.last_success = 0.0
 
+    MAX_SERVERS = 8
+
+    def add_server(self, addr: str, peer_id: str = "") -> bool:
+        """Add a learned rendezvous server (a peer's embedded DHT).
+        Every peer runs one (reference: libp2p ModeServer on every peer),
+        so after bootstrap the mesh has N rendezvous targets, not one."""
+        if not addr or len(self.clients) >= self.MAX_SERVERS:
+            return False
+        if any(c.addr == addr for c in self.clients):
+            return False
+        self.clients.append(RendezvousClient(addr, self.identity,
+                                             timeout=self.metadata_timeout))
+        self.log.info("learned rendezvous server %s", addr)
+        return True
+
     async def close(self):
         for c in self.clients:
             await c.close()

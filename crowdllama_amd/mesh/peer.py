@@ -53,6 +53,7 @@ class Peer:
                                  worker_mode=worker_mode,
                                  version=f"{__version__}+{commit_hash()}")
         self._server: asyncio.base_events.Server | None = None
+        self._dht = None              # embedded rendezvous server
         self._conn_writers: set = set()
         self._tasks: list[asyncio.Task] = []
         self.port: int | None = None
@@ -64,6 +65,21 @@ class Peer:
         self._server = await asyncio.start_server(
             self._on_conn, self.cfg.listen_host, self.cfg.listen_port)
         self.port = self._server.sockets[0].getsockname()[1]
+        if self.cfg.peer_dht:
+            # every peer runs an embedded rendezvous server (reference:
+            # libp2p DHT ModeServer on every peer, pkg/dht/dht.go:106-112)
+            # and advertises it in its signed record; consumers add it as
+            # a rendezvous target, so the mesh keeps N live servers after
+            # the bootstrap node dies instead of relying only on gossip
+            from .dhtnode import DHTServer
+            self._dht = DHTServer(self.cfg, identity=self.identity)
+            dht_port = await self._dht.start(self.cfg.listen_host, 0)
+            host = "127.0.0.1" if self.cfg.listen_host in ("0.0.0.0", "::")                 else self.cfg.listen_host
+            self.resource.dht_addr = f"{host}:{dht_port}"
+            # self-register: the advertise loop then keeps this peer's
+            # record live on its OWN server, so a consumer that learned
+            # it can resolve providers there after the bootstrap dies
+            self.discovery.add_server(self.resource.dht_addr)
         self.update_metadata()
         await self.peer_manager.start()
         self._tasks = [
@@ -85,6 +101,9 @@ class Peer:
         self._tasks.clear()
         await self.peer_manager.stop()
         await self.discovery.close()
+        if self._dht is not None:
+            await self._dht.stop()
+            self._dht = None
         if self._server:
             self._server.close()
         for w in list(self._conn_writers):   # see DHTServer.stop: close

@@ -61,6 +61,10 @@ class PeerManager:
             else:
                 pi.resource = res
                 pi.last_seen = time.time()
+            if res.dht_addr:
+                # the record is signature-verified upstream, so the
+                # embedded-server address is as trustworthy as the rest
+                self.discovery.add_server(res.dht_addr, res.peer_id)
 
     async def remove_peer(self, peer_id: str, tombstone: bool = True) -> None:
         async with self._lock:

@@ -26,6 +26,11 @@ class Resource:
     version: str = ""
     worker_mode: bool = False
     addrs: list[str] = field(default_factory=list)  # "host:port"
+    # this peer's embedded rendezvous server (reference parity: every
+    # libp2p peer runs the DHT in ModeServer, pkg/dht/dht.go:106-112);
+    # consumers add it as a rendezvous target, so losing the bootstrap
+    # node leaves N live servers, not one gossip fallback
+    dht_addr: str = ""
     # record authentication: pubkey is the advertising
     # peer's ed25519 public key (hex); sig signs the canonical payload.
     # Receivers verify sig AND that peer_id == hash(pubkey), so a record
@@ -50,6 +55,7 @@ class Resource:
             "version": self.version,
             "worker_mode": self.worker_mode,
             "addrs": self.addrs,
+            "dht_addr": self.dht_addr,
         }
 
     def _signing_bytes(self) -> bytes:
@@ -93,6 +99,7 @@ class Resource:
         r.version = d.get("version", "")
         r.worker_mode = bool(d.get("worker_mode", False))
         r.addrs = list(d.get("addrs", []))
+        r.dht_addr = d.get("dht_addr", "")
         r.pubkey = d.get("pubkey", "")
         r.sig = d.get("sig", "")
         return r

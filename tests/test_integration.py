@@ -494,3 +494,47 @@ def test_mesh_survives_bootstrap_death(mesh_cfg):
             await consumer.stop()
             await worker.stop()
     asyncio.run(go())
+
+
+def test_embedded_peer_dht_takes_over(mesh_cfg):
+    """Every peer runs an embedded rendezvous server advertised in its
+    signed record (reference parity: every libp2p peer is a DHT server,
+    pkg/dht/dht.go:106-112). A consumer that has learned a worker's
+    server keeps resolving providers through it after the bootstrap node
+    dies — no gossip fallback needed."""
+    async def go():
+        dht = DHTServer(mesh_cfg("dht"), "CLADHT")
+        dht_port = await dht.start("127.0.0.1", 0)
+        boot = [f"127.0.0.1:{dht_port}"]
+        wcfg = mesh_cfg("worker")
+        wcfg.bootstrap_peers = boot
+        worker = Peer(wcfg, worker_mode=True,
+                      engines={"m": MockEngine("m")})
+        await worker.start()
+        assert worker.resource.dht_addr, "embedded server not advertised"
+        ccfg = mesh_cfg("consumer")
+        ccfg.bootstrap_peers = boot
+        consumer = Peer(ccfg, worker_mode=False)
+        await consumer.start()
+        try:
+            # consumer discovers the worker and learns its embedded server
+            await _poll(lambda: any(
+                c.addr == worker.resource.dht_addr
+                for c in consumer.discovery.clients),
+                desc="embedded server learned")
+            await dht.stop()  # bootstrap dies
+            consumer.discovery.fallback_addrs = None  # NO gossip crutch
+            # the worker keeps advertising to its own server; a fresh
+            # resolution round through the learned server still finds it
+            deadline = time.time() + 20
+            found = False
+            while time.time() < deadline and not found:
+                provs = await consumer.discovery.find_providers()
+                found = any(p["peer_id"] == worker.peer_id for p in provs)
+                if not found:
+                    await asyncio.sleep(0.3)
+            assert found, "worker not resolvable via embedded server"
+        finally:
+            await consumer.stop()
+            await worker.stop()
+    asyncio.run(go())
