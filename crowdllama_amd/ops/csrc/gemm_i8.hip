@@ -116,7 +116,19 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     constexpr int FM = (BM_ / WMW) / 16;     // a fragments per wave (1 or 4)
     constexpr bool XSHARED = BM_ > 32;       // one X image vs per-wave
     constexpr int RAWB = (W == DT::DQ4K) ? BK / 2 : BK;  // raw bytes/row
-    constexpr int NGW = (BN / 4) * RAWB / 1024;  // W glds per wave (1 or 2)
+    // BM<=32: W staging is wave-PRIVATE — each wave DMAs exactly the
+    // (BN/WNW) fragment columns it consumes, so NO cross-wave LDS
+    // dependency exists anywhere and the k-loop runs with counted
+    // per-wave vmcnt and NO BARRIER AT ALL (the s_barrier arrival skew
+    // was the dominant parked share: 52.6% of decode GEMM wave cycles).
+    // BM=16 (WMW=1) duplicates nothing (4 waves own 4 disjoint column
+    // quarters); BM=32 (WMW=2) stages each column half twice (+WSZ LDS
+    // per buffer, L2 absorbs the 2x reads). BM>=64 keeps the shared
+    // image + barrier.
+    constexpr bool PRIVW = !XSHARED;
+    constexpr int WPRIV = (BN / WNW) * RAWB;     // private W bytes/wave
+    constexpr int NGW = PRIVW ? WPRIV / 1024
+                              : (BN / 4) * RAWB / 1024;  // W glds/wave
     constexpr int NGX = XSHARED ? BM_ * BK / 4096
                                 : BM_ * BK / 1024;   // X glds per wave
     // hdr2 glds/wave: Q4K JF (f32 pairs per 32-block), Q8 JF/2 (f32 per
@@ -131,7 +143,7 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     //   [WSZ, +4*XSZ)                 per-wave X int8 copies, [BM_][64]
     //   [.., +4*1024) (Q4K) / +4*512 (Q8)  per-wave hdr copies
     //   [.., +4*256)                  per-wave xsc copies [kg2][2][16] f32
-    constexpr int WSZ = BN * RAWB;
+    constexpr int WSZ = PRIVW ? 4 * WPRIV : BN * RAWB;
     constexpr int XSZ = BM_ * BK;
     constexpr int XTOT = XSHARED ? XSZ : 4 * XSZ;
     constexpr int HSZ = (W == DT::DQ8) ? JF * 128 : JF * 256;
@@ -170,7 +182,10 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     // 32 of every 128 B here: the round-2 bandwidth wall).
     const int64_t ktiles = K / BK;
     const uint8_t* wtile0 = qs2 + (int64_t)bn * ktiles * (BN * RAWB);
-    const int wlocal = wid * (NGW * 1024) + lane * 16;  // == LDS offset
+    // cooperative (XSHARED) source/dest; private: slice by wn, copy by wid
+    const int wlocal = PRIVW ? wn * WPRIV + lane * 16
+                             : wid * (NGW * 1024) + lane * 16;
+    const int wldst = PRIVW ? wid * WPRIV : wid * (NGW * 1024);
     // X staging rows. BM<=32: each wave DMAs its own full copy (glds gx
     // covers rows gx*16 + lane>>2). BM=128: ONE shared image; wave wid
     // DMAs rows [wid*32, wid*32+32) (glds gx covers wid*32 + gx*16 + ...).
@@ -208,7 +223,7 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
             for (int g = 0; g < NGW; g++)
                 glds16_nt(wt + wlocal + g * 1024,
                           __builtin_amdgcn_readfirstlane((unsigned)(size_t)
-                              &lds[pb][wid * (NGW * 1024) + g * 1024]));
+                              &lds[pb][wldst + g * 1024]));
         }
         // X tile (copy per wave, or this wave's quarter of the shared one)
         #pragma unroll
@@ -317,12 +332,14 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
         (void)0;
         #pragma unroll
         for (int j = 0; j < JF; j++) {
-            const int r = wn * (BN / WNW) + j * 16 + lrow;
+            const int r = PRIVW ? j * 16 + lrow            // private image
+                                : wn * (BN / WNW) + j * 16 + lrow;
+            const int wbase = PRIVW ? wid * WPRIV : 0;
             rawj[j][0] = *reinterpret_cast<const long*>(
-                &lds[pb][r * RAWB + lk * 8]);
+                &lds[pb][wbase + r * RAWB + lk * 8]);
             if constexpr (W == DT::DQ8 || W == DT::DQ6K)
                 rawj[j][1] = *reinterpret_cast<const long*>(
-                    &lds[pb][r * RAWB + 32 + lk * 8]);
+                    &lds[pb][wbase + r * RAWB + 32 + lk * 8]);
         }
         #pragma unroll
         for (int kb2 = 0; kb2 < 2; kb2++) {
@@ -418,7 +435,8 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     for (int kb = kb_lo; kb < kb_hi; kb += BK) {
         asm volatile("s_waitcnt vmcnt(%0)" ::"i"((NBUF - 2) * NGLT)
                      : "memory");
-        __builtin_amdgcn_s_barrier();
+        if constexpr (!PRIVW)   // private staging: no cross-wave deps
+            __builtin_amdgcn_s_barrier();
         if (kb + (NBUF - 1) * BK <= kb_last) {
             const int nb = pb + NBUF - 1 >= NBUF ? pb - 1 : pb + NBUF - 1;
             issue_tile(kb + (NBUF - 1) * BK, nb);
