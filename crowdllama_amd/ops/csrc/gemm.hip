@@ -625,8 +625,9 @@ __global__ __launch_bounds__(256) void k_attn_prefill(
 
 // split-K workgroup target: >=2 WGs per CU keeps all 8 XCDs fed when M is
 // small; overridable for on-hardware sweeps (CLA_SPLITK_TARGET).
-// BM=16 decode tiles (M<=16) like deeper splits (768 ~ 3 WGs/CU measured
-// +4% at B=16 with the i8 DMA kernel); BM=32 tiles regressed at 768
+// BM=16 decode tiles (M<=16) like deeper splits (1024 once the k-loop
+// went barrier-free: extra WGs no longer cost arrival skew — B=8 2822
+// vs 2778 tok/s, B=16 tied; 1536 regresses); BM=32 tiles regressed deep
 // (B=32: 5549 vs 6260 tok/s) and keep 512.
 static int splitk_target(int M) {
     static int env = [] {
@@ -634,7 +635,7 @@ static int splitk_target(int M) {
         return e ? atoi(e) : 0;
     }();
     if (env > 0) return env;
-    return M <= 16 ? 768 : 512;
+    return M <= 16 ? 1024 : 512;
 }
 
 // Single source of truth for the split-K factor: launch_gemm_ex and
