@@ -149,9 +149,9 @@ __global__ __launch_bounds__(256) void k_gemm_i8(
     constexpr int HSZ = (W == DT::DQ8) ? JF * 128 : JF * 256;
     constexpr int SSZ = FM * 256;
     constexpr int BUFSZ = WSZ + XTOT + 4 * (HSZ + SSZ);
-    // 3-deep DMA ring: 4 was latency-equivalent (PMC: the park is
-    // barrier/arrival, not DMA) and its LDS cost one WG/CU of residency
-    // (44 KB -> 3 WGs/CU; 34 KB -> 4, letting split-K target 1024)
+    // 3-deep DMA ring: depth 4 re-tested after the barrier removal and
+    // still loses (B=32 7028 -> 5607: the extra buffer's LDS costs a
+    // WG/CU of residency, which now matters MORE at split-K 1024)
     constexpr int NBUF = 3;
     __shared__ __attribute__((aligned(16))) int8_t lds[NBUF][BUFSZ];
 
