@@ -699,9 +699,14 @@ void Engine::step(hipStream_t s) {
     };
 
     launch_embed(embed_, cur_ids_, x_, B, s);
-    // B=1 only: at B=2 the split-K MFMA GEMM already beats two GEMV
-    // passes (measured 340 tok/s GEMV vs ~440 GEMM at B=2)
-    const bool gemv_path = B <= 1;
+    // B=1 always; B=2 when the register-x GEMV covers the projections
+    // (quant weights, hidden a single 2048/4096 stripe): its BB=2 form
+    // shares one weight stream between both rows, beating the split-K
+    // GEMM that previously won at B=2 (~440 tok/s agg)
+    const bool gemv_path =
+        B <= 1 ||
+        (B == 2 && is_quant_dt(layers_[0].qkv.parts[0].w.dtype) &&
+         (meta_.hidden == 2048 || meta_.hidden == 4096));
     const QBufs qbufs{xq_, xsc_};
     const QBufs* qb = cfg_.act_q8 ? &qbufs : nullptr;
     int li = 0;

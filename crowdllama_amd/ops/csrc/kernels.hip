@@ -619,14 +619,15 @@ __global__ __launch_bounds__(256) void k_fill_res(
     if (i < n) y[i] = res ? res[i] : 0.f;
 }
 
-template <DT W, int P, int SEGF>
+template <DT W, int P, int SEGF, int BB = 1>
 __global__ __launch_bounds__(256) void k_gemv_r(
     const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
     const float* __restrict__ xin, const float* __restrict__ gw,
     const float* __restrict__ res, float* __restrict__ y,
-    int N, int K, int kbeg0, int accum, float eps) {
+    int N, int K, int kbeg0, int accum, int ldy, float eps) {
     static_assert(W == DT::DQ4K || W == DT::DQ6K || W == DT::DQ8,
                   "register-x GEMV covers quantized weights only");
+    static_assert(BB == 1 || BB == 2, "register-x GEMV: 1 or 2 rows");
     constexpr int CPL = (W == DT::DQ4K) ? SEGF / 32 : SEGF / 16;
     // pipeline depth: NB-1 rows of weight loads in flight per lane. The
     // binding constraint at 8 waves/CU is outstanding bytes (~2 us
@@ -635,8 +636,11 @@ __global__ __launch_bounds__(256) void k_gemv_r(
     // PRE_RMS, where the extra live state tipped occupancy and measured
     // slower (qkv 8.4 -> 10.7 us) — and 3 for DQ8/DQ6K (19 VGPRs/slot;
     // the Q6_K head hit 6.0 TB/s = 93% of the streaming ceiling there).
+    // BB=2 (both decode rows share one weight stream) doubles the x
+    // registers, so back the depth off a notch.
     constexpr int NB =
-        (W == DT::DQ4K) ? (P == PRE_RMS ? 3 : 5) : 3;
+        BB == 2 ? ((W == DT::DQ4K) ? 3 : 2)
+                : ((W == DT::DQ4K) ? (P == PRE_RMS ? 3 : 5) : 3);
     static_assert(CPL >= 1 && CPL <= 4, "stripe must be 2048 or 4096");
 
     const int lane = threadIdx.x & 63;
@@ -661,85 +665,107 @@ __global__ __launch_bounds__(256) void k_gemv_r(
     // lane-LINEARLY (perfectly coalesced), applies SILU / gw and the
     // block-wide RMS reduce in that linear pass, bounces through LDS, and
     // each lane then gathers its mapped segment with ds_reads.
-    __shared__ float4 xs4[KL / 4];
-    __shared__ float redw[8];
+    __shared__ float4 xs4[BB * KL / 4];
+    __shared__ float redw[BB * 4];
     {
         const int tid = threadIdx.x;
-        float ssp = 0.f;
         #pragma unroll
-        for (int i = 0; i < KL / 4 / 256; i++) {
-            const int k4 = i * 256 + tid;          // float4 index in stripe
-            const int k = kbeg + k4 * 4;
-            float4 v;
-            if constexpr (P == PRE_SILU) {
-                const float4 g = *reinterpret_cast<const float4*>(xin + k);
-                const float4 u = *reinterpret_cast<const float4*>(xin + K + k);
-                v.x = (g.x / (1.f + __expf(-g.x))) * u.x;
-                v.y = (g.y / (1.f + __expf(-g.y))) * u.y;
-                v.z = (g.z / (1.f + __expf(-g.z))) * u.z;
-                v.w = (g.w / (1.f + __expf(-g.w))) * u.w;
-            } else {
-                v = *reinterpret_cast<const float4*>(xin + k);
+        for (int b = 0; b < BB; b++) {
+            const float* xrow = xin + (int64_t)b * (P == PRE_SILU ? 2 : 1) * K;
+            float ssp = 0.f;
+            #pragma unroll
+            for (int i = 0; i < KL / 4 / 256; i++) {
+                const int k4 = i * 256 + tid;      // float4 index in stripe
+                const int k = kbeg + k4 * 4;
+                float4 v;
+                if constexpr (P == PRE_SILU) {
+                    const float4 g = *reinterpret_cast<const float4*>(xrow + k);
+                    const float4 u =
+                        *reinterpret_cast<const float4*>(xrow + K + k);
+                    v.x = (g.x / (1.f + __expf(-g.x))) * u.x;
+                    v.y = (g.y / (1.f + __expf(-g.y))) * u.y;
+                    v.z = (g.z / (1.f + __expf(-g.z))) * u.z;
+                    v.w = (g.w / (1.f + __expf(-g.w))) * u.w;
+                } else {
+                    v = *reinterpret_cast<const float4*>(xrow + k);
+                }
+                if constexpr (P == PRE_RMS) {
+                    ssp += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+                    const float4 g = *reinterpret_cast<const float4*>(gw + k);
+                    v.x *= g.x; v.y *= g.y; v.z *= g.z; v.w *= g.w;
+                }
+                xs4[b * (KL / 4) + k4] = v;
             }
             if constexpr (P == PRE_RMS) {
-                ssp += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
-                const float4 g = *reinterpret_cast<const float4*>(gw + k);
-                v.x *= g.x; v.y *= g.y; v.z *= g.z; v.w *= g.w;
+                ssp = wave_reduce_sum(ssp);
+                if ((tid & 63) == 0) redw[b * 4 + (tid >> 6)] = ssp;
             }
-            xs4[k4] = v;
-        }
-        if constexpr (P == PRE_RMS) {
-            ssp = wave_reduce_sum(ssp);
-            if ((tid & 63) == 0) redw[tid >> 6] = ssp;
         }
         __syncthreads();
     }
-    float inv = 1.f;
-    if constexpr (P == PRE_RMS)   // single-stripe only: K == KL
-        inv = rsqrtf((redw[0] + redw[1] + redw[2] + redw[3]) / (float)K + eps);
+    float inv[BB];
+    #pragma unroll
+    for (int b = 0; b < BB; b++) {
+        inv[b] = 1.f;
+        if constexpr (P == PRE_RMS)   // single-stripe only: K == KL
+            inv[b] = rsqrtf((redw[b * 4] + redw[b * 4 + 1] + redw[b * 4 + 2] +
+                             redw[b * 4 + 3]) / (float)K + eps);
+    }
 
     // gather this lane's mapped segment out of LDS (xr holds x*gw*inv)
-    float4 xr[SEGF / 4];
+    float4 xr[BB * SEGF / 4];
     #pragma unroll
-    for (int j = 0; j < CPL; j++) {
-        const int c = lane * CPL + j;
-        if constexpr (W == DT::DQ4K) {
-            const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
-            const int k0 = sb * 256 + q * 64 + h * 16;   // stripe-local
-            #pragma unroll
-            for (int t = 0; t < 4; t++) {
-                xr[j * 8 + t] = xs4[(k0 >> 2) + t];
-                xr[j * 8 + 4 + t] = xs4[((k0 + 32) >> 2) + t];
-            }
-        } else {
-            const int k0 = c * 16;
-            #pragma unroll
-            for (int t = 0; t < 4; t++) xr[j * 4 + t] = xs4[(k0 >> 2) + t];
-        }
-    }
-    if constexpr (P == PRE_RMS) {
+    for (int b = 0; b < BB; b++) {
+        constexpr int XB = SEGF / 4;
         #pragma unroll
-        for (int t = 0; t < SEGF / 4; t++) {
-            xr[t].x *= inv; xr[t].y *= inv; xr[t].z *= inv; xr[t].w *= inv;
+        for (int j = 0; j < CPL; j++) {
+            const int c = lane * CPL + j;
+            const int sb4 = b * (KL / 4);
+            if constexpr (W == DT::DQ4K) {
+                const int sb = c >> 3, p = c & 7, q = p >> 1, h = p & 1;
+                const int k0 = sb * 256 + q * 64 + h * 16;   // stripe-local
+                #pragma unroll
+                for (int t = 0; t < 4; t++) {
+                    xr[b * XB + j * 8 + t] = xs4[sb4 + (k0 >> 2) + t];
+                    xr[b * XB + j * 8 + 4 + t] =
+                        xs4[sb4 + ((k0 + 32) >> 2) + t];
+                }
+            } else {
+                const int k0 = c * 16;
+                #pragma unroll
+                for (int t = 0; t < 4; t++)
+                    xr[b * XB + j * 4 + t] = xs4[sb4 + (k0 >> 2) + t];
+            }
+        }
+        if constexpr (P == PRE_RMS) {
+            #pragma unroll
+            for (int t = 0; t < XB; t++) {
+                xr[b * XB + t].x *= inv[b]; xr[b * XB + t].y *= inv[b];
+                xr[b * XB + t].z *= inv[b]; xr[b * XB + t].w *= inv[b];
+            }
         }
     }
     // Q4_K: per-16-run x sums for the d*sc*sum(qx) - dmin*mn*sum(x) regroup
-    float sxl[CPL], sxh[CPL];
-    if constexpr (W == DT::DQ4K) {
+    float sxl[BB * CPL], sxh[BB * CPL];
+    #pragma unroll
+    for (int b = 0; b < BB; b++) {
+        constexpr int XB = SEGF / 4;
         #pragma unroll
         for (int j = 0; j < CPL; j++) {
-            float a = 0.f, b = 0.f;
-            #pragma unroll
-            for (int t = 0; t < 4; t++) {
-                const float4 l = xr[j * 8 + t], h = xr[j * 8 + 4 + t];
-                a += l.x + l.y + l.z + l.w;
-                b += h.x + h.y + h.z + h.w;
+            if constexpr (W == DT::DQ4K) {
+                float a = 0.f, bs = 0.f;
+                #pragma unroll
+                for (int t = 0; t < 4; t++) {
+                    const float4 l = xr[b * XB + j * 8 + t];
+                    const float4 h = xr[b * XB + j * 8 + 4 + t];
+                    a += l.x + l.y + l.z + l.w;
+                    bs += h.x + h.y + h.z + h.w;
+                }
+                sxl[b * CPL + j] = a; sxh[b * CPL + j] = bs;
+            } else {
+                sxl[b * CPL + j] = 0.f; sxh[b * CPL + j] = 0.f;
             }
-            sxl[j] = a; sxh[j] = b;
         }
-    } else {
-        #pragma unroll
-        for (int j = 0; j < CPL; j++) { sxl[j] = 0.f; sxh[j] = 0.f; }
     }
 
     // rows this wave owns: the CONTIGUOUS block [wid*rpw, wid*rpw+rows_my).
@@ -755,7 +781,7 @@ __global__ __launch_bounds__(256) void k_gemv_r(
     uint2 hd4[NB];      // DQ4K pair header
     uint32_t hs[NB];    // DQ8 packed f16 d's / DQ6K packed sc bytes
     uint32_t hdd[NB];   // DQ6K f16 d bits
-    uint32_t rv[NB];    // res[r] bits, prefetched with the row's weights:
+    uint32_t rv[NB * BB];  // res[r] bits, prefetched with the row's weights:
                         // a res load at the store join costs a vmcnt(0)
                         // drain of the whole pipeline EVERY row (seen in
                         // disassembly) even when res == nullptr. When res
@@ -772,7 +798,9 @@ __global__ __launch_bounds__(256) void k_gemv_r(
         const int r = r0 + ic;
         const uint8_t* qrow = qsb + (int64_t)r * qs_rb;
         const uint8_t* hrow = hdb + (int64_t)r * hdr_rb;
-        rv[slot] = __float_as_uint(resl[r]);
+        #pragma unroll
+        for (int b = 0; b < BB; b++)
+            rv[slot * BB + b] = __float_as_uint(resl[(int64_t)b * ldy + r]);
         #pragma unroll
         for (int j = 0; j < CPL; j++)
             qv[slot][j] = *(
@@ -810,7 +838,10 @@ __global__ __launch_bounds__(256) void k_gemv_r(
              + (float)(int)(int8_t)(dw >> 24) * x.w;
     };
     auto compute = [&](int slot, int i) {
-        float acc = 0.f;
+        constexpr int XB = SEGF / 4;
+        float acc[BB];
+        #pragma unroll
+        for (int b = 0; b < BB; b++) acc[b] = 0.f;
         if constexpr (W == DT::DQ4K) {
             const uint2 hd = hd4[slot];
             const float d = f16_bits_to_f32(hd.x & 0xFFFF);
@@ -821,44 +852,59 @@ __global__ __launch_bounds__(256) void k_gemv_r(
             const float mh = dmin * (float)(hd.y >> 24);
             #pragma unroll
             for (int j = 0; j < CPL; j++) {
-                float ql = 0.f, qh = 0.f;
                 #pragma unroll
-                for (int t = 0; t < 4; t++) {
-                    const uint32_t dw = qv[slot][j][t];
-                    ql += dot16(dw & 0x0F0F0F0Fu, xr[j * 8 + t]);
-                    qh += dot16((dw >> 4) & 0x0F0F0F0Fu, xr[j * 8 + 4 + t]);
+                for (int b = 0; b < BB; b++) {
+                    float ql = 0.f, qh = 0.f;
+                    #pragma unroll
+                    for (int t = 0; t < 4; t++) {
+                        const uint32_t dw = qv[slot][j][t];
+                        ql += dot16(dw & 0x0F0F0F0Fu, xr[b * XB + j * 8 + t]);
+                        qh += dot16((dw >> 4) & 0x0F0F0F0Fu,
+                                    xr[b * XB + j * 8 + 4 + t]);
+                    }
+                    acc[b] += dl * ql - ml * sxl[b * CPL + j]
+                            + dh * qh - mh * sxh[b * CPL + j];
                 }
-                acc += dl * ql - ml * sxl[j] + dh * qh - mh * sxh[j];
             }
         } else if constexpr (W == DT::DQ8) {
             #pragma unroll
             for (int j = 0; j < CPL; j++) {
-                float q = 0.f;
-                #pragma unroll
-                for (int t = 0; t < 4; t++)
-                    q += sdot16(qv[slot][j][t], xr[j * 4 + t]);
                 const float dj = f16_bits_to_f32(
                     (hs[slot] >> (CPL == 4 ? (j >> 1) * 16 : 0)) & 0xFFFF);
-                acc += dj * q;
+                #pragma unroll
+                for (int b = 0; b < BB; b++) {
+                    float q = 0.f;
+                    #pragma unroll
+                    for (int t = 0; t < 4; t++)
+                        q += sdot16(qv[slot][j][t], xr[b * XB + j * 4 + t]);
+                    acc[b] += dj * q;
+                }
             }
         } else {   // DQ6K
             const float d = f16_bits_to_f32(hdd[slot]);
             #pragma unroll
             for (int j = 0; j < CPL; j++) {
-                float q = 0.f;
-                #pragma unroll
-                for (int t = 0; t < 4; t++)
-                    q += sdot16(qv[slot][j][t], xr[j * 4 + t]);
                 const float dj =
                     d * (float)(int)(int8_t)((hs[slot] >> (8 * j)) & 0xFF);
-                acc += dj * q;
+                #pragma unroll
+                for (int b = 0; b < BB; b++) {
+                    float q = 0.f;
+                    #pragma unroll
+                    for (int t = 0; t < 4; t++)
+                        q += sdot16(qv[slot][j][t], xr[b * XB + j * 4 + t]);
+                    acc[b] += dj * q;
+                }
             }
         }
-        const float v = wave_reduce_sum(acc);
-        if (i < rows_my && lane == 0) {
-            const int r = r0 + i;
-            if (accum) atomicAdd(y + r, v);
-            else y[r] = v + __uint_as_float(rv[slot] & rmask);
+        #pragma unroll
+        for (int b = 0; b < BB; b++) {
+            const float v = wave_reduce_sum(acc[b]);
+            if (i < rows_my && lane == 0) {
+                const int r = r0 + i;
+                if (accum) atomicAdd(y + (int64_t)b * ldy + r, v);
+                else y[(int64_t)b * ldy + r] =
+                    v + __uint_as_float(rv[slot * BB + b] & rmask);
+            }
         }
     };
 
@@ -1779,7 +1825,7 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
     if (B > 2) throw std::runtime_error("GEMV path supports B<=2");
     const bool quant_w = (w.dtype == DT::DQ4K || w.dtype == DT::DQ6K ||
                           w.dtype == DT::DQ8);
-    if (B == 1 && quant_w && gemv_r_mode() &&
+    if (B <= 2 && quant_w && gemv_r_mode() &&
         (K == 2048 || K == 4096)) {
         // Single-stripe shapes only: for K > 4096 (the down projection)
         // a gridDim.z-striped atomic variant measured ~33 us vs the
@@ -1787,6 +1833,7 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
         // cheap at B=1), so multi-stripe stays unselected.
         // nwg: ~8 waves/CU saturates the stream for big N; smaller N
         // can't afford the per-wave ramp at 512 WGs (rows/wave < 4).
+        // B=2 shares one weight stream between both rows (BB=2).
         int nwg = (N >= 16384) ? 512 : 256;
         nwg = std::min(nwg, (N + 3) / 4);
         static int wgs_env = [] {
@@ -1794,19 +1841,27 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
             return e ? atoi(e) : 0;
         }();
         if (wgs_env > 0) nwg = std::min(wgs_env, (N + 3) / 4);
-        #define GEMVR_SEG(WT, SEGF)                                            \
+        #define GEMVR_SEG(WT, SEGF, BBV)                                       \
             do {                                                               \
-                auto kern = (pre == PRE_RMS) ? k_gemv_r<WT, PRE_RMS, SEGF>     \
-                           : (pre == PRE_SILU) ? k_gemv_r<WT, PRE_SILU, SEGF>  \
-                           : k_gemv_r<WT, PRE_NONE, SEGF>;                     \
+                auto kern = (pre == PRE_RMS)                                   \
+                        ? k_gemv_r<WT, PRE_RMS, SEGF, BBV>                     \
+                        : (pre == PRE_SILU)                                    \
+                        ? k_gemv_r<WT, PRE_SILU, SEGF, BBV>                    \
+                        : k_gemv_r<WT, PRE_NONE, SEGF, BBV>;                   \
                 hipLaunchKernelGGL(kern, dim3(nwg, 1, 1), dim3(256), 0,        \
                     stream, (const uint8_t*)w.qs, (const uint8_t*)w.hdr,       \
-                    xin, gw, res, y, N, K, 0, 0, eps);                         \
+                    xin, gw, res, y, N, K, 0, 0, ldy, eps);                    \
             } while (0)
         #define GEMVR_CASE(WT)                                                 \
             do {                                                               \
-                if (K == 4096) GEMVR_SEG(WT, 64);                              \
-                else GEMVR_SEG(WT, 32);                                        \
+                if (B == 2) {                                                  \
+                    if (K == 4096) GEMVR_SEG(WT, 64, 2);                       \
+                    else GEMVR_SEG(WT, 32, 2);                                 \
+                } else if (K == 4096) {                                        \
+                    GEMVR_SEG(WT, 64, 1);                                      \
+                } else {                                                       \
+                    GEMVR_SEG(WT, 32, 1);                                      \
+                }                                                              \
             } while (0)
         switch (w.dtype) {
             case DT::DQ4K: GEMVR_CASE(DT::DQ4K); return;

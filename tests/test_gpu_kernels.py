@@ -157,19 +157,20 @@ def test_gemv_pre_silu(core):
 
 
 @pytest.mark.parametrize("name", list(CASES))
+@pytest.mark.parametrize("B", [1, 2])
 @pytest.mark.parametrize("K,pre", [
     (2048, PRE_NONE), (2048, PRE_RMS), (2048, PRE_SILU),
     (4096, PRE_NONE), (4096, PRE_RMS), (4096, PRE_SILU),
-    (14336, PRE_NONE), (14336, PRE_SILU),   # striped: fill + atomicAdd
-    (6144, PRE_SILU),                       # 4096 stripe + 2048 tail
+    (14336, PRE_NONE), (14336, PRE_SILU),   # B=1: legacy long-K path
+    (6144, PRE_SILU),                       # ineligible: legacy fallback
 ])
-def test_gemv_r_path(core, name, K, pre):
-    """Register-x B=1 GEMV (k_gemv_r): every quant dtype on the stripe
-    shapes the engine actually launches (K=2048 CPL-half, K=4096 full,
-    K>4096 multi-stripe atomic accumulation). K>=2048 at B=1 selects the
-    new kernel inside launch_gemv."""
+def test_gemv_r_path(core, name, B, K, pre):
+    """Register-x GEMV (k_gemv_r): every quant dtype on the stripe
+    shapes the engine actually launches (K=2048 CPL-half, K=4096 full),
+    at B=1 and the shared-weight-stream BB=2 form. K>4096 covers the
+    legacy/rl fallbacks at both batch sizes."""
     dt, quant, dequant, repack_fn = CASES[name]
-    rng = np.random.default_rng(K * 7 + pre)
+    rng = np.random.default_rng(K * 7 + pre + B)
     N = 192
     w = rng.standard_normal((N, K)).astype(np.float32) * 0.1
     raw = quant(w)
@@ -177,11 +178,11 @@ def test_gemv_r_path(core, name, K, pre):
     qs, hdr = repack_fn(raw.reshape(N, -1), N, K)
     gw = rng.standard_normal(K).astype(np.float32)
     if pre == PRE_SILU:
-        x = rng.standard_normal((1, 2 * K)).astype(np.float32)
+        x = rng.standard_normal((B, 2 * K)).astype(np.float32)
         g, u = x[:, :K], x[:, K:]
         xe = (g / (1 + np.exp(-g))) * u
     else:
-        x = rng.standard_normal((1, K)).astype(np.float32)
+        x = rng.standard_normal((B, K)).astype(np.float32)
         xe = x
         if pre == PRE_RMS:
             xe = x / np.sqrt((x * x).mean(axis=1, keepdims=True) + 1e-5) * gw
