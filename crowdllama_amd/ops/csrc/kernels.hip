@@ -606,25 +606,20 @@ __global__ __launch_bounds__(256) void k_gemv_q8(
 //    d*sc*sum(q*x) - dmin*mn*sum(x) with sum(x) per 16-run precomputed
 //    once per lane — ~2.3 VALU/weight, under the HBM-bound time.
 //  - Selected for single-stripe shapes (K = 2048 or 4096) only. A
-//    gridDim.z-striped atomicAdd variant for K > 4096 (kept: accum/kbeg
-//    args + k_fill_res) measured SLOWER than the legacy LDS kernel on
-//    the down projection (~33 vs 23 us: x re-staging there is L2-served
-//    and the stripe launches don't amortize), so K > 4096 falls back.
+//    gridDim.z-striped atomicAdd variant for K > 4096 was built and
+//    measured SLOWER than the legacy LDS kernel on the down projection
+//    (~33 vs 23 us: x re-staging there is L2-served and the stripe
+//    launches don't amortize), so it was removed; K > 4096 uses
+//    k_gemv_rl (K = 14336) or the legacy kernel.
 // Reference parity: same mat-vec the reference delegates to llama.cpp
 // (SURVEY.md §2.3); numerics = plain f32 dot of dequantized weights.
-
-__global__ __launch_bounds__(256) void k_fill_res(
-    const float* __restrict__ res, float* __restrict__ y, int n) {
-    const int i = (int)blockIdx.x * 256 + threadIdx.x;
-    if (i < n) y[i] = res ? res[i] : 0.f;
-}
 
 template <DT W, int P, int SEGF, int BB = 1>
 __global__ __launch_bounds__(256) void k_gemv_r(
     const uint8_t* __restrict__ qs, const uint8_t* __restrict__ hdr,
     const float* __restrict__ xin, const float* __restrict__ gw,
     const float* __restrict__ res, float* __restrict__ y,
-    int N, int K, int kbeg0, int accum, int ldy, float eps) {
+    int N, int K, int kbeg0, int ldy, float eps) {
     static_assert(W == DT::DQ4K || W == DT::DQ6K || W == DT::DQ8,
                   "register-x GEMV covers quantized weights only");
     static_assert(BB == 1 || BB == 2, "register-x GEMV: 1 or 2 rows");
@@ -901,8 +896,7 @@ __global__ __launch_bounds__(256) void k_gemv_r(
             const float v = wave_reduce_sum(acc[b]);
             if (i < rows_my && lane == 0) {
                 const int r = r0 + i;
-                if (accum) atomicAdd(y + (int64_t)b * ldy + r, v);
-                else y[(int64_t)b * ldy + r] =
+                y[(int64_t)b * ldy + r] =
                     v + __uint_as_float(rv[slot * BB + b] & rmask);
             }
         }
@@ -1850,7 +1844,7 @@ void launch_gemv(const WTensor& w, int pre, const float* xin, const float* gw,
                         : k_gemv_r<WT, PRE_NONE, SEGF, BBV>;                   \
                 hipLaunchKernelGGL(kern, dim3(nwg, 1, 1), dim3(256), 0,        \
                     stream, (const uint8_t*)w.qs, (const uint8_t*)w.hdr,       \
-                    xin, gw, res, y, N, K, 0, 0, ldy, eps);                    \
+                    xin, gw, res, y, N, K, 0, ldy, eps);                       \
             } while (0)
         #define GEMVR_CASE(WT)                                                 \
             do {                                                               \
