@@ -265,3 +265,19 @@ def test_nat_reachability_classification(cfg):
             await cli.close()
             await srv.stop()
     run(go())
+
+
+def test_discovery_learned_servers(cfg):
+    """Discovery.add_server: dedupes against bootstrap + learned addrs and
+    caps the client list (embedded per-peer rendezvous servers)."""
+    ident = identity_from_seed(b"\x07" * 32)
+    disco = Discovery(["127.0.0.1:9000"], ident)
+    assert disco.add_server("127.0.0.1:9100")          # learned
+    assert not disco.add_server("127.0.0.1:9100")      # duplicate
+    assert not disco.add_server("127.0.0.1:9000")      # bootstrap dup
+    assert not disco.add_server("")                    # empty
+    n0 = len(disco.clients)
+    for i in range(Discovery.MAX_SERVERS + 4):
+        disco.add_server(f"127.0.0.1:{9200 + i}")
+    assert len(disco.clients) == Discovery.MAX_SERVERS
+    assert n0 <= Discovery.MAX_SERVERS
